@@ -1,0 +1,255 @@
+"""Cluster configuration.
+
+Single-TOML configuration tree mirroring the reference's `curvine-config`
+crate (/root/reference/crates/common/curvine-config: `ClusterConf` root with
+`MasterConf`/`WorkerConf`/`ClientConf`/`FuseConf`/`JournalConf`..., defaults
+in etc/curvine-cluster.toml). Data-dir entries use the same
+``"[MEM:30GB]/path"`` syntax (worker_conf.rs:26), extended with the MI355X
+``HBM`` tier: ``"[HBM:200GB]/gpu0"`` (path component is a label; HBM blocks
+live in a hipMalloc arena, not a filesystem).
+"""
+from __future__ import annotations
+
+import dataclasses
+import os
+import re
+from dataclasses import dataclass, field
+from typing import Any, Optional
+
+try:
+    import tomli as _toml
+except ImportError:  # pragma: no cover
+    _toml = None
+
+_UNITS = {
+    "B": 1, "KB": 1 << 10, "MB": 1 << 20, "GB": 1 << 30, "TB": 1 << 40,
+    "K": 1 << 10, "M": 1 << 20, "G": 1 << 30, "T": 1 << 40,
+    "KIB": 1 << 10, "MIB": 1 << 20, "GIB": 1 << 30, "TIB": 1 << 40,
+}
+
+
+def parse_bytes(s: "str | int | float") -> int:
+    """'64MB' → 67108864. Accepts int passthrough."""
+    if isinstance(s, (int, float)):
+        return int(s)
+    m = re.fullmatch(r"\s*([0-9.]+)\s*([A-Za-z]*)\s*", s)
+    if not m:
+        raise ValueError(f"bad byte size: {s!r}")
+    val = float(m.group(1))
+    unit = m.group(2).upper() or "B"
+    if unit not in _UNITS:
+        raise ValueError(f"bad byte unit: {s!r}")
+    return int(val * _UNITS[unit])
+
+
+def fmt_bytes(n: float) -> str:
+    for unit, div in (("TiB", 1 << 40), ("GiB", 1 << 30), ("MiB", 1 << 20), ("KiB", 1 << 10)):
+        if abs(n) >= div:
+            return f"{n / div:.2f} {unit}"
+    return f"{int(n)} B"
+
+
+# storage tiers, hottest first.  HBM is the MI355X addition (288 GB/GPU).
+TIER_HBM = "HBM"
+TIER_MEM = "MEM"
+TIER_SSD = "SSD"
+TIER_HDD = "HDD"
+TIERS = (TIER_HBM, TIER_MEM, TIER_SSD, TIER_HDD)
+TIER_ORDER = {t: i for i, t in enumerate(TIERS)}
+
+_DATA_DIR_RE = re.compile(r"^\[(\w+)(?::([^\]]+))?\](.*)$")
+
+
+@dataclass
+class DataDir:
+    tier: str = TIER_MEM
+    capacity: int = 1 << 30
+    path: str = "/tmp/curvine/data"
+    device_id: int = 0  # GPU ordinal for HBM dirs
+
+    @staticmethod
+    def parse(s: str) -> "DataDir":
+        """Parse "[MEM:30GB]/path" / "[HBM:200GB:0]label" / bare path."""
+        m = _DATA_DIR_RE.match(s)
+        if not m:
+            return DataDir(tier=TIER_SSD, capacity=0, path=s)
+        tier = m.group(1).upper()
+        if tier not in TIERS:
+            raise ValueError(f"unknown storage tier in {s!r}")
+        cap, dev = 1 << 30, 0
+        if m.group(2):
+            parts = m.group(2).split(":")
+            cap = parse_bytes(parts[0])
+            if len(parts) > 1:
+                dev = int(parts[1])
+        return DataDir(tier=tier, capacity=cap, path=m.group(3) or f"/tmp/curvine/{tier.lower()}", device_id=dev)
+
+
+@dataclass
+class MasterConf:
+    hostname: str = "127.0.0.1"
+    rpc_port: int = 8995
+    web_port: int = 9000
+    meta_dir: str = "/tmp/curvine/meta"
+    # worker placement policy: local | round_robin | random | load_based
+    worker_policy: str = "local"
+    min_replication: int = 1
+    max_replication: int = 16
+    block_size: int = 64 << 20
+    heartbeat_check_ms: int = 5_000
+    worker_expire_ms: int = 60_000
+    retry_cache_size: int = 100_000
+    retry_cache_ttl_ms: int = 600_000
+    audit_log: bool = False
+    # capacity watermarks for quota/eviction (fractions of cluster capacity)
+    eviction_high_watermark: float = 0.95
+    eviction_low_watermark: float = 0.85
+    eviction_policy: str = "lru"  # lru | lfu | none
+    ttl_check_ms: int = 5_000
+
+
+@dataclass
+class JournalConf:
+    enable: bool = True
+    journal_dir: str = "/tmp/curvine/journal"
+    # raft peers "id@host:port"; single entry = standalone (no election)
+    peers: list[str] = field(default_factory=list)
+    snapshot_interval_entries: int = 100_000
+    segment_max_bytes: int = 256 << 20
+    flush_batch: int = 256
+    flush_interval_ms: int = 10
+    rpc_port: int = 8996
+    election_timeout_ms: int = 1500
+    heartbeat_interval_ms: int = 300
+
+
+@dataclass
+class WorkerConf:
+    hostname: str = "127.0.0.1"
+    rpc_port: int = 8997
+    web_port: int = 9001
+    data_dirs: list[str] = field(default_factory=lambda: ["[MEM:1GB]/tmp/curvine/mem"])
+    heartbeat_interval_ms: int = 3_000
+    io_slow_us: int = 300_000
+    # device staging: pinned ring buffers for HBM<->host movement
+    staging_buf_bytes: int = 4 << 20
+    staging_buf_count: int = 8
+    replication_concurrency: int = 4
+
+    def parsed_dirs(self) -> list[DataDir]:
+        return [DataDir.parse(s) for s in self.data_dirs]
+
+
+@dataclass
+class ClientConf:
+    master_addrs: list[str] = field(default_factory=lambda: ["127.0.0.1:8995"])
+    short_circuit: bool = True
+    write_chunk_size: int = 1 << 20
+    read_chunk_size: int = 1 << 20
+    read_chunk_num: int = 8
+    read_parallel: int = 1
+    read_slice_size: int = 16 << 20
+    write_buffer_chunks: int = 8
+    replicas: int = 1
+    block_size: int = 64 << 20
+    rpc_timeout_ms: int = 60_000
+    conn_retry: int = 3
+    storage_tier: str = TIER_HBM  # preferred tier for new blocks
+    enable_crc: bool = False
+    # auto-cache UFS files on miss (unified fs)
+    auto_cache: bool = True
+    auto_cache_max_inflight: int = 4
+
+
+@dataclass
+class FuseConf:
+    mnt_path: str = "/tmp/curvine-fuse"
+    mnt_number: int = 1          # parallel fuse channels (clone_fd analog)
+    io_threads: int = 4
+    max_write: int = 1 << 20
+    max_readahead: int = 8 << 20
+    attr_ttl_ms: int = 1_000
+    entry_ttl_ms: int = 1_000
+    negative_ttl_ms: int = 0
+    direct_io: bool = False
+    allow_other: bool = True
+    state_file: str = "/tmp/curvine/fuse.state"
+    kernel_cache: bool = True
+
+
+@dataclass
+class UfsConf:
+    endpoint: str = ""
+    access_key: str = ""
+    secret_key: str = ""
+    region: str = ""
+    extra: dict = field(default_factory=dict)
+
+
+@dataclass
+class JobConf:
+    worker_task_concurrency: int = 4
+    task_chunk_size: int = 8 << 20
+    store: str = "memory"  # memory | sqlite
+    store_path: str = "/tmp/curvine/jobs.db"
+
+
+@dataclass
+class ClusterConf:
+    cluster_id: str = "curvine-amd"
+    master: MasterConf = field(default_factory=MasterConf)
+    journal: JournalConf = field(default_factory=JournalConf)
+    worker: WorkerConf = field(default_factory=WorkerConf)
+    client: ClientConf = field(default_factory=ClientConf)
+    fuse: FuseConf = field(default_factory=FuseConf)
+    ufs: UfsConf = field(default_factory=UfsConf)
+    job: JobConf = field(default_factory=JobConf)
+    testing: bool = False
+
+    # ---------------- loading ----------------
+    @staticmethod
+    def from_dict(d: dict) -> "ClusterConf":
+        conf = ClusterConf()
+        for section, value in d.items():
+            if not hasattr(conf, section):
+                continue
+            cur = getattr(conf, section)
+            if dataclasses.is_dataclass(cur) and isinstance(value, dict):
+                for k, v in value.items():
+                    if hasattr(cur, k):
+                        setattr(cur, k, v)
+            else:
+                setattr(conf, section, value)
+        conf.apply_env()
+        return conf
+
+    @staticmethod
+    def from_file(path: str) -> "ClusterConf":
+        if _toml is None:
+            raise RuntimeError("tomli not available")
+        with open(path, "rb") as f:
+            return ClusterConf.from_dict(_toml.load(f))
+
+    def apply_env(self) -> None:
+        """CURVINE_MASTER_HOSTNAME etc. overrides (cluster_conf.rs analog)."""
+        env = os.environ
+        if "CURVINE_MASTER_HOSTNAME" in env:
+            self.master.hostname = env["CURVINE_MASTER_HOSTNAME"]
+        if "CURVINE_WORKER_HOSTNAME" in env:
+            self.worker.hostname = env["CURVINE_WORKER_HOSTNAME"]
+        if "CURVINE_MASTER_PORT" in env:
+            self.master.rpc_port = int(env["CURVINE_MASTER_PORT"])
+
+    def master_addr(self) -> tuple[str, int]:
+        return (self.master.hostname, self.master.rpc_port)
+
+    def overlay(self, **kv: Any) -> "ClusterConf":
+        """CLI overlay (ClientCliArgs analog): dotted keys, e.g.
+        overlay(**{"client.replicas": 2})."""
+        for key, v in kv.items():
+            obj = self
+            parts = key.split("__") if "__" in key else key.split(".")
+            for p in parts[:-1]:
+                obj = getattr(obj, p)
+            setattr(obj, parts[-1], v)
+        return self
