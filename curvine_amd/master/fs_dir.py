@@ -1,0 +1,504 @@
+"""In-memory inode tree with journal emission on every mutation.
+
+Analog of the reference's `FsDir`
+(/root/reference/curvine-master/src/master/meta/fs_dir.rs:42-46; mkdir
+:130-164, rename :346-368, acquire_new_block :703-731, complete_file
+:737-763) and the `InodeView` file/dir variants (meta/inode/).
+
+Every mutation is implemented as a pure ``_apply_*`` function driven by a
+journal entry; public methods validate, build the entry, apply it and hand
+it to the journal writer — so replay (`apply_entry`) and live execution are
+the same deterministic code path (op_id ordering, fs_dir.rs:116-120).
+"""
+from __future__ import annotations
+
+import logging
+from typing import Iterator, Optional
+
+from curvine_amd import errors as err
+from curvine_amd.conf import TIER_MEM
+from curvine_amd.master.journal import JournalWriter, Op
+from curvine_amd.model import (BlockInfo, FileStatus, FileType, now_ms)
+
+log = logging.getLogger("curvine.fsdir")
+
+ROOT_ID = 1
+
+
+class Inode:
+    __slots__ = ("id", "parent_id", "name", "file_type", "children",
+                 "length", "blocks", "block_size", "replicas", "storage_tier",
+                 "complete", "mtime_ms", "atime_ms", "mode", "uid", "gid",
+                 "ttl_ms", "ttl_action", "symlink_target", "nlink", "xattrs",
+                 "create_ms")
+
+    def __init__(self, id: int, name: str, file_type: int, mode: int = 0o755):
+        self.id = id
+        self.parent_id = 0
+        self.name = name
+        self.file_type = file_type
+        self.children: Optional[dict[str, int]] = {} if file_type == FileType.DIR else None
+        self.length = 0
+        self.blocks: list[list[int]] = []   # [block_id, length]
+        self.block_size = 64 << 20
+        self.replicas = 1
+        self.storage_tier = TIER_MEM
+        self.complete = True
+        t = now_ms()
+        self.create_ms = t
+        self.mtime_ms = t
+        self.atime_ms = t
+        self.mode = mode
+        self.uid = 0
+        self.gid = 0
+        self.ttl_ms = 0
+        self.ttl_action = "none"
+        self.symlink_target = ""
+        self.nlink = 1
+        self.xattrs: dict[str, bytes] = {}
+
+    @property
+    def is_dir(self) -> bool:
+        return self.file_type == FileType.DIR
+
+    def to_state(self) -> dict:
+        return {s: (getattr(self, s) if s != "children" else
+                    (dict(self.children) if self.children is not None else None))
+                for s in self.__slots__}
+
+    @staticmethod
+    def from_state(d: dict) -> "Inode":
+        ino = Inode(d["id"], d["name"], d["file_type"])
+        for s in Inode.__slots__:
+            if s in d:
+                setattr(ino, s, d[s])
+        if ino.file_type == FileType.DIR and ino.children is None:
+            ino.children = {}
+        return ino
+
+
+def norm_path(path: str) -> str:
+    if not path.startswith("/"):
+        raise err.InvalidPath(f"path must be absolute: {path!r}")
+    parts = [p for p in path.split("/") if p and p != "."]
+    for p in parts:
+        if p == "..":
+            raise err.InvalidPath(f"'..' not allowed: {path!r}")
+    return "/" + "/".join(parts)
+
+
+def split_path(path: str) -> tuple[str, str]:
+    path = norm_path(path)
+    if path == "/":
+        return "/", ""
+    parent, _, name = path.rpartition("/")
+    return parent or "/", name
+
+
+class FsDir:
+    def __init__(self, journal: JournalWriter):
+        self.journal = journal
+        self.inodes: dict[int, Inode] = {}
+        self.block_index: dict[int, int] = {}   # block_id -> inode_id
+        self.next_inode_id = ROOT_ID
+        self.next_block_id = 0
+        root = Inode(ROOT_ID, "", FileType.DIR, 0o755)
+        self.inodes[ROOT_ID] = root
+
+    # ---------------- lookup ----------------
+    def resolve(self, path: str) -> Optional[Inode]:
+        path = norm_path(path)
+        node = self.inodes[ROOT_ID]
+        if path == "/":
+            return node
+        for part in path.strip("/").split("/"):
+            if not node.is_dir or node.children is None:
+                return None
+            cid = node.children.get(part)
+            if cid is None:
+                return None
+            node = self.inodes[cid]
+        return node
+
+    def must_resolve(self, path: str) -> Inode:
+        node = self.resolve(path)
+        if node is None:
+            raise err.FileNotFound(path)
+        return node
+
+    def path_of(self, inode_id: int) -> str:
+        parts: list[str] = []
+        node = self.inodes.get(inode_id)
+        while node is not None and node.id != ROOT_ID:
+            parts.append(node.name)
+            node = self.inodes.get(node.parent_id)
+        return "/" + "/".join(reversed(parts))
+
+    def status_of(self, node: Inode, path: str | None = None) -> FileStatus:
+        return FileStatus(
+            inode_id=node.id,
+            path=path if path is not None else self.path_of(node.id),
+            name=node.name, file_type=int(node.file_type),
+            length=node.length, is_complete=node.complete,
+            block_size=node.block_size, replicas=node.replicas,
+            storage_tier=node.storage_tier,
+            mtime_ms=node.mtime_ms, atime_ms=node.atime_ms,
+            mode=node.mode, uid=node.uid, gid=node.gid,
+            ttl_ms=node.ttl_ms, ttl_action=node.ttl_action,
+            symlink_target=node.symlink_target, nlink=node.nlink,
+            xattrs={k: bytes(v) for k, v in node.xattrs.items()})
+
+    def iter_files(self) -> Iterator[Inode]:
+        for node in self.inodes.values():
+            if node.file_type == FileType.FILE:
+                yield node
+
+    # ---------------- mutations ----------------
+    def mkdir(self, path: str, mode: int = 0o755, create_parents: bool = False) -> Inode:
+        path = norm_path(path)
+        if self.resolve(path) is not None:
+            node = self.resolve(path)
+            if node.is_dir:
+                return node
+            raise err.FileAlreadyExists(path)
+        parent_path, name = split_path(path)
+        parent = self.resolve(parent_path)
+        if parent is None:
+            if not create_parents:
+                raise err.FileNotFound(parent_path)
+            parent = self.mkdir(parent_path, mode, True)
+        if not parent.is_dir:
+            raise err.NotDirectory(parent_path)
+        entry = self.journal.log(Op.MKDIR, parent_id=parent.id, name=name,
+                                 inode_id=self.next_inode_id + 1, mode=mode)
+        return self._apply_mkdir(entry)
+
+    def _apply_mkdir(self, e: dict) -> Inode:
+        node = Inode(e["inode_id"], e["name"], FileType.DIR, e.get("mode", 0o755))
+        node.parent_id = e["parent_id"]
+        self.inodes[node.id] = node
+        self.inodes[e["parent_id"]].children[e["name"]] = node.id
+        self.next_inode_id = max(self.next_inode_id, node.id)
+        return node
+
+    def create(self, path: str, block_size: int, replicas: int,
+               storage_tier: str, overwrite: bool = False,
+               mode: int = 0o644, create_parents: bool = True,
+               file_type: int = int(FileType.FILE)) -> tuple[Inode, list[int]]:
+        """Returns (inode, blocks_to_delete_of_overwritten_file)."""
+        path = norm_path(path)
+        existing = self.resolve(path)
+        removed_blocks: list[int] = []
+        if existing is not None:
+            if existing.is_dir:
+                raise err.IsDirectory(path)
+            if not overwrite:
+                raise err.FileAlreadyExists(path)
+        parent_path, name = split_path(path)
+        parent = self.resolve(parent_path)
+        if parent is None:
+            if not create_parents:
+                raise err.FileNotFound(parent_path)
+            parent = self.mkdir(parent_path, 0o755, True)
+        if not parent.is_dir:
+            raise err.NotDirectory(parent_path)
+        entry = self.journal.log(
+            Op.CREATE, parent_id=parent.id, name=name,
+            inode_id=self.next_inode_id + 1, block_size=block_size,
+            replicas=replicas, storage_tier=storage_tier, mode=mode,
+            overwrite=overwrite, file_type=file_type)
+        node, removed_blocks = self._apply_create(entry)
+        return node, removed_blocks
+
+    def _apply_create(self, e: dict) -> tuple[Inode, list[int]]:
+        parent = self.inodes[e["parent_id"]]
+        removed: list[int] = []
+        old_id = parent.children.get(e["name"])
+        if old_id is not None:
+            removed = self._drop_inode(self.inodes[old_id])
+        node = Inode(e["inode_id"], e["name"], e.get("file_type", int(FileType.FILE)),
+                     e.get("mode", 0o644))
+        node.parent_id = parent.id
+        node.block_size = e["block_size"]
+        node.replicas = e["replicas"]
+        node.storage_tier = e["storage_tier"]
+        node.complete = False
+        self.inodes[node.id] = node
+        parent.children[e["name"]] = node.id
+        parent.mtime_ms = now_ms()
+        self.next_inode_id = max(self.next_inode_id, node.id)
+        return node, removed
+
+    def add_block(self, node: Inode, commit_prev_len: int = -1) -> int:
+        """Allocate a new block id for an incomplete file
+        (acquire_new_block analog). commit_prev_len >= 0 finalizes the
+        previous block's length."""
+        if node.complete:
+            raise err.FileInWriting(f"{self.path_of(node.id)} is complete")
+        entry = self.journal.log(Op.ADD_BLOCK, inode_id=node.id,
+                                 block_id=self.next_block_id + 1,
+                                 commit_prev_len=commit_prev_len)
+        return self._apply_add_block(entry)
+
+    def _apply_add_block(self, e: dict) -> int:
+        node = self.inodes[e["inode_id"]]
+        if e.get("commit_prev_len", -1) >= 0 and node.blocks:
+            node.blocks[-1][1] = e["commit_prev_len"]
+        bid = e["block_id"]
+        node.blocks.append([bid, 0])
+        self.block_index[bid] = node.id
+        self.next_block_id = max(self.next_block_id, bid)
+        return bid
+
+    def complete_file(self, node: Inode, length: int,
+                      block_lens: list[int] | None = None) -> None:
+        entry = self.journal.log(Op.COMPLETE_FILE, inode_id=node.id,
+                                 length=length, block_lens=block_lens)
+        self._apply_complete(entry)
+
+    def _apply_complete(self, e: dict) -> None:
+        node = self.inodes[e["inode_id"]]
+        node.length = e["length"]
+        if e.get("block_lens"):
+            for blk, ln in zip(node.blocks, e["block_lens"]):
+                blk[1] = ln
+        else:
+            # derive block lengths from total length/block_size
+            rem = node.length
+            for blk in node.blocks:
+                blk[1] = min(rem, node.block_size)
+                rem -= blk[1]
+        node.complete = True
+        node.mtime_ms = now_ms()
+
+    def delete(self, path: str, recursive: bool = False) -> list[int]:
+        """Returns deleted block ids (caller schedules worker deletes)."""
+        path = norm_path(path)
+        if path == "/":
+            raise err.InvalidPath("cannot delete /")
+        node = self.must_resolve(path)
+        if node.is_dir and node.children and not recursive:
+            raise err.DirNotEmpty(path)
+        entry = self.journal.log(Op.DELETE, inode_id=node.id)
+        return self._apply_delete(entry)
+
+    def _apply_delete(self, e: dict) -> list[int]:
+        node = self.inodes.get(e["inode_id"])
+        if node is None:
+            return []
+        parent = self.inodes.get(node.parent_id)
+        if parent is not None and parent.children is not None:
+            parent.children.pop(node.name, None)
+            parent.mtime_ms = now_ms()
+        return self._drop_inode(node)
+
+    def _drop_inode(self, node: Inode) -> list[int]:
+        removed: list[int] = []
+        stack = [node]
+        while stack:
+            n = stack.pop()
+            if n.is_dir and n.children:
+                stack.extend(self.inodes[c] for c in n.children.values())
+            for bid, _ in n.blocks:
+                self.block_index.pop(bid, None)
+                removed.append(bid)
+            self.inodes.pop(n.id, None)
+        return removed
+
+    def rename(self, src: str, dst: str) -> None:
+        src, dst = norm_path(src), norm_path(dst)
+        if src == "/" or dst == "/":
+            raise err.InvalidPath("cannot rename /")
+        if dst == src:
+            return
+        if dst.startswith(src + "/"):
+            raise err.InvalidPath(f"cannot rename {src} into itself")
+        node = self.must_resolve(src)
+        dst_parent_path, dst_name = split_path(dst)
+        dst_parent = self.must_resolve(dst_parent_path)
+        if not dst_parent.is_dir:
+            raise err.NotDirectory(dst_parent_path)
+        existing = self.resolve(dst)
+        if existing is not None:
+            if existing.is_dir:
+                if existing.children:
+                    raise err.DirNotEmpty(dst)
+            elif node.is_dir:
+                raise err.NotDirectory(dst)
+        entry = self.journal.log(Op.RENAME, inode_id=node.id,
+                                 dst_parent=dst_parent.id, dst_name=dst_name)
+        self._apply_rename(entry)
+
+    def _apply_rename(self, e: dict) -> list[int]:
+        node = self.inodes[e["inode_id"]]
+        dst_parent = self.inodes[e["dst_parent"]]
+        removed: list[int] = []
+        old_id = dst_parent.children.get(e["dst_name"])
+        if old_id is not None and old_id != node.id:
+            removed = self._drop_inode(self.inodes[old_id])
+        src_parent = self.inodes.get(node.parent_id)
+        if src_parent is not None and src_parent.children is not None:
+            src_parent.children.pop(node.name, None)
+            src_parent.mtime_ms = now_ms()
+        node.parent_id = dst_parent.id
+        node.name = e["dst_name"]
+        dst_parent.children[node.name] = node.id
+        dst_parent.mtime_ms = now_ms()
+        return removed
+
+    def set_attr(self, node: Inode, **attrs) -> None:
+        entry = self.journal.log(Op.SET_ATTR, inode_id=node.id, attrs=attrs)
+        self._apply_set_attr(entry)
+
+    def _apply_set_attr(self, e: dict) -> None:
+        node = self.inodes.get(e["inode_id"])
+        if node is None:
+            return
+        for k, v in e["attrs"].items():
+            if k in ("mode", "uid", "gid", "atime_ms", "mtime_ms", "ttl_ms",
+                     "ttl_action", "replicas", "storage_tier"):
+                setattr(node, k, v)
+
+    def set_xattr(self, node: Inode, name: str, value: bytes) -> None:
+        entry = self.journal.log(Op.SET_XATTR, inode_id=node.id,
+                                 name=name, value=value)
+        self._apply_set_xattr(entry)
+
+    def _apply_set_xattr(self, e: dict) -> None:
+        node = self.inodes.get(e["inode_id"])
+        if node is not None:
+            node.xattrs[e["name"]] = e["value"]
+
+    def remove_xattr(self, node: Inode, name: str) -> None:
+        entry = self.journal.log(Op.REMOVE_XATTR, inode_id=node.id, name=name)
+        self._apply_remove_xattr(entry)
+
+    def _apply_remove_xattr(self, e: dict) -> None:
+        node = self.inodes.get(e["inode_id"])
+        if node is not None:
+            node.xattrs.pop(e["name"], None)
+
+    def symlink(self, link_path: str, target: str) -> Inode:
+        link_path = norm_path(link_path)
+        if self.resolve(link_path) is not None:
+            raise err.FileAlreadyExists(link_path)
+        parent_path, name = split_path(link_path)
+        parent = self.must_resolve(parent_path)
+        if not parent.is_dir:
+            raise err.NotDirectory(parent_path)
+        entry = self.journal.log(Op.SYMLINK, parent_id=parent.id, name=name,
+                                 inode_id=self.next_inode_id + 1, target=target)
+        return self._apply_symlink(entry)
+
+    def _apply_symlink(self, e: dict) -> Inode:
+        node = Inode(e["inode_id"], e["name"], int(FileType.SYMLINK), 0o777)
+        node.parent_id = e["parent_id"]
+        node.symlink_target = e["target"]
+        node.complete = True
+        self.inodes[node.id] = node
+        self.inodes[e["parent_id"]].children[e["name"]] = node.id
+        self.next_inode_id = max(self.next_inode_id, node.id)
+        return node
+
+    def link(self, src: str, dst: str) -> Inode:
+        """Hard link: a second dentry to the same inode."""
+        src_node = self.must_resolve(src)
+        if src_node.is_dir:
+            raise err.IsDirectory(src)
+        dst = norm_path(dst)
+        if self.resolve(dst) is not None:
+            raise err.FileAlreadyExists(dst)
+        parent_path, name = split_path(dst)
+        parent = self.must_resolve(parent_path)
+        entry = self.journal.log(Op.LINK, inode_id=src_node.id,
+                                 dst_parent=parent.id, dst_name=name)
+        self._apply_link(entry)
+        return src_node
+
+    def _apply_link(self, e: dict) -> None:
+        node = self.inodes[e["inode_id"]]
+        parent = self.inodes[e["dst_parent"]]
+        parent.children[e["dst_name"]] = node.id
+        node.nlink += 1
+        parent.mtime_ms = now_ms()
+
+    def resize(self, node: Inode, new_length: int) -> list[int]:
+        """Truncate. Returns block ids fully beyond the new length."""
+        entry = self.journal.log(Op.RESIZE, inode_id=node.id, length=new_length)
+        return self._apply_resize(entry)
+
+    def _apply_resize(self, e: dict) -> list[int]:
+        node = self.inodes.get(e["inode_id"])
+        if node is None:
+            return []
+        new_len = e["length"]
+        removed: list[int] = []
+        node.length = new_len
+        keep = []
+        off = 0
+        for bid, blen in node.blocks:
+            if off >= new_len:
+                removed.append(bid)
+                self.block_index.pop(bid, None)
+            else:
+                keep.append([bid, min(blen if blen else node.block_size,
+                                      new_len - off)])
+            off += blen if blen else node.block_size
+        node.blocks = keep
+        node.mtime_ms = now_ms()
+        return removed
+
+    def free(self, node: Inode) -> list[int]:
+        """Drop cached blocks, keep metadata (cv free / ttl 'free')."""
+        entry = self.journal.log(Op.FREE, inode_id=node.id)
+        return self._apply_free(entry)
+
+    def _apply_free(self, e: dict) -> list[int]:
+        node = self.inodes.get(e["inode_id"])
+        if node is None:
+            return []
+        removed = [bid for bid, _ in node.blocks]
+        for bid in removed:
+            self.block_index.pop(bid, None)
+        node.blocks = []
+        # not complete anymore in the cache sense; length metadata kept
+        return removed
+
+    # ---------------- replay & snapshot ----------------
+    APPLY = {
+        Op.MKDIR: "_apply_mkdir", Op.CREATE: "_apply_create",
+        Op.ADD_BLOCK: "_apply_add_block", Op.COMPLETE_FILE: "_apply_complete",
+        Op.DELETE: "_apply_delete", Op.RENAME: "_apply_rename",
+        Op.SET_ATTR: "_apply_set_attr", Op.SYMLINK: "_apply_symlink",
+        Op.LINK: "_apply_link", Op.RESIZE: "_apply_resize",
+        Op.FREE: "_apply_free", Op.SET_XATTR: "_apply_set_xattr",
+        Op.REMOVE_XATTR: "_apply_remove_xattr",
+    }
+
+    def apply_entry(self, e: dict) -> None:
+        fn = self.APPLY.get(e["op"])
+        if fn is None:
+            return
+        getattr(self, fn)(e)
+        self.journal.op_id = max(self.journal.op_id, e["op_id"])
+
+    def to_snapshot(self) -> dict:
+        return {
+            "op_id": self.journal.op_id,
+            "next_inode_id": self.next_inode_id,
+            "next_block_id": self.next_block_id,
+            "inodes": [n.to_state() for n in self.inodes.values()],
+        }
+
+    def load_snapshot(self, state: dict) -> int:
+        self.inodes = {}
+        self.block_index = {}
+        for s in state["inodes"]:
+            node = Inode.from_state(s)
+            self.inodes[node.id] = node
+            for bid, _ in node.blocks:
+                self.block_index[bid] = node.id
+        self.next_inode_id = state["next_inode_id"]
+        self.next_block_id = state["next_block_id"]
+        self.journal.op_id = state["op_id"]
+        return state["op_id"]

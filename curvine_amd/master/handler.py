@@ -1,0 +1,200 @@
+"""Master RPC dispatch.
+
+Analog of the reference's `MasterHandler`
+(/root/reference/curvine-master/src/master/master_handler.rs:1030-1151):
+per-connection dispatch over the RpcCode surface.  Header field names follow
+proto/master.proto & worker.proto.
+"""
+from __future__ import annotations
+
+import logging
+import time
+from typing import Optional
+
+from curvine_amd import errors as err
+from curvine_amd.master.filesystem import MasterFilesystem
+from curvine_amd.model import WorkerInfo
+from curvine_amd.rpc.codes import RpcCode
+from curvine_amd.rpc.message import Message
+
+log = logging.getLogger("curvine.master.handler")
+audit = logging.getLogger("audit")
+
+
+class MasterHandler:
+    def __init__(self, master):
+        self.master = master
+        self.fs: MasterFilesystem = master.fs
+
+    async def handle(self, msg: Message, conn) -> Optional[Message]:
+        code = RpcCode(msg.code) if msg.code in RpcCode._value2member_map_ else RpcCode.Undefined
+        fn = getattr(self, f"op_{code.name.lower()}", None)
+        if fn is None:
+            raise err.Unsupported(f"rpc code {msg.code}")
+        if self.master.raft is not None and not self.master.raft.is_leader \
+                and code not in _READONLY_OK:
+            raise err.NotLeader(f"leader={self.master.raft.leader_addr or ''}")
+        t0 = time.perf_counter()
+        try:
+            reply = fn(msg.header, msg.data)
+            return msg.reply(reply or {})
+        finally:
+            if self.fs.conf.master.audit_log:
+                audit.info("cmd=%s used_us=%d", code.name,
+                           int((time.perf_counter() - t0) * 1e6))
+
+    # ---------------- filesystem ----------------
+    def op_heartbeat(self, h, d):
+        return {"ts": int(time.time() * 1000)}
+
+    def op_mkdir(self, h, d):
+        st = self.fs.mkdir(h["path"], h.get("mode", 0o755),
+                           h.get("create_parents", True))
+        return {"status": st.to_dict()}
+
+    def op_createfile(self, h, d):
+        st = self.fs.create(h["path"], h.get("block_size", 0),
+                            h.get("replicas", 0), h.get("storage_tier", ""),
+                            h.get("overwrite", False), h.get("mode", 0o644))
+        return {"status": st.to_dict()}
+
+    def op_appendfile(self, h, d):
+        return {"file_blocks": self.fs.append(h["path"]).to_dict()}
+
+    def op_openfile(self, h, d):
+        return {"file_blocks": self.fs.open(h["path"]).to_dict()}
+
+    def op_filestatus(self, h, d):
+        return {"status": self.fs.file_status(h["path"]).to_dict()}
+
+    def op_liststatus(self, h, d):
+        return {"statuses": [s.to_dict() for s in self.fs.list_status(h["path"])]}
+
+    def op_exists(self, h, d):
+        return {"exists": self.fs.exists(h["path"])}
+
+    def op_delete(self, h, d):
+        n = self.fs.delete(h["path"], h.get("recursive", False))
+        return {"deleted_blocks": n}
+
+    def op_rename(self, h, d):
+        self.fs.rename(h["src"], h["dst"])
+        return {}
+
+    def op_addblock(self, h, d):
+        lb = self.fs.add_block(h["path"], h.get("commit_prev_len", -1),
+                               h.get("client_host", ""),
+                               h.get("client_worker_id", -1),
+                               h.get("exclude_workers"))
+        return {"block": lb.to_dict()}
+
+    def op_completefile(self, h, d):
+        st = self.fs.complete_file(h["path"], h["length"], h.get("block_lens"))
+        return {"status": st.to_dict()}
+
+    def op_getblocklocations(self, h, d):
+        return {"file_blocks": self.fs.get_block_locations(h["path"]).to_dict()}
+
+    def op_getfilesysteminfo(self, h, d):
+        return self.fs.master_info()
+
+    def op_setattr(self, h, d):
+        st = self.fs.set_attr(h["path"], **{
+            k: h.get(k) for k in ("mode", "uid", "gid", "atime_ms", "mtime_ms",
+                                  "ttl_ms", "ttl_action", "replicas",
+                                  "storage_tier", "xattrs")})
+        return {"status": st.to_dict()}
+
+    def op_symlink(self, h, d):
+        return {"status": self.fs.symlink(h["path"], h["target"]).to_dict()}
+
+    def op_link(self, h, d):
+        return {"status": self.fs.link(h["src"], h["dst"]).to_dict()}
+
+    def op_resizefile(self, h, d):
+        return {"status": self.fs.resize(h["path"], h["length"]).to_dict()}
+
+    def op_free(self, h, d):
+        return {"freed_blocks": self.fs.free(h["path"], h.get("recursive", False))}
+
+    def op_createfilesbatch(self, h, d):
+        out = []
+        for req in h["files"]:
+            st = self.fs.create(req["path"], req.get("block_size", 0),
+                                req.get("replicas", 0),
+                                req.get("storage_tier", ""),
+                                req.get("overwrite", False))
+            out.append(st.to_dict())
+        return {"statuses": out}
+
+    def op_completefilesbatch(self, h, d):
+        for req in h["files"]:
+            self.fs.complete_file(req["path"], req["length"], req.get("block_lens"))
+        return {}
+
+    # ---------------- mounts ----------------
+    def op_mount(self, h, d):
+        mi = self.master.mounts.mount(h["curvine_path"], h["ufs_path"],
+                                      h.get("properties", {}),
+                                      h.get("cache_mode", "cache"),
+                                      h.get("auto_cache", True))
+        return {"mount": mi.to_dict()}
+
+    def op_unmount(self, h, d):
+        self.master.mounts.unmount(h["curvine_path"])
+        return {}
+
+    def op_updatemount(self, h, d):
+        mi = self.master.mounts.update(h["curvine_path"], h.get("properties", {}),
+                                       h.get("cache_mode"), h.get("auto_cache"))
+        return {"mount": mi.to_dict()}
+
+    def op_getmounttable(self, h, d):
+        return {"mounts": [m.to_dict() for m in self.master.mounts.table()]}
+
+    def op_getmountinfo(self, h, d):
+        mi = self.master.mounts.lookup(h["path"])
+        return {"mount": mi.to_dict() if mi else None}
+
+    # ---------------- jobs ----------------
+    def op_submitjob(self, h, d):
+        return self.master.jobs.submit(h)
+
+    def op_getjobstatus(self, h, d):
+        return self.master.jobs.status(h["job_id"])
+
+    def op_canceljob(self, h, d):
+        return self.master.jobs.cancel(h["job_id"])
+
+    def op_reporttask(self, h, d):
+        self.master.jobs.report_task(h)
+        return {}
+
+    # ---------------- worker plane ----------------
+    def op_workerheartbeat(self, h, d):
+        info = WorkerInfo.from_dict(h["worker"])
+        cmds = self.fs.worker_heartbeat(info, h.get("added_blocks", []),
+                                        h.get("removed_blocks", []))
+        return {"commands": cmds}
+
+    def op_workerblockreport(self, h, d):
+        to_delete = self.fs.block_report(h["worker_id"], h.get("blocks", []))
+        return {"delete_blocks": to_delete}
+
+    # ---------------- replication ----------------
+    def op_reportblockreplicationresult(self, h, d):
+        self.master.replication.report_result(h)
+        return {}
+
+    def op_requestreplacementworker(self, h, d):
+        ws = self.fs.workers.choose_workers(
+            1, "load_based", exclude=set(h.get("exclude", [])))
+        return {"worker": ws[0].address.to_dict()}
+
+    def op_reportunderreplicatedblocks(self, h, d):
+        for bid in h.get("block_ids", []):
+            self.master.replication.mark_under_replicated(bid)
+        return {}
+
+    def op_metricsreport(self, h, d):
+        return {}

@@ -1,0 +1,173 @@
+"""Master process composition.
+
+Analog of /root/reference/curvine-master/src/master/master_server.rs
+(:164-276): journal system -> filesystem -> mount/job/replication managers
+-> background actor -> RPC server, in that start order.
+"""
+from __future__ import annotations
+
+import asyncio
+import logging
+from typing import Optional
+
+from curvine_amd.conf import ClusterConf
+from curvine_amd.master.filesystem import MasterFilesystem
+from curvine_amd.master.handler import MasterHandler
+from curvine_amd.master.jobs import JobManager
+from curvine_amd.master.journal import JournalWriter, Op
+from curvine_amd.master.mount_manager import MountManager
+from curvine_amd.master.replication import MasterReplicationManager
+from curvine_amd.rpc.codes import RpcCode
+from curvine_amd.rpc.server import HandlerService, RpcServer
+
+log = logging.getLogger("curvine.master")
+
+# codes a raft follower may serve (read-only surface)
+_READONLY_OK = {
+    RpcCode.Heartbeat, RpcCode.FileStatus, RpcCode.ListStatus,
+    RpcCode.Exists, RpcCode.GetFilesystemInfo, RpcCode.GetMountTable,
+    RpcCode.GetMountInfo, RpcCode.GetJobStatus, RpcCode.MetricsReport,
+}
+import curvine_amd.master.handler as _handler_mod
+_handler_mod._READONLY_OK = _READONLY_OK
+
+
+class MasterService(HandlerService):
+    def __init__(self, master: "Master"):
+        self.master = master
+
+    def get_message_handler(self):
+        return MasterHandler(self.master)
+
+
+class Master:
+    def __init__(self, conf: ClusterConf):
+        self.conf = conf
+        self.journal = JournalWriter(conf.journal.journal_dir,
+                                     conf.journal.segment_max_bytes)
+        self.fs = MasterFilesystem(conf, self.journal)
+        self.mounts = MountManager(self.journal)
+        self.jobs = JobManager(self)
+        self.replication = MasterReplicationManager(self.fs)
+        self.raft = None  # set by journal system when peers configured
+        self.rpc = RpcServer("master", conf.master.hostname,
+                             conf.master.rpc_port, MasterService(self))
+        self._actor_task: Optional[asyncio.Task] = None
+        self._stopped = asyncio.Event()
+        self._mutation_count = 0
+
+    # ---------------- lifecycle ----------------
+    async def start(self) -> "Master":
+        self._restore()
+        await self.rpc.start()
+        self.conf.master.rpc_port = self.rpc.port
+        self._actor_task = asyncio.create_task(self._actor_loop())
+        log.info("master started on %s:%d", self.conf.master.hostname, self.rpc.port)
+        return self
+
+    async def stop(self) -> None:
+        self._stopped.set()
+        if self._actor_task:
+            self._actor_task.cancel()
+            try:
+                await self._actor_task
+            except (asyncio.CancelledError, Exception):  # noqa: BLE001
+                pass
+        await self.rpc.stop()
+        self.journal.close()
+
+    def _restore(self) -> None:
+        """Replay snapshot + journal: fs entries and mount entries share the
+        WAL, so dispatch by entry kind."""
+        def apply(e: dict) -> None:
+            if not self.mounts.apply_entry(e):
+                self.fs.fs_dir.apply_entry(e)
+            self.journal.op_id = max(self.journal.op_id, e["op_id"])
+
+        def load_snap(state: dict) -> int:
+            op = self.fs.fs_dir.load_snapshot(state)
+            self.mounts.load_snapshot(state.get("mounts", []))
+            return op
+
+        self.fs.loader.load(apply, load_snap)
+
+    def checkpoint(self) -> None:
+        state = self.fs.fs_dir.to_snapshot()
+        state["mounts"] = self.mounts.to_snapshot()
+        self.fs.loader.save_snapshot(state)
+        self.journal.purge_through(self.journal.op_id)
+
+    # ---------------- background actor ----------------
+    async def _actor_loop(self) -> None:
+        """ScheduledExecutor analog (master/fs/master_actor.rs:30-149):
+        heartbeat expiry, TTL cleanup, replication scan, quota/eviction,
+        periodic checkpoint."""
+        check_ms = self.conf.master.heartbeat_check_ms
+        tick = 0
+        while not self._stopped.is_set():
+            try:
+                await asyncio.sleep(check_ms / 1000.0)
+            except asyncio.CancelledError:
+                return
+            tick += 1
+            try:
+                lost = self.fs.workers.check_expired()
+                if lost:
+                    for bid in self.fs.handle_lost_workers(lost):
+                        self.replication.mark_under_replicated(bid)
+                self.replication.check_all()
+                self.replication.scan()
+                self._ttl_sweep()
+                self._eviction_sweep()
+                if self.journal.op_id and tick % 60 == 0:
+                    self.checkpoint()
+            except Exception as e:  # noqa: BLE001
+                log.exception("master actor tick failed: %s", e)
+
+    def _ttl_sweep(self) -> None:
+        """TTL subsystem analog (meta/inode/ttl/ttl_manager.rs:35-49)."""
+        from curvine_amd.model import now_ms
+        now = now_ms()
+        expired = []
+        for node in list(self.fs.fs_dir.inodes.values()):
+            if node.ttl_ms > 0 and node.create_ms + node.ttl_ms < now \
+                    and node.ttl_action in ("delete", "free"):
+                expired.append(node)
+        for node in expired:
+            path = self.fs.fs_dir.path_of(node.id)
+            try:
+                if node.ttl_action == "delete":
+                    self.fs.delete(path, recursive=True)
+                else:
+                    self.fs.free(path, recursive=True)
+                log.info("ttl %s %s", node.ttl_action, path)
+            except Exception:  # noqa: BLE001
+                pass
+
+    def _eviction_sweep(self) -> None:
+        """Quota/eviction analog (quota/quota_manager.rs:31-83 +
+        eviction/evictor.rs:36-107): when used capacity crosses the high
+        watermark, free least-recently/least-frequently used complete
+        files until below the low watermark."""
+        policy = self.conf.master.eviction_policy
+        if policy == "none":
+            return
+        cap = self.fs.workers.total_capacity()
+        used = self.fs.workers.total_used()
+        if cap <= 0 or used < cap * self.conf.master.eviction_high_watermark:
+            return
+        target = cap * self.conf.master.eviction_low_watermark
+        files = [n for n in self.fs.fs_dir.iter_files()
+                 if n.complete and n.blocks and n.id not in self.fs.writing]
+        files.sort(key=lambda n: n.atime_ms)
+        freed = 0
+        for node in files:
+            if used - freed <= target:
+                break
+            freed += sum(b[1] for b in node.blocks)
+            path = self.fs.fs_dir.path_of(node.id)
+            try:
+                self.fs.free(path)
+                log.info("evicted (freed) %s", path)
+            except Exception:  # noqa: BLE001
+                pass
