@@ -1,0 +1,646 @@
+// curvine_amd native data plane: HBM block arenas, pinned staging
+// pipelines, and CDNA4 kernel wrappers (kernels.hip).
+//
+// This is the MI355X-era equivalent of the reference's native layer
+// (/root/reference/crates/core/curvine-sys: sendfile/splice/fadvise are the
+// CPU fallback path there; here the hot path is hipMemcpyAsync through a
+// pinned ring + device kernels).  Torch-free on purpose: consumers pass raw
+// device pointers (e.g. torch tensor.data_ptr()) for GPU-to-GPU reads.
+//
+// Arena model: one big hipMalloc per (device, arena) — the HBM tier of the
+// worker block store (BdevLayout/BdevOffsetAllocator analog,
+// /root/reference/crates/adapters/curvine-storage-local/src/layout/
+// bdev_layout.rs:30-111); offset allocation lives in Python
+// (curvine_amd/worker/arena_alloc.py), this layer moves bytes.
+
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include <atomic>
+#include <cstring>
+#include <mutex>
+#include <stdexcept>
+#include <string>
+#include <vector>
+
+#include <hip/hip_runtime.h>
+
+namespace py = pybind11;
+
+#define HIP_CHECK(expr)                                                        \
+  do {                                                                         \
+    hipError_t _e = (expr);                                                    \
+    if (_e != hipSuccess)                                                      \
+      throw std::runtime_error(std::string(#expr) + ": " +                     \
+                               hipGetErrorString(_e));                         \
+  } while (0)
+
+// ---------------------------------------------------------------------------
+// CRC32C (Castagnoli): host tables, SW slice-by-8, GF(2) combine
+// ---------------------------------------------------------------------------
+
+static uint32_t crc_tab[8][256];
+static uint32_t comb_mat[24][32];  // shift by (CRC_SUB << k) bytes
+static const uint32_t CRC_POLY = 0x82F63B78u;  // reflected Castagnoli
+static const int CRC_SUB = 4096;
+static const int WG = 256;
+
+static void gf2_square(uint32_t dst[32], const uint32_t src[32]) {
+  for (int i = 0; i < 32; ++i) {
+    uint32_t v = src[i], out = 0;
+    for (int b = 0; b < 32; ++b)
+      if ((v >> b) & 1) out ^= src[b];
+    dst[i] = out;
+  }
+}
+
+static uint32_t gf2_apply(const uint32_t mat[32], uint32_t crc) {
+  uint32_t out = 0;
+  for (int i = 0; i < 32; ++i)
+    if ((crc >> i) & 1) out ^= mat[i];
+  return out;
+}
+
+static void crc_init_tables() {
+  for (uint32_t i = 0; i < 256; ++i) {
+    uint32_t c = i;
+    for (int k = 0; k < 8; ++k) c = (c >> 1) ^ ((c & 1) ? CRC_POLY : 0);
+    crc_tab[0][i] = c;
+  }
+  for (int t = 1; t < 8; ++t)
+    for (uint32_t i = 0; i < 256; ++i)
+      crc_tab[t][i] = (crc_tab[t - 1][i] >> 8) ^ crc_tab[0][crc_tab[t - 1][i] & 0xFF];
+
+  // M1 = append-one-zero-bit operator; comb_mat[0] = M1^(8*CRC_SUB)
+  uint32_t m[32], tmp[32];
+  m[0] = CRC_POLY;
+  for (int i = 1; i < 32; ++i) m[i] = 1u << (i - 1);
+  // raise to 8*CRC_SUB = 2^15 -> square 15 times
+  int shift_bits_log2 = 0;
+  {
+    uint64_t bits = 8ull * CRC_SUB;  // 32768 = 2^15
+    while ((1ull << shift_bits_log2) < bits) shift_bits_log2++;
+  }
+  for (int s = 0; s < shift_bits_log2; ++s) {
+    gf2_square(tmp, m);
+    std::memcpy(m, tmp, sizeof(m));
+  }
+  std::memcpy(comb_mat[0], m, sizeof(m));
+  for (int k = 1; k < 24; ++k) gf2_square(comb_mat[k], comb_mat[k - 1]);
+}
+
+static uint32_t crc32c_sw(const uint8_t* p, size_t n, uint32_t crc_in) {
+  uint32_t crc = crc_in ^ 0xFFFFFFFFu;
+  while (n && ((uintptr_t)p & 7)) { crc = (crc >> 8) ^ crc_tab[0][(crc ^ *p++) & 0xFF]; --n; }
+#if defined(__SSE4_2__)
+  while (n >= 8) {
+    crc = (uint32_t)__builtin_ia32_crc32di(crc, *(const uint64_t*)p);
+    p += 8; n -= 8;
+  }
+  while (n) { crc = __builtin_ia32_crc32qi(crc, *p++); --n; }
+#else
+  while (n >= 8) {
+    uint64_t v = *(const uint64_t*)p;
+    uint32_t lo = (uint32_t)v ^ crc, hi = (uint32_t)(v >> 32);
+    crc = crc_tab[7][lo & 0xFF] ^ crc_tab[6][(lo >> 8) & 0xFF] ^
+          crc_tab[5][(lo >> 16) & 0xFF] ^ crc_tab[4][lo >> 24] ^
+          crc_tab[3][hi & 0xFF] ^ crc_tab[2][(hi >> 8) & 0xFF] ^
+          crc_tab[1][(hi >> 16) & 0xFF] ^ crc_tab[0][hi >> 24];
+    p += 8; n -= 8;
+  }
+  while (n) { crc = (crc >> 8) ^ crc_tab[0][(crc ^ *p++) & 0xFF]; --n; }
+#endif
+  return crc ^ 0xFFFFFFFFu;
+}
+
+// shift a finalized crc over `len` zero bytes (zlib crc32_combine algebra)
+static uint32_t crc32c_shift(uint32_t crc, uint64_t len) {
+  if (!len) return crc;
+  uint32_t m[32], sq[32];
+  m[0] = CRC_POLY;
+  for (int i = 1; i < 32; ++i) m[i] = 1u << (i - 1);
+  // m = M1 (one bit). apply for each set bit of 8*len.
+  uint64_t bits = 8ull * len;
+  while (bits) {
+    if (bits & 1) crc = gf2_apply(m, crc);
+    bits >>= 1;
+    if (bits) { gf2_square(sq, m); std::memcpy(m, sq, sizeof(m)); }
+  }
+  return crc;
+}
+
+static uint32_t crc32c_combine(uint32_t crc1, uint32_t crc2, uint64_t len2) {
+  return crc32c_shift(crc1, len2) ^ crc2;
+}
+
+// ---------------------------------------------------------------------------
+// Kernels — single translation unit (device symbols + hipMemcpyToSymbol
+// need no -fgpu-rdc this way)
+// ---------------------------------------------------------------------------
+#include "kernels.hip"
+
+// ---------------------------------------------------------------------------
+// GPU availability
+// ---------------------------------------------------------------------------
+
+static int cached_device_count = -2;
+static int device_count() {
+  if (cached_device_count == -2) {
+    int n = 0;
+    if (hipGetDeviceCount(&n) != hipSuccess) n = 0;
+    cached_device_count = n;
+  }
+  return cached_device_count;
+}
+
+// ---------------------------------------------------------------------------
+// Arena
+// ---------------------------------------------------------------------------
+
+struct Arena {
+  void* base = nullptr;
+  size_t cap = 0;
+  int device = -1;       // -1 = host memory (CPU fallback / MEM tier)
+  bool pinned = false;   // host arena allocated with hipHostMalloc
+  hipStream_t stream = nullptr;       // copy stream
+  hipStream_t kstream = nullptr;      // kernel stream
+  // pinned staging ring (device arenas)
+  std::vector<void*> pin;
+  std::vector<hipEvent_t> ev;
+  size_t pin_sz = 0;
+  // device scratch for gather/crc results
+  void* scratch = nullptr;
+  size_t scratch_sz = 0;
+  bool crc_tables_uploaded = false;
+  std::mutex mu;
+
+  bool is_dev() const { return device >= 0; }
+};
+
+static std::vector<Arena*> g_arenas;
+static std::mutex g_arenas_mu;
+
+static Arena* get_arena(int h) {
+  std::lock_guard<std::mutex> g(g_arenas_mu);
+  if (h < 0 || h >= (int)g_arenas.size() || !g_arenas[h])
+    throw std::runtime_error("bad arena handle");
+  return g_arenas[h];
+}
+
+static int arena_create(int device, size_t cap, size_t staging_bytes,
+                        int staging_count, bool host_pinned) {
+  auto* a = new Arena();
+  a->device = device;
+  a->cap = cap;
+  if (device >= 0) {
+    if (device >= device_count()) {
+      delete a;
+      throw std::runtime_error("no such GPU device");
+    }
+    HIP_CHECK(hipSetDevice(device));
+    HIP_CHECK(hipMalloc(&a->base, cap));
+    HIP_CHECK(hipStreamCreateWithFlags(&a->stream, hipStreamNonBlocking));
+    HIP_CHECK(hipStreamCreateWithFlags(&a->kstream, hipStreamNonBlocking));
+    a->pin_sz = staging_bytes;
+    a->pin.resize(staging_count);
+    a->ev.resize(staging_count);
+    for (int i = 0; i < staging_count; ++i) {
+      HIP_CHECK(hipHostMalloc(&a->pin[i], staging_bytes, hipHostMallocDefault));
+      HIP_CHECK(hipEventCreateWithFlags(&a->ev[i], hipEventDisableTiming));
+    }
+  } else {
+    if (host_pinned && device_count() > 0) {
+      HIP_CHECK(hipHostMalloc(&a->base, cap, hipHostMallocDefault));
+      a->pinned = true;
+    } else {
+      a->base = std::malloc(cap);
+      if (!a->base) { delete a; throw std::bad_alloc(); }
+    }
+  }
+  std::lock_guard<std::mutex> g(g_arenas_mu);
+  for (size_t i = 0; i < g_arenas.size(); ++i)
+    if (!g_arenas[i]) { g_arenas[i] = a; return (int)i; }
+  g_arenas.push_back(a);
+  return (int)g_arenas.size() - 1;
+}
+
+static void arena_destroy(int h) {
+  Arena* a;
+  {
+    std::lock_guard<std::mutex> g(g_arenas_mu);
+    if (h < 0 || h >= (int)g_arenas.size() || !g_arenas[h]) return;
+    a = g_arenas[h];
+    g_arenas[h] = nullptr;
+  }
+  if (a->is_dev()) {
+    hipSetDevice(a->device);
+    hipStreamSynchronize(a->stream);
+    for (auto* p : a->pin) hipHostFree(p);
+    for (auto e : a->ev) hipEventDestroy(e);
+    if (a->scratch) hipFree(a->scratch);
+    hipFree(a->base);
+    hipStreamDestroy(a->stream);
+    hipStreamDestroy(a->kstream);
+  } else if (a->pinned) {
+    hipHostFree(a->base);
+  } else {
+    std::free(a->base);
+  }
+  delete a;
+}
+
+static void ensure_scratch(Arena* a, size_t n) {
+  if (a->scratch_sz >= n) return;
+  if (a->scratch) HIP_CHECK(hipFree(a->scratch));
+  HIP_CHECK(hipMalloc(&a->scratch, n));
+  a->scratch_sz = n;
+}
+
+static void ensure_crc_tables(Arena* a) {
+  if (a->crc_tables_uploaded) return;
+  crc_init_tables();
+  HIP_CHECK(hipMemcpyToSymbol(HIP_SYMBOL(g_crc_tab), crc_tab, sizeof(crc_tab)));
+  HIP_CHECK(hipMemcpyToSymbol(HIP_SYMBOL(g_comb_mat), comb_mat, sizeof(comb_mat)));
+  a->crc_tables_uploaded = true;
+}
+
+// ---- host<->arena copies with pinned-ring pipelining ----
+
+static void check_range(Arena* a, uint64_t off, uint64_t n) {
+  if (off + n > a->cap) throw std::runtime_error("arena range out of bounds");
+}
+
+// device -> host buffer, chunked through the pinned ring: D2H DMA of chunk
+// k overlaps the CPU memcpy of chunk k-1 (the staging pipeline of
+// BASELINE.json's "pinned hipMemcpyAsync on a side stream").
+static void dev_read(Arena* a, uint64_t off, uint8_t* dst, uint64_t n) {
+  std::lock_guard<std::mutex> g(a->mu);
+  HIP_CHECK(hipSetDevice(a->device));
+  const uint8_t* src = (const uint8_t*)a->base + off;
+  size_t nb = a->pin.size();
+  uint64_t nchunks = (n + a->pin_sz - 1) / a->pin_sz;
+  // issue up to nb asynchronous D2H copies ahead, drain in order
+  for (uint64_t c = 0; c < nchunks; ++c) {
+    uint64_t coff = c * a->pin_sz;
+    uint64_t clen = std::min<uint64_t>(a->pin_sz, n - coff);
+    size_t slot = c % nb;
+    if (c >= nb) HIP_CHECK(hipEventSynchronize(a->ev[slot]));
+    HIP_CHECK(hipMemcpyAsync(a->pin[slot], src + coff, clen,
+                             hipMemcpyDeviceToHost, a->stream));
+    HIP_CHECK(hipEventRecord(a->ev[slot], a->stream));
+    if (c + 1 == nchunks || ((c + 1) % nb) == 0) {
+      // drain the window and memcpy out
+      uint64_t first = (c / nb) * nb;
+      for (uint64_t d = first; d <= c; ++d) {
+        size_t ds = d % nb;
+        HIP_CHECK(hipEventSynchronize(a->ev[ds]));
+        uint64_t doff = d * a->pin_sz;
+        std::memcpy(dst + doff, a->pin[ds],
+                    std::min<uint64_t>(a->pin_sz, n - doff));
+      }
+    }
+  }
+}
+
+static void dev_write(Arena* a, uint64_t off, const uint8_t* src, uint64_t n) {
+  std::lock_guard<std::mutex> g(a->mu);
+  HIP_CHECK(hipSetDevice(a->device));
+  uint8_t* dst = (uint8_t*)a->base + off;
+  size_t nb = a->pin.size();
+  uint64_t nchunks = (n + a->pin_sz - 1) / a->pin_sz;
+  for (uint64_t c = 0; c < nchunks; ++c) {
+    uint64_t coff = c * a->pin_sz;
+    uint64_t clen = std::min<uint64_t>(a->pin_sz, n - coff);
+    size_t slot = c % nb;
+    if (c >= nb) HIP_CHECK(hipEventSynchronize(a->ev[slot]));
+    std::memcpy(a->pin[slot], src + coff, clen);
+    HIP_CHECK(hipMemcpyAsync(dst + coff, a->pin[slot], clen,
+                             hipMemcpyHostToDevice, a->stream));
+    HIP_CHECK(hipEventRecord(a->ev[slot], a->stream));
+  }
+  HIP_CHECK(hipStreamSynchronize(a->stream));
+}
+
+static void arena_read(int h, uint64_t off, py::buffer buf, uint64_t buf_off,
+                       uint64_t n) {
+  Arena* a = get_arena(h);
+  py::buffer_info info = buf.request(true);
+  uint64_t cap = (uint64_t)info.size * (uint64_t)info.itemsize;
+  if (buf_off + n > cap) throw std::runtime_error("dst buffer too small");
+  check_range(a, off, n);
+  uint8_t* dst = (uint8_t*)info.ptr + buf_off;
+  py::gil_scoped_release rel;
+  if (a->is_dev()) dev_read(a, off, dst, n);
+  else std::memcpy(dst, (uint8_t*)a->base + off, n);
+}
+
+static void arena_write(int h, uint64_t off, py::buffer buf, uint64_t buf_off,
+                        uint64_t n) {
+  Arena* a = get_arena(h);
+  py::buffer_info info = buf.request(false);
+  uint64_t cap = (uint64_t)info.size * (uint64_t)info.itemsize;
+  if (buf_off + n > cap) throw std::runtime_error("src buffer too small");
+  check_range(a, off, n);
+  const uint8_t* src = (const uint8_t*)info.ptr + buf_off;
+  py::gil_scoped_release rel;
+  if (a->is_dev()) dev_write(a, off, src, n);
+  else std::memcpy((uint8_t*)a->base + off, src, n);
+}
+
+// raw-pointer variants (torch tensors, other arenas' memory)
+static void arena_read_ptr(int h, uint64_t off, uintptr_t dst, uint64_t n,
+                           bool dst_is_device) {
+  Arena* a = get_arena(h);
+  check_range(a, off, n);
+  py::gil_scoped_release rel;
+  if (a->is_dev()) {
+    std::lock_guard<std::mutex> g(a->mu);
+    HIP_CHECK(hipSetDevice(a->device));
+    HIP_CHECK(hipMemcpyAsync((void*)dst, (uint8_t*)a->base + off, n,
+                             dst_is_device ? hipMemcpyDeviceToDevice
+                                           : hipMemcpyDeviceToHost,
+                             a->stream));
+    HIP_CHECK(hipStreamSynchronize(a->stream));
+  } else if (dst_is_device) {
+    HIP_CHECK(hipMemcpy((void*)dst, (uint8_t*)a->base + off, n,
+                        hipMemcpyHostToDevice));
+  } else {
+    std::memcpy((void*)dst, (uint8_t*)a->base + off, n);
+  }
+}
+
+static void arena_write_ptr(int h, uint64_t off, uintptr_t src, uint64_t n,
+                            bool src_is_device) {
+  Arena* a = get_arena(h);
+  check_range(a, off, n);
+  py::gil_scoped_release rel;
+  if (a->is_dev()) {
+    std::lock_guard<std::mutex> g(a->mu);
+    HIP_CHECK(hipSetDevice(a->device));
+    HIP_CHECK(hipMemcpyAsync((uint8_t*)a->base + off, (const void*)src, n,
+                             src_is_device ? hipMemcpyDeviceToDevice
+                                           : hipMemcpyHostToDevice,
+                             a->stream));
+    HIP_CHECK(hipStreamSynchronize(a->stream));
+  } else if (src_is_device) {
+    HIP_CHECK(hipMemcpy((uint8_t*)a->base + off, (const void*)src, n,
+                        hipMemcpyDeviceToHost));
+  } else {
+    std::memcpy((uint8_t*)a->base + off, (const void*)src, n);
+  }
+}
+
+// arena -> arena (same or cross device / host tiers)
+static void arena_copy(int dst_h, uint64_t dst_off, int src_h, uint64_t src_off,
+                       uint64_t n) {
+  Arena* d = get_arena(dst_h);
+  Arena* s = get_arena(src_h);
+  check_range(d, dst_off, n);
+  check_range(s, src_off, n);
+  py::gil_scoped_release rel;
+  uint8_t* dp = (uint8_t*)d->base + dst_off;
+  uint8_t* sp = (uint8_t*)s->base + src_off;
+  if (!d->is_dev() && !s->is_dev()) {
+    std::memcpy(dp, sp, n);
+    return;
+  }
+  Arena* deva = d->is_dev() ? d : s;
+  std::lock_guard<std::mutex> g(deva->mu);
+  HIP_CHECK(hipSetDevice(deva->device));
+  hipMemcpyKind kind =
+      d->is_dev() ? (s->is_dev() ? hipMemcpyDeviceToDevice : hipMemcpyHostToDevice)
+                  : hipMemcpyDeviceToHost;
+  if (d->is_dev() && s->is_dev() && d->device != s->device) {
+    HIP_CHECK(hipMemcpyPeerAsync(dp, d->device, sp, s->device, n, deva->stream));
+  } else {
+    HIP_CHECK(hipMemcpyAsync(dp, sp, n, kind, deva->stream));
+  }
+  HIP_CHECK(hipStreamSynchronize(deva->stream));
+}
+
+// ---- kernels ----
+
+static int grid_for(uint64_t work_items) {
+  // >> 256 workgroups to fill 8 XCDs; cap for grid-stride loops
+  uint64_t g = (work_items + WG - 1) / WG;
+  if (g > 8192) g = 8192;
+  if (g < 1) g = 1;
+  return (int)g;
+}
+
+static void arena_fill(int h, uint64_t off, uint64_t n, int value) {
+  Arena* a = get_arena(h);
+  check_range(a, off, n);
+  py::gil_scoped_release rel;
+  if (!a->is_dev()) {
+    std::memset((uint8_t*)a->base + off, value, n);
+    return;
+  }
+  std::lock_guard<std::mutex> g(a->mu);
+  HIP_CHECK(hipSetDevice(a->device));
+  hipLaunchKernelGGL(fill_kernel, dim3(grid_for(n / 16 + 1)), dim3(WG), 0,
+                     a->kstream, (uint8_t*)a->base + off, (uint8_t)value, n);
+  HIP_CHECK(hipGetLastError());
+  HIP_CHECK(hipStreamSynchronize(a->kstream));
+}
+
+static uint32_t arena_crc32c(int h, uint64_t off, uint64_t n) {
+  Arena* a = get_arena(h);
+  check_range(a, off, n);
+  if (!a->is_dev()) {
+    const uint8_t* p = (const uint8_t*)a->base + off;
+    py::gil_scoped_release rel;
+    crc_init_tables();
+    return crc32c_sw(p, n, 0);
+  }
+  py::gil_scoped_release rel;
+  std::lock_guard<std::mutex> g(a->mu);
+  HIP_CHECK(hipSetDevice(a->device));
+  ensure_crc_tables(a);
+  const uint8_t* p = (const uint8_t*)a->base + off;
+  if (((uintptr_t)p & 7) != 0) throw std::runtime_error("crc32c needs 8B-aligned offset");
+  uint64_t n_sub = n / CRC_SUB;
+  uint64_t tail = n - n_sub * CRC_SUB;
+  uint32_t result = 0;
+  if (n_sub) {
+    uint32_t n_wg = (uint32_t)((n_sub + WG - 1) / WG);
+    ensure_scratch(a, n_wg * sizeof(uint32_t) + a->pin_sz);
+    uint32_t* d_out = (uint32_t*)a->scratch;
+    hipLaunchKernelGGL(crc32c_kernel, dim3(n_wg), dim3(WG), 0, a->kstream, p,
+                       n_sub, d_out);
+    HIP_CHECK(hipGetLastError());
+    std::vector<uint32_t> host_out(n_wg);
+    HIP_CHECK(hipMemcpyAsync(host_out.data(), d_out, n_wg * sizeof(uint32_t),
+                             hipMemcpyDeviceToHost, a->kstream));
+    HIP_CHECK(hipStreamSynchronize(a->kstream));
+    // combine per-workgroup crcs (each covers WG*CRC_SUB except the last)
+    uint64_t remain = n_sub;
+    result = host_out[0];
+    uint64_t covered0 = std::min<uint64_t>(WG, remain);
+    remain -= covered0;
+    for (uint32_t w = 1; w < n_wg; ++w) {
+      uint64_t cov = std::min<uint64_t>(WG, remain);
+      remain -= cov;
+      result = crc32c_combine(result, host_out[w], cov * CRC_SUB);
+    }
+  }
+  if (tail) {
+    // read the tail back through a pinned slot and crc on host
+    uint8_t* pin0 = (uint8_t*)(a->pin.empty() ? nullptr : a->pin[0]);
+    std::vector<uint8_t> tmp;
+    uint8_t* dst;
+    if (pin0 && tail <= a->pin_sz) dst = pin0;
+    else { tmp.resize(tail); dst = tmp.data(); }
+    HIP_CHECK(hipMemcpyAsync(dst, p + n_sub * CRC_SUB, tail,
+                             hipMemcpyDeviceToHost, a->kstream));
+    HIP_CHECK(hipStreamSynchronize(a->kstream));
+    uint32_t tail_crc = crc32c_sw(dst, tail, 0);
+    result = n_sub ? crc32c_combine(result, tail_crc, tail) : tail_crc;
+  }
+  return result;
+}
+
+static uint32_t crc32c_buf(py::buffer buf, uint32_t init) {
+  py::buffer_info info = buf.request(false);
+  const uint8_t* p = (const uint8_t*)info.ptr;
+  size_t n = (size_t)info.size * info.itemsize;
+  py::gil_scoped_release rel;
+  crc_init_tables();
+  return crc32c_sw(p, n, init);
+}
+
+// gather device extents into a host buffer: kernel packs extents into
+// contiguous scratch on-device, then one pipelined D2H.
+static void arena_gather(int h, std::vector<std::pair<uint64_t, uint64_t>> ext,
+                         py::buffer out, uint64_t out_off) {
+  Arena* a = get_arena(h);
+  py::buffer_info info = out.request(true);
+  uint64_t total = 0;
+  for (auto& e : ext) { check_range(a, e.first, e.second); total += e.second; }
+  if (out_off + total > (uint64_t)info.size * info.itemsize)
+    throw std::runtime_error("gather dst too small");
+  uint8_t* dst = (uint8_t*)info.ptr + out_off;
+  py::gil_scoped_release rel;
+  if (!a->is_dev()) {
+    uint64_t o = 0;
+    for (auto& e : ext) {
+      std::memcpy(dst + o, (uint8_t*)a->base + e.first, e.second);
+      o += e.second;
+    }
+    return;
+  }
+  std::lock_guard<std::mutex> g(a->mu);
+  HIP_CHECK(hipSetDevice(a->device));
+  // build extent + tile tables
+  std::vector<Extent> exts(ext.size());
+  std::vector<uint32_t> tile_ext;
+  std::vector<uint64_t> tile_off;
+  uint64_t o = 0;
+  for (size_t i = 0; i < ext.size(); ++i) {
+    exts[i] = {ext[i].first, o, ext[i].second};
+    for (uint64_t t = 0; t < ext[i].second; t += TILE) {
+      tile_ext.push_back((uint32_t)i);
+      tile_off.push_back(t);
+    }
+    o += ext[i].second;
+  }
+  size_t meta = exts.size() * sizeof(Extent) +
+                tile_ext.size() * (sizeof(uint32_t) + sizeof(uint64_t));
+  ensure_scratch(a, total + meta + 64);
+  uint8_t* d_pack = (uint8_t*)a->scratch;
+  Extent* d_ext = (Extent*)(d_pack + ((total + 63) & ~63ull));
+  uint32_t* d_te = (uint32_t*)(d_ext + exts.size());
+  uint64_t* d_to = (uint64_t*)(((uintptr_t)(d_te + tile_ext.size()) + 7) & ~7ull);
+  HIP_CHECK(hipMemcpyAsync(d_ext, exts.data(), exts.size() * sizeof(Extent),
+                           hipMemcpyHostToDevice, a->kstream));
+  HIP_CHECK(hipMemcpyAsync(d_te, tile_ext.data(),
+                           tile_ext.size() * sizeof(uint32_t),
+                           hipMemcpyHostToDevice, a->kstream));
+  HIP_CHECK(hipMemcpyAsync(d_to, tile_off.data(),
+                           tile_off.size() * sizeof(uint64_t),
+                           hipMemcpyHostToDevice, a->kstream));
+  int grid = (int)std::min<size_t>(tile_ext.size(), 8192);
+  hipLaunchKernelGGL(copy_extents_kernel, dim3(grid), dim3(WG), 0, a->kstream,
+                     (const uint8_t*)a->base, d_pack, d_ext, d_te, d_to,
+                     (uint32_t)tile_ext.size());
+  HIP_CHECK(hipGetLastError());
+  HIP_CHECK(hipStreamSynchronize(a->kstream));
+  // pipelined D2H of the packed region
+  size_t nb = a->pin.size();
+  uint64_t nchunks = (total + a->pin_sz - 1) / a->pin_sz;
+  for (uint64_t c = 0; c < nchunks; ++c) {
+    uint64_t coff = c * a->pin_sz;
+    uint64_t clen = std::min<uint64_t>(a->pin_sz, total - coff);
+    size_t slot = c % nb;
+    if (c >= nb) HIP_CHECK(hipEventSynchronize(a->ev[slot]));
+    HIP_CHECK(hipMemcpyAsync(a->pin[slot], d_pack + coff, clen,
+                             hipMemcpyDeviceToHost, a->stream));
+    HIP_CHECK(hipEventRecord(a->ev[slot], a->stream));
+    if (c + 1 == nchunks || ((c + 1) % nb) == 0) {
+      uint64_t first = (c / nb) * nb;
+      for (uint64_t d2 = first; d2 <= c; ++d2) {
+        size_t ds = d2 % nb;
+        HIP_CHECK(hipEventSynchronize(a->ev[ds]));
+        uint64_t doff = d2 * a->pin_sz;
+        std::memcpy(dst + doff, a->pin[ds],
+                    std::min<uint64_t>(a->pin_sz, total - doff));
+      }
+    }
+  }
+}
+
+static uintptr_t arena_base_ptr(int h) {
+  return (uintptr_t)get_arena(h)->base;
+}
+
+static py::dict arena_info(int h) {
+  Arena* a = get_arena(h);
+  py::dict d;
+  d["capacity"] = a->cap;
+  d["device"] = a->device;
+  d["pinned"] = a->pinned;
+  d["staging_bytes"] = a->pin_sz;
+  d["staging_count"] = a->pin.size();
+  return d;
+}
+
+static void device_sync(int device) {
+  py::gil_scoped_release rel;
+  HIP_CHECK(hipSetDevice(device));
+  HIP_CHECK(hipDeviceSynchronize());
+}
+
+static py::dict device_mem_info(int device) {
+  size_t free_b = 0, total_b = 0;
+  HIP_CHECK(hipSetDevice(device));
+  HIP_CHECK(hipMemGetInfo(&free_b, &total_b));
+  py::dict d;
+  d["free"] = free_b;
+  d["total"] = total_b;
+  return d;
+}
+
+PYBIND11_MODULE(_native, m) {
+  m.doc() = "curvine_amd native data plane (HIP/CDNA4, gfx950)";
+  m.def("device_count", &device_count);
+  m.def("device_sync", &device_sync);
+  m.def("device_mem_info", &device_mem_info);
+  m.def("arena_create", &arena_create, py::arg("device"), py::arg("capacity"),
+        py::arg("staging_bytes") = 4 << 20, py::arg("staging_count") = 8,
+        py::arg("host_pinned") = false);
+  m.def("arena_destroy", &arena_destroy);
+  m.def("arena_read", &arena_read);
+  m.def("arena_write", &arena_write);
+  m.def("arena_read_ptr", &arena_read_ptr);
+  m.def("arena_write_ptr", &arena_write_ptr);
+  m.def("arena_copy", &arena_copy);
+  m.def("arena_fill", &arena_fill);
+  m.def("arena_crc32c", &arena_crc32c);
+  m.def("arena_gather", &arena_gather);
+  m.def("arena_base_ptr", &arena_base_ptr);
+  m.def("arena_info", &arena_info);
+  m.def("crc32c", &crc32c_buf, py::arg("buf"), py::arg("init") = 0);
+  m.def("crc32c_combine", &crc32c_combine);
+  m.attr("CRC_SUB") = CRC_SUB;
+  m.attr("__hip_arch__") = "gfx950";
+}

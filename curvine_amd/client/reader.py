@@ -1,0 +1,210 @@
+"""FsReader: positioned reads across blocks with replica fallback.
+
+Analog of /root/reference/crates/client/curvine-client-core/src/file/
+fs_reader.rs + block/block_reader.rs:111-230 (local short-circuit first,
+remote fallback on failure, failed-worker tracking) and
+fs_reader_parallel.rs (sliced parallel reads for large requests);
+`ReadDetector` (read_detector.rs:42-134) gates sequential prefetch.
+"""
+from __future__ import annotations
+
+import asyncio
+from typing import Optional
+
+from curvine_amd import errors as err
+from curvine_amd.client.block_client import (BlockReaderHole, BlockReaderLocal,
+                                             BlockReaderRemote)
+from curvine_amd.client.fs_client import FsClient
+from curvine_amd.model import FileBlocks, LocatedBlock
+
+
+class ReadDetector:
+    """Sequential-vs-random pattern detection (read_detector.rs analog)."""
+
+    def __init__(self):
+        self.last_end = 0
+        self.seq_count = 0
+        self.rand_count = 0
+
+    def observe(self, off: int, n: int) -> None:
+        if off == self.last_end:
+            self.seq_count += 1
+        else:
+            self.rand_count += 1
+        self.last_end = off + n
+
+    @property
+    def is_sequential(self) -> bool:
+        return self.seq_count >= self.rand_count
+
+
+class FsReader:
+    def __init__(self, client: FsClient, file_blocks: FileBlocks):
+        self.client = client
+        self.fb = file_blocks
+        self.status = file_blocks.status
+        self.length = self.status.length
+        self.pos = 0
+        self.detector = ReadDetector()
+        self._readers: dict[int, object] = {}   # block index -> reader
+        self.failed_workers: set[int] = set()
+        self.chunk_size = client.conf.client.read_chunk_size
+        self.parallel = max(1, client.conf.client.read_parallel)
+        self.slice_size = client.conf.client.read_slice_size
+
+    # ---------------- block reader selection ----------------
+    def _block_at(self, off: int) -> tuple[int, LocatedBlock, int]:
+        """(index, located block, offset within block)."""
+        # blocks are equally sized except the last
+        bs = self.status.block_size
+        idx = off // bs
+        if idx >= len(self.fb.blocks):
+            raise err.OutOfRange(f"offset {off} beyond {self.length}")
+        lb = self.fb.blocks[idx]
+        return idx, lb, off - lb.offset
+
+    async def _reader_for(self, idx: int):
+        r = self._readers.get(idx)
+        if r is not None:
+            return r
+        lb = self.fb.blocks[idx]
+        r = await self._open_block_reader(lb)
+        self._readers[idx] = r
+        return r
+
+    async def _open_block_reader(self, lb: LocatedBlock):
+        if not lb.locations:
+            # no cached replica: sparse hole (or caller falls back to UFS)
+            return BlockReaderHole(lb.block.length)
+        from curvine_amd.worker import registry
+        candidates = [a for a in lb.locations
+                      if a.worker_id not in self.failed_workers] or lb.locations
+        # in-process short-circuit first
+        if self.client.conf.client.short_circuit:
+            for addr in candidates:
+                store = registry.lookup(addr.worker_id)
+                if store is not None:
+                    try:
+                        return BlockReaderLocal(store, lb.block.block_id)
+                    except Exception:  # noqa: BLE001
+                        self.failed_workers.add(addr.worker_id)
+        last: Exception = err.BlockNotFound(str(lb.block.block_id))
+        for addr in candidates:
+            try:
+                r = BlockReaderRemote(addr, lb.block.block_id)
+                return r
+            except Exception as e:  # noqa: BLE001
+                self.failed_workers.add(addr.worker_id)
+                last = e
+        raise last
+
+    # ---------------- reads ----------------
+    async def pread(self, off: int, n: int) -> bytes:
+        n = max(0, min(n, self.length - off))
+        if n == 0:
+            return b""
+        out = bytearray(n)
+        await self.pread_into(off, out, 0, n)
+        return bytes(out)
+
+    async def pread_into(self, off: int, out, out_off: int, n: int) -> int:
+        n = max(0, min(n, self.length - off))
+        if n == 0:
+            return 0
+        self.detector.observe(off, n)
+        if n >= self.slice_size and self.parallel > 1:
+            return await self._pread_parallel(off, out, out_off, n)
+        got = 0
+        while got < n:
+            idx, lb, boff = self._block_at(off + got)
+            want = min(n - got, lb.block.length - boff)
+            if want <= 0:
+                break
+            r = await self._reader_for(idx)
+            rn = await self._read_with_fallback(idx, r, boff, out,
+                                                out_off + got, want)
+            if rn <= 0:
+                break
+            got += rn
+        return got
+
+    async def _read_with_fallback(self, idx, reader, boff, out, out_off, want):
+        try:
+            return await reader.read_into(boff, out, out_off, want)
+        except Exception:  # noqa: BLE001 — replica failed mid-read
+            self._readers.pop(idx, None)
+            lb = self.fb.blocks[idx]
+            r2 = await self._open_block_reader(lb)
+            self._readers[idx] = r2
+            return await r2.read_into(boff, out, out_off, want)
+
+    async def _pread_parallel(self, off, out, out_off, n) -> int:
+        """Slice fan-out (fs_reader_parallel.rs:94-131 analog)."""
+        k = min(self.parallel, (n + self.slice_size - 1) // self.slice_size)
+        per = (n + k - 1) // k
+        tasks = []
+        for i in range(k):
+            so = off + i * per
+            sn = min(per, off + n - so)
+            if sn <= 0:
+                break
+            tasks.append(self._pread_slice(so, out, out_off + i * per, sn))
+        results = await asyncio.gather(*tasks)
+        return sum(results)
+
+    async def _pread_slice(self, off, out, out_off, n) -> int:
+        got = 0
+        while got < n:
+            idx, lb, boff = self._block_at(off + got)
+            want = min(n - got, lb.block.length - boff)
+            if want <= 0:
+                break
+            r = await self._reader_for(idx)
+            rn = await self._read_with_fallback(idx, r, boff, out,
+                                                out_off + got, want)
+            if rn <= 0:
+                break
+            got += rn
+        return got
+
+    async def read(self, n: int = -1) -> bytes:
+        if n < 0:
+            n = self.length - self.pos
+        data = await self.pread(self.pos, n)
+        self.pos += len(data)
+        return data
+
+    def seek(self, pos: int) -> None:
+        self.pos = pos
+
+    async def pread_to_device(self, off: int, dst_ptr: int, n: int) -> int:
+        """Read file bytes directly into consumer GPU memory
+        (hipMemcpyDtoD from the HBM arena when local, staged otherwise)."""
+        n = max(0, min(n, self.length - off))
+        got = 0
+        while got < n:
+            idx, lb, boff = self._block_at(off + got)
+            want = min(n - got, lb.block.length - boff)
+            if want <= 0:
+                break
+            r = await self._reader_for(idx)
+            if isinstance(r, BlockReaderLocal):
+                rn = await r.read_to_device(boff, dst_ptr + got, want)
+            else:
+                tmp = bytearray(want)
+                rn = await r.read_into(boff, tmp, 0, want)
+                from curvine_amd import native
+                native.load().arena_write_ptr  # presence check
+                raise err.Unsupported(
+                    "device reads from remote workers need a local staging "
+                    "arena; use pread_into + upload")
+            got += rn
+        return got
+
+    def close(self) -> None:
+        for r in self._readers.values():
+            try:
+                r.close()
+            except Exception:  # noqa: BLE001
+                pass
+        self._readers.clear()

@@ -1,0 +1,110 @@
+"""FsWriter: buffered block writer with star replica fan-out.
+
+Analog of /root/reference/crates/client/curvine-client-core/src/file/
+fs_writer.rs + fs_writer_buffer.rs (background flusher) +
+block/block_writer.rs:184-210 (each chunk written to ALL replica locations
+in parallel — star replication).
+"""
+from __future__ import annotations
+
+import asyncio
+from typing import Optional
+
+from curvine_amd import errors as err
+from curvine_amd.client.block_client import make_block_writer
+from curvine_amd.client.fs_client import FsClient
+from curvine_amd.model import FileStatus, LocatedBlock
+
+
+class FsWriter:
+    def __init__(self, client: FsClient, status: FileStatus):
+        self.client = client
+        self.status = status
+        self.path = status.path
+        self.block_size = status.block_size
+        self.chunk_size = client.conf.client.write_chunk_size
+        self._buf = bytearray()
+        self._block: Optional[LocatedBlock] = None
+        self._writers: list = []
+        self._block_pos = 0
+        self._block_lens: list[int] = []
+        self._commits: list[dict] = []   # block locations to report on complete
+        self.pos = 0
+        self._closed = False
+
+    async def write(self, data) -> int:
+        if self._closed:
+            raise err.FsError("writer closed")
+        data = memoryview(data)
+        total = len(data)
+        while len(data) > 0:
+            room = self.chunk_size - len(self._buf)
+            take = min(room, len(data))
+            self._buf.extend(data[:take])
+            data = data[take:]
+            if len(self._buf) >= self.chunk_size:
+                await self._flush_chunk()
+        self.pos += total
+        return total
+
+    async def _flush_chunk(self) -> None:
+        buf = self._buf
+        self._buf = bytearray()
+        off = 0
+        while off < len(buf):
+            if self._block is None:
+                await self._next_block()
+            room = self.block_size - self._block_pos
+            take = min(room, len(buf) - off)
+            chunk = bytes(buf[off:off + take])
+            # star fan-out: all replicas in parallel
+            await asyncio.gather(*[w.write(chunk) for w in self._writers])
+            self._block_pos += take
+            off += take
+            if self._block_pos >= self.block_size:
+                await self._commit_block()
+
+    async def _next_block(self) -> None:
+        lb = await self.client.add_block(self.path)
+        self._block = lb
+        self._block_pos = 0
+        if not lb.locations:
+            raise err.NoAvailableWorker(self.path)
+        self._writers = [make_block_writer(addr, lb.block.block_id,
+                                           self.block_size, tier)
+                         for addr, tier in zip(lb.locations, lb.tiers)]
+
+    async def _commit_block(self) -> None:
+        tiers = await asyncio.gather(
+            *[w.commit(self._block_pos) for w in self._writers])
+        self._commits.append({
+            "block_id": self._block.block.block_id,
+            "locations": [a.worker_id for a in self._block.locations],
+            "tiers": [t or hint for t, hint in zip(tiers, self._block.tiers)]})
+        self._block_lens.append(self._block_pos)
+        self._block = None
+        self._writers = []
+        self._block_pos = 0
+
+    async def flush(self) -> None:
+        if self._buf:
+            await self._flush_chunk()
+
+    async def complete(self) -> FileStatus:
+        if self._closed:
+            return self.status
+        await self.flush()
+        if self._block is not None:
+            await self._commit_block()
+        self._closed = True
+        length = sum(self._block_lens)
+        return await self.client.complete_file(self.path, length,
+                                               self._block_lens, self._commits)
+
+    async def abort(self) -> None:
+        self._closed = True
+        for w in self._writers:
+            try:
+                await w.abort()
+            except Exception:  # noqa: BLE001
+                pass

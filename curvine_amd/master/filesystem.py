@@ -121,10 +121,17 @@ class MasterFilesystem:
             tiers=[node.storage_tier] * len(workers))
 
     def complete_file(self, path: str, length: int,
-                      block_lens: list[int] | None = None) -> FileStatus:
+                      block_lens: list[int] | None = None,
+                      commits: list[dict] | None = None) -> FileStatus:
         node = self.fs_dir.must_resolve(path)
         self.fs_dir.complete_file(node, length, block_lens)
         self.writing.pop(node.id, None)
+        # client-reported block locations (commit metadata,
+        # block_writer.rs:324-337 analog) — soft state, ahead of heartbeats
+        for c in commits or []:
+            for wid, tier in zip(c.get("locations", []),
+                                 c.get("tiers", []) or ["MEM"] * len(c.get("locations", []))):
+                self.workers.block_locs.setdefault(c["block_id"], {})[wid] = tier
         return self.fs_dir.status_of(node, norm_path(path))
 
     def delete(self, path: str, recursive: bool = False) -> int:

@@ -89,7 +89,8 @@ class MasterHandler:
         return {"block": lb.to_dict()}
 
     def op_completefile(self, h, d):
-        st = self.fs.complete_file(h["path"], h["length"], h.get("block_lens"))
+        st = self.fs.complete_file(h["path"], h["length"], h.get("block_lens"),
+                                   h.get("commits"))
         return {"status": st.to_dict()}
 
     def op_getblocklocations(self, h, d):

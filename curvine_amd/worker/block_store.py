@@ -1,0 +1,203 @@
+"""Worker block store: multi-tier facade with reservation/rollback.
+
+Analog of the reference's `BlockStore` + `VfsDataset`
+(/root/reference/curvine-worker/src/worker/block/block_store.rs:27-330:
+open/finalize with reservation-prepare-publish and rollback :128-193,
+generation-safe readers :234-248, short-circuit disclosure :253-271;
+crates/adapters/curvine-storage-local/src/vfs_dataset.rs:31-213 multi-dir
+multi-tier dataset with startup scan).
+
+Tier order is hottest-first: HBM (GPU arenas) > MEM (host) > SSD > HDD;
+allocation falls through to colder tiers on capacity pressure.  Metadata
+mutations hold a lock; byte movement happens outside it.
+"""
+from __future__ import annotations
+
+import logging
+import threading
+from typing import Optional
+
+from curvine_amd import errors as err
+from curvine_amd.conf import TIERS, TIER_ORDER, WorkerConf
+from curvine_amd.model import BlockState, StorageInfo
+from curvine_amd.worker.layout import (BlockLayout, BlockReader, BlockWriter,
+                                       make_layout)
+
+log = logging.getLogger("curvine.blockstore")
+
+
+class _Block:
+    __slots__ = ("block_id", "state", "meta", "layout", "generation",
+                 "readers", "pending_delete")
+
+    def __init__(self, block_id: int, layout: BlockLayout, meta: dict):
+        self.block_id = block_id
+        self.state = BlockState.WRITING
+        self.meta = meta
+        self.layout = layout
+        self.generation = 0
+        self.readers = 0
+        self.pending_delete = False
+
+
+class BlockStore:
+    def __init__(self, conf: WorkerConf):
+        self.conf = conf
+        self.layouts: list[BlockLayout] = []
+        for i, dd in enumerate(conf.parsed_dirs()):
+            self.layouts.append(make_layout(dd, i, conf.staging_buf_bytes,
+                                            conf.staging_buf_count))
+        # hottest-first iteration order
+        self.layouts.sort(key=lambda l: TIER_ORDER.get(l.tier, 9))
+        self.blocks: dict[int, _Block] = {}
+        self.lock = threading.Lock()
+        # incremental heartbeat deltas
+        self._added: list[dict] = []
+        self._removed: list[int] = []
+        self.scan()
+
+    # ---------------- startup ----------------
+    def scan(self) -> None:
+        for layout in self.layouts:
+            for meta in layout.scan():
+                b = _Block(meta["block_id"], layout, meta)
+                b.state = BlockState.FINALIZED
+                self.blocks[b.block_id] = b
+                self._added.append({"block_id": b.block_id, "tier": layout.tier})
+        if self.blocks:
+            log.info("block store recovered %d blocks", len(self.blocks))
+
+    # ---------------- write path ----------------
+    def create_writer(self, block_id: int, reserve: int,
+                      tier_hint: str = "") -> BlockWriter:
+        with self.lock:
+            existing = self.blocks.get(block_id)
+            if existing is not None:
+                if existing.state == BlockState.WRITING:
+                    raise err.BlockInWriting(str(block_id))
+                raise err.FileAlreadyExists(f"block {block_id}")
+        # allocation outside the map lock (layouts have their own state)
+        start = 0
+        if tier_hint in TIER_ORDER:
+            start = TIER_ORDER[tier_hint]
+        last_exc: Exception = err.CapacityExceeded("no data dirs")
+        for layout in self.layouts:
+            if TIER_ORDER.get(layout.tier, 9) < start:
+                continue
+            try:
+                meta = layout.allocate(block_id, reserve)
+            except err.CapacityExceeded as e:
+                last_exc = e
+                continue
+            b = _Block(block_id, layout, meta)
+            with self.lock:
+                if block_id in self.blocks:   # raced: rollback
+                    layout.deallocate(meta)
+                    raise err.BlockInWriting(str(block_id))
+                self.blocks[block_id] = b
+            return BlockWriter(layout, block_id, meta)
+        raise last_exc
+
+    def finalize(self, block_id: int, length: int) -> str:
+        """Publish a written block; returns its tier."""
+        with self.lock:
+            b = self.blocks.get(block_id)
+            if b is None:
+                raise err.BlockNotFound(str(block_id))
+            if b.state == BlockState.FINALIZED:
+                return b.layout.tier
+        b.layout.finalize(b.meta, length)
+        with self.lock:
+            b.state = BlockState.FINALIZED
+            b.generation += 1
+            self._added.append({"block_id": block_id, "tier": b.layout.tier})
+        return b.layout.tier
+
+    def abort(self, block_id: int) -> None:
+        with self.lock:
+            b = self.blocks.pop(block_id, None)
+        if b is not None and b.state == BlockState.WRITING:
+            b.layout.deallocate(b.meta)
+        elif b is not None:
+            # was finalized; put it back
+            with self.lock:
+                self.blocks[block_id] = b
+
+    # ---------------- read path ----------------
+    def open_reader(self, block_id: int) -> BlockReader:
+        with self.lock:
+            b = self.blocks.get(block_id)
+            if b is None or b.pending_delete:
+                raise err.BlockNotFound(str(block_id))
+            if b.state != BlockState.FINALIZED:
+                raise err.BlockInWriting(str(block_id))
+            b.readers += 1
+            meta = b.meta
+        reader = BlockReader(b.layout, block_id, meta)
+        store = self
+
+        def close():
+            with store.lock:
+                b.readers -= 1
+                if b.pending_delete and b.readers == 0:
+                    store._do_delete(b)
+        reader.close = close  # type: ignore[method-assign]
+        return reader
+
+    def local_info(self, block_id: int) -> dict:
+        with self.lock:
+            b = self.blocks.get(block_id)
+            if b is None or b.state != BlockState.FINALIZED or b.pending_delete:
+                raise err.BlockNotFound(str(block_id))
+            return b.layout.local_info(b.meta)
+
+    # ---------------- delete ----------------
+    def delete(self, block_id: int) -> None:
+        with self.lock:
+            b = self.blocks.get(block_id)
+            if b is None:
+                return
+            if b.readers > 0:
+                b.pending_delete = True
+                return
+            self._do_delete(b)
+
+    def _do_delete(self, b: _Block) -> None:
+        # caller holds self.lock
+        self.blocks.pop(b.block_id, None)
+        self._removed.append(b.block_id)
+        try:
+            b.layout.deallocate(b.meta)
+        except Exception as e:  # noqa: BLE001
+            log.warning("deallocate block %d: %s", b.block_id, e)
+
+    # ---------------- reporting ----------------
+    def take_deltas(self) -> tuple[list[dict], list[int]]:
+        with self.lock:
+            added, self._added = self._added, []
+            removed, self._removed = self._removed, []
+        return added, removed
+
+    def full_report(self) -> list[dict]:
+        with self.lock:
+            return [{"block_id": b.block_id, "tier": b.layout.tier,
+                     "length": b.meta.get("length", 0)}
+                    for b in self.blocks.values()
+                    if b.state == BlockState.FINALIZED and not b.pending_delete]
+
+    def storages(self) -> list[StorageInfo]:
+        with self.lock:
+            counts: dict[int, int] = {}
+            for b in self.blocks.values():
+                counts[b.layout.dir_id] = counts.get(b.layout.dir_id, 0) + 1
+        return [StorageInfo(tier=l.tier, dir_id=l.dir_id, capacity=l.capacity,
+                            used=l.used, block_num=counts.get(l.dir_id, 0))
+                for l in self.layouts]
+
+    def block_count(self) -> int:
+        with self.lock:
+            return len(self.blocks)
+
+    def close(self) -> None:
+        for l in self.layouts:
+            l.close()

@@ -1,0 +1,202 @@
+"""End-to-end MiniCluster tests: write/read through the full client-worker
+path, tiering, replication, heartbeats, block deletes, load jobs."""
+import asyncio
+import hashlib
+import os
+
+import pytest
+
+from curvine_amd import errors as err
+from curvine_amd.testing import MiniCluster
+
+
+@pytest.fixture
+def loop():
+    loop = asyncio.new_event_loop()
+    asyncio.set_event_loop(loop)
+    yield loop
+    loop.close()
+
+
+def run(loop, coro):
+    return loop.run_until_complete(coro)
+
+
+def test_write_read_small(loop, tmp_path):
+    async def main():
+        async with MiniCluster(tmp_dir=str(tmp_path)) as mc:
+            fs = mc.fs()
+            data = os.urandom(1000)
+            st = await fs.write_all("/hello.bin", data)
+            assert st.length == 1000
+            back = await fs.read_all("/hello.bin")
+            assert back == data
+            await fs.close()
+    run(loop, main())
+
+
+def test_write_read_multiblock(loop, tmp_path):
+    async def main():
+        async with MiniCluster(tmp_dir=str(tmp_path)) as mc:
+            fs = mc.fs()
+            # 3.5 blocks of 4 MB
+            data = os.urandom(14 << 20)
+            await fs.write_all("/big.bin", data)
+            st = await fs.file_status("/big.bin")
+            assert st.length == len(data)
+            back = await fs.read_all("/big.bin")
+            assert hashlib.md5(back).digest() == hashlib.md5(data).digest()
+            # positioned reads across block boundary
+            r = await fs.open("/big.bin")
+            chunk = await r.pread((4 << 20) - 100, 200)
+            assert chunk == data[(4 << 20) - 100:(4 << 20) + 100]
+            r.close()
+            await fs.close()
+    run(loop, main())
+
+
+def test_remote_read_no_short_circuit(loop, tmp_path):
+    async def main():
+        async with MiniCluster(tmp_dir=str(tmp_path)) as mc:
+            conf = mc.client_conf()
+            conf.client.short_circuit = False
+            from curvine_amd.client.filesystem import CurvineFileSystem
+            fs = CurvineFileSystem(conf)
+            data = os.urandom(6 << 20)
+            await fs.write_all("/remote.bin", data)
+            back = await fs.read_all("/remote.bin")
+            assert back == data
+            await fs.close()
+    run(loop, main())
+
+
+def test_tier_fallback_to_ssd(loop, tmp_path):
+    async def main():
+        # MEM dir only 64MB; write 100MB -> must spill to SSD
+        async with MiniCluster(tmp_dir=str(tmp_path)) as mc:
+            fs = mc.fs()
+            data = os.urandom(100 << 20)
+            await fs.write_all("/spill.bin", data, storage_tier="MEM")
+            back = await fs.read_all("/spill.bin")
+            assert back == data
+            tiers = {s.tier: s.used for s in mc.workers[0].store.storages()}
+            assert tiers.get("SSD", 0) > 0
+            await fs.close()
+    run(loop, main())
+
+
+def test_replication_star_write(loop, tmp_path):
+    async def main():
+        async with MiniCluster(tmp_dir=str(tmp_path), workers=3) as mc:
+            fs = mc.fs()
+            data = os.urandom(2 << 20)
+            await fs.write_all("/rep.bin", data, replicas=2)
+            # allow heartbeat to register locations
+            await asyncio.sleep(0.3)
+            fb = await fs.client.open("/rep.bin")
+            assert len(fb.blocks[0].locations) == 2
+            back = await fs.read_all("/rep.bin")
+            assert back == data
+            await fs.close()
+    run(loop, main())
+
+
+def test_delete_propagates_to_worker(loop, tmp_path):
+    async def main():
+        async with MiniCluster(tmp_dir=str(tmp_path)) as mc:
+            fs = mc.fs()
+            await fs.write_all("/gone.bin", os.urandom(1 << 20))
+            await asyncio.sleep(0.25)
+            assert mc.workers[0].store.block_count() == 1
+            await fs.delete("/gone.bin")
+            # worker executes delete on next heartbeat
+            for _ in range(40):
+                await asyncio.sleep(0.1)
+                if mc.workers[0].store.block_count() == 0:
+                    break
+            assert mc.workers[0].store.block_count() == 0
+            await fs.close()
+    run(loop, main())
+
+
+def test_re_replication_after_worker_loss(loop, tmp_path):
+    async def main():
+        async with MiniCluster(tmp_dir=str(tmp_path), workers=3) as mc:
+            mc.conf.master.worker_expire_ms = 600
+            mc.master.fs.workers.expire_ms = 600
+            fs = mc.fs()
+            data = os.urandom(1 << 20)
+            await fs.write_all("/ha.bin", data, replicas=2)
+            await asyncio.sleep(0.3)
+            fb = await fs.client.open("/ha.bin")
+            locs = {a.worker_id for a in fb.blocks[0].locations}
+            assert len(locs) == 2
+            # kill one replica holder
+            victim = next(w for w in mc.workers if w.worker_id in locs)
+            await victim.stop()
+            mc.workers.remove(victim)
+            # wait for expiry + re-replication
+            for _ in range(100):
+                await asyncio.sleep(0.2)
+                fb = await fs.client.open("/ha.bin")
+                live = {a.worker_id for a in fb.blocks[0].locations}
+                if len(live) >= 2 and victim.worker_id not in live:
+                    break
+            assert len(live) >= 2 and victim.worker_id not in live
+            back = await fs.read_all("/ha.bin")
+            assert back == data
+            await fs.close()
+    run(loop, main())
+
+
+def test_load_job_from_local_ufs(loop, tmp_path):
+    async def main():
+        ufs_root = tmp_path / "ufs"
+        (ufs_root / "sub").mkdir(parents=True)
+        payload = {}
+        for i in range(3):
+            data = os.urandom(300_000 + i)
+            (ufs_root / "sub" / f"f{i}.bin").write_bytes(data)
+            payload[f"/sub/f{i}.bin"] = data
+        async with MiniCluster(tmp_dir=str(tmp_path / "cv")) as mc:
+            fs = mc.fs()
+            await fs.mount("/mnt", f"file://{ufs_root}")
+            job = await fs.submit_job("/mnt/sub")
+            for _ in range(100):
+                await asyncio.sleep(0.1)
+                st = await fs.job_status(job["job_id"])
+                if st["state"].startswith("completed"):
+                    break
+            assert st["state"] == "completed", st
+            assert st["done"] == 3
+            for rel, data in payload.items():
+                back = await fs.read_all("/mnt" + rel)
+                assert back == data
+            await fs.close()
+    run(loop, main())
+
+
+def test_worker_restart_rescan(loop, tmp_path):
+    async def main():
+        # SSD-tier blocks survive worker restart (startup scan)
+        dirs = [[f"[SSD:1GB]{tmp_path}/w0/ssd"]]
+        async with MiniCluster(tmp_dir=str(tmp_path), worker_dirs=dirs) as mc:
+            fs = mc.fs()
+            data = os.urandom(2 << 20)
+            await fs.write_all("/persist.bin", data, storage_tier="SSD")
+            w = mc.workers[0]
+            await w.stop()
+            mc.workers.remove(w)
+            from curvine_amd.worker.server import Worker
+            import copy
+            wc = copy.deepcopy(mc.conf)
+            wc.worker.rpc_port = 0
+            wc.worker.data_dirs = dirs[0]
+            w2 = await Worker(wc, worker_id=1).start()
+            mc.workers.append(w2)
+            assert w2.store.block_count() == 1
+            await asyncio.sleep(0.3)   # heartbeat reports recovered blocks
+            back = await fs.read_all("/persist.bin")
+            assert back == data
+            await fs.close()
+    run(loop, main())
