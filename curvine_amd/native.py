@@ -48,6 +48,14 @@ def load():
     with _lock:
         if _mod is not None:
             return _mod
+        # Load torch's bundled HIP runtime first: loading /opt/rocm's
+        # libamdhip64 ahead of torch breaks torch.cuda init ("No HIP GPUs
+        # are available") — observed on the MI355X pool (torch 2.10+rocm7.0
+        # vs /opt/rocm 7.2).  Harmless when torch is absent.
+        try:
+            import torch  # noqa: F401
+        except ImportError:
+            pass
         try:
             from curvine_amd import _native as mod  # type: ignore
         except ImportError:

@@ -67,6 +67,8 @@ class MiniCluster:
             wc.worker.rpc_port = 0
             if self.worker_dirs and i < len(self.worker_dirs):
                 wc.worker.data_dirs = self.worker_dirs[i]
+            elif self.n_workers == 1:
+                pass   # single worker: honour conf.worker.data_dirs as given
             else:
                 wc.worker.data_dirs = [
                     f"[MEM:64MB]{self.tmp_dir}/w{i}/mem",
