@@ -54,11 +54,15 @@ class FsReader:
 
     # ---------------- block reader selection ----------------
     def _block_at(self, off: int) -> tuple[int, LocatedBlock, int]:
-        """(index, located block, offset within block)."""
-        # blocks are equally sized except the last
-        bs = self.status.block_size
-        idx = off // bs
-        if idx >= len(self.fb.blocks):
+        """(index, located block, offset within block).  Blocks can be
+        shorter than block_size mid-file (appends), so search by offset."""
+        import bisect
+        offs = getattr(self, "_block_offs", None)
+        if offs is None or len(offs) != len(self.fb.blocks):
+            offs = [b.offset for b in self.fb.blocks]
+            self._block_offs = offs
+        idx = bisect.bisect_right(offs, off) - 1
+        if idx < 0 or idx >= len(self.fb.blocks):
             raise err.OutOfRange(f"offset {off} beyond {self.length}")
         lb = self.fb.blocks[idx]
         return idx, lb, off - lb.offset
@@ -104,8 +108,8 @@ class FsReader:
         if n == 0:
             return b""
         out = bytearray(n)
-        await self.pread_into(off, out, 0, n)
-        return bytes(out)
+        got = await self.pread_into(off, out, 0, n)
+        return bytes(out[:got])
 
     async def pread_into(self, off: int, out, out_off: int, n: int) -> int:
         n = max(0, min(n, self.length - off))

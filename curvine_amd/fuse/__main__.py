@@ -1,0 +1,70 @@
+"""cv-fuse: standalone FUSE daemon process.
+
+Usage:
+    python -m curvine_amd.fuse --mnt /mnt/curvine --master 127.0.0.1:8995 \
+        [--conf cluster.toml] [--embed-worker] [--device N] [--channels K]
+
+Runs the FUSE daemon in its own process — the production deployment shape
+(and the only deadlock-free one: an in-process daemon starves its own
+mount's FLUSH requests whenever the embedding app spawns subprocesses,
+because fork/vfork suspends the parent with the GIL held while the child's
+pre-exec fd-closing sends FUSE requests).
+
+Prints "READY <mnt>" on stdout once mounted; exits cleanly on SIGTERM.
+"""
+from __future__ import annotations
+
+import argparse
+import logging
+import signal
+import sys
+import threading
+
+from curvine_amd.conf import ClusterConf
+from curvine_amd.fuse.daemon import FuseDaemon
+
+
+def main(argv=None) -> int:
+    p = argparse.ArgumentParser(prog="cv-fuse")
+    p.add_argument("--mnt", required=True)
+    p.add_argument("--master", default=None, help="host:port")
+    p.add_argument("--conf", default=None, help="cluster TOML")
+    p.add_argument("--embed-worker", action="store_true",
+                   help="run a worker (HBM arena owner) inside this daemon")
+    p.add_argument("--device", type=int, default=-1, help="GPU ordinal")
+    p.add_argument("--channels", type=int, default=0)
+    p.add_argument("--data-dir", action="append", default=[],
+                   help='worker data dir, e.g. "[HBM:200GB:0]gpu0"')
+    p.add_argument("--log-level", default="INFO")
+    args = p.parse_args(argv)
+
+    logging.basicConfig(
+        level=getattr(logging, args.log_level.upper(), logging.INFO),
+        format="%(asctime)s %(name)s %(levelname)s %(message)s")
+
+    conf = ClusterConf.from_file(args.conf) if args.conf else ClusterConf()
+    if args.master:
+        conf.client.master_addrs = [args.master]
+        host, _, port = args.master.rpartition(":")
+        conf.master.hostname, conf.master.rpc_port = host, int(port)
+    if args.channels:
+        conf.fuse.mnt_number = args.channels
+    if args.data_dir:
+        conf.worker.data_dirs = args.data_dir
+    conf.worker.rpc_port = 0
+    conf.fuse.mnt_path = args.mnt
+
+    daemon = FuseDaemon(conf, args.mnt, embed_worker=args.embed_worker,
+                        device_id=args.device).start()
+    print(f"READY {args.mnt}", flush=True)
+
+    stop = threading.Event()
+    signal.signal(signal.SIGTERM, lambda *a: stop.set())
+    signal.signal(signal.SIGINT, lambda *a: stop.set())
+    stop.wait()
+    daemon.stop()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -1,0 +1,770 @@
+"""FUSE operations over the curvine client.
+
+Analog of the reference's `CurvineFileSystem` (fuse side,
+/root/reference/curvine-fuse/src/fs/curvine_file_system.rs, 3.7k lines)
+with its dcache (fs/dcache/dir_tree.rs:44-93 ino<->path map), handle state
+(fs/state/), FuseReader/FuseWriter stream bridges (fs/fuse_reader.rs:37-122,
+fuse_writer.rs:52) — re-designed for the MI355X deployment: the FUSE daemon
+embeds the local GPU's worker, so READ on an HBM-cached block is a
+synchronous in-process arena read (hipMemcpyAsync through the pinned ring)
+with no event-loop hop; metadata ops bridge to the client's asyncio loop.
+"""
+from __future__ import annotations
+
+import asyncio
+import errno
+import logging
+import os
+import stat as stat_m
+import threading
+import time
+from typing import Optional
+
+from curvine_amd import errors as cverr
+from curvine_amd.fuse import abi
+from curvine_amd.model import FileStatus, FileType
+
+log = logging.getLogger("curvine.fuse.ops")
+
+
+class Node:
+    __slots__ = ("id", "parent", "name", "nlookup", "children")
+
+    def __init__(self, id: int, parent: int, name: str):
+        self.id = id
+        self.parent = parent
+        self.name = name
+        self.nlookup = 0
+        self.children: dict[str, int] = {}
+
+
+class FileHandle:
+    __slots__ = ("fh", "node_id", "path", "status", "reader", "writer",
+                 "write_pos", "flags", "dir_entries", "lock")
+
+    def __init__(self, fh: int, node_id: int, path: str):
+        self.fh = fh
+        self.node_id = node_id
+        self.path = path
+        self.status: Optional[FileStatus] = None
+        self.reader = None          # SyncReadState
+        self.writer = None          # FsWriter (async, bridged)
+        self.write_pos = 0
+        self.flags = 0
+        self.dir_entries: Optional[list] = None
+        self.lock = threading.Lock()
+
+
+class SyncReadState:
+    """Per-handle block readers resolved once at open; in-process blocks are
+    read synchronously straight from the worker's store."""
+
+    def __init__(self, fs: "CurvineFuseFs", file_blocks):
+        self.fs = fs
+        self.fb = file_blocks
+        self.length = file_blocks.status.length
+        self.block_size = file_blocks.status.block_size
+        self._offs = None                     # block offset index (lazy)
+        self._local: dict[int, object] = {}   # block idx -> store reader
+        self._remote = None                   # lazy async FsReader
+
+    def read_into(self, off: int, out, out_off: int, n: int) -> int:
+        import bisect
+        n = max(0, min(n, self.length - off))
+        if self._offs is None:
+            self._offs = [b.offset for b in self.fb.blocks]
+        got = 0
+        while got < n:
+            idx = bisect.bisect_right(self._offs, off + got) - 1
+            if idx < 0 or idx >= len(self.fb.blocks):
+                break
+            lb = self.fb.blocks[idx]
+            boff = off + got - lb.offset
+            want = min(n - got, lb.block.length - boff)
+            if want <= 0:
+                break
+            r = self._local.get(idx)
+            if r is None:
+                r = self._open_local(idx, lb)
+            if r is not None:
+                got += r.read_into(boff, out, out_off + got, want)
+            else:
+                got += self._read_remote(off + got, out, out_off + got, want)
+        return got
+
+    def _open_local(self, idx: int, lb):
+        from curvine_amd.worker import registry
+        for addr in lb.locations:
+            store = registry.lookup(addr.worker_id)
+            if store is not None:
+                try:
+                    r = store.open_reader(lb.block.block_id)
+                    self._local[idx] = r
+                    return r
+                except Exception:  # noqa: BLE001
+                    continue
+        return None
+
+    def _read_remote(self, off: int, out, out_off: int, n: int) -> int:
+        # remote / hole blocks: bridge to the asyncio FsReader
+        if self._remote is None:
+            from curvine_amd.client.reader import FsReader
+            self._remote = FsReader(self.fs.fs.client, self.fb)
+        buf = bytearray(n)
+        got = self.fs.call(self._remote.pread_into(off, buf, 0, n))
+        out[out_off:out_off + got] = memoryview(buf)[:got]
+        return got
+
+    def close(self) -> None:
+        for r in self._local.values():
+            try:
+                r.close()
+            except Exception:  # noqa: BLE001
+                pass
+        self._local.clear()
+        if self._remote is not None:
+            self._remote.close()
+
+
+class CurvineFuseFs:
+    """Opcode handler table over CurvineFileSystem."""
+
+    def __init__(self, fs, conf, loop: asyncio.AbstractEventLoop):
+        self.fs = fs               # CurvineFileSystem (async)
+        self.conf = conf
+        self.loop = loop
+        self.session = None
+        self.nodes: dict[int, Node] = {1: Node(1, 0, "")}
+        self.nodes_lock = threading.Lock()
+        self.next_node = 2
+        self.handles: dict[int, FileHandle] = {}
+        self.next_fh = 1
+        self.handles_lock = threading.Lock()
+        self.attr_cache: dict[int, tuple[FileStatus, float]] = {}
+        self.attr_ttl = conf.fuse.attr_ttl_ms / 1000.0
+        self.entry_ttl = conf.fuse.entry_ttl_ms / 1000.0
+        # POSIX advisory locks: (node_id) -> list of (start, end, type, owner, pid)
+        self.plocks: dict[int, list] = {}
+        self.plock_mu = threading.Lock()
+
+    # ---------------- helpers ----------------
+    def call(self, coro, timeout: float = 120.0):
+        return asyncio.run_coroutine_threadsafe(coro, self.loop).result(timeout)
+
+    def node_path(self, node_id: int) -> str:
+        with self.nodes_lock:
+            parts = []
+            node = self.nodes.get(node_id)
+            if node is None:
+                raise OSError(errno.ESTALE, "stale nodeid")
+            while node.id != 1:
+                parts.append(node.name)
+                node = self.nodes[node.parent]
+            return "/" + "/".join(reversed(parts))
+
+    def child_node(self, parent_id: int, name: str) -> Node:
+        with self.nodes_lock:
+            parent = self.nodes[parent_id]
+            nid = parent.children.get(name)
+            if nid is not None:
+                return self.nodes[nid]
+            node = Node(self.next_node, parent_id, name)
+            self.next_node += 1
+            self.nodes[node.id] = node
+            parent.children[name] = node.id
+            return node
+
+    def drop_child(self, parent_id: int, name: str) -> None:
+        with self.nodes_lock:
+            parent = self.nodes.get(parent_id)
+            if parent:
+                parent.children.pop(name, None)
+
+    def cached_status(self, node_id: int) -> Optional[FileStatus]:
+        ent = self.attr_cache.get(node_id)
+        if ent and ent[1] > time.monotonic():
+            return ent[0]
+        return None
+
+    def cache_status(self, node_id: int, st: FileStatus) -> None:
+        self.attr_cache[node_id] = (st, time.monotonic() + self.attr_ttl)
+
+    def invalidate(self, node_id: int) -> None:
+        self.attr_cache.pop(node_id, None)
+
+    def stat_path(self, node_id: int, path: str) -> FileStatus:
+        st = self.cached_status(node_id)
+        if st is None:
+            st = self.call(self.fs.file_status(path))
+            self.cache_status(node_id, st)
+        return st
+
+    def entry_out(self, node: Node, st: FileStatus) -> bytes:
+        node.nlookup += 1
+        self.cache_status(node.id, st)
+        attr = abi.pack_attr(self._with_ino(st, node.id))
+        ev = int(self.entry_ttl)
+        evn = int((self.entry_ttl - ev) * 1e9)
+        return abi.ENTRY_OUT.pack(node.id, 0, ev, ev, evn, evn) + attr
+
+    def attr_out(self, node_id: int, st: FileStatus) -> bytes:
+        av = int(self.attr_ttl)
+        avn = int((self.attr_ttl - av) * 1e9)
+        return abi.ATTR_OUT.pack(av, avn, 0) + abi.pack_attr(
+            self._with_ino(st, node_id))
+
+    @staticmethod
+    def _with_ino(st: FileStatus, ino: int) -> FileStatus:
+        import copy
+        st2 = copy.copy(st)
+        st2.inode_id = ino
+        return st2
+
+    def new_handle(self, node_id: int, path: str) -> FileHandle:
+        with self.handles_lock:
+            fh = self.next_fh
+            self.next_fh += 1
+            h = FileHandle(fh, node_id, path)
+            self.handles[fh] = h
+            return h
+
+    def get_handle(self, fh: int) -> FileHandle:
+        h = self.handles.get(fh)
+        if h is None:
+            raise OSError(errno.EBADF, f"fh {fh}")
+        return h
+
+    # ---------------- ops ----------------
+    def op_init(self, nodeid, body, ctx):
+        major, minor, max_readahead, flags = abi.INIT_IN.unpack_from(body, 0)
+        log.info("fuse INIT kernel %d.%d flags=%#x", major, minor, flags)
+        out_flags = (abi.FUSE_ASYNC_READ | abi.FUSE_BIG_WRITES |
+                     abi.FUSE_PARALLEL_DIROPS | abi.FUSE_ATOMIC_O_TRUNC |
+                     abi.FUSE_MAX_PAGES | abi.FUSE_CACHE_SYMLINKS |
+                     abi.FUSE_POSIX_LOCKS) & flags | abi.FUSE_MAX_PAGES
+        max_write = self.conf.fuse.max_write
+        return abi.INIT_OUT.pack(
+            7, min(minor, abi.FUSE_KERNEL_MINOR_VERSION),
+            self.conf.fuse.max_readahead, out_flags,
+            64, 48, max_write, 1,
+            max(1, max_write // 4096), 0, 0, 0)
+
+    def op_destroy(self, nodeid, body, ctx):
+        return b""
+
+    def op_lookup(self, nodeid, body, ctx):
+        name = bytes(body).split(b"\x00", 1)[0].decode()
+        parent_path = self.node_path(nodeid)
+        path = (parent_path.rstrip("/") + "/" + name)
+        try:
+            node = self.child_node(nodeid, name)
+            st = self.stat_path(node.id, path)
+        except cverr.FileNotFound:
+            raise OSError(errno.ENOENT, path)
+        return self.entry_out(node, st)
+
+    def op_forget(self, nodeid, body, ctx):
+        (nlookup,) = abi.FORGET_IN.unpack_from(body, 0)
+        self._forget_one(nodeid, nlookup)
+        return None
+
+    def op_batch_forget(self, nodeid, body, ctx):
+        count, _ = abi.BATCH_FORGET_IN.unpack_from(body, 0)
+        off = abi.BATCH_FORGET_IN.size
+        for _ in range(count):
+            nid, nl = abi.FORGET_ONE.unpack_from(body, off)
+            off += abi.FORGET_ONE.size
+            self._forget_one(nid, nl)
+        return None
+
+    def _forget_one(self, nid: int, nlookup: int) -> None:
+        with self.nodes_lock:
+            node = self.nodes.get(nid)
+            if node is None or nid == 1:
+                return
+            node.nlookup -= nlookup
+            if node.nlookup <= 0 and not node.children:
+                self.nodes.pop(nid, None)
+                parent = self.nodes.get(node.parent)
+                if parent and parent.children.get(node.name) == nid:
+                    parent.children.pop(node.name, None)
+                self.attr_cache.pop(nid, None)
+
+    def op_getattr(self, nodeid, body, ctx):
+        path = self.node_path(nodeid)
+        try:
+            st = self.stat_path(nodeid, path)
+        except cverr.FileNotFound:
+            raise OSError(errno.ENOENT, path)
+        # live write handle: report current write position as size
+        st = self._adjust_writing_size(nodeid, st)
+        return self.attr_out(nodeid, st)
+
+    def _adjust_writing_size(self, nodeid: int, st: FileStatus) -> FileStatus:
+        with self.handles_lock:
+            for h in self.handles.values():
+                if h.node_id == nodeid and h.writer is not None:
+                    import copy
+                    st2 = copy.copy(st)
+                    st2.length = max(st.length, h.write_pos)
+                    return st2
+        return st
+
+    def op_setattr(self, nodeid, body, ctx):
+        (valid, _pad, fh, size, _lo, at, mt, ct, atn, mtn, ctn, mode,
+         _u4, uid, gid, _u5) = abi.SETATTR_IN.unpack_from(body, 0)
+        path = self.node_path(nodeid)
+        attrs = {}
+        if valid & abi.FATTR_MODE:
+            attrs["mode"] = mode & 0o7777
+        if valid & abi.FATTR_UID:
+            attrs["uid"] = uid
+        if valid & abi.FATTR_GID:
+            attrs["gid"] = gid
+        if valid & (abi.FATTR_ATIME | abi.FATTR_ATIME_NOW):
+            attrs["atime_ms"] = int(time.time() * 1000) \
+                if valid & abi.FATTR_ATIME_NOW else at * 1000 + atn // 1_000_000
+        if valid & (abi.FATTR_MTIME | abi.FATTR_MTIME_NOW):
+            attrs["mtime_ms"] = int(time.time() * 1000) \
+                if valid & abi.FATTR_MTIME_NOW else mt * 1000 + mtn // 1_000_000
+        self.invalidate(nodeid)
+        if valid & abi.FATTR_SIZE:
+            st = self._truncate(nodeid, path, size, fh if valid & abi.FATTR_FH else 0)
+        if attrs:
+            st = self.call(self.fs.set_attr(path, **attrs))
+        else:
+            st = self.stat_path(nodeid, path)
+        st = self._adjust_writing_size(nodeid, st)
+        self.cache_status(nodeid, st)
+        return self.attr_out(nodeid, st)
+
+    def _truncate(self, nodeid, path, size, fh) -> FileStatus:
+        # open write handle truncating to its own position: no-op
+        if fh:
+            h = self.handles.get(fh)
+            if h is not None and h.writer is not None and h.write_pos == size:
+                return self.stat_path(nodeid, path)
+        try:
+            cur = self.call(self.fs.file_status(path))
+        except cverr.FileNotFound:
+            raise OSError(errno.ENOENT, path)
+        if size == cur.length:
+            return cur
+        if size < cur.length:
+            return self.call(self.fs.resize(path, size))
+        raise OSError(errno.ENOTSUP, "extending truncate not supported yet")
+
+    def op_mkdir(self, nodeid, body, ctx):
+        mode, _umask = abi.MKDIR_IN.unpack_from(body, 0)
+        name = bytes(body[abi.MKDIR_IN.size:]).split(b"\x00", 1)[0].decode()
+        path = self.node_path(nodeid).rstrip("/") + "/" + name
+        st = self.call(self.fs.mkdir(path, mode & 0o7777, create_parents=False))
+        node = self.child_node(nodeid, name)
+        return self.entry_out(node, st)
+
+    def op_unlink(self, nodeid, body, ctx):
+        name = bytes(body).split(b"\x00", 1)[0].decode()
+        path = self.node_path(nodeid).rstrip("/") + "/" + name
+        self.call(self.fs.delete(path, recursive=False))
+        self.drop_child(nodeid, name)
+        return b""
+
+    def op_rmdir(self, nodeid, body, ctx):
+        name = bytes(body).split(b"\x00", 1)[0].decode()
+        path = self.node_path(nodeid).rstrip("/") + "/" + name
+        st = self.call(self.fs.file_status(path))
+        if not st.is_dir:
+            raise OSError(errno.ENOTDIR, path)
+        self.call(self.fs.delete(path, recursive=False))
+        self.drop_child(nodeid, name)
+        return b""
+
+    def _rename(self, nodeid, newdir, oldname, newname, flags=0):
+        src = self.node_path(nodeid).rstrip("/") + "/" + oldname
+        dst = self.node_path(newdir).rstrip("/") + "/" + newname
+        if flags & 1:   # RENAME_NOREPLACE
+            if self.call(self.fs.exists(dst)):
+                raise OSError(errno.EEXIST, dst)
+        if flags & 2:   # RENAME_EXCHANGE
+            tmp = dst + f".xchg.{time.monotonic_ns()}"
+            self.call(self.fs.rename(dst, tmp))
+            self.call(self.fs.rename(src, dst))
+            self.call(self.fs.rename(tmp, src))
+        else:
+            self.call(self.fs.rename(src, dst))
+        with self.nodes_lock:
+            src_parent = self.nodes.get(nodeid)
+            nid = src_parent.children.pop(oldname, None) if src_parent else None
+            if nid is not None and not flags & 2:
+                node = self.nodes[nid]
+                node.parent = newdir
+                node.name = newname
+                dst_parent = self.nodes.get(newdir)
+                if dst_parent is not None:
+                    old = dst_parent.children.get(newname)
+                    if old is not None:
+                        self.attr_cache.pop(old, None)
+                    dst_parent.children[newname] = nid
+                self.attr_cache.pop(nid, None)
+        return b""
+
+    def op_rename(self, nodeid, body, ctx):
+        (newdir,) = abi.RENAME_IN.unpack_from(body, 0)
+        names = bytes(body[abi.RENAME_IN.size:]).split(b"\x00")
+        return self._rename(nodeid, newdir, names[0].decode(), names[1].decode())
+
+    def op_rename2(self, nodeid, body, ctx):
+        newdir, flags, _ = abi.RENAME2_IN.unpack_from(body, 0)
+        names = bytes(body[abi.RENAME2_IN.size:]).split(b"\x00")
+        return self._rename(nodeid, newdir, names[0].decode(),
+                            names[1].decode(), flags)
+
+    def op_symlink(self, nodeid, body, ctx):
+        parts = bytes(body).split(b"\x00")
+        name, target = parts[0].decode(), parts[1].decode()
+        path = self.node_path(nodeid).rstrip("/") + "/" + name
+        st = self.call(self.fs.symlink(path, target))
+        node = self.child_node(nodeid, name)
+        return self.entry_out(node, st)
+
+    def op_readlink(self, nodeid, body, ctx):
+        st = self.stat_path(nodeid, self.node_path(nodeid))
+        if not st.is_symlink:
+            raise OSError(errno.EINVAL, "not a symlink")
+        return st.symlink_target.encode()
+
+    def op_link(self, nodeid, body, ctx):
+        (oldnode,) = abi.LINK_IN.unpack_from(body, 0)
+        name = bytes(body[abi.LINK_IN.size:]).split(b"\x00", 1)[0].decode()
+        src = self.node_path(oldnode)
+        dst = self.node_path(nodeid).rstrip("/") + "/" + name
+        st = self.call(self.fs.link(src, dst))
+        node = self.child_node(nodeid, name)
+        self.invalidate(oldnode)
+        return self.entry_out(node, st)
+
+    def op_mknod(self, nodeid, body, ctx):
+        mode, rdev, _umask, _ = abi.MKNOD_IN.unpack_from(body, 0)
+        if not stat_m.S_ISREG(mode):
+            raise OSError(errno.ENOTSUP, "only regular files")
+        name = bytes(body[abi.MKNOD_IN.size:]).split(b"\x00", 1)[0].decode()
+        path = self.node_path(nodeid).rstrip("/") + "/" + name
+        w = self.call(self.fs.create(path, overwrite=False))
+        st = self.call(w.complete())
+        node = self.child_node(nodeid, name)
+        return self.entry_out(node, st)
+
+    # ---------------- open/create/read/write ----------------
+    def op_create(self, nodeid, body, ctx):
+        flags, mode, _umask, _of = abi.CREATE_IN.unpack_from(body, 0)
+        name = bytes(body[abi.CREATE_IN.size:]).split(b"\x00", 1)[0].decode()
+        path = self.node_path(nodeid).rstrip("/") + "/" + name
+        excl = bool(flags & os.O_EXCL)
+        writer = self.call(self.fs.create(path, overwrite=not excl))
+        h = self.new_handle(0, path)
+        h.writer = writer
+        h.flags = flags
+        h.status = writer.status
+        node = self.child_node(nodeid, name)
+        h.node_id = node.id
+        self.invalidate(node.id)
+        entry = self.entry_out(node, writer.status)
+        open_out = abi.OPEN_OUT.pack(h.fh, 0, 0)
+        return entry + open_out
+
+    def op_open(self, nodeid, body, ctx):
+        flags, _ = abi.OPEN_IN.unpack_from(body, 0)
+        path = self.node_path(nodeid)
+        accmode = flags & os.O_ACCMODE
+        h = self.new_handle(nodeid, path)
+        h.flags = flags
+        try:
+            if accmode == os.O_RDONLY:
+                fb = self.call(self.fs.client.open(path))
+                h.status = fb.status
+                h.reader = SyncReadState(self, fb)
+            else:
+                if flags & os.O_TRUNC:
+                    h.writer = self.call(self.fs.create(path, overwrite=True))
+                    h.write_pos = 0
+                elif flags & os.O_APPEND or accmode in (os.O_WRONLY, os.O_RDWR):
+                    st = self.call(self.fs.file_status(path))
+                    if st.length == 0:
+                        h.writer = self.call(self.fs.create(path, overwrite=True))
+                    else:
+                        h.writer = self.call(self.fs.append(path))
+                    h.write_pos = h.writer.pos
+                h.status = h.writer.status
+                if accmode == os.O_RDWR:
+                    try:
+                        fb = self.call(self.fs.client.open(path))
+                        h.reader = SyncReadState(self, fb)
+                    except cverr.FsError:
+                        pass
+            self.invalidate(nodeid)
+            return abi.OPEN_OUT.pack(h.fh, 0, 0)
+        except Exception:
+            with self.handles_lock:
+                self.handles.pop(h.fh, None)
+            raise
+
+    def op_read(self, nodeid, body, ctx):
+        fh, offset, size, _rf, _lo, _fl, _ = abi.READ_IN.unpack_from(body, 0)
+        h = self.get_handle(fh)
+        if h.reader is None:
+            # O_WRONLY handle read, or reader not yet available
+            raise OSError(errno.EBADF, "not open for read")
+        buf = bytearray(size)
+        n = h.reader.read_into(offset, buf, 0, size)
+        return memoryview(buf)[:n]
+
+    def op_write(self, nodeid, body, ctx):
+        fh, offset, size, _wf, _lo, _fl, _ = abi.WRITE_IN.unpack_from(body, 0)
+        h = self.get_handle(fh)
+        data = bytes(body[abi.WRITE_IN.size:abi.WRITE_IN.size + size])
+        with h.lock:
+            if h.writer is None:
+                raise OSError(errno.EBADF, "not open for write")
+            if offset != h.write_pos:
+                raise OSError(errno.ENOTSUP,
+                              f"non-sequential write at {offset} (pos {h.write_pos})")
+            self.call(h.writer.write(data))
+            h.write_pos += len(data)
+        return abi.WRITE_OUT.pack(len(data), 0)
+
+    def op_flush(self, nodeid, body, ctx):
+        fh, _u, _p, _lo = abi.FLUSH_IN.unpack_from(body, 0)
+        h = self.handles.get(fh)
+        if h is not None and h.writer is not None:
+            with h.lock:
+                self.call(h.writer.flush())
+        return b""
+
+    def op_fsync(self, nodeid, body, ctx):
+        return b""
+
+    def op_release(self, nodeid, body, ctx):
+        fh, _f, _rf, _lo = abi.RELEASE_IN.unpack_from(body, 0)
+        with self.handles_lock:
+            h = self.handles.pop(fh, None)
+        if h is None:
+            return b""
+        if h.writer is not None:
+            with h.lock:
+                st = self.call(h.writer.complete())
+                self.cache_status(h.node_id, st)
+        if h.reader is not None:
+            h.reader.close()
+        with self.plock_mu:
+            self.plocks.pop(fh, None)
+        self.invalidate(h.node_id)
+        return b""
+
+    # ---------------- dirs ----------------
+    def op_opendir(self, nodeid, body, ctx):
+        path = self.node_path(nodeid)
+        h = self.new_handle(nodeid, path)
+        return abi.OPEN_OUT.pack(h.fh, 0, 0)
+
+    def op_readdir(self, nodeid, body, ctx):
+        fh, offset, size, _rf, _lo, _fl, _ = abi.READ_IN.unpack_from(body, 0)
+        h = self.get_handle(fh)
+        if h.dir_entries is None:
+            sts = self.call(self.fs.list_status(h.path))
+            entries = [(".", None), ("..", None)]
+            entries += [(s.name, s) for s in sts]
+            h.dir_entries = entries
+        out = bytearray()
+        idx = offset
+        while idx < len(h.dir_entries):
+            name, st = h.dir_entries[idx]
+            if st is None:
+                ino, dtype = nodeid, abi.DT_DIR
+            else:
+                node = self.child_node(h.node_id, name)
+                ino = node.id
+                dtype = abi.DT_DIR if st.is_dir else (
+                    abi.DT_LNK if st.is_symlink else abi.DT_REG)
+            ent = abi.pack_dirent(ino, idx + 1, name.encode(), dtype)
+            if len(out) + len(ent) > size:
+                break
+            out += ent
+            idx += 1
+        return bytes(out)
+
+    def op_releasedir(self, nodeid, body, ctx):
+        fh, _f, _rf, _lo = abi.RELEASE_IN.unpack_from(body, 0)
+        with self.handles_lock:
+            self.handles.pop(fh, None)
+        return b""
+
+    def op_fsyncdir(self, nodeid, body, ctx):
+        return b""
+
+    # ---------------- misc ----------------
+    def op_statfs(self, nodeid, body, ctx):
+        info = self.call(self.fs.get_master_info())
+        bsize = 4096
+        blocks = max(1, info.get("capacity", 0) // bsize)
+        bfree = max(0, (info.get("capacity", 0) - info.get("used", 0)) // bsize)
+        return abi.KSTATFS.pack(blocks, bfree, bfree,
+                                1 << 30, (1 << 30) - info.get("inode_num", 0),
+                                bsize, 255, bsize, 0)
+
+    def op_access(self, nodeid, body, ctx):
+        return b""
+
+    def op_getxattr(self, nodeid, body, ctx):
+        size, _ = abi.GETXATTR_IN.unpack_from(body, 0)
+        name = bytes(body[abi.GETXATTR_IN.size:]).split(b"\x00", 1)[0].decode()
+        st = self.stat_path(nodeid, self.node_path(nodeid))
+        val = st.xattrs.get(name)
+        if val is None:
+            raise OSError(errno.ENODATA, name)
+        val = bytes(val)
+        if size == 0:
+            return abi.GETXATTR_OUT.pack(len(val), 0)
+        if len(val) > size:
+            raise OSError(errno.ERANGE, name)
+        return val
+
+    def op_listxattr(self, nodeid, body, ctx):
+        size, _ = abi.GETXATTR_IN.unpack_from(body, 0)
+        st = self.stat_path(nodeid, self.node_path(nodeid))
+        blob = b"".join(k.encode() + b"\x00" for k in st.xattrs)
+        if size == 0:
+            return abi.GETXATTR_OUT.pack(len(blob), 0)
+        if len(blob) > size:
+            raise OSError(errno.ERANGE, "listxattr")
+        return blob
+
+    def op_setxattr(self, nodeid, body, ctx):
+        vsize, _flags = abi.SETXATTR_IN.unpack_from(body, 0)
+        rest = bytes(body[abi.SETXATTR_IN.size:])
+        name, _, tail = rest.partition(b"\x00")
+        value = tail[:vsize]
+        path = self.node_path(nodeid)
+        self.call(self.fs.set_attr(path, xattrs={name.decode(): value}))
+        self.invalidate(nodeid)
+        return b""
+
+    def op_removexattr(self, nodeid, body, ctx):
+        name = bytes(body).split(b"\x00", 1)[0].decode()
+        path = self.node_path(nodeid)
+        st = self.stat_path(nodeid, path)
+        if name not in st.xattrs:
+            raise OSError(errno.ENODATA, name)
+        self.call(self.fs.set_attr(path, xattrs={name: None}))
+        self.invalidate(nodeid)
+        return b""
+
+    def op_fallocate(self, nodeid, body, ctx):
+        fh, offset, length, mode, _ = abi.FALLOCATE_IN.unpack_from(body, 0)
+        if mode != 0:
+            raise OSError(errno.ENOTSUP, "fallocate mode")
+        return b""
+
+    def op_lseek(self, nodeid, body, ctx):
+        fh, offset, whence, _ = abi.LSEEK_IN.unpack_from(body, 0)
+        h = self.get_handle(fh)
+        length = h.status.length if h.status else 0
+        SEEK_DATA, SEEK_HOLE = 3, 4
+        if whence == SEEK_DATA:
+            if offset >= length:
+                raise OSError(errno.ENXIO, "past eof")
+            return abi.LSEEK_OUT.pack(offset)
+        if whence == SEEK_HOLE:
+            return abi.LSEEK_OUT.pack(length)
+        raise OSError(errno.EINVAL, "whence")
+
+    def op_interrupt(self, nodeid, body, ctx):
+        return None
+
+    # ---------------- POSIX advisory locks ----------------
+    F_RDLCK, F_WRLCK, F_UNLCK = 0, 1, 2
+
+    def _lock_conflicts(self, node_id, start, end, ltype, owner):
+        for (s, e, t, o, _pid) in self.plocks.get(node_id, []):
+            if o == owner:
+                continue
+            if s <= end and start <= e and (t == self.F_WRLCK or
+                                            ltype == self.F_WRLCK):
+                return (s, e, t, o)
+        return None
+
+    def op_getlk(self, nodeid, body, ctx):
+        fh, owner, start, end, ltype, pid, _fl, _ = abi.LK_IN.unpack_from(body, 0)
+        with self.plock_mu:
+            c = self._lock_conflicts(nodeid, start, end or (1 << 63), ltype, owner)
+        if c is None:
+            return abi.LK_OUT.pack(0, 0, self.F_UNLCK, 0)
+        return abi.LK_OUT.pack(c[0], c[1], c[2], 0)
+
+    def op_setlk(self, nodeid, body, ctx, wait=False):
+        fh, owner, start, end, ltype, pid, _fl, _ = abi.LK_IN.unpack_from(body, 0)
+        end = end or (1 << 63)
+        deadline = time.monotonic() + 30 if wait else 0
+        while True:
+            with self.plock_mu:
+                if ltype == self.F_UNLCK:
+                    locks = self.plocks.get(nodeid, [])
+                    self.plocks[nodeid] = [
+                        l for l in locks
+                        if not (l[3] == owner and l[0] <= end and start <= l[1])]
+                    return b""
+                c = self._lock_conflicts(nodeid, start, end, ltype, owner)
+                if c is None:
+                    self.plocks.setdefault(nodeid, []).append(
+                        (start, end, ltype, owner, pid))
+                    return b""
+            if not wait or time.monotonic() > deadline:
+                raise OSError(errno.EAGAIN, "lock conflict")
+            time.sleep(0.02)
+
+    def op_setlkw(self, nodeid, body, ctx):
+        return self.op_setlk(nodeid, body, ctx, wait=True)
+
+    HANDLERS = {}
+
+
+CurvineFuseFs.HANDLERS = {
+    abi.Op.INIT: CurvineFuseFs.op_init,
+    abi.Op.DESTROY: CurvineFuseFs.op_destroy,
+    abi.Op.LOOKUP: CurvineFuseFs.op_lookup,
+    abi.Op.FORGET: CurvineFuseFs.op_forget,
+    abi.Op.BATCH_FORGET: CurvineFuseFs.op_batch_forget,
+    abi.Op.GETATTR: CurvineFuseFs.op_getattr,
+    abi.Op.SETATTR: CurvineFuseFs.op_setattr,
+    abi.Op.MKDIR: CurvineFuseFs.op_mkdir,
+    abi.Op.UNLINK: CurvineFuseFs.op_unlink,
+    abi.Op.RMDIR: CurvineFuseFs.op_rmdir,
+    abi.Op.RENAME: CurvineFuseFs.op_rename,
+    abi.Op.RENAME2: CurvineFuseFs.op_rename2,
+    abi.Op.SYMLINK: CurvineFuseFs.op_symlink,
+    abi.Op.READLINK: CurvineFuseFs.op_readlink,
+    abi.Op.LINK: CurvineFuseFs.op_link,
+    abi.Op.MKNOD: CurvineFuseFs.op_mknod,
+    abi.Op.CREATE: CurvineFuseFs.op_create,
+    abi.Op.OPEN: CurvineFuseFs.op_open,
+    abi.Op.READ: CurvineFuseFs.op_read,
+    abi.Op.WRITE: CurvineFuseFs.op_write,
+    abi.Op.FLUSH: CurvineFuseFs.op_flush,
+    abi.Op.FSYNC: CurvineFuseFs.op_fsync,
+    abi.Op.RELEASE: CurvineFuseFs.op_release,
+    abi.Op.OPENDIR: CurvineFuseFs.op_opendir,
+    abi.Op.READDIR: CurvineFuseFs.op_readdir,
+    abi.Op.RELEASEDIR: CurvineFuseFs.op_releasedir,
+    abi.Op.FSYNCDIR: CurvineFuseFs.op_fsyncdir,
+    abi.Op.STATFS: CurvineFuseFs.op_statfs,
+    abi.Op.ACCESS: CurvineFuseFs.op_access,
+    abi.Op.GETXATTR: CurvineFuseFs.op_getxattr,
+    abi.Op.LISTXATTR: CurvineFuseFs.op_listxattr,
+    abi.Op.SETXATTR: CurvineFuseFs.op_setxattr,
+    abi.Op.REMOVEXATTR: CurvineFuseFs.op_removexattr,
+    abi.Op.FALLOCATE: CurvineFuseFs.op_fallocate,
+    abi.Op.LSEEK: CurvineFuseFs.op_lseek,
+    abi.Op.INTERRUPT: CurvineFuseFs.op_interrupt,
+    abi.Op.GETLK: CurvineFuseFs.op_getlk,
+    abi.Op.SETLK: CurvineFuseFs.op_setlk,
+    abi.Op.SETLKW: CurvineFuseFs.op_setlkw,
+}

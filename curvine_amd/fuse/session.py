@@ -1,0 +1,182 @@
+"""FUSE session: pure mount (no libfuse), channels, dispatch loop.
+
+Analog of the reference's libfuse-free mount + session
+(/root/reference/curvine-fuse/src/raw/fuse_pure.rs:17-160 mount via
+libc::mount with fd=..., session/fuse_session.rs:154-260 multi-channel run
+loop, session/channel/fuse_receiver.rs dispatch).  Channels are cloned
+with FUSE_DEV_IOC_CLONE (clone_fd analog) and each gets an OS thread; the
+hot READ/WRITE path never touches the asyncio loop when the target block
+lives in this process (HBM short-circuit).
+"""
+from __future__ import annotations
+
+import ctypes
+import errno
+import fcntl
+import logging
+import os
+import struct
+import threading
+from typing import Optional
+
+from curvine_amd.fuse import abi
+
+log = logging.getLogger("curvine.fuse")
+
+libc = ctypes.CDLL("libc.so.6", use_errno=True)
+FUSE_DEV_IOC_CLONE = 0x8004E500
+MNT_DETACH = 2
+
+
+def mount_fuse(mnt_path: str, allow_other: bool = True,
+               default_permissions: bool = False,
+               rootmode: int = 0o40755) -> int:
+    """Open /dev/fuse and mount it at mnt_path; returns the session fd."""
+    os.makedirs(mnt_path, exist_ok=True)
+    fd = os.open("/dev/fuse", os.O_RDWR)
+    opts = f"fd={fd},rootmode={rootmode & 0o170000:o},user_id=0,group_id=0"
+    if allow_other:
+        opts += ",allow_other"
+    if default_permissions:
+        opts += ",default_permissions"
+    ret = libc.mount(b"curvinefs", mnt_path.encode(), b"fuse.curvinefs",
+                     0, opts.encode())
+    if ret != 0:
+        e = ctypes.get_errno()
+        os.close(fd)
+        raise OSError(e, f"fuse mount at {mnt_path}: {os.strerror(e)}")
+    return fd
+
+
+def clone_channel(session_fd: int) -> int:
+    """FUSE_DEV_IOC_CLONE: a second queue fd on the same session."""
+    fd = os.open("/dev/fuse", os.O_RDWR)
+    buf = struct.pack("I", session_fd)
+    fcntl.ioctl(fd, FUSE_DEV_IOC_CLONE, buf)
+    return fd
+
+
+def umount(mnt_path: str) -> None:
+    libc.umount2(mnt_path.encode(), MNT_DETACH)
+
+
+class FuseChannel(threading.Thread):
+    """One /dev/fuse queue consumer."""
+
+    def __init__(self, session: "FuseSession", fd: int, idx: int):
+        super().__init__(daemon=True, name=f"fuse-ch{idx}")
+        self.session = session
+        self.fd = fd
+        self.idx = idx
+        self.bufsize = session.max_write + (64 << 10)
+        self.reply_buf = bytearray(self.bufsize)
+
+    def run(self) -> None:
+        fs = self.session.fs
+        while not self.session.stopped:
+            try:
+                req = os.read(self.fd, self.bufsize)
+            except OSError as e:
+                if e.errno == errno.EINTR:
+                    continue
+                if e.errno in (errno.ENODEV, errno.EBADF):
+                    break   # unmounted
+                if e.errno == errno.ENOENT:
+                    continue  # request aborted before we read it
+                log.error("fuse read ch%d: %s", self.idx, e)
+                break
+            if not req:
+                break
+            try:
+                self.dispatch(req)
+            except Exception as e:  # noqa: BLE001
+                log.exception("fuse dispatch failed: %s", e)
+        log.info("fuse channel %d exiting", self.idx)
+
+    def dispatch(self, req: bytes) -> None:
+        (length, opcode, unique, nodeid, uid, gid, pid, _extlen, _pad) = \
+            abi.IN_HEADER.unpack_from(req, 0)
+        body = memoryview(req)[abi.IN_HEADER_SIZE:length]
+        handler = self.session.fs.HANDLERS.get(opcode)
+        if handler is None:
+            log.debug("fuse op %s unimplemented",
+                      abi.Op.NAMES.get(opcode, opcode))
+            self.reply_error(unique, errno.ENOSYS)
+            return
+        try:
+            result = handler(self.session.fs, nodeid, body,
+                             (uid, gid, pid, unique, self))
+        except OSError as e:
+            self.reply_error(unique, e.errno or errno.EIO)
+            return
+        except Exception as e:  # noqa: BLE001
+            from curvine_amd.errors import to_errno
+            eno = to_errno(e)
+            if eno == errno.EIO:
+                log.warning("fuse op %s error: %s",
+                            abi.Op.NAMES.get(opcode, opcode), e)
+            self.reply_error(unique, eno)
+            return
+        if result is None:
+            return   # no reply (FORGET, INTERRUPT, or handler replied itself)
+        self.reply(unique, result)
+
+    # ---------------- replies ----------------
+    def reply(self, unique: int, body) -> None:
+        if isinstance(body, (list, tuple)):
+            parts = body
+        else:
+            parts = [body]
+        total = abi.OUT_HEADER_SIZE + sum(len(p) for p in parts)
+        hdr = abi.OUT_HEADER.pack(total, 0, unique)
+        try:
+            os.writev(self.fd, [hdr, *parts])
+        except OSError as e:
+            if e.errno not in (errno.ENOENT, errno.ENODEV):
+                log.warning("fuse reply failed: %s", e)
+
+    def reply_error(self, unique: int, eno: int) -> None:
+        hdr = abi.OUT_HEADER.pack(abi.OUT_HEADER_SIZE, -eno, unique)
+        try:
+            os.write(self.fd, hdr)
+        except OSError:
+            pass
+
+
+class FuseSession:
+    def __init__(self, fs, mnt_path: str, channels: int = 1,
+                 max_write: int = 1 << 20, allow_other: bool = True):
+        self.fs = fs
+        self.mnt_path = mnt_path
+        self.n_channels = max(1, channels)
+        self.max_write = max_write
+        self.stopped = False
+        self.session_fd = -1
+        self.channels: list[FuseChannel] = []
+        fs.session = self
+
+    def start(self) -> "FuseSession":
+        self.session_fd = mount_fuse(self.mnt_path,
+                                     allow_other=True)
+        fds = [self.session_fd]
+        for i in range(1, self.n_channels):
+            fds.append(clone_channel(self.session_fd))
+        for i, fd in enumerate(fds):
+            ch = FuseChannel(self, fd, i)
+            self.channels.append(ch)
+            ch.start()
+        log.info("fuse mounted at %s (%d channels)", self.mnt_path,
+                 len(self.channels))
+        return self
+
+    def stop(self) -> None:
+        self.stopped = True
+        umount(self.mnt_path)
+        for ch in self.channels:
+            try:
+                os.close(ch.fd)
+            except OSError:
+                pass
+        for ch in self.channels:
+            ch.join(timeout=3)
+        self.channels = []

@@ -104,3 +104,45 @@ class MiniCluster:
 
     async def __aexit__(self, *a) -> None:
         await self.stop()
+
+
+class SyncMiniCluster:
+    """MiniCluster running on a dedicated event-loop thread, for callers
+    that are not themselves async (FUSE tests, CLI tests, benchmarks)."""
+
+    def __init__(self, **kw):
+        import threading
+        self.loop = asyncio.new_event_loop()
+        self._thread = threading.Thread(target=self._run, daemon=True,
+                                        name="curvine-minicluster")
+        self._thread.start()
+        self.mc = MiniCluster(**kw)
+
+    def _run(self):
+        asyncio.set_event_loop(self.loop)
+        self.loop.run_forever()
+
+    def call(self, coro, timeout: float = 120.0):
+        return asyncio.run_coroutine_threadsafe(coro, self.loop).result(timeout)
+
+    def start(self) -> "SyncMiniCluster":
+        self.call(self.mc.start())
+        return self
+
+    def stop(self) -> None:
+        try:
+            self.call(self.mc.stop(), timeout=30)
+        finally:
+            self.loop.call_soon_threadsafe(self.loop.stop)
+            self._thread.join(timeout=5)
+
+    def client_conf(self) -> ClusterConf:
+        return self.mc.client_conf()
+
+    @property
+    def workers(self):
+        return self.mc.workers
+
+    @property
+    def master(self):
+        return self.mc.master

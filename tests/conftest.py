@@ -1,9 +1,18 @@
 import os
+import subprocess
 import sys
 
 import pytest
 
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+# FUSE tests run the daemon IN the pytest process (convenient for asserts).
+# glibc posix_spawn suspends the vfork'ing thread WITH the GIL held while the
+# child execs; the child's exec closes inherited CLOEXEC fds on the fuse
+# mount, which sends FLUSH to our daemon — whose channel thread then can't
+# take the GIL: a deadlock triangle.  Plain fork avoids the suspension.
+# (Production deployments run the daemon as its own process: cv-fuse.)
+subprocess._USE_POSIX_SPAWN = False
 
 
 def pytest_configure(config):
