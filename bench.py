@@ -1,15 +1,17 @@
 #!/usr/bin/env python3
-"""Flagship benchmark: cached sequential-read GB/s on MI355X HBM tiers.
+"""Flagship benchmark: FUSE sequential-read GB/s (and 4K random IOPS) on
+MI355X HBM cache tiers — the BASELINE.json metric.
 
-Measures the BASELINE.json metric (FUSE/client sequential-read throughput
-on synthetic random-byte files cached in HBM) on N GPUs of one node —
-weak scaling: each rank owns one GPU's worker (HBM arena) and reads its
-own shard through the cache client (short-circuit HBM -> pinned -> host),
-fio-style with T concurrent streams of 1 MiB reads.
+Per rank (one per GPU, weak scaling): rank 0 hosts the master; every rank
+spawns a cv-fuse daemon process embedding the worker that owns its GPU's
+HBM arena, writes a synthetic random-byte dataset through the mount, then
+times fio-style reads through the kernel mount (default: T threads x 1 MiB
+sequential; --workload randread4k for the IOPS path).
 
-Driver contract: rank 0 prints ONE JSON line; timing brackets are
+Driver contract: rank 0 prints ONE JSON line; timing is bracketed by
 barrier + torch.cuda.synchronize on both sides; value is the whole-job
-aggregate GiB/s (max elapsed over ranks).
+aggregate (max elapsed over ranks).  --path client bypasses FUSE and reads
+via the in-process client (plumbing mode / containers without /dev/fuse).
 """
 from __future__ import annotations
 
@@ -17,27 +19,48 @@ import argparse
 import asyncio
 import json
 import os
+import subprocess
 import sys
 import tempfile
+import threading
 import time
 
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
-REFERENCE_PEAK_GIBPS = 9.5   # 256KB seq read, 32 threads
+REFERENCE_PEAK_GIBPS = 9.5     # 256 KiB seq read, 32 threads
+REFERENCE_PEAK_RAND_GIBPS = 9.0
 
 
 def log(rank, *a):
     print(f"[bench r{rank}]", *a, file=sys.stderr, flush=True)
 
 
-async def setup_cluster(args, rank, world, dist):
-    """Rank 0: master. Every rank: one worker on its GPU + client."""
-    from curvine_amd.testing import test_conf
-    from curvine_amd.master.server import Master
-    from curvine_amd.worker.server import Worker
-    from curvine_amd.client.filesystem import CurvineFileSystem
+class ClusterRuntime:
+    """Master (rank0) + optional in-process worker + client on a dedicated
+    asyncio loop thread."""
 
-    tmp = tempfile.mkdtemp(prefix=f"curvine-bench-r{rank}-")
+    def __init__(self):
+        self.loop = asyncio.new_event_loop()
+        self._t = threading.Thread(target=self._run, daemon=True)
+        self._t.start()
+        self.master = None
+        self.worker = None
+        self.fs = None
+
+    def _run(self):
+        asyncio.set_event_loop(self.loop)
+        self.loop.run_forever()
+
+    def call(self, coro, timeout=600.0):
+        return asyncio.run_coroutine_threadsafe(coro, self.loop).result(timeout)
+
+    def stop(self):
+        self.loop.call_soon_threadsafe(self.loop.stop)
+        self._t.join(timeout=5)
+
+
+def build_conf(args, tmp, rank):
+    from curvine_amd.testing import test_conf
     conf = test_conf(tmp)
     conf.master.block_size = args.block_size
     conf.client.block_size = args.block_size
@@ -47,14 +70,23 @@ async def setup_cluster(args, rank, world, dist):
     conf.client.read_chunk_size = args.read_chunk
     conf.worker.staging_buf_bytes = args.staging_bytes
     conf.worker.staging_buf_count = args.staging_count
+    conf.fuse.mnt_number = args.fuse_channels
+    return conf
 
-    master = None
+
+def setup(args, rank, world, dist, has_gpu):
+    from curvine_amd.master.server import Master
+    from curvine_amd.worker.server import Worker
+    from curvine_amd.client.filesystem import CurvineFileSystem
+
+    tmp = tempfile.mkdtemp(prefix=f"curvine-bench-r{rank}-")
+    conf = build_conf(args, tmp, rank)
+    rt = ClusterRuntime()
+    master_port = 0
     if rank == 0:
         conf.master.rpc_port = 0
-        master = await Master(conf).start()
-        master_port = master.rpc.port
-    else:
-        master_port = 0
+        rt.master = rt.call(Master(conf).start())
+        master_port = rt.master.rpc.port
     if world > 1:
         obj = [master_port]
         dist.broadcast_object_list(obj, src=0)
@@ -63,65 +95,177 @@ async def setup_cluster(args, rank, world, dist):
     conf.master.rpc_port = master_port
 
     local_rank = int(os.environ.get("LOCAL_RANK", rank))
-    from curvine_amd import native
-    hbm_gb = args.hbm_gb
-    if native.gpu_available():
-        conf.worker.data_dirs = [f"[HBM:{hbm_gb}GB:{local_rank}]gpu{local_rank}"]
+    if has_gpu:
+        data_dirs = [f"[HBM:{args.hbm_gb}GB:{local_rank}]gpu{local_rank}"]
     else:
-        # CPU fallback for plumbing runs in the dev container
-        conf.worker.data_dirs = [f"[MEM:{hbm_gb}GB]{tmp}/mem"]
-    conf.worker.rpc_port = 0
-    worker = await Worker(conf, worker_id=rank + 1, device_id=local_rank).start()
+        data_dirs = [f"[MEM:{args.hbm_gb}GB]{tmp}/mem"]
 
-    fs = CurvineFileSystem(conf)
-    fs.client.local_worker_id = worker.worker_id
-    return master, worker, fs, conf
+    daemon_proc, mnt = None, None
+    if args.path == "fuse":
+        mnt = f"/tmp/curvine-bench-mnt-r{rank}"
+        daemon_proc = subprocess.Popen(
+            [sys.executable, "-m", "curvine_amd.fuse", "--mnt", mnt,
+             "--master", f"127.0.0.1:{master_port}",
+             "--embed-worker", "--device", str(local_rank if has_gpu else -1),
+             "--channels", str(args.fuse_channels),
+             "--log-level", "WARNING"] +
+            [f"--data-dir={d}" for d in data_dirs],
+            stdout=subprocess.PIPE, stderr=sys.stderr, text=True,
+            cwd=os.path.dirname(os.path.abspath(__file__)))
+        line = daemon_proc.stdout.readline()
+        if not line.startswith("READY"):
+            raise RuntimeError(f"cv-fuse failed to start: {line!r}")
+    else:
+        conf.worker.data_dirs = data_dirs
+        conf.worker.rpc_port = 0
+        rt.worker = rt.call(Worker(conf, worker_id=rank + 1,
+                                   device_id=local_rank).start())
+
+    async def mkfs():
+        return CurvineFileSystem(conf)
+    rt.fs = rt.call(mkfs())
+    if rt.worker is not None:
+        rt.fs.client.local_worker_id = rt.worker.worker_id
+    return rt, conf, daemon_proc, mnt
 
 
-async def write_dataset(args, rank, fs):
-    """Synthetic random-byte files, one directory per rank."""
+def write_dataset(args, rank, rt, mnt):
     import numpy as np
     rng = np.random.default_rng(1234 + rank)
-    base_chunk = rng.integers(0, 256, size=8 << 20, dtype=np.uint8).tobytes()
-    for i in range(args.files):
-        w = await fs.create(f"/bench/r{rank}/f{i}", overwrite=True,
-                            storage_tier="HBM")
-        remaining = args.file_size
-        while remaining > 0:
-            n = min(len(base_chunk), remaining)
-            await w.write(base_chunk[:n] if n < len(base_chunk) else base_chunk)
-            remaining -= n
-        await w.complete()
+    base = rng.integers(0, 256, size=8 << 20, dtype=np.uint8).tobytes()
+    if mnt is not None:
+        os.makedirs(f"{mnt}/bench/r{rank}", exist_ok=True)
+        for i in range(args.files):
+            with open(f"{mnt}/bench/r{rank}/f{i}", "wb") as f:
+                remaining = args.file_size
+                while remaining > 0:
+                    n = min(len(base), remaining)
+                    f.write(base[:n])
+                    remaining -= n
+    else:
+        async def write_one(i):
+            w = await rt.fs.create(f"/bench/r{rank}/f{i}", overwrite=True,
+                                   storage_tier="HBM")
+            remaining = args.file_size
+            while remaining > 0:
+                n = min(len(base), remaining)
+                await w.write(base[:n] if n < len(base) else base)
+                remaining -= n
+            await w.complete()
+        for i in range(args.files):
+            rt.call(write_one(i))
 
 
-async def one_step(args, rank, fs) -> int:
-    """Read the whole rank-local dataset with T concurrent streams of
-    `read_chunk` sequential reads. Returns bytes read."""
-    total = 0
-    sem = asyncio.Semaphore(args.threads)
-    results = []
+# ---------------------------------------------------------------------------
+# workloads
+# ---------------------------------------------------------------------------
 
-    async def read_file(i):
-        async with sem:
-            r = await fs.open(f"/bench/r{rank}/f{i}")
-            got, pos = 0, 0
+def step_fuse_seq(args, rank, mnt) -> int:
+    """T threads, each sequentially reading its share of files in
+    read_chunk chunks through the kernel mount."""
+    paths = [f"{mnt}/bench/r{rank}/f{i}" for i in range(args.files)]
+    total = [0] * args.threads
+    errs = []
+
+    def worker(t):
+        try:
             buf = bytearray(args.read_chunk)
-            while pos < r.length:
-                n = await r.pread_into(pos, buf, 0, min(args.read_chunk,
-                                                        r.length - pos))
-                if n <= 0:
-                    break
-                pos += n
-                got += n
-            r.close()
-            return got
+            mv = memoryview(buf)
+            for i in range(t, len(paths), args.threads):
+                fd = os.open(paths[i], os.O_RDONLY)
+                try:
+                    while True:
+                        n = os.readv(fd, [mv])
+                        if n <= 0:
+                            break
+                        total[t] += n
+                finally:
+                    os.close(fd)
+        except Exception as e:  # noqa: BLE001
+            errs.append(e)
 
-    results = await asyncio.gather(*[read_file(i) for i in range(args.files)])
-    total = sum(results)
+    threads = [threading.Thread(target=worker, args=(t,))
+               for t in range(args.threads)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    if errs:
+        raise errs[0]
+    got = sum(total)
     expect = args.files * args.file_size
-    if total != expect:
-        raise RuntimeError(f"step read {total} != {expect}")
-    return total
+    if got != expect:
+        raise RuntimeError(f"step read {got} != {expect}")
+    return got
+
+
+def step_fuse_rand4k(args, rank, mnt, lat_out: list) -> int:
+    """T threads x N random 4 KiB preads (the IOPS path)."""
+    paths = [f"{mnt}/bench/r{rank}/f{i}" for i in range(args.files)]
+    import random
+    per_thread = args.rand_reads // args.threads
+    total = [0] * args.threads
+    lats: list[list[float]] = [[] for _ in range(args.threads)]
+    errs = []
+
+    def worker(t):
+        try:
+            rng = random.Random(t * 7919 + rank)
+            fds = [os.open(p, os.O_RDONLY) for p in paths]
+            try:
+                for _ in range(per_thread):
+                    fd = fds[rng.randrange(len(fds))]
+                    off = rng.randrange(max(1, args.file_size - 4096))
+                    t0 = time.perf_counter_ns()
+                    data = os.pread(fd, 4096, off)
+                    lats[t].append((time.perf_counter_ns() - t0) / 1000.0)
+                    total[t] += len(data)
+            finally:
+                for fd in fds:
+                    os.close(fd)
+        except Exception as e:  # noqa: BLE001
+            errs.append(e)
+
+    threads = [threading.Thread(target=worker, args=(t,))
+               for t in range(args.threads)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    if errs:
+        raise errs[0]
+    alll = sorted(x for l in lats for x in l)
+    if alll:
+        lat_out.append({"p50_us": alll[len(alll) // 2],
+                        "p99_us": alll[int(len(alll) * 0.99)]})
+    return sum(total)
+
+
+def step_client_seq(args, rank, rt) -> int:
+    async def run_step():
+        sem = asyncio.Semaphore(args.threads)
+
+        async def read_file(i):
+            async with sem:
+                r = await rt.fs.open(f"/bench/r{rank}/f{i}")
+                got, pos = 0, 0
+                buf = bytearray(args.read_chunk)
+                while pos < r.length:
+                    n = await r.pread_into(pos, buf, 0,
+                                           min(args.read_chunk, r.length - pos))
+                    if n <= 0:
+                        break
+                    pos += n
+                    got += n
+                r.close()
+                return got
+        res = await asyncio.gather(*[read_file(i) for i in range(args.files)])
+        return sum(res)
+    got = rt.call(run_step())
+    expect = args.files * args.file_size
+    if got != expect:
+        raise RuntimeError(f"step read {got} != {expect}")
+    return got
 
 
 def main():
@@ -129,14 +273,19 @@ def main():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=5)
     p.add_argument("--warmup", type=int, default=2)
+    p.add_argument("--path", choices=["fuse", "client"], default="fuse")
+    p.add_argument("--workload", choices=["seqread", "randread4k"],
+                   default="seqread")
     p.add_argument("--files", type=int, default=8)
     p.add_argument("--file-size", type=int, default=1 << 30)
     p.add_argument("--block-size", type=int, default=256 << 20)
     p.add_argument("--read-chunk", type=int, default=1 << 20)
-    p.add_argument("--threads", type=int, default=8)
+    p.add_argument("--threads", type=int, default=16)
+    p.add_argument("--rand-reads", type=int, default=200_000)
     p.add_argument("--hbm-gb", type=int, default=16)
     p.add_argument("--staging-bytes", type=int, default=8 << 20)
     p.add_argument("--staging-count", type=int, default=8)
+    p.add_argument("--fuse-channels", type=int, default=8)
     args = p.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -147,16 +296,20 @@ def main():
     if world > 1:
         import torch.distributed as dist_mod
         dist = dist_mod
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
-        dist.init_process_group(backend=backend)
+        dist.init_process_group(
+            "nccl" if torch.cuda.is_available() else "gloo")
         if torch.cuda.is_available():
             torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank)))
-
     has_gpu = torch.cuda.is_available()
-    # size the dataset to the machine: CPU plumbing runs use small files
+
+    if not os.path.exists("/dev/fuse") or os.geteuid() != 0:
+        if args.path == "fuse":
+            log(rank, "no /dev/fuse; falling back to client path")
+            args.path = "client"
     if not has_gpu and args.file_size > 64 << 20:
-        args.files, args.file_size = 4, 32 << 20
-        args.hbm_gb = 1
+        args.files, args.file_size = 4, 32 << 20   # CPU plumbing scale
+        args.hbm_gb = 2
+        args.rand_reads = 20_000
 
     def barrier_sync():
         if dist is not None:
@@ -164,28 +317,32 @@ def main():
         if has_gpu:
             torch.cuda.synchronize()
 
-    loop = asyncio.new_event_loop()
-    asyncio.set_event_loop(loop)
-
-    master, worker, fs, conf = loop.run_until_complete(
-        setup_cluster(args, rank, world, dist))
-    log(rank, f"cluster up; writing {args.files}x{args.file_size >> 20}MiB")
+    rt, conf, daemon_proc, mnt = setup(args, rank, world, dist, has_gpu)
+    log(rank, f"cluster up ({args.path}); writing "
+        f"{args.files}x{args.file_size >> 20}MiB")
     t0 = time.perf_counter()
-    loop.run_until_complete(write_dataset(args, rank, fs))
+    write_dataset(args, rank, rt, mnt)
     log(rank, f"dataset written in {time.perf_counter() - t0:.1f}s")
 
-    for _ in range(args.warmup):
-        loop.run_until_complete(one_step(args, rank, fs))
+    lat_out: list = []
 
+    def one_step():
+        if args.path == "fuse" and args.workload == "seqread":
+            return step_fuse_seq(args, rank, mnt)
+        if args.path == "fuse":
+            return step_fuse_rand4k(args, rank, mnt, lat_out)
+        return step_client_seq(args, rank, rt)
+
+    for _ in range(args.warmup):
+        one_step()
     barrier_sync()
     t_start = time.perf_counter()
     bytes_per_step = 0
     for _ in range(args.steps):
-        bytes_per_step = loop.run_until_complete(one_step(args, rank, fs))
+        bytes_per_step = one_step()
     barrier_sync()
     elapsed = time.perf_counter() - t_start
 
-    # max elapsed over ranks
     if dist is not None:
         t = torch.tensor([elapsed], dtype=torch.float64)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
@@ -195,40 +352,63 @@ def main():
     gibps = total_bytes / elapsed / (1 << 30)
 
     if rank == 0:
+        if args.workload == "randread4k":
+            iops = (args.rand_reads * args.steps * world) / elapsed
+            metric, value, unit = "fuse_rand_read_4k_IOPS", round(iops, 1), "IOPS"
+            vs = round(gibps / REFERENCE_PEAK_RAND_GIBPS, 3)
+        else:
+            metric = ("fuse_seq_read_GiBps" if args.path == "fuse"
+                      else "cached_seq_read_GiBps")
+            value, unit = round(gibps, 3), "GiB/s"
+            vs = round(gibps / REFERENCE_PEAK_GIBPS, 3)
         result = {
-            "metric": "cached_seq_read_GiBps",
-            "value": round(gibps, 3),
-            "unit": "GiB/s",
+            "metric": metric,
+            "value": value,
+            "unit": unit,
             "n_gpus": world,
             "steps": args.steps,
             "warmup": args.warmup,
             "ms_per_step": round(elapsed / args.steps * 1000, 2),
             "higher_is_better": True,
             "scaling": "weak",
-            "vs_baseline": round(gibps / REFERENCE_PEAK_GIBPS, 3),
+            "vs_baseline": vs,
             "dtype": "bytes",
-            "data": "synthetic random-byte files (numpy PRNG), cached in HBM tier",
+            "data": "synthetic random-byte files (numpy PRNG), cached in "
+                    + ("HBM tier" if has_gpu else "MEM tier (cpu plumbing)"),
             "config": {
                 "model": "curvine-amd cache engine",
-                "workload": "fio-style sequential read, cached",
-                "path": "client_short_circuit",
-                "tier": "HBM" if has_gpu else "MEM(cpu plumbing)",
+                "workload": args.workload,
+                "path": args.path,
+                "tier": "HBM" if has_gpu else "MEM(cpu)",
                 "files_per_rank": args.files,
                 "file_size": args.file_size,
                 "read_chunk": args.read_chunk,
                 "threads": args.threads,
                 "block_size": args.block_size,
+                "fuse_channels": args.fuse_channels,
+                "GiBps": round(gibps, 3),
+                "latency": lat_out[-1] if lat_out else None,
                 "parallelism": f"shard-per-gpu x{world}",
             },
         }
         print(json.dumps(result), flush=True)
 
+    # teardown
+    if daemon_proc is not None:
+        daemon_proc.terminate()
+        try:
+            daemon_proc.wait(timeout=10)
+        except subprocess.TimeoutExpired:
+            daemon_proc.kill()
     async def teardown():
-        await fs.close()
-        await worker.stop()
-        if master:
-            await master.stop()
-    loop.run_until_complete(teardown())
+        if rt.fs:
+            await rt.fs.close()
+        if rt.worker:
+            await rt.worker.stop()
+        if rt.master:
+            await rt.master.stop()
+    rt.call(teardown())
+    rt.stop()
     if dist is not None:
         dist.destroy_process_group()
 

@@ -85,8 +85,11 @@ FUSE_READDIRPLUS_AUTO = 1 << 14
 FUSE_ASYNC_DIO = 1 << 15
 FUSE_WRITEBACK_CACHE = 1 << 16
 FUSE_PARALLEL_DIROPS = 1 << 18
+FUSE_HANDLE_KILLPRIV = 1 << 19
 FUSE_MAX_PAGES = 1 << 22
 FUSE_CACHE_SYMLINKS = 1 << 23
+FUSE_HANDLE_KILLPRIV_V2 = 1 << 28   # kernel skips security.capability xattr
+                                    # lookups before every WRITE
 
 # headers
 IN_HEADER = struct.Struct("<IIQQIIIHH")      # len opcode unique nodeid uid gid pid total_extlen pad

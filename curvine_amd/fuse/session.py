@@ -30,11 +30,12 @@ MNT_DETACH = 2
 
 def mount_fuse(mnt_path: str, allow_other: bool = True,
                default_permissions: bool = False,
-               rootmode: int = 0o40755) -> int:
+               rootmode: int = 0o40755, max_read: int = 1 << 20) -> int:
     """Open /dev/fuse and mount it at mnt_path; returns the session fd."""
     os.makedirs(mnt_path, exist_ok=True)
     fd = os.open("/dev/fuse", os.O_RDWR)
-    opts = f"fd={fd},rootmode={rootmode & 0o170000:o},user_id=0,group_id=0"
+    opts = (f"fd={fd},rootmode={rootmode & 0o170000:o},user_id=0,group_id=0"
+            f",max_read={max_read}")
     if allow_other:
         opts += ",allow_other"
     if default_permissions:
@@ -46,6 +47,29 @@ def mount_fuse(mnt_path: str, allow_other: bool = True,
         os.close(fd)
         raise OSError(e, f"fuse mount at {mnt_path}: {os.strerror(e)}")
     return fd
+
+
+def tune_readahead(mnt_path: str, readahead_kb: int) -> None:
+    """Raise the mount's BDI read_ahead_kb so the kernel pipelines larger
+    readahead windows (fuse_session.rs:154-190 max_readahead_kb analog).
+
+    The device id comes from /proc/self/mountinfo, NOT from stat(mnt): the
+    daemon statting its own mount before serving requests deadlocks.
+    """
+    try:
+        dev = None
+        with open("/proc/self/mountinfo") as f:
+            for line in f:
+                parts = line.split()
+                if len(parts) > 4 and parts[4] == mnt_path:
+                    dev = parts[2]   # "major:minor"
+        if dev is None:
+            return
+        with open(f"/sys/class/bdi/{dev}/read_ahead_kb", "w") as f:
+            f.write(str(readahead_kb))
+        log.info("bdi %s read_ahead_kb=%d", dev, readahead_kb)
+    except OSError as e:
+        log.debug("bdi readahead tuning failed: %s", e)
 
 
 def clone_channel(session_fd: int) -> int:
@@ -156,8 +180,10 @@ class FuseSession:
         fs.session = self
 
     def start(self) -> "FuseSession":
-        self.session_fd = mount_fuse(self.mnt_path,
-                                     allow_other=True)
+        self.session_fd = mount_fuse(self.mnt_path, allow_other=True,
+                                     max_read=self.max_write)
+        tune_readahead(self.mnt_path,
+                       getattr(self.fs.conf.fuse, "max_readahead", 8 << 20) >> 10)
         fds = [self.session_fd]
         for i in range(1, self.n_channels):
             fds.append(clone_channel(self.session_fd))
