@@ -157,6 +157,8 @@ static int device_count() {
 // Arena
 // ---------------------------------------------------------------------------
 
+#define SPOOL 8   // parallel copy streams per device arena
+
 struct Arena {
   void* base = nullptr;
   size_t cap = 0;
@@ -164,6 +166,11 @@ struct Arena {
   bool pinned = false;   // host arena allocated with hipHostMalloc
   hipStream_t stream = nullptr;       // copy stream
   hipStream_t kstream = nullptr;      // kernel stream
+  // stream pool: concurrent readers each grab a slot so large copies
+  // don't serialize on one stream/lock
+  hipStream_t spool[SPOOL] = {};
+  std::mutex spool_mu[SPOOL];
+  std::atomic<uint32_t> srr{0};
   // pinned staging ring (device arenas)
   std::vector<void*> pin;
   std::vector<hipEvent_t> ev;
@@ -201,6 +208,8 @@ static int arena_create(int device, size_t cap, size_t staging_bytes,
     HIP_CHECK(hipMalloc(&a->base, cap));
     HIP_CHECK(hipStreamCreateWithFlags(&a->stream, hipStreamNonBlocking));
     HIP_CHECK(hipStreamCreateWithFlags(&a->kstream, hipStreamNonBlocking));
+    for (int i = 0; i < SPOOL; ++i)
+      HIP_CHECK(hipStreamCreateWithFlags(&a->spool[i], hipStreamNonBlocking));
     a->pin_sz = staging_bytes;
     a->pin.resize(staging_count);
     a->ev.resize(staging_count);
@@ -241,6 +250,8 @@ static void arena_destroy(int h) {
     hipFree(a->base);
     hipStreamDestroy(a->stream);
     hipStreamDestroy(a->kstream);
+    for (int i = 0; i < SPOOL; ++i)
+      if (a->spool[i]) hipStreamDestroy(a->spool[i]);
   } else if (a->pinned) {
     hipHostFree(a->base);
   } else {
@@ -347,20 +358,23 @@ static void arena_write(int h, uint64_t off, py::buffer buf, uint64_t buf_off,
   else std::memcpy((uint8_t*)a->base + off, src, n);
 }
 
-// raw-pointer variants (torch tensors, other arenas' memory)
+// raw-pointer variants (torch tensors, pinned reply buffers, other arenas'
+// memory).  Uses the stream POOL: concurrent FUSE channel threads each get
+// their own stream, so large D2H copies overlap instead of serializing.
 static void arena_read_ptr(int h, uint64_t off, uintptr_t dst, uint64_t n,
                            bool dst_is_device) {
   Arena* a = get_arena(h);
   check_range(a, off, n);
   py::gil_scoped_release rel;
   if (a->is_dev()) {
-    std::lock_guard<std::mutex> g(a->mu);
+    uint32_t slot = a->srr.fetch_add(1) % SPOOL;
+    std::lock_guard<std::mutex> g(a->spool_mu[slot]);
     HIP_CHECK(hipSetDevice(a->device));
     HIP_CHECK(hipMemcpyAsync((void*)dst, (uint8_t*)a->base + off, n,
                              dst_is_device ? hipMemcpyDeviceToDevice
                                            : hipMemcpyDeviceToHost,
-                             a->stream));
-    HIP_CHECK(hipStreamSynchronize(a->stream));
+                             a->spool[slot]));
+    HIP_CHECK(hipStreamSynchronize(a->spool[slot]));
   } else if (dst_is_device) {
     HIP_CHECK(hipMemcpy((void*)dst, (uint8_t*)a->base + off, n,
                         hipMemcpyHostToDevice));
@@ -375,13 +389,14 @@ static void arena_write_ptr(int h, uint64_t off, uintptr_t src, uint64_t n,
   check_range(a, off, n);
   py::gil_scoped_release rel;
   if (a->is_dev()) {
-    std::lock_guard<std::mutex> g(a->mu);
+    uint32_t slot = a->srr.fetch_add(1) % SPOOL;
+    std::lock_guard<std::mutex> g(a->spool_mu[slot]);
     HIP_CHECK(hipSetDevice(a->device));
     HIP_CHECK(hipMemcpyAsync((uint8_t*)a->base + off, (const void*)src, n,
                              src_is_device ? hipMemcpyDeviceToDevice
                                            : hipMemcpyHostToDevice,
-                             a->stream));
-    HIP_CHECK(hipStreamSynchronize(a->stream));
+                             a->spool[slot]));
+    HIP_CHECK(hipStreamSynchronize(a->spool[slot]));
   } else if (src_is_device) {
     HIP_CHECK(hipMemcpy((uint8_t*)a->base + off, (const void*)src, n,
                         hipMemcpyDeviceToHost));
@@ -593,6 +608,54 @@ static uintptr_t arena_base_ptr(int h) {
   return (uintptr_t)get_arena(h)->base;
 }
 
+// ---------------------------------------------------------------------------
+// Pinned host buffers (FUSE reply/request buffers: DMA lands directly in
+// the buffer that is writev'd to /dev/fuse — no staging-ring hop)
+// ---------------------------------------------------------------------------
+
+struct PinnedBuf { void* ptr = nullptr; size_t n = 0; bool pinned = false; };
+static std::vector<PinnedBuf> g_pins;
+static std::mutex g_pins_mu;
+
+static int pinned_alloc(size_t n) {
+  PinnedBuf b;
+  b.n = n;
+  if (device_count() > 0) {
+    HIP_CHECK(hipHostMalloc(&b.ptr, n, hipHostMallocDefault));
+    b.pinned = true;
+  } else {
+    b.ptr = std::malloc(n);
+    if (!b.ptr) throw std::bad_alloc();
+  }
+  std::lock_guard<std::mutex> g(g_pins_mu);
+  for (size_t i = 0; i < g_pins.size(); ++i)
+    if (!g_pins[i].ptr) { g_pins[i] = b; return (int)i; }
+  g_pins.push_back(b);
+  return (int)g_pins.size() - 1;
+}
+
+static void pinned_free(int id) {
+  std::lock_guard<std::mutex> g(g_pins_mu);
+  if (id < 0 || id >= (int)g_pins.size() || !g_pins[id].ptr) return;
+  if (g_pins[id].pinned) hipHostFree(g_pins[id].ptr);
+  else std::free(g_pins[id].ptr);
+  g_pins[id] = PinnedBuf{};
+}
+
+static py::memoryview pinned_view(int id) {
+  std::lock_guard<std::mutex> g(g_pins_mu);
+  if (id < 0 || id >= (int)g_pins.size() || !g_pins[id].ptr)
+    throw std::runtime_error("bad pinned buffer id");
+  return py::memoryview::from_memory(g_pins[id].ptr, g_pins[id].n, false);
+}
+
+static uintptr_t pinned_ptr(int id) {
+  std::lock_guard<std::mutex> g(g_pins_mu);
+  if (id < 0 || id >= (int)g_pins.size() || !g_pins[id].ptr)
+    throw std::runtime_error("bad pinned buffer id");
+  return (uintptr_t)g_pins[id].ptr;
+}
+
 static py::dict arena_info(int h) {
   Arena* a = get_arena(h);
   py::dict d;
@@ -639,6 +702,10 @@ PYBIND11_MODULE(_native, m) {
   m.def("arena_gather", &arena_gather);
   m.def("arena_base_ptr", &arena_base_ptr);
   m.def("arena_info", &arena_info);
+  m.def("pinned_alloc", &pinned_alloc);
+  m.def("pinned_free", &pinned_free);
+  m.def("pinned_view", &pinned_view);
+  m.def("pinned_ptr", &pinned_ptr);
   m.def("crc32c", &crc32c_buf, py::arg("buf"), py::arg("init") = 0);
   m.def("crc32c_combine", &crc32c_combine);
   m.attr("CRC_SUB") = CRC_SUB;

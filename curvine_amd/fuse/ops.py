@@ -105,6 +105,35 @@ class SyncReadState:
                     continue
         return None
 
+    def read_into_px(self, off: int, ptr: int, view, n: int) -> int:
+        """read_into variant for a pinned destination buffer: HBM-resident
+        blocks DMA directly to `ptr` (hipMemcpyAsync D2H onto a pooled
+        stream); file/remote blocks fall back to the view."""
+        import bisect
+        n = max(0, min(n, self.length - off))
+        if self._offs is None:
+            self._offs = [b.offset for b in self.fb.blocks]
+        got = 0
+        while got < n:
+            idx = bisect.bisect_right(self._offs, off + got) - 1
+            if idx < 0 or idx >= len(self.fb.blocks):
+                break
+            lb = self.fb.blocks[idx]
+            boff = off + got - lb.offset
+            want = min(n - got, lb.block.length - boff)
+            if want <= 0:
+                break
+            r = self._local.get(idx)
+            if r is None:
+                r = self._open_local(idx, lb)
+            if r is not None and r.meta.get("kind") == "arena":
+                got += r.read_to_ptr(boff, ptr + got, want, False)
+            elif r is not None:
+                got += r.read_into(boff, view, got, want)
+            else:
+                got += self._read_remote(off + got, view, got, want)
+        return got
+
     def _read_remote(self, off: int, out, out_off: int, n: int) -> int:
         # remote / hole blocks: bridge to the asyncio FsReader
         if self._remote is None:
@@ -152,23 +181,28 @@ class SyncWriteState:
         self._cur_pos = 0
         self._done = False
 
-    def write(self, data) -> int:
+    def write(self, data, ptr: int | None = None) -> int:
+        """`ptr`, when given, is the raw address of `data` inside a pinned
+        buffer — HBM-bound bytes then DMA host->device directly."""
         data = memoryview(data)
         total = len(data)
-        while len(data) > 0:
+        consumed = 0
+        while consumed < total:
             if self._cur is None and self._cur_async is None:
                 self._next_block()
             room = self.block_size - self._cur_pos
-            take = min(room, len(data))
-            chunk = data[:take]
+            take = min(room, total - consumed)
+            chunk = data[consumed:consumed + take]
             if self._cur is not None:
-                self._cur.write(chunk, take)
+                if ptr is not None and self._cur.meta.get("kind") == "arena":
+                    self._cur.write_from_ptr(ptr + consumed, take, False)
+                else:
+                    self._cur.write(chunk, take)
             else:
-                import asyncio as _a
                 payload = bytes(chunk)
                 self.fs.call(_gather_writes(self._cur_async, payload))
             self._cur_pos += take
-            data = data[take:]
+            consumed += take
             if self._cur_pos >= self.block_size:
                 self._commit_block()
         self.pos += total
@@ -632,6 +666,12 @@ class CurvineFuseFs:
         if h.reader is None:
             # O_WRONLY handle read, or reader not yet available
             raise OSError(errno.EBADF, "not open for read")
+        ch = ctx[4]
+        pin = getattr(ch, "reply_pin", None)
+        if pin is not None and size <= pin.nbytes - 64 and \
+                hasattr(h.reader, "read_into_px"):
+            n = h.reader.read_into_px(offset, pin.ptr, pin.view, size)
+            return pin.view[:n]
         buf = bytearray(size)
         n = h.reader.read_into(offset, buf, 0, size)
         return memoryview(buf)[:n]
@@ -639,16 +679,20 @@ class CurvineFuseFs:
     def op_write(self, nodeid, body, ctx):
         fh, offset, size, _wf, _lo, _fl, _ = abi.WRITE_IN.unpack_from(body, 0)
         h = self.get_handle(fh)
-        data = bytes(body[abi.WRITE_IN.size:abi.WRITE_IN.size + size])
+        data = body[abi.WRITE_IN.size:abi.WRITE_IN.size + size]
+        ch = ctx[4]
+        req = getattr(ch, "req_buf", None)
+        ptr = (req.ptr + abi.IN_HEADER_SIZE + abi.WRITE_IN.size
+               if req is not None else None)
         with h.lock:
             if h.writer is None:
                 raise OSError(errno.EBADF, "not open for write")
             if offset != h.write_pos:
                 raise OSError(errno.ENOTSUP,
                               f"non-sequential write at {offset} (pos {h.write_pos})")
-            h.writer.write(data)
-            h.write_pos += len(data)
-        return abi.WRITE_OUT.pack(len(data), 0)
+            h.writer.write(data, ptr=ptr)
+            h.write_pos += size
+        return abi.WRITE_OUT.pack(size, 0)
 
     def op_flush(self, nodeid, body, ctx):
         fh, _u, _p, _lo = abi.FLUSH_IN.unpack_from(body, 0)

@@ -93,13 +93,18 @@ class FuseChannel(threading.Thread):
         self.fd = fd
         self.idx = idx
         self.bufsize = session.max_write + (64 << 10)
-        self.reply_buf = bytearray(self.bufsize)
+        # pinned request/reply buffers: WRITE payloads DMA host->HBM straight
+        # from the request buffer; READ replies DMA HBM->host straight into
+        # the buffer handed to writev — no staging hop either way.
+        from curvine_amd.native import PinnedBuffer
+        self.req_buf = PinnedBuffer(self.bufsize)
+        self.reply_pin = PinnedBuffer(self.bufsize)
 
     def run(self) -> None:
-        fs = self.session.fs
+        req_view = self.req_buf.view
         while not self.session.stopped:
             try:
-                req = os.read(self.fd, self.bufsize)
+                n = os.readv(self.fd, [req_view])
             except OSError as e:
                 if e.errno == errno.EINTR:
                     continue
@@ -109,10 +114,10 @@ class FuseChannel(threading.Thread):
                     continue  # request aborted before we read it
                 log.error("fuse read ch%d: %s", self.idx, e)
                 break
-            if not req:
+            if n <= 0:
                 break
             try:
-                self.dispatch(req)
+                self.dispatch(req_view[:n])
             except Exception as e:  # noqa: BLE001
                 log.exception("fuse dispatch failed: %s", e)
         log.info("fuse channel %d exiting", self.idx)

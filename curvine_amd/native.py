@@ -145,6 +145,32 @@ class Arena:
             pass
 
 
+class PinnedBuffer:
+    """hipHostMalloc'd buffer (plain malloc without a GPU): DMA-fast host
+    memory exposed as a writable memoryview.  Used for FUSE request/reply
+    buffers so device reads land directly in the bytes handed to writev."""
+
+    def __init__(self, nbytes: int):
+        self._n = load()
+        self.nbytes = nbytes
+        self.id = self._n.pinned_alloc(nbytes)
+        self.view = self._n.pinned_view(self.id)
+        self.ptr = self._n.pinned_ptr(self.id)
+        self._closed = False
+
+    def close(self) -> None:
+        if not self._closed:
+            self.view = None
+            self._n.pinned_free(self.id)
+            self._closed = True
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
+
+
 def crc32c(buf, init: int = 0) -> int:
     return load().crc32c(buf, init)
 
