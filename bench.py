@@ -241,6 +241,40 @@ def step_fuse_rand4k(args, rank, mnt, lat_out: list) -> int:
     return sum(total)
 
 
+def step_client_rand4k(args, rank, rt, lat_out: list) -> int:
+    """Random 4 KiB preads through the client short-circuit path."""
+    import random
+
+    async def run_step():
+        readers = []
+        for i in range(args.files):
+            readers.append(await rt.fs.open(f"/bench/r{rank}/f{i}"))
+        lats: list[float] = []
+        total = 0
+        rng = random.Random(rank)
+
+        async def one_reader(t):
+            nonlocal total
+            buf = bytearray(4096)
+            r2 = random.Random(t * 7919 + rank)
+            for _ in range(args.rand_reads // args.threads):
+                r = readers[r2.randrange(len(readers))]
+                off = r2.randrange(max(1, r.length - 4096))
+                t0 = time.perf_counter_ns()
+                n = await r.pread_into(off, buf, 0, 4096)
+                lats.append((time.perf_counter_ns() - t0) / 1000.0)
+                total += n
+        await asyncio.gather(*[one_reader(t) for t in range(args.threads)])
+        for r in readers:
+            r.close()
+        lats.sort()
+        if lats:
+            lat_out.append({"p50_us": lats[len(lats) // 2],
+                            "p99_us": lats[int(len(lats) * 0.99)]})
+        return total
+    return rt.call(run_step())
+
+
 def step_client_seq(args, rank, rt) -> int:
     async def run_step():
         sem = asyncio.Semaphore(args.threads)
@@ -302,9 +336,28 @@ def main():
             torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank)))
     has_gpu = torch.cuda.is_available()
 
-    if not os.path.exists("/dev/fuse") or os.geteuid() != 0:
-        if args.path == "fuse":
-            log(rank, "no /dev/fuse; falling back to client path")
+    if args.path == "fuse" and not os.path.exists("/dev/fuse") \
+            and os.geteuid() == 0:
+        # GPU pool boxes ship without the device node; create it (char 10:229)
+        try:
+            os.mknod("/dev/fuse", 0o666 | 0o020000, os.makedev(10, 229))
+            log(rank, "created /dev/fuse")
+        except OSError as e:
+            log(rank, f"mknod /dev/fuse failed: {e}")
+    if args.path == "fuse":
+        ok = False
+        if os.path.exists("/dev/fuse") and os.geteuid() == 0:
+            try:  # probe a real mount: the node may exist without the module
+                from curvine_amd.fuse.session import mount_fuse, umount
+                probe = f"/tmp/curvine-fuse-probe-{os.getpid()}"
+                fd = mount_fuse(probe)
+                umount(probe)
+                os.close(fd)
+                ok = True
+            except OSError as e:
+                log(rank, f"fuse probe failed: {e}")
+        if not ok:
+            log(rank, "FUSE unavailable; falling back to client path")
             args.path = "client"
     if not has_gpu and args.file_size > 64 << 20:
         args.files, args.file_size = 4, 32 << 20   # CPU plumbing scale
@@ -327,10 +380,12 @@ def main():
     lat_out: list = []
 
     def one_step():
-        if args.path == "fuse" and args.workload == "seqread":
-            return step_fuse_seq(args, rank, mnt)
-        if args.path == "fuse":
+        if args.workload == "randread4k":
+            if args.path != "fuse":
+                return step_client_rand4k(args, rank, rt, lat_out)
             return step_fuse_rand4k(args, rank, mnt, lat_out)
+        if args.path == "fuse":
+            return step_fuse_seq(args, rank, mnt)
         return step_client_seq(args, rank, rt)
 
     for _ in range(args.warmup):
@@ -353,7 +408,7 @@ def main():
 
     if rank == 0:
         if args.workload == "randread4k":
-            iops = (args.rand_reads * args.steps * world) / elapsed
+            iops = (total_bytes / 4096) / elapsed
             metric, value, unit = "fuse_rand_read_4k_IOPS", round(iops, 1), "IOPS"
             vs = round(gibps / REFERENCE_PEAK_RAND_GIBPS, 3)
         else:

@@ -109,6 +109,10 @@ class FsClient:
         h = await self._rpc(RpcCode.ResizeFile, {"path": path, "length": length})
         return FileStatus.from_dict(h["status"])
 
+    async def get_block_locations(self, path: str) -> FileBlocks:
+        h = await self._rpc(RpcCode.GetBlockLocations, {"path": path})
+        return FileBlocks.from_dict(h["file_blocks"])
+
     async def free(self, path: str, recursive: bool = False) -> int:
         h = await self._rpc(RpcCode.Free, {"path": path, "recursive": recursive})
         return h.get("freed_blocks", 0)
