@@ -409,7 +409,8 @@ def main():
     if rank == 0:
         if args.workload == "randread4k":
             iops = (total_bytes / 4096) / elapsed
-            metric, value, unit = "fuse_rand_read_4k_IOPS", round(iops, 1), "IOPS"
+            metric = (f"{args.path}_rand_read_4k_IOPS")
+            value, unit = round(iops, 1), "IOPS"
             vs = round(gibps / REFERENCE_PEAK_RAND_GIBPS, 3)
         else:
             metric = ("fuse_seq_read_GiBps" if args.path == "fuse"

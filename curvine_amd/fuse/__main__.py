@@ -58,10 +58,18 @@ def main(argv=None) -> int:
                         device_id=args.device).start()
     print(f"READY {args.mnt}", flush=True)
 
+    def dump_stats(*_a):
+        import json as _json
+        import sys as _sys
+        print(_json.dumps({"fuse_op_stats": daemon.session.stats()}),
+              file=_sys.stderr, flush=True)
+
     stop = threading.Event()
     signal.signal(signal.SIGTERM, lambda *a: stop.set())
     signal.signal(signal.SIGINT, lambda *a: stop.set())
+    signal.signal(signal.SIGUSR1, dump_stats)
     stop.wait()
+    dump_stats()
     daemon.stop()
     return 0
 

@@ -97,8 +97,10 @@ class FuseChannel(threading.Thread):
         # from the request buffer; READ replies DMA HBM->host straight into
         # the buffer handed to writev — no staging hop either way.
         from curvine_amd.native import PinnedBuffer
+        from curvine_amd.metrics import OpStats
         self.req_buf = PinnedBuffer(self.bufsize)
         self.reply_pin = PinnedBuffer(self.bufsize)
+        self.stats = OpStats()
 
     def run(self) -> None:
         req_view = self.req_buf.view
@@ -122,7 +124,9 @@ class FuseChannel(threading.Thread):
                 log.exception("fuse dispatch failed: %s", e)
         log.info("fuse channel %d exiting", self.idx)
 
-    def dispatch(self, req: bytes) -> None:
+    def dispatch(self, req) -> None:
+        import time as _time
+        t0 = _time.perf_counter()
         (length, opcode, unique, nodeid, uid, gid, pid, _extlen, _pad) = \
             abi.IN_HEADER.unpack_from(req, 0)
         body = memoryview(req)[abi.IN_HEADER_SIZE:length]
@@ -149,6 +153,10 @@ class FuseChannel(threading.Thread):
         if result is None:
             return   # no reply (FORGET, INTERRUPT, or handler replied itself)
         self.reply(unique, result)
+        self.stats.record(abi.Op.NAMES.get(opcode, str(opcode)),
+                          _time.perf_counter() - t0,
+                          len(result) if not isinstance(result, (list, tuple))
+                          else sum(len(p) for p in result))
 
     # ---------------- replies ----------------
     def reply(self, unique: int, body) -> None:
@@ -199,6 +207,10 @@ class FuseSession:
         log.info("fuse mounted at %s (%d channels)", self.mnt_path,
                  len(self.channels))
         return self
+
+    def stats(self) -> dict:
+        from curvine_amd.metrics import OpStats
+        return OpStats.merge([ch.stats for ch in self.channels])
 
     def stop(self) -> None:
         self.stopped = True
