@@ -281,10 +281,46 @@ static void check_range(Arena* a, uint64_t off, uint64_t n) {
   if (off + n > a->cap) throw std::runtime_error("arena range out of bounds");
 }
 
+static bool is_pinned_host(const void* p) {
+  hipPointerAttribute_t at;
+  if (hipPointerGetAttributes(&at, p) != hipSuccess) {
+    (void)hipGetLastError();   // clear the error
+    return false;
+  }
+  return at.type == hipMemoryTypeHost;
+}
+
+// small or pinned-destination copies skip the ring: one hipMemcpyAsync on
+// a pooled stream (parallel across caller threads, no global lock)
+static const uint64_t RING_THRESHOLD = 2u << 20;
+
+static void dev_read_direct(Arena* a, uint64_t off, uint8_t* dst, uint64_t n) {
+  uint32_t slot = a->srr.fetch_add(1) % SPOOL;
+  std::lock_guard<std::mutex> g(a->spool_mu[slot]);
+  HIP_CHECK(hipSetDevice(a->device));
+  HIP_CHECK(hipMemcpyAsync(dst, (const uint8_t*)a->base + off, n,
+                           hipMemcpyDeviceToHost, a->spool[slot]));
+  HIP_CHECK(hipStreamSynchronize(a->spool[slot]));
+}
+
+static void dev_write_direct(Arena* a, uint64_t off, const uint8_t* src,
+                             uint64_t n) {
+  uint32_t slot = a->srr.fetch_add(1) % SPOOL;
+  std::lock_guard<std::mutex> g(a->spool_mu[slot]);
+  HIP_CHECK(hipSetDevice(a->device));
+  HIP_CHECK(hipMemcpyAsync((uint8_t*)a->base + off, src, n,
+                           hipMemcpyHostToDevice, a->spool[slot]));
+  HIP_CHECK(hipStreamSynchronize(a->spool[slot]));
+}
+
 // device -> host buffer, chunked through the pinned ring: D2H DMA of chunk
 // k overlaps the CPU memcpy of chunk k-1 (the staging pipeline of
 // BASELINE.json's "pinned hipMemcpyAsync on a side stream").
 static void dev_read(Arena* a, uint64_t off, uint8_t* dst, uint64_t n) {
+  if (n <= RING_THRESHOLD || is_pinned_host(dst)) {
+    dev_read_direct(a, off, dst, n);
+    return;
+  }
   std::lock_guard<std::mutex> g(a->mu);
   HIP_CHECK(hipSetDevice(a->device));
   const uint8_t* src = (const uint8_t*)a->base + off;
@@ -314,6 +350,10 @@ static void dev_read(Arena* a, uint64_t off, uint8_t* dst, uint64_t n) {
 }
 
 static void dev_write(Arena* a, uint64_t off, const uint8_t* src, uint64_t n) {
+  if (n <= RING_THRESHOLD || is_pinned_host(src)) {
+    dev_write_direct(a, off, src, n);
+    return;
+  }
   std::lock_guard<std::mutex> g(a->mu);
   HIP_CHECK(hipSetDevice(a->device));
   uint8_t* dst = (uint8_t*)a->base + off;
@@ -609,6 +649,69 @@ static uintptr_t arena_base_ptr(int h) {
 }
 
 // ---------------------------------------------------------------------------
+// DLPack export: zero-copy torch tensors over arena memory (uint8, 1-D).
+// Lets RCCL collectives (torch.distributed "nccl" on ROCm) broadcast
+// HBM-resident cache blocks over xGMI without a staging copy, and lets
+// DataLoaders consume cached bytes as device tensors.
+// ABI structs per dlpack v0.8 (stable).
+// ---------------------------------------------------------------------------
+
+extern "C" {
+typedef struct { int32_t device_type; int32_t device_id; } DLDevice;
+typedef struct { uint8_t code; uint8_t bits; uint16_t lanes; } DLDataType;
+typedef struct {
+  void* data; DLDevice device; int32_t ndim; DLDataType dtype;
+  int64_t* shape; int64_t* strides; uint64_t byte_offset;
+} DLTensor;
+typedef struct DLManagedTensor {
+  DLTensor dl_tensor; void* manager_ctx;
+  void (*deleter)(struct DLManagedTensor*);
+} DLManagedTensor;
+}
+static const int kDLCPU = 1, kDLROCM = 10;
+
+struct DLWrap { DLManagedTensor mt; int64_t shape[1]; };
+
+static void dl_deleter(DLManagedTensor* mt) {
+  delete reinterpret_cast<DLWrap*>(mt->manager_ctx);
+}
+
+static void dl_capsule_destructor(PyObject* cap) {
+  // torch renames the capsule to "used_dltensor" after consuming it
+  if (PyCapsule_IsValid(cap, "dltensor")) {
+    auto* mt = (DLManagedTensor*)PyCapsule_GetPointer(cap, "dltensor");
+    if (mt && mt->deleter) mt->deleter(mt);
+  }
+}
+
+static py::object arena_dlpack(int h, uint64_t off, uint64_t n) {
+  Arena* a = get_arena(h);
+  check_range(a, off, n);
+  auto* w = new DLWrap();
+  w->shape[0] = (int64_t)n;
+  w->mt.dl_tensor.data = (uint8_t*)a->base + off;
+  w->mt.dl_tensor.device = {a->is_dev() ? kDLROCM : kDLCPU,
+                            a->is_dev() ? a->device : 0};
+  w->mt.dl_tensor.ndim = 1;
+  w->mt.dl_tensor.dtype = {0 /*kDLInt? 0=int*/, 8, 1};
+  w->mt.dl_tensor.dtype.code = 1;  // kDLUInt
+  w->mt.dl_tensor.shape = w->shape;
+  w->mt.dl_tensor.strides = nullptr;
+  w->mt.dl_tensor.byte_offset = 0;
+  w->mt.manager_ctx = w;
+  w->mt.deleter = dl_deleter;
+  PyObject* cap = PyCapsule_New(&w->mt, "dltensor", dl_capsule_destructor);
+  return py::reinterpret_steal<py::object>(cap);
+}
+
+static py::memoryview arena_host_view(int h, uint64_t off, uint64_t n) {
+  Arena* a = get_arena(h);
+  if (a->is_dev()) throw std::runtime_error("arena_host_view: device arena");
+  check_range(a, off, n);
+  return py::memoryview::from_memory((uint8_t*)a->base + off, n, false);
+}
+
+// ---------------------------------------------------------------------------
 // Pinned host buffers (FUSE reply/request buffers: DMA lands directly in
 // the buffer that is writev'd to /dev/fuse — no staging-ring hop)
 // ---------------------------------------------------------------------------
@@ -702,6 +805,8 @@ PYBIND11_MODULE(_native, m) {
   m.def("arena_gather", &arena_gather);
   m.def("arena_base_ptr", &arena_base_ptr);
   m.def("arena_info", &arena_info);
+  m.def("arena_dlpack", &arena_dlpack);
+  m.def("arena_host_view", &arena_host_view);
   m.def("pinned_alloc", &pinned_alloc);
   m.def("pinned_free", &pinned_free);
   m.def("pinned_view", &pinned_view);
