@@ -50,8 +50,9 @@ class ClusterRuntime:
     def _run(self):
         from concurrent.futures import ThreadPoolExecutor
         asyncio.set_event_loop(self.loop)
+        workers = min(64, (os.cpu_count() or 8) * 2)
         self.loop.set_default_executor(
-            ThreadPoolExecutor(max_workers=64, thread_name_prefix="cv-io"))
+            ThreadPoolExecutor(max_workers=workers, thread_name_prefix="cv-io"))
         self.loop.run_forever()
 
     def call(self, coro, timeout=600.0):
@@ -282,15 +283,17 @@ def step_client_seq(args, rank, rt) -> int:
     async def run_step():
         sem = asyncio.Semaphore(args.threads)
 
+        from curvine_amd import native
+        use_pinned = native.gpu_available()
+
         async def read_file(i):
-            from curvine_amd.native import PinnedBuffer
             async with sem:
                 r = await rt.fs.open(f"/bench/r{rank}/f{i}")
                 got, pos = 0, 0
-                # pinned destination: D2H DMA lands directly here (57 GB/s
-                # link rate) instead of the pageable staging path
-                pbuf = PinnedBuffer(args.read_chunk)
-                buf = pbuf.view
+                # pinned destination on GPU boxes: D2H DMA lands directly
+                # here (57 GB/s link rate) instead of the pageable path
+                pbuf = native.PinnedBuffer(args.read_chunk) if use_pinned else None
+                buf = pbuf.view if pbuf else bytearray(args.read_chunk)
                 while pos < r.length:
                     n = await r.pread_into(pos, buf, 0,
                                            min(args.read_chunk, r.length - pos))
@@ -299,7 +302,8 @@ def step_client_seq(args, rank, rt) -> int:
                     pos += n
                     got += n
                 r.close()
-                pbuf.close()
+                if pbuf:
+                    pbuf.close()
                 return got
         res = await asyncio.gather(*[read_file(i) for i in range(args.files)])
         return sum(res)
