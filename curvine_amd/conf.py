@@ -112,8 +112,10 @@ class MasterConf:
 class JournalConf:
     enable: bool = True
     journal_dir: str = "/tmp/curvine/journal"
-    # raft peers "id@host:port"; single entry = standalone (no election)
+    # raft peers "id@host:port" (port = the master RPC port); single entry
+    # or empty = standalone (no election)
     peers: list[str] = field(default_factory=list)
+    node_id: int = 0          # this master's id within `peers`
     snapshot_interval_entries: int = 100_000
     segment_max_bytes: int = 256 << 20
     flush_batch: int = 256

@@ -31,12 +31,17 @@ class MasterHandler:
         fn = getattr(self, f"op_{code.name.lower()}", None)
         if fn is None:
             raise err.Unsupported(f"rpc code {msg.code}")
-        if self.master.raft is not None and not self.master.raft.is_leader \
-                and code not in _READONLY_OK:
-            raise err.NotLeader(f"leader={self.master.raft.leader_addr or ''}")
+        raft = self.master.raft
+        if raft is not None and not raft.is_leader and code not in _READONLY_OK:
+            raise err.NotLeader(f"leader={raft.leader_addr or ''}")
         t0 = time.perf_counter()
         try:
+            op_before = self.master.journal.op_id
             reply = fn(msg.header, msg.data)
+            if raft is not None and raft.is_leader \
+                    and self.master.journal.op_id > op_before:
+                # withhold the reply until the mutation's entries commit
+                await raft.wait_commit(self.master.journal.op_id)
             return msg.reply(reply or {})
         finally:
             if self.fs.conf.master.audit_log:
@@ -199,3 +204,13 @@ class MasterHandler:
 
     def op_metricsreport(self, h, d):
         return {}
+
+    # ---------------- raft protocol ----------------
+    def op_raftvote(self, h, d):
+        return self.master.raft.on_vote(h)
+
+    def op_raftappendentries(self, h, d):
+        return self.master.raft.on_append(h)
+
+    def op_raftinstallsnapshot(self, h, d):
+        return self.master.raft.on_install_snapshot(h)

@@ -1,0 +1,466 @@
+"""Raft journal replication for master HA.
+
+Analog of the reference's curvine-raft crate
+(/root/reference/crates/metadata/curvine-raft/src/raft/raft_node.rs:47-186
+RawNode wrapper, raft/storage/ log storage, raft/snapshot/ install) and its
+journal integration (journal_system.rs:43-419): the journal's entries ARE
+the raft log; followers replay them into their FsDir.
+
+Speaks the same RPC framing as everything else (codes RaftVote /
+RaftAppendEntries / RaftInstallSnapshot) on the master's RPC server.
+
+Design notes (deviations, by design, from textbook Raft):
+* the leader applies entries to its state machine at append time (the
+  master's mutation API is synchronous); client REPLIES are withheld until
+  commit (MasterHandler awaits wait_commit).  On losing leadership with
+  uncommitted tail entries the node rebuilds its state machine from
+  snapshot + committed log (`_rebuild` callback) — the dirty optimistic
+  state never becomes visible.
+* log entries are the journal's msgpack entries; index == op_id.
+"""
+from __future__ import annotations
+
+import asyncio
+import logging
+import os
+import random
+import struct
+import time
+from typing import Callable, Optional
+
+import msgpack
+
+from curvine_amd import errors as err
+from curvine_amd.master.journal import crc32c_sw, decode_stream
+from curvine_amd.rpc.client import RpcClient
+from curvine_amd.rpc.codes import RpcCode
+
+log = logging.getLogger("curvine.raft")
+
+FOLLOWER, CANDIDATE, LEADER = "follower", "candidate", "leader"
+_FRAME = struct.Struct(">II")
+
+
+class RaftLog:
+    """Durable log: entries[i] = (term, payload-dict). 1-indexed."""
+
+    def __init__(self, path: str):
+        self.path = path
+        os.makedirs(os.path.dirname(path) or ".", exist_ok=True)
+        self.entries: list[tuple[int, dict]] = []   # in-memory copy
+        self.snapshot_index = 0
+        self.snapshot_term = 0
+        self._f = None
+        self._load()
+
+    def _load(self):
+        if os.path.exists(self.path):
+            with open(self.path, "rb") as f:
+                for rec in decode_stream(f):
+                    if rec.get("_snap"):
+                        self.snapshot_index = rec["index"]
+                        self.snapshot_term = rec["term"]
+                        self.entries = []
+                    else:
+                        self.entries.append((rec["_t"], rec["e"]))
+        self._f = open(self.path, "ab")
+
+    def _append_file(self, rec: dict):
+        payload = msgpack.packb(rec, use_bin_type=True)
+        self._f.write(_FRAME.pack(len(payload), crc32c_sw(payload)) + payload)
+
+    def flush(self):
+        self._f.flush()
+        os.fsync(self._f.fileno())
+
+    @property
+    def last_index(self) -> int:
+        return self.snapshot_index + len(self.entries)
+
+    def term_at(self, index: int) -> int:
+        if index == self.snapshot_index:
+            return self.snapshot_term
+        if index <= self.snapshot_index or index > self.last_index:
+            return -1
+        return self.entries[index - self.snapshot_index - 1][0]
+
+    def entry_at(self, index: int) -> dict:
+        return self.entries[index - self.snapshot_index - 1][1]
+
+    def append(self, term: int, entry: dict) -> int:
+        self.entries.append((term, entry))
+        self._append_file({"_t": term, "e": entry})
+        return self.last_index
+
+    def truncate_from(self, index: int) -> None:
+        """Drop entries >= index (conflict resolution); rewrites the file."""
+        keep = index - self.snapshot_index - 1
+        if keep < 0:
+            keep = 0
+        self.entries = self.entries[:keep]
+        self._rewrite()
+
+    def compact_to(self, index: int, term: int) -> None:
+        n = index - self.snapshot_index
+        if n <= 0:
+            return
+        self.entries = self.entries[n:]
+        self.snapshot_index = index
+        self.snapshot_term = term
+        self._rewrite()
+
+    def _rewrite(self):
+        self._f.close()
+        tmp = self.path + ".tmp"
+        with open(tmp, "wb") as f:
+            payload = msgpack.packb({"_snap": True, "index": self.snapshot_index,
+                                     "term": self.snapshot_term},
+                                    use_bin_type=True)
+            f.write(_FRAME.pack(len(payload), crc32c_sw(payload)) + payload)
+            for t, e in self.entries:
+                payload = msgpack.packb({"_t": t, "e": e}, use_bin_type=True)
+                f.write(_FRAME.pack(len(payload), crc32c_sw(payload)) + payload)
+            f.flush()
+            os.fsync(f.fileno())
+        os.replace(tmp, self.path)
+        self._f = open(self.path, "ab")
+
+
+class RaftNode:
+    """One raft participant, driven by the master's asyncio loop."""
+
+    def __init__(self, node_id: int, peers: dict[int, tuple[str, int]],
+                 state_dir: str,
+                 apply_entry: Callable[[dict], None],
+                 make_snapshot: Callable[[], dict],
+                 load_snapshot: Callable[[dict], None],
+                 rebuild: Callable[[], None],
+                 election_timeout_ms: int = 1500,
+                 heartbeat_ms: int = 300):
+        self.id = node_id
+        self.peers = {pid: addr for pid, addr in peers.items() if pid != node_id}
+        self.state = FOLLOWER
+        self.term = 0
+        self.voted_for: Optional[int] = None
+        self.leader_id: Optional[int] = None
+        self.log = RaftLog(os.path.join(state_dir, f"raft_{node_id}.log"))
+        self.meta_path = os.path.join(state_dir, f"raft_{node_id}.meta")
+        self._load_meta()
+        self.commit_index = self.log.snapshot_index
+        self.last_applied = self.log.snapshot_index
+        self.apply_entry = apply_entry
+        self.make_snapshot = make_snapshot
+        self.load_snapshot_cb = load_snapshot
+        self.rebuild = rebuild
+        self.election_timeout = election_timeout_ms / 1000.0
+        self.heartbeat = heartbeat_ms / 1000.0
+        self.next_index: dict[int, int] = {}
+        self.match_index: dict[int, int] = {}
+        self._last_heard = time.monotonic()
+        self._commit_waiters: list[tuple[int, asyncio.Future]] = []
+        self._clients: dict[int, RpcClient] = {}
+        self._stopped = False
+        self._tasks: list[asyncio.Task] = []
+        self._applying_remote = False
+
+    # ---------------- persistence ----------------
+    def _load_meta(self):
+        if os.path.exists(self.meta_path):
+            with open(self.meta_path, "rb") as f:
+                m = msgpack.unpackb(f.read(), raw=False)
+            self.term = m["term"]
+            self.voted_for = m["voted_for"]
+
+    def _save_meta(self):
+        tmp = self.meta_path + ".tmp"
+        with open(tmp, "wb") as f:
+            f.write(msgpack.packb({"term": self.term,
+                                   "voted_for": self.voted_for}))
+            f.flush()
+            os.fsync(f.fileno())
+        os.replace(tmp, self.meta_path)
+
+    # ---------------- lifecycle ----------------
+    def start(self) -> "RaftNode":
+        # replay committed log into the state machine at boot happens via
+        # journal restore; raft applies from last_applied onward
+        self._tasks.append(asyncio.create_task(self._ticker()))
+        if not self.peers:   # single-node group: immediate leader
+            self._become_leader()
+        return self
+
+    async def stop(self):
+        self._stopped = True
+        for t in self._tasks:
+            t.cancel()
+        await asyncio.gather(*self._tasks, return_exceptions=True)
+        for c in self._clients.values():
+            await c.close()
+        self.log.flush()
+
+    @property
+    def is_leader(self) -> bool:
+        return self.state == LEADER
+
+    @property
+    def leader_addr(self) -> str:
+        if self.leader_id is not None and self.leader_id in self.peers:
+            h, p = self.peers[self.leader_id]
+            return f"{h}:{p}"
+        return ""
+
+    # ---------------- timers ----------------
+    async def _ticker(self):
+        while not self._stopped:
+            try:
+                await asyncio.sleep(self.heartbeat / 2)
+                now = time.monotonic()
+                if self.state == LEADER:
+                    await self._broadcast_append()
+                elif now - self._last_heard > \
+                        self.election_timeout * (1 + random.random()):
+                    await self._run_election()
+            except asyncio.CancelledError:
+                return
+            except Exception as e:  # noqa: BLE001
+                log.exception("raft tick: %s", e)
+
+    async def _client(self, pid: int) -> RpcClient:
+        c = self._clients.get(pid)
+        if c is None or not c.is_connected:
+            h, p = self.peers[pid]
+            c = await RpcClient(h, p, timeout_ms=2000).connect()
+            self._clients[pid] = c
+        return c
+
+    # ---------------- election ----------------
+    async def _run_election(self):
+        self.state = CANDIDATE
+        self.term += 1
+        self.voted_for = self.id
+        self._save_meta()
+        self.leader_id = None
+        self._last_heard = time.monotonic()
+        term = self.term
+        votes = 1
+        log.info("node %d starting election term %d", self.id, term)
+
+        async def ask(pid):
+            try:
+                c = await self._client(pid)
+                r = await c.rpc(RpcCode.RaftVote, {
+                    "term": term, "candidate": self.id,
+                    "last_log_index": self.log.last_index,
+                    "last_log_term": self.log.term_at(self.log.last_index)},
+                    timeout=1.5)
+                return r.header
+            except Exception:  # noqa: BLE001
+                return None
+
+        results = await asyncio.gather(*[ask(p) for p in self.peers])
+        if self.term != term or self.state != CANDIDATE:
+            return
+        for r in results:
+            if r is None:
+                continue
+            if r.get("term", 0) > self.term:
+                self._become_follower(r["term"])
+                return
+            if r.get("granted"):
+                votes += 1
+        if votes * 2 > len(self.peers) + 1:
+            self._become_leader()
+
+    def _become_leader(self):
+        self.state = LEADER
+        self.leader_id = self.id
+        for pid in self.peers:
+            self.next_index[pid] = self.log.last_index + 1
+            self.match_index[pid] = 0
+        # apply any follower-era backlog so our state machine is current
+        # (leader applies-at-append from here on)
+        while self.last_applied < self.log.last_index:
+            self.last_applied += 1
+            try:
+                self.apply_entry(self.log.entry_at(self.last_applied))
+            except Exception as e:  # noqa: BLE001
+                log.exception("backlog apply %d: %s", self.last_applied, e)
+        # no-op entry in the new term: commits the previous-term backlog
+        # (raft's no-commit-of-old-terms rule needs a current-term entry)
+        self.append_local({"op": "noop", "op_id": self.log.last_index + 1})
+        log.info("node %d is LEADER (term %d, last_index %d)",
+                 self.id, self.term, self.log.last_index)
+
+    def _become_follower(self, term: int):
+        was_leader = self.state == LEADER
+        dirty = self.log.last_index > self.commit_index
+        self.state = FOLLOWER
+        if term > self.term:
+            self.term = term
+            self.voted_for = None
+            self._save_meta()
+        self._last_heard = time.monotonic()
+        if was_leader and dirty:
+            # optimistic applies beyond commit: rebuild the state machine
+            log.warning("node %d stepping down with uncommitted tail; rebuild",
+                        self.id)
+            self.log.truncate_from(self.commit_index + 1)
+            self.rebuild()
+        self._fail_waiters()
+
+    def _fail_waiters(self):
+        for idx, fut in self._commit_waiters:
+            if not fut.done():
+                fut.set_exception(err.NotLeader(f"leader={self.leader_addr}"))
+        self._commit_waiters.clear()
+
+    # ---------------- leader: propose + replicate ----------------
+    def append_local(self, entry: dict) -> int:
+        """Leader-side append; returns the new index (== op_id).  The
+        caller's state machine applies at append time, so last_applied
+        advances with the append."""
+        if self.state != LEADER:
+            raise err.NotLeader(f"leader={self.leader_addr}")
+        idx = self.log.append(self.term, entry)
+        self.last_applied = idx
+        return idx
+
+    async def wait_commit(self, index: int, timeout: float = 10.0):
+        if index <= self.commit_index:
+            return
+        fut = asyncio.get_event_loop().create_future()
+        self._commit_waiters.append((index, fut))
+        await self._broadcast_append()
+        await asyncio.wait_for(fut, timeout)
+
+    async def _broadcast_append(self):
+        if self.state != LEADER:
+            return
+        if not self.peers:
+            self._advance_commit(self.log.last_index)
+            return
+        await asyncio.gather(*[self._append_to(p) for p in self.peers],
+                             return_exceptions=True)
+        # commit = median of match indexes (including self)
+        matches = sorted([self.log.last_index] +
+                         [self.match_index.get(p, 0) for p in self.peers],
+                        reverse=True)
+        majority = matches[len(matches) // 2]
+        if majority > self.commit_index and \
+                self.log.term_at(majority) == self.term:
+            self._advance_commit(majority)
+
+    def _advance_commit(self, index: int):
+        self.commit_index = index
+        self.log.flush()
+        remaining = []
+        for idx, fut in self._commit_waiters:
+            if idx <= index:
+                if not fut.done():
+                    fut.set_result(None)
+            else:
+                remaining.append((idx, fut))
+        self._commit_waiters = remaining
+
+    async def _append_to(self, pid: int):
+        ni = self.next_index.get(pid, self.log.last_index + 1)
+        if ni <= self.log.snapshot_index:
+            await self._send_snapshot(pid)
+            return
+        prev = ni - 1
+        entries = []
+        for i in range(ni, min(self.log.last_index, ni + 255) + 1):
+            entries.append({"_t": self.log.term_at(i), "e": self.log.entry_at(i)})
+        try:
+            c = await self._client(pid)
+            r = await c.rpc(RpcCode.RaftAppendEntries, {
+                "term": self.term, "leader": self.id,
+                "prev_index": prev, "prev_term": self.log.term_at(prev),
+                "entries": entries, "commit": self.commit_index}, timeout=2.0)
+        except Exception:  # noqa: BLE001
+            return
+        h = r.header
+        if h.get("term", 0) > self.term:
+            self._become_follower(h["term"])
+            return
+        if h.get("success"):
+            self.match_index[pid] = prev + len(entries)
+            self.next_index[pid] = self.match_index[pid] + 1
+        else:
+            self.next_index[pid] = max(1, h.get("hint", ni - 1))
+
+    async def _send_snapshot(self, pid: int):
+        snap = self.make_snapshot()
+        try:
+            c = await self._client(pid)
+            await c.rpc(RpcCode.RaftInstallSnapshot, {
+                "term": self.term, "leader": self.id,
+                "index": self.log.last_index,
+                "snap_term": self.log.term_at(self.log.last_index),
+                "state": snap}, timeout=10.0)
+            self.next_index[pid] = self.log.last_index + 1
+            self.match_index[pid] = self.log.last_index
+        except Exception as e:  # noqa: BLE001
+            log.debug("snapshot to %d failed: %s", pid, e)
+
+    # ---------------- RPC handlers (called from MasterHandler) ----------------
+    def on_vote(self, h: dict) -> dict:
+        if h["term"] > self.term:
+            self._become_follower(h["term"])
+        granted = False
+        if h["term"] >= self.term and \
+                self.voted_for in (None, h["candidate"]):
+            my_last_term = self.log.term_at(self.log.last_index)
+            if (h["last_log_term"], h["last_log_index"]) >= \
+                    (my_last_term, self.log.last_index):
+                granted = True
+                self.voted_for = h["candidate"]
+                self._save_meta()
+                self._last_heard = time.monotonic()
+        return {"term": self.term, "granted": granted}
+
+    def on_append(self, h: dict) -> dict:
+        if h["term"] < self.term:
+            return {"term": self.term, "success": False}
+        if h["term"] > self.term or self.state != FOLLOWER:
+            self._become_follower(h["term"])
+        self._last_heard = time.monotonic()
+        self.leader_id = h["leader"]
+        prev = h["prev_index"]
+        if prev > self.log.last_index or \
+                (prev > self.log.snapshot_index and
+                 self.log.term_at(prev) != h["prev_term"]):
+            return {"term": self.term, "success": False,
+                    "hint": min(prev, self.log.last_index + 1)}
+        idx = prev
+        for rec in h["entries"]:
+            idx += 1
+            if idx <= self.log.last_index:
+                if self.log.term_at(idx) == rec["_t"]:
+                    continue
+                self.log.truncate_from(idx)
+            self.log.append(rec["_t"], rec["e"])
+        if h["commit"] > self.commit_index:
+            self.commit_index = min(h["commit"], self.log.last_index)
+            self._apply_committed()
+        return {"term": self.term, "success": True}
+
+    def on_install_snapshot(self, h: dict) -> dict:
+        if h["term"] < self.term:
+            return {"term": self.term}
+        self._become_follower(h["term"])
+        self.leader_id = h["leader"]
+        self.load_snapshot_cb(h["state"])
+        self.log.compact_to(h["index"], h["snap_term"])
+        self.commit_index = h["index"]
+        self.last_applied = h["index"]
+        log.info("node %d installed snapshot at index %d", self.id, h["index"])
+        return {"term": self.term}
+
+    def _apply_committed(self):
+        while self.last_applied < self.commit_index:
+            self.last_applied += 1
+            try:
+                self.apply_entry(self.log.entry_at(self.last_applied))
+            except Exception as e:  # noqa: BLE001
+                log.exception("apply entry %d: %s", self.last_applied, e)

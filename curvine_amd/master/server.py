@@ -22,11 +22,12 @@ from curvine_amd.rpc.server import HandlerService, RpcServer
 
 log = logging.getLogger("curvine.master")
 
-# codes a raft follower may serve (read-only surface)
+# codes a raft follower may serve.  Client READS also go to the leader
+# (linearizable read-your-writes; followers may lag) — only the raft
+# protocol itself and liveness probes are served by anyone.
 _READONLY_OK = {
-    RpcCode.Heartbeat, RpcCode.FileStatus, RpcCode.ListStatus,
-    RpcCode.Exists, RpcCode.GetFilesystemInfo, RpcCode.GetMountTable,
-    RpcCode.GetMountInfo, RpcCode.GetJobStatus, RpcCode.MetricsReport,
+    RpcCode.Heartbeat, RpcCode.MetricsReport,
+    RpcCode.RaftVote, RpcCode.RaftAppendEntries, RpcCode.RaftInstallSnapshot,
 }
 import curvine_amd.master.handler as _handler_mod
 _handler_mod._READONLY_OK = _READONLY_OK
@@ -40,29 +41,129 @@ class MasterService(HandlerService):
         return MasterHandler(self.master)
 
 
+class RaftJournalWriter(JournalWriter):
+    """Journal writer whose durable log is the raft log: log() appends to
+    the raft leader's log (index == op_id); the raft node replicates and
+    the RPC handler withholds replies until commit."""
+
+    def __init__(self, journal_dir: str):
+        super().__init__(journal_dir)
+        self.raft = None   # wired by Master after construction
+
+    def log(self, op: str, **fields) -> dict:
+        if not self.enabled:
+            return {}
+        # index is assigned deterministically before append so the entry is
+        # serialized into the raft log with its final op_id
+        index = self.raft.log.last_index + 1
+        entry = {"op": op, "op_id": index, **fields}
+        got = self.raft.append_local(entry)
+        assert got == index
+        self.op_id = index
+        return entry
+
+    def flush(self) -> None:
+        if self.raft is not None:
+            self.raft.log.flush()
+
+    def close(self) -> None:
+        self.flush()
+
+    def purge_through(self, op_id: int) -> None:
+        pass   # raft log compaction handles retention
+
+
 class Master:
     def __init__(self, conf: ClusterConf):
         self.conf = conf
-        self.journal = JournalWriter(conf.journal.journal_dir,
-                                     conf.journal.segment_max_bytes)
+        self._raft_mode = len(conf.journal.peers) > 1
+        if self._raft_mode:
+            self.journal = RaftJournalWriter(conf.journal.journal_dir)
+        else:
+            self.journal = JournalWriter(conf.journal.journal_dir,
+                                         conf.journal.segment_max_bytes)
         self.fs = MasterFilesystem(conf, self.journal)
         self.mounts = MountManager(self.journal)
         self.jobs = JobManager(self)
         self.replication = MasterReplicationManager(self.fs)
-        self.raft = None  # set by journal system when peers configured
+        self.raft = None
         self.rpc = RpcServer("master", conf.master.hostname,
                              conf.master.rpc_port, MasterService(self))
         self._actor_task: Optional[asyncio.Task] = None
         self._stopped = asyncio.Event()
         self._mutation_count = 0
 
+    def _make_raft(self):
+        from curvine_amd.master.raft import RaftNode
+        peers = {}
+        for spec in self.conf.journal.peers:
+            pid, _, addr = spec.partition("@")
+            host, _, port = addr.rpartition(":")
+            peers[int(pid)] = (host, int(port))
+        return RaftNode(
+            self.conf.journal.node_id, peers,
+            self.conf.journal.journal_dir,
+            apply_entry=self._apply_entry,
+            make_snapshot=self._snapshot_state,
+            load_snapshot=self._load_snapshot_state,
+            rebuild=self._rebuild_state,
+            election_timeout_ms=self.conf.journal.election_timeout_ms,
+            heartbeat_ms=self.conf.journal.heartbeat_interval_ms)
+
+    def _apply_entry(self, e: dict) -> None:
+        """Raft follower apply path (leader applied at append time)."""
+        if not self.mounts.apply_entry(e):
+            self.fs.fs_dir.apply_entry(e)
+        self.journal.op_id = max(self.journal.op_id, e["op_id"])
+
+    def _snapshot_state(self) -> dict:
+        state = self.fs.fs_dir.to_snapshot()
+        state["mounts"] = self.mounts.to_snapshot()
+        return state
+
+    def _load_snapshot_state(self, state: dict) -> None:
+        self.fs.fs_dir.load_snapshot(state)
+        self.mounts.load_snapshot(state.get("mounts", []))
+
+    def _rebuild_state(self) -> None:
+        """Re-derive the state machine from snapshot + committed raft log
+        (called when stepping down with optimistic uncommitted applies)."""
+        from curvine_amd.master.fs_dir import FsDir
+        self.fs.fs_dir = FsDir(self.journal)
+        self.mounts.mounts = {}
+        snap = self.fs.loader.snapshot_path()
+        import os as _os
+        if _os.path.exists(snap):
+            with open(snap, "rb") as f:
+                from curvine_amd.master.journal import decode_stream
+                entries = list(decode_stream(f))
+            if entries:
+                self._load_snapshot_state(entries[0])
+        start = max(self.raft.log.snapshot_index, self.journal.op_id)
+        for i in range(start + 1, self.raft.commit_index + 1):
+            self._apply_entry(self.raft.log.entry_at(i))
+        self.journal.op_id = self.raft.commit_index
+
     # ---------------- lifecycle ----------------
     async def start(self) -> "Master":
-        self._restore()
+        if self._raft_mode:
+            self.raft = self._make_raft()
+            self.journal.raft = self.raft
+            # replay the local raft log into the state machine
+            lg = self.raft.log
+            for i in range(lg.snapshot_index + 1, lg.last_index + 1):
+                self._apply_entry(lg.entry_at(i))
+            self.raft.last_applied = lg.last_index
+            self.journal.op_id = lg.last_index
+        else:
+            self._restore()
         await self.rpc.start()
         self.conf.master.rpc_port = self.rpc.port
+        if self.raft is not None:
+            self.raft.start()
         self._actor_task = asyncio.create_task(self._actor_loop())
-        log.info("master started on %s:%d", self.conf.master.hostname, self.rpc.port)
+        log.info("master started on %s:%d%s", self.conf.master.hostname,
+                 self.rpc.port, " (raft)" if self.raft else "")
         return self
 
     async def stop(self) -> None:
@@ -73,6 +174,8 @@ class Master:
                 await self._actor_task
             except (asyncio.CancelledError, Exception):  # noqa: BLE001
                 pass
+        if self.raft is not None:
+            await self.raft.stop()
         await self.rpc.stop()
         self.journal.close()
 
@@ -92,10 +195,13 @@ class Master:
         self.fs.loader.load(apply, load_snap)
 
     def checkpoint(self) -> None:
-        state = self.fs.fs_dir.to_snapshot()
-        state["mounts"] = self.mounts.to_snapshot()
+        state = self._snapshot_state()
         self.fs.loader.save_snapshot(state)
-        self.journal.purge_through(self.journal.op_id)
+        if self.raft is not None:
+            la = self.raft.last_applied
+            self.raft.log.compact_to(la, self.raft.log.term_at(la))
+        else:
+            self.journal.purge_through(self.journal.op_id)
 
     # ---------------- background actor ----------------
     async def _actor_loop(self) -> None:
@@ -110,6 +216,8 @@ class Master:
             except asyncio.CancelledError:
                 return
             tick += 1
+            if self.raft is not None and not self.raft.is_leader:
+                continue   # followers don't drive cluster mutations
             try:
                 lost = self.fs.workers.check_expired()
                 if lost:

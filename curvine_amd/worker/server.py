@@ -92,9 +92,15 @@ class Worker:
 
     # ---------------- heartbeats ----------------
     async def _connect_master(self) -> None:
-        host, port = self.conf.master_addr()
-        self._master = await RpcClient(host, port,
-                                       self.conf.client.rpc_timeout_ms).connect()
+        # ClusterConnector: follows the raft leader across masters
+        from curvine_amd.rpc.client import ClusterConnector
+        addrs = list(self.conf.client.master_addrs)
+        own = f"{self.conf.master.hostname}:{self.conf.master.rpc_port}"
+        if own not in addrs:
+            addrs.append(own)
+        if self._master is None:
+            self._master = ClusterConnector(addrs,
+                                            self.conf.client.rpc_timeout_ms)
 
     async def _heartbeat_loop(self) -> None:
         interval = self.conf.worker.heartbeat_interval_ms / 1000.0
