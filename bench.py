@@ -245,69 +245,100 @@ def step_fuse_rand4k(args, rank, mnt, lat_out: list) -> int:
     return sum(total)
 
 
+def _sync_readers(args, rank, rt):
+    """Open all rank-local files as synchronous short-circuit readers."""
+    readers = []
+    for i in range(args.files):
+        r = rt.call(rt.fs.open(f"/bench/r{rank}/f{i}"))
+        readers.append(r.to_sync())
+        r.close()
+    return readers
+
+
 def step_client_rand4k(args, rank, rt, lat_out: list) -> int:
-    """Random 4 KiB preads through the client short-circuit path."""
+    """Random 4 KiB reads via the SYNC short-circuit path: plain OS
+    threads, no event loop in the per-op path (this is the IOPS metric)."""
     import random
+    readers = _sync_readers(args, rank, rt)
+    per_thread = args.rand_reads // args.threads
+    total = [0] * args.threads
+    lats: list[list[float]] = [[] for _ in range(args.threads)]
+    errs = []
 
-    async def run_step():
-        readers = []
-        for i in range(args.files):
-            readers.append(await rt.fs.open(f"/bench/r{rank}/f{i}"))
-        lats: list[float] = []
-        total = 0
-        rng = random.Random(rank)
-
-        async def one_reader(t):
-            nonlocal total
+    def worker(t):
+        try:
+            rng = random.Random(t * 7919 + rank)
             buf = bytearray(4096)
-            r2 = random.Random(t * 7919 + rank)
-            for _ in range(args.rand_reads // args.threads):
-                r = readers[r2.randrange(len(readers))]
-                off = r2.randrange(max(1, r.length - 4096))
+            for _ in range(per_thread):
+                r = readers[rng.randrange(len(readers))]
+                off = rng.randrange(max(1, r.length - 4096))
                 t0 = time.perf_counter_ns()
-                n = await r.pread_into(off, buf, 0, 4096)
-                lats.append((time.perf_counter_ns() - t0) / 1000.0)
-                total += n
-        await asyncio.gather(*[one_reader(t) for t in range(args.threads)])
-        for r in readers:
-            r.close()
-        lats.sort()
-        if lats:
-            lat_out.append({"p50_us": lats[len(lats) // 2],
-                            "p99_us": lats[int(len(lats) * 0.99)]})
-        return total
-    return rt.call(run_step())
+                n = r.pread_into(off, buf, 0, 4096)
+                lats[t].append((time.perf_counter_ns() - t0) / 1000.0)
+                total[t] += n
+        except Exception as e:  # noqa: BLE001
+            errs.append(e)
+
+    ths = [threading.Thread(target=worker, args=(t,))
+           for t in range(args.threads)]
+    for t in ths:
+        t.start()
+    for t in ths:
+        t.join()
+    for r in readers:
+        r.close()
+    if errs:
+        raise errs[0]
+    alll = sorted(x for l in lats for x in l)
+    if alll:
+        lat_out.append({"p50_us": alll[len(alll) // 2],
+                        "p99_us": alll[int(len(alll) * 0.99)]})
+    return sum(total)
 
 
 def step_client_seq(args, rank, rt) -> int:
-    async def run_step():
-        sem = asyncio.Semaphore(args.threads)
+    """Sequential reads via the SYNC short-circuit path: OS threads with
+    pinned destination buffers (direct D2H DMA, no event loop per op)."""
+    from curvine_amd import native
+    use_pinned = native.gpu_available()
+    readers = _sync_readers(args, rank, rt)
+    nthreads = min(args.threads, max(1, len(readers)))
+    total = [0] * nthreads
+    errs = []
 
-        from curvine_amd import native
-        use_pinned = native.gpu_available()
-
-        async def read_file(i):
-            async with sem:
-                r = await rt.fs.open(f"/bench/r{rank}/f{i}")
-                got, pos = 0, 0
-                # pinned destination on GPU boxes: D2H DMA lands directly
-                # here (57 GB/s link rate) instead of the pageable path
-                pbuf = native.PinnedBuffer(args.read_chunk) if use_pinned else None
-                buf = pbuf.view if pbuf else bytearray(args.read_chunk)
+    def worker(t):
+        try:
+            pbuf = native.PinnedBuffer(args.read_chunk) if use_pinned else None
+            buf = None if pbuf else bytearray(args.read_chunk)
+            # thread t reads files t, t+T, ... fully (fio numjobs analog)
+            for i in range(t, len(readers), nthreads):
+                r = readers[i]
+                pos = 0
                 while pos < r.length:
-                    n = await r.pread_into(pos, buf, 0,
-                                           min(args.read_chunk, r.length - pos))
+                    want = min(args.read_chunk, r.length - pos)
+                    if pbuf is not None:
+                        n = r.pread_into_ptr(pos, pbuf.ptr, want)
+                    else:
+                        n = r.pread_into(pos, buf, 0, want)
                     if n <= 0:
                         break
                     pos += n
-                    got += n
-                r.close()
-                if pbuf:
-                    pbuf.close()
-                return got
-        res = await asyncio.gather(*[read_file(i) for i in range(args.files)])
-        return sum(res)
-    got = rt.call(run_step())
+                    total[t] += n
+            if pbuf is not None:
+                pbuf.close()
+        except Exception as e:  # noqa: BLE001
+            errs.append(e)
+
+    ths = [threading.Thread(target=worker, args=(t,)) for t in range(nthreads)]
+    for t in ths:
+        t.start()
+    for t in ths:
+        t.join()
+    for r in readers:
+        r.close()
+    if errs:
+        raise errs[0]
+    got = sum(total)
     expect = args.files * args.file_size
     if got != expect:
         raise RuntimeError(f"step read {got} != {expect}")

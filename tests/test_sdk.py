@@ -1,0 +1,101 @@
+"""Python SDK: fsspec filesystem + torch tensor reads."""
+import os
+
+import pytest
+
+from curvine_amd.testing import SyncMiniCluster
+
+
+@pytest.fixture
+def cluster(tmp_path):
+    smc = SyncMiniCluster(tmp_dir=str(tmp_path)).start()
+    yield smc
+    smc.stop()
+
+
+def test_fsspec_roundtrip(cluster):
+    import fsspec
+
+    from curvine_amd.sdk.fsspec_fs import register
+    register()
+    master = f"127.0.0.1:{cluster.master.rpc.port}"
+    fs = fsspec.filesystem("cv", master=master, skip_instance_cache=True)
+    fs.mkdir("/sdk/dir")
+    data = os.urandom(5 << 20)
+    with fs.open("/sdk/dir/blob.bin", "wb") as f:
+        f.write(data)
+    assert fs.exists("/sdk/dir/blob.bin")
+    info = fs.info("/sdk/dir/blob.bin")
+    assert info["size"] == len(data) and info["type"] == "file"
+    with fs.open("/sdk/dir/blob.bin", "rb") as f:
+        assert f.read(100) == data[:100]
+        f.seek(1 << 20)
+        assert f.read(50) == data[1 << 20:(1 << 20) + 50]
+    assert fs.cat_file("/sdk/dir/blob.bin", 10, 20) == data[10:20]
+    names = fs.ls("/sdk/dir", detail=False)
+    assert any(n.endswith("blob.bin") for n in names)
+    fs.mv("/sdk/dir/blob.bin", "/sdk/moved.bin")
+    assert fs.exists("/sdk/moved.bin")
+    fs.rm("/sdk", recursive=True)
+    assert not fs.exists("/sdk/moved.bin")
+
+
+def test_pandas_over_fsspec(cluster):
+    pd = pytest.importorskip("pandas")
+    import fsspec
+
+    from curvine_amd.sdk.fsspec_fs import register
+    register()
+    master = f"127.0.0.1:{cluster.master.rpc.port}"
+    fs = fsspec.filesystem("cv", master=master, skip_instance_cache=True)
+    with fs.open("/df.csv", "wb") as f:
+        f.write(b"a,b\n1,2\n3,4\n")
+    df = pd.read_csv(f"cv://df.csv", storage_options={
+        "master": master, "skip_instance_cache": True})
+    assert list(df.columns) == ["a", "b"] and len(df) == 2
+
+
+def test_tensor_read_cpu(cluster):
+    import torch
+
+    from curvine_amd.client.filesystem import SyncFs
+    from curvine_amd.sdk.torch_io import CurvineTensorReader
+    sf = SyncFs(cluster.client_conf())
+    data = os.urandom(2 << 20)
+    sf.write_file("/tensor.bin", data)
+    r = CurvineTensorReader(sf, "/tensor.bin")
+    t = torch.zeros(len(data), dtype=torch.uint8)
+    n = r.read_into_tensor(t)
+    assert n == len(data)
+    assert bytes(t.numpy().tobytes()) == data
+    # offset read
+    t2 = r.to_tensor(device="cpu", file_off=1000, n=500)
+    assert t2.numpy().tobytes() == data[1000:1500]
+    r.close()
+    sf.shutdown()
+
+
+@pytest.mark.gpu
+def test_tensor_read_hbm_to_device(tmp_path):
+    """HBM-cached file -> cuda tensor: pure device-to-device copy."""
+    import torch
+
+    from curvine_amd.client.filesystem import SyncFs
+    from curvine_amd.sdk.torch_io import CurvineTensorReader
+    from curvine_amd.testing import SyncMiniCluster, test_conf
+    conf = test_conf(str(tmp_path))
+    conf.worker.data_dirs = ["[HBM:512MB:0]gpu0"]
+    smc = SyncMiniCluster(conf=conf, tmp_dir=str(tmp_path)).start()
+    try:
+        sf = SyncFs(smc.client_conf())
+        data = os.urandom(32 << 20)
+        sf.write_file("/hbm.bin", data, storage_tier="HBM")
+        r = CurvineTensorReader(sf, "/hbm.bin")
+        t = torch.zeros(len(data), dtype=torch.uint8, device="cuda:0")
+        n = r.read_into_tensor(t)
+        assert n == len(data)
+        assert t.cpu().numpy().tobytes() == data
+        r.close()
+        sf.shutdown()
+    finally:
+        smc.stop()
