@@ -106,6 +106,8 @@ class MasterFilesystem:
     def add_block(self, path: str, commit_prev_len: int = -1,
                   client_host: str = "", client_worker_id: int = -1,
                   exclude_workers: list[int] | None = None) -> LocatedBlock:
+        from curvine_amd.fault import fault_point
+        fault_point("master.add_block")
         node = self.fs_dir.must_resolve(path)
         if node.id not in self.writing:
             raise err.FsError(f"no write lease on {path}")

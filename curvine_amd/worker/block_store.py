@@ -70,6 +70,8 @@ class BlockStore:
     # ---------------- write path ----------------
     def create_writer(self, block_id: int, reserve: int,
                       tier_hint: str = "") -> BlockWriter:
+        from curvine_amd.fault import fault_point
+        fault_point("worker.block.create")
         with self.lock:
             existing = self.blocks.get(block_id)
             if existing is not None:
@@ -100,6 +102,8 @@ class BlockStore:
 
     def finalize(self, block_id: int, length: int) -> str:
         """Publish a written block; returns its tier."""
+        from curvine_amd.fault import fault_point
+        fault_point("worker.block.finalize")
         with self.lock:
             b = self.blocks.get(block_id)
             if b is None:

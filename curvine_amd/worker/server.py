@@ -187,6 +187,8 @@ class Worker:
         async with self._load_sem:
             ok, error = True, ""
             try:
+                from curvine_amd.fault import fault_point
+                fault_point("worker.load_task")
                 ufs = get_ufs(cmd["ufs_path"], cmd.get("properties", {}))
                 fs = CurvineFileSystem(self.conf)
                 fs.client.local_worker_id = self.worker_id
