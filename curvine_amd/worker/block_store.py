@@ -28,7 +28,7 @@ log = logging.getLogger("curvine.blockstore")
 
 class _Block:
     __slots__ = ("block_id", "state", "meta", "layout", "generation",
-                 "readers", "pending_delete")
+                 "readers", "pending_delete", "last_access")
 
     def __init__(self, block_id: int, layout: BlockLayout, meta: dict):
         self.block_id = block_id
@@ -38,6 +38,7 @@ class _Block:
         self.generation = 0
         self.readers = 0
         self.pending_delete = False
+        self.last_access = 0.0
 
 
 class BlockStore:
@@ -136,6 +137,8 @@ class BlockStore:
             if b.state != BlockState.FINALIZED:
                 raise err.BlockInWriting(str(block_id))
             b.readers += 1
+            import time as _time
+            b.last_access = _time.monotonic()
             meta = b.meta
         reader = BlockReader(b.layout, block_id, meta)
         store = self
@@ -201,6 +204,85 @@ class BlockStore:
     def block_count(self) -> int:
         with self.lock:
             return len(self.blocks)
+
+    # ---------------- tier demotion ----------------
+    def demote_coldest(self, high_watermark: float = 0.90,
+                       low_watermark: float = 0.75) -> int:
+        """Move least-recently-read blocks from a pressured tier to the
+        next colder one (HBM -> MEM/SSD demotion; the MI355X analog of the
+        reference's multi-tier rebalancing).  Finalized blocks are
+        immutable, so the copy runs without blocking readers; the swap is
+        skipped if a reader appeared meanwhile.  Returns blocks moved."""
+        moved = 0
+        for i, src in enumerate(self.layouts[:-1]):
+            if src.capacity <= 0 or src.used < src.capacity * high_watermark:
+                continue
+            target_used = src.capacity * low_watermark
+            with self.lock:
+                cands = sorted(
+                    (b for b in self.blocks.values()
+                     if b.layout is src and b.state == BlockState.FINALIZED
+                     and not b.pending_delete),
+                    key=lambda b: b.last_access)
+            for b in cands:
+                if src.used <= target_used:
+                    break
+                if self._demote_one(b, self.layouts[i + 1:]):
+                    moved += 1
+        return moved
+
+    def _demote_one(self, b: _Block, colder: list[BlockLayout]) -> bool:
+        length = b.meta.get("length", 0)
+        for dst in colder:
+            if dst.available < length:
+                continue
+            try:
+                new_meta = dst.allocate(b.block_id, length)
+            except err.CapacityExceeded:
+                continue
+            try:
+                self._copy_block(b.layout, b.meta, dst, new_meta, length)
+                dst.finalize(new_meta, length)
+            except Exception as e:  # noqa: BLE001
+                log.warning("demotion copy of block %d failed: %s",
+                            b.block_id, e)
+                dst.deallocate(new_meta)
+                return False
+            with self.lock:
+                if b.readers > 0 or b.pending_delete or \
+                        self.blocks.get(b.block_id) is not b:
+                    swap = False
+                else:
+                    old_layout, old_meta = b.layout, b.meta
+                    b.layout, b.meta = dst, new_meta
+                    self._added.append({"block_id": b.block_id,
+                                        "tier": dst.tier})
+                    swap = True
+            if swap:
+                old_layout.deallocate(old_meta)
+                log.info("demoted block %d %s -> %s (%d bytes)",
+                         b.block_id, old_layout.tier, dst.tier, length)
+                return True
+            dst.deallocate(new_meta)
+            return False
+        return False
+
+    @staticmethod
+    def _copy_block(src_layout, src_meta, dst_layout, dst_meta,
+                    length: int, chunk: int = 16 << 20) -> None:
+        from curvine_amd.worker.layout import ArenaLayout
+        if isinstance(src_layout, ArenaLayout) and \
+                isinstance(dst_layout, ArenaLayout):
+            # device/host arena direct copy (D2D 5 TB/s on-chip, or DMA)
+            src_layout.arena.copy_to(dst_layout.arena, dst_meta["offset"],
+                                     src_meta["offset"], length)
+            return
+        pos = 0
+        while pos < length:
+            n = min(chunk, length - pos)
+            data = src_layout._read_at(src_meta, pos, n)
+            dst_layout._write_at(dst_meta, pos, data, len(data))
+            pos += n
 
     def close(self) -> None:
         for l in self.layouts:

@@ -104,10 +104,16 @@ class Worker:
 
     async def _heartbeat_loop(self) -> None:
         interval = self.conf.worker.heartbeat_interval_ms / 1000.0
+        tick = 0
         while not self._stopped.is_set():
             try:
                 await asyncio.sleep(interval)
                 await self._heartbeat_once()
+                tick += 1
+                if tick % 5 == 0:   # tier-pressure demotion sweep
+                    loop = asyncio.get_event_loop()
+                    await loop.run_in_executor(None,
+                                               self.store.demote_coldest)
             except asyncio.CancelledError:
                 return
             except Exception as e:  # noqa: BLE001
