@@ -81,6 +81,86 @@ class CurvineFileSystem:
         fb = await self.client.open(path)
         return FsReader(self.client, fb)
 
+    async def write_files_batch(self, files: dict, overwrite: bool = True,
+                                storage_tier: str = "") -> list[FileStatus]:
+        """Small-file fast path (BatchBlockWriter analog,
+        /root/reference/.../block/batch_block_writer.rs +
+        batch_write_handler.rs:32-150): one CreateFilesBatch, one
+        AddBlocksBatch, payloads grouped per worker into WriteBlocksBatch
+        frames, one CompleteFilesBatch."""
+        from curvine_amd.rpc.codes import RpcCode
+        from curvine_amd.rpc.message import MAX_DATA_SIZE
+        paths = list(files.keys())
+        tier = storage_tier or self.conf.client.storage_tier
+        reply = await self.client.connector.rpc(RpcCode.CreateFilesBatch, {
+            "files": [{"path": p, "overwrite": overwrite,
+                       "storage_tier": tier,
+                       "block_size": self.conf.client.block_size}
+                      for p in paths]})
+        reply = await self.client.connector.rpc(RpcCode.AddBlocksBatch, {
+            "blocks": [{"path": p} for p in paths],
+            "client_host": self.client.client_host,
+            "client_worker_id": self.client.local_worker_id})
+        from curvine_amd.model import LocatedBlock
+        lbs = [LocatedBlock.from_dict(b) for b in reply.header["blocks"]]
+        # group by (first) target worker
+        groups: dict = {}
+        for p, lb in zip(paths, lbs):
+            addr = lb.locations[0]
+            groups.setdefault(addr.key(), (addr, []))[1].append((p, lb))
+        from curvine_amd.client.block_client import factory
+        from curvine_amd.worker import registry
+        commits: dict[str, dict] = {}
+        for addr, items in groups.values():
+            store = registry.lookup(addr.worker_id)
+            if store is not None:
+                loop = asyncio.get_event_loop()
+                for p, lb in items:
+                    data = files[p]
+
+                    def write_one(bid=lb.block.block_id, data=data):
+                        w = store.create_writer(bid, max(len(data), 1), tier)
+                        if data:
+                            w.write(data)
+                        return store.finalize(bid, len(data))
+                    t = await loop.run_in_executor(None, write_one)
+                    commits[p] = {"block_id": lb.block.block_id,
+                                  "locations": [addr.worker_id], "tiers": [t]}
+                continue
+            client = await factory().get(addr.hostname, addr.rpc_port)
+            batch, payload, size = [], [], 0
+            pend = list(items)
+            while pend:
+                p, lb = pend.pop(0)
+                data = files[p]
+                if size + len(data) > MAX_DATA_SIZE and batch:
+                    await self._flush_batch(client, batch, payload, commits,
+                                            addr, tier)
+                    batch, payload, size = [], [], 0
+                batch.append((p, lb))
+                payload.append(data)
+                size += len(data)
+            if batch:
+                await self._flush_batch(client, batch, payload, commits,
+                                        addr, tier)
+        reply = await self.client.connector.rpc(RpcCode.CompleteFilesBatch, {
+            "files": [{"path": p, "length": len(files[p]),
+                       "block_lens": [len(files[p])] if files[p] else [],
+                       "commits": [commits[p]] if p in commits else []}
+                      for p in paths]})
+        return [FileStatus.from_dict(s) for s in reply.header["statuses"]]
+
+    async def _flush_batch(self, client, batch, payload, commits, addr, tier):
+        from curvine_amd.rpc.codes import RpcCode
+        blocks = [{"block_id": lb.block.block_id, "length": len(d),
+                   "tier": tier}
+                  for (p, lb), d in zip(batch, payload)]
+        reply = await client.rpc(RpcCode.WriteBlocksBatch,
+                                 {"blocks": blocks}, b"".join(payload))
+        for p, lb in batch:
+            commits[p] = {"block_id": lb.block.block_id,
+                          "locations": [addr.worker_id], "tiers": [tier]}
+
     async def write_all(self, path: str, data, overwrite: bool = True,
                         **kw) -> FileStatus:
         w = await self.create(path, overwrite=overwrite, **kw)

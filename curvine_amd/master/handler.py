@@ -133,10 +133,23 @@ class MasterHandler:
             out.append(st.to_dict())
         return {"statuses": out}
 
+    def op_addblocksbatch(self, h, d):
+        out = []
+        for req in h["blocks"]:
+            lb = self.fs.add_block(req["path"], req.get("commit_prev_len", -1),
+                                   h.get("client_host", ""),
+                                   h.get("client_worker_id", -1))
+            out.append(lb.to_dict())
+        return {"blocks": out}
+
     def op_completefilesbatch(self, h, d):
+        out = []
         for req in h["files"]:
-            self.fs.complete_file(req["path"], req["length"], req.get("block_lens"))
-        return {}
+            st = self.fs.complete_file(req["path"], req["length"],
+                                       req.get("block_lens"),
+                                       req.get("commits"))
+            out.append(st.to_dict())
+        return {"statuses": out}
 
     # ---------------- mounts ----------------
     def op_mount(self, h, d):
