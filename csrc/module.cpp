@@ -157,7 +157,7 @@ static int device_count() {
 // Arena
 // ---------------------------------------------------------------------------
 
-#define SPOOL 8   // parallel copy streams per device arena
+#define SPOOL 16  // parallel copy streams per device arena
 
 struct Arena {
   void* base = nullptr;
@@ -294,10 +294,19 @@ static bool is_pinned_host(const void* p) {
 // a pooled stream (parallel across caller threads, no global lock)
 static const uint64_t RING_THRESHOLD = 2u << 20;
 
+// tiny copies (the 4K-IOPS path): synchronous hipMemcpy — no stream
+// slot, no lock, no event round trip; thread-safe and latency-optimal
+static const uint64_t TINY_THRESHOLD = 256u << 10;
+
 static void dev_read_direct(Arena* a, uint64_t off, uint8_t* dst, uint64_t n) {
+  HIP_CHECK(hipSetDevice(a->device));
+  if (n <= TINY_THRESHOLD) {
+    HIP_CHECK(hipMemcpy(dst, (const uint8_t*)a->base + off, n,
+                        hipMemcpyDeviceToHost));
+    return;
+  }
   uint32_t slot = a->srr.fetch_add(1) % SPOOL;
   std::lock_guard<std::mutex> g(a->spool_mu[slot]);
-  HIP_CHECK(hipSetDevice(a->device));
   HIP_CHECK(hipMemcpyAsync(dst, (const uint8_t*)a->base + off, n,
                            hipMemcpyDeviceToHost, a->spool[slot]));
   HIP_CHECK(hipStreamSynchronize(a->spool[slot]));
@@ -305,9 +314,14 @@ static void dev_read_direct(Arena* a, uint64_t off, uint8_t* dst, uint64_t n) {
 
 static void dev_write_direct(Arena* a, uint64_t off, const uint8_t* src,
                              uint64_t n) {
+  HIP_CHECK(hipSetDevice(a->device));
+  if (n <= TINY_THRESHOLD) {
+    HIP_CHECK(hipMemcpy((uint8_t*)a->base + off, src, n,
+                        hipMemcpyHostToDevice));
+    return;
+  }
   uint32_t slot = a->srr.fetch_add(1) % SPOOL;
   std::lock_guard<std::mutex> g(a->spool_mu[slot]);
-  HIP_CHECK(hipSetDevice(a->device));
   HIP_CHECK(hipMemcpyAsync((uint8_t*)a->base + off, src, n,
                            hipMemcpyHostToDevice, a->spool[slot]));
   HIP_CHECK(hipStreamSynchronize(a->spool[slot]));

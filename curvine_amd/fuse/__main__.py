@@ -35,6 +35,9 @@ def main(argv=None) -> int:
     p.add_argument("--channels", type=int, default=0)
     p.add_argument("--data-dir", action="append", default=[],
                    help='worker data dir, e.g. "[HBM:200GB:0]gpu0"')
+    p.add_argument("--takeover", action="store_true",
+                   help="hot upgrade: adopt the running daemon's session "
+                        "fd + open-handle state instead of mounting")
     p.add_argument("--log-level", default="INFO")
     args = p.parse_args(argv)
 
@@ -55,7 +58,7 @@ def main(argv=None) -> int:
     conf.fuse.mnt_path = args.mnt
 
     daemon = FuseDaemon(conf, args.mnt, embed_worker=args.embed_worker,
-                        device_id=args.device).start()
+                        device_id=args.device).start(takeover=args.takeover)
     print(f"READY {args.mnt}", flush=True)
 
     def dump_stats(*_a):

@@ -8,6 +8,7 @@ reads.  `mount()` is the programmatic entry; `cv-fuse` CLI wraps it.
 from __future__ import annotations
 
 import asyncio
+import os
 import logging
 import threading
 from typing import Optional
@@ -42,7 +43,12 @@ class FuseDaemon:
     def call(self, coro, timeout: float = 120.0):
         return asyncio.run_coroutine_threadsafe(coro, self.loop).result(timeout)
 
-    def start(self) -> "FuseDaemon":
+    def control_socket_path(self) -> str:
+        import hashlib
+        h = hashlib.sha1(self.mnt_path.encode()).hexdigest()[:10]
+        return f"/tmp/cv-fuse-ctl-{h}.sock"
+
+    def start(self, takeover: bool = False) -> "FuseDaemon":
         self._loop_thread.start()
         if self.embed_worker:
             from curvine_amd.worker.server import Worker
@@ -57,15 +63,96 @@ class FuseDaemon:
         if self.worker is not None:
             self.fs.client.local_worker_id = self.worker.worker_id
         self.fuse_fs = CurvineFuseFs(self.fs, self.conf, self.loop)
+        session_fd = None
+        if takeover:
+            session_fd, state = self._request_takeover()
+            self.fuse_fs.restore_state(state)
         self.session = FuseSession(
             self.fuse_fs, self.mnt_path,
             channels=self.conf.fuse.mnt_number,
-            max_write=self.conf.fuse.max_write).start()
+            max_write=self.conf.fuse.max_write).start(session_fd=session_fd)
+        self._start_control_server()
         return self
 
+    # ---------------- hot upgrade (fd + state handover) ----------------
+    def _request_takeover(self) -> tuple[int, dict]:
+        """Connect to the running daemon's control socket; receive the
+        /dev/fuse session fd (SCM_RIGHTS) + serialized handle state."""
+        import json
+        import socket as sock
+        import struct
+        s = sock.socket(sock.AF_UNIX, sock.SOCK_STREAM)
+        s.connect(self.control_socket_path())
+        s.sendall(b"TAKEOVER")
+        msg, fds, _flags, _addr = sock.recv_fds(s, 8, 1)
+        (length,) = struct.unpack(">I", msg[:4]) if len(msg) >= 4 else (0,)
+        buf = msg[4:]
+        while len(buf) < length:
+            chunk = s.recv(65536)
+            if not chunk:
+                break
+            buf += chunk
+        s.close()
+        if not fds:
+            raise RuntimeError("takeover: no fd received")
+        state = json.loads(buf.decode()) if buf else {}
+        log.info("takeover: received session fd %d + %d handles",
+                 fds[0], len(state.get("handles", [])))
+        return fds[0], state
+
+    def _start_control_server(self) -> None:
+        import json
+        import socket as sock
+        import struct
+        import threading
+        path = self.control_socket_path()
+        try:
+            os.unlink(path)
+        except OSError:
+            pass
+        srv = sock.socket(sock.AF_UNIX, sock.SOCK_STREAM)
+        srv.bind(path)
+        srv.listen(1)
+        self._ctl_sock = srv
+
+        def serve():
+            while True:
+                try:
+                    conn, _ = srv.accept()
+                except OSError:
+                    return
+                try:
+                    req = conn.recv(64)
+                    if req.startswith(b"TAKEOVER"):
+                        state = self.fuse_fs.dump_state()
+                        payload = json.dumps(state).encode()
+                        sock.send_fds(conn,
+                                      [struct.pack(">I", len(payload)) + payload],
+                                      [self.session.session_fd])
+                        log.info("handed session fd to successor; draining")
+                        self.handed_over = True
+                    elif req.startswith(b"STATS"):
+                        conn.sendall(json.dumps(self.session.stats()).encode())
+                except Exception as e:  # noqa: BLE001
+                    log.warning("control socket: %s", e)
+                finally:
+                    conn.close()
+
+        threading.Thread(target=serve, daemon=True,
+                         name="cv-fuse-ctl").start()
+
+    handed_over = False
+
     def stop(self) -> None:
+        if getattr(self, "_ctl_sock", None):
+            try:
+                self._ctl_sock.close()
+                os.unlink(self.control_socket_path())
+            except OSError:
+                pass
         if self.session:
-            self.session.stop()
+            # after a handover the successor owns the mount: never umount
+            self.session.stop(umount_fs=not self.handed_over)
         if self.fs:
             try:
                 self.call(self.fs.close(), timeout=10)

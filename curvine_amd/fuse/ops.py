@@ -663,8 +663,13 @@ class CurvineFuseFs:
     def op_read(self, nodeid, body, ctx):
         fh, offset, size, _rf, _lo, _fl, _ = abi.READ_IN.unpack_from(body, 0)
         h = self.get_handle(fh)
+        if h.reader is None and h.writer is None:
+            # restored handle (hot upgrade): rebuild the reader lazily
+            fb = self.call(self.fs.client.open(h.path))
+            h.status = fb.status
+            h.reader = SyncReadState(self, fb)
         if h.reader is None:
-            # O_WRONLY handle read, or reader not yet available
+            # O_WRONLY handle read
             raise OSError(errno.EBADF, "not open for read")
         ch = ctx[4]
         pin = getattr(ch, "reply_pin", None)
@@ -885,6 +890,57 @@ class CurvineFuseFs:
 
     def op_setlkw(self, nodeid, body, ctx):
         return self.op_setlk(nodeid, body, ctx, wait=True)
+
+    # ---------------- hot-upgrade state persist/restore ----------------
+    # (fs/file_system.rs:219-223 + state/backend_handle.rs:288-310 +
+    #  fs-api StateFile analog)
+    def dump_state(self) -> dict:
+        """Serialize kernel-visible state (node ids, open handles, locks).
+        Writers are completed before dumping (freeze)."""
+        with self.handles_lock:
+            for h in list(self.handles.values()):
+                if h.writer is not None:
+                    with h.lock:
+                        st = h.writer.complete()
+                        h.writer = None
+                        h.status = st
+            handles = [{"fh": h.fh, "node_id": h.node_id, "path": h.path,
+                        "flags": h.flags, "write_pos": h.write_pos,
+                        "is_dir": h.dir_entries is not None or
+                        (h.status.is_dir if h.status else False)}
+                       for h in self.handles.values()]
+        with self.nodes_lock:
+            nodes = [{"id": n.id, "parent": n.parent, "name": n.name,
+                      "nlookup": n.nlookup}
+                     for n in self.nodes.values() if n.id != 1]
+        with self.plock_mu:
+            plocks = {str(k): v for k, v in self.plocks.items()}
+        return {"nodes": nodes, "next_node": self.next_node,
+                "handles": handles, "next_fh": self.next_fh,
+                "plocks": plocks}
+
+    def restore_state(self, state: dict) -> None:
+        with self.nodes_lock:
+            for nd in state.get("nodes", []):
+                node = Node(nd["id"], nd["parent"], nd["name"])
+                node.nlookup = nd["nlookup"]
+                self.nodes[node.id] = node
+            for node in self.nodes.values():
+                parent = self.nodes.get(node.parent)
+                if parent is not None and node.id != 1:
+                    parent.children[node.name] = node.id
+            self.next_node = state.get("next_node", self.next_node)
+        with self.handles_lock:
+            for hd in state.get("handles", []):
+                h = FileHandle(hd["fh"], hd["node_id"], hd["path"])
+                h.flags = hd.get("flags", 0)
+                h.write_pos = hd.get("write_pos", 0)
+                # readers rebuilt lazily on first READ
+                self.handles[h.fh] = h
+            self.next_fh = state.get("next_fh", self.next_fh)
+        with self.plock_mu:
+            for k, v in state.get("plocks", {}).items():
+                self.plocks[int(k)] = [tuple(x) for x in v]
 
     HANDLERS = {}
 

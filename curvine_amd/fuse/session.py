@@ -192,11 +192,18 @@ class FuseSession:
         self.channels: list[FuseChannel] = []
         fs.session = self
 
-    def start(self) -> "FuseSession":
-        self.session_fd = mount_fuse(self.mnt_path, allow_other=True,
-                                     max_read=self.max_write)
-        tune_readahead(self.mnt_path,
-                       getattr(self.fs.conf.fuse, "max_readahead", 8 << 20) >> 10)
+    def start(self, session_fd: int | None = None) -> "FuseSession":
+        """Mount fresh, or adopt an existing session fd (hot upgrade:
+        the kernel mount persists; we clone our channels off the passed
+        fd and serve the same connection)."""
+        if session_fd is None:
+            self.session_fd = mount_fuse(self.mnt_path, allow_other=True,
+                                         max_read=self.max_write)
+            tune_readahead(
+                self.mnt_path,
+                getattr(self.fs.conf.fuse, "max_readahead", 8 << 20) >> 10)
+        else:
+            self.session_fd = session_fd
         fds = [self.session_fd]
         for i in range(1, self.n_channels):
             fds.append(clone_channel(self.session_fd))
@@ -204,17 +211,21 @@ class FuseSession:
             ch = FuseChannel(self, fd, i)
             self.channels.append(ch)
             ch.start()
-        log.info("fuse mounted at %s (%d channels)", self.mnt_path,
-                 len(self.channels))
+        log.info("fuse %s at %s (%d channels)",
+                 "adopted" if session_fd is not None else "mounted",
+                 self.mnt_path, len(self.channels))
         return self
 
     def stats(self) -> dict:
         from curvine_amd.metrics import OpStats
         return OpStats.merge([ch.stats for ch in self.channels])
 
-    def stop(self) -> None:
+    def stop(self, umount_fs: bool = True) -> None:
+        """umount_fs=False: hand-over shutdown — the kernel mount stays
+        alive and is served by the adopting daemon."""
         self.stopped = True
-        umount(self.mnt_path)
+        if umount_fs:
+            umount(self.mnt_path)
         for ch in self.channels:
             try:
                 os.close(ch.fd)

@@ -218,6 +218,42 @@ class MasterHandler:
     def op_metricsreport(self, h, d):
         return {}
 
+    # ---------------- transfer service (codes 46-54) ----------------
+    # the standalone curvine-data-transfer surface; backed by the same
+    # job manager (transfer/mod.rs analog with the in-memory store)
+    def op_submittransfer(self, h, d):
+        return self.master.jobs.submit(h)
+
+    def op_gettransferstatus(self, h, d):
+        return self.master.jobs.status(h.get("job_id") or h.get("transfer_id"))
+
+    def op_canceltransfer(self, h, d):
+        return self.master.jobs.cancel(h.get("job_id") or h.get("transfer_id"))
+
+    def op_reporttransfertask(self, h, d):
+        self.master.jobs.report_task(h)
+        return {}
+
+    def op_querytransfertask(self, h, d):
+        job = self.master.jobs.jobs.get(h.get("job_id"))
+        if job is None:
+            raise err.JobNotFound(str(h.get("job_id")))
+        return {"tasks": list(job["tasks"].values())}
+
+    def op_listtransfers(self, h, d):
+        return {"transfers": [
+            {k: j[k] for k in ("job_id", "state", "total", "done", "failed",
+                               "path")}
+            for j in self.master.jobs.jobs.values()]}
+
+    def op_retrytransfer(self, h, d):
+        job = self.master.jobs.jobs.get(h.get("job_id"))
+        if job is None:
+            raise err.JobNotFound(str(h.get("job_id")))
+        return self.master.jobs.submit({"path": job["path"],
+                                        "recursive": job["recursive"],
+                                        "replicas": job["replicas"]})
+
     # ---------------- raft protocol ----------------
     def op_raftvote(self, h, d):
         return self.master.raft.on_vote(h)
