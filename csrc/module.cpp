@@ -138,6 +138,7 @@ static uint32_t crc32c_combine(uint32_t crc1, uint32_t crc2, uint64_t len2) {
 // need no -fgpu-rdc this way)
 // ---------------------------------------------------------------------------
 #include "kernels.hip"
+#include "lz4.hip"
 
 // ---------------------------------------------------------------------------
 // GPU availability
@@ -663,6 +664,146 @@ static uintptr_t arena_base_ptr(int h) {
 }
 
 // ---------------------------------------------------------------------------
+// LZ4 container: compress/decompress (host) + decompress-into-arena (GPU)
+// ---------------------------------------------------------------------------
+
+static py::bytes lz4_compress_py(py::buffer buf) {
+  py::buffer_info info = buf.request(false);
+  const uint8_t* src = (const uint8_t*)info.ptr;
+  size_t n = (size_t)info.size * info.itemsize;
+  uint32_t n_chunks = (uint32_t)((n + LZ4_CHUNK - 1) / LZ4_CHUNK);
+  std::vector<uint32_t> sizes(n_chunks);
+  std::vector<uint8_t> out;
+  out.reserve(n / 2 + 1024);
+  size_t header = 4 + 8 + 4 + 4 + 4ull * n_chunks;
+  out.resize(header);
+  {
+    py::gil_scoped_release rel;
+    std::vector<uint8_t> tmp(LZ4_CHUNK + LZ4_CHUNK / 128 + 64);
+    for (uint32_t c = 0; c < n_chunks; ++c) {
+      size_t cn = std::min<size_t>(LZ4_CHUNK, n - (size_t)c * LZ4_CHUNK);
+      size_t cs = lz4_compress_block(src + (size_t)c * LZ4_CHUNK, cn,
+                                     tmp.data());
+      sizes[c] = (uint32_t)cs;
+      out.insert(out.end(), tmp.data(), tmp.data() + cs);
+    }
+    uint8_t* h = out.data();
+    memcpy(h, &LZ4_MAGIC, 4);
+    uint64_t raw = n;
+    memcpy(h + 4, &raw, 8);
+    uint32_t ck = LZ4_CHUNK;
+    memcpy(h + 12, &ck, 4);
+    memcpy(h + 16, &n_chunks, 4);
+    memcpy(h + 20, sizes.data(), 4ull * n_chunks);
+  }
+  return py::bytes((const char*)out.data(), out.size());
+}
+
+struct Lz4Header {
+  uint64_t raw_size;
+  uint32_t chunk_size, n_chunks;
+  const uint32_t* sizes;
+  const uint8_t* payload;
+};
+
+static Lz4Header lz4_parse(const uint8_t* p, size_t n) {
+  if (n < 20) throw std::runtime_error("lz4: short container");
+  uint32_t magic;
+  memcpy(&magic, p, 4);
+  if (magic != LZ4_MAGIC) throw std::runtime_error("lz4: bad magic");
+  Lz4Header h;
+  memcpy(&h.raw_size, p + 4, 8);
+  memcpy(&h.chunk_size, p + 12, 4);
+  memcpy(&h.n_chunks, p + 16, 4);
+  if (n < 20 + 4ull * h.n_chunks) throw std::runtime_error("lz4: truncated");
+  h.sizes = (const uint32_t*)(p + 20);
+  h.payload = p + 20 + 4ull * h.n_chunks;
+  return h;
+}
+
+static py::bytes lz4_decompress_py(py::buffer buf) {
+  py::buffer_info info = buf.request(false);
+  const uint8_t* src = (const uint8_t*)info.ptr;
+  size_t n = (size_t)info.size * info.itemsize;
+  Lz4Header h = lz4_parse(src, n);
+  std::string out(h.raw_size, '\0');
+  {
+    py::gil_scoped_release rel;
+    const uint8_t* p = h.payload;
+    for (uint32_t c = 0; c < h.n_chunks; ++c) {
+      size_t cap = std::min<uint64_t>(h.chunk_size,
+                                      h.raw_size - (uint64_t)c * h.chunk_size);
+      size_t got = lz4_decompress_block_host(
+          p, h.sizes[c], (uint8_t*)out.data() + (uint64_t)c * h.chunk_size,
+          cap);
+      if (got != cap) throw std::runtime_error("lz4: corrupt chunk");
+      p += h.sizes[c];
+    }
+  }
+  return py::bytes(out);
+}
+
+// decompress a CVLZ container directly into an arena (GPU kernel on device
+// arenas: one workgroup per 64K chunk)
+static uint64_t arena_lz4_decompress(int ah, uint64_t dst_off, py::buffer buf) {
+  Arena* a = get_arena(ah);
+  py::buffer_info info = buf.request(false);
+  const uint8_t* src = (const uint8_t*)info.ptr;
+  size_t n = (size_t)info.size * info.itemsize;
+  Lz4Header h = lz4_parse(src, n);
+  check_range(a, dst_off, h.raw_size);
+  py::gil_scoped_release rel;
+  if (!a->is_dev()) {
+    const uint8_t* p = h.payload;
+    for (uint32_t c = 0; c < h.n_chunks; ++c) {
+      size_t cap = std::min<uint64_t>(h.chunk_size,
+                                      h.raw_size - (uint64_t)c * h.chunk_size);
+      size_t got = lz4_decompress_block_host(
+          p, h.sizes[c],
+          (uint8_t*)a->base + dst_off + (uint64_t)c * h.chunk_size, cap);
+      if (got != cap) throw std::runtime_error("lz4: corrupt chunk");
+      p += h.sizes[c];
+    }
+    return h.raw_size;
+  }
+  std::lock_guard<std::mutex> g(a->mu);
+  HIP_CHECK(hipSetDevice(a->device));
+  std::vector<uint32_t> offs(h.n_chunks);
+  uint32_t acc = 0;
+  for (uint32_t c = 0; c < h.n_chunks; ++c) {
+    offs[c] = acc;
+    acc += h.sizes[c];
+  }
+  size_t payload_n = acc;
+  size_t meta = 8ull * h.n_chunks + 64;
+  ensure_scratch(a, payload_n + meta + 64);
+  uint8_t* d_comp = (uint8_t*)a->scratch;
+  uint32_t* d_off = (uint32_t*)(d_comp + ((payload_n + 63) & ~63ull));
+  uint32_t* d_len = d_off + h.n_chunks;
+  int* d_err = (int*)(d_len + h.n_chunks);
+  HIP_CHECK(hipMemcpyAsync(d_comp, h.payload, payload_n,
+                           hipMemcpyHostToDevice, a->kstream));
+  HIP_CHECK(hipMemcpyAsync(d_off, offs.data(), 4ull * h.n_chunks,
+                           hipMemcpyHostToDevice, a->kstream));
+  HIP_CHECK(hipMemcpyAsync(d_len, h.sizes, 4ull * h.n_chunks,
+                           hipMemcpyHostToDevice, a->kstream));
+  HIP_CHECK(hipMemsetAsync(d_err, 0, 4, a->kstream));
+  int grid = (int)std::min<uint32_t>(h.n_chunks, 16384);
+  hipLaunchKernelGGL(lz4_decompress_kernel, dim3(grid), dim3(64), 0,
+                     a->kstream, d_comp, d_off, d_len,
+                     (uint8_t*)a->base + dst_off, h.chunk_size, h.raw_size,
+                     h.n_chunks, d_err);
+  HIP_CHECK(hipGetLastError());
+  int err_host = 0;
+  HIP_CHECK(hipMemcpyAsync(&err_host, d_err, 4, hipMemcpyDeviceToHost,
+                           a->kstream));
+  HIP_CHECK(hipStreamSynchronize(a->kstream));
+  if (err_host) throw std::runtime_error(
+      "lz4: corrupt chunk " + std::to_string(err_host - 1) + " (device)");
+  return h.raw_size;
+}
+
+// ---------------------------------------------------------------------------
 // DLPack export: zero-copy torch tensors over arena memory (uint8, 1-D).
 // Lets RCCL collectives (torch.distributed "nccl" on ROCm) broadcast
 // HBM-resident cache blocks over xGMI without a staging copy, and lets
@@ -825,6 +966,9 @@ PYBIND11_MODULE(_native, m) {
   m.def("pinned_free", &pinned_free);
   m.def("pinned_view", &pinned_view);
   m.def("pinned_ptr", &pinned_ptr);
+  m.def("lz4_compress", &lz4_compress_py);
+  m.def("lz4_decompress", &lz4_decompress_py);
+  m.def("arena_lz4_decompress", &arena_lz4_decompress);
   m.def("crc32c", &crc32c_buf, py::arg("buf"), py::arg("init") = 0);
   m.def("crc32c_combine", &crc32c_combine);
   m.attr("CRC_SUB") = CRC_SUB;

@@ -20,6 +20,7 @@ _REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 _SO = os.path.join(_REPO, "curvine_amd", "_native.so")
 _SRC = os.path.join(_REPO, "csrc", "module.cpp")
 _SRC2 = os.path.join(_REPO, "csrc", "kernels.hip")
+_SRC3 = os.path.join(_REPO, "csrc", "lz4.hip")
 _lock = threading.Lock()
 _mod = None
 
@@ -28,7 +29,8 @@ def build_native(force: bool = False) -> str:
     """Compile csrc/ into curvine_amd/_native.so for gfx950."""
     if not force and os.path.exists(_SO) and os.path.exists(_SRC):
         if os.path.getmtime(_SO) >= max(os.path.getmtime(_SRC),
-                                        os.path.getmtime(_SRC2)):
+                                        os.path.getmtime(_SRC2),
+                                        os.path.getmtime(_SRC3)):
             return _SO
     import pybind11
     hipcc = os.environ.get("HIPCC", "/opt/rocm/bin/hipcc")
@@ -177,3 +179,11 @@ def crc32c(buf, init: int = 0) -> int:
 
 def crc32c_combine(crc1: int, crc2: int, len2: int) -> int:
     return load().crc32c_combine(crc1, crc2, len2)
+
+
+def lz4_compress(buf) -> bytes:
+    return load().lz4_compress(buf)
+
+
+def lz4_decompress(buf) -> bytes:
+    return load().lz4_decompress(buf)

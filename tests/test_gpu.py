@@ -145,3 +145,20 @@ def test_e2e_hbm_cache(tmp_path):
             await mc.stop()
 
     asyncio.new_event_loop().run_until_complete(main())
+
+
+def test_lz4_device_decompress(dev_arena):
+    """GPU LZ4 decompress into the HBM arena == host decompress."""
+    import random
+    rng = random.Random(7)
+    for i, data in enumerate([
+            bytes(rng.choices(b"abcdefgh", k=3_000_000)),
+            os.urandom(500_000),
+            b"\x00" * (64 << 10) * 5 + os.urandom(1000),
+            (b"pattern123" * 7000)[: (64 << 10) * 2 + 13]]):
+        comp = native.lz4_compress(data)
+        n = native.load().arena_lz4_decompress(dev_arena.handle, 0, comp)
+        assert n == len(data)
+        out = np.zeros(len(data), dtype=np.uint8)
+        dev_arena.read(0, out, 0, len(data))
+        assert out.tobytes() == data, f"lz4 case {i} device mismatch"
