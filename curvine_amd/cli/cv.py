@@ -214,6 +214,16 @@ def cmd_node_list(fs, args):
               f"[{tiers}]")
 
 
+def cmd_node_decommission(fs, args):
+    from curvine_amd.rpc.codes import RpcCode
+
+    async def run():
+        r = await fs.fs.client.connector.rpc(
+            RpcCode.DecommissionWorker, {"worker_id": args.worker_id})
+        return r.header
+    print(json.dumps(fs.call(run())))
+
+
 def cmd_load(fs, args):
     job = fs.submit_job(args.path, recursive=True, replicas=args.replicas)
     print(json.dumps(job))
@@ -340,6 +350,9 @@ def build_parser() -> argparse.ArgumentParser:
     nodesub = node.add_subparsers(dest="nodecmd", required=True)
     nl = nodesub.add_parser("list")
     nl.set_defaults(fn=cmd_node_list)
+    nd = nodesub.add_parser("decommission")
+    nd.add_argument("worker_id", type=int)
+    nd.set_defaults(fn=cmd_node_decommission)
     return p
 
 

@@ -86,6 +86,7 @@ class MasterFilesystem:
         if node.is_dir:
             raise err.IsDirectory(path)
         node.atime_ms = now_ms()
+        node.access_count += 1
         return self._file_blocks(node, path)
 
     def _file_blocks(self, node: Inode, path: str) -> FileBlocks:

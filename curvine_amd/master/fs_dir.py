@@ -30,7 +30,7 @@ class Inode:
                  "length", "blocks", "block_size", "replicas", "storage_tier",
                  "complete", "mtime_ms", "atime_ms", "mode", "uid", "gid",
                  "ttl_ms", "ttl_action", "symlink_target", "nlink", "xattrs",
-                 "create_ms")
+                 "create_ms", "access_count")
 
     def __init__(self, id: int, name: str, file_type: int, mode: int = 0o755):
         self.id = id
@@ -56,6 +56,7 @@ class Inode:
         self.symlink_target = ""
         self.nlink = 1
         self.xattrs: dict[str, bytes] = {}
+        self.access_count = 0
 
     @property
     def is_dir(self) -> bool:

@@ -21,6 +21,16 @@ log = logging.getLogger("curvine.master.handler")
 audit = logging.getLogger("audit")
 
 
+_MUTATIONS = {
+    RpcCode.Mkdir, RpcCode.Delete, RpcCode.CreateFile, RpcCode.AppendFile,
+    RpcCode.Rename, RpcCode.AddBlock, RpcCode.CompleteFile, RpcCode.SetAttr,
+    RpcCode.Symlink, RpcCode.Link, RpcCode.ResizeFile, RpcCode.Free,
+    RpcCode.CreateFilesBatch, RpcCode.AddBlocksBatch,
+    RpcCode.CompleteFilesBatch, RpcCode.Mount, RpcCode.UnMount,
+    RpcCode.UpdateMount, RpcCode.SubmitJob,
+}
+
+
 class MasterHandler:
     def __init__(self, master):
         self.master = master
@@ -34,6 +44,15 @@ class MasterHandler:
         raft = self.master.raft
         if raft is not None and not raft.is_leader and code not in _READONLY_OK:
             raise err.NotLeader(f"leader={raft.leader_addr or ''}")
+        # mutation retry cache (fs_retry_cache.rs analog): a replayed
+        # request (connector retry after timeout/failover) returns the
+        # original reply instead of re-executing (double add_block etc.)
+        rkey = None
+        if code in _MUTATIONS and msg.header.get("cid") is not None:
+            rkey = (msg.header["cid"], msg.header.get("rid"))
+            cached = self.master.retry_cache.get(rkey)
+            if cached is not None:
+                return msg.reply(cached)
         t0 = time.perf_counter()
         try:
             op_before = self.master.journal.op_id
@@ -42,6 +61,8 @@ class MasterHandler:
                     and self.master.journal.op_id > op_before:
                 # withhold the reply until the mutation's entries commit
                 await raft.wait_commit(self.master.journal.op_id)
+            if rkey is not None:
+                self.master.retry_cache.put(rkey, reply or {})
             return msg.reply(reply or {})
         finally:
             if self.fs.conf.master.audit_log:
@@ -217,6 +238,14 @@ class MasterHandler:
 
     def op_metricsreport(self, h, d):
         return {}
+
+    def op_decommissionworker(self, h, d):
+        self.fs.workers.decommission(h["worker_id"])
+        # proactively queue re-replication of its blocks
+        for bid, locs in self.fs.workers.block_locs.items():
+            if h["worker_id"] in locs:
+                self.master.replication.mark_under_replicated(bid)
+        return {"state": "decommissioning"}
 
     # ---------------- transfer service (codes 46-54) ----------------
     # the standalone curvine-data-transfer surface; backed by the same

@@ -40,12 +40,17 @@ class MasterReplicationManager:
 
     def check_all(self) -> None:
         """Periodic sweep: compare live locations vs required replicas."""
+        from curvine_amd.model import WorkerState
+        decom_exists = any(w.state == WorkerState.DECOMMISSIONING
+                           for w in self.fs.workers.workers.values())
         for node in self.fs.fs_dir.iter_files():
-            if node.replicas <= 1 or not node.complete:
+            if not node.complete or (node.replicas <= 1 and not decom_exists):
                 continue
             for bid, _ in node.blocks:
-                live = self.fs.workers.locations_of(bid)
-                if 0 < len(live) < node.replicas:
+                locs = self.fs.workers.locations_of(bid)
+                healthy = [1 for w, _ in locs
+                           if w.state != WorkerState.DECOMMISSIONING]
+                if 0 < len(locs) and len(healthy) < node.replicas:
                     self.mark_under_replicated(bid)
 
     def _submit(self, block_id: int, job: dict) -> int:
@@ -58,13 +63,17 @@ class MasterReplicationManager:
         if node is None or not locs:
             self.pending.pop(block_id, None)
             return 0
-        if len(locs) >= node.replicas:
+        # replicas on decommissioning workers don't count toward the goal
+        from curvine_amd.model import WorkerState
+        healthy = [(w, t) for w, t in locs
+                   if w.state != WorkerState.DECOMMISSIONING]
+        if len(healthy) >= node.replicas:
             self.pending.pop(block_id, None)
             return 0
         have = {w.address.worker_id for w, _ in locs}
         try:
             targets = self.fs.workers.choose_workers(
-                node.replicas - len(locs), "load_based", exclude=have)
+                node.replicas - len(healthy), "load_based", exclude=have)
         except Exception:  # noqa: BLE001 — not enough workers yet
             return 0
         src = locs[0][0]

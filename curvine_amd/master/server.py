@@ -33,6 +33,33 @@ import curvine_amd.master.handler as _handler_mod
 _handler_mod._READONLY_OK = _READONLY_OK
 
 
+class RetryCache:
+    """Bounded (cid, rid) -> reply cache for mutation dedup
+    (fs/fs_retry_cache.rs analog)."""
+
+    def __init__(self, size: int = 100_000, ttl_ms: int = 600_000):
+        from collections import OrderedDict
+        self._d: "OrderedDict[tuple, tuple[float, dict]]" = OrderedDict()
+        self.size = size
+        self.ttl = ttl_ms / 1000.0
+
+    def get(self, key):
+        import time as _t
+        ent = self._d.get(key)
+        if ent is None:
+            return None
+        if _t.monotonic() - ent[0] > self.ttl:
+            self._d.pop(key, None)
+            return None
+        return ent[1]
+
+    def put(self, key, reply: dict) -> None:
+        import time as _t
+        self._d[key] = (_t.monotonic(), reply)
+        while len(self._d) > self.size:
+            self._d.popitem(last=False)
+
+
 class MasterService(HandlerService):
     def __init__(self, master: "Master"):
         self.master = master
@@ -87,6 +114,8 @@ class Master:
         self.jobs = JobManager(self)
         self.replication = MasterReplicationManager(self.fs)
         self.raft = None
+        self.retry_cache = RetryCache(conf.master.retry_cache_size,
+                                      conf.master.retry_cache_ttl_ms)
         self.rpc = RpcServer("master", conf.master.hostname,
                              conf.master.rpc_port, MasterService(self))
         self._actor_task: Optional[asyncio.Task] = None
@@ -267,7 +296,10 @@ class Master:
         target = cap * self.conf.master.eviction_low_watermark
         files = [n for n in self.fs.fs_dir.iter_files()
                  if n.complete and n.blocks and n.id not in self.fs.writing]
-        files.sort(key=lambda n: n.atime_ms)
+        if policy == "lfu":
+            files.sort(key=lambda n: (n.access_count, n.atime_ms))
+        else:
+            files.sort(key=lambda n: n.atime_ms)
         freed = 0
         for node in files:
             if used - freed <= target:

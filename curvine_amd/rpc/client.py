@@ -211,13 +211,23 @@ class ClusterConnector:
     (client/cluster_connector.rs analog)."""
 
     def __init__(self, addrs: list[str], timeout_ms: int = 60_000, retries: int = 3):
+        import uuid
         self.addrs = [(a.split(":")[0], int(a.split(":")[1])) for a in addrs]
         self.factory = ClientFactory(timeout_ms)
         self.retries = retries
         self._leader: Optional[tuple[str, int]] = None
+        # client identity for the master's mutation retry-cache
+        self.cid = uuid.uuid4().hex[:16]
+        self._rid = 0
 
     async def rpc(self, code: RpcCode, header: dict | None = None,
                   data: bytes = b"", timeout: float | None = None) -> Message:
+        # stable retry key: a re-sent logical call carries the same (cid,
+        # rid) so the master's retry cache dedups replayed mutations
+        header = dict(header or {})
+        self._rid += 1
+        header.setdefault("cid", self.cid)
+        header.setdefault("rid", self._rid)
         last: Exception = ConnectError("no master addresses")
         order = ([self._leader] if self._leader else []) + \
                 [a for a in self.addrs if a != self._leader]

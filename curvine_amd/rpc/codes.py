@@ -80,6 +80,7 @@ class RpcCode(IntEnum):
     RetryTransfer = 54
 
     MetricsReport = 60
+    DecommissionWorker = 61   # curvine_amd admin extension
 
     # raft journal (curvine_amd: raft speaks the same RPC framing)
     RaftVote = 70
