@@ -265,17 +265,26 @@ def step_client_rand4k(args, rank, rt, lat_out: list) -> int:
     lats: list[list[float]] = [[] for _ in range(args.threads)]
     errs = []
 
+    from curvine_amd import native
+    use_pinned = native.gpu_available()
+
     def worker(t):
         try:
             rng = random.Random(t * 7919 + rank)
-            buf = bytearray(4096)
+            pbuf = native.PinnedBuffer(4096) if use_pinned else None
+            buf = bytearray(4096) if pbuf is None else None
             for _ in range(per_thread):
                 r = readers[rng.randrange(len(readers))]
                 off = rng.randrange(max(1, r.length - 4096))
                 t0 = time.perf_counter_ns()
-                n = r.pread_into(off, buf, 0, 4096)
+                if pbuf is not None:
+                    n = r.pread_into_ptr(off, pbuf.ptr, 4096)
+                else:
+                    n = r.pread_into(off, buf, 0, 4096)
                 lats[t].append((time.perf_counter_ns() - t0) / 1000.0)
                 total[t] += n
+            if pbuf is not None:
+                pbuf.close()
         except Exception as e:  # noqa: BLE001
             errs.append(e)
 

@@ -295,37 +295,34 @@ static bool is_pinned_host(const void* p) {
 // a pooled stream (parallel across caller threads, no global lock)
 static const uint64_t RING_THRESHOLD = 2u << 20;
 
-// tiny copies (the 4K-IOPS path): synchronous hipMemcpy — no stream
-// slot, no lock, no event round trip; thread-safe and latency-optimal
-static const uint64_t TINY_THRESHOLD = 256u << 10;
+// per-caller-thread stream: zero contention for concurrent small copies
+// (the 4K-IOPS path).  hipMemcpy on the null stream and ROCm's internal
+// pageable-staging lock both serialize; a thread_local stream does not.
+static hipStream_t thread_stream(int device) {
+  thread_local hipStream_t tls[64] = {};
+  if (device < 0 || device >= 64) throw std::runtime_error("bad device");
+  if (!tls[device]) {
+    HIP_CHECK(hipSetDevice(device));
+    HIP_CHECK(hipStreamCreateWithFlags(&tls[device], hipStreamNonBlocking));
+  }
+  return tls[device];
+}
 
 static void dev_read_direct(Arena* a, uint64_t off, uint8_t* dst, uint64_t n) {
   HIP_CHECK(hipSetDevice(a->device));
-  if (n <= TINY_THRESHOLD) {
-    HIP_CHECK(hipMemcpy(dst, (const uint8_t*)a->base + off, n,
-                        hipMemcpyDeviceToHost));
-    return;
-  }
-  uint32_t slot = a->srr.fetch_add(1) % SPOOL;
-  std::lock_guard<std::mutex> g(a->spool_mu[slot]);
+  hipStream_t s = thread_stream(a->device);
   HIP_CHECK(hipMemcpyAsync(dst, (const uint8_t*)a->base + off, n,
-                           hipMemcpyDeviceToHost, a->spool[slot]));
-  HIP_CHECK(hipStreamSynchronize(a->spool[slot]));
+                           hipMemcpyDeviceToHost, s));
+  HIP_CHECK(hipStreamSynchronize(s));
 }
 
 static void dev_write_direct(Arena* a, uint64_t off, const uint8_t* src,
                              uint64_t n) {
   HIP_CHECK(hipSetDevice(a->device));
-  if (n <= TINY_THRESHOLD) {
-    HIP_CHECK(hipMemcpy((uint8_t*)a->base + off, src, n,
-                        hipMemcpyHostToDevice));
-    return;
-  }
-  uint32_t slot = a->srr.fetch_add(1) % SPOOL;
-  std::lock_guard<std::mutex> g(a->spool_mu[slot]);
+  hipStream_t s = thread_stream(a->device);
   HIP_CHECK(hipMemcpyAsync((uint8_t*)a->base + off, src, n,
-                           hipMemcpyHostToDevice, a->spool[slot]));
-  HIP_CHECK(hipStreamSynchronize(a->spool[slot]));
+                           hipMemcpyHostToDevice, s));
+  HIP_CHECK(hipStreamSynchronize(s));
 }
 
 // device -> host buffer, chunked through the pinned ring: D2H DMA of chunk
@@ -422,14 +419,12 @@ static void arena_read_ptr(int h, uint64_t off, uintptr_t dst, uint64_t n,
   check_range(a, off, n);
   py::gil_scoped_release rel;
   if (a->is_dev()) {
-    uint32_t slot = a->srr.fetch_add(1) % SPOOL;
-    std::lock_guard<std::mutex> g(a->spool_mu[slot]);
     HIP_CHECK(hipSetDevice(a->device));
+    hipStream_t s = thread_stream(a->device);
     HIP_CHECK(hipMemcpyAsync((void*)dst, (uint8_t*)a->base + off, n,
                              dst_is_device ? hipMemcpyDeviceToDevice
-                                           : hipMemcpyDeviceToHost,
-                             a->spool[slot]));
-    HIP_CHECK(hipStreamSynchronize(a->spool[slot]));
+                                           : hipMemcpyDeviceToHost, s));
+    HIP_CHECK(hipStreamSynchronize(s));
   } else if (dst_is_device) {
     HIP_CHECK(hipMemcpy((void*)dst, (uint8_t*)a->base + off, n,
                         hipMemcpyHostToDevice));
@@ -444,14 +439,12 @@ static void arena_write_ptr(int h, uint64_t off, uintptr_t src, uint64_t n,
   check_range(a, off, n);
   py::gil_scoped_release rel;
   if (a->is_dev()) {
-    uint32_t slot = a->srr.fetch_add(1) % SPOOL;
-    std::lock_guard<std::mutex> g(a->spool_mu[slot]);
     HIP_CHECK(hipSetDevice(a->device));
+    hipStream_t s = thread_stream(a->device);
     HIP_CHECK(hipMemcpyAsync((uint8_t*)a->base + off, (const void*)src, n,
                              src_is_device ? hipMemcpyDeviceToDevice
-                                           : hipMemcpyHostToDevice,
-                             a->spool[slot]));
-    HIP_CHECK(hipStreamSynchronize(a->spool[slot]));
+                                           : hipMemcpyHostToDevice, s));
+    HIP_CHECK(hipStreamSynchronize(s));
   } else if (src_is_device) {
     HIP_CHECK(hipMemcpy((uint8_t*)a->base + off, (const void*)src, n,
                         hipMemcpyDeviceToHost));
