@@ -241,6 +241,18 @@ class SyncWriteState:
         self._cur = self._cur_store = self._cur_lb = self._cur_async = None
         self._cur_pos = 0
 
+    _ZEROS = bytes(1 << 20)
+
+    def write_zeros(self, n: int) -> int:
+        """Sparse forward seek: fill the hole with zeros
+        (sparse_hole_flush_read_test analog)."""
+        remaining = n
+        while remaining > 0:
+            take = min(len(self._ZEROS), remaining)
+            self.write(memoryview(self._ZEROS)[:take])
+            remaining -= take
+        return n
+
     def flush(self) -> None:
         pass   # partial blocks are committed at complete()
 
@@ -692,9 +704,13 @@ class CurvineFuseFs:
         with h.lock:
             if h.writer is None:
                 raise OSError(errno.EBADF, "not open for write")
-            if offset != h.write_pos:
+            if offset > h.write_pos:
+                # forward seek: zero-fill the sparse hole
+                h.writer.write_zeros(offset - h.write_pos)
+                h.write_pos = offset
+            elif offset != h.write_pos:
                 raise OSError(errno.ENOTSUP,
-                              f"non-sequential write at {offset} (pos {h.write_pos})")
+                              f"backward write at {offset} (pos {h.write_pos})")
             h.writer.write(data, ptr=ptr)
             h.write_pos += size
         return abi.WRITE_OUT.pack(size, 0)

@@ -230,3 +230,17 @@ def test_shell_tools(mount):
     out = subprocess.run(["dd", f"if={src}", "of=/dev/null", "bs=256K"],
                          capture_output=True, check=True)
     assert b"1048576 bytes" in out.stderr
+
+
+def test_sparse_forward_seek_write(mount):
+    """cp --sparse / seek-past-EOF writes: the hole reads back as zeros."""
+    mnt, *_ = mount
+    with open(f"{mnt}/sparse.bin", "wb") as f:
+        f.write(b"head")
+        f.seek(3 << 20)
+        f.write(b"tail")
+    data = open(f"{mnt}/sparse.bin", "rb").read()
+    assert len(data) == (3 << 20) + 4
+    assert data[:4] == b"head"
+    assert data[-4:] == b"tail"
+    assert data[4:3 << 20] == b"\x00" * ((3 << 20) - 4)
