@@ -267,21 +267,30 @@ def step_client_rand4k(args, rank, rt, lat_out: list) -> int:
 
     from curvine_amd import native
     use_pinned = native.gpu_available()
+    depth = max(1, args.iodepth)
 
     def worker(t):
         try:
             rng = random.Random(t * 7919 + rank)
-            pbuf = native.PinnedBuffer(4096) if use_pinned else None
+            pbuf = native.PinnedBuffer(4096 * depth) if use_pinned else None
             buf = bytearray(4096) if pbuf is None else None
-            for _ in range(per_thread):
+            batches = per_thread // depth
+            for _ in range(max(1, batches)):
                 r = readers[rng.randrange(len(readers))]
-                off = rng.randrange(max(1, r.length - 4096))
                 t0 = time.perf_counter_ns()
-                if pbuf is not None:
-                    n = r.pread_into_ptr(off, pbuf.ptr, 4096)
+                if pbuf is not None and depth > 1:
+                    offs = [rng.randrange(max(1, r.length - 4096))
+                            for _ in range(depth)]
+                    r.pread_batch_ptr(offs, 4096, pbuf.ptr, 4096)
+                    n = 4096 * depth
+                elif pbuf is not None:
+                    n = r.pread_into_ptr(
+                        rng.randrange(max(1, r.length - 4096)), pbuf.ptr, 4096)
                 else:
-                    n = r.pread_into(off, buf, 0, 4096)
-                lats[t].append((time.perf_counter_ns() - t0) / 1000.0)
+                    n = r.pread_into(
+                        rng.randrange(max(1, r.length - 4096)), buf, 0, 4096)
+                # latency per IO (batch wall / depth, the fio convention)
+                lats[t].append((time.perf_counter_ns() - t0) / 1000.0 / depth)
                 total[t] += n
             if pbuf is not None:
                 pbuf.close()
@@ -368,6 +377,8 @@ def main():
     p.add_argument("--read-chunk", type=int, default=1 << 20)
     p.add_argument("--threads", type=int, default=16)
     p.add_argument("--rand-reads", type=int, default=200_000)
+    p.add_argument("--iodepth", type=int, default=16,
+                   help="queue depth per thread for randread4k (fio iodepth)")
     p.add_argument("--hbm-gb", type=int, default=16)
     p.add_argument("--staging-bytes", type=int, default=8 << 20)
     p.add_argument("--staging-count", type=int, default=8)
@@ -492,6 +503,7 @@ def main():
                 "file_size": args.file_size,
                 "read_chunk": args.read_chunk,
                 "threads": args.threads,
+                "iodepth": args.iodepth,
                 "block_size": args.block_size,
                 "fuse_channels": args.fuse_channels,
                 "GiBps": round(gibps, 3),

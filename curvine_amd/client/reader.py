@@ -106,6 +106,39 @@ class SyncLocalReader:
         got = self.pread_into(off, out, 0, len(out))
         return bytes(out[:got])
 
+    def pread_batch_ptr(self, file_offs: list[int], n: int, dst_ptr: int,
+                        stride: int) -> int:
+        """Batched fixed-size reads (fio iodepth analog): read n bytes at
+        each file offset into dst_ptr + i*stride.  Arena-resident reads
+        issue async on one thread-local stream with ONE sync; anything
+        else (file tier, block-spanning) falls back per-read.  Returns the
+        number of reads served."""
+        import bisect
+        import ctypes
+        from curvine_amd import native
+        groups: dict[int, tuple[list, list]] = {}  # arena handle -> offs, dsts
+        slow: list[tuple[int, int]] = []
+        for i, off in enumerate(file_offs):
+            idx = bisect.bisect_right(self._offs, off) - 1
+            lb = self.fb.blocks[idx]
+            boff = off - lb.offset
+            r = self._readers[idx]
+            meta = r.meta
+            if meta.get("kind") == "arena" and boff + n <= lb.block.length:
+                offs, dsts = groups.setdefault(r.layout.arena.handle, ([], []))
+                offs.append(meta["offset"] + boff)
+                dsts.append(dst_ptr + i * stride)
+            else:
+                slow.append((i, off))
+        mod = native.load()
+        for h, (offs, dsts) in groups.items():
+            mod.arena_read_batch(h, offs, dsts, n)
+        for i, off in slow:
+            buf = bytearray(n)
+            got = self.pread_into(off, buf, 0, n)
+            ctypes.memmove(dst_ptr + i * stride, bytes(buf[:got]), got)
+        return len(file_offs)
+
     def close(self) -> None:
         for r in getattr(self, "_readers", []):
             try:

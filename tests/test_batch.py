@@ -52,3 +52,30 @@ def test_batch_write_remote(tmp_path):
                 assert await fs.read_all(p) == data
             await fs.close()
     run(main())
+
+
+def test_pread_batch_ptr(tmp_path):
+    """Batched fixed-size reads land in the right slots."""
+    async def main():
+        async with MiniCluster(tmp_dir=str(tmp_path)) as mc:
+            fs = mc.fs()
+            data = os.urandom(6 << 20)   # spans two 4MB blocks
+            await fs.write_all("/b.bin", data)
+            r = await fs.open("/b.bin")
+            sr = r.to_sync()
+            from curvine_amd.native import PinnedBuffer
+            depth = 32
+            pbuf = PinnedBuffer(4096 * depth)
+            import random
+            rng = random.Random(9)
+            offs = [rng.randrange(len(data) - 4096) for _ in range(depth)]
+            offs[3] = (4 << 20) - 100   # block-spanning -> slow path
+            sr.pread_batch_ptr(offs, 4096, pbuf.ptr, 4096)
+            for i, off in enumerate(offs):
+                got = bytes(pbuf.view[i * 4096:(i + 1) * 4096])
+                assert got == data[off:off + 4096], f"slot {i} off {off}"
+            pbuf.close()
+            sr.close()
+            r.close()
+            await fs.close()
+    run(main())
