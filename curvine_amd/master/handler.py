@@ -239,6 +239,31 @@ class MasterHandler:
     def op_metricsreport(self, h, d):
         return {}
 
+    def op_getmetadatasnapshotpage(self, h, d):
+        """Page through the full inode table (CV-metadata sync,
+        master_filesystem.rs:836-875 analog)."""
+        offset = h.get("page_token", 0)
+        limit = min(h.get("limit", 1000), 10_000)
+        ids = sorted(self.fs.fs_dir.inodes)
+        page = [self.fs.fs_dir.inodes[i].to_state()
+                for i in ids[offset:offset + limit]]
+        nxt = offset + limit if offset + limit < len(ids) else None
+        return {"inodes": page, "next_token": nxt,
+                "op_id": self.master.journal.op_id}
+
+    def op_getmetadatadeltapage(self, h, d):
+        """Journal entries since op_id; requests older than the retained
+        window must fall back to a snapshot page sweep."""
+        since = h.get("since_op_id", 0)
+        recent = self.master.recent_entries
+        if recent and recent[0]["op_id"] > since + 1:
+            return {"snapshot_required": True,
+                    "op_id": self.master.journal.op_id}
+        limit = min(h.get("limit", 1000), 10_000)
+        out = [e for e in recent if e["op_id"] > since][:limit]
+        return {"entries": out, "op_id": self.master.journal.op_id,
+                "snapshot_required": False}
+
     def op_decommissionworker(self, h, d):
         self.fs.workers.decommission(h["worker_id"])
         # proactively queue re-replication of its blocks

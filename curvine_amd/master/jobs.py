@@ -18,11 +18,44 @@ from curvine_amd import errors as err
 log = logging.getLogger("curvine.jobs")
 
 
+class JobStore:
+    """Job persistence (transfer/store.rs analog): memory or sqlite."""
+
+    def __init__(self, kind: str = "memory", path: str = ""):
+        self.kind = kind
+        self._db = None
+        if kind == "sqlite":
+            import sqlite3
+            import os
+            os.makedirs(os.path.dirname(path) or ".", exist_ok=True)
+            self._db = sqlite3.connect(path)
+            self._db.execute(
+                "CREATE TABLE IF NOT EXISTS jobs (job_id TEXT PRIMARY KEY,"
+                " payload TEXT)")
+            self._db.commit()
+
+    def save(self, job: dict) -> None:
+        if self._db is not None:
+            import json
+            self._db.execute(
+                "INSERT OR REPLACE INTO jobs (job_id, payload) VALUES (?, ?)",
+                (job["job_id"], json.dumps(job, default=str)))
+            self._db.commit()
+
+    def load_all(self) -> dict:
+        if self._db is None:
+            return {}
+        import json
+        return {jid: json.loads(p) for jid, p in
+                self._db.execute("SELECT job_id, payload FROM jobs")}
+
+
 class JobManager:
     def __init__(self, master):
         self.master = master
-        self.jobs: dict[str, dict] = {}
-        self.next_id = 0
+        self.store = JobStore(master.conf.job.store, master.conf.job.store_path)
+        self.jobs: dict[str, dict] = self.store.load_all()
+        self.next_id = max((int(j.rsplit("-", 1)[-1]) for j in self.jobs), default=0)
 
     def submit(self, h: dict) -> dict:
         """h: {path, recursive, replicas}.  The path must be under a mount
@@ -41,6 +74,7 @@ class JobManager:
         }
         self.jobs[job_id] = job
         self._plan(job)
+        self.store.save(job)
         return {"job_id": job_id, "state": job["state"], "total": job["total"]}
 
     def _plan(self, job: dict) -> None:
@@ -94,6 +128,7 @@ class JobManager:
             raise err.JobNotFound(job_id)
         if job["state"] in ("planning", "running"):
             job["state"] = "cancelled"
+            self.store.save(job)
         return {"job_id": job_id, "state": job["state"]}
 
     def report_task(self, h: dict) -> None:
@@ -112,3 +147,4 @@ class JobManager:
             job["failed"] += 1
         if job["done"] + job["failed"] >= job["total"] and job["state"] == "running":
             job["state"] = "completed" if job["failed"] == 0 else "completed_with_failures"
+            self.store.save(job)

@@ -94,6 +94,7 @@ class JournalWriter:
         self._seg_start_op = 0
         self._seg_bytes = 0
         self.enabled = True
+        self.on_log = None   # optional hook(entry) after each append
 
     # ---- segment files: seg_<first_op_id>.wal ----
     def _segments(self) -> list[tuple[int, str]]:
@@ -126,6 +127,8 @@ class JournalWriter:
         if self.sync:
             self._f.flush()
             os.fsync(self._f.fileno())
+        if self.on_log is not None:
+            self.on_log(entry)
         return entry
 
     def flush(self) -> None:

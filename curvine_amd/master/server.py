@@ -87,6 +87,8 @@ class RaftJournalWriter(JournalWriter):
         got = self.raft.append_local(entry)
         assert got == index
         self.op_id = index
+        if self.on_log is not None:
+            self.on_log(entry)
         return entry
 
     def flush(self) -> None:
@@ -116,6 +118,10 @@ class Master:
         self.raft = None
         self.retry_cache = RetryCache(conf.master.retry_cache_size,
                                       conf.master.retry_cache_ttl_ms)
+        # recent journal entries for metadata delta paging (codes 28/29)
+        import collections
+        self.recent_entries = collections.deque(maxlen=20_000)
+        self.journal.on_log = self.recent_entries.append
         self.rpc = RpcServer("master", conf.master.hostname,
                              conf.master.rpc_port, MasterService(self))
         self._actor_task: Optional[asyncio.Task] = None

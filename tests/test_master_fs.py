@@ -162,3 +162,45 @@ def test_free_keeps_metadata(tmp_path):
     st = fs.file_status("/f")
     assert st.length == 10
     assert fs.open("/f").blocks == []
+
+
+def test_metadata_snapshot_and_delta_pages(tmp_path):
+    """CV-metadata sync paging (codes 28/29)."""
+    import asyncio
+
+    from curvine_amd.master.server import Master
+    from curvine_amd.conf import ClusterConf
+    from curvine_amd.rpc.client import RpcClient
+    from curvine_amd.rpc.codes import RpcCode
+
+    async def main():
+        conf = ClusterConf()
+        conf.master.rpc_port = 0
+        conf.journal.journal_dir = str(tmp_path / "j2")
+        m = await Master(conf).start()
+        c = await RpcClient("127.0.0.1", m.rpc.port).connect()
+        for i in range(25):
+            await c.rpc(RpcCode.Mkdir, {"path": f"/pg/d{i}",
+                                        "create_parents": True})
+        # snapshot pages
+        seen = []
+        token = 0
+        while token is not None:
+            r = await c.rpc(RpcCode.GetMetadataSnapshotPage,
+                            {"page_token": token, "limit": 10})
+            seen += r.header["inodes"]
+            token = r.header["next_token"]
+        assert len(seen) == len(m.fs.fs_dir.inodes)
+        op_now = r.header["op_id"]
+        # delta since op_now: empty
+        r = await c.rpc(RpcCode.GetMetadataDeltaPage, {"since_op_id": op_now})
+        assert r.header["entries"] == []
+        # mutate, delta catches it
+        await c.rpc(RpcCode.Mkdir, {"path": "/pg/extra",
+                                    "create_parents": True})
+        r = await c.rpc(RpcCode.GetMetadataDeltaPage, {"since_op_id": op_now})
+        assert any(e["op"] == "mkdir" and e.get("name") == "extra"
+                   for e in r.header["entries"])
+        await c.close()
+        await m.stop()
+    asyncio.new_event_loop().run_until_complete(main())
