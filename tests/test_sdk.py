@@ -99,3 +99,50 @@ def test_tensor_read_hbm_to_device(tmp_path):
         sf.shutdown()
     finally:
         smc.stop()
+
+
+def test_torch_dataloader_over_shards(cluster, tmp_path):
+    """WebDataset-style tar shards through a torch DataLoader."""
+    import io
+    import tarfile
+
+    import torch
+    from torch.utils.data import DataLoader
+
+    from curvine_amd.client.filesystem import SyncFs
+    from curvine_amd.sdk.dataset import CurvineShardDataset
+    sf = SyncFs(cluster.client_conf())
+    samples = {}
+    shard_paths = []
+    for s in range(3):
+        buf = io.BytesIO()
+        with tarfile.open(fileobj=buf, mode="w") as tf:
+            for i in range(10):
+                name = f"sample-{s}-{i}.bin"
+                payload = os.urandom(1000 + i)
+                samples[name] = payload
+                info = tarfile.TarInfo(name)
+                info.size = len(payload)
+                tf.addfile(info, io.BytesIO(payload))
+        path = f"/shards/shard-{s:03d}.tar"
+        sf.write_file(path, buf.getvalue())
+        shard_paths.append(path)
+    ds = CurvineShardDataset(cluster.client_conf(), shard_paths)
+    loader = DataLoader(ds, batch_size=None, num_workers=0)
+    seen = {}
+    for name, payload in loader:
+        seen[name] = bytes(payload)
+    assert seen == samples
+    sf.shutdown()
+
+
+def test_file_dataset(cluster):
+    from curvine_amd.sdk.dataset import CurvineFileDataset
+    from curvine_amd.client.filesystem import SyncFs
+    sf = SyncFs(cluster.client_conf())
+    for i in range(5):
+        sf.write_file(f"/ds/item{i}.bin", bytes([i]) * 100)
+    ds = CurvineFileDataset(cluster.client_conf(), "/ds")
+    assert len(ds) == 5
+    assert ds[2] == bytes([2]) * 100
+    sf.shutdown()
