@@ -58,7 +58,14 @@ class CurvineFileDataset:
         return self.transform(data) if self.transform else data
 
 
-class CurvineShardDataset:
+try:
+    from torch.utils.data import IterableDataset as _TorchIterable
+except ImportError:  # pragma: no cover
+    class _TorchIterable:  # type: ignore[no-redef]
+        pass
+
+
+class CurvineShardDataset(_TorchIterable):
     """Iterable dataset over tar shards (WebDataset layout): yields
     (name, bytes) per tar member, sharded across DataLoader workers."""
 
