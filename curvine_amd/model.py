@@ -168,8 +168,20 @@ class FileStatus:
     def is_symlink(self) -> bool:
         return self.file_type == FileType.SYMLINK
 
+    # manual dict (dataclasses.asdict deep-copies: ~10x slower on the
+    # metadata QPS hot path)
     def to_dict(self) -> dict:
-        return asdict(self)
+        return {
+            "inode_id": self.inode_id, "path": self.path, "name": self.name,
+            "file_type": self.file_type, "length": self.length,
+            "is_complete": self.is_complete, "block_size": self.block_size,
+            "replicas": self.replicas, "storage_tier": self.storage_tier,
+            "mtime_ms": self.mtime_ms, "atime_ms": self.atime_ms,
+            "mode": self.mode, "uid": self.uid, "gid": self.gid,
+            "ttl_ms": self.ttl_ms, "ttl_action": self.ttl_action,
+            "symlink_target": self.symlink_target, "nlink": self.nlink,
+            "xattrs": self.xattrs,
+        }
 
     @staticmethod
     def from_dict(d: dict) -> "FileStatus":
