@@ -372,10 +372,14 @@ class CurvineFuseFs:
             self._with_ino(st, node_id))
 
     @staticmethod
-    def _with_ino(st: FileStatus, ino: int) -> FileStatus:
+    def _with_ino(st: FileStatus, node_id: int) -> FileStatus:
+        # displayed st_ino = the MASTER's inode id (hardlinked names show
+        # the same ino); the FUSE nodeid (protocol handle) stays ours
+        if st.inode_id:
+            return st
         import copy
         st2 = copy.copy(st)
-        st2.inode_id = ino
+        st2.inode_id = node_id
         return st2
 
     def new_handle(self, node_id: int, path: str) -> FileHandle:
@@ -549,6 +553,24 @@ class CurvineFuseFs:
             self.call(self.fs.rename(dst, tmp))
             self.call(self.fs.rename(src, dst))
             self.call(self.fs.rename(tmp, src))
+            # the kernel swaps the dentry<->inode associations, so swap our
+            # nodeid<->path mapping the same way: each nodeid keeps naming
+            # its (moved) content and page caches stay coherent
+            with self.nodes_lock:
+                sp = self.nodes.get(nodeid)
+                dp = self.nodes.get(newdir)
+                na = sp.children.get(oldname) if sp else None
+                nb = dp.children.get(newname) if dp else None
+                if na is not None and nb is not None:
+                    sp.children[oldname] = nb
+                    dp.children[newname] = na
+                    self.nodes[na].parent = newdir
+                    self.nodes[na].name = newname
+                    self.nodes[nb].parent = nodeid
+                    self.nodes[nb].name = oldname
+            for nid in (na, nb):
+                if nid is not None:
+                    self.invalidate(nid)
         else:
             self.call(self.fs.rename(src, dst))
         with self.nodes_lock:

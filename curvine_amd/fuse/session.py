@@ -216,6 +216,20 @@ class FuseSession:
                  self.mnt_path, len(self.channels))
         return self
 
+    NOTIFY_INVAL_INODE = 2
+
+    def notify_inval_inode(self, nodeid: int) -> None:
+        """Push a kernel page/attr-cache invalidation for one inode
+        (needed when content moves beneath a path, e.g. RENAME_EXCHANGE)."""
+        import struct as _s
+        payload = _s.pack("<Qqq", nodeid, 0, -1)
+        hdr = abi.OUT_HEADER.pack(abi.OUT_HEADER_SIZE + len(payload),
+                                  self.NOTIFY_INVAL_INODE, 0)
+        try:
+            os.write(self.session_fd, hdr + payload)
+        except OSError:
+            pass   # ENOENT: kernel has nothing cached for it
+
     def stats(self) -> dict:
         from curvine_amd.metrics import OpStats
         return OpStats.merge([ch.stats for ch in self.channels])

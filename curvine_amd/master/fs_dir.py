@@ -280,17 +280,25 @@ class FsDir:
         node = self.must_resolve(path)
         if node.is_dir and node.children and not recursive:
             raise err.DirNotEmpty(path)
-        entry = self.journal.log(Op.DELETE, inode_id=node.id)
+        parent_path, name = split_path(path)
+        parent = self.must_resolve(parent_path)
+        entry = self.journal.log(Op.DELETE, inode_id=node.id,
+                                 parent_id=parent.id, name=name)
         return self._apply_delete(entry)
 
     def _apply_delete(self, e: dict) -> list[int]:
         node = self.inodes.get(e["inode_id"])
         if node is None:
             return []
-        parent = self.inodes.get(node.parent_id)
+        # the journal entry records WHICH dentry was unlinked (hardlinks)
+        parent = self.inodes.get(e.get("parent_id", node.parent_id))
+        name = e.get("name", node.name)
         if parent is not None and parent.children is not None:
-            parent.children.pop(node.name, None)
+            parent.children.pop(name, None)
             parent.mtime_ms = now_ms()
+        if node.file_type == FileType.FILE and node.nlink > 1:
+            node.nlink -= 1   # other hardlinked names keep the data
+            return []
         return self._drop_inode(node)
 
     def _drop_inode(self, node: Inode) -> list[int]:

@@ -1,0 +1,183 @@
+"""POSIX conformance corpus over the FUSE mount — the analog of the
+reference's targeted repro scripts (build/tests/scripts/*.py: mmap,
+renameat2 flags, O_TRUNC shared writer, read-past-EOF, hardlink
+semantics...)."""
+import ctypes
+import errno
+import mmap
+import os
+import stat
+import subprocess
+import sys
+
+import pytest
+
+requires_fuse = pytest.mark.skipif(
+    not os.path.exists("/dev/fuse") or os.geteuid() != 0,
+    reason="needs /dev/fuse and root")
+
+pytestmark = requires_fuse
+
+from tests.test_fuse import mount  # noqa: E402,F401 — reuse the fixture
+
+
+def test_read_past_eof(mount):
+    """fuse_read_past_eof_test.rs analog."""
+    mnt, *_ = mount
+    p = f"{mnt}/eof.bin"
+    with open(p, "wb") as f:
+        f.write(b"12345")
+    fd = os.open(p, os.O_RDONLY)
+    try:
+        assert os.pread(fd, 100, 0) == b"12345"
+        assert os.pread(fd, 10, 5) == b""
+        assert os.pread(fd, 10, 1000) == b""
+    finally:
+        os.close(fd)
+
+
+def test_mmap_read(mount):
+    """mmap over FUSE (kernel satisfies faults via READ).  The fault must
+    run in a SEPARATE process: this test process embeds the cache worker,
+    and a page fault holds the GIL (it is not a syscall), so a same-process
+    fault would deadlock against the worker serving it — a constraint of
+    embedding workers, not of the FUSE server (real apps are separate
+    processes)."""
+    mnt, *_ = mount
+    p = f"{mnt}/mm.bin"
+    data = os.urandom(128 << 10)
+    with open(p, "wb") as f:
+        f.write(data)
+    prog = (
+        "import mmap, sys\n"
+        f"f = open({p!r}, 'rb')\n"
+        "m = mmap.mmap(f.fileno(), 0, prot=mmap.PROT_READ)\n"
+        "raw = bytes(m)\n"
+        "sys.stdout.buffer.write(raw[:100] + raw[-10:])\n")
+    r = subprocess.run([sys.executable, "-c", prog], capture_output=True,
+                       timeout=60)
+    assert r.returncode == 0, r.stderr.decode()
+    assert r.stdout == data[:100] + data[-10:]
+
+
+def test_renameat2_noreplace(mount):
+    mnt, *_ = mount
+    a, b = f"{mnt}/ra.txt", f"{mnt}/rb.txt"
+    open(a, "wb").write(b"a")
+    open(b, "wb").write(b"b")
+    libc = ctypes.CDLL("libc.so.6", use_errno=True)
+    RENAME_NOREPLACE = 1
+    ret = libc.renameat2(-100, a.encode(), -100, b.encode(), RENAME_NOREPLACE)
+    assert ret != 0 and ctypes.get_errno() == errno.EEXIST
+    os.unlink(b)
+    ret = libc.renameat2(-100, a.encode(), -100, b.encode(), RENAME_NOREPLACE)
+    assert ret == 0
+    assert open(b, "rb").read() == b"a"
+
+
+def test_renameat2_exchange(mount):
+    mnt, *_ = mount
+    a, b = f"{mnt}/xa.txt", f"{mnt}/xb.txt"
+    open(a, "wb").write(b"AAA")
+    open(b, "wb").write(b"BBB")
+    libc = ctypes.CDLL("libc.so.6", use_errno=True)
+    RENAME_EXCHANGE = 2
+    ret = libc.renameat2(-100, a.encode(), -100, b.encode(), RENAME_EXCHANGE)
+    assert ret == 0, os.strerror(ctypes.get_errno())
+    assert open(a, "rb").read() == b"BBB"
+    assert open(b, "rb").read() == b"AAA"
+
+
+def test_hardlink_semantics(mount):
+    """link count, data shared, unlink one name keeps the data."""
+    mnt, *_ = mount
+    a, b = f"{mnt}/h1.txt", f"{mnt}/h2.txt"
+    with open(a, "wb") as f:
+        f.write(b"linked")
+    os.link(a, b)
+    assert os.stat(a).st_nlink == 2
+    assert os.stat(a).st_ino == os.stat(b).st_ino
+    os.unlink(a)
+    assert open(b, "rb").read() == b"linked"
+
+
+def test_o_excl(mount):
+    mnt, *_ = mount
+    p = f"{mnt}/excl.txt"
+    fd = os.open(p, os.O_CREAT | os.O_EXCL | os.O_WRONLY, 0o600)
+    os.close(fd)
+    with pytest.raises(OSError) as ei:
+        os.open(p, os.O_CREAT | os.O_EXCL | os.O_WRONLY)
+    assert ei.value.errno == errno.EEXIST
+
+
+def test_mode_bits_on_create(mount):
+    mnt, *_ = mount
+    p = f"{mnt}/modes.bin"
+    fd = os.open(p, os.O_CREAT | os.O_WRONLY, 0o640)
+    os.close(fd)
+    assert stat.S_IMODE(os.stat(p).st_mode) == 0o640
+
+
+def test_directory_mtime_updates_on_child_create(mount):
+    mnt, *_ = mount
+    d = f"{mnt}/mtdir"
+    os.mkdir(d)
+    m1 = os.stat(d).st_mtime_ns
+    import time
+    time.sleep(1.1)   # attr TTL + ms resolution
+    open(f"{d}/child", "wb").close()
+    m2 = os.stat(d).st_mtime_ns
+    assert m2 >= m1
+
+
+def test_many_small_files_listing(mount):
+    mnt, *_ = mount
+    d = f"{mnt}/many"
+    os.mkdir(d)
+    for i in range(300):
+        with open(f"{d}/f{i:04d}", "wb") as f:
+            f.write(b"x")
+    names = sorted(os.listdir(d))
+    assert len(names) == 300
+    assert names[0] == "f0000" and names[-1] == "f0299"
+
+
+def test_deep_paths(mount):
+    mnt, *_ = mount
+    path = mnt
+    for i in range(20):
+        path = f"{path}/d{i}"
+    os.makedirs(path)
+    with open(f"{path}/leaf.txt", "wb") as f:
+        f.write(b"deep")
+    assert open(f"{path}/leaf.txt", "rb").read() == b"deep"
+    # readdir down the chain
+    p = mnt
+    for i in range(20):
+        assert f"d{i}" in os.listdir(p)
+        p = f"{p}/d{i}"
+
+
+def test_fsync_and_datasync(mount):
+    mnt, *_ = mount
+    with open(f"{mnt}/sync.bin", "wb") as f:
+        f.write(b"data")
+        f.flush()
+        os.fsync(f.fileno())
+        os.fdatasync(f.fileno())
+
+
+def test_rename_open_file_keeps_reading(mount):
+    """POSIX: an open fd survives rename of its path."""
+    mnt, *_ = mount
+    p = f"{mnt}/moving.bin"
+    data = os.urandom(2 << 20)
+    with open(p, "wb") as f:
+        f.write(data)
+    f = open(p, "rb", buffering=0)
+    first = f.read(1 << 20)
+    os.rename(p, f"{mnt}/moved.bin")
+    rest = f.read()
+    f.close()
+    assert first + rest == data
