@@ -75,8 +75,12 @@ class RpcClient:
         self._fail_pending(ConnectError(f"{self.addr} closed"))
 
     def _fail_pending(self, err: Exception) -> None:
-        for q in self._pending.values():
-            q.put_nowait(err)
+        for sink in self._pending.values():
+            if isinstance(sink, asyncio.Future):
+                if not sink.done():
+                    sink.set_exception(err)
+            else:
+                sink.put_nowait(err)
         self._pending.clear()
 
     async def _rx_loop(self) -> None:
@@ -88,11 +92,15 @@ class RpcClient:
                     msg.set_header_bytes(await self._reader.readexactly(hlen))
                 if dlen:
                     msg.data = await self._reader.readexactly(dlen)
-                q = self._pending.get(msg.req_id)
-                if q is not None:
-                    q.put_nowait(msg)
+                sink = self._pending.get(msg.req_id)
+                if sink is None:
+                    log.debug("drop orphan reply req_id=%d code=%d",
+                              msg.req_id, msg.code)
+                elif isinstance(sink, asyncio.Future):
+                    if not sink.done():
+                        sink.set_result(msg)
                 else:
-                    log.debug("drop orphan reply req_id=%d code=%d", msg.req_id, msg.code)
+                    sink.put_nowait(msg)
         except asyncio.CancelledError:
             raise
         except Exception as e:  # noqa: BLE001 — propagate to callers
@@ -113,16 +121,14 @@ class RpcClient:
         if not self.is_connected:
             raise ConnectError(f"{self.addr} not connected")
         msg = Message.request(code, header, data)
-        q: asyncio.Queue = asyncio.Queue()
-        self._pending[msg.req_id] = q
+        fut = asyncio.get_running_loop().create_future()
+        self._pending[msg.req_id] = fut
         try:
             await self._send(msg)
             try:
-                reply = await asyncio.wait_for(q.get(), timeout or self.timeout)
+                reply = await asyncio.wait_for(fut, timeout or self.timeout)
             except asyncio.TimeoutError as e:
                 raise RpcTimeout(f"{RpcCode(code).name} to {self.addr} timed out") from e
-            if isinstance(reply, Exception):
-                raise reply
             return reply.raise_if_error()
         finally:
             self._pending.pop(msg.req_id, None)

@@ -37,7 +37,11 @@ class ServerConn:
             self.writer.write(parts[0])
             if parts[1]:
                 self.writer.write(parts[1])
-            await self.writer.drain()
+            # drain (an extra task switch) only when backpressure matters:
+            # large payloads, or the socket buffer has actually filled up
+            if len(parts[1]) >= (64 << 10) or \
+                    self.writer.transport.get_write_buffer_size() > (1 << 20):
+                await self.writer.drain()
 
     async def recv(self) -> Optional[Message]:
         try:
