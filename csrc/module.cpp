@@ -947,6 +947,9 @@ static void device_sync(int device) {
   HIP_CHECK(hipDeviceSynchronize());
 }
 
+// native FUSE data loop (needs Arena/get_arena/thread_stream above)
+#include "fuse_loop.hip"
+
 static py::dict device_mem_info(int device) {
   size_t free_b = 0, total_b = 0;
   HIP_CHECK(hipSetDevice(device));
@@ -983,6 +986,13 @@ PYBIND11_MODULE(_native, m) {
   m.def("pinned_free", &pinned_free);
   m.def("pinned_view", &pinned_view);
   m.def("pinned_ptr", &pinned_ptr);
+  m.def("fuse_loop_create", &fuse_loop_create);
+  m.def("fuse_loop_add_channel", &fuse_loop_add_channel);
+  m.def("fuse_loop_register", &fuse_loop_register);
+  m.def("fuse_loop_unregister", &fuse_loop_unregister);
+  m.def("fuse_loop_next_forward", &fuse_loop_next_forward);
+  m.def("fuse_loop_stats", &fuse_loop_stats);
+  m.def("fuse_loop_stop", &fuse_loop_stop);
   m.def("lz4_compress", &lz4_compress_py);
   m.def("lz4_decompress", &lz4_decompress_py);
   m.def("arena_lz4_decompress", &arena_lz4_decompress);

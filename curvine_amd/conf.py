@@ -177,6 +177,7 @@ class FuseConf:
     allow_other: bool = True
     state_file: str = "/tmp/curvine/fuse.state"
     kernel_cache: bool = True
+    native_loop: bool = True   # GIL-free C++ READ channels
 
 
 @dataclass

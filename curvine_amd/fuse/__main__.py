@@ -71,6 +71,9 @@ def main(argv=None) -> int:
     signal.signal(signal.SIGTERM, lambda *a: stop.set())
     signal.signal(signal.SIGINT, lambda *a: stop.set())
     signal.signal(signal.SIGUSR1, dump_stats)
+    import faulthandler
+    signal.signal(signal.SIGUSR2,
+                  lambda *a: faulthandler.dump_traceback(file=sys.stderr))
     stop.wait()
     dump_stats()
     daemon.stop()

@@ -70,7 +70,9 @@ class FuseDaemon:
         self.session = FuseSession(
             self.fuse_fs, self.mnt_path,
             channels=self.conf.fuse.mnt_number,
-            max_write=self.conf.fuse.max_write).start(session_fd=session_fd)
+            max_write=self.conf.fuse.max_write,
+            native_loop=self.conf.fuse.native_loop).start(
+                session_fd=session_fd)
         self._start_control_server()
         return self
 

@@ -666,6 +666,13 @@ class CurvineFuseFs:
                 fb = self.call(self.fs.client.open(path))
                 h.status = fb.status
                 h.reader = SyncReadState(self, fb)
+                if self.session is not None:
+                    # HBM/host-arena-resident files: READs served by the
+                    # GIL-free native loop from here on
+                    try:
+                        self.session.try_register_read(h.fh, h.reader)
+                    except Exception as e:  # noqa: BLE001
+                        log.debug("native read registration failed: %s", e)
             else:
                 if flags & os.O_TRUNC:
                     st = self.call(self.fs.client.create(path, overwrite=True))
@@ -726,6 +733,12 @@ class CurvineFuseFs:
         with h.lock:
             if h.writer is None:
                 raise OSError(errno.EBADF, "not open for write")
+            if h.writer._done:
+                # a dup'd fd wrote after an earlier close flushed the file:
+                # transparently reopen for append
+                fb = self.call(self.fs.client.append(h.path))
+                h.writer = SyncWriteState(self, fb.status, fb.blocks)
+                h.write_pos = h.writer.pos
             if offset > h.write_pos:
                 # forward seek: zero-fill the sparse hole
                 h.writer.write_zeros(offset - h.write_pos)
@@ -738,11 +751,16 @@ class CurvineFuseFs:
         return abi.WRITE_OUT.pack(size, 0)
 
     def op_flush(self, nodeid, body, ctx):
+        """FLUSH completes the file: it is the only write-side op that is
+        synchronous with close(2) (RELEASE is async, so completing there
+        races with an immediately following OPEN on another channel)."""
         fh, _u, _p, _lo = abi.FLUSH_IN.unpack_from(body, 0)
         h = self.handles.get(fh)
         if h is not None and h.writer is not None:
             with h.lock:
-                h.writer.flush()
+                if not h.writer._done:
+                    st = h.writer.complete()
+                    self.cache_status(h.node_id, st)
         return b""
 
     def op_fsync(self, nodeid, body, ctx):
@@ -759,6 +777,10 @@ class CurvineFuseFs:
                 st = h.writer.complete()
                 self.cache_status(h.node_id, st)
         if h.reader is not None:
+            # unregister from the native loop BEFORE closing store readers
+            # (registration pins the blocks via reader refcounts)
+            if self.session is not None:
+                self.session.unregister_read(fh)
             h.reader.close()
         with self.plock_mu:
             self.plocks.pop(fh, None)

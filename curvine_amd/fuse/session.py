@@ -180,16 +180,66 @@ class FuseChannel(threading.Thread):
             pass
 
 
+class ForwardChannel(threading.Thread):
+    """Python worker draining the native loop's forward queue (metadata
+    ops + unregistered reads); replies directly on the session fd."""
+
+    def __init__(self, session: "FuseSession", loop_id: int, idx: int):
+        super().__init__(daemon=True, name=f"fuse-fwd{idx}")
+        self.session = session
+        self.loop_id = loop_id   # NB: Thread already owns 'native_id'
+        self.fd = session.session_fd
+        self.idx = idx
+        from curvine_amd.metrics import OpStats
+        from curvine_amd.native import PinnedBuffer, load
+        self._native = load()
+        self.stats = OpStats()
+        self.reply_pin = PinnedBuffer(session.max_write + (64 << 10))
+        self.req_buf = None   # forwarded requests are plain bytes
+
+    def run(self) -> None:
+        import os as _os
+        import time as _t
+        trace = _os.environ.get("CV_FUSE_TRACE") == "1"
+        dispatch = FuseChannel.dispatch
+        while not self.session.stopped:
+            origin_fd, raw = self._native.fuse_loop_next_forward(
+                self.loop_id, 0.5)
+            if not raw:
+                continue
+            self.fd = origin_fd   # replies go to the fd that read the request
+            if trace:
+                (ln, opc, uniq, *_r) = abi.IN_HEADER.unpack_from(raw, 0)
+                t0 = _t.perf_counter()
+                log.info("fwd%d pull %s u=%d", self.idx,
+                         abi.Op.NAMES.get(opc, opc), uniq)
+            try:
+                dispatch(self, memoryview(raw))
+            except Exception as e:  # noqa: BLE001
+                log.exception("fuse forward dispatch: %s", e)
+            if trace:
+                log.info("fwd%d done %s u=%d %.1fms", self.idx,
+                         abi.Op.NAMES.get(opc, opc), uniq,
+                         ( _t.perf_counter() - t0) * 1000)
+
+    # reply plumbing shared with FuseChannel
+    reply = FuseChannel.reply
+    reply_error = FuseChannel.reply_error
+
+
 class FuseSession:
     def __init__(self, fs, mnt_path: str, channels: int = 1,
-                 max_write: int = 1 << 20, allow_other: bool = True):
+                 max_write: int = 1 << 20, allow_other: bool = True,
+                 native_loop: bool = True):
         self.fs = fs
         self.mnt_path = mnt_path
         self.n_channels = max(1, channels)
         self.max_write = max_write
+        self.native_loop = native_loop
+        self.native_id = None
         self.stopped = False
         self.session_fd = -1
-        self.channels: list[FuseChannel] = []
+        self.channels: list = []
         fs.session = self
 
     def start(self, session_fd: int | None = None) -> "FuseSession":
@@ -207,14 +257,53 @@ class FuseSession:
         fds = [self.session_fd]
         for i in range(1, self.n_channels):
             fds.append(clone_channel(self.session_fd))
-        for i, fd in enumerate(fds):
-            ch = FuseChannel(self, fd, i)
-            self.channels.append(ch)
-            ch.start()
-        log.info("fuse %s at %s (%d channels)",
+        if self.native_loop:
+            # GIL-free C++ channel threads serve registered READs; the rest
+            # is forwarded to Python ForwardChannel workers
+            from curvine_amd.native import load
+            native = load()
+            self.native_id = native.fuse_loop_create(self.max_write)
+            for fd in fds:
+                native.fuse_loop_add_channel(self.native_id, fd)
+            n_fwd = max(2, min(4, self.n_channels))
+            for i in range(n_fwd):
+                ch = ForwardChannel(self, self.native_id, i)
+                self.channels.append(ch)
+                ch.start()
+        else:
+            for i, fd in enumerate(fds):
+                ch = FuseChannel(self, fd, i)
+                self.channels.append(ch)
+                ch.start()
+        log.info("fuse %s at %s (%d channels%s)",
                  "adopted" if session_fd is not None else "mounted",
-                 self.mnt_path, len(self.channels))
+                 self.mnt_path, len(fds),
+                 ", native loop" if self.native_id is not None else "")
         return self
+
+    # ---------------- native read registration ----------------
+    def try_register_read(self, fh: int, srs) -> bool:
+        """Register an open read handle's arena extents with the native
+        loop (all blocks must be in-process arena-resident)."""
+        if self.native_id is None:
+            return False
+        exts = []
+        for idx, lb in enumerate(srs.fb.blocks):
+            r = srs._local.get(idx)
+            if r is None:
+                r = srs._open_local(idx, lb)
+            if r is None or r.meta.get("kind") != "arena":
+                return False
+            exts.append((lb.offset, lb.block.length,
+                         r.layout.arena.handle, r.meta["offset"]))
+        from curvine_amd.native import load
+        load().fuse_loop_register(self.native_id, fh, srs.length, exts)
+        return True
+
+    def unregister_read(self, fh: int) -> None:
+        if self.native_id is not None:
+            from curvine_amd.native import load
+            load().fuse_loop_unregister(self.native_id, fh)
 
     NOTIFY_INVAL_INODE = 2
 
@@ -232,7 +321,11 @@ class FuseSession:
 
     def stats(self) -> dict:
         from curvine_amd.metrics import OpStats
-        return OpStats.merge([ch.stats for ch in self.channels])
+        out = OpStats.merge([ch.stats for ch in self.channels])
+        if self.native_id is not None:
+            from curvine_amd.native import load
+            out["_native_loop"] = load().fuse_loop_stats(self.native_id)
+        return out
 
     def stop(self, umount_fs: bool = True) -> None:
         """umount_fs=False: hand-over shutdown — the kernel mount stays
@@ -240,11 +333,15 @@ class FuseSession:
         self.stopped = True
         if umount_fs:
             umount(self.mnt_path)
-        for ch in self.channels:
-            try:
-                os.close(ch.fd)
-            except OSError:
-                pass
+        if self.native_id is not None:
+            from curvine_amd.native import load
+            load().fuse_loop_stop(self.native_id)
+        else:
+            for ch in self.channels:
+                try:
+                    os.close(ch.fd)
+                except OSError:
+                    pass
         for ch in self.channels:
             ch.join(timeout=3)
         self.channels = []

@@ -21,6 +21,7 @@ _SO = os.path.join(_REPO, "curvine_amd", "_native.so")
 _SRC = os.path.join(_REPO, "csrc", "module.cpp")
 _SRC2 = os.path.join(_REPO, "csrc", "kernels.hip")
 _SRC3 = os.path.join(_REPO, "csrc", "lz4.hip")
+_SRC4 = os.path.join(_REPO, "csrc", "fuse_loop.hip")
 _lock = threading.Lock()
 _mod = None
 
@@ -30,7 +31,8 @@ def build_native(force: bool = False) -> str:
     if not force and os.path.exists(_SO) and os.path.exists(_SRC):
         if os.path.getmtime(_SO) >= max(os.path.getmtime(_SRC),
                                         os.path.getmtime(_SRC2),
-                                        os.path.getmtime(_SRC3)):
+                                        os.path.getmtime(_SRC3),
+                                        os.path.getmtime(_SRC4)):
             return _SO
     import pybind11
     hipcc = os.environ.get("HIPCC", "/opt/rocm/bin/hipcc")

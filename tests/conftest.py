@@ -13,6 +13,7 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 # take the GIL: a deadlock triangle.  Plain fork avoids the suspension.
 # (Production deployments run the daemon as its own process: cv-fuse.)
 subprocess._USE_POSIX_SPAWN = False
+subprocess._USE_VFORK = False   # fork_exec also vforks (bpo-35823)
 
 
 def pytest_configure(config):
