@@ -84,14 +84,26 @@ class RpcClient:
         self._pending.clear()
 
     async def _rx_loop(self) -> None:
+        buf = bytearray()
         try:
             while True:
-                proto = await self._reader.readexactly(PROTO_SIZE)
-                hlen, dlen, msg = Message.decode_proto(proto)
+                while len(buf) < PROTO_SIZE:
+                    chunk = await self._reader.read(256 << 10)
+                    if not chunk:
+                        raise ConnectionResetError("eof")
+                    buf += chunk
+                hlen, dlen, msg = Message.decode_proto(bytes(buf[:PROTO_SIZE]))
+                total = PROTO_SIZE + hlen + dlen
+                while len(buf) < total:
+                    chunk = await self._reader.read(256 << 10)
+                    if not chunk:
+                        raise ConnectionResetError("eof")
+                    buf += chunk
                 if hlen:
-                    msg.set_header_bytes(await self._reader.readexactly(hlen))
+                    msg.set_header_bytes(bytes(buf[PROTO_SIZE:PROTO_SIZE + hlen]))
                 if dlen:
-                    msg.data = await self._reader.readexactly(dlen)
+                    msg.data = bytes(buf[PROTO_SIZE + hlen:total])
+                del buf[:total]
                 sink = self._pending.get(msg.req_id)
                 if sink is None:
                     log.debug("drop orphan reply req_id=%d code=%d",
