@@ -31,6 +31,14 @@
 struct NExtent { uint64_t file_off, len; int arena; uint64_t arena_off; };
 struct NHandle { uint64_t length; std::vector<NExtent> ext; };
 
+// registered write window: sequential WRITEs append into the current
+// block's arena extent GIL-free; block boundaries forward to Python
+struct WHandle {
+  int arena;
+  uint64_t arena_off, capacity, start_off;
+  std::atomic<uint64_t> written{0};
+};
+
 struct FuseInHeader {
   uint32_t len, opcode;
   uint64_t unique, nodeid;
@@ -44,19 +52,28 @@ struct FuseReadIn {
   uint32_t flags, pad;
 };
 struct FuseOutHeader { uint32_t len; int32_t error; uint64_t unique; };
+struct FuseWriteIn {
+  uint64_t fh, offset;
+  uint32_t size, write_flags;
+  uint64_t lock_owner;
+  uint32_t flags, pad;
+};
+struct FuseWriteOut { uint32_t size, pad; };
 static const uint32_t FUSE_OP_READ = 15;
+static const uint32_t FUSE_OP_WRITE = 16;
 
 struct FuseLoop {
   std::vector<int> fds;
   size_t bufsize = (1u << 20) + (64u << 10);
   std::unordered_map<uint64_t, NHandle> handles;
+  std::unordered_map<uint64_t, std::unique_ptr<WHandle>> whandles;
   std::shared_mutex hmu;
   std::vector<std::thread> threads;
   std::deque<std::pair<int, std::string>> fwd;   // (origin fd, raw request)
   std::mutex fwd_mu;
   std::condition_variable fwd_cv;
   std::atomic<bool> stopping{false};
-  std::atomic<uint64_t> native_reads{0}, forwarded{0};
+  std::atomic<uint64_t> native_reads{0}, native_writes{0}, forwarded{0};
 };
 
 static std::vector<FuseLoop*> g_loops;
@@ -120,6 +137,48 @@ static bool serve_read(FuseLoop* L, int fd, const FuseInHeader* h,
   return true;
 }
 
+// serve one sequential WRITE natively; false -> forward to Python
+static bool serve_write(FuseLoop* L, int fd, const FuseInHeader* h,
+                        const FuseWriteIn* w, const uint8_t* payload,
+                        size_t payload_avail) {
+  if (w->size > payload_avail) return false;
+  std::shared_lock<std::shared_mutex> lk(L->hmu);
+  auto it = L->whandles.find(w->fh);
+  if (it == L->whandles.end()) return false;
+  WHandle* W = it->second.get();
+  uint64_t cur = W->written.load(std::memory_order_acquire);
+  if (w->offset != W->start_off + cur) return false;       // non-sequential
+  if (cur + w->size > W->capacity) return false;           // block boundary
+  if (!W->written.compare_exchange_strong(cur, cur + w->size))
+    return false;   // concurrent writer (dup fd): let Python arbitrate
+  Arena* a = get_arena(W->arena);
+  uint8_t* dst = (uint8_t*)a->base + W->arena_off + cur;
+  bool ok = true;
+  if (a->is_dev()) {
+    hipSetDevice(a->device);
+    hipStream_t s = thread_stream(a->device);
+    ok = hipMemcpyAsync(dst, payload, w->size, hipMemcpyHostToDevice, s) ==
+             hipSuccess &&
+         hipStreamSynchronize(s) == hipSuccess;
+  } else {
+    std::memcpy(dst, payload, w->size);
+  }
+  if (!ok) {   // roll the reservation back and let Python handle it
+    W->written.fetch_sub(w->size);
+    return false;
+  }
+  struct { FuseOutHeader oh; FuseWriteOut wo; } rep;
+  rep.oh.len = sizeof(rep);
+  rep.oh.error = 0;
+  rep.oh.unique = h->unique;
+  rep.wo.size = w->size;
+  rep.wo.pad = 0;
+  ssize_t wr = write(fd, &rep, sizeof(rep));
+  (void)wr;
+  L->native_writes.fetch_add(1, std::memory_order_relaxed);
+  return true;
+}
+
 static void loop_thread(FuseLoop* L, int fd) {
   const bool trace = getenv("CV_FUSE_NATIVE_TRACE") != nullptr;
   std::vector<uint8_t> req(L->bufsize);
@@ -158,6 +217,24 @@ static void loop_thread(FuseLoop* L, int fd) {
       if (r->size + sizeof(FuseOutHeader) <= L->bufsize &&
           serve_read(L, fd, h, r, reply))
         continue;
+    }
+    if (h->opcode == FUSE_OP_WRITE &&
+        n >= (ssize_t)(sizeof(FuseInHeader) + sizeof(FuseWriteIn))) {
+      const FuseWriteIn* w =
+          (const FuseWriteIn*)(req.data() + sizeof(FuseInHeader));
+      const uint8_t* payload =
+          req.data() + sizeof(FuseInHeader) + sizeof(FuseWriteIn);
+      size_t avail = n - sizeof(FuseInHeader) - sizeof(FuseWriteIn);
+      struct timespec t0, t1;
+      if (trace) clock_gettime(CLOCK_MONOTONIC, &t0);
+      bool served = serve_write(L, fd, h, w, payload, avail);
+      if (trace) {
+        clock_gettime(CLOCK_MONOTONIC, &t1);
+        fprintf(stderr, "[nloop fd%d] WRITE off=%llu size=%u served=%d %.2fms\n",
+                fd, (unsigned long long)w->offset, w->size, (int)served,
+                (t1.tv_sec - t0.tv_sec) * 1e3 + (t1.tv_nsec - t0.tv_nsec) / 1e6);
+      }
+      if (served) continue;
     }
     // anything else: forward the raw request to the Python control plane.
     // The origin fd travels with it: FUSE replies MUST be written to the
@@ -205,6 +282,38 @@ static void fuse_loop_register(int id, uint64_t fh, uint64_t length,
   L->handles[fh] = std::move(H);
 }
 
+static void fuse_loop_register_write(int id, uint64_t fh, int arena,
+                                     uint64_t arena_off, uint64_t capacity,
+                                     uint64_t start_off) {
+  FuseLoop* L = get_loop(id);
+  auto W = std::make_unique<WHandle>();
+  W->arena = arena;
+  W->arena_off = arena_off;
+  W->capacity = capacity;
+  W->start_off = start_off;
+  std::unique_lock<std::shared_mutex> lk(L->hmu);
+  L->whandles[fh] = std::move(W);
+}
+
+// remove the write window; returns bytes natively written into it
+static uint64_t fuse_loop_unregister_write(int id, uint64_t fh) {
+  FuseLoop* L = get_loop(id);
+  py::gil_scoped_release rel;
+  std::unique_lock<std::shared_mutex> lk(L->hmu);
+  auto it = L->whandles.find(fh);
+  if (it == L->whandles.end()) return 0;
+  uint64_t w = it->second->written.load();
+  L->whandles.erase(it);
+  return w;
+}
+
+static uint64_t fuse_loop_write_state(int id, uint64_t fh) {
+  FuseLoop* L = get_loop(id);
+  std::shared_lock<std::shared_mutex> lk(L->hmu);
+  auto it = L->whandles.find(fh);
+  return it == L->whandles.end() ? 0 : it->second->written.load();
+}
+
 static void fuse_loop_unregister(int id, uint64_t fh) {
   FuseLoop* L = get_loop(id);
   py::gil_scoped_release rel;   // may wait for in-flight reads
@@ -234,6 +343,7 @@ static py::dict fuse_loop_stats(int id) {
   FuseLoop* L = get_loop(id);
   py::dict d;
   d["native_reads"] = L->native_reads.load();
+  d["native_writes"] = L->native_writes.load();
   d["forwarded"] = L->forwarded.load();
   {
     std::shared_lock<std::shared_mutex> lk(L->hmu);

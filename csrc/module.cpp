@@ -990,6 +990,9 @@ PYBIND11_MODULE(_native, m) {
   m.def("fuse_loop_add_channel", &fuse_loop_add_channel);
   m.def("fuse_loop_register", &fuse_loop_register);
   m.def("fuse_loop_unregister", &fuse_loop_unregister);
+  m.def("fuse_loop_register_write", &fuse_loop_register_write);
+  m.def("fuse_loop_unregister_write", &fuse_loop_unregister_write);
+  m.def("fuse_loop_write_state", &fuse_loop_write_state);
   m.def("fuse_loop_next_forward", &fuse_loop_next_forward);
   m.def("fuse_loop_stats", &fuse_loop_stats);
   m.def("fuse_loop_stop", &fuse_loop_stop);

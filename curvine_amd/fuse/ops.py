@@ -40,7 +40,7 @@ class Node:
 
 class FileHandle:
     __slots__ = ("fh", "node_id", "path", "status", "reader", "writer",
-                 "write_pos", "flags", "dir_entries", "lock")
+                 "write_pos", "flags", "dir_entries", "lock", "nw_registered")
 
     def __init__(self, fh: int, node_id: int, path: str):
         self.fh = fh
@@ -53,6 +53,7 @@ class FileHandle:
         self.flags = 0
         self.dir_entries: Optional[list] = None
         self.lock = threading.Lock()
+        self.nw_registered = False   # native write window active
 
 
 class SyncReadState:
@@ -243,6 +244,14 @@ class SyncWriteState:
 
     _ZEROS = bytes(1 << 20)
 
+    def advance(self, n: int) -> None:
+        """Account bytes appended directly into the current block by the
+        native FUSE loop (the window never exceeds the block boundary)."""
+        self._cur_pos += n
+        self.pos += n
+        if self._cur_pos >= self.block_size:
+            self._commit_block()
+
     def write_zeros(self, n: int) -> int:
         """Sparse forward seek: fill the hole with zeros
         (sparse_hole_flush_read_test analog)."""
@@ -382,6 +391,43 @@ class CurvineFuseFs:
         st2.inode_id = node_id
         return st2
 
+    # ---------------- native write-window plumbing ----------------
+    def _sync_native_write(self, h: FileHandle) -> None:
+        """Fold natively-appended bytes back into the Python writer state
+        (destructive: the window is removed; re-registered after the next
+        Python-side write)."""
+        if not h.nw_registered or self.session is None or \
+                self.session.native_id is None:
+            return
+        from curvine_amd.native import load
+        n = load().fuse_loop_unregister_write(self.session.native_id, h.fh)
+        h.nw_registered = False
+        if n and h.writer is not None:
+            h.writer.advance(n)
+            h.write_pos += n
+
+    def _native_write_extra(self, h: FileHandle) -> int:
+        if not h.nw_registered or self.session is None or \
+                self.session.native_id is None:
+            return 0
+        from curvine_amd.native import load
+        return load().fuse_loop_write_state(self.session.native_id, h.fh)
+
+    def _register_native_write(self, h: FileHandle) -> None:
+        w = h.writer
+        if self.session is None or self.session.native_id is None or \
+                w is None or w._done or w._cur is None or \
+                w._cur.meta.get("kind") != "arena":
+            return
+        room = w.block_size - w._cur_pos
+        if room <= 0:
+            return
+        from curvine_amd.native import load
+        load().fuse_loop_register_write(
+            self.session.native_id, h.fh, w._cur.layout.arena.handle,
+            w._cur.meta["offset"] + w._cur_pos, room, h.write_pos)
+        h.nw_registered = True
+
     def new_handle(self, node_id: int, path: str) -> FileHandle:
         with self.handles_lock:
             fh = self.next_fh
@@ -469,7 +515,8 @@ class CurvineFuseFs:
                 if h.node_id == nodeid and h.writer is not None:
                     import copy
                     st2 = copy.copy(st)
-                    st2.length = max(st.length, h.write_pos)
+                    st2.length = max(st.length,
+                                     h.write_pos + self._native_write_extra(h))
                     return st2
         return st
 
@@ -731,6 +778,7 @@ class CurvineFuseFs:
         ptr = (req.ptr + abi.IN_HEADER_SIZE + abi.WRITE_IN.size
                if req is not None else None)
         with h.lock:
+            self._sync_native_write(h)
             if h.writer is None:
                 raise OSError(errno.EBADF, "not open for write")
             if h.writer._done:
@@ -748,6 +796,8 @@ class CurvineFuseFs:
                               f"backward write at {offset} (pos {h.write_pos})")
             h.writer.write(data, ptr=ptr)
             h.write_pos += size
+            # subsequent sequential WRITEs append GIL-free in C++
+            self._register_native_write(h)
         return abi.WRITE_OUT.pack(size, 0)
 
     def op_flush(self, nodeid, body, ctx):
@@ -758,6 +808,7 @@ class CurvineFuseFs:
         h = self.handles.get(fh)
         if h is not None and h.writer is not None:
             with h.lock:
+                self._sync_native_write(h)
                 if not h.writer._done:
                     st = h.writer.complete()
                     self.cache_status(h.node_id, st)
@@ -774,6 +825,7 @@ class CurvineFuseFs:
             return b""
         if h.writer is not None:
             with h.lock:
+                self._sync_native_write(h)
                 st = h.writer.complete()
                 self.cache_status(h.node_id, st)
         if h.reader is not None:
@@ -961,6 +1013,7 @@ class CurvineFuseFs:
             for h in list(self.handles.values()):
                 if h.writer is not None:
                     with h.lock:
+                        self._sync_native_write(h)
                         st = h.writer.complete()
                         h.writer = None
                         h.status = st
