@@ -941,6 +941,12 @@ static py::dict arena_info(int h) {
   return d;
 }
 
+// raw host->device copy for staged remote reads into consumer tensors
+static void memcpy_h2d(uintptr_t dst, uintptr_t src, uint64_t n) {
+  py::gil_scoped_release rel;
+  HIP_CHECK(hipMemcpy((void*)dst, (const void*)src, n, hipMemcpyHostToDevice));
+}
+
 static void device_sync(int device) {
   py::gil_scoped_release rel;
   HIP_CHECK(hipSetDevice(device));
@@ -964,6 +970,7 @@ PYBIND11_MODULE(_native, m) {
   m.doc() = "curvine_amd native data plane (HIP/CDNA4, gfx950)";
   m.def("device_count", &device_count);
   m.def("device_sync", &device_sync);
+  m.def("memcpy_h2d", &memcpy_h2d);
   m.def("device_mem_info", &device_mem_info);
   m.def("arena_create", &arena_create, py::arg("device"), py::arg("capacity"),
         py::arg("staging_bytes") = 4 << 20, py::arg("staging_count") = 8,

@@ -305,13 +305,18 @@ class FsReader:
             if isinstance(r, BlockReaderLocal):
                 rn = await r.read_to_device(boff, dst_ptr + got, want)
             else:
-                tmp = bytearray(want)
-                rn = await r.read_into(boff, tmp, 0, want)
+                # remote block: fetch to pinned host memory, then H2D
                 from curvine_amd import native
-                native.load().arena_write_ptr  # presence check
-                raise err.Unsupported(
-                    "device reads from remote workers need a local staging "
-                    "arena; use pread_into + upload")
+                pbuf = native.PinnedBuffer(want)
+                try:
+                    rn = await r.read_into(boff, pbuf.view, 0, want)
+                    if rn > 0:
+                        loop = asyncio.get_event_loop()
+                        await loop.run_in_executor(
+                            None, native.load().memcpy_h2d,
+                            dst_ptr + got, pbuf.ptr, rn)
+                finally:
+                    pbuf.close()
             got += rn
         return got
 

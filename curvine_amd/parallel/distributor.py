@@ -107,16 +107,16 @@ class BlockDistributor:
                 store.abort(block_id)
             raise
 
-    def broadcast_file(self, fs_sync_call, store, path: str,
+    def broadcast_file(self, open_file, store, path: str,
                        src_rank: int) -> dict:
         """Replicate every block of a cached file to all ranks' stores.
 
-        `fs_sync_call(coro)` runs client coroutines (rank-local master RPC).
+        `open_file(path) -> FileBlocks` resolves the block list on the
+        source rank (e.g. ``lambda p: sync_fs.call(fs.client.open(p))``).
         Returns {block_id: length}. The master learns the new replicas from
         each worker's next heartbeat (incremental added_blocks report)."""
-        # src resolves the block list; everyone gets it via broadcast_object
         if self.rank == src_rank:
-            fb = fs_sync_call("open", path)
+            fb = open_file(path)
             blocks = [(b.block.block_id, b.block.length) for b in fb.blocks]
             tier = fb.status.storage_tier
             obj = [blocks, tier]

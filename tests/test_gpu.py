@@ -162,3 +162,41 @@ def test_lz4_device_decompress(dev_arena):
         out = np.zeros(len(data), dtype=np.uint8)
         dev_arena.read(0, out, 0, len(data))
         assert out.tobytes() == data, f"lz4 case {i} device mismatch"
+
+
+def test_pread_to_device_mixed_remote(tmp_path):
+    """pread_to_device works for remote blocks via pinned staging."""
+    import asyncio
+
+    import torch
+
+    from curvine_amd.testing import MiniCluster, test_conf
+    from curvine_amd.worker import registry
+
+    async def main():
+        conf = test_conf(str(tmp_path))
+        mc = MiniCluster(conf=conf, tmp_dir=str(tmp_path),
+                         worker_dirs=[["[HBM:128MB:0]gpu0"]])
+        await mc.start()
+        try:
+            fs = mc.fs()
+            data = os.urandom(8 << 20)
+            await fs.write_all("/mix.bin", data, storage_tier="HBM")
+            t = torch.zeros(len(data), dtype=torch.uint8, device="cuda:0")
+            r = await fs.open("/mix.bin")
+            # force the remote path by hiding the in-process store
+            saved = dict(registry._stores)
+            registry._stores.clear()
+            try:
+                n = await r.pread_to_device(0, t.data_ptr(), len(data))
+            finally:
+                registry._stores.update(saved)
+            torch.cuda.synchronize()
+            assert n == len(data)
+            assert t.cpu().numpy().tobytes() == data
+            r.close()
+            await fs.close()
+        finally:
+            await mc.stop()
+
+    asyncio.new_event_loop().run_until_complete(main())
