@@ -10,18 +10,26 @@ dmabuf IPC, so colocated-in-process is the designed-for fast path.
 """
 from __future__ import annotations
 
+import os
 from typing import Optional
 
 _stores: dict[int, "object"] = {}
+_pids: dict[int, int] = {}
 
 
 def register(worker_id: int, store) -> None:
     _stores[worker_id] = store
+    _pids[worker_id] = os.getpid()
 
 
 def unregister(worker_id: int) -> None:
     _stores.pop(worker_id, None)
+    _pids.pop(worker_id, None)
 
 
 def lookup(worker_id: int):
+    """None after fork: a child inherits the registry dict but not usable
+    HIP contexts/arenas — forked DataLoader workers must go remote."""
+    if _pids.get(worker_id) != os.getpid():
+        return None
     return _stores.get(worker_id)

@@ -24,6 +24,9 @@ def main():
     p.add_argument("--shard-mb", type=int, default=512)
     p.add_argument("--sample-kb", type=int, default=256)
     p.add_argument("--epochs", type=int, default=3)
+    p.add_argument("--workers", type=int, default=0,
+                   help="DataLoader num_workers (forked workers read via "
+                        "RPC; in-process short-circuit is parent-only)")
     p.add_argument("--to-device", action="store_true",
                    help="land each sample in a cuda tensor")
     args = p.parse_args()
@@ -73,16 +76,19 @@ def main():
     total_mb = args.shards * args.shard_mb
 
     dev = torch.device("cuda:0") if has_gpu and args.to_device else None
+    # device transfer must happen in the PARENT when num_workers > 0
+    # (forked children cannot touch the HIP context)
+    move_in_transform = dev is not None and args.workers == 0
 
     def xform(item):
         name, payload = item
-        if dev is not None:
+        if move_in_transform:
             return torch.frombuffer(bytearray(payload), dtype=torch.uint8) \
                 .to(dev, non_blocking=True)
         return payload
 
     ds = CurvineShardDataset(cconf, shard_paths, transform=xform)
-    loader = DataLoader(ds, batch_size=None, num_workers=0)
+    loader = DataLoader(ds, batch_size=None, num_workers=args.workers)
 
     results = {"shards": args.shards, "shard_mb": args.shard_mb,
                "sample_kb": args.sample_kb,
@@ -93,6 +99,9 @@ def main():
         n = 0
         t0 = time.perf_counter()
         for item in loader:
+            if dev is not None and not move_in_transform:
+                item = torch.frombuffer(bytearray(item), dtype=torch.uint8) \
+                    .to(dev, non_blocking=True)
             n += item.numel() if hasattr(item, "numel") else len(item)
         if has_gpu:
             torch.cuda.synchronize()
