@@ -35,6 +35,8 @@ def main(argv=None) -> int:
     p.add_argument("--channels", type=int, default=0)
     p.add_argument("--data-dir", action="append", default=[],
                    help='worker data dir, e.g. "[HBM:200GB:0]gpu0"')
+    p.add_argument("--web-port", type=int, default=0,
+                   help="serve /metrics and /api/fuse on this port")
     p.add_argument("--takeover", action="store_true",
                    help="hot upgrade: adopt the running daemon's session "
                         "fd + open-handle state instead of mounting")
@@ -59,6 +61,16 @@ def main(argv=None) -> int:
 
     daemon = FuseDaemon(conf, args.mnt, embed_worker=args.embed_worker,
                         device_id=args.device).start(takeover=args.takeover)
+    web = None
+    if args.web_port:
+        from curvine_amd.web.server import WebServer
+
+        async def mkweb():
+            return await WebServer(conf, worker=daemon.worker,
+                                   fuse_session=daemon.session,
+                                   port=args.web_port,
+                                   host="127.0.0.1").start()
+        web = daemon.call(mkweb())
     print(f"READY {args.mnt}", flush=True)
 
     def dump_stats(*_a):
@@ -76,6 +88,11 @@ def main(argv=None) -> int:
                   lambda *a: faulthandler.dump_traceback(file=sys.stderr))
     stop.wait()
     dump_stats()
+    if web is not None:
+        try:
+            daemon.call(web.stop(), timeout=5)
+        except Exception:
+            pass
     daemon.stop()
     return 0
 

@@ -44,18 +44,19 @@ fetch('/api/info').then(r=>r.json()).then(d=>{
 
 class WebServer:
     def __init__(self, conf, master=None, worker=None, fuse_session=None,
-                 port: int | None = None):
+                 port: int | None = None, host: str | None = None):
         self.conf = conf
         self.master = master
         self.worker = worker
         self.fuse_session = fuse_session
+        self.host = host if host is not None else conf.master.hostname
         self.port = port if port is not None else (
             conf.master.web_port if master else conf.worker.web_port)
         self._server: Optional[asyncio.AbstractServer] = None
 
     async def start(self) -> "WebServer":
         self._server = await asyncio.start_server(
-            self._on_conn, self.conf.master.hostname, self.port,
+            self._on_conn, self.host, self.port,
             reuse_address=True)
         if self.port == 0:
             self.port = self._server.sockets[0].getsockname()[1]
@@ -151,6 +152,8 @@ class WebServer:
                 lines.append(f"curvine_worker_blocks{lbl} {s.block_num}")
         if self.fuse_session:
             for op, d in self.fuse_session.stats().items():
+                if not isinstance(d, dict) or "count" not in d:
+                    continue  # e.g. "_native_loop" raw counters
                 lines.append(f'curvine_fuse_ops_total{{op="{op}"}} {d["count"]}')
                 lines.append(
                     f'curvine_fuse_op_seconds_total{{op="{op}"}} {d["time_s"]}')

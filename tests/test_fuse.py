@@ -244,3 +244,48 @@ def test_sparse_forward_seek_write(mount):
     assert data[:4] == b"head"
     assert data[-4:] == b"tail"
     assert data[4:3 << 20] == b"\x00" * ((3 << 20) - 4)
+
+
+def test_web_metrics_endpoint(tmp_path):
+    """cv-fuse --web-port serves /api/fuse stats and prometheus /metrics."""
+    import json
+    import socket
+    import urllib.request
+
+    from curvine_amd.testing import SyncMiniCluster
+
+    smc = SyncMiniCluster(tmp_dir=str(tmp_path / "cv")).start()
+    mnt = f"/tmp/curvine-fuse-web-{os.getpid()}"
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "curvine_amd.fuse", "--mnt", mnt,
+         "--master", f"127.0.0.1:{smc.master.rpc.port}",
+         "--web-port", str(port), "--log-level", "WARNING"],
+        stdout=subprocess.PIPE, stderr=subprocess.DEVNULL, text=True,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    try:
+        line = proc.stdout.readline()
+        assert line.startswith("READY"), f"daemon failed: {line!r}"
+        with open(f"{mnt}/web.bin", "wb") as f:
+            f.write(b"x" * 65536)
+        assert open(f"{mnt}/web.bin", "rb").read(16) == b"x" * 16
+        stats = json.load(urllib.request.urlopen(
+            f"http://127.0.0.1:{port}/api/fuse", timeout=10))
+        assert any(isinstance(d, dict) and d.get("count", 0) > 0
+                   for d in stats.values()), stats
+        body = urllib.request.urlopen(
+            f"http://127.0.0.1:{port}/metrics", timeout=10).read().decode()
+        assert "curvine" in body
+    finally:
+        proc.terminate()
+        try:
+            proc.wait(timeout=10)
+        except subprocess.TimeoutExpired:
+            proc.kill()
+        smc.stop()
+        try:
+            os.rmdir(mnt)
+        except OSError:
+            pass
