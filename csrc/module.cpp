@@ -956,6 +956,9 @@ static void device_sync(int device) {
 // native FUSE data loop (needs Arena/get_arena/thread_stream above)
 #include "fuse_loop.hip"
 
+// native metadata RPC frontend (pure epoll/C++, no GPU involvement)
+#include "meta_server.cpp"
+
 static py::dict device_mem_info(int device) {
   size_t free_b = 0, total_b = 0;
   HIP_CHECK(hipSetDevice(device));
@@ -1003,6 +1006,17 @@ PYBIND11_MODULE(_native, m) {
   m.def("fuse_loop_next_forward", &fuse_loop_next_forward);
   m.def("fuse_loop_stats", &fuse_loop_stats);
   m.def("fuse_loop_stop", &fuse_loop_stop);
+  m.def("meta_create", &meta_create);
+  m.def("meta_stop", &meta_stop_srv);
+  m.def("meta_set_serving", &meta_set_serving);
+  m.def("meta_upsert", &meta_upsert);
+  m.def("meta_add_child", &meta_add_child);
+  m.def("meta_remove_child", &meta_remove_child);
+  m.def("meta_drop", &meta_drop);
+  m.def("meta_clear", &meta_clear);
+  m.def("meta_forward_pop", &meta_forward_pop);
+  m.def("meta_send", &meta_send);
+  m.def("meta_stats", &meta_stats);
   m.def("lz4_compress", &lz4_compress_py);
   m.def("lz4_decompress", &lz4_decompress_py);
   m.def("arena_lz4_decompress", &arena_lz4_decompress);

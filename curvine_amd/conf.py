@@ -106,6 +106,12 @@ class MasterConf:
     eviction_low_watermark: float = 0.85
     eviction_policy: str = "lru"  # lru | lfu | none
     ttl_check_ms: int = 5_000
+    # native C++ metadata frontend (csrc/meta_server.cpp): epoll threads
+    # serve FileStatus/ListStatus/Exists GIL-free, everything else is
+    # forwarded to the Python handler.  Falls back to the asyncio server
+    # if the native extension is unavailable.
+    native_meta: bool = True
+    meta_threads: int = 4
 
 
 @dataclass

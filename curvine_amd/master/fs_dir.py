@@ -105,6 +105,10 @@ class FsDir:
         self.next_block_id = 0
         root = Inode(ROOT_ID, "", FileType.DIR, 0o755)
         self.inodes[ROOT_ID] = root
+        # native metadata mirror (master/native_meta.MetaMirror): notified
+        # synchronously from every _apply_* so the C++ read path stays
+        # coherent with this tree
+        self.mirror = None
 
     # ---------------- lookup ----------------
     def resolve(self, path: str) -> Optional[Inode]:
@@ -180,6 +184,9 @@ class FsDir:
         self.inodes[node.id] = node
         self.inodes[e["parent_id"]].children[e["name"]] = node.id
         self.next_inode_id = max(self.next_inode_id, node.id)
+        if self.mirror:
+            self.mirror.upsert(node)
+            self.mirror.add_child(e["parent_id"], e["name"], node.id)
         return node
 
     def create(self, path: str, block_size: int, replicas: int,
@@ -228,6 +235,10 @@ class FsDir:
         parent.children[e["name"]] = node.id
         parent.mtime_ms = now_ms()
         self.next_inode_id = max(self.next_inode_id, node.id)
+        if self.mirror:
+            self.mirror.upsert(node)
+            self.mirror.upsert(parent)          # mtime changed
+            self.mirror.add_child(parent.id, e["name"], node.id)
         return node, removed
 
     def add_block(self, node: Inode, commit_prev_len: int = -1) -> int:
@@ -271,6 +282,8 @@ class FsDir:
                 rem -= blk[1]
         node.complete = True
         node.mtime_ms = now_ms()
+        if self.mirror:
+            self.mirror.upsert(node)
 
     def delete(self, path: str, recursive: bool = False) -> list[int]:
         """Returns deleted block ids (caller schedules worker deletes)."""
@@ -296,8 +309,13 @@ class FsDir:
         if parent is not None and parent.children is not None:
             parent.children.pop(name, None)
             parent.mtime_ms = now_ms()
+            if self.mirror:
+                self.mirror.remove_child(parent.id, name)
+                self.mirror.upsert(parent)
         if node.file_type == FileType.FILE and node.nlink > 1:
             node.nlink -= 1   # other hardlinked names keep the data
+            if self.mirror:
+                self.mirror.upsert(node)
             return []
         return self._drop_inode(node)
 
@@ -312,6 +330,8 @@ class FsDir:
                 self.block_index.pop(bid, None)
                 removed.append(bid)
             self.inodes.pop(n.id, None)
+            if self.mirror:
+                self.mirror.drop(n.id)
         return removed
 
     def rename(self, src: str, dst: str) -> None:
@@ -323,6 +343,8 @@ class FsDir:
         if dst.startswith(src + "/"):
             raise err.InvalidPath(f"cannot rename {src} into itself")
         node = self.must_resolve(src)
+        src_parent_path, src_name = split_path(src)
+        src_parent = self.must_resolve(src_parent_path)
         dst_parent_path, dst_name = split_path(dst)
         dst_parent = self.must_resolve(dst_parent_path)
         if not dst_parent.is_dir:
@@ -334,7 +356,10 @@ class FsDir:
                     raise err.DirNotEmpty(dst)
             elif node.is_dir:
                 raise err.NotDirectory(dst)
+        # record WHICH dentry moves: renaming a hardlink via a secondary
+        # name must not touch the primary dentry (same rule as delete)
         entry = self.journal.log(Op.RENAME, inode_id=node.id,
+                                 src_parent=src_parent.id, src_name=src_name,
                                  dst_parent=dst_parent.id, dst_name=dst_name)
         self._apply_rename(entry)
 
@@ -345,14 +370,22 @@ class FsDir:
         old_id = dst_parent.children.get(e["dst_name"])
         if old_id is not None and old_id != node.id:
             removed = self._drop_inode(self.inodes[old_id])
-        src_parent = self.inodes.get(node.parent_id)
+        src_parent = self.inodes.get(e.get("src_parent", node.parent_id))
+        old_name = e.get("src_name", node.name)
         if src_parent is not None and src_parent.children is not None:
-            src_parent.children.pop(node.name, None)
+            src_parent.children.pop(old_name, None)
             src_parent.mtime_ms = now_ms()
         node.parent_id = dst_parent.id
         node.name = e["dst_name"]
         dst_parent.children[node.name] = node.id
         dst_parent.mtime_ms = now_ms()
+        if self.mirror:
+            if src_parent is not None:
+                self.mirror.remove_child(src_parent.id, old_name)
+                self.mirror.upsert(src_parent)
+            self.mirror.upsert(node)            # name changed
+            self.mirror.add_child(dst_parent.id, node.name, node.id)
+            self.mirror.upsert(dst_parent)
         return removed
 
     def set_attr(self, node: Inode, **attrs) -> None:
@@ -367,6 +400,8 @@ class FsDir:
             if k in ("mode", "uid", "gid", "atime_ms", "mtime_ms", "ttl_ms",
                      "ttl_action", "replicas", "storage_tier"):
                 setattr(node, k, v)
+        if self.mirror:
+            self.mirror.upsert(node)
 
     def set_xattr(self, node: Inode, name: str, value: bytes) -> None:
         entry = self.journal.log(Op.SET_XATTR, inode_id=node.id,
@@ -377,6 +412,8 @@ class FsDir:
         node = self.inodes.get(e["inode_id"])
         if node is not None:
             node.xattrs[e["name"]] = e["value"]
+            if self.mirror:
+                self.mirror.upsert(node)
 
     def remove_xattr(self, node: Inode, name: str) -> None:
         entry = self.journal.log(Op.REMOVE_XATTR, inode_id=node.id, name=name)
@@ -386,6 +423,8 @@ class FsDir:
         node = self.inodes.get(e["inode_id"])
         if node is not None:
             node.xattrs.pop(e["name"], None)
+            if self.mirror:
+                self.mirror.upsert(node)
 
     def symlink(self, link_path: str, target: str) -> Inode:
         link_path = norm_path(link_path)
@@ -407,6 +446,9 @@ class FsDir:
         self.inodes[node.id] = node
         self.inodes[e["parent_id"]].children[e["name"]] = node.id
         self.next_inode_id = max(self.next_inode_id, node.id)
+        if self.mirror:
+            self.mirror.upsert(node)
+            self.mirror.add_child(e["parent_id"], e["name"], node.id)
         return node
 
     def link(self, src: str, dst: str) -> Inode:
@@ -430,6 +472,10 @@ class FsDir:
         parent.children[e["dst_name"]] = node.id
         node.nlink += 1
         parent.mtime_ms = now_ms()
+        if self.mirror:
+            self.mirror.upsert(node)
+            self.mirror.upsert(parent)
+            self.mirror.add_child(parent.id, e["dst_name"], node.id)
 
     def resize(self, node: Inode, new_length: int) -> list[int]:
         """Truncate. Returns block ids fully beyond the new length."""
@@ -455,6 +501,8 @@ class FsDir:
             off += blen if blen else node.block_size
         node.blocks = keep
         node.mtime_ms = now_ms()
+        if self.mirror:
+            self.mirror.upsert(node)
         return removed
 
     def free(self, node: Inode) -> list[int]:
@@ -471,6 +519,8 @@ class FsDir:
             self.block_index.pop(bid, None)
         node.blocks = []
         # not complete anymore in the cache sense; length metadata kept
+        if self.mirror:
+            self.mirror.upsert(node)
         return removed
 
     # ---------------- replay & snapshot ----------------

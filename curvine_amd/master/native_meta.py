@@ -1,0 +1,202 @@
+"""Native metadata RPC frontend glue.
+
+Pairs with csrc/meta_server.cpp: the C++ epoll threads own the master's
+listening socket and serve FileStatus/ListStatus/Exists (and keepalives)
+from a GIL-free mirror of the inode tree; every other frame is forwarded
+here and handled by the ordinary Python MasterHandler, the reply written
+back through ``meta_send``.
+
+This is the MI355X answer to the reference's tokio-native RPC server
+(crates/core/rpc/src/server/rpc_server.rs): the metadata read path —
+the QPS benchmark surface (curvine-tests fs bench) — never crosses the
+interpreter.
+
+Mirror maintenance: ``MetaMirror`` hangs off ``FsDir.mirror``; every
+``_apply_*`` (the shared live/replay choke point, fs_dir.py) notifies it
+synchronously, so a mutation is visible in the C++ tree before the
+mutating RPC's reply leaves the master.  Whole-state swaps (snapshot
+install, step-down rebuild) call ``attach()`` to re-prime the mirror.
+"""
+from __future__ import annotations
+
+import asyncio
+import logging
+import socket
+import threading
+from typing import Optional
+
+import msgpack
+
+from curvine_amd import native
+from curvine_amd.rpc.message import Message, PROTO_SIZE
+
+log = logging.getLogger("curvine.meta.native")
+
+
+def _pack_pairs(d: dict) -> tuple[bytes, int]:
+    """msgpack-encode a dict, returning (pairs-without-map-header, npairs)
+    so C++ can splice extra keys (``path``) into the same map."""
+    b = msgpack.packb(d, use_bin_type=True)
+    first = b[0]
+    if 0x80 <= first <= 0x8F:
+        return b[1:], first & 0xF
+    if first == 0xDE:
+        return b[3:], int.from_bytes(b[1:3], "big")
+    if first == 0xDF:
+        return b[5:], int.from_bytes(b[1:5], "big")
+    raise ValueError("not a msgpack map")
+
+
+def _node_blob(node) -> tuple[bytes, int]:
+    """FileStatus fields (fs_dir.status_of / model.FileStatus.to_dict)
+    minus the lookup-dependent ``path``."""
+    return _pack_pairs({
+        "inode_id": node.id, "name": node.name,
+        "file_type": int(node.file_type), "length": node.length,
+        "is_complete": node.complete, "block_size": node.block_size,
+        "replicas": node.replicas, "storage_tier": node.storage_tier,
+        "mtime_ms": node.mtime_ms, "atime_ms": node.atime_ms,
+        "mode": node.mode, "uid": node.uid, "gid": node.gid,
+        "ttl_ms": node.ttl_ms, "ttl_action": node.ttl_action,
+        "symlink_target": node.symlink_target, "nlink": node.nlink,
+        "xattrs": {k: bytes(v) for k, v in node.xattrs.items()},
+    })
+
+
+class MetaMirror:
+    """FsDir observer pushing inode state into the C++ tree."""
+
+    def __init__(self, lib, sid: int):
+        self.lib = lib
+        self.sid = sid
+
+    def upsert(self, node) -> None:
+        blob, n = _node_blob(node)
+        self.lib.meta_upsert(self.sid, node.id, node.is_dir, blob, n)
+
+    def add_child(self, parent_id: int, name: str, child_id: int) -> None:
+        self.lib.meta_add_child(self.sid, parent_id, name, child_id)
+
+    def remove_child(self, parent_id: int, name: str) -> None:
+        self.lib.meta_remove_child(self.sid, parent_id, name)
+
+    def drop(self, inode_id: int) -> None:
+        self.lib.meta_drop(self.sid, inode_id)
+
+
+class _FwdConn:
+    """Shim standing in for rpc.server.ServerConn on forwarded frames
+    (master handlers are unary and never touch it beyond attributes)."""
+    __slots__ = ("peer", "state")
+
+    def __init__(self, conn_id: int):
+        self.peer = f"native:{conn_id}"
+        self.state = {}
+
+
+class NativeMetaFrontend:
+    """Owns the listening socket + C++ server; drop-in for the master's
+    RpcServer slot (same .port / .start() / .stop() surface)."""
+
+    def __init__(self, master, nthreads: int = 4, fwd_batch: int = 256):
+        self.master = master
+        self.lib = native.load()
+        if not hasattr(self.lib, "meta_create"):
+            raise RuntimeError("_native.so lacks meta server (rebuild)")
+        conf = master.conf.master
+        sock = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
+        sock.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
+        sock.bind((conf.hostname, conf.rpc_port))
+        sock.listen(1024)
+        self.port = sock.getsockname()[1]
+        self.sid = self.lib.meta_create(sock.detach(), nthreads)
+        self.fwd_batch = fwd_batch
+        self._fwd_thread: Optional[threading.Thread] = None
+        self._loop: Optional[asyncio.AbstractEventLoop] = None
+        self._handler = None
+        self._stopped = False
+        self._conns: dict[int, _FwdConn] = {}
+
+    # ---------------- mirror ----------------
+    def attach(self) -> None:
+        """(Re)prime the C++ tree from the current FsDir and hook future
+        mutations.  Called at start and after snapshot-install/rebuild."""
+        fs_dir = self.master.fs.fs_dir
+        fs_dir.mirror = MetaMirror(self.lib, self.sid)
+        self.lib.meta_clear(self.sid)
+        up, ac = self.lib.meta_upsert, self.lib.meta_add_child
+        for node in fs_dir.inodes.values():
+            blob, n = _node_blob(node)
+            up(self.sid, node.id, node.is_dir, blob, n)
+            if node.children:
+                for name, cid in node.children.items():
+                    ac(self.sid, node.id, name, cid)
+
+    def set_serving(self, on: bool) -> None:
+        self.lib.meta_set_serving(self.sid, on)
+
+    def stats(self) -> dict:
+        return self.lib.meta_stats(self.sid)
+
+    # ---------------- lifecycle ----------------
+    async def start(self) -> None:
+        self._loop = asyncio.get_running_loop()
+        self._handler = self.master.rpc_service.get_message_handler()
+        self.attach()
+        self._fwd_thread = threading.Thread(
+            target=self._fwd_loop, name="meta-fwd", daemon=True)
+        self._fwd_thread.start()
+        raft = self.master.raft
+        self.set_serving(raft is None or raft.is_leader)
+        log.info("native meta frontend on :%d", self.port)
+
+    async def stop(self) -> None:
+        self._stopped = True
+        self.lib.meta_stop(self.sid)
+        if self._fwd_thread is not None:
+            await asyncio.get_running_loop().run_in_executor(
+                None, self._fwd_thread.join, 5.0)
+
+    # ---------------- forwarded frames ----------------
+    def _fwd_loop(self) -> None:
+        pop = self.lib.meta_forward_pop
+        while not self._stopped:
+            try:
+                items = pop(self.sid, 500, self.fwd_batch)
+            except Exception:
+                if self._stopped:
+                    return
+                raise
+            if not items or self._stopped:
+                continue
+            self._loop.call_soon_threadsafe(self._dispatch, items)
+
+    def _dispatch(self, items) -> None:
+        for conn_id, raw in items:
+            asyncio.ensure_future(self._handle_raw(conn_id, raw))
+
+    async def _handle_raw(self, conn_id: int, raw: bytes) -> None:
+        try:
+            hlen, dlen, msg = Message.decode_proto(raw[:PROTO_SIZE])
+            if hlen:
+                msg.set_header_bytes(raw[PROTO_SIZE:PROTO_SIZE + hlen])
+            if dlen:
+                msg.data = raw[PROTO_SIZE + hlen:PROTO_SIZE + hlen + dlen]
+            conn = self._conns.get(conn_id)
+            if conn is None:
+                conn = self._conns[conn_id] = _FwdConn(conn_id)
+                if len(self._conns) > 4096:   # bounded scratch
+                    self._conns.pop(next(iter(self._conns)))
+            try:
+                reply = await self._handler.handle(msg, conn)
+            except asyncio.CancelledError:
+                raise
+            except Exception as e:  # noqa: BLE001 — errors cross the wire
+                log.debug("fwd handler error code=%s: %s", msg.code, e)
+                reply = msg.error_reply(e)
+            if reply is not None:
+                self.lib.meta_send(self.sid, conn_id, reply.encode())
+        except asyncio.CancelledError:
+            raise
+        except Exception:  # noqa: BLE001
+            log.exception("forwarded frame dispatch failed")

@@ -152,6 +152,8 @@ class RaftNode:
         self.make_snapshot = make_snapshot
         self.load_snapshot_cb = load_snapshot
         self.rebuild = rebuild
+        # optional leadership observer (native meta frontend serving flag)
+        self.on_role_change = None
         self.election_timeout = election_timeout_ms / 1000.0
         self.heartbeat = heartbeat_ms / 1000.0
         self.next_index: dict[int, int] = {}
@@ -236,6 +238,8 @@ class RaftNode:
     # ---------------- election ----------------
     async def _run_election(self):
         self.state = CANDIDATE
+        if self.on_role_change is not None:
+            self.on_role_change(False)
         self.term += 1
         self.voted_for = self.id
         self._save_meta()
@@ -288,6 +292,8 @@ class RaftNode:
         # no-op entry in the new term: commits the previous-term backlog
         # (raft's no-commit-of-old-terms rule needs a current-term entry)
         self.append_local({"op": "noop", "op_id": self.log.last_index + 1})
+        if self.on_role_change is not None:
+            self.on_role_change(True)
         log.info("node %d is LEADER (term %d, last_index %d)",
                  self.id, self.term, self.log.last_index)
 
@@ -295,6 +301,10 @@ class RaftNode:
         was_leader = self.state == LEADER
         dirty = self.log.last_index > self.commit_index
         self.state = FOLLOWER
+        if self.on_role_change is not None:
+            # flip serving off BEFORE any state-machine rebuild below: the
+            # native read path must not race a mirror re-prime
+            self.on_role_change(False)
         if term > self.term:
             self.term = term
             self.voted_for = None

@@ -122,8 +122,10 @@ class Master:
         import collections
         self.recent_entries = collections.deque(maxlen=20_000)
         self.journal.on_log = self.recent_entries.append
+        self.rpc_service = MasterService(self)
         self.rpc = RpcServer("master", conf.master.hostname,
-                             conf.master.rpc_port, MasterService(self))
+                             conf.master.rpc_port, self.rpc_service)
+        self.native_meta = None   # NativeMetaFrontend when enabled
         self._actor_task: Optional[asyncio.Task] = None
         self._stopped = asyncio.Event()
         self._mutation_count = 0
@@ -159,6 +161,8 @@ class Master:
     def _load_snapshot_state(self, state: dict) -> None:
         self.fs.fs_dir.load_snapshot(state)
         self.mounts.load_snapshot(state.get("mounts", []))
+        if self.native_meta is not None:   # whole-state swap: re-prime
+            self.native_meta.attach()
 
     def _rebuild_state(self) -> None:
         """Re-derive the state machine from snapshot + committed raft log
@@ -178,6 +182,8 @@ class Master:
         for i in range(start + 1, self.raft.commit_index + 1):
             self._apply_entry(self.raft.log.entry_at(i))
         self.journal.op_id = self.raft.commit_index
+        if self.native_meta is not None:   # fs_dir was replaced: re-prime
+            self.native_meta.attach()
 
     # ---------------- lifecycle ----------------
     async def start(self) -> "Master":
@@ -192,9 +198,21 @@ class Master:
             self.journal.op_id = lg.last_index
         else:
             self._restore()
+        if self.conf.master.native_meta:
+            try:
+                from curvine_amd.master.native_meta import NativeMetaFrontend
+                self.native_meta = NativeMetaFrontend(
+                    self, nthreads=self.conf.master.meta_threads)
+                self.rpc = self.native_meta
+            except Exception as e:  # noqa: BLE001 — CPU-only fallback
+                log.warning("native meta frontend unavailable (%s); "
+                            "using asyncio rpc server", e)
+                self.native_meta = None
         await self.rpc.start()
         self.conf.master.rpc_port = self.rpc.port
         if self.raft is not None:
+            if self.native_meta is not None:
+                self.raft.on_role_change = self.native_meta.set_serving
             self.raft.start()
         self._actor_task = asyncio.create_task(self._actor_loop())
         log.info("master started on %s:%d%s", self.conf.master.hostname,
