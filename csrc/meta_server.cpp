@@ -332,8 +332,16 @@ static void meta_close_conn(MetaServer* S, const std::shared_ptr<MetaConn>& c) {
   if (was) return;
   epoll_ctl(S->epfd, EPOLL_CTL_DEL, c->fd, nullptr);
   close(c->fd);
-  std::lock_guard<std::mutex> g(S->conns_mu);
-  S->conns.erase(c->id);
+  {
+    std::lock_guard<std::mutex> g(S->conns_mu);
+    S->conns.erase(c->id);
+  }
+  {
+    // empty-frame sentinel: tells Python to reap this conn's drain task
+    std::lock_guard<std::mutex> g(S->fq_mu);
+    S->fq.emplace_back(c->id, std::string());
+  }
+  S->fq_cv.notify_one();
 }
 
 static std::string meta_proto(uint32_t hlen, uint32_t dlen, uint8_t code,

@@ -115,7 +115,11 @@ class NativeMetaFrontend:
         self._loop: Optional[asyncio.AbstractEventLoop] = None
         self._handler = None
         self._stopped = False
-        self._conns: dict[int, _FwdConn] = {}
+        # per-connection serial drain: pipelined requests on one socket
+        # must execute in arrival order (the asyncio server's per-conn
+        # loop gave that for free)
+        self._queues: dict[int, asyncio.Queue] = {}
+        self._tasks: dict[int, asyncio.Task] = {}
 
     # ---------------- mirror ----------------
     def attach(self) -> None:
@@ -156,6 +160,13 @@ class NativeMetaFrontend:
         if self._fwd_thread is not None:
             await asyncio.get_running_loop().run_in_executor(
                 None, self._fwd_thread.join, 5.0)
+        for t in self._tasks.values():
+            t.cancel()
+        if self._tasks:
+            await asyncio.gather(*self._tasks.values(),
+                                 return_exceptions=True)
+        self._tasks.clear()
+        self._queues.clear()
 
     # ---------------- forwarded frames ----------------
     def _fwd_loop(self) -> None:
@@ -173,20 +184,52 @@ class NativeMetaFrontend:
 
     def _dispatch(self, items) -> None:
         for conn_id, raw in items:
-            asyncio.ensure_future(self._handle_raw(conn_id, raw))
+            if not raw:                      # close sentinel from C++
+                t = self._tasks.pop(conn_id, None)
+                if t is not None:
+                    t.cancel()
+                self._queues.pop(conn_id, None)
+                continue
+            q = self._queues.get(conn_id)
+            if q is None:
+                q = self._queues[conn_id] = asyncio.Queue()
+                self._tasks[conn_id] = asyncio.ensure_future(
+                    self._drain(conn_id, q))
+            q.put_nowait(raw)
 
-    async def _handle_raw(self, conn_id: int, raw: bytes) -> None:
+    async def _drain(self, conn_id: int, q: asyncio.Queue) -> None:
+        """Serial handler loop for one connection; replies for frames that
+        were queued together go out in ONE meta_send."""
+        conn = _FwdConn(conn_id)
+        buf: list[bytes] = []
+        try:
+            while True:
+                raw = await q.get()
+                while True:
+                    enc = await self._handle_raw(raw, conn)
+                    if enc is not None:
+                        buf.append(enc)
+                    if q.empty():
+                        break
+                    raw = q.get_nowait()
+                if buf:
+                    ok = self.lib.meta_send(self.sid, conn_id, b"".join(buf))
+                    buf.clear()
+                    if not ok:
+                        break
+        except asyncio.CancelledError:
+            pass
+        finally:
+            self._tasks.pop(conn_id, None)
+            self._queues.pop(conn_id, None)
+
+    async def _handle_raw(self, raw: bytes, conn) -> Optional[bytes]:
         try:
             hlen, dlen, msg = Message.decode_proto(raw[:PROTO_SIZE])
             if hlen:
                 msg.set_header_bytes(raw[PROTO_SIZE:PROTO_SIZE + hlen])
             if dlen:
                 msg.data = raw[PROTO_SIZE + hlen:PROTO_SIZE + hlen + dlen]
-            conn = self._conns.get(conn_id)
-            if conn is None:
-                conn = self._conns[conn_id] = _FwdConn(conn_id)
-                if len(self._conns) > 4096:   # bounded scratch
-                    self._conns.pop(next(iter(self._conns)))
             try:
                 reply = await self._handler.handle(msg, conn)
             except asyncio.CancelledError:
@@ -194,9 +237,9 @@ class NativeMetaFrontend:
             except Exception as e:  # noqa: BLE001 — errors cross the wire
                 log.debug("fwd handler error code=%s: %s", msg.code, e)
                 reply = msg.error_reply(e)
-            if reply is not None:
-                self.lib.meta_send(self.sid, conn_id, reply.encode())
+            return reply.encode() if reply is not None else None
         except asyncio.CancelledError:
             raise
         except Exception:  # noqa: BLE001
             log.exception("forwarded frame dispatch failed")
+            return None
