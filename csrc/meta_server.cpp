@@ -30,6 +30,7 @@
 #include <sys/socket.h>
 #include <unistd.h>
 
+#include <algorithm>
 #include <atomic>
 #include <chrono>
 #include <condition_variable>
@@ -50,10 +51,13 @@
 static constexpr size_t kMetaProto = 22;
 static constexpr uint32_t kMetaMaxLen = 16u << 20;
 static constexpr uint8_t kCodeKeepalive = 1;
+static constexpr uint8_t kCodeOpenFile = 5;
 static constexpr uint8_t kCodeFileStatus = 7;
 static constexpr uint8_t kCodeListStatus = 8;
 static constexpr uint8_t kCodeExists = 9;
+static constexpr uint8_t kCodeGetBlockLocations = 13;
 static constexpr uint8_t kErrFileNotFound = 3;  // errors.ErrorCode
+static constexpr uint8_t kErrIsDirectory = 7;
 static constexpr int64_t kRootId = 1;
 
 static inline uint32_t rd_u32be(const uint8_t* p) {
@@ -262,6 +266,18 @@ struct MetaNode {
   uint32_t npairs = 0;
   std::string blob;                      // msgpack pairs sans "path"
   std::map<std::string, int64_t> children;  // sorted == Python sorted()
+  std::vector<std::pair<int64_t, int64_t>> blocks;  // (block_id, length)
+};
+
+struct LocEnt {
+  int64_t wid;
+  uint8_t ord;        // TIER_ORDER rank (stable-sort key, hottest first)
+  std::string tier;
+};
+
+struct WorkerRec {
+  std::string blob;   // msgpack map of WorkerAddress.to_dict()
+  bool lost = false;
 };
 
 struct MetaConn {
@@ -278,6 +294,10 @@ struct MetaServer {
   bool serving = false;  // guarded by tree_mu
   std::shared_mutex tree_mu;
   std::unordered_map<int64_t, MetaNode> nodes;
+  std::unordered_map<int64_t, WorkerRec> workers;        // guarded by tree_mu
+  std::unordered_map<int64_t, std::vector<LocEnt>> block_locs;
+  std::mutex acc_mu;   // open() access counters, drained by Python
+  std::unordered_map<int64_t, uint64_t> access_counts;
   std::mutex conns_mu;
   std::unordered_map<uint64_t, std::shared_ptr<MetaConn>> conns;
   std::atomic<uint64_t> next_conn{1};
@@ -286,7 +306,8 @@ struct MetaServer {
   std::condition_variable fq_cv;
   std::deque<std::pair<uint64_t, std::string>> fq;
   std::atomic<uint64_t> served_status{0}, served_list{0}, served_exists{0},
-      served_ping{0}, served_notfound{0}, forwarded{0}, conns_total{0};
+      served_open{0}, served_ping{0}, served_notfound{0}, forwarded{0},
+      conns_total{0};
 };
 
 static std::mutex g_meta_mu;
@@ -367,15 +388,16 @@ static void meta_reply(MetaServer* S, MetaConn* c, uint8_t code,
   if (!meta_write_all(c, out.data(), out.size())) c->dead.store(true);
 }
 
-static void meta_reply_notfound(MetaServer* S, MetaConn* c, uint8_t code,
-                                uint8_t req_status, uint64_t req_id,
-                                uint32_t seq, const std::string& path) {
+static void meta_reply_fserr(MetaServer* S, MetaConn* c, uint8_t code,
+                             uint8_t req_status, uint64_t req_id,
+                             uint32_t seq, uint8_t err_code,
+                             const std::string& msg) {
   std::string h;
   h.push_back(char(0x82));
   mp_str(h, "error_code", 10);
-  mp_uint(h, kErrFileNotFound);
+  mp_uint(h, err_code);
   mp_str(h, "error_msg", 9);
-  mp_str(h, path);
+  mp_str(h, msg);
   S->served_notfound.fetch_add(1, std::memory_order_relaxed);
   meta_reply(S, c, code, req_status, req_id, seq, h, /*error=*/true);
 }
@@ -452,11 +474,91 @@ static bool meta_serve(MetaServer* S, MetaConn* c, const uint8_t* frame,
       return true;
     }
     lk.unlock();
-    meta_reply_notfound(S, c, code, req_status, req_id, seq, path);
+    meta_reply_fserr(S, c, code, req_status, req_id, seq, kErrFileNotFound,
+                     path);
     return true;
   }
   const MetaNode& nd = S->nodes.find(id)->second;
   std::string h;
+  if (code == kCodeOpenFile || code == kCodeGetBlockLocations) {
+    if (nd.is_dir) {
+      lk.unlock();
+      meta_reply_fserr(S, c, code, req_status, req_id, seq, kErrIsDirectory,
+                       path);
+      return true;
+    }
+    h.push_back(char(0x81));
+    mp_str(h, "file_blocks", 11);
+    h.push_back(char(0x82));
+    mp_str(h, "status", 6);
+    meta_append_status(h, nd, norm);
+    mp_str(h, "blocks", 6);
+    size_t nb = nd.blocks.size();
+    if (nb < 16) {
+      h.push_back(char(0x90 | nb));
+    } else if (nb < 65536) {
+      h.push_back(char(0xdc));
+      wr_u16be(h, uint16_t(nb));
+    } else {
+      h.push_back(char(0xdd));
+      wr_u32be(h, uint32_t(nb));
+    }
+    uint64_t off = 0;
+    std::vector<const LocEnt*> locs;
+    for (const auto& blk : nd.blocks) {
+      // live locations, hottest tier first (locations_of analog; stable
+      // sort keeps heartbeat insertion order within a tier)
+      locs.clear();
+      auto bl = S->block_locs.find(blk.first);
+      if (bl != S->block_locs.end()) {
+        for (const auto& e : bl->second) {
+          auto w = S->workers.find(e.wid);
+          if (w != S->workers.end() && !w->second.lost) locs.push_back(&e);
+        }
+        std::stable_sort(locs.begin(), locs.end(),
+                         [](const LocEnt* a, const LocEnt* b) {
+                           return a->ord < b->ord;
+                         });
+      }
+      h.push_back(char(0x84));
+      mp_str(h, "block", 5);
+      h.push_back(char(0x83));
+      mp_str(h, "block_id", 8);
+      mp_uint(h, uint64_t(blk.first));
+      mp_str(h, "length", 6);
+      mp_uint(h, uint64_t(blk.second));
+      mp_str(h, "state", 5);
+      h.push_back(char(0x01));              // BlockState.FINALIZED
+      mp_str(h, "offset", 6);
+      mp_uint(h, off);
+      mp_str(h, "locations", 9);
+      size_t nl = locs.size();
+      if (nl < 16) {
+        h.push_back(char(0x90 | nl));
+      } else {
+        h.push_back(char(0xdc));
+        wr_u16be(h, uint16_t(nl));
+      }
+      for (const LocEnt* e : locs) h += S->workers.find(e->wid)->second.blob;
+      mp_str(h, "tiers", 5);
+      if (nl < 16) {
+        h.push_back(char(0x90 | nl));
+      } else {
+        h.push_back(char(0xdc));
+        wr_u16be(h, uint16_t(nl));
+      }
+      for (const LocEnt* e : locs) mp_str(h, e->tier);
+      off += uint64_t(blk.second);
+    }
+    S->served_open.fetch_add(1, std::memory_order_relaxed);
+    lk.unlock();
+    {
+      std::lock_guard<std::mutex> g(S->acc_mu);
+      S->access_counts[id]++;
+    }
+    meta_reply(S, c, code, req_status, req_id, seq, h);
+    return true;
+  }
   if (code == kCodeExists) {
     h.push_back(char(0x81));
     mp_str(h, "exists", 6);
@@ -514,7 +616,8 @@ static void meta_handle_frame(MetaServer* S, const std::shared_ptr<MetaConn>& c,
     return;
   }
   if ((code == kCodeFileStatus || code == kCodeListStatus ||
-       code == kCodeExists) &&
+       code == kCodeExists || code == kCodeOpenFile ||
+       code == kCodeGetBlockLocations) &&
       dlen == 0) {
     if (meta_serve(S, c.get(), frame, hlen, code, status, req_id, seq)) return;
   }
@@ -675,14 +778,20 @@ static void meta_set_serving(int64_t sid, bool on) {
 }
 
 static void meta_upsert(int64_t sid, int64_t id, bool is_dir,
-                        py::bytes blob, uint32_t npairs) {
+                        py::bytes blob, uint32_t npairs, py::bytes blocks) {
   MetaServer* S = meta_get(sid);
   std::string b = blob;
+  std::string bb = blocks;  // packed little-endian (block_id, length) i64 pairs
   std::unique_lock<std::shared_mutex> lk(S->tree_mu);
   MetaNode& nd = S->nodes[id];
   nd.is_dir = is_dir;
   nd.npairs = npairs;
   nd.blob = std::move(b);
+  nd.blocks.clear();
+  const int64_t* p = (const int64_t*)bb.data();
+  size_t n = bb.size() / 16;
+  nd.blocks.reserve(n);
+  for (size_t i = 0; i < n; i++) nd.blocks.emplace_back(p[2 * i], p[2 * i + 1]);
 }
 
 static void meta_add_child(int64_t sid, int64_t parent,
@@ -711,6 +820,61 @@ static void meta_clear(int64_t sid) {
   MetaServer* S = meta_get(sid);
   std::unique_lock<std::shared_mutex> lk(S->tree_mu);
   S->nodes.clear();
+}
+
+static void meta_worker_upsert(int64_t sid, int64_t wid, py::bytes blob,
+                               bool lost) {
+  MetaServer* S = meta_get(sid);
+  std::string b = blob;
+  std::unique_lock<std::shared_mutex> lk(S->tree_mu);
+  WorkerRec& w = S->workers[wid];
+  w.blob = std::move(b);
+  w.lost = lost;
+}
+
+static void meta_block_add_loc(int64_t sid, int64_t bid, int64_t wid,
+                               const std::string& tier, int ord) {
+  MetaServer* S = meta_get(sid);
+  std::unique_lock<std::shared_mutex> lk(S->tree_mu);
+  auto& v = S->block_locs[bid];
+  for (auto& e : v) {
+    if (e.wid == wid) {   // tier update (demotion) keeps position
+      e.tier = tier;
+      e.ord = uint8_t(ord);
+      return;
+    }
+  }
+  v.push_back(LocEnt{wid, uint8_t(ord), tier});
+}
+
+static void meta_block_remove_loc(int64_t sid, int64_t bid, int64_t wid) {
+  MetaServer* S = meta_get(sid);
+  std::unique_lock<std::shared_mutex> lk(S->tree_mu);
+  auto it = S->block_locs.find(bid);
+  if (it == S->block_locs.end()) return;
+  auto& v = it->second;
+  v.erase(std::remove_if(v.begin(), v.end(),
+                         [&](const LocEnt& e) { return e.wid == wid; }),
+          v.end());
+  if (v.empty()) S->block_locs.erase(it);
+}
+
+static void meta_block_drop(int64_t sid, int64_t bid) {
+  MetaServer* S = meta_get(sid);
+  std::unique_lock<std::shared_mutex> lk(S->tree_mu);
+  S->block_locs.erase(bid);
+}
+
+static py::dict meta_take_access(int64_t sid) {
+  MetaServer* S = meta_get(sid);
+  std::unordered_map<int64_t, uint64_t> taken;
+  {
+    std::lock_guard<std::mutex> g(S->acc_mu);
+    taken.swap(S->access_counts);
+  }
+  py::dict d;
+  for (auto& kv : taken) d[py::int_(kv.first)] = kv.second;
+  return d;
 }
 
 static py::list meta_forward_pop(int64_t sid, int timeout_ms, int max_items) {
@@ -752,6 +916,7 @@ static py::dict meta_stats(int64_t sid) {
   d["served_status"] = S->served_status.load();
   d["served_list"] = S->served_list.load();
   d["served_exists"] = S->served_exists.load();
+  d["served_open"] = S->served_open.load();
   d["served_ping"] = S->served_ping.load();
   d["served_notfound"] = S->served_notfound.load();
   d["forwarded"] = S->forwarded.load();

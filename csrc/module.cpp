@@ -1017,6 +1017,11 @@ PYBIND11_MODULE(_native, m) {
   m.def("meta_forward_pop", &meta_forward_pop);
   m.def("meta_send", &meta_send);
   m.def("meta_stats", &meta_stats);
+  m.def("meta_worker_upsert", &meta_worker_upsert);
+  m.def("meta_block_add_loc", &meta_block_add_loc);
+  m.def("meta_block_remove_loc", &meta_block_remove_loc);
+  m.def("meta_block_drop", &meta_block_drop);
+  m.def("meta_take_access", &meta_take_access);
   m.def("lz4_compress", &lz4_compress_py);
   m.def("lz4_decompress", &lz4_decompress_py);
   m.def("arena_lz4_decompress", &arena_lz4_decompress);

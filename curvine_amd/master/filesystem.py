@@ -134,7 +134,7 @@ class MasterFilesystem:
         for c in commits or []:
             for wid, tier in zip(c.get("locations", []),
                                  c.get("tiers", []) or ["MEM"] * len(c.get("locations", []))):
-                self.workers.block_locs.setdefault(c["block_id"], {})[wid] = tier
+                self.workers.add_location(c["block_id"], wid, tier)
         return self.fs_dir.status_of(node, norm_path(path))
 
     def delete(self, path: str, recursive: bool = False) -> int:

@@ -260,6 +260,8 @@ class FsDir:
         node.blocks.append([bid, 0])
         self.block_index[bid] = node.id
         self.next_block_id = max(self.next_block_id, bid)
+        if self.mirror:
+            self.mirror.upsert(node)   # block list feeds native open()
         return bid
 
     def complete_file(self, node: Inode, length: int,

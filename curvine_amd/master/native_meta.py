@@ -22,12 +22,15 @@ from __future__ import annotations
 import asyncio
 import logging
 import socket
+import struct
 import threading
 from typing import Optional
 
 import msgpack
 
 from curvine_amd import native
+from curvine_amd.conf import TIER_ORDER
+from curvine_amd.model import WorkerState, now_ms
 from curvine_amd.rpc.message import Message, PROTO_SIZE
 
 log = logging.getLogger("curvine.meta.native")
@@ -63,6 +66,16 @@ def _node_blob(node) -> tuple[bytes, int]:
     })
 
 
+def _pack_blocks(node) -> bytes:
+    if not node.blocks:
+        return b""
+    flat = []
+    for bid, blen in node.blocks:
+        flat.append(bid)
+        flat.append(blen)
+    return struct.pack(f"<{len(flat)}q", *flat)
+
+
 class MetaMirror:
     """FsDir observer pushing inode state into the C++ tree."""
 
@@ -72,7 +85,8 @@ class MetaMirror:
 
     def upsert(self, node) -> None:
         blob, n = _node_blob(node)
-        self.lib.meta_upsert(self.sid, node.id, node.is_dir, blob, n)
+        self.lib.meta_upsert(self.sid, node.id, node.is_dir, blob, n,
+                             _pack_blocks(node))
 
     def add_child(self, parent_id: int, name: str, child_id: int) -> None:
         self.lib.meta_add_child(self.sid, parent_id, name, child_id)
@@ -82,6 +96,30 @@ class MetaMirror:
 
     def drop(self, inode_id: int) -> None:
         self.lib.meta_drop(self.sid, inode_id)
+
+
+class WorkerMirror:
+    """WorkerManager observer: worker addresses + block locations, so the
+    C++ frontend can assemble OpenFile/GetBlockLocations replies."""
+
+    def __init__(self, lib, sid: int):
+        self.lib = lib
+        self.sid = sid
+
+    def upsert_worker(self, info) -> None:
+        blob = msgpack.packb(info.address.to_dict(), use_bin_type=True)
+        self.lib.meta_worker_upsert(self.sid, info.address.worker_id, blob,
+                                    int(info.state) == int(WorkerState.LOST))
+
+    def add_loc(self, block_id: int, worker_id: int, tier: str) -> None:
+        self.lib.meta_block_add_loc(self.sid, block_id, worker_id, tier,
+                                    TIER_ORDER.get(tier, 9))
+
+    def remove_loc(self, block_id: int, worker_id: int) -> None:
+        self.lib.meta_block_remove_loc(self.sid, block_id, worker_id)
+
+    def drop_block(self, block_id: int) -> None:
+        self.lib.meta_block_drop(self.sid, block_id)
 
 
 class _FwdConn:
@@ -123,21 +161,45 @@ class NativeMetaFrontend:
 
     # ---------------- mirror ----------------
     def attach(self) -> None:
-        """(Re)prime the C++ tree from the current FsDir and hook future
-        mutations.  Called at start and after snapshot-install/rebuild."""
+        """(Re)prime the C++ tree from the current FsDir + WorkerManager
+        and hook future mutations.  Called at start and after
+        snapshot-install/rebuild."""
         fs_dir = self.master.fs.fs_dir
-        fs_dir.mirror = MetaMirror(self.lib, self.sid)
+        mirror = MetaMirror(self.lib, self.sid)
+        fs_dir.mirror = mirror
         self.lib.meta_clear(self.sid)
         up, ac = self.lib.meta_upsert, self.lib.meta_add_child
         for node in fs_dir.inodes.values():
             blob, n = _node_blob(node)
-            up(self.sid, node.id, node.is_dir, blob, n)
+            up(self.sid, node.id, node.is_dir, blob, n, _pack_blocks(node))
             if node.children:
                 for name, cid in node.children.items():
                     ac(self.sid, node.id, name, cid)
+        workers = self.master.fs.workers
+        wmirror = WorkerMirror(self.lib, self.sid)
+        workers.mirror = wmirror
+        for info in workers.workers.values():
+            wmirror.upsert_worker(info)
+        for bid, locs in workers.block_locs.items():
+            for wid, tier in locs.items():
+                wmirror.add_loc(bid, wid, tier)
 
     def set_serving(self, on: bool) -> None:
         self.lib.meta_set_serving(self.sid, on)
+
+    def drain_access(self) -> None:
+        """Fold native open() access counters back into the Python inodes
+        (LFU/LRU eviction inputs; atime approximated to drain time)."""
+        counts = self.lib.meta_take_access(self.sid)
+        if not counts:
+            return
+        fs_dir = self.master.fs.fs_dir
+        t = now_ms()
+        for inode_id, n in counts.items():
+            node = fs_dir.inodes.get(inode_id)
+            if node is not None:
+                node.access_count += n
+                node.atime_ms = t
 
     def stats(self) -> dict:
         return self.lib.meta_stats(self.sid)

@@ -278,6 +278,9 @@ class Master:
                         self.replication.mark_under_replicated(bid)
                 self.replication.check_all()
                 self.replication.scan()
+                if self.native_meta is not None:
+                    # fold native open() counters into LFU/LRU inputs
+                    self.native_meta.drain_access()
                 self._ttl_sweep()
                 self._eviction_sweep()
                 if self.journal.op_id and tick % 60 == 0:

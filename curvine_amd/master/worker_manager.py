@@ -33,6 +33,9 @@ class WorkerManager:
         # pending commands per worker (delivered on heartbeat)
         self.commands: dict[int, list[dict]] = {}
         self._rr = 0
+        # native metadata mirror (master/native_meta.WorkerMirror): keeps
+        # the C++ OpenFile reply path's location map coherent
+        self.mirror = None
 
     # ---------------- registry ----------------
     def heartbeat(self, info: WorkerInfo,
@@ -44,22 +47,34 @@ class WorkerManager:
         if prev is not None and prev.state == WorkerState.DECOMMISSIONING:
             info.state = prev.state
         self.workers[wid] = info
+        if self.mirror:
+            self.mirror.upsert_worker(info)
         for b in added_blocks or []:
-            self.block_locs.setdefault(b["block_id"], {})[wid] = b.get("tier", "MEM")
+            self.add_location(b["block_id"], wid, b.get("tier", "MEM"))
         for bid in removed_blocks or []:
             locs = self.block_locs.get(bid)
             if locs:
                 locs.pop(wid, None)
                 if not locs:
                     self.block_locs.pop(bid, None)
+                if self.mirror:
+                    self.mirror.remove_loc(bid, wid)
         return self.commands.pop(wid, [])
+
+    def add_location(self, block_id: int, worker_id: int, tier: str) -> None:
+        """Single entry point for location inserts (heartbeat deltas,
+        block reports, client commit metadata) so the native mirror sees
+        every one."""
+        self.block_locs.setdefault(block_id, {})[worker_id] = tier
+        if self.mirror:
+            self.mirror.add_loc(block_id, worker_id, tier)
 
     def block_report(self, worker_id: int, blocks: list[dict]) -> list[int]:
         """Full report: reconcile; returns block ids the worker should NOT
         have (master has no record) so it can delete them
         (master_filesystem.rs:1289 analog). Caller supplies the valid set."""
         for b in blocks:
-            self.block_locs.setdefault(b["block_id"], {})[worker_id] = b.get("tier", "MEM")
+            self.add_location(b["block_id"], worker_id, b.get("tier", "MEM"))
         return []
 
     def check_expired(self) -> list[int]:
@@ -69,6 +84,8 @@ class WorkerManager:
                 if w.last_heartbeat_ms < deadline and w.state != WorkerState.LOST]
         for wid in lost:
             self.workers[wid].state = int(WorkerState.LOST)
+            if self.mirror:
+                self.mirror.upsert_worker(self.workers[wid])
             log.warning("worker %d expired -> LOST", wid)
         # callers run handle_lost_workers() to drop locations + re-replicate
         return lost
@@ -83,6 +100,8 @@ class WorkerManager:
                 affected.append(bid)
                 if not locs:
                     self.block_locs.pop(bid, None)
+                if self.mirror:
+                    self.mirror.remove_loc(bid, worker_id)
         return affected
 
     def decommission(self, worker_id: int) -> None:
@@ -90,6 +109,8 @@ class WorkerManager:
         if w is None:
             raise err.WorkerNotFound(str(worker_id))
         w.state = int(WorkerState.DECOMMISSIONING)
+        if self.mirror:
+            self.mirror.upsert_worker(w)
 
     def live_workers(self) -> list[WorkerInfo]:
         return [w for w in self.workers.values()
@@ -107,6 +128,8 @@ class WorkerManager:
             for wid in self.block_locs.get(bid, {}):
                 self.add_command(wid, {"cmd": CMD_DELETE_BLOCK, "block_id": bid})
             self.block_locs.pop(bid, None)
+            if self.mirror:
+                self.mirror.drop_block(bid)
 
     # ---------------- placement ----------------
     def choose_workers(self, count: int, policy: str = "local",
