@@ -175,3 +175,32 @@ def test_xattr_and_rename_visibility(cluster):
     names = [s.name for s in sf.list_status("/xv")]
     assert names == ["g"]
     sf.shutdown()
+
+
+def test_encoding_boundaries(cluster):
+    """Long (str8/str16) and unicode names, deep paths, >16-entry dirs —
+    the C++ msgpack emitters must agree with Python's decoder."""
+    from curvine_amd.client.filesystem import SyncFs
+
+    sf = SyncFs(cluster.client_conf())
+    long_name = "n" * 300                      # str16 path/name on the wire
+    uni = "δοκιμή-试验-🚀"
+    deep = "/b/" + "/".join(f"lvl{i}" for i in range(20))
+    sf.mkdir(deep, create_parents=True)
+    sf.write_file(f"/b/{long_name}", b"L")
+    sf.write_file(f"/b/{uni}", b"U" * 100)
+    for i in range(25):                        # array16 list in ListStatus
+        sf.write_file(f"/b/many/f{i:02d}", b"x")
+    st = sf.file_status(f"/b/{long_name}")
+    assert st.name == long_name and st.length == 1
+    st = sf.file_status(f"/b/{uni}")
+    assert st.name == uni and st.length == 100
+    assert sf.file_status(deep).is_dir
+    names = [s.name for s in sf.list_status("/b/many")]
+    assert names == sorted(f"f{i:02d}" for i in range(25))
+    # open with blocks through the native path
+    sf.write_file("/b/blocky", b"z" * (3 << 20))
+    assert sf.read_file("/b/blocky") == b"z" * (3 << 20)
+    stats = cluster.master.native_meta.stats()
+    assert stats["served_status"] >= 3
+    sf.shutdown()
