@@ -377,7 +377,7 @@ def main():
     p.add_argument("--read-chunk", type=int, default=1 << 20)
     p.add_argument("--threads", type=int, default=16)
     p.add_argument("--rand-reads", type=int, default=200_000)
-    p.add_argument("--iodepth", type=int, default=16,
+    p.add_argument("--iodepth", type=int, default=64,
                    help="queue depth per thread for randread4k (fio iodepth)")
     p.add_argument("--hbm-gb", type=int, default=16)
     p.add_argument("--staging-bytes", type=int, default=8 << 20)
@@ -474,7 +474,9 @@ def main():
             iops = (total_bytes / 4096) / elapsed
             metric = (f"{args.path}_rand_read_4k_IOPS")
             value, unit = round(iops, 1), "IOPS"
-            vs = round(gibps / REFERENCE_PEAK_RAND_GIBPS, 3)
+            # the reference publishes no 4 KiB IOPS figure (only 256 KiB
+            # random GiB/s and a ~100 µs latency class) — no ratio to quote
+            vs = None
         else:
             metric = ("fuse_seq_read_GiBps" if args.path == "fuse"
                       else "cached_seq_read_GiBps")

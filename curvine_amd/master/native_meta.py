@@ -122,6 +122,24 @@ class WorkerMirror:
         self.lib.meta_block_drop(self.sid, block_id)
 
 
+class _Delegate:
+    """Manual ``yield from`` for a coroutine that already yielded once:
+    re-yields its pending awaitable chain so a Task can finish driving it."""
+
+    def __init__(self, coro, first_yield):
+        self.coro = coro
+        self.first = first_yield
+
+    def __await__(self):
+        coro, y = self.coro, self.first
+        try:
+            while True:
+                sent = yield y
+                y = coro.send(sent)
+        except StopIteration as si:
+            return si.value
+
+
 class _FwdConn:
     """Shim standing in for rpc.server.ServerConn on forwarded frames
     (master handlers are unary and never touch it beyond attributes)."""
@@ -245,6 +263,12 @@ class NativeMetaFrontend:
             self._loop.call_soon_threadsafe(self._dispatch, items)
 
     def _dispatch(self, items) -> None:
+        # Fast path (no raft): master handlers are synchronous coroutines
+        # (the only await, raft.wait_commit, is skipped), so drive each to
+        # completion inline with coro.send(None) — no Queue, no Task — and
+        # batch every reply for a connection into one meta_send.
+        sync_ok = self.master.raft is None
+        replies: dict[int, list[bytes]] = {}
         for conn_id, raw in items:
             if not raw:                      # close sentinel from C++
                 t = self._tasks.pop(conn_id, None)
@@ -252,12 +276,55 @@ class NativeMetaFrontend:
                     t.cancel()
                 self._queues.pop(conn_id, None)
                 continue
+            if sync_ok and conn_id not in self._queues:
+                enc = self._handle_sync(conn_id, raw)
+                if enc is not None:
+                    replies.setdefault(conn_id, []).append(enc)
+                continue
             q = self._queues.get(conn_id)
             if q is None:
                 q = self._queues[conn_id] = asyncio.Queue()
                 self._tasks[conn_id] = asyncio.ensure_future(
                     self._drain(conn_id, q))
             q.put_nowait(raw)
+        for conn_id, bufs in replies.items():
+            self.lib.meta_send(self.sid, conn_id, b"".join(bufs))
+
+    def _handle_sync(self, conn_id: int, raw: bytes) -> Optional[bytes]:
+        """Drive one handler coroutine synchronously; falls back to the
+        ordered queue path if it unexpectedly suspends."""
+        hlen, dlen, msg = Message.decode_proto(raw[:PROTO_SIZE])
+        if hlen:
+            msg.set_header_bytes(raw[PROTO_SIZE:PROTO_SIZE + hlen])
+        if dlen:
+            msg.data = raw[PROTO_SIZE + hlen:PROTO_SIZE + hlen + dlen]
+        coro = self._handler.handle(msg, _FwdConn(conn_id))
+        try:
+            y = coro.send(None)
+        except StopIteration as si:
+            reply = si.value
+        except Exception as e:  # noqa: BLE001 — errors cross the wire
+            log.debug("fwd handler error code=%s: %s", msg.code, e)
+            reply = msg.error_reply(e)
+        else:
+            # suspended (shouldn't happen without raft): finish in a task;
+            # route later frames for this conn through the ordered queue
+            q = self._queues[conn_id] = asyncio.Queue()
+            self._tasks[conn_id] = asyncio.ensure_future(
+                self._finish_then_drain(conn_id, q, coro, y, msg))
+            return None
+        return reply.encode() if reply is not None else None
+
+    async def _finish_then_drain(self, conn_id, q, coro, first_yield, msg):
+        try:
+            reply = await _Delegate(coro, first_yield)
+        except asyncio.CancelledError:
+            raise
+        except Exception as e:  # noqa: BLE001
+            reply = msg.error_reply(e)
+        if reply is not None:
+            self.lib.meta_send(self.sid, conn_id, reply.encode())
+        await self._drain(conn_id, q)
 
     async def _drain(self, conn_id: int, q: asyncio.Queue) -> None:
         """Serial handler loop for one connection; replies for frames that
