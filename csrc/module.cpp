@@ -652,6 +652,58 @@ static void arena_gather(int h, std::vector<std::pair<uint64_t, uint64_t>> ext,
   }
 }
 
+// gather arena extents into a caller-supplied pointer with per-extent dst
+// offsets.  Device arena + device dst = pure on-chip D2D via
+// copy_extents_kernel (no host hop — the training-ingest fast path: tar
+// sample payloads living in the HBM cache land directly in a torch
+// device tensor).  Host arena = plain memcpy scatter into a host dst.
+static void arena_gather_ptr(
+    int h, std::vector<std::tuple<uint64_t, uint64_t, uint64_t>> ext,
+    uintptr_t dst_ptr) {
+  Arena* a = get_arena(h);
+  for (auto& e : ext) check_range(a, std::get<0>(e), std::get<2>(e));
+  uint8_t* dst = (uint8_t*)dst_ptr;
+  py::gil_scoped_release rel;
+  if (!a->is_dev()) {
+    for (auto& e : ext)
+      std::memcpy(dst + std::get<1>(e), (uint8_t*)a->base + std::get<0>(e),
+                  std::get<2>(e));
+    return;
+  }
+  std::lock_guard<std::mutex> g(a->mu);
+  HIP_CHECK(hipSetDevice(a->device));
+  std::vector<Extent> exts(ext.size());
+  std::vector<uint32_t> tile_ext;
+  std::vector<uint64_t> tile_off;
+  for (size_t i = 0; i < ext.size(); ++i) {
+    exts[i] = {std::get<0>(ext[i]), std::get<1>(ext[i]), std::get<2>(ext[i])};
+    for (uint64_t t = 0; t < exts[i].len; t += TILE) {
+      tile_ext.push_back((uint32_t)i);
+      tile_off.push_back(t);
+    }
+  }
+  size_t meta = exts.size() * sizeof(Extent) +
+                tile_ext.size() * (sizeof(uint32_t) + sizeof(uint64_t)) + 64;
+  ensure_scratch(a, meta);
+  Extent* d_ext = (Extent*)a->scratch;
+  uint32_t* d_te = (uint32_t*)(d_ext + exts.size());
+  uint64_t* d_to = (uint64_t*)(((uintptr_t)(d_te + tile_ext.size()) + 7) & ~7ull);
+  HIP_CHECK(hipMemcpyAsync(d_ext, exts.data(), exts.size() * sizeof(Extent),
+                           hipMemcpyHostToDevice, a->kstream));
+  HIP_CHECK(hipMemcpyAsync(d_te, tile_ext.data(),
+                           tile_ext.size() * sizeof(uint32_t),
+                           hipMemcpyHostToDevice, a->kstream));
+  HIP_CHECK(hipMemcpyAsync(d_to, tile_off.data(),
+                           tile_off.size() * sizeof(uint64_t),
+                           hipMemcpyHostToDevice, a->kstream));
+  int grid = (int)std::min<size_t>(tile_ext.size(), 8192);
+  hipLaunchKernelGGL(copy_extents_kernel, dim3(grid), dim3(WG), 0, a->kstream,
+                     (const uint8_t*)a->base, dst, d_ext, d_te, d_to,
+                     (uint32_t)tile_ext.size());
+  HIP_CHECK(hipGetLastError());
+  HIP_CHECK(hipStreamSynchronize(a->kstream));
+}
+
 static uintptr_t arena_base_ptr(int h) {
   return (uintptr_t)get_arena(h)->base;
 }
@@ -987,6 +1039,7 @@ PYBIND11_MODULE(_native, m) {
   m.def("arena_fill", &arena_fill);
   m.def("arena_crc32c", &arena_crc32c);
   m.def("arena_gather", &arena_gather);
+  m.def("arena_gather_ptr", &arena_gather_ptr);
   m.def("arena_base_ptr", &arena_base_ptr);
   m.def("arena_read_batch", &arena_read_batch);
   m.def("arena_info", &arena_info);

@@ -139,6 +139,49 @@ class SyncLocalReader:
             ctypes.memmove(dst_ptr + i * stride, bytes(buf[:got]), got)
         return len(file_offs)
 
+    def pread_gather(self, samples: list[tuple[int, int, int]],
+                     dst_ptr: int, dst_on_device: bool = False) -> int:
+        """Scatter-gather variable-size reads: for each (file_off, n,
+        dst_off), land bytes at dst_ptr + dst_off.  Arena-resident extents
+        are grouped per arena into ONE gather call (device arenas: the
+        on-chip copy_extents_kernel — the HBM-cache -> torch-device-tensor
+        training-ingest path with no host hop); everything else falls back
+        to pread_into_ptr DMA.  Returns total bytes gathered."""
+        import bisect
+        groups: dict[int, tuple[object, list]] = {}  # handle -> (arena, tri)
+        slow: list[tuple[int, int, int]] = []
+        total = 0
+        for off, n, doff in samples:
+            n = max(0, min(n, self.length - off))
+            got = 0
+            while got < n:
+                idx = bisect.bisect_right(self._offs, off + got) - 1
+                lb = self.fb.blocks[idx]
+                boff = off + got - lb.offset
+                want = min(n - got, lb.block.length - boff)
+                if want <= 0:
+                    break
+                r = self._readers[idx]
+                meta = r.meta
+                arena = getattr(r.layout, "arena", None)
+                is_arena = meta.get("kind") == "arena" and arena is not None
+                # gather_ptr needs src and dst on the same side (device
+                # kernel vs memcpy); cross-side extents use DMA fallback
+                same_side = is_arena and \
+                    ((arena.device >= 0) == bool(dst_on_device))
+                if same_side:
+                    _, tri = groups.setdefault(arena.handle, (arena, []))
+                    tri.append((meta["offset"] + boff, doff + got, want))
+                else:
+                    slow.append((off + got, want, doff + got))
+                got += want
+            total += got
+        for arena, tri in groups.values():
+            arena.gather_ptr(tri, dst_ptr)
+        for off, want, doff in slow:
+            self.pread_into_ptr(off, dst_ptr + doff, want)
+        return total
+
     def close(self) -> None:
         for r in getattr(self, "_readers", []):
             try:

@@ -136,6 +136,13 @@ class Arena:
         pipelined D2H on device arenas)."""
         self._n.arena_gather(self.handle, extents, out, out_off)
 
+    def gather_ptr(self, triples: list[tuple[int, int, int]],
+                   dst_ptr: int) -> None:
+        """Scatter-gather [(src_off, dst_off, len)...] into a raw pointer.
+        Device arena requires a device dst (on-chip D2D kernel, no host
+        hop); host arena requires a host dst (memcpy)."""
+        self._n.arena_gather_ptr(self.handle, triples, dst_ptr)
+
     def base_ptr(self) -> int:
         return self._n.arena_base_ptr(self.handle)
 

@@ -29,6 +29,10 @@ def main():
                         "RPC; in-process short-circuit is parent-only)")
     p.add_argument("--to-device", action="store_true",
                    help="land each sample in a cuda tensor")
+    p.add_argument("--device-loader", action="store_true",
+                   help="CurvineDeviceLoader: batched HBM->device gather "
+                        "(no host hop) instead of torch DataLoader")
+    p.add_argument("--batch-size", type=int, default=64)
     args = p.parse_args()
 
     import numpy as np
@@ -74,6 +78,35 @@ def main():
         shard_paths.append(path)
     ingest_s = time.perf_counter() - t0
     total_mb = args.shards * args.shard_mb
+
+    if args.device_loader:
+        from curvine_amd.sdk.dataset import CurvineDeviceLoader
+        dl = CurvineDeviceLoader(
+            cconf, shard_paths,
+            device="cuda:0" if has_gpu else "cpu",
+            batch_size=args.batch_size)
+        results = {"shards": args.shards, "shard_mb": args.shard_mb,
+                   "sample_kb": args.sample_kb, "mode": "device_loader",
+                   "samples": dl.num_samples,
+                   "ingest_GBps": round(total_mb / 1024 / ingest_s, 2),
+                   "tier": "HBM" if has_gpu else "MEM"}
+        epochs = []
+        for _ in range(args.epochs):
+            n = 0
+            t0 = time.perf_counter()
+            for tensor, sections, names in dl:
+                n += tensor.numel()
+            if has_gpu:
+                torch.cuda.synchronize()
+            dt = time.perf_counter() - t0
+            epochs.append(round(n / dt / 2**30, 3))
+        results["epoch_GiBps"] = epochs
+        results["sustained_GiBps"] = max(epochs)
+        print(json.dumps(results))
+        dl.close()
+        sf.shutdown()
+        smc.stop()
+        return
 
     dev = torch.device("cuda:0") if has_gpu and args.to_device else None
     # device transfer must happen in the PARENT when num_workers > 0

@@ -200,3 +200,51 @@ def test_pread_to_device_mixed_remote(tmp_path):
             await mc.stop()
 
     asyncio.new_event_loop().run_until_complete(main())
+
+
+def test_device_loader_hbm(tmp_path):
+    """CurvineDeviceLoader: tar samples cached in the HBM tier gathered
+    straight into a cuda tensor (copy_extents_kernel, no host hop)."""
+    import io
+    import tarfile
+
+    import torch
+
+    from curvine_amd.client.filesystem import SyncFs
+    from curvine_amd.sdk.dataset import CurvineDeviceLoader
+    from curvine_amd.testing import SyncMiniCluster, test_conf
+
+    conf = test_conf(str(tmp_path))
+    smc = SyncMiniCluster(conf=conf, tmp_dir=str(tmp_path),
+                          worker_dirs=[["[HBM:512MB:0]gpu0"]]).start()
+    try:
+        sf = SyncFs(smc.client_conf())
+        samples = {}
+        shard_paths = []
+        for s in range(2):
+            buf = io.BytesIO()
+            with tarfile.open(fileobj=buf, mode="w") as tf:
+                for i in range(40):
+                    name = f"g{s}-{i}"
+                    payload = os.urandom(64 << 10)
+                    samples[name] = payload
+                    info = tarfile.TarInfo(name)
+                    info.size = len(payload)
+                    tf.addfile(info, io.BytesIO(payload))
+            path = f"/gdl/shard-{s}.tar"
+            sf.write_file(path, buf.getvalue(), storage_tier="HBM")
+            shard_paths.append(path)
+        dl = CurvineDeviceLoader(smc.client_conf(), shard_paths,
+                                 device="cuda:0", batch_size=16)
+        assert dl.num_samples == 80
+        seen = {}
+        for tensor, sections, names in dl:
+            assert tensor.device.type == "cuda"
+            host = tensor.cpu().numpy().tobytes()
+            for (start, ln), name in zip(sections, names):
+                seen[name] = host[start:start + ln]
+        assert seen == samples
+        dl.close()
+        sf.shutdown()
+    finally:
+        smc.stop()

@@ -146,3 +146,44 @@ def test_file_dataset(cluster):
     assert len(ds) == 5
     assert ds[2] == bytes([2]) * 100
     sf.shutdown()
+
+
+def test_device_loader_cpu(cluster):
+    """CurvineDeviceLoader on the MEM tier with a cpu tensor: tar headers
+    indexed via short-circuit preads, payloads gathered host-side."""
+    import io
+    import tarfile
+
+    import torch
+
+    from curvine_amd.client.filesystem import SyncFs
+    from curvine_amd.sdk.dataset import CurvineDeviceLoader
+
+    sf = SyncFs(cluster.client_conf())
+    samples = {}
+    shard_paths = []
+    for s in range(3):
+        buf = io.BytesIO()
+        with tarfile.open(fileobj=buf, mode="w") as tf:
+            for i in range(12):
+                name = f"dl-{s}-{i}" + ("x" * 120 if i == 5 else "")
+                payload = os.urandom(700 + 37 * i)
+                samples[name] = payload
+                info = tarfile.TarInfo(name)
+                info.size = len(payload)
+                tf.addfile(info, io.BytesIO(payload))
+        path = f"/dl/shard-{s:03d}.tar"
+        sf.write_file(path, buf.getvalue())
+        shard_paths.append(path)
+
+    dl = CurvineDeviceLoader(cluster.client_conf(), shard_paths,
+                             device="cpu", batch_size=7, shuffle=True, seed=3)
+    assert dl.num_samples == len(samples)
+    seen = {}
+    for tensor, sections, names in dl:
+        assert tensor.dtype == torch.uint8
+        for (start, ln), name in zip(sections, names):
+            seen[name] = bytes(tensor[start:start + ln].numpy().tobytes())
+    assert seen == samples
+    dl.close()
+    sf.shutdown()
