@@ -450,6 +450,7 @@ class CurvineFuseFs:
                      abi.FUSE_PARALLEL_DIROPS | abi.FUSE_ATOMIC_O_TRUNC |
                      abi.FUSE_MAX_PAGES | abi.FUSE_CACHE_SYMLINKS |
                      abi.FUSE_HANDLE_KILLPRIV_V2 |
+                     abi.FUSE_DO_READDIRPLUS | abi.FUSE_READDIRPLUS_AUTO |
                      abi.FUSE_POSIX_LOCKS) & flags | abi.FUSE_MAX_PAGES
         max_write = self.conf.fuse.max_write
         return abi.INIT_OUT.pack(
@@ -871,6 +872,40 @@ class CurvineFuseFs:
             idx += 1
         return bytes(out)
 
+    def op_readdirplus(self, nodeid, body, ctx):
+        """One round trip for `ls -l`: each entry carries a full
+        fuse_entry_out (lookup + attrs) ahead of the dirent
+        (fuse_lowlevel readdirplus analog; entries count as lookups)."""
+        fh, offset, size, _rf, _lo, _fl, _ = abi.READ_IN.unpack_from(body, 0)
+        h = self.get_handle(fh)
+        if h.dir_entries is None:
+            sts = self.call(self.fs.list_status(h.path))
+            entries = [(".", None), ("..", None)]
+            entries += [(s.name, s) for s in sts]
+            h.dir_entries = entries
+        empty_entry = abi.ENTRY_OUT.pack(0, 0, 0, 0, 0, 0) + \
+            b"\x00" * abi.ATTR.size
+        out = bytearray()
+        idx = offset
+        while idx < len(h.dir_entries):
+            name, st = h.dir_entries[idx]
+            if st is None:
+                # "." / "..": dirent only, zeroed entry (no lookup taken)
+                entry = empty_entry
+                ino, dtype = nodeid, abi.DT_DIR
+            else:
+                node = self.child_node(h.node_id, name)
+                entry = self.entry_out(node, st)
+                ino = node.id
+                dtype = abi.DT_DIR if st.is_dir else (
+                    abi.DT_LNK if st.is_symlink else abi.DT_REG)
+            ent = entry + abi.pack_dirent(ino, idx + 1, name.encode(), dtype)
+            if len(out) + len(ent) > size:
+                break
+            out += ent
+            idx += 1
+        return bytes(out)
+
     def op_releasedir(self, nodeid, body, ctx):
         fh, _f, _rf, _lo = abi.RELEASE_IN.unpack_from(body, 0)
         with self.handles_lock:
@@ -1084,6 +1119,7 @@ CurvineFuseFs.HANDLERS = {
     abi.Op.RELEASE: CurvineFuseFs.op_release,
     abi.Op.OPENDIR: CurvineFuseFs.op_opendir,
     abi.Op.READDIR: CurvineFuseFs.op_readdir,
+    abi.Op.READDIRPLUS: CurvineFuseFs.op_readdirplus,
     abi.Op.RELEASEDIR: CurvineFuseFs.op_releasedir,
     abi.Op.FSYNCDIR: CurvineFuseFs.op_fsyncdir,
     abi.Op.STATFS: CurvineFuseFs.op_statfs,

@@ -171,6 +171,8 @@ class NativeMetaFrontend:
         self._loop: Optional[asyncio.AbstractEventLoop] = None
         self._handler = None
         self._stopped = False
+        import os
+        self._inline = os.environ.get("CURVINE_META_INLINE", "1") != "0"
         # per-connection serial drain: pipelined requests on one socket
         # must execute in arrival order (the asyncio server's per-conn
         # loop gave that for free)
@@ -267,7 +269,8 @@ class NativeMetaFrontend:
         # (the only await, raft.wait_commit, is skipped), so drive each to
         # completion inline with coro.send(None) — no Queue, no Task — and
         # batch every reply for a connection into one meta_send.
-        sync_ok = self.master.raft is None
+        # CURVINE_META_INLINE=0 forces the ordered-queue path (A/B knob).
+        sync_ok = self._inline and self.master.raft is None
         replies: dict[int, list[bytes]] = {}
         for conn_id, raw in items:
             if not raw:                      # close sentinel from C++

@@ -289,3 +289,25 @@ def test_web_metrics_endpoint(tmp_path):
             os.rmdir(mnt)
         except OSError:
             pass
+
+
+def test_readdirplus_ls_l(mount):
+    """`ls -l` via READDIRPLUS: names + sizes in one op (daemon advertises
+    FUSE_DO_READDIRPLUS); entries must carry correct attrs."""
+    mnt, *_ = mount
+    os.makedirs(f"{mnt}/plus/sub")
+    sizes = {}
+    for i in range(30):
+        with open(f"{mnt}/plus/f{i:02d}", "wb") as f:
+            f.write(b"a" * (100 + i))
+        sizes[f"f{i:02d}"] = 100 + i
+    out = subprocess.run(["ls", "-l", f"{mnt}/plus"],
+                         capture_output=True, text=True, check=True).stdout
+    for name, sz in sizes.items():
+        assert name in out
+        assert f" {sz} " in out.split(name)[0].rsplit("\n", 1)[-1] + " ", \
+            f"size {sz} missing for {name}"
+    # entries remain stat-able (nlookup bookkeeping sane across forgets)
+    with os.scandir(f"{mnt}/plus") as it:
+        got = {e.name: e.stat().st_size for e in it if e.is_file()}
+    assert got == sizes
