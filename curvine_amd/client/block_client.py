@@ -40,9 +40,12 @@ def factory() -> ClientFactory:
 class BlockWriterLocal:
     """Writes straight into the colocated worker's block store."""
 
-    def __init__(self, store, block_id: int, reserve: int, tier: str):
+    def __init__(self, store, block_id: int, reserve: int, tier: str,
+                 reopen: bool = False):
         self.store = store
-        self.writer = store.create_writer(block_id, reserve, tier)
+        self.reopen = reopen
+        self.writer = (store.reopen_writer(block_id) if reopen
+                       else store.create_writer(block_id, reserve, tier))
         self.block_id = block_id
         self.pos = 0
 
@@ -51,13 +54,20 @@ class BlockWriterLocal:
         await loop.run_in_executor(None, self.writer.write, data)
         self.pos += len(data)
 
-    async def commit(self, length: int) -> str:
+    async def pwrite(self, off: int, data) -> None:
+        loop = asyncio.get_event_loop()
+        await loop.run_in_executor(None, self.writer.pwrite, off, data)
+
+    async def commit(self, length: int | None = None) -> str:
+        if length is None:       # in-place rewrite: nothing to publish
+            return ""
         loop = asyncio.get_event_loop()
         return await loop.run_in_executor(None, self.store.finalize,
                                           self.block_id, length)
 
     async def abort(self) -> None:
-        self.store.abort(self.block_id)
+        if not self.reopen:
+            self.store.abort(self.block_id)
 
 
 class BlockWriterRemote:
@@ -66,11 +76,12 @@ class BlockWriterRemote:
     WINDOW = 4
 
     def __init__(self, addr: WorkerAddress, block_id: int, reserve: int,
-                 tier: str):
+                 tier: str, reopen: bool = False):
         self.addr = addr
         self.block_id = block_id
         self.reserve = reserve
         self.tier = tier
+        self.reopen = reopen
         self.stream: Optional[RpcStream] = None
         self.inflight = 0
         self.pos = 0
@@ -82,7 +93,7 @@ class BlockWriterRemote:
         self.stream = client.stream(RpcCode.WriteBlock)
         reply = await self.stream.call(
             {"block_id": self.block_id, "reserve": self.reserve,
-             "tier": self.tier}, status=Status.Open)
+             "tier": self.tier, "reopen": self.reopen}, status=Status.Open)
 
     async def write(self, data) -> None:
         await self._ensure_open()
@@ -93,12 +104,21 @@ class BlockWriterRemote:
             await self.stream.recv()
             self.inflight -= 1
 
-    async def commit(self, length: int) -> str:
+    async def pwrite(self, off: int, data) -> None:
+        await self._ensure_open()
+        await self.stream.send({"off": off}, bytes(data), Status.Running)
+        self.inflight += 1
+        while self.inflight >= self.WINDOW:
+            await self.stream.recv()
+            self.inflight -= 1
+
+    async def commit(self, length: int | None = None) -> str:
         await self._ensure_open()
         while self.inflight > 0:
             await self.stream.recv()
             self.inflight -= 1
-        await self.stream.send({"length": length}, b"", Status.Complete)
+        hdr = {"no_finalize": True} if length is None else {"length": length}
+        await self.stream.send(hdr, b"", Status.Complete)
         reply = await self.stream.recv()
         self.stream.close()
         return reply.header.get("tier", "")
@@ -113,12 +133,12 @@ class BlockWriterRemote:
 
 
 def make_block_writer(addr: WorkerAddress, block_id: int, reserve: int,
-                      tier: str):
+                      tier: str, reopen: bool = False):
     from curvine_amd.worker import registry
     store = registry.lookup(addr.worker_id)
     if store is not None:
-        return BlockWriterLocal(store, block_id, reserve, tier)
-    return BlockWriterRemote(addr, block_id, reserve, tier)
+        return BlockWriterLocal(store, block_id, reserve, tier, reopen)
+    return BlockWriterRemote(addr, block_id, reserve, tier, reopen)
 
 
 # ---------------------------------------------------------------------------

@@ -174,6 +174,11 @@ class SyncWriteState:
         self.block_size = status.block_size
         self.pos = status.length if existing_blocks else 0
         self._block_lens = [b.block.length for b in (existing_blocks or [])]
+        # (block_id, locations, tiers) per settled block — random writes
+        # reopen these for in-place rewrite
+        self._block_addrs = [(b.block.block_id, b.locations, b.tiers)
+                             for b in (existing_blocks or [])]
+        self._rw: dict[int, list] = {}   # block idx -> rewrite writers
         self._commits: list[dict] = []
         self._cur = None           # sync store BlockWriter
         self._cur_store = None
@@ -239,8 +244,64 @@ class SyncWriteState:
             "locations": [a.worker_id for a in lb.locations],
             "tiers": [t or hint for t, hint in zip(tiers, lb.tiers)]})
         self._block_lens.append(self._cur_pos)
+        self._block_addrs.append((lb.block.block_id, list(lb.locations),
+                                  [t or hint for t, hint
+                                   in zip(tiers, lb.tiers)]))
         self._cur = self._cur_store = self._cur_lb = self._cur_async = None
         self._cur_pos = 0
+
+    # ---------------- random (backward) writes ----------------
+    def pwrite_back(self, off: int, data) -> int:
+        """In-place rewrite of already-written bytes [off, off+len) with
+        off+len <= self.pos (fs_writer_base.rs seek-write analog).
+        Settled blocks are reopened for positional rewrite on every
+        replica; the current open block is patched directly."""
+        data = memoryview(data)
+        n = len(data)
+        if off + n > self.pos:
+            raise cverr.OutOfRange(
+                f"rewrite [{off},{off + n}) past {self.pos}")
+        cur_start = sum(self._block_lens)
+        consumed = 0
+        while consumed < n:
+            o = off + consumed
+            if o >= cur_start:                     # current open block
+                boff = o - cur_start
+                take = min(n - consumed, self._cur_pos - boff)
+                chunk = data[consumed:consumed + take]
+                if self._cur is not None:
+                    self._cur.pwrite(boff, chunk, take)
+                else:
+                    self.fs.call(_gather_pwrites(
+                        self._cur_async, boff, bytes(chunk)))
+            else:                                  # settled block
+                import bisect
+                starts = []
+                s = 0
+                for ln in self._block_lens:
+                    starts.append(s)
+                    s += ln
+                idx = bisect.bisect_right(starts, o) - 1
+                boff = o - starts[idx]
+                take = min(n - consumed, self._block_lens[idx] - boff)
+                chunk = data[consumed:consumed + take]
+                ws = self._rewrite_writers(idx)
+                if len(ws) == 1 and hasattr(ws[0], "writer"):
+                    ws[0].writer.pwrite(boff, chunk, take)   # local, no hop
+                else:
+                    self.fs.call(_gather_pwrites(ws, boff, bytes(chunk)))
+            consumed += take
+        return n
+
+    def _rewrite_writers(self, idx: int) -> list:
+        ws = self._rw.get(idx)
+        if ws is None:
+            from curvine_amd.client.block_client import make_block_writer
+            bid, addrs, tiers = self._block_addrs[idx]
+            ws = [make_block_writer(a, bid, 0, t, reopen=True)
+                  for a, t in zip(addrs, tiers)]
+            self._rw[idx] = ws
+        return ws
 
     _ZEROS = bytes(1 << 20)
 
@@ -270,6 +331,11 @@ class SyncWriteState:
             return self.status
         if self._cur is not None or self._cur_async is not None:
             self._commit_block()
+        if self._rw:
+            # close rewrite streams (no finalize: blocks stay as-is)
+            for ws in self._rw.values():
+                self.fs.call(_gather_commits(ws, None))
+            self._rw = {}
         self._done = True
         length = sum(self._block_lens)
         st = self.fs.call(self.fs.fs.client.complete_file(
@@ -286,6 +352,11 @@ class SyncWriteState:
 async def _gather_writes(writers, payload):
     import asyncio
     await asyncio.gather(*[w.write(payload) for w in writers])
+
+
+async def _gather_pwrites(writers, off, payload):
+    import asyncio
+    await asyncio.gather(*[w.pwrite(off, payload) for w in writers])
 
 
 async def _gather_commits(writers, length):
@@ -792,9 +863,16 @@ class CurvineFuseFs:
                 # forward seek: zero-fill the sparse hole
                 h.writer.write_zeros(offset - h.write_pos)
                 h.write_pos = offset
-            elif offset != h.write_pos:
-                raise OSError(errno.ENOTSUP,
-                              f"backward write at {offset} (pos {h.write_pos})")
+            elif offset < h.write_pos:
+                # random write: rewrite the overlap in place, append any
+                # tail past the current end (linkers, rsync --inplace)
+                overlap = min(size, h.write_pos - offset)
+                h.writer.pwrite_back(offset, data[:overlap])
+                if overlap < size:
+                    h.writer.write(data[overlap:])
+                    h.write_pos = offset + size
+                self._register_native_write(h)
+                return abi.WRITE_OUT.pack(size, 0)
             h.writer.write(data, ptr=ptr)
             h.write_pos += size
             # subsequent sequential WRITEs append GIL-free in C++

@@ -101,6 +101,21 @@ class BlockStore:
             return BlockWriter(layout, block_id, meta)
         raise last_exc
 
+    def reopen_writer(self, block_id: int) -> BlockWriter:
+        """Positional-rewrite access to an existing block (random writes:
+        the reference reopens a block writer at the seek position,
+        fs_writer_base.rs:466-473).  FINALIZED blocks may be rewritten IN
+        PLACE within their length (concurrent readers see unspecified
+        interleaving, as POSIX allows); growth requires the append path."""
+        with self.lock:
+            b = self.blocks.get(block_id)
+            if b is None or b.pending_delete:
+                raise err.BlockNotFound(str(block_id))
+        w = BlockWriter(b.layout, block_id, b.meta)
+        if b.state == BlockState.FINALIZED:
+            w.pos = b.meta.get("length", 0)
+        return w
+
     def finalize(self, block_id: int, length: int) -> str:
         """Publish a written block; returns its tier."""
         from curvine_amd.fault import fault_point

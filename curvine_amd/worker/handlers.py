@@ -55,11 +55,17 @@ class WorkerHandler:
         loop = asyncio.get_event_loop()
         if msg.req_status == Status.Open:
             h = msg.header
-            writer = await loop.run_in_executor(
-                None, self.store.create_writer, h["block_id"],
-                h.get("reserve", 64 << 20), h.get("tier", ""))
+            if h.get("reopen"):
+                # positional rewrite of an existing block (random writes)
+                writer = await loop.run_in_executor(
+                    None, self.store.reopen_writer, h["block_id"])
+            else:
+                writer = await loop.run_in_executor(
+                    None, self.store.create_writer, h["block_id"],
+                    h.get("reserve", 64 << 20), h.get("tier", ""))
             self.writes[msg.req_id] = {"writer": writer,
                                        "block_id": h["block_id"],
+                                       "reopen": bool(h.get("reopen")),
                                        "t0": time.perf_counter()}
             return msg.reply({"ok": True}, resp_status=Status.Running)
         sess = self.writes.get(msg.req_id)
@@ -67,9 +73,20 @@ class WorkerHandler:
             raise err.FsError("write stream not open")
         if msg.req_status == Status.Running:
             if msg.data:
-                await loop.run_in_executor(None, sess["writer"].write, msg.data)
+                off = msg.header.get("off")
+                if off is not None:
+                    await loop.run_in_executor(None, sess["writer"].pwrite,
+                                               off, msg.data)
+                else:
+                    await loop.run_in_executor(None, sess["writer"].write,
+                                               msg.data)
             return msg.reply({}, resp_status=Status.Running)
         if msg.req_status == Status.Complete:
+            if msg.header.get("no_finalize") or (
+                    sess["reopen"] and "length" not in msg.header):
+                # in-place rewrite: block metadata (length/state) unchanged
+                self.writes.pop(msg.req_id, None)
+                return msg.reply({}, resp_status=Status.Complete)
             length = msg.header.get("length", sess["writer"].pos)
             tier = await loop.run_in_executor(
                 None, self.store.finalize, sess["block_id"], length)

@@ -40,6 +40,15 @@ class BlockWriter:
         self.pos += n
         return n
 
+    def pwrite(self, off: int, data, n: int | None = None) -> int:
+        """Positional write within the reserved allocation (random-write
+        support, fs_writer_base.rs:455-478 seek analog); `pos` tracks the
+        high watermark used as the default finalize length."""
+        n = len(data) if n is None else n
+        self.layout._write_at(self.meta, off, data, n)
+        self.pos = max(self.pos, off + n)
+        return n
+
 
 class BlockReader:
     def __init__(self, layout: "BlockLayout", block_id: int, meta: dict):

@@ -311,3 +311,63 @@ def test_readdirplus_ls_l(mount):
     with os.scandir(f"{mnt}/plus") as it:
         got = {e.name: e.stat().st_size for e in it if e.is_file()}
     assert got == sizes
+
+
+def test_random_writes_same_handle(mount):
+    """Backward writes within one open handle (linker pattern): rewrite
+    bytes inside already-written data, spanning a block boundary."""
+    mnt, *_ = mount
+    p = f"{mnt}/rw.bin"
+    base = bytearray(os.urandom(10 << 20))   # 3 blocks at 4 MiB
+    with open(p, "wb") as f:
+        f.write(base)
+        # patch inside the first (committed) block
+        f.seek(100)
+        f.write(b"HEADER-PATCH")
+        base[100:112] = b"HEADER-PATCH"
+        # patch across the block 0/1 boundary
+        bnd = 4 * 1024 * 1024 - 6
+        f.seek(bnd)
+        f.write(b"BOUNDARY-SPAN")
+        base[bnd:bnd + 13] = b"BOUNDARY-SPAN"
+        # patch in the still-open tail block
+        f.seek(len(base) - 50)
+        f.write(b"TAIL")
+        base[len(base) - 50:len(base) - 46] = b"TAIL"
+    assert os.path.getsize(p) == len(base)
+    with open(p, "rb") as f:
+        got = f.read()
+    assert got == bytes(base)
+
+
+def test_rewrite_overlap_extends_eof(mount):
+    """A write overlapping the end and extending past it: rewrite the
+    overlap in place, append the tail."""
+    mnt, *_ = mount
+    p = f"{mnt}/ext.bin"
+    with open(p, "wb") as f:
+        f.write(b"A" * 1000)
+        f.seek(990)
+        f.write(b"B" * 30)        # 10 overlap + 20 append
+    data = open(p, "rb").read()
+    assert len(data) == 1020
+    assert data[:990] == b"A" * 990 and data[990:] == b"B" * 30
+
+
+def test_inplace_rewrite_existing_file(mount):
+    """r+b on a closed file: patch the middle without truncation
+    (rsync --inplace pattern); length and surrounding bytes intact."""
+    mnt, *_ = mount
+    p = f"{mnt}/inplace.bin"
+    base = bytearray(os.urandom(6 << 20))
+    with open(p, "wb") as f:
+        f.write(base)
+    with open(p, "r+b") as f:
+        f.seek(2 << 20)
+        f.write(b"MIDDLE-REWRITE")
+        base[2 << 20:(2 << 20) + 14] = b"MIDDLE-REWRITE"
+        f.seek(0)
+        f.write(b"FRONT")
+        base[:5] = b"FRONT"
+    assert os.path.getsize(p) == len(base)
+    assert open(p, "rb").read() == bytes(base)
