@@ -232,6 +232,41 @@ class SyncFs:
     def read_file(self, path: str) -> bytes:
         return self.call(self.fs.read_all(path))
 
+    def pread(self, path: str, start: int, length: int | None = None) -> bytes:
+        async def go():
+            r = await self.fs.open(path)
+            try:
+                n = (r.length - start) if length is None else length
+                return await r.pread(start, max(0, n))
+            finally:
+                r.close()
+        return self.call(go())
+
+    def open_writer(self, path: str, overwrite: bool = True):
+        """Sync streaming writer: .write(bytes) / .close() -> FileStatus."""
+        w = self.call(self.fs.create(path, overwrite=overwrite))
+        sync = self
+
+        class _W:
+            def write(self, data):
+                return sync.call(w.write(data))
+
+            def close(self):
+                return sync.call(w.complete())
+
+            def abort(self):
+                return sync.call(w.abort())
+
+            def __enter__(self):
+                return self
+
+            def __exit__(self, et, ev, tb):
+                if et is None:
+                    self.close()
+                else:
+                    self.abort()
+        return _W()
+
     def write_file(self, path: str, data, **kw) -> FileStatus:
         return self.call(self.fs.write_all(path, data, **kw))
 

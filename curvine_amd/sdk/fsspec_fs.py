@@ -76,7 +76,9 @@ class CurvineFileSystemSpec(AbstractFileSystem):
 
     # ---------------- metadata ----------------
     def _status_to_info(self, st) -> dict:
-        return {"name": st.path.lstrip("/"),
+        # rooted names, matching _strip_protocol (pyarrow's dataset
+        # discovery compares them against the normalized base dir)
+        return {"name": st.path,
                 "size": st.length,
                 "type": "directory" if st.is_dir else "file",
                 "mtime": st.mtime_ms / 1000.0}

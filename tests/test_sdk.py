@@ -187,3 +187,78 @@ def test_device_loader_cpu(cluster):
     assert seen == samples
     dl.close()
     sf.shutdown()
+
+
+def test_object_store_and_safe_commit(cluster):
+    """curvine-lancedb analog: object-store surface + conditional-put
+    commit (atomic create is the linearization point)."""
+    from curvine_amd import errors as err
+    from curvine_amd.client.filesystem import SyncFs
+    from curvine_amd.sdk.object_store import (CommitConflict,
+                                              ConditionalPutCommitter,
+                                              CurvineObjectStore)
+
+    sf = SyncFs(cluster.client_conf())
+    store = CurvineObjectStore(sf, prefix="/tables")
+    store.put("t1/data/part-0.bin", b"D" * 5000)
+    store.put("t1/data/part-1.bin", b"E" * 100)
+    assert store.head("t1/data/part-0.bin").size == 5000
+    assert store.get("t1/data/part-0.bin", 10, 20) == b"D" * 20
+    keys = [m.key for m in store.list("t1")]
+    assert keys == ["t1/data/part-0.bin", "t1/data/part-1.bin"]
+    dirs, objs = store.list_with_delimiter("t1")
+    assert dirs == ["t1/data"] and objs == []
+
+    # create mode: second put of the same key must lose
+    store.put("t1/lock", b"w1", mode="create")
+    with pytest.raises(err.FileAlreadyExists):
+        store.put("t1/lock", b"w2", mode="create")
+    assert store.get("t1/lock") == b"w1"
+
+    # copy / rename
+    store.copy("t1/data/part-1.bin", "t1/data/part-1.copy")
+    assert store.get("t1/data/part-1.copy") == b"E" * 100
+    store.rename("t1/data/part-1.copy", "t1/data/part-2.bin")
+    assert not store.exists("t1/data/part-1.copy")
+
+    # multipart writer
+    with store.put_multipart("t1/big.bin") as w:
+        for _ in range(8):
+            w.write(b"x" * (1 << 20))
+    assert store.head("t1/big.bin").size == 8 << 20
+
+    # conditional-put commit handler
+    c = ConditionalPutCommitter(store, "t1")
+    assert c.latest_version() == 0
+    c.commit(1, b"manifest-v1")
+    c.commit(2, b"manifest-v2")
+    with pytest.raises(CommitConflict):
+        c.commit(2, b"manifest-v2-loser")
+    assert c.latest_version() == 2
+    assert c.read_manifest(2) == b"manifest-v2"
+    sf.shutdown()
+
+
+def test_pyarrow_parquet_dataset(cluster):
+    """Arrow analytics on cv://: write a partitioned parquet dataset
+    through fsspec, read it back with pyarrow.dataset."""
+    pa = pytest.importorskip("pyarrow")
+    import pyarrow.dataset as pads
+    import pyarrow.parquet as pq
+    import fsspec
+
+    from curvine_amd.sdk.fsspec_fs import register
+    register()
+    master = f"127.0.0.1:{cluster.master.rpc.port}"
+    fs = fsspec.filesystem("cv", master=master, skip_instance_cache=True)
+    table = pa.table({"k": list(range(1000)),
+                      "part": [i % 4 for i in range(1000)],
+                      "v": [float(i) * 0.5 for i in range(1000)]})
+    pq.write_to_dataset(table, "/pq/ds", partition_cols=["part"],
+                        filesystem=fs)
+    ds = pads.dataset("/pq/ds", filesystem=fs, partitioning="hive")
+    got = ds.to_table()
+    assert got.num_rows == 1000
+    assert sorted(got.column("k").to_pylist()) == list(range(1000))
+    filt = ds.to_table(filter=pads.field("part") == 2)
+    assert filt.num_rows == 250
