@@ -23,6 +23,7 @@
 
 #include <arpa/inet.h>
 #include <fcntl.h>
+#include <sys/eventfd.h>
 #include <netinet/in.h>
 #include <netinet/tcp.h>
 #include <poll.h>
@@ -305,6 +306,7 @@ struct MetaServer {
   std::mutex fq_mu;
   std::condition_variable fq_cv;
   std::deque<std::pair<uint64_t, std::string>> fq;
+  int fq_efd = -1;   // eventfd: wakes the Python loop's add_reader
   std::atomic<uint64_t> served_status{0}, served_list{0}, served_exists{0},
       served_open{0}, served_ping{0}, served_notfound{0}, forwarded{0},
       conns_total{0};
@@ -363,6 +365,11 @@ static void meta_close_conn(MetaServer* S, const std::shared_ptr<MetaConn>& c) {
     S->fq.emplace_back(c->id, std::string());
   }
   S->fq_cv.notify_one();
+  if (S->fq_efd >= 0) {
+    uint64_t one = 1;
+    ssize_t r = write(S->fq_efd, &one, 8);
+    (void)r;
+  }
 }
 
 static std::string meta_proto(uint32_t hlen, uint32_t dlen, uint8_t code,
@@ -622,13 +629,20 @@ static void meta_handle_frame(MetaServer* S, const std::shared_ptr<MetaConn>& c,
     if (meta_serve(S, c.get(), frame, hlen, code, status, req_id, seq)) return;
   }
   S->forwarded.fetch_add(1, std::memory_order_relaxed);
+  bool was_empty;
   {
     std::lock_guard<std::mutex> g(S->fq_mu);
+    was_empty = S->fq.empty();
     S->fq.emplace_back(c->id,
                        std::string((const char*)frame,
                                    kMetaProto + hlen + dlen));
   }
   S->fq_cv.notify_one();
+  if (was_empty && S->fq_efd >= 0) {
+    uint64_t one = 1;
+    ssize_t r = write(S->fq_efd, &one, 8);
+    (void)r;
+  }
 }
 
 // ---------------------------------------------------------------- loop
@@ -732,6 +746,7 @@ static void meta_thread(MetaServer* S) {
 static int64_t meta_create(int listen_fd, int nthreads) {
   auto S = std::make_unique<MetaServer>();
   S->listen_fd = listen_fd;
+  S->fq_efd = eventfd(0, EFD_NONBLOCK | EFD_CLOEXEC);
   int fl = fcntl(listen_fd, F_GETFL, 0);
   fcntl(listen_fd, F_SETFL, fl | O_NONBLOCK);
   S->epfd = epoll_create1(EPOLL_CLOEXEC);
@@ -767,6 +782,7 @@ static void meta_stop_srv(int64_t sid) {
   }
   close(S->listen_fd);
   close(S->epfd);
+  if (S->fq_efd >= 0) close(S->fq_efd);
   for (auto& kv : S->conns) close(kv.second->fd);
 }
 
@@ -876,6 +892,8 @@ static py::dict meta_take_access(int64_t sid) {
   for (auto& kv : taken) d[py::int_(kv.first)] = kv.second;
   return d;
 }
+
+static int meta_eventfd(int64_t sid) { return meta_get(sid)->fq_efd; }
 
 static py::list meta_forward_pop(int64_t sid, int timeout_ms, int max_items) {
   MetaServer* S = meta_get(sid);

@@ -1068,6 +1068,7 @@ PYBIND11_MODULE(_native, m) {
   m.def("meta_drop", &meta_drop);
   m.def("meta_clear", &meta_clear);
   m.def("meta_forward_pop", &meta_forward_pop);
+  m.def("meta_eventfd", &meta_eventfd);
   m.def("meta_send", &meta_send);
   m.def("meta_stats", &meta_stats);
   m.def("meta_worker_upsert", &meta_worker_upsert);

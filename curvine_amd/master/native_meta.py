@@ -167,8 +167,8 @@ class NativeMetaFrontend:
         self.port = sock.getsockname()[1]
         self.sid = self.lib.meta_create(sock.detach(), nthreads)
         self.fwd_batch = fwd_batch
-        self._fwd_thread: Optional[threading.Thread] = None
         self._loop: Optional[asyncio.AbstractEventLoop] = None
+        self._efd: Optional[int] = None
         self._handler = None
         self._stopped = False
         import os
@@ -229,19 +229,20 @@ class NativeMetaFrontend:
         self._loop = asyncio.get_running_loop()
         self._handler = self.master.rpc_service.get_message_handler()
         self.attach()
-        self._fwd_thread = threading.Thread(
-            target=self._fwd_loop, name="meta-fwd", daemon=True)
-        self._fwd_thread.start()
+        # forwarded frames wake the loop through an eventfd the C++ side
+        # signals on empty->non-empty — the same epoll wake path a socket
+        # read gives the asyncio server, with no relay thread in between
+        self._efd = self.lib.meta_eventfd(self.sid)
+        self._loop.add_reader(self._efd, self._on_forward_ready)
         raft = self.master.raft
         self.set_serving(raft is None or raft.is_leader)
         log.info("native meta frontend on :%d", self.port)
 
     async def stop(self) -> None:
         self._stopped = True
+        if self._loop is not None and self._efd is not None:
+            self._loop.remove_reader(self._efd)
         self.lib.meta_stop(self.sid)
-        if self._fwd_thread is not None:
-            await asyncio.get_running_loop().run_in_executor(
-                None, self._fwd_thread.join, 5.0)
         for t in self._tasks.values():
             t.cancel()
         if self._tasks:
@@ -251,18 +252,19 @@ class NativeMetaFrontend:
         self._queues.clear()
 
     # ---------------- forwarded frames ----------------
-    def _fwd_loop(self) -> None:
-        pop = self.lib.meta_forward_pop
-        while not self._stopped:
-            try:
-                items = pop(self.sid, 500, self.fwd_batch)
-            except Exception:
-                if self._stopped:
-                    return
-                raise
-            if not items or self._stopped:
-                continue
-            self._loop.call_soon_threadsafe(self._dispatch, items)
+    def _on_forward_ready(self) -> None:
+        import os
+        try:
+            os.read(self._efd, 8)          # clear the signal first
+        except BlockingIOError:
+            pass
+        except OSError:
+            return                          # efd closed at stop
+        while True:
+            items = self.lib.meta_forward_pop(self.sid, 0, self.fwd_batch)
+            if not items:
+                break
+            self._dispatch(items)
 
     def _dispatch(self, items) -> None:
         # Fast path (no raft): master handlers are synchronous coroutines
