@@ -112,6 +112,10 @@ class MasterConf:
     # if the native extension is unavailable.
     native_meta: bool = True
     meta_threads: int = 4
+    # sqlite-backed durable inode store (RocksInodeStore analog): dirty
+    # inodes batched per actor tick; restart = table scan + WAL tail.
+    # Non-raft masters only (raft nodes rebuild from the raft log).
+    inode_db: bool = True
 
 
 @dataclass

@@ -96,6 +96,30 @@ def split_path(path: str) -> tuple[str, str]:
     return parent or "/", name
 
 
+class MirrorFanout:
+    """Several FsDir observers behind the single ``mirror`` slot (e.g.
+    the native metadata mirror + the sqlite inode store)."""
+
+    def __init__(self, mirrors: list):
+        self.mirrors = mirrors
+
+    def upsert(self, node) -> None:
+        for m in self.mirrors:
+            m.upsert(node)
+
+    def add_child(self, parent_id: int, name: str, child_id: int) -> None:
+        for m in self.mirrors:
+            m.add_child(parent_id, name, child_id)
+
+    def remove_child(self, parent_id: int, name: str) -> None:
+        for m in self.mirrors:
+            m.remove_child(parent_id, name)
+
+    def drop(self, inode_id: int) -> None:
+        for m in self.mirrors:
+            m.drop(inode_id)
+
+
 class FsDir:
     def __init__(self, journal: JournalWriter):
         self.journal = journal

@@ -161,14 +161,17 @@ class JournalLoader:
         return os.path.join(self.dir, "snapshot.bin")
 
     def load(self, apply_entry: Callable[[dict], None],
-             load_snapshot: Callable[[dict], int]) -> int:
-        """Returns the last op_id applied."""
-        last_op = 0
+             load_snapshot: Callable[[dict], int],
+             start_op: int = 0) -> int:
+        """Returns the last op_id applied.  ``start_op`` is pre-restored
+        state (e.g. the sqlite inode store): the snapshot is only loaded
+        when it is newer, and WAL replay begins past whichever won."""
+        last_op = start_op
         snap = self.snapshot_path()
         if os.path.exists(snap):
             with open(snap, "rb") as f:
                 entries = list(decode_stream(f))
-            if entries:
+            if entries and entries[0].get("op_id", 0) > start_op:
                 last_op = load_snapshot(entries[0])
         if not os.path.isdir(self.dir):
             return last_op

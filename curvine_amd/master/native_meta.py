@@ -184,9 +184,17 @@ class NativeMetaFrontend:
         """(Re)prime the C++ tree from the current FsDir + WorkerManager
         and hook future mutations.  Called at start and after
         snapshot-install/rebuild."""
+        from curvine_amd.master.fs_dir import MirrorFanout
         fs_dir = self.master.fs.fs_dir
         mirror = MetaMirror(self.lib, self.sid)
-        fs_dir.mirror = mirror
+        # preserve co-observers (e.g. the sqlite inode store)
+        cur = fs_dir.mirror
+        others = []
+        if isinstance(cur, MirrorFanout):
+            others = [m for m in cur.mirrors if not isinstance(m, MetaMirror)]
+        elif cur is not None and not isinstance(cur, MetaMirror):
+            others = [cur]
+        fs_dir.mirror = MirrorFanout([mirror] + others) if others else mirror
         self.lib.meta_clear(self.sid)
         up, ac = self.lib.meta_upsert, self.lib.meta_add_child
         for node in fs_dir.inodes.values():

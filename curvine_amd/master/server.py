@@ -126,6 +126,7 @@ class Master:
         self.rpc = RpcServer("master", conf.master.hostname,
                              conf.master.rpc_port, self.rpc_service)
         self.native_meta = None   # NativeMetaFrontend when enabled
+        self.inode_db = None      # SqliteInodeStore when enabled
         self._actor_task: Optional[asyncio.Task] = None
         self._stopped = asyncio.Event()
         self._mutation_count = 0
@@ -230,11 +231,15 @@ class Master:
         if self.raft is not None:
             await self.raft.stop()
         await self.rpc.stop()
+        if self.inode_db is not None:
+            self.inode_db.flush(self.fs.fs_dir, self.mounts.to_snapshot(),
+                                self.journal.op_id)
+            self.inode_db.close()
         self.journal.close()
 
     def _restore(self) -> None:
-        """Replay snapshot + journal: fs entries and mount entries share the
-        WAL, so dispatch by entry kind."""
+        """Restore state, newest source first: sqlite inode store (one
+        table scan), else snapshot; then replay the WAL tail."""
         def apply(e: dict) -> None:
             if not self.mounts.apply_entry(e):
                 self.fs.fs_dir.apply_entry(e)
@@ -245,7 +250,27 @@ class Master:
             self.mounts.load_snapshot(state.get("mounts", []))
             return op
 
-        self.fs.loader.load(apply, load_snap)
+        start = 0
+        if self.conf.master.inode_db:
+            import os as _os
+            from curvine_amd.master.inode_db import SqliteInodeStore
+            self.inode_db = SqliteInodeStore(
+                _os.path.join(self.conf.journal.journal_dir, "inodes.db"))
+            start = self.inode_db.load(self.fs.fs_dir, self.mounts) or 0
+        self.fs.loader.load(apply, load_snap, start_op=start)
+        if self.inode_db is not None:
+            # hook future mutations; WAL-tail entries replayed above were
+            # applied without the mirror, so mark everything once
+            fs_dir = self.fs.fs_dir
+            if fs_dir.journal.op_id > start:
+                self.inode_db.resync(fs_dir)
+            cur = fs_dir.mirror
+            if cur is None:
+                fs_dir.mirror = self.inode_db
+            else:
+                from curvine_amd.master.fs_dir import MirrorFanout
+                ms = cur.mirrors if isinstance(cur, MirrorFanout) else [cur]
+                fs_dir.mirror = MirrorFanout(ms + [self.inode_db])
 
     def checkpoint(self) -> None:
         state = self._snapshot_state()
@@ -283,6 +308,10 @@ class Master:
                     self.native_meta.drain_access()
                 self._ttl_sweep()
                 self._eviction_sweep()
+                if self.inode_db is not None:
+                    self.inode_db.flush(self.fs.fs_dir,
+                                        self.mounts.to_snapshot(),
+                                        self.journal.op_id)
                 if self.journal.op_id and tick % 60 == 0:
                     self.checkpoint()
             except Exception as e:  # noqa: BLE001

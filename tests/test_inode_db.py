@@ -1,0 +1,129 @@
+"""Sqlite-backed inode store: journal-free restarts, WAL-tail reconcile,
+coexistence with the native meta mirror."""
+import asyncio
+import os
+
+import pytest
+
+
+def _run(coro):
+    return asyncio.new_event_loop().run_until_complete(coro)
+
+
+def test_restart_from_db(tmp_path):
+    """Stop flushes to sqlite; a fresh master restores from the DB (no
+    snapshot file involved) with identical metadata."""
+    from curvine_amd.master.server import Master
+    from curvine_amd.testing import test_conf
+
+    conf = test_conf(str(tmp_path))
+
+    async def phase1():
+        m = await Master(conf).start()
+        m.fs.mkdir("/db/dir", 0o750, True)
+        st = m.fs.create("/db/f1", 0, 1, "MEM", False)
+        m.fs.complete_file("/db/f1", 123, [123])
+        m.fs.set_attr("/db/f1", xattrs={"user.a": b"v"})
+        m.fs.symlink("/db/ln", "/db/f1")
+        m.fs.create("/db/gone", 0, 1, "", False)
+        m.fs.complete_file("/db/gone", 1, [1])
+        m.fs.delete("/db/gone")
+        await m.stop()
+        return st
+
+    _run(phase1())
+    db = os.path.join(conf.journal.journal_dir, "inodes.db")
+    assert os.path.exists(db)
+    # remove WAL segments AND snapshot: restart must come from sqlite only
+    for n in os.listdir(conf.journal.journal_dir):
+        if n.startswith("seg_") or n == "snapshot.bin":
+            os.remove(os.path.join(conf.journal.journal_dir, n))
+
+    async def phase2():
+        from curvine_amd.master.server import Master
+        m = await Master(conf).start()
+        st = m.fs.file_status("/db/f1")
+        assert st.length == 123 and st.xattrs.get("user.a") == b"v"
+        assert m.fs.file_status("/db/dir").mode == 0o750
+        assert m.fs.file_status("/db/ln").symlink_target == "/db/f1"
+        assert not m.fs.exists("/db/gone")
+        names = [s.name for s in m.fs.list_status("/db")]
+        assert names == ["dir", "f1", "ln"]
+        # ids keep advancing (no reuse after restart)
+        st2 = m.fs.create("/db/f2", 0, 1, "", False)
+        assert st2.inode_id > st.inode_id
+        await m.stop()
+
+    _run(phase2())
+
+
+def test_wal_tail_reconcile(tmp_path):
+    """Mutations after the last flush live only in the WAL; restart must
+    replay them over the DB state and reconcile deletions."""
+    from curvine_amd.master.server import Master
+    from curvine_amd.testing import test_conf
+
+    conf = test_conf(str(tmp_path))
+
+    async def phase1():
+        m = await Master(conf).start()
+        m.fs.create("/t/keep", 0, 1, "", False)
+        m.fs.complete_file("/t/keep", 5, [5])
+        m.fs.create("/t/stale", 0, 1, "", False)
+        m.fs.complete_file("/t/stale", 5, [5])
+        # force a flush (as the actor tick would)
+        m.inode_db.flush(m.fs.fs_dir, m.mounts.to_snapshot(),
+                         m.journal.op_id)
+        # post-flush mutations: only in the WAL
+        m.fs.delete("/t/stale")
+        m.fs.create("/t/tail", 0, 1, "", False)
+        m.fs.complete_file("/t/tail", 9, [9])
+        # crash: no final flush of these (simulate by closing db directly)
+        m.inode_db.close()
+        m.inode_db = None
+        await m.stop()
+
+    _run(phase1())
+
+    async def phase2():
+        m = await Master(conf).start()
+        assert m.fs.exists("/t/keep")
+        assert m.fs.exists("/t/tail")
+        assert not m.fs.exists("/t/stale")
+        # second clean restart now comes purely from the reconciled DB
+        await m.stop()
+        for n in os.listdir(conf.journal.journal_dir):
+            if n.startswith("seg_") or n == "snapshot.bin":
+                os.remove(os.path.join(conf.journal.journal_dir, n))
+        m2 = await Master(conf).start()
+        assert m2.fs.exists("/t/tail") and not m2.fs.exists("/t/stale")
+        await m2.stop()
+
+    _run(phase2())
+
+
+def test_db_and_native_mirror_coexist(tmp_path):
+    """Both observers hang off FsDir.mirror (fanout): native reads stay
+    correct while the DB tracks dirt."""
+    from curvine_amd.client.filesystem import SyncFs
+    from curvine_amd.testing import SyncMiniCluster, test_conf
+
+    conf = test_conf(str(tmp_path / "cv"))
+    smc = SyncMiniCluster(conf=conf, tmp_dir=str(tmp_path / "cv")).start()
+    try:
+        master = smc.master
+        assert master.inode_db is not None
+        assert master.native_meta is not None
+        sf = SyncFs(smc.client_conf())
+        sf.write_file("/co/x", b"abc")
+        assert sf.file_status("/co/x").length == 3    # native-served
+        assert master.native_meta.stats()["served_status"] >= 1
+        master.inode_db.flush(master.fs.fs_dir,
+                              master.mounts.to_snapshot(),
+                              master.journal.op_id)
+        rows = master.inode_db.conn.execute(
+            "SELECT COUNT(*) FROM inodes").fetchone()[0]
+        assert rows == len(master.fs.fs_dir.inodes)
+        sf.shutdown()
+    finally:
+        smc.stop()
