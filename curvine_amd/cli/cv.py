@@ -300,6 +300,51 @@ def cmd_bench(fs, args):
         "write_MBps": round(wbps / 1e6, 1), "read_MBps": round(rbps / 1e6, 1)}))
 
 
+def cmd_validate(_fs, args):
+    """Offline config validation (cli/validate_run.rs analog): parse the
+    TOML, check tier specs / ports / peer addresses, print a summary."""
+    from curvine_amd.conf import ClusterConf, TIERS
+    path = args.conf_path or args.conf
+    conf = ClusterConf.from_file(path) if path else ClusterConf()
+    problems = []
+    try:
+        dirs = conf.worker.parsed_dirs()
+    except Exception as e:  # noqa: BLE001
+        problems.append(f"data_dirs: {e}")
+        dirs = []
+    for dd in dirs:
+        if dd.tier not in TIERS:
+            problems.append(f"unknown tier {dd.tier!r}")
+        if dd.tier == "HBM" and dd.device_id < 0:
+            problems.append(f"HBM dir {dd.path!r} needs a device ordinal")
+    for port_name in ("rpc_port", "web_port"):
+        v = getattr(conf.master, port_name)
+        if not (0 <= v < 65536):
+            problems.append(f"master.{port_name}={v} out of range")
+    for spec in conf.journal.peers:
+        pid, sep, addr = spec.partition("@")
+        if not sep or not pid.isdigit() or ":" not in addr:
+            problems.append(f"journal peer {spec!r}: want 'id@host:port'")
+    if conf.master.block_size <= 0:
+        problems.append("master.block_size must be positive")
+    if len(conf.journal.peers) not in (0, 1) and \
+            len(conf.journal.peers) % 2 == 0:
+        problems.append("raft peer count should be odd")
+    summary = {
+        "conf": path or "(defaults)",
+        "master": f"{conf.master.hostname}:{conf.master.rpc_port}",
+        "raft_peers": len(conf.journal.peers),
+        "tiers": [f"{d.tier}:{d.capacity >> 30}GB" for d in dirs],
+        "block_size_mb": conf.master.block_size >> 20,
+        "fuse_mnt": conf.fuse.mnt_path,
+        "native_meta": conf.master.native_meta,
+        "problems": problems,
+    }
+    print(json.dumps(summary, indent=2))
+    if problems:
+        raise SystemExit(1)
+
+
 def build_parser() -> argparse.ArgumentParser:
     p = argparse.ArgumentParser(prog="cv", description=__doc__)
     p.add_argument("--conf", default=os.environ.get("CURVINE_CONF"))
@@ -346,6 +391,9 @@ def build_parser() -> argparse.ArgumentParser:
     add("mount-table", cmd_mount_table)
     add("bench", cmd_bench, A("--num", type=int, default=200),
         A("--size", type=int, default=16 << 20))
+    vp = sub.add_parser("validate", help="validate a cluster TOML offline")
+    vp.add_argument("conf_path", nargs="?", default=None)
+    vp.set_defaults(fn=cmd_validate, offline=True)
     node = sub.add_parser("node")
     nodesub = node.add_subparsers(dest="nodecmd", required=True)
     nl = nodesub.add_parser("list")
@@ -358,6 +406,15 @@ def build_parser() -> argparse.ArgumentParser:
 
 def main(argv=None) -> int:
     args = build_parser().parse_args(argv)
+    if getattr(args, "offline", False):
+        try:
+            args.fn(None, args)
+            return 0
+        except SystemExit as e:
+            return int(e.code or 0)
+        except Exception as e:  # noqa: BLE001
+            print(f"cv: {e}", file=sys.stderr)
+            return 1
     fs = make_fs(args)
     try:
         args.fn(fs, args)

@@ -68,3 +68,14 @@ def test_cli_bench(cluster, capsys):
     assert cv(cluster, "bench", "--num", "20", "--size", str(1 << 20)) == 0
     out = json.loads(capsys.readouterr().out)
     assert out["create_qps"] > 0 and out["read_MBps"] > 0
+
+
+def test_validate_subcommand(tmp_path):
+    from curvine_amd.cli.cv import main
+    # valid config
+    assert main(["validate", "etc/curvine-cluster.toml"]) == 0
+    # broken config: bad tier + even raft peer count
+    bad = tmp_path / "bad.toml"
+    bad.write_text('[worker]\ndata_dirs = ["[HBM]gpu0"]\n'
+                   '[journal]\npeers = ["1@h:1", "2@h:2"]\n')
+    assert main(["validate", str(bad)]) == 1
