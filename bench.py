@@ -324,15 +324,29 @@ def step_client_seq(args, rank, rt) -> int:
     total = [0] * nthreads
     errs = []
 
+    depth = max(1, args.seq_batch)
+
     def worker(t):
         try:
-            pbuf = native.PinnedBuffer(args.read_chunk) if use_pinned else None
+            pbuf = native.PinnedBuffer(args.read_chunk * depth) \
+                if use_pinned else None
             buf = None if pbuf else bytearray(args.read_chunk)
             # thread t reads files t, t+T, ... fully (fio numjobs analog)
             for i in range(t, len(readers), nthreads):
                 r = readers[i]
                 pos = 0
                 while pos < r.length:
+                    if pbuf is not None and \
+                            pos + args.read_chunk * depth <= r.length:
+                        # depth chunks issued async on one stream, ONE
+                        # sync — overlaps successive D2H DMAs
+                        offs = [pos + k * args.read_chunk
+                                for k in range(depth)]
+                        r.pread_batch_ptr(offs, args.read_chunk, pbuf.ptr,
+                                          args.read_chunk)
+                        pos += args.read_chunk * depth
+                        total[t] += args.read_chunk * depth
+                        continue
                     want = min(args.read_chunk, r.length - pos)
                     if pbuf is not None:
                         n = r.pread_into_ptr(pos, pbuf.ptr, want)
@@ -379,6 +393,8 @@ def main():
     p.add_argument("--rand-reads", type=int, default=200_000)
     p.add_argument("--iodepth", type=int, default=64,
                    help="queue depth per thread for randread4k (fio iodepth)")
+    p.add_argument("--seq-batch", type=int, default=4,
+                   help="chunks per sync in seqread (DMA pipelining depth)")
     p.add_argument("--hbm-gb", type=int, default=16)
     p.add_argument("--staging-bytes", type=int, default=8 << 20)
     p.add_argument("--staging-count", type=int, default=8)
