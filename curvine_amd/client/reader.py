@@ -116,10 +116,12 @@ class SyncLocalReader:
         import bisect
         import ctypes
         from curvine_amd import native
+        if not self.fb.blocks:
+            return 0
         groups: dict[int, tuple[list, list]] = {}  # arena handle -> offs, dsts
         slow: list[tuple[int, int]] = []
         for i, off in enumerate(file_offs):
-            idx = bisect.bisect_right(self._offs, off) - 1
+            idx = max(0, bisect.bisect_right(self._offs, off) - 1)
             lb = self.fb.blocks[idx]
             boff = off - lb.offset
             r = self._readers[idx]

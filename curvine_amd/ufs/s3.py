@@ -89,17 +89,43 @@ class _S3Reader(UfsReader):
 
 
 class _S3Writer(UfsWriter):
-    """Buffered whole-object PUT (multipart upload left for large files)."""
+    """Objects below ``part_size`` go up as one PUT; larger ones switch
+    to multipart (CreateMultipartUpload / UploadPart / Complete), so a
+    cache write-back never buffers more than one part in memory."""
 
-    def __init__(self, fs: "S3Ufs", key: str):
-        self.fs, self.key, self.parts = fs, key, []
+    def __init__(self, fs: "S3Ufs", key: str, part_size: int = 8 << 20):
+        self.fs, self.key = fs, key
+        self.part_size = part_size
+        self._buf = bytearray()
+        self.upload_id: Optional[str] = None
+        self._etags: list[str] = []
 
     def write(self, data: bytes) -> int:
-        self.parts.append(bytes(data))
+        self._buf.extend(data)
+        while len(self._buf) >= self.part_size:
+            self._flush_part(self.part_size)
         return len(data)
 
+    def _flush_part(self, n: int) -> None:
+        if self.upload_id is None:
+            self.upload_id = self.fs._create_multipart(self.key)
+        chunk = bytes(self._buf[:n])
+        del self._buf[:n]
+        etag = self.fs._upload_part(self.key, self.upload_id,
+                                    len(self._etags) + 1, chunk)
+        self._etags.append(etag)
+
     def close(self) -> None:
-        self.fs._put(self.key, b"".join(self.parts))
+        if self.upload_id is None:
+            self.fs._put(self.key, bytes(self._buf))
+            return
+        if self._buf:
+            self._flush_part(len(self._buf))
+        self.fs._complete_multipart(self.key, self.upload_id, self._etags)
+
+    def abort(self) -> None:
+        if self.upload_id is not None:
+            self.fs._abort_multipart(self.key, self.upload_id)
 
 
 class S3Ufs(UnderFs):
@@ -144,6 +170,31 @@ class S3Ufs(UnderFs):
     def _put(self, key: str, data: bytes) -> None:
         self._req("PUT", self._url(key), data=data)
 
+    # ---------------- multipart upload ----------------
+    def _create_multipart(self, key: str) -> str:
+        r = self._req("POST", self._url(key, "uploads="))
+        root = ET.fromstring(r.content)
+        ns = root.tag.split("}")[0] + "}" if "}" in root.tag else ""
+        return root.find(f"{ns}UploadId").text
+
+    def _upload_part(self, key: str, upload_id: str, part_no: int,
+                     data: bytes) -> str:
+        q = f"partNumber={part_no}&uploadId={urllib.parse.quote(upload_id)}"
+        r = self._req("PUT", self._url(key, q), data=data)
+        return r.headers.get("etag", "").strip('"')
+
+    def _complete_multipart(self, key: str, upload_id: str,
+                            etags: list[str]) -> None:
+        body = "<CompleteMultipartUpload>" + "".join(
+            f"<Part><PartNumber>{i + 1}</PartNumber><ETag>\"{e}\"</ETag></Part>"
+            for i, e in enumerate(etags)) + "</CompleteMultipartUpload>"
+        q = f"uploadId={urllib.parse.quote(upload_id)}"
+        self._req("POST", self._url(key, q), data=body.encode())
+
+    def _abort_multipart(self, key: str, upload_id: str) -> None:
+        q = f"uploadId={urllib.parse.quote(upload_id)}"
+        self._req("DELETE", self._url(key, q))
+
     def list_files(self, path: str = "/", recursive: bool = True) -> list[dict]:
         prefix = self._key(path).lstrip("/")
         if prefix:
@@ -155,8 +206,7 @@ class S3Ufs(UnderFs):
                 q += "&delimiter=%2F"
             if token:
                 q += f"&continuation-token={urllib.parse.quote(token)}"
-            r = self._req("GET", self._url("", q) if False else
-                          f"{self.endpoint}/{self.bucket}?{q}")
+            r = self._req("GET", f"{self.endpoint}/{self.bucket}?{q}")
             root = ET.fromstring(r.content)
             ns = root.tag.split("}")[0] + "}" if "}" in root.tag else ""
             for c in root.findall(f"{ns}Contents"):
