@@ -141,6 +141,7 @@ class S3Ufs(UnderFs):
         self.region = properties.get("region", "us-east-1")
         self.access_key = properties.get("access_key", "")
         self.secret_key = properties.get("secret_key", "")
+        self.part_size = int(properties.get("multipart_part_size", 8 << 20))
         self.session = requests.Session()
 
     def _url(self, key: str, query: str = "") -> str:
@@ -196,7 +197,7 @@ class S3Ufs(UnderFs):
         self._req("DELETE", self._url(key, q))
 
     def list_files(self, path: str = "/", recursive: bool = True) -> list[dict]:
-        prefix = self._key(path).lstrip("/")
+        prefix = self._key(path).strip("/")
         if prefix:
             prefix += "/"
         out, token = [], None
@@ -235,7 +236,7 @@ class S3Ufs(UnderFs):
         return _S3Reader(self, self._key(path), offset, st["length"])
 
     def create(self, path: str) -> UfsWriter:
-        return _S3Writer(self, self._key(path))
+        return _S3Writer(self, self._key(path), self.part_size)
 
     def delete(self, path: str, recursive: bool = False) -> None:
         self._req("DELETE", self._url(self._key(path)))

@@ -211,6 +211,18 @@ class UnifiedFileSystem(CurvineFileSystem):
         self._cache_inflight = 0
         self._cache_pending: set[str] = set()
 
+    async def mount(self, curvine_path, ufs_path, properties=None,
+                    cache_mode="cache", auto_cache=True):
+        out = await super().mount(curvine_path, ufs_path, properties,
+                                  cache_mode, auto_cache)
+        self.mounts.invalidate()   # own mutations must be visible at once
+        return out
+
+    async def unmount(self, curvine_path):
+        out = await super().unmount(curvine_path)
+        self.mounts.invalidate()
+        return out
+
     async def _route(self, path: str):
         """(mount_info, ufs, rel) or (None, None, None)."""
         mi = await self.mounts.lookup(path)
