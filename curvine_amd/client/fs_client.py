@@ -28,8 +28,25 @@ class FsClient:
         await self.connector.close()
 
     async def _rpc(self, code: RpcCode, header: dict | None = None) -> dict:
-        reply = await self.connector.rpc(code, header)
-        return reply.header
+        if not self.conf.client.audit_log:
+            reply = await self.connector.rpc(code, header)
+            return reply.header
+        # client audit stream (unified_filesystem.rs:144-169 analog)
+        import logging
+        import time
+        t0 = time.perf_counter()
+        ok = True
+        try:
+            reply = await self.connector.rpc(code, header)
+            return reply.header
+        except Exception:
+            ok = False
+            raise
+        finally:
+            logging.getLogger("audit.client").info(
+                "cmd=%s path=%s ok=%s used_us=%d", code.name,
+                (header or {}).get("path", ""), ok,
+                int((time.perf_counter() - t0) * 1e6))
 
     # ---------------- namespace ----------------
     async def mkdir(self, path: str, mode: int = 0o755,

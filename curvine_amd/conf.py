@@ -171,6 +171,9 @@ class ClientConf:
     # auto-cache UFS files on miss (unified fs)
     auto_cache: bool = True
     auto_cache_max_inflight: int = 4
+    # client-side audit stream (unified_filesystem.rs:144-169 analog):
+    # cmd/path/ok/used_us per metadata RPC on logger "audit.client"
+    audit_log: bool = False
 
 
 @dataclass

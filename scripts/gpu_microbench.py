@@ -85,6 +85,35 @@ def main():
     dt = timeit(lambda: native.load().arena_copy(a.handle, 2 * GB, a.handle, 0, GB))
     results["d2d_copy_GBps"] = round(2 * GB / dt / 1e9, 2)  # rd+wr bytes
 
+    # on-device sample gather (CurvineDeviceLoader hot path): 4096 x
+    # 256 KiB scattered extents packed to a device destination
+    mod = native.load()
+    n_s, s_sz = 4096, 256 << 10
+    src_offs = rng.choice(range(0, (2 * GB) // s_sz), n_s,
+                          replace=False).tolist()
+    triples = [(int(o) * s_sz, i * s_sz, s_sz)
+               for i, o in enumerate(src_offs)]
+    dst_ptr = a.base_ptr() + 3 * GB
+    dt = timeit(lambda: mod.arena_gather_ptr(a.handle, triples, dst_ptr))
+    results["gather_dev_4096x256KiB_GBps"] = round(
+        2 * n_s * s_sz / dt / 1e9, 2)   # rd+wr bytes
+
+    # LZ4 device decompress: 512 MiB of ~3:1-compressible data
+    rep = np.frombuffer((b"curvine-amd lz4 block payload %06d " % 42) * 64,
+                        dtype=np.uint8)
+    block = np.tile(rep, (512 << 20) // len(rep) + 1)[:512 << 20].copy()
+    noise_idx = rng.choice(len(block), len(block) // 64, replace=False)
+    block[noise_idx] = rng.integers(0, 256, len(noise_idx), dtype=np.uint8)
+    comp = native.lz4_compress(block.tobytes())
+    results["lz4_ratio"] = round(len(block) / len(comp), 2)
+    dt = timeit(lambda: mod.arena_lz4_decompress(a.handle, 0, comp))
+    results["lz4_device_decompress_GBps"] = round(len(block) / dt / 1e9, 2)
+    back = np.zeros(1 << 20, dtype=np.uint8)
+    a.read(0, back, 0, 1 << 20)
+    assert back.tobytes() == block[:1 << 20].tobytes()
+    dt = timeit(lambda: native.lz4_decompress(comp))
+    results["lz4_host_decompress_GBps"] = round(len(block) / dt / 1e9, 2)
+
     print(json.dumps(results, indent=1))
     a.close()
     pin.close()

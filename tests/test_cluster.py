@@ -200,3 +200,29 @@ def test_worker_restart_rescan(loop, tmp_path):
             assert back == data
             await fs.close()
     run(loop, main())
+
+
+def test_client_audit_log(tmp_path, caplog):
+    """client.audit_log: every metadata RPC emits cmd/path/ok/used_us on
+    the audit.client logger."""
+    import asyncio
+    import logging
+
+    from curvine_amd.testing import MiniCluster
+
+    async def main():
+        mc = await MiniCluster(tmp_dir=str(tmp_path)).start()
+        conf = mc.client_conf()
+        conf.client.audit_log = True
+        from curvine_amd.client.filesystem import CurvineFileSystem
+        fs = CurvineFileSystem(conf)
+        with caplog.at_level(logging.INFO, logger="audit.client"):
+            await fs.mkdir("/aud", create_parents=True)
+            await fs.file_status("/aud")
+        await fs.close()
+        await mc.stop()
+
+    asyncio.new_event_loop().run_until_complete(main())
+    cmds = [r.message for r in caplog.records if r.name == "audit.client"]
+    assert any("cmd=Mkdir" in m and "ok=True" in m for m in cmds)
+    assert any("cmd=FileStatus" in m and "path=/aud" in m for m in cmds)
