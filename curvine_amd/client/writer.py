@@ -28,6 +28,8 @@ class FsWriter:
         self._writers: list = []
         self._block_pos = 0
         self._block_lens: list[int] = []
+        self._block_addrs: list[tuple] = []   # (block_id, locations, tiers)
+        self._rw: dict[int, list] = {}        # rewrite writers per block idx
         self._commits: list[dict] = []   # block locations to report on complete
         self.pos = 0
         self._closed = False
@@ -82,6 +84,9 @@ class FsWriter:
             "locations": [a.worker_id for a in self._block.locations],
             "tiers": [t or hint for t, hint in zip(tiers, self._block.tiers)]})
         self._block_lens.append(self._block_pos)
+        self._block_addrs.append(
+            (self._block.block.block_id, list(self._block.locations),
+             [t or hint for t, hint in zip(tiers, self._block.tiers)]))
         self._block = None
         self._writers = []
         self._block_pos = 0
@@ -90,12 +95,60 @@ class FsWriter:
         if self._buf:
             await self._flush_chunk()
 
+    # ---------------- positional rewrite (writer seek analog) ----------------
+    async def pwrite_at(self, off: int, data) -> int:
+        """Rewrite already-written bytes in place on every replica
+        (fs_writer_base.rs:455-478 seek-write analog; settled blocks are
+        reopened positionally).  [off, off+len) must lie within bytes
+        this handle has written; appending stays with write()."""
+        await self.flush()
+        data = memoryview(bytes(data))
+        n = len(data)
+        if off + n > self.pos:
+            raise err.OutOfRange(f"rewrite [{off},{off + n}) past {self.pos}")
+        cur_start = sum(self._block_lens)
+        consumed = 0
+        while consumed < n:
+            o = off + consumed
+            if o >= cur_start:                   # current open block
+                boff = o - cur_start
+                take = min(n - consumed, self._block_pos - boff)
+                chunk = bytes(data[consumed:consumed + take])
+                await asyncio.gather(
+                    *[w.pwrite(boff, chunk) for w in self._writers])
+            else:                                # settled block
+                import bisect
+                starts, s = [], 0
+                for ln in self._block_lens:
+                    starts.append(s)
+                    s += ln
+                idx = bisect.bisect_right(starts, o) - 1
+                boff = o - starts[idx]
+                take = min(n - consumed, self._block_lens[idx] - boff)
+                chunk = bytes(data[consumed:consumed + take])
+                ws = self._rewrite_writers(idx)
+                await asyncio.gather(*[w.pwrite(boff, chunk) for w in ws])
+            consumed += take
+        return n
+
+    def _rewrite_writers(self, idx: int) -> list:
+        ws = self._rw.get(idx)
+        if ws is None:
+            bid, addrs, tiers = self._block_addrs[idx]
+            ws = [make_block_writer(a, bid, 0, t, reopen=True)
+                  for a, t in zip(addrs, tiers)]
+            self._rw[idx] = ws
+        return ws
+
     async def complete(self) -> FileStatus:
         if self._closed:
             return self.status
         await self.flush()
         if self._block is not None:
             await self._commit_block()
+        for ws in self._rw.values():   # close rewrite streams, no finalize
+            await asyncio.gather(*[w.commit(None) for w in ws])
+        self._rw = {}
         self._closed = True
         length = sum(self._block_lens)
         return await self.client.complete_file(self.path, length,

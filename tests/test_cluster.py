@@ -226,3 +226,40 @@ def test_client_audit_log(tmp_path, caplog):
     cmds = [r.message for r in caplog.records if r.name == "audit.client"]
     assert any("cmd=Mkdir" in m and "ok=True" in m for m in cmds)
     assert any("cmd=FileStatus" in m and "path=/aud" in m for m in cmds)
+
+
+def test_fswriter_pwrite_at(tmp_path):
+    """Async client writer positional rewrite: patch committed and open
+    blocks before complete(), spanning a block boundary."""
+    import asyncio
+    import os as _os
+
+    from curvine_amd.testing import MiniCluster, test_conf
+
+    async def main():
+        conf = test_conf(str(tmp_path))
+        conf.master.block_size = 1 << 20
+        conf.client.block_size = 1 << 20
+        mc = await MiniCluster(conf=conf, tmp_dir=str(tmp_path)).start()
+        fs = mc.fs()
+        base = bytearray(_os.urandom(3 * (1 << 20) + 500))
+        w = await fs.create("/pw/f", overwrite=True)
+        await w.write(base)
+        # patch inside committed block 0
+        await w.pwrite_at(100, b"PATCH-A")
+        base[100:107] = b"PATCH-A"
+        # patch across the block 1/2 boundary
+        bnd = 2 * (1 << 20) - 3
+        await w.pwrite_at(bnd, b"PATCH-B")
+        base[bnd:bnd + 7] = b"PATCH-B"
+        # patch the open tail block
+        await w.pwrite_at(len(base) - 20, b"Z")
+        base[len(base) - 20:len(base) - 19] = b"Z"
+        st = await w.complete()
+        assert st.length == len(base)
+        got = await fs.read_all("/pw/f")
+        assert got == bytes(base)
+        await fs.close()
+        await mc.stop()
+
+    asyncio.new_event_loop().run_until_complete(main())
