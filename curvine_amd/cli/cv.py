@@ -15,7 +15,7 @@ import os
 import sys
 import time
 
-from curvine_amd.conf import ClusterConf, fmt_bytes, parse_bytes
+from curvine_amd.conf import ClusterConf, fmt_bytes
 from curvine_amd.client.filesystem import SyncFs
 from curvine_amd.unified import UnifiedFileSystem
 

@@ -9,14 +9,13 @@ from __future__ import annotations
 
 import asyncio
 import threading
-from typing import Optional
 
 from curvine_amd import errors as err
 from curvine_amd.client.fs_client import FsClient
 from curvine_amd.client.reader import FsReader
 from curvine_amd.client.writer import FsWriter
 from curvine_amd.conf import ClusterConf
-from curvine_amd.model import FileBlocks, FileStatus
+from curvine_amd.model import FileStatus
 
 
 class CurvineFileSystem:

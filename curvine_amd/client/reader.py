@@ -9,7 +9,6 @@ fs_reader_parallel.rs (sliced parallel reads for large requests);
 from __future__ import annotations
 
 import asyncio
-from typing import Optional
 
 from curvine_amd import errors as err
 from curvine_amd.client.block_client import (BlockReaderHole, BlockReaderLocal,

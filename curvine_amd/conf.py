@@ -14,7 +14,7 @@ import dataclasses
 import os
 import re
 from dataclasses import dataclass, field
-from typing import Any, Optional
+from typing import Any
 
 try:
     import tomli as _toml

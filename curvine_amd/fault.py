@@ -12,7 +12,7 @@ from __future__ import annotations
 import fnmatch
 import threading
 import time
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Optional
 
 _rules: list["FaultRule"] = []

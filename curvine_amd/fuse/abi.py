@@ -9,7 +9,6 @@ session/fuse_op_code.rs.  Layouts follow include/uapi/linux/fuse.h
 from __future__ import annotations
 
 import struct
-from dataclasses import dataclass
 
 FUSE_KERNEL_VERSION = 7
 FUSE_KERNEL_MINOR_VERSION = 36

@@ -22,7 +22,7 @@ from typing import Optional
 
 from curvine_amd import errors as cverr
 from curvine_amd.fuse import abi
-from curvine_amd.model import FileStatus, FileType
+from curvine_amd.model import FileStatus
 
 log = logging.getLogger("curvine.fuse.ops")
 

@@ -17,7 +17,6 @@ import logging
 import os
 import struct
 import threading
-from typing import Optional
 
 from curvine_amd.fuse import abi
 

@@ -13,7 +13,6 @@ RwLock'd SyncFsDir fast path).
 from __future__ import annotations
 
 import logging
-from typing import Optional
 
 from curvine_amd import errors as err
 from curvine_amd.conf import ClusterConf

@@ -6,7 +6,6 @@ URI; persisted through the journal (Mount/UnMount entries).
 """
 from __future__ import annotations
 
-import itertools
 
 from curvine_amd import errors as err
 from curvine_amd.master.fs_dir import norm_path

@@ -23,7 +23,6 @@ import asyncio
 import logging
 import socket
 import struct
-import threading
 from typing import Optional
 
 import msgpack

@@ -13,12 +13,11 @@ from __future__ import annotations
 
 import logging
 import random
-import time
 from typing import Optional
 
 from curvine_amd import errors as err
 from curvine_amd.conf import TIER_ORDER
-from curvine_amd.model import (CMD_DELETE_BLOCK, StorageInfo, WorkerAddress,
+from curvine_amd.model import (CMD_DELETE_BLOCK,
                                WorkerInfo, WorkerState, now_ms)
 
 log = logging.getLogger("curvine.workers")

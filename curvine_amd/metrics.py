@@ -10,10 +10,8 @@ Uses prometheus_client when present; hot paths keep plain dict counters
 """
 from __future__ import annotations
 
-import json
 import threading
 import time
-from typing import Optional
 
 try:
     import prometheus_client as prom

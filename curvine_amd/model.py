@@ -10,7 +10,6 @@ from __future__ import annotations
 import time
 from dataclasses import dataclass, field, asdict
 from enum import IntEnum
-from typing import Optional
 
 from curvine_amd.conf import TIER_MEM
 

@@ -13,7 +13,7 @@ import logging
 import socket
 from typing import Optional
 
-from curvine_amd.errors import ConnectError, FsError, NotLeader, RpcTimeout, RetryError
+from curvine_amd.errors import ConnectError, FsError, NotLeader, RpcTimeout
 from curvine_amd.rpc.codes import RpcCode
 from curvine_amd.rpc.message import Message, Status, PROTO_SIZE, MAX_DATA_SIZE
 

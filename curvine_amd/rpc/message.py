@@ -17,7 +17,6 @@ import itertools
 import struct
 from dataclasses import dataclass, field
 from enum import IntEnum
-from typing import Any, Optional
 
 import msgpack
 

@@ -14,7 +14,7 @@ import logging
 import socket
 from typing import Optional
 
-from curvine_amd.rpc.message import Message, Status, PROTO_SIZE, MAX_DATA_SIZE
+from curvine_amd.rpc.message import Message, PROTO_SIZE, MAX_DATA_SIZE
 
 log = logging.getLogger("curvine.rpc")
 

@@ -16,7 +16,7 @@ from __future__ import annotations
 
 import io
 import tarfile
-from typing import Iterator, Optional
+from typing import Iterator
 
 from curvine_amd.conf import ClusterConf
 

@@ -13,7 +13,6 @@ unchanged.
 from __future__ import annotations
 
 import io
-import os
 from typing import Optional
 
 try:

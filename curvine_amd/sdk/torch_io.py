@@ -8,7 +8,6 @@ trip); host-tier blocks are uploaded through the pinned path.
 """
 from __future__ import annotations
 
-from typing import Optional
 
 from curvine_amd import errors as err
 

@@ -12,8 +12,8 @@ import os
 import tempfile
 from typing import Optional
 
-from curvine_amd.conf import ClusterConf, DataDir
-from curvine_amd.client.filesystem import CurvineFileSystem, SyncFs
+from curvine_amd.conf import ClusterConf
+from curvine_amd.client.filesystem import CurvineFileSystem
 from curvine_amd.master.server import Master
 from curvine_amd.worker.server import Worker
 

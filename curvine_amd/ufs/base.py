@@ -1,7 +1,7 @@
 """UFS trait (curvine-ufs-api `UnderFs`/`AsyncChunkReader` analog)."""
 from __future__ import annotations
 
-from typing import Iterator, Optional
+from typing import Optional
 
 
 class UfsReader:

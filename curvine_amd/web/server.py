@@ -13,7 +13,7 @@ import html
 import json
 import logging
 import urllib.parse
-from typing import Callable, Optional
+from typing import Optional
 
 log = logging.getLogger("curvine.web")
 

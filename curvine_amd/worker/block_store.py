@@ -15,10 +15,9 @@ from __future__ import annotations
 
 import logging
 import threading
-from typing import Optional
 
 from curvine_amd import errors as err
-from curvine_amd.conf import TIERS, TIER_ORDER, WorkerConf
+from curvine_amd.conf import TIER_ORDER, WorkerConf
 from curvine_amd.model import BlockState, StorageInfo
 from curvine_amd.worker.layout import (BlockLayout, BlockReader, BlockWriter,
                                        make_layout)

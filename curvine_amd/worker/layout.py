@@ -13,7 +13,6 @@ raw-extent): here
 from __future__ import annotations
 
 import os
-from typing import Optional
 
 from curvine_amd import errors as err
 from curvine_amd.conf import DataDir, TIER_HBM, TIER_MEM

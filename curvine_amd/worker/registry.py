@@ -11,7 +11,6 @@ dmabuf IPC, so colocated-in-process is the designed-for fast path.
 from __future__ import annotations
 
 import os
-from typing import Optional
 
 _stores: dict[int, "object"] = {}
 _pids: dict[int, int] = {}
