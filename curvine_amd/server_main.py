@@ -33,8 +33,13 @@ async def run(args) -> None:
             conf.worker.data_dirs = args.data_dir
         servers.append(await Worker(conf, device_id=args.device).start())
     if args.web:
+        from curvine_amd.master.server import Master
         from curvine_amd.web.server import WebServer
-        servers.append(await WebServer(conf, servers[0]).start())
+        for s in list(servers):
+            if isinstance(s, Master):
+                servers.append(await WebServer(conf, master=s).start())
+            else:   # worker dashboard + /metrics on its own port
+                servers.append(await WebServer(conf, worker=s).start())
 
     stop = asyncio.Event()
     loop = asyncio.get_event_loop()
