@@ -661,6 +661,16 @@ class CurvineFuseFs:
         self.drop_child(nodeid, name)
         return b""
 
+    def _rewrite_handle_paths(self, old: str, new: str) -> None:
+        """Retarget open handles after a rename (covers files and whole
+        directory subtrees)."""
+        with self.handles_lock:
+            for h in self.handles.values():
+                if h.path == old or h.path.startswith(old + "/"):
+                    h.path = new + h.path[len(old):]
+                    if h.writer is not None:
+                        h.writer.path = h.path
+
     def _rename(self, nodeid, newdir, oldname, newname, flags=0):
         src = self.node_path(nodeid).rstrip("/") + "/" + oldname
         dst = self.node_path(newdir).rstrip("/") + "/" + newname
@@ -692,6 +702,9 @@ class CurvineFuseFs:
                     self.invalidate(nid)
         else:
             self.call(self.fs.rename(src, dst))
+            # open handles are path-keyed (add_block/complete_file RPCs):
+            # retarget them so a write fd keeps working under the new name
+            self._rewrite_handle_paths(src, dst)
         with self.nodes_lock:
             src_parent = self.nodes.get(nodeid)
             nid = src_parent.children.pop(oldname, None) if src_parent else None

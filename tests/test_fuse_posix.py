@@ -181,3 +181,35 @@ def test_rename_open_file_keeps_reading(mount):
     rest = f.read()
     f.close()
     assert first + rest == data
+
+
+def test_read_after_unlink(mount):
+    """POSIX: an open fd keeps serving data after unlink (store reader
+    refcounts defer block deletion)."""
+    mnt, *_ = mount
+    p = f"{mnt}/unlinked.bin"
+    data = os.urandom(2 << 20)
+    with open(p, "wb") as f:
+        f.write(data)
+    f = open(p, "rb")
+    head = f.read(1024)
+    os.unlink(p)
+    assert not os.path.exists(p)
+    f.seek(0)
+    assert f.read() == data          # full read after unlink
+    assert head == data[:1024]
+    f.close()
+
+
+def test_write_fd_survives_rename(mount):
+    """An open write fd keeps appending to the file under its new name."""
+    mnt, *_ = mount
+    src, dst = f"{mnt}/wr-a.bin", f"{mnt}/wr-b.bin"
+    f = open(src, "wb")
+    f.write(b"part1-")
+    f.flush()
+    os.rename(src, dst)
+    f.write(b"part2")
+    f.close()
+    assert not os.path.exists(src)
+    assert open(dst, "rb").read() == b"part1-part2"
