@@ -112,3 +112,44 @@ def test_s3_mount_read_through(s3, tmp_path):
             await mc.stop()
 
     asyncio.new_event_loop().run_until_complete(main())
+
+
+def test_s3_load_job_into_cache(s3, tmp_path):
+    """BASELINE config[3] flow on the CPU tier: `cv load` pulls S3
+    objects into the cache; after the S3 endpoint dies, reads still
+    serve from cache."""
+    import time
+
+    from curvine_amd.testing import MiniCluster
+    from curvine_amd.unified import UnifiedFileSystem
+
+    objs = {f"bkt/warm/f{i}.bin": os.urandom(400_000 + i * 31)
+            for i in range(4)}
+    s3.objects.update(objs)
+
+    async def main():
+        mc = await MiniCluster(tmp_dir=str(tmp_path / "cv")).start()
+        fs = UnifiedFileSystem(mc.client_conf())
+        try:
+            await fs.mount("/warm", "s3://bkt/warm", _props(s3),
+                           auto_cache=False)
+            job = await fs.submit_job("/warm", recursive=True)
+            deadline = time.time() + 30
+            while time.time() < deadline:
+                st = await fs.job_status(job["job_id"])
+                if st["state"] not in ("planning", "running"):
+                    break
+                await asyncio.sleep(0.2)
+            assert st["state"] == "completed", st
+            assert st["done"] == len(objs)
+            # S3 goes away; the cache must now be authoritative
+            s3.stop()
+            for key, data in objs.items():
+                name = key.rsplit("/", 1)[-1]
+                got = await fs.read_all(f"/warm/{name}")
+                assert got == data, name
+        finally:
+            await fs.close()
+            await mc.stop()
+
+    asyncio.new_event_loop().run_until_complete(main())
