@@ -185,8 +185,8 @@ class FsDir:
     # ---------------- mutations ----------------
     def mkdir(self, path: str, mode: int = 0o755, create_parents: bool = False) -> Inode:
         path = norm_path(path)
-        if self.resolve(path) is not None:
-            node = self.resolve(path)
+        node = self.resolve(path)
+        if node is not None:
             if node.is_dir:
                 return node
             raise err.FileAlreadyExists(path)
