@@ -261,11 +261,20 @@ class ClusterConf:
 
     def overlay(self, **kv: Any) -> "ClusterConf":
         """CLI overlay (ClientCliArgs analog): dotted keys, e.g.
-        overlay(**{"client.replicas": 2})."""
+        overlay(**{"client.replicas": 2}).  String values are coerced to
+        the field's existing type (mount -o options arrive as text)."""
         for key, v in kv.items():
             obj = self
             parts = key.split("__") if "__" in key else key.split(".")
             for p in parts[:-1]:
                 obj = getattr(obj, p)
+            cur = getattr(obj, parts[-1], None)
+            if isinstance(v, str) and not isinstance(cur, str):
+                if isinstance(cur, bool):
+                    v = v.lower() in ("1", "true", "yes", "on", "")
+                elif isinstance(cur, int):
+                    v = parse_bytes(v)
+                elif isinstance(cur, float):
+                    v = float(v)
             setattr(obj, parts[-1], v)
         return self

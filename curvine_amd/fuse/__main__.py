@@ -37,6 +37,11 @@ def main(argv=None) -> int:
                    help='worker data dir, e.g. "[HBM:200GB:0]gpu0"')
     p.add_argument("--web-port", type=int, default=0,
                    help="serve /metrics and /api/fuse on this port")
+    p.add_argument("-o", "--opt", action="append", default=[],
+                   help="mount-style options (cli/mount_args.rs analog): "
+                        "-o master=host:port,channels=8,web_port=9100,"
+                        "embed_worker,device=0, or dotted conf overlays "
+                        "like fuse.max_write=1048576")
     p.add_argument("--takeover", action="store_true",
                    help="hot upgrade: adopt the running daemon's session "
                         "fd + open-handle state instead of mounting")
@@ -47,7 +52,34 @@ def main(argv=None) -> int:
         level=getattr(logging, args.log_level.upper(), logging.INFO),
         format="%(asctime)s %(name)s %(levelname)s %(message)s")
 
+    # -o options fold into the flag namespace / conf overlay
+    overlays = {}
+    for group in args.opt:
+        for kv in group.split(","):
+            if not kv:
+                continue
+            k, _, v = kv.partition("=")
+            if k == "master":
+                args.master = args.master or v
+            elif k == "channels":
+                args.channels = args.channels or int(v)
+            elif k == "web_port":
+                args.web_port = args.web_port or int(v)
+            elif k == "embed_worker":
+                args.embed_worker = True
+            elif k == "device":
+                args.device = int(v)
+            elif k == "data_dir":
+                args.data_dir.append(v)
+            elif "." in k:
+                overlays[k] = v
+            else:
+                print(f"cv-fuse: unknown -o option {k!r}", file=sys.stderr)
+                return 2
+
     conf = ClusterConf.from_file(args.conf) if args.conf else ClusterConf()
+    if overlays:
+        conf.overlay(**overlays)
     if args.master:
         conf.client.master_addrs = [args.master]
         host, _, port = args.master.rpartition(":")

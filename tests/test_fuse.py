@@ -371,3 +371,39 @@ def test_inplace_rewrite_existing_file(mount):
         base[:5] = b"FRONT"
     assert os.path.getsize(p) == len(base)
     assert open(p, "rb").read() == bytes(base)
+
+
+def test_mount_o_options(tmp_path):
+    """cv-fuse -o option parsing (mount_args analog): master/channels and
+    dotted conf overlays with type coercion."""
+    from curvine_amd.testing import SyncMiniCluster
+
+    smc = SyncMiniCluster(tmp_dir=str(tmp_path / "cv")).start()
+    mnt = f"/tmp/curvine-fuse-oopt-{os.getpid()}"
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "curvine_amd.fuse", "--mnt", mnt,
+         "-o", f"master=127.0.0.1:{smc.master.rpc.port},channels=2",
+         "-o", "fuse.max_write=524288,client.enable_crc=true",
+         "--log-level", "WARNING"],
+        stdout=subprocess.PIPE, stderr=subprocess.DEVNULL, text=True,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    try:
+        line = proc.stdout.readline()
+        assert line.startswith("READY"), f"daemon failed: {line!r}"
+        with open(f"{mnt}/o.bin", "wb") as f:
+            f.write(b"k" * 100000)
+        assert open(f"{mnt}/o.bin", "rb").read() == b"k" * 100000
+    finally:
+        proc.terminate()
+        try:
+            proc.wait(timeout=10)
+        except subprocess.TimeoutExpired:
+            proc.kill()
+        smc.stop()
+        try:
+            os.rmdir(mnt)
+        except OSError:
+            pass
+    # unknown option rejected
+    from curvine_amd.fuse.__main__ import main as fmain
+    assert fmain(["--mnt", "/tmp/x", "-o", "bogus=1"]) == 2
