@@ -213,3 +213,62 @@ def test_write_fd_survives_rename(mount):
     f.close()
     assert not os.path.exists(src)
     assert open(dst, "rb").read() == b"part1-part2"
+
+
+def test_read_past_eof(mount):
+    """fuse_read_past_eof_test analog: preads at and beyond EOF return
+    empty, short tail reads clamp."""
+    mnt, *_ = mount
+    p = f"{mnt}/eof.bin"
+    data = os.urandom(10_000)
+    with open(p, "wb") as f:
+        f.write(data)
+    fd = os.open(p, os.O_RDONLY)
+    try:
+        assert os.pread(fd, 100, 10_000) == b""          # at EOF
+        assert os.pread(fd, 100, 50_000) == b""          # far past EOF
+        assert os.pread(fd, 1000, 9_500) == data[9_500:] # clamped tail
+    finally:
+        os.close(fd)
+
+
+def test_truncate_storm_under_readers(mount):
+    """resize_lock_p95_test analog: concurrent readers during a truncate
+    storm never deadlock or corrupt (bounded by the test timeout)."""
+    import threading as _th
+    mnt, *_ = mount
+    p = f"{mnt}/storm.bin"
+    with open(p, "wb") as f:
+        f.write(b"S" * 1_000_000)
+    stop = _th.Event()
+    errors = []
+
+    def reader():
+        while not stop.is_set():
+            try:
+                with open(p, "rb") as f:
+                    chunk = f.read(65536)
+                    assert set(chunk) <= {ord("S")}
+            except FileNotFoundError:
+                pass
+            except OSError:
+                pass   # racing a shrink is allowed to return EIO once
+            except Exception as e:  # noqa: BLE001
+                errors.append(e)
+                return
+
+    threads = [_th.Thread(target=reader) for _ in range(4)]
+    for t in threads:
+        t.start()
+    try:
+        for i in range(30):
+            os.truncate(p, 500_000 if i % 2 else 1_000_000)
+            if i % 2:
+                with open(p, "ab") as f:
+                    f.write(b"S" * 500_000)
+    finally:
+        stop.set()
+        for t in threads:
+            t.join(timeout=30)
+    assert not errors
+    assert os.path.getsize(p) == 1_000_000
