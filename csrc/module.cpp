@@ -732,6 +732,110 @@ static void arena_read_batch(int h, const std::vector<uint64_t>& offs,
 }
 
 // ---------------------------------------------------------------------------
+// Registered readers: a file's arena extent table pinned in C++ so batched
+// random preads resolve + issue + sync entirely GIL-free (the 4 KiB IOPS
+// path; the client-side sibling of the native FUSE read registration).
+// Safety contract (curvine_amd/client/reader.py): the SyncLocalReader
+// holds its store readers open (block refcounts defer delete/demote) for
+// the registration's whole lifetime and unregisters before closing them.
+// ---------------------------------------------------------------------------
+
+struct RdExt { uint64_t file_off, len; int arena; uint64_t arena_off; };
+struct RdReader { uint64_t length; std::vector<RdExt> exts; };
+static std::mutex g_rd_mu;
+static std::unordered_map<int64_t, RdReader> g_rd;
+static int64_t g_rd_next = 1;
+
+static int64_t reader_register(
+    std::vector<std::tuple<uint64_t, uint64_t, int, uint64_t>> exts,
+    uint64_t length) {
+  RdReader r;
+  r.length = length;
+  r.exts.reserve(exts.size());
+  uint64_t prev = 0;
+  for (auto& e : exts) {
+    if (std::get<0>(e) < prev)
+      throw std::runtime_error("reader_register: extents must be sorted");
+    prev = std::get<0>(e);
+    get_arena(std::get<2>(e));   // validate handle
+    r.exts.push_back({std::get<0>(e), std::get<1>(e), std::get<2>(e),
+                      std::get<3>(e)});
+  }
+  std::lock_guard<std::mutex> g(g_rd_mu);
+  int64_t rid = g_rd_next++;
+  g_rd[rid] = std::move(r);
+  return rid;
+}
+
+static void reader_unregister(int64_t rid) {
+  std::lock_guard<std::mutex> g(g_rd_mu);
+  g_rd.erase(rid);
+}
+
+// fixed-size batched preads: for each file offset, land n bytes at
+// dst_ptr + i*stride.  Returns the indices it could NOT serve (reads
+// spanning extent boundaries or past EOF) so the caller falls back for
+// exactly those.
+static py::list reader_pread_batch(int64_t rid,
+                                   const std::vector<uint64_t>& offs,
+                                   uint64_t n, uintptr_t dst_ptr,
+                                   uint64_t stride) {
+  RdReader* r;
+  {
+    std::lock_guard<std::mutex> g(g_rd_mu);
+    auto it = g_rd.find(rid);
+    if (it == g_rd.end()) throw std::runtime_error("bad reader id");
+    r = &it->second;
+  }
+  std::vector<uint32_t> skipped;
+  {
+    py::gil_scoped_release rel;
+    int used_dev[8];
+    int n_dev = 0;
+    const auto& exts = r->exts;
+    for (size_t i = 0; i < offs.size(); ++i) {
+      uint64_t off = offs[i];
+      const RdExt* e = nullptr;
+      if (off + n <= r->length && !exts.empty()) {
+        // binary search: last extent with file_off <= off
+        size_t lo = 0, hi = exts.size();
+        while (lo < hi) {
+          size_t mid = (lo + hi) / 2;
+          if (exts[mid].file_off <= off) lo = mid + 1; else hi = mid;
+        }
+        if (lo > 0 && off - exts[lo - 1].file_off + n <= exts[lo - 1].len)
+          e = &exts[lo - 1];
+      }
+      if (e == nullptr) {
+        skipped.push_back((uint32_t)i);
+        continue;
+      }
+      uint64_t boff = off - e->file_off;
+      Arena* a = get_arena(e->arena);
+      uint8_t* dst = (uint8_t*)dst_ptr + i * stride;
+      const uint8_t* src = (uint8_t*)a->base + e->arena_off + boff;
+      if (!a->is_dev()) {
+        std::memcpy(dst, src, n);
+      } else {
+        HIP_CHECK(hipSetDevice(a->device));
+        HIP_CHECK(hipMemcpyAsync(dst, src, n, hipMemcpyDeviceToHost,
+                                 thread_stream(a->device)));
+        bool seen = false;
+        for (int d = 0; d < n_dev; d++) seen |= (used_dev[d] == a->device);
+        if (!seen && n_dev < 8) used_dev[n_dev++] = a->device;
+      }
+    }
+    for (int d = 0; d < n_dev; d++) {
+      HIP_CHECK(hipSetDevice(used_dev[d]));
+      HIP_CHECK(hipStreamSynchronize(thread_stream(used_dev[d])));
+    }
+  }
+  py::list out;
+  for (uint32_t i : skipped) out.append(i);
+  return out;
+}
+
+// ---------------------------------------------------------------------------
 // LZ4 container: compress/decompress (host) + decompress-into-arena (GPU)
 // ---------------------------------------------------------------------------
 
@@ -1042,6 +1146,9 @@ PYBIND11_MODULE(_native, m) {
   m.def("arena_gather_ptr", &arena_gather_ptr);
   m.def("arena_base_ptr", &arena_base_ptr);
   m.def("arena_read_batch", &arena_read_batch);
+  m.def("reader_register", &reader_register);
+  m.def("reader_unregister", &reader_unregister);
+  m.def("reader_pread_batch", &reader_pread_batch);
   m.def("arena_info", &arena_info);
   m.def("arena_dlpack", &arena_dlpack);
   m.def("arena_host_view", &arena_host_view);

@@ -51,6 +51,7 @@ class SyncLocalReader:
         self.length = file_blocks.status.length
         self._offs = [b.offset for b in file_blocks.blocks]
         self._readers = []
+        self._native_rid = None
         try:
             for lb in file_blocks.blocks:
                 r = None
@@ -66,6 +67,30 @@ class SyncLocalReader:
         except Exception:
             self.close()
             raise
+        self._try_register_native()
+
+    def _try_register_native(self) -> None:
+        """Pin the extent table in C++ when every block is arena-backed:
+        batched preads then resolve+issue+sync GIL-free (reader_register
+        in csrc/module.cpp).  The store readers stay open for the whole
+        registration (refcounts defer delete/demote), mirroring the
+        native FUSE read contract."""
+        exts = []
+        for lb, r in zip(self.fb.blocks, self._readers):
+            meta = r.meta
+            arena = getattr(r.layout, "arena", None)
+            if meta.get("kind") != "arena" or arena is None:
+                return
+            exts.append((lb.offset, lb.block.length, arena.handle,
+                         meta["offset"]))
+        if not exts:
+            return
+        try:
+            from curvine_amd import native
+            self._native_rid = native.load().reader_register(
+                exts, self.length)
+        except Exception:  # noqa: BLE001 — extension absent on this host
+            self._native_rid = None
 
     def pread_into(self, off: int, out, out_off: int, n: int) -> int:
         import bisect
@@ -117,6 +142,14 @@ class SyncLocalReader:
         from curvine_amd import native
         if not self.fb.blocks:
             return 0
+        if self._native_rid is not None:
+            skipped = native.load().reader_pread_batch(
+                self._native_rid, file_offs, n, dst_ptr, stride)
+            for i in skipped:
+                buf = bytearray(n)
+                got = self.pread_into(file_offs[i], buf, 0, n)
+                ctypes.memmove(dst_ptr + i * stride, bytes(buf[:got]), got)
+            return len(file_offs)
         groups: dict[int, tuple[list, list]] = {}  # arena handle -> offs, dsts
         slow: list[tuple[int, int]] = []
         for i, off in enumerate(file_offs):
@@ -184,6 +217,15 @@ class SyncLocalReader:
         return total
 
     def close(self) -> None:
+        if getattr(self, "_native_rid", None) is not None:
+            # unregister BEFORE releasing the store readers that pin the
+            # registered extents
+            try:
+                from curvine_amd import native
+                native.load().reader_unregister(self._native_rid)
+            except Exception:  # noqa: BLE001
+                pass
+            self._native_rid = None
         for r in getattr(self, "_readers", []):
             try:
                 r.close()
