@@ -181,6 +181,15 @@ class SyncLocalReader:
         on-chip copy_extents_kernel — the HBM-cache -> torch-device-tensor
         training-ingest path with no host hop); everything else falls back
         to pread_into_ptr DMA.  Returns total bytes gathered."""
+        groups, slow, total = self.resolve_gather(samples, dst_on_device)
+        self.exec_gather(groups, slow, dst_ptr)
+        return total
+
+    def resolve_gather(self, samples, dst_on_device: bool):
+        """Resolution half of pread_gather, separable so callers with a
+        static access pattern (CurvineDeviceLoader) resolve once and
+        replay `exec_gather` per epoch.  Valid while this reader is open
+        (its store readers pin the blocks)."""
         import bisect
         groups: dict[int, tuple[object, list]] = {}  # handle -> (arena, tri)
         slow: list[tuple[int, int, int]] = []
@@ -210,11 +219,13 @@ class SyncLocalReader:
                     slow.append((off + got, want, doff + got))
                 got += want
             total += got
-        for arena, tri in groups.values():
+        return list(groups.values()), slow, total
+
+    def exec_gather(self, groups, slow, dst_ptr: int) -> None:
+        for arena, tri in groups:
             arena.gather_ptr(tri, dst_ptr)
         for off, want, doff in slow:
             self.pread_into_ptr(off, dst_ptr + doff, want)
-        return total
 
     def close(self) -> None:
         if getattr(self, "_native_rid", None) is not None:

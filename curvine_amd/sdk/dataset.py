@@ -167,6 +167,7 @@ class CurvineDeviceLoader:
         self._fs = SyncFs(conf)
         self._readers = []
         self._samples = []             # (reader_idx, name, off, size)
+        self._plans = None             # cached batch gather plans
         for sp in shard_paths:
             fb = self._fs.call(self._fs.fs.client.open(sp))
             r = SyncLocalReader(fb)
@@ -182,30 +183,47 @@ class CurvineDeviceLoader:
     def num_samples(self) -> int:
         return len(self._samples)
 
-    def __iter__(self):
-        import torch
+    def _plan(self, on_dev: bool):
+        """Batch plans with sample extents pre-resolved down to arena
+        triples (valid while the readers stay open — they pin the
+        blocks): iteration is just torch.empty + replayed gather calls.
+        The gather kernel runs ~13 µs per 16 MiB batch; Python must not
+        dominate it."""
         order = list(range(len(self._samples)))
         if self.shuffle:
             import random
             random.Random(self.seed).shuffle(order)
-        dev = torch.device(self.device)
-        on_dev = dev.type != "cpu"
+        plans = []
         bs = self.batch_size
         for b0 in range(0, len(order), bs):
             batch = [self._samples[i] for i in order[b0:b0 + bs]]
-            total = sum(s[3] for s in batch)
-            out = torch.empty(total, dtype=torch.uint8, device=dev)
-            dst = out.data_ptr()
-            sections, names = [], []
             per_reader: dict[int, list] = {}
+            sections, names = [], []
             pos = 0
             for ridx, name, off, size in batch:
                 per_reader.setdefault(ridx, []).append((off, size, pos))
                 sections.append((pos, size))
                 names.append(name)
                 pos += size
+            execs = []
             for ridx, samples in per_reader.items():
-                self._readers[ridx].pread_gather(samples, dst, on_dev)
+                groups, slow, _ = self._readers[ridx].resolve_gather(
+                    samples, on_dev)
+                execs.append((self._readers[ridx], groups, slow))
+            plans.append((pos, execs, sections, names))
+        return plans
+
+    def __iter__(self):
+        import torch
+        dev = torch.device(self.device)
+        on_dev = dev.type != "cpu"
+        if self._plans is None:
+            self._plans = self._plan(on_dev)
+        for total, execs, sections, names in self._plans:
+            out = torch.empty(total, dtype=torch.uint8, device=dev)
+            dst = out.data_ptr()
+            for reader, groups, slow in execs:
+                reader.exec_gather(groups, slow, dst)
             if on_dev:
                 torch.cuda.synchronize(dev)
             yield out, sections, names
