@@ -255,6 +255,35 @@ def _sync_readers(args, rank, rt):
     return readers
 
 
+def step_client_seq_remote(args, rank, rt) -> int:
+    """Sequential reads with short-circuit DISABLED: every byte crosses
+    the worker's streaming RPC (the inter-node data plane, exercised
+    over loopback — one concurrent stream per file)."""
+    async def run():
+        rt.fs.client.conf.client.short_circuit = False
+        readers = [await rt.fs.open(f"/bench/r{rank}/f{i}")
+                   for i in range(args.files)]
+
+        async def read_file(r):
+            buf = bytearray(args.read_chunk)
+            pos = 0
+            while pos < r.length:
+                want = min(args.read_chunk, r.length - pos)
+                got = await r.pread_into(pos, buf, 0, want)
+                if got <= 0:
+                    break
+                pos += got
+            return pos
+        try:
+            totals = await asyncio.gather(*(read_file(r) for r in readers))
+        finally:
+            for r in readers:
+                r.close()
+            rt.fs.client.conf.client.short_circuit = True
+        return sum(totals)
+    return rt.call(run())
+
+
 def step_client_rand4k(args, rank, rt, lat_out: list) -> int:
     """Random 4 KiB reads via the SYNC short-circuit path: plain OS
     threads, no event loop in the per-op path (this is the IOPS metric)."""
@@ -395,6 +424,8 @@ def main():
                    help="queue depth per thread for randread4k (fio iodepth)")
     p.add_argument("--seq-batch", type=int, default=4,
                    help="chunks per sync in seqread (DMA pipelining depth)")
+    p.add_argument("--no-short-circuit", action="store_true",
+                   help="force the streaming worker-RPC read path")
     p.add_argument("--hbm-gb", type=int, default=16)
     p.add_argument("--staging-bytes", type=int, default=8 << 20)
     p.add_argument("--staging-count", type=int, default=8)
@@ -465,6 +496,8 @@ def main():
             return step_fuse_rand4k(args, rank, mnt, lat_out)
         if args.path == "fuse":
             return step_fuse_seq(args, rank, mnt)
+        if args.no_short_circuit:
+            return step_client_seq_remote(args, rank, rt)
         return step_client_seq(args, rank, rt)
 
     for _ in range(args.warmup):
@@ -524,6 +557,7 @@ def main():
                 "iodepth": args.iodepth,
                 "block_size": args.block_size,
                 "fuse_channels": args.fuse_channels,
+                "short_circuit": not args.no_short_circuit,
                 "GiBps": round(gibps, 3),
                 "latency": lat_out[-1] if lat_out else None,
                 "parallelism": f"shard-per-gpu x{world}",
