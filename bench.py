@@ -119,6 +119,30 @@ def setup(args, rank, world, dist, has_gpu):
         line = daemon_proc.stdout.readline()
         if not line.startswith("READY"):
             raise RuntimeError(f"cv-fuse failed to start: {line!r}")
+    elif args.no_short_circuit:
+        # faithful remote plane: the worker is its OWN process, every
+        # byte crosses its streaming RPC over loopback
+        wp = subprocess.Popen(
+            [sys.executable, "-m", "curvine_amd.server_main",
+             "--service", "worker", "--master-port", str(master_port),
+             "--device", str(local_rank if has_gpu else -1),
+             "--log-level", "WARNING"] +
+            [f"--data-dir={d}" for d in data_dirs],
+            stderr=sys.stderr,
+            cwd=os.path.dirname(os.path.abspath(__file__)))
+        daemon_proc = wp
+
+        async def wait_worker():
+            from curvine_amd.client.filesystem import CurvineFileSystem as _C
+            f = _C(conf)
+            for _ in range(100):
+                info = await f.get_master_info()
+                if info["live_workers"]:
+                    await f.close()
+                    return
+                await asyncio.sleep(0.2)
+            raise RuntimeError("worker process never registered")
+        rt.call(wait_worker())
     else:
         conf.worker.data_dirs = data_dirs
         conf.worker.rpc_port = 0
