@@ -181,7 +181,7 @@ class BlockReaderRemote:
         self.block_id = block_id
         self.length = 0
 
-    async def read_range(self, off: int, n: int, chunk_size: int = 1 << 20):
+    async def read_range(self, off: int, n: int, chunk_size: int = 4 << 20):
         """Async iterator of chunks covering [off, off+n)."""
         client = await factory().get(self.addr.hostname, self.addr.rpc_port)
         stream = client.stream(RpcCode.ReadBlock)
