@@ -139,6 +139,16 @@ class _Delegate:
             return si.value
 
 
+class _BadFrame(Exception):
+    """Frame whose msgpack header failed to decode: reply with a wire
+    error instead of letting the exception escape the dispatch loop."""
+
+    def __init__(self, msg, cause):
+        super().__init__(str(cause))
+        self.msg = msg
+        self.cause = cause if isinstance(cause, Exception) else Exception(cause)
+
+
 class _FwdConn:
     """Shim standing in for rpc.server.ServerConn on forwarded frames
     (master handlers are unary and never touch it beyond attributes)."""
@@ -302,14 +312,25 @@ class NativeMetaFrontend:
         for conn_id, bufs in replies.items():
             self.lib.meta_send(self.sid, conn_id, b"".join(bufs))
 
+    @staticmethod
+    def _decode_frame(raw: bytes) -> Message:
+        hlen, dlen, msg = Message.decode_proto(raw[:PROTO_SIZE])
+        try:
+            if hlen:
+                msg.set_header_bytes(raw[PROTO_SIZE:PROTO_SIZE + hlen])
+        except Exception as e:  # noqa: BLE001 — undecodable header
+            raise _BadFrame(msg, e)
+        if dlen:
+            msg.data = raw[PROTO_SIZE + hlen:PROTO_SIZE + hlen + dlen]
+        return msg
+
     def _handle_sync(self, conn_id: int, raw: bytes) -> Optional[bytes]:
         """Drive one handler coroutine synchronously; falls back to the
         ordered queue path if it unexpectedly suspends."""
-        hlen, dlen, msg = Message.decode_proto(raw[:PROTO_SIZE])
-        if hlen:
-            msg.set_header_bytes(raw[PROTO_SIZE:PROTO_SIZE + hlen])
-        if dlen:
-            msg.data = raw[PROTO_SIZE + hlen:PROTO_SIZE + hlen + dlen]
+        try:
+            msg = self._decode_frame(raw)
+        except _BadFrame as bf:
+            return bf.msg.error_reply(bf.cause).encode()
         coro = self._handler.handle(msg, _FwdConn(conn_id))
         try:
             y = coro.send(None)
@@ -366,11 +387,10 @@ class NativeMetaFrontend:
 
     async def _handle_raw(self, raw: bytes, conn) -> Optional[bytes]:
         try:
-            hlen, dlen, msg = Message.decode_proto(raw[:PROTO_SIZE])
-            if hlen:
-                msg.set_header_bytes(raw[PROTO_SIZE:PROTO_SIZE + hlen])
-            if dlen:
-                msg.data = raw[PROTO_SIZE + hlen:PROTO_SIZE + hlen + dlen]
+            try:
+                msg = self._decode_frame(raw)
+            except _BadFrame as bf:
+                return bf.msg.error_reply(bf.cause).encode()
             try:
                 reply = await self._handler.handle(msg, conn)
             except asyncio.CancelledError:

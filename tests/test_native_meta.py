@@ -204,3 +204,45 @@ def test_encoding_boundaries(cluster):
     stats = cluster.master.native_meta.stats()
     assert stats["served_status"] >= 3
     sf.shutdown()
+
+
+def test_garbage_frames_do_not_wedge(cluster):
+    """Malformed input on the native meta port: oversized lengths close
+    the connection; valid clients keep working throughout."""
+    import socket
+    import struct
+
+    from curvine_amd.client.filesystem import SyncFs
+
+    port = cluster.master.rpc.port
+    sf = SyncFs(cluster.client_conf())
+    sf.mkdir("/gz", create_parents=True)
+
+    # oversized header length -> server closes the conn
+    s1 = socket.create_connection(("127.0.0.1", port), timeout=5)
+    s1.sendall(struct.pack(">IIBBQI", 1 << 30, 0, 7, 0, 1, 0))
+    s1.settimeout(5)
+    assert s1.recv(1) == b""          # closed on us
+    s1.close()
+
+    # truncated frame then disconnect mid-header
+    s2 = socket.create_connection(("127.0.0.1", port), timeout=5)
+    s2.sendall(b"\x00\x00\x00\x10")
+    s2.close()
+
+    # undecodable msgpack header on a hot code -> forwarded to Python,
+    # which answers with an error reply rather than hanging
+    s3 = socket.create_connection(("127.0.0.1", port), timeout=5)
+    junk = b"\xc1\xff\xfe"            # 0xc1 is an invalid msgpack byte
+    s3.sendall(struct.pack(">IIBBQI", len(junk), 0, 7, 0, 42, 0) + junk)
+    s3.settimeout(10)
+    hdr = s3.recv(22)
+    assert len(hdr) == 22
+    _hl, _dl, code, status, req_id, _seq = struct.unpack(">IIBBQI", hdr)
+    assert req_id == 42 and (status >> 4) == 5   # error reply
+    s3.close()
+
+    # the real client is unaffected
+    assert sf.exists("/gz")
+    assert sf.file_status("/gz").is_dir
+    sf.shutdown()
