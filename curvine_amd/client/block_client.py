@@ -62,8 +62,10 @@ class BlockWriterLocal:
         if length is None:       # in-place rewrite: nothing to publish
             return ""
         loop = asyncio.get_event_loop()
-        return await loop.run_in_executor(None, self.store.finalize,
+        tier = await loop.run_in_executor(None, self.store.finalize,
                                           self.block_id, length)
+        self.last_crc = self.store.block_crc(self.block_id)
+        return tier
 
     async def abort(self) -> None:
         if not self.reopen:
@@ -121,6 +123,7 @@ class BlockWriterRemote:
         await self.stream.send(hdr, b"", Status.Complete)
         reply = await self.stream.recv()
         self.stream.close()
+        self.last_crc = reply.header.get("crc32c")
         return reply.header.get("tier", "")
 
     async def abort(self) -> None:

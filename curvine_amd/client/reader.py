@@ -227,6 +227,20 @@ class SyncLocalReader:
         for off, want, doff in slow:
             self.pread_into_ptr(off, dst_ptr + doff, want)
 
+    def verify(self) -> list[int]:
+        """Recompute every resident block's CRC32C (HBM: device kernel)
+        against the worker's publish-time value; returns block ids that
+        no longer match (bit rot / unexpected mutation).  Blocks without
+        a stored CRC (rewritten in place) are skipped."""
+        bad = []
+        for lb, r in zip(self.fb.blocks, self._readers):
+            stored = r.meta.get("crc32c")
+            if stored is None:
+                continue
+            if r.crc32c(0, lb.block.length) != stored:
+                bad.append(lb.block.block_id)
+        return bad
+
     def close(self) -> None:
         if getattr(self, "_native_rid", None) is not None:
             # unregister BEFORE releasing the store readers that pin the

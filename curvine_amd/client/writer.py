@@ -33,6 +33,11 @@ class FsWriter:
         self._commits: list[dict] = []   # block locations to report on complete
         self.pos = 0
         self._closed = False
+        # end-to-end write integrity: keep a running CRC32C per block and
+        # cross-check the worker's publish-time CRC at commit
+        self._crc_on = bool(client.conf.client.enable_crc)
+        self._blk_crc = 0
+        self._crc_valid = True
 
     async def write(self, data) -> int:
         if self._closed:
@@ -61,6 +66,9 @@ class FsWriter:
             chunk = bytes(buf[off:off + take])
             # star fan-out: all replicas in parallel
             await asyncio.gather(*[w.write(chunk) for w in self._writers])
+            if self._crc_on:
+                from curvine_amd import native
+                self._blk_crc = native.crc32c(chunk, self._blk_crc)
             self._block_pos += take
             off += take
             if self._block_pos >= self.block_size:
@@ -79,6 +87,15 @@ class FsWriter:
     async def _commit_block(self) -> None:
         tiers = await asyncio.gather(
             *[w.commit(self._block_pos) for w in self._writers])
+        if self._crc_on and self._crc_valid:
+            for w in self._writers:
+                got = getattr(w, "last_crc", None)
+                if got is not None and got != self._blk_crc:
+                    raise err.ChecksumMismatch(
+                        f"block {self._block.block.block_id}: worker crc "
+                        f"{got:#010x} != client {self._blk_crc:#010x}")
+        self._blk_crc = 0
+        self._crc_valid = True
         self._commits.append({
             "block_id": self._block.block.block_id,
             "locations": [a.worker_id for a in self._block.locations],
@@ -106,6 +123,7 @@ class FsWriter:
         n = len(data)
         if off + n > self.pos:
             raise err.OutOfRange(f"rewrite [{off},{off + n}) past {self.pos}")
+        self._crc_valid = False   # running CRC no longer linear
         cur_start = sum(self._block_lens)
         consumed = 0
         while consumed < n:

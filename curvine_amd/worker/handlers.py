@@ -91,7 +91,9 @@ class WorkerHandler:
             tier = await loop.run_in_executor(
                 None, self.store.finalize, sess["block_id"], length)
             self.writes.pop(msg.req_id, None)
-            return msg.reply({"tier": tier}, resp_status=Status.Complete)
+            return msg.reply({"tier": tier,
+                              "crc32c": self.store.block_crc(sess["block_id"])},
+                             resp_status=Status.Complete)
         if msg.req_status == Status.Cancel:
             await loop.run_in_executor(None, self.store.abort, sess["block_id"])
             self.writes.pop(msg.req_id, None)
