@@ -316,4 +316,4 @@ class MasterHandler:
         return self.master.raft.on_append(h)
 
     def op_raftinstallsnapshot(self, h, d):
-        return self.master.raft.on_install_snapshot(h)
+        return self.master.raft.on_install_snapshot(h, d)

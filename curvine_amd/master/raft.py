@@ -399,17 +399,30 @@ class RaftNode:
         else:
             self.next_index[pid] = max(1, h.get("hint", ni - 1))
 
+    SNAP_CHUNK = 8 << 20   # stay under the 16 MiB frame payload cap
+
     async def _send_snapshot(self, pid: int):
-        snap = self.make_snapshot()
+        """Chunked install (raft/snapshot streaming analog): the state is
+        msgpack-serialized once and streamed in SNAP_CHUNK frames — a
+        multi-GB namespace must never need a single oversized frame."""
+        blob = msgpack.packb(self.make_snapshot(), use_bin_type=True)
+        index = self.log.last_index
+        snap_term = self.log.term_at(index)
+        nchunks = max(1, (len(blob) + self.SNAP_CHUNK - 1) // self.SNAP_CHUNK)
         try:
             c = await self._client(pid)
-            await c.rpc(RpcCode.RaftInstallSnapshot, {
-                "term": self.term, "leader": self.id,
-                "index": self.log.last_index,
-                "snap_term": self.log.term_at(self.log.last_index),
-                "state": snap}, timeout=10.0)
-            self.next_index[pid] = self.log.last_index + 1
-            self.match_index[pid] = self.log.last_index
+            for seq in range(nchunks):
+                chunk = blob[seq * self.SNAP_CHUNK:(seq + 1) * self.SNAP_CHUNK]
+                r = await c.rpc(RpcCode.RaftInstallSnapshot, {
+                    "term": self.term, "leader": self.id,
+                    "index": index, "snap_term": snap_term,
+                    "seq": seq, "nchunks": nchunks},
+                    data=chunk, timeout=10.0)
+                if r.header.get("term", 0) > self.term:
+                    self._become_follower(r.header["term"])
+                    return
+            self.next_index[pid] = index + 1
+            self.match_index[pid] = index
         except Exception as e:  # noqa: BLE001
             log.debug("snapshot to %d failed: %s", pid, e)
 
@@ -455,12 +468,29 @@ class RaftNode:
             self._apply_committed()
         return {"term": self.term, "success": True}
 
-    def on_install_snapshot(self, h: dict) -> dict:
+    def on_install_snapshot(self, h: dict, data: bytes = b"") -> dict:
         if h["term"] < self.term:
             return {"term": self.term}
         self._become_follower(h["term"])
         self.leader_id = h["leader"]
-        self.load_snapshot_cb(h["state"])
+        if "state" in h:            # legacy single-frame install
+            state = h["state"]
+        else:
+            key = (h["leader"], h["index"], h.get("nchunks", 1))
+            if getattr(self, "_snap_rx_key", None) != key:
+                self._snap_rx_key = key
+                self._snap_rx = []
+            if h.get("seq", 0) != len(self._snap_rx):
+                # out-of-order / replayed chunk: restart the transfer
+                self._snap_rx_key = None
+                return {"term": self.term, "restart": True}
+            self._snap_rx.append(bytes(data))
+            if len(self._snap_rx) < h.get("nchunks", 1):
+                return {"term": self.term, "ok": True}
+            blob = b"".join(self._snap_rx)
+            self._snap_rx_key, self._snap_rx = None, []
+            state = msgpack.unpackb(blob, raw=False)
+        self.load_snapshot_cb(state)
         self.log.compact_to(h["index"], h["snap_term"])
         self.commit_index = h["index"]
         self.last_applied = h["index"]

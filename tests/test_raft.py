@@ -134,3 +134,54 @@ def test_restart_catches_up(loop, tmp_path):
         for m in live + [restarted]:
             await m.stop()
     run(loop, main())
+
+
+def test_chunked_snapshot_install(loop, tmp_path, monkeypatch):
+    """A follower far behind a compacted log catches up via the CHUNKED
+    snapshot stream (multi-frame; single-frame installs would cap the
+    namespace at the 16 MiB frame limit)."""
+    from curvine_amd.master.raft import RaftNode
+    monkeypatch.setattr(RaftNode, "SNAP_CHUNK", 4096)   # force many chunks
+
+    async def main():
+        masters, ports, peers = await start_group(tmp_path)
+        leader = await wait_leader(masters)
+        conn = ClusterConnector([f"127.0.0.1:{p}" for p in ports],
+                                timeout_ms=8000, retries=6)
+        follower = next(m for m in masters if not m.raft.is_leader)
+        fid = follower.conf.journal.node_id
+        fconf = copy.deepcopy(follower.conf)
+        await follower.stop()
+        live = [m for m in masters if m is not follower]
+
+        # grow the namespace (long names -> a multi-chunk snapshot blob),
+        # then compact the leader's log so append catch-up is impossible
+        for i in range(40):
+            await conn.rpc(RpcCode.Mkdir, {
+                "path": f"/snapdir/{'x' * 200}-{i}",
+                "create_parents": True})
+        leader = await wait_leader(live)
+        leader.checkpoint()
+        assert leader.raft.log.snapshot_index > 0
+
+        fconf.master.rpc_port = ports[fid - 1]
+        restarted = await Master(fconf).start()
+        for _ in range(150):
+            await asyncio.sleep(0.1)
+            if restarted.fs.fs_dir.resolve(
+                    f"/snapdir/{'x' * 200}-39") is not None:
+                break
+        assert restarted.fs.fs_dir.resolve(
+            f"/snapdir/{'x' * 200}-39") is not None
+        # and it keeps participating
+        await conn.rpc(RpcCode.Mkdir, {"path": "/after_snap",
+                                       "create_parents": True})
+        for _ in range(50):
+            await asyncio.sleep(0.1)
+            if restarted.fs.fs_dir.resolve("/after_snap") is not None:
+                break
+        assert restarted.fs.fs_dir.resolve("/after_snap") is not None
+        await conn.close()
+        for m in live + [restarted]:
+            await m.stop()
+    run(loop, main())
