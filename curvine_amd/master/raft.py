@@ -236,7 +236,34 @@ class RaftNode:
         return c
 
     # ---------------- election ----------------
+    async def _pre_vote(self) -> bool:
+        """Pre-vote (raft §9.6 / PreVote extension): ask peers whether
+        they WOULD grant term+1 before bumping our term — a partitioned
+        node rejoining can no longer term-inflate a healthy cluster."""
+        term = self.term + 1
+        grants = 1
+
+        async def ask(pid):
+            try:
+                c = await self._client(pid)
+                r = await c.rpc(RpcCode.RaftVote, {
+                    "term": term, "candidate": self.id, "prevote": True,
+                    "last_log_index": self.log.last_index,
+                    "last_log_term": self.log.term_at(self.log.last_index)},
+                    timeout=1.5)
+                return r.header
+            except Exception:  # noqa: BLE001
+                return None
+
+        for r in await asyncio.gather(*[ask(p) for p in self.peers]):
+            if r and r.get("granted"):
+                grants += 1
+        return grants * 2 > len(self.peers) + 1
+
     async def _run_election(self):
+        if not await self._pre_vote():
+            self._last_heard = time.monotonic()   # back off, stay follower
+            return
         self.state = CANDIDATE
         if self.on_role_change is not None:
             self.on_role_change(False)
@@ -428,14 +455,24 @@ class RaftNode:
 
     # ---------------- RPC handlers (called from MasterHandler) ----------------
     def on_vote(self, h: dict) -> dict:
+        my_last_term = self.log.term_at(self.log.last_index)
+        log_ok = (h["last_log_term"], h["last_log_index"]) >= \
+            (my_last_term, self.log.last_index)
+        if h.get("prevote"):
+            # no term bump, no persisted vote: grant only if the log is
+            # current AND we have not heard from a live leader recently
+            leader_quiet = (time.monotonic() - self._last_heard) \
+                >= self.election_timeout
+            return {"term": self.term,
+                    "granted": bool(h["term"] >= self.term and log_ok
+                                    and (leader_quiet
+                                         or self.state != FOLLOWER))}
         if h["term"] > self.term:
             self._become_follower(h["term"])
         granted = False
         if h["term"] >= self.term and \
                 self.voted_for in (None, h["candidate"]):
-            my_last_term = self.log.term_at(self.log.last_index)
-            if (h["last_log_term"], h["last_log_index"]) >= \
-                    (my_last_term, self.log.last_index):
+            if log_ok:
                 granted = True
                 self.voted_for = h["candidate"]
                 self._save_meta()

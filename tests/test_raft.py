@@ -185,3 +185,31 @@ def test_chunked_snapshot_install(loop, tmp_path, monkeypatch):
         for m in live + [restarted]:
             await m.stop()
     run(loop, main())
+
+
+def test_prevote_blocks_term_inflation(loop, tmp_path):
+    """A node without quorum must not inflate its term while isolated
+    (pre-vote): when peers return, the cluster resumes at a sane term."""
+    async def main():
+        masters, ports, peers = await start_group(tmp_path)
+        leader = await wait_leader(masters)
+        term0 = leader.raft.term
+        survivor = next(m for m in masters if not m.raft.is_leader)
+        others = [m for m in masters if m is not survivor]
+        confs = [copy.deepcopy(m.conf) for m in others]
+        for m in others:
+            await m.stop()
+        # several election timeouts pass with no quorum
+        await asyncio.sleep(2.5)
+        assert survivor.raft.term <= term0 + 1, \
+            f"term inflated to {survivor.raft.term}"
+        assert not survivor.raft.is_leader
+        # peers return: a leader emerges without a huge term jump
+        restarted = []
+        for c, p in zip(confs, [m.conf.master.rpc_port for m in others]):
+            restarted.append(await Master(c).start())
+        leader2 = await wait_leader([survivor] + restarted, timeout=15)
+        assert leader2.raft.term <= term0 + 3
+        for m in [survivor] + restarted:
+            await m.stop()
+    run(loop, main())
