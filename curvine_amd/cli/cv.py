@@ -224,6 +224,16 @@ def cmd_node_decommission(fs, args):
     print(json.dumps(fs.call(run())))
 
 
+def cmd_transfer_leader(fs, args):
+    from curvine_amd.rpc.codes import RpcCode
+
+    async def run():
+        r = await fs.fs.client.connector.rpc(
+            RpcCode.RaftTransferLeader, {"target": args.node_id})
+        return r.header
+    print(json.dumps(fs.call(run())))
+
+
 def cmd_load(fs, args):
     job = fs.submit_job(args.path, recursive=True, replicas=args.replicas)
     print(json.dumps(job))
@@ -401,6 +411,9 @@ def build_parser() -> argparse.ArgumentParser:
     nd = nodesub.add_parser("decommission")
     nd.add_argument("worker_id", type=int)
     nd.set_defaults(fn=cmd_node_decommission)
+    nt = nodesub.add_parser("transfer-leader")
+    nt.add_argument("node_id", type=int)
+    nt.set_defaults(fn=cmd_transfer_leader)
     return p
 
 

@@ -7,6 +7,7 @@ proto/master.proto & worker.proto.
 """
 from __future__ import annotations
 
+import asyncio
 import logging
 import time
 from typing import Optional
@@ -57,6 +58,8 @@ class MasterHandler:
         try:
             op_before = self.master.journal.op_id
             reply = fn(msg.header, msg.data)
+            if asyncio.iscoroutine(reply):
+                reply = await reply
             if raft is not None and raft.is_leader \
                     and self.master.journal.op_id > op_before:
                 # withhold the reply until the mutation's entries commit
@@ -317,3 +320,9 @@ class MasterHandler:
 
     def op_raftinstallsnapshot(self, h, d):
         return self.master.raft.on_install_snapshot(h, d)
+
+    def op_rafttransferleader(self, h, d):
+        if h.get("leader") is not None:   # inbound TimeoutNow from leader
+            return self.master.raft.on_transfer_leader(h)
+        # client/CLI request: we are the leader, hand off to target
+        return self.master.raft.transfer_leadership(h["target"])

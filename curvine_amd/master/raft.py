@@ -260,8 +260,8 @@ class RaftNode:
                 grants += 1
         return grants * 2 > len(self.peers) + 1
 
-    async def _run_election(self):
-        if not await self._pre_vote():
+    async def _run_election(self, skip_prevote: bool = False):
+        if not skip_prevote and not await self._pre_vote():
             self._last_heard = time.monotonic()   # back off, stay follower
             return
         self.state = CANDIDATE
@@ -452,6 +452,29 @@ class RaftNode:
             self.match_index[pid] = index
         except Exception as e:  # noqa: BLE001
             log.debug("snapshot to %d failed: %s", pid, e)
+
+    # ---------------- leadership transfer ----------------
+    async def transfer_leadership(self, target: int) -> dict:
+        """TimeoutNow-style transfer: bring the target fully up to date,
+        then tell it to elect itself immediately (no pre-vote); we step
+        down as soon as its higher term reaches us."""
+        if not self.is_leader:
+            raise err.NotLeader(f"leader={self.leader_addr or ''}")
+        if target not in self.peers:
+            raise err.InvalidArgument(f"unknown peer {target}")
+        # push any missing entries first so its log can win the election
+        await self._append_to(target)
+        c = await self._client(target)
+        r = await c.rpc(RpcCode.RaftTransferLeader,
+                        {"term": self.term, "leader": self.id}, timeout=5.0)
+        return {"target": target, "accepted": bool(r.header.get("ok"))}
+
+    def on_transfer_leader(self, h: dict) -> dict:
+        if h["term"] < self.term or self.state == LEADER:
+            return {"term": self.term, "ok": False}
+        # elect immediately: skip pre-vote (the current leader asked)
+        asyncio.ensure_future(self._run_election(skip_prevote=True))
+        return {"term": self.term, "ok": True}
 
     # ---------------- RPC handlers (called from MasterHandler) ----------------
     def on_vote(self, h: dict) -> dict:

@@ -28,6 +28,7 @@ log = logging.getLogger("curvine.master")
 _READONLY_OK = {
     RpcCode.Heartbeat, RpcCode.MetricsReport,
     RpcCode.RaftVote, RpcCode.RaftAppendEntries, RpcCode.RaftInstallSnapshot,
+    RpcCode.RaftTransferLeader,
 }
 import curvine_amd.master.handler as _handler_mod
 _handler_mod._READONLY_OK = _READONLY_OK

@@ -86,6 +86,7 @@ class RpcCode(IntEnum):
     RaftVote = 70
     RaftAppendEntries = 71
     RaftInstallSnapshot = 72
+    RaftTransferLeader = 73
 
     # block interface (worker data plane)
     WriteBlock = 80

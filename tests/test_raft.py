@@ -213,3 +213,31 @@ def test_prevote_blocks_term_inflation(loop, tmp_path):
         for m in [survivor] + restarted:
             await m.stop()
     run(loop, main())
+
+
+def test_leadership_transfer(loop, tmp_path):
+    """RaftTransferLeader: the leader hands off to a chosen follower
+    (TimeoutNow analog) and steps down when its term arrives."""
+    async def main():
+        masters, ports, peers = await start_group(tmp_path)
+        leader = await wait_leader(masters)
+        conn = ClusterConnector([f"127.0.0.1:{p}" for p in ports],
+                                timeout_ms=8000, retries=6)
+        await conn.rpc(RpcCode.Mkdir, {"path": "/xfer", "create_parents": True})
+        target = next(m for m in masters if not m.raft.is_leader)
+        tid = target.conf.journal.node_id
+        r = await conn.rpc(RpcCode.RaftTransferLeader, {"target": tid})
+        assert r.header.get("accepted") is True
+        for _ in range(100):
+            await asyncio.sleep(0.1)
+            if target.raft.is_leader:
+                break
+        assert target.raft.is_leader
+        # the cluster keeps mutating under the new leader
+        await conn.rpc(RpcCode.Mkdir, {"path": "/xfer/after",
+                                       "create_parents": True})
+        assert target.fs.fs_dir.resolve("/xfer/after") is not None
+        await conn.close()
+        for m in masters:
+            await m.stop()
+    run(loop, main())
