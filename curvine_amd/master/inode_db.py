@@ -60,8 +60,12 @@ class SqliteInodeStore:
         self._deleted.add(inode_id)
 
     # ---------------- persistence ----------------
+    MAX_BATCH = 20_000   # bound the per-tick pause on mutation storms
+
     def flush(self, fs_dir, mounts_state, op_id: int) -> int:
-        """One transaction: dirty upserts + deletes + watermarks.
+        """One transaction: dirty upserts + deletes + watermarks, capped
+        at MAX_BATCH rows (the remainder stays dirty for the next tick —
+        correctness is carried by the WAL tail + restart reconcile).
         Returns the number of rows written."""
         if not self._dirty and not self._deleted:
             cur = self.conn.execute("SELECT v FROM meta WHERE k='op_id'")
@@ -69,12 +73,17 @@ class SqliteInodeStore:
             if row is not None and int.from_bytes(row[0], "little") == op_id:
                 return 0
         rows = []
+        batch = []
         for iid in self._dirty:
+            batch.append(iid)
             node = fs_dir.inodes.get(iid)
             if node is None:
                 continue
             rows.append((iid, msgpack.packb(node.to_state(),
                                             use_bin_type=True)))
+            if len(rows) >= self.MAX_BATCH:
+                break
+        partial = len(batch) < len(self._dirty)
         dels = [(iid,) for iid in self._deleted]
         with self.conn:
             if rows:
@@ -82,17 +91,20 @@ class SqliteInodeStore:
                     "REPLACE INTO inodes (id, state) VALUES (?, ?)", rows)
             if dels:
                 self.conn.executemany("DELETE FROM inodes WHERE id=?", dels)
-            meta = [("op_id", op_id.to_bytes(8, "little")),
-                    ("next_inode_id",
+            meta = [("next_inode_id",
                      fs_dir.next_inode_id.to_bytes(8, "little")),
                     ("next_block_id",
                      fs_dir.next_block_id.to_bytes(8, "little")),
                     ("mounts", msgpack.packb(mounts_state,
                                              use_bin_type=True))]
+            if not partial:
+                # advance the restart watermark only once every dirty
+                # inode as of this op is actually on disk
+                meta.append(("op_id", op_id.to_bytes(8, "little")))
             self.conn.executemany(
                 "REPLACE INTO meta (k, v) VALUES (?, ?)", meta)
         n = len(rows) + len(dels)
-        self._dirty.clear()
+        self._dirty.difference_update(batch)
         self._deleted.clear()
         return n
 
