@@ -182,6 +182,15 @@ class FsDir:
             if node.file_type == FileType.FILE:
                 yield node
 
+    @staticmethod
+    def _stamp(node: "Inode", e: dict) -> None:
+        """Creation/mtime from the entry's log-time stamp: replay (WAL,
+        raft followers, partial-flush restarts) reproduces identical
+        timestamps instead of re-reading the clock."""
+        ts = e.get("ts")
+        if ts:
+            node.create_ms = node.mtime_ms = node.atime_ms = ts
+
     # ---------------- mutations ----------------
     def mkdir(self, path: str, mode: int = 0o755, create_parents: bool = False) -> Inode:
         path = norm_path(path)
@@ -204,6 +213,7 @@ class FsDir:
 
     def _apply_mkdir(self, e: dict) -> Inode:
         node = Inode(e["inode_id"], e["name"], FileType.DIR, e.get("mode", 0o755))
+        self._stamp(node, e)
         node.parent_id = e["parent_id"]
         self.inodes[node.id] = node
         self.inodes[e["parent_id"]].children[e["name"]] = node.id
@@ -250,6 +260,7 @@ class FsDir:
             removed = self._drop_inode(self.inodes[old_id])
         node = Inode(e["inode_id"], e["name"], e.get("file_type", int(FileType.FILE)),
                      e.get("mode", 0o644))
+        self._stamp(node, e)
         node.parent_id = parent.id
         node.block_size = e["block_size"]
         node.replicas = e["replicas"]
@@ -257,7 +268,7 @@ class FsDir:
         node.complete = False
         self.inodes[node.id] = node
         parent.children[e["name"]] = node.id
-        parent.mtime_ms = now_ms()
+        parent.mtime_ms = e.get("ts") or now_ms()
         self.next_inode_id = max(self.next_inode_id, node.id)
         if self.mirror:
             self.mirror.upsert(node)
@@ -307,7 +318,7 @@ class FsDir:
                 blk[1] = min(rem, node.block_size)
                 rem -= blk[1]
         node.complete = True
-        node.mtime_ms = now_ms()
+        node.mtime_ms = e.get("ts") or now_ms()
         if self.mirror:
             self.mirror.upsert(node)
 
@@ -334,7 +345,7 @@ class FsDir:
         name = e.get("name", node.name)
         if parent is not None and parent.children is not None:
             parent.children.pop(name, None)
-            parent.mtime_ms = now_ms()
+            parent.mtime_ms = e.get("ts") or now_ms()
             if self.mirror:
                 self.mirror.remove_child(parent.id, name)
                 self.mirror.upsert(parent)
@@ -400,11 +411,11 @@ class FsDir:
         old_name = e.get("src_name", node.name)
         if src_parent is not None and src_parent.children is not None:
             src_parent.children.pop(old_name, None)
-            src_parent.mtime_ms = now_ms()
+            src_parent.mtime_ms = e.get("ts") or now_ms()
         node.parent_id = dst_parent.id
         node.name = e["dst_name"]
         dst_parent.children[node.name] = node.id
-        dst_parent.mtime_ms = now_ms()
+        dst_parent.mtime_ms = e.get("ts") or now_ms()
         if self.mirror:
             if src_parent is not None:
                 self.mirror.remove_child(src_parent.id, old_name)
@@ -466,6 +477,7 @@ class FsDir:
 
     def _apply_symlink(self, e: dict) -> Inode:
         node = Inode(e["inode_id"], e["name"], int(FileType.SYMLINK), 0o777)
+        self._stamp(node, e)
         node.parent_id = e["parent_id"]
         node.symlink_target = e["target"]
         node.complete = True
@@ -497,7 +509,7 @@ class FsDir:
         parent = self.inodes[e["dst_parent"]]
         parent.children[e["dst_name"]] = node.id
         node.nlink += 1
-        parent.mtime_ms = now_ms()
+        parent.mtime_ms = e.get("ts") or now_ms()
         if self.mirror:
             self.mirror.upsert(node)
             self.mirror.upsert(parent)
@@ -526,7 +538,7 @@ class FsDir:
                                       new_len - off)])
             off += blen if blen else node.block_size
         node.blocks = keep
-        node.mtime_ms = now_ms()
+        node.mtime_ms = e.get("ts") or now_ms()
         if self.mirror:
             self.mirror.upsert(node)
         return removed

@@ -118,7 +118,8 @@ class JournalWriter:
         if not self.enabled:
             return {}
         self.op_id += 1
-        entry = {"op": op, "op_id": self.op_id, **fields}
+        from curvine_amd.model import now_ms
+        entry = {"op": op, "op_id": self.op_id, "ts": now_ms(), **fields}
         buf = encode_entry(entry)
         if self._f is None or self._seg_bytes + len(buf) > self.segment_max:
             self._roll(self.op_id)

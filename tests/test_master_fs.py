@@ -204,3 +204,35 @@ def test_metadata_snapshot_and_delta_pages(tmp_path):
         await c.close()
         await m.stop()
     asyncio.new_event_loop().run_until_complete(main())
+
+
+def test_replay_reproduces_timestamps(tmp_path):
+    """Journal entries carry a log-time stamp: a WAL replay yields the
+    SAME mtimes/create times as the live run (deterministic replay —
+    raft followers, sqlite partial-flush restarts and crash recovery all
+    converge bit-for-bit on time fields)."""
+    import asyncio
+    import copy
+
+    from curvine_amd.master.server import Master
+    from curvine_amd.testing import test_conf as _tc
+
+    conf = _tc(str(tmp_path))
+    conf.master.inode_db = False       # force pure WAL replay
+
+    async def main():
+        m = await Master(conf).start()
+        m.fs.mkdir("/ts/dir", 0o755, True)
+        m.fs.create("/ts/f", 0, 1, "", False)
+        m.fs.complete_file("/ts/f", 7, [7])
+        m.fs.rename("/ts/f", "/ts/g")
+        live = {p: (m.fs.file_status(p).mtime_ms)
+                for p in ("/ts", "/ts/dir", "/ts/g")}
+        await m.stop()
+        await asyncio.sleep(0.05)      # ensure the clock moved on
+        m2 = await Master(copy.deepcopy(conf)).start()
+        for p, mt in live.items():
+            assert m2.fs.file_status(p).mtime_ms == mt, p
+        await m2.stop()
+
+    asyncio.new_event_loop().run_until_complete(main())
