@@ -256,7 +256,12 @@ class FsDir:
         parent = self.inodes[e["parent_id"]]
         removed: list[int] = []
         old_id = parent.children.get(e["name"])
-        if old_id is not None:
+        if old_id is not None and old_id != e["inode_id"] \
+                and old_id in self.inodes:
+            # overwrite of a different pre-existing file.  old_id ==
+            # inode_id means idempotent replay of this very create; a
+            # dangling edge (old_id not in inodes) can come from a
+            # partially flushed store whose WAL tail we are replaying.
             removed = self._drop_inode(self.inodes[old_id])
         node = Inode(e["inode_id"], e["name"], e.get("file_type", int(FileType.FILE)),
                      e.get("mode", 0o644))
@@ -289,9 +294,15 @@ class FsDir:
 
     def _apply_add_block(self, e: dict) -> int:
         node = self.inodes[e["inode_id"]]
+        bid = e["block_id"]
+        if any(b[0] == bid for b in node.blocks):
+            # idempotent replay: WAL tail re-applied over a partially
+            # flushed store already contains this block
+            self.block_index[bid] = node.id
+            self.next_block_id = max(self.next_block_id, bid)
+            return bid
         if e.get("commit_prev_len", -1) >= 0 and node.blocks:
             node.blocks[-1][1] = e["commit_prev_len"]
-        bid = e["block_id"]
         node.blocks.append([bid, 0])
         self.block_index[bid] = node.id
         self.next_block_id = max(self.next_block_id, bid)
@@ -507,6 +518,8 @@ class FsDir:
     def _apply_link(self, e: dict) -> None:
         node = self.inodes[e["inode_id"]]
         parent = self.inodes[e["dst_parent"]]
+        if parent.children.get(e["dst_name"]) == node.id:
+            return   # idempotent replay: dentry already present
         parent.children[e["dst_name"]] = node.id
         node.nlink += 1
         parent.mtime_ms = e.get("ts") or now_ms()

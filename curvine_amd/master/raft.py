@@ -146,7 +146,7 @@ class RaftNode:
         self.log = RaftLog(os.path.join(state_dir, f"raft_{node_id}.log"))
         self.meta_path = os.path.join(state_dir, f"raft_{node_id}.meta")
         self._load_meta()
-        self.commit_index = self.log.snapshot_index
+        self.commit_index = self.boot_commit
         self.last_applied = self.log.snapshot_index
         self.apply_entry = apply_entry
         self.make_snapshot = make_snapshot
@@ -167,20 +167,34 @@ class RaftNode:
 
     # ---------------- persistence ----------------
     def _load_meta(self):
+        self._persisted_commit = 0
         if os.path.exists(self.meta_path):
             with open(self.meta_path, "rb") as f:
                 m = msgpack.unpackb(f.read(), raw=False)
             self.term = m["term"]
             self.voted_for = m["voted_for"]
+            self._persisted_commit = m.get("commit", 0)
 
-    def _save_meta(self):
+    def _save_meta(self, fsync: bool = True):
+        """Persist term/vote (fsynced — raft safety requires it) and the
+        commit watermark (advisory: a stale LOWER value only makes boot
+        replay conservatively short, so commit-only updates skip fsync)."""
         tmp = self.meta_path + ".tmp"
         with open(tmp, "wb") as f:
             f.write(msgpack.packb({"term": self.term,
-                                   "voted_for": self.voted_for}))
+                                   "voted_for": self.voted_for,
+                                   "commit": self.commit_index}))
             f.flush()
-            os.fsync(f.fileno())
+            if fsync:
+                os.fsync(f.fileno())
         os.replace(tmp, self.meta_path)
+
+    @property
+    def boot_commit(self) -> int:
+        """Safe boot-replay bound: committed watermark clamped to the log
+        (never below the snapshot the log starts at)."""
+        return max(self.log.snapshot_index,
+                   min(self._persisted_commit, self.log.last_index))
 
     # ---------------- lifecycle ----------------
     def start(self) -> "RaftNode":
@@ -343,6 +357,7 @@ class RaftNode:
                         self.id)
             self.log.truncate_from(self.commit_index + 1)
             self.rebuild()
+            self.last_applied = self.commit_index
         self._fail_waiters()
 
     def _fail_waiters(self):
@@ -390,6 +405,7 @@ class RaftNode:
     def _advance_commit(self, index: int):
         self.commit_index = index
         self.log.flush()
+        self._save_meta(fsync=False)   # advisory boot-replay watermark
         remaining = []
         for idx, fut in self._commit_waiters:
             if idx <= index:
@@ -516,16 +532,33 @@ class RaftNode:
             return {"term": self.term, "success": False,
                     "hint": min(prev, self.log.last_index + 1)}
         idx = prev
+        appended = False
+        conflict_below_applied = False
         for rec in h["entries"]:
             idx += 1
             if idx <= self.log.last_index:
                 if self.log.term_at(idx) == rec["_t"]:
                     continue
+                if idx <= self.last_applied:
+                    # conflicting entries were already applied (e.g. a
+                    # boot that optimistically replayed an uncommitted
+                    # tail): the state machine must be rebuilt from
+                    # snapshot + surviving committed log
+                    conflict_below_applied = True
                 self.log.truncate_from(idx)
             self.log.append(rec["_t"], rec["e"])
+            appended = True
+        if appended:
+            # durability before ack: the leader counts this ack toward
+            # commit, so the entries must survive a crash here
+            self.log.flush()
         if h["commit"] > self.commit_index:
             self.commit_index = min(h["commit"], self.log.last_index)
-            self._apply_committed()
+            self._save_meta(fsync=False)
+        if conflict_below_applied:
+            self.rebuild()
+            self.last_applied = min(self.commit_index, self.log.last_index)
+        self._apply_committed()
         return {"term": self.term, "success": True}
 
     def on_install_snapshot(self, h: dict, data: bytes = b"") -> dict:

@@ -82,9 +82,12 @@ class RaftJournalWriter(JournalWriter):
         if not self.enabled:
             return {}
         # index is assigned deterministically before append so the entry is
-        # serialized into the raft log with its final op_id
+        # serialized into the raft log with its final op_id; 'ts' is the
+        # log-time stamp every replica replays identically (deterministic
+        # create/mtime and TTL expiry across leader and followers)
+        from curvine_amd.model import now_ms
         index = self.raft.log.last_index + 1
-        entry = {"op": op, "op_id": index, **fields}
+        entry = {"op": op, "op_id": index, "ts": now_ms(), **fields}
         got = self.raft.append_local(entry)
         assert got == index
         self.op_id = index
@@ -192,12 +195,18 @@ class Master:
         if self._raft_mode:
             self.raft = self._make_raft()
             self.journal.raft = self.raft
-            # replay the local raft log into the state machine
+            # replay the local raft log ONLY up to the persisted commit
+            # watermark: the uncommitted tail may be truncated by a new
+            # leader, and entries applied at boot could then never be
+            # rolled back (state divergence).  The tail is applied later:
+            # by _apply_committed when a leader re-commits it, or by
+            # _become_leader's backlog apply if WE win the election.
             lg = self.raft.log
-            for i in range(lg.snapshot_index + 1, lg.last_index + 1):
+            safe = self.raft.boot_commit
+            for i in range(lg.snapshot_index + 1, safe + 1):
                 self._apply_entry(lg.entry_at(i))
-            self.raft.last_applied = lg.last_index
-            self.journal.op_id = lg.last_index
+            self.raft.last_applied = safe
+            self.journal.op_id = safe
         else:
             self._restore()
         if self.conf.master.native_meta:

@@ -93,6 +93,10 @@ class RpcClient:
                         raise ConnectionResetError("eof")
                     buf += chunk
                 hlen, dlen, msg = Message.decode_proto(bytes(buf[:PROTO_SIZE]))
+                if hlen > MAX_DATA_SIZE or dlen > MAX_DATA_SIZE:
+                    raise ValueError(
+                        f"frame hlen={hlen} dlen={dlen} exceeds "
+                        f"{MAX_DATA_SIZE}")
                 total = PROTO_SIZE + hlen + dlen
                 while len(buf) < total:
                     chunk = await self._reader.read(256 << 10)

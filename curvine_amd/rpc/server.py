@@ -58,6 +58,11 @@ class ServerConn:
                 if dlen > MAX_DATA_SIZE:
                     raise ValueError(
                         f"frame data_len {dlen} exceeds {MAX_DATA_SIZE}")
+                if hlen > MAX_DATA_SIZE:
+                    # corrupt/malicious frame: never buffer an unbounded
+                    # header
+                    raise ValueError(
+                        f"frame header_len {hlen} exceeds {MAX_DATA_SIZE}")
                 total = PROTO_SIZE + hlen + dlen
                 if len(buf) >= total:
                     if hlen:

@@ -21,8 +21,9 @@ pytestmark = requires_fuse
 from tests.test_fuse import mount  # noqa: E402,F401 — reuse the fixture
 
 
-def test_read_past_eof(mount):
-    """fuse_read_past_eof_test.rs analog."""
+def test_read_past_eof_small(mount):
+    """fuse_read_past_eof_test.rs analog (small file; the 10 KB variant
+    below covers clamped tail reads)."""
     mnt, *_ = mount
     p = f"{mnt}/eof.bin"
     with open(p, "wb") as f:

@@ -171,3 +171,58 @@ def test_partial_flush_watermark(tmp_path, monkeypatch):
         await m.stop()
 
     _run(phase2())
+
+
+def test_partial_flush_wal_replay_idempotent(tmp_path, monkeypatch):
+    """after a capped flush the sqlite rows are NEWER
+    than the op_id watermark; restart replays the WAL tail over them.
+    add_block and link replay must be idempotent (no duplicate block
+    entries, no double nlink)."""
+    from curvine_amd.master.inode_db import SqliteInodeStore
+    from curvine_amd.master.server import Master
+    from curvine_amd.testing import test_conf
+
+    monkeypatch.setattr(SqliteInodeStore, "MAX_BATCH", 2)
+    conf = test_conf(str(tmp_path))
+
+    async def phase1():
+        m = await Master(conf).start()
+        m.fs.mkdir("/pfi", create_parents=True)
+        st = m.fs.create("/pfi/a", 0, 1, "", False)
+        node = m.fs.fs_dir.inodes[st.inode_id]
+        for _ in range(3):
+            m.fs.fs_dir.add_block(node, commit_prev_len=64)
+        m.fs.fs_dir.complete_file(node, 192, [64, 64, 64])
+        m.fs.fs_dir.link("/pfi/a", "/pfi/hard")
+        # flush everything, then do ONE more mutation batch and a single
+        # capped flush: rows newer than the watermark now exist
+        while m.inode_db.flush(m.fs.fs_dir, m.mounts.to_snapshot(),
+                               m.journal.op_id) or m.inode_db._dirty:
+            pass
+        st2 = m.fs.create("/pfi/b", 0, 1, "", False)
+        node2 = m.fs.fs_dir.inodes[st2.inode_id]
+        m.fs.fs_dir.add_block(node2)
+        m.fs.fs_dir.link("/pfi/a", "/pfi/hard2")
+        m.inode_db.flush(m.fs.fs_dir, m.mounts.to_snapshot(),
+                         m.journal.op_id)   # capped: watermark stays back
+        # crash WITHOUT the clean stop-flush (stop() would drain)
+        m.journal.close()
+        m.inode_db.close()
+        await m.rpc.stop()
+        if m._actor_task:
+            m._actor_task.cancel()
+
+    _run(phase1())
+
+    async def phase2():
+        m = await Master(conf).start()
+        a = m.fs.fs_dir.must_resolve("/pfi/a")
+        assert [b[1] for b in a.blocks] == [64, 64, 64], \
+            f"duplicate/incorrect blocks after replay: {a.blocks}"
+        assert len({b[0] for b in a.blocks}) == 3
+        assert a.nlink == 3, f"nlink={a.nlink} (link replay not idempotent)"
+        b = m.fs.fs_dir.must_resolve("/pfi/b")
+        assert len(b.blocks) == 1
+        await m.stop()
+
+    _run(phase2())

@@ -241,3 +241,55 @@ def test_leadership_transfer(loop, tmp_path):
         for m in masters:
             await m.stop()
     run(loop, main())
+
+
+def test_crashed_leader_uncommitted_tail_converges(loop, tmp_path):
+    """a leader that applied an uncommitted tail
+    optimistically, crashed, and restarts as a follower must NOT keep the
+    phantom state.  Boot replays only up to the persisted commit
+    watermark; the new leader's conflicting entries truncate the tail and
+    the replica converges."""
+    async def main():
+        masters, ports, peers = await start_group(tmp_path)
+        leader = await wait_leader(masters)
+        confs = {m.conf.journal.node_id: copy.deepcopy(m.conf)
+                 for m in masters}
+        lid = leader.conf.journal.node_id
+        # isolate the leader: stop both followers so nothing commits
+        followers = [m for m in masters if m is not leader]
+        for f in followers:
+            await f.stop()
+        # leader applies locally at append time; never reaches a majority
+        leader.fs.fs_dir.mkdir("/phantom")
+        assert leader.fs.fs_dir.resolve("/phantom") is not None
+        await leader.stop()
+        # restart the two followers; they elect a leader without /phantom
+        live = []
+        for f in followers:
+            c = confs[f.conf.journal.node_id]
+            c.master.rpc_port = ports[c.journal.node_id - 1]
+            live.append(await Master(c).start())
+        new_leader = await wait_leader(live)
+        conn = ClusterConnector(
+            [f"127.0.0.1:{ports[m.conf.journal.node_id - 1]}" for m in live],
+            timeout_ms=8000, retries=6)
+        for i in range(5):
+            await conn.rpc(RpcCode.Mkdir, {"path": f"/real/{i}",
+                                           "create_parents": True})
+        # restart the crashed leader: boot must NOT apply the phantom tail
+        lc = confs[lid]
+        lc.master.rpc_port = ports[lid - 1]
+        restarted = await Master(lc).start()
+        assert restarted.fs.fs_dir.resolve("/phantom") is None, \
+            "boot replayed an uncommitted tail"
+        for _ in range(100):
+            await asyncio.sleep(0.1)
+            if restarted.fs.fs_dir.resolve("/real/4") is not None:
+                break
+        assert restarted.fs.fs_dir.resolve("/real/4") is not None
+        assert restarted.fs.fs_dir.resolve("/phantom") is None, \
+            "conflicting-entry truncation left phantom state applied"
+        await conn.close()
+        for m in live + [restarted]:
+            await m.stop()
+    run(loop, main())
