@@ -45,6 +45,11 @@ class MasterHandler:
         raft = self.master.raft
         if raft is not None and not raft.is_leader and code not in _READONLY_OK:
             raise err.NotLeader(f"leader={raft.leader_addr or ''}")
+        if raft is not None and raft.is_leader and not raft.read_ready \
+                and code not in _READONLY_OK:
+            # ReadIndex rule: a fresh leader may lag entries committed by
+            # the old leader; serve nothing until our no-op commits
+            await raft.wait_commit(raft.term_start_index)
         # mutation retry cache (fs_retry_cache.rs analog): a replayed
         # request (connector retry after timeout/failover) returns the
         # original reply instead of re-executing (double add_block etc.)

@@ -84,8 +84,12 @@ class MetaMirror:
 
     def upsert(self, node) -> None:
         blob, n = _node_blob(node)
-        self.lib.meta_upsert(self.sid, node.id, node.is_dir, blob, n,
-                             _pack_blocks(node))
+        self._raw_upsert(node.id, node.is_dir, blob, n, _pack_blocks(node))
+
+    def _raw_upsert(self, inode_id: int, is_dir: bool, blob: bytes,
+                    n: int, blocks: bytes) -> None:
+        """Pre-serialized upsert (CommitGatedMirror replays these)."""
+        self.lib.meta_upsert(self.sid, inode_id, is_dir, blob, n, blocks)
 
     def add_child(self, parent_id: int, name: str, child_id: int) -> None:
         self.lib.meta_add_child(self.sid, parent_id, name, child_id)
@@ -95,6 +99,63 @@ class MetaMirror:
 
     def drop(self, inode_id: int) -> None:
         self.lib.meta_drop(self.sid, inode_id)
+
+
+class CommitGatedMirror:
+    """Linearizable native reads under raft: the leader applies entries
+    to FsDir at APPEND time (optimistic), but the C++ tree must only see
+    COMMITTED state — a concurrent native FileStatus would otherwise
+    observe data a step-down rebuild later erases.
+
+    Each mirror op is serialized eagerly (capturing the node's state at
+    apply time, before any later mutation touches it), tagged with the
+    entry's op_id (== raft index), and replayed into the C++ tree when
+    the commit index passes it.  Follower applies are always committed
+    already, so they pass straight through."""
+
+    def __init__(self, inner: MetaMirror, current_op, current_commit):
+        self.inner = inner
+        self._op = current_op            # () -> op_id being applied
+        self._commit = current_commit    # () -> raft commit index
+        self.pending: list[tuple[int, tuple]] = []   # (op_id, call)
+
+    def _gate(self, call: tuple) -> None:
+        op = self._op()
+        if op <= self._commit():
+            self._run(call)
+        else:
+            self.pending.append((op, call))
+
+    def _run(self, call: tuple) -> None:
+        fn, *args = call
+        getattr(self.inner, fn)(*args)
+
+    def flush(self, commit: int) -> None:
+        if not self.pending:
+            return
+        i = 0
+        for i, (op, call) in enumerate(self.pending):  # noqa: B007
+            if op > commit:
+                break
+            self._run(call)
+        else:
+            i += 1
+        del self.pending[:i]
+
+    # MetaMirror surface — serialize NOW, apply at commit
+    def upsert(self, node) -> None:
+        blob, n = _node_blob(node)
+        self._gate(("_raw_upsert", node.id, node.is_dir, blob, n,
+                    _pack_blocks(node)))
+
+    def add_child(self, parent_id: int, name: str, child_id: int) -> None:
+        self._gate(("add_child", parent_id, name, child_id))
+
+    def remove_child(self, parent_id: int, name: str) -> None:
+        self._gate(("remove_child", parent_id, name))
+
+    def drop(self, inode_id: int) -> None:
+        self._gate(("drop", inode_id))
 
 
 class WorkerMirror:
@@ -175,6 +236,7 @@ class NativeMetaFrontend:
         sock.listen(1024)
         self.port = sock.getsockname()[1]
         self.sid = self.lib.meta_create(sock.detach(), nthreads)
+        self.mirror_gate: Optional[CommitGatedMirror] = None
         self.fwd_batch = fwd_batch
         self._loop: Optional[asyncio.AbstractEventLoop] = None
         self._efd: Optional[int] = None
@@ -196,12 +258,22 @@ class NativeMetaFrontend:
         from curvine_amd.master.fs_dir import MirrorFanout
         fs_dir = self.master.fs.fs_dir
         mirror = MetaMirror(self.lib, self.sid)
+        raft = self.master.raft
+        if raft is not None:
+            # attach() runs from a clean (all-committed) state: rebuild,
+            # snapshot install, and start all discard any optimistic tail
+            # first, so an empty pending queue is correct here
+            mirror = self.mirror_gate = CommitGatedMirror(
+                mirror, lambda: self.master.journal.op_id,
+                lambda: raft.commit_index)
+            raft.on_commit = self.mirror_gate.flush
         # preserve co-observers (e.g. the sqlite inode store)
         cur = fs_dir.mirror
         others = []
+        _native = (MetaMirror, CommitGatedMirror)
         if isinstance(cur, MirrorFanout):
-            others = [m for m in cur.mirrors if not isinstance(m, MetaMirror)]
-        elif cur is not None and not isinstance(cur, MetaMirror):
+            others = [m for m in cur.mirrors if not isinstance(m, _native)]
+        elif cur is not None and not isinstance(cur, _native):
             others = [cur]
         fs_dir.mirror = MirrorFanout([mirror] + others) if others else mirror
         self.lib.meta_clear(self.sid)

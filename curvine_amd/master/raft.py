@@ -154,6 +154,12 @@ class RaftNode:
         self.rebuild = rebuild
         # optional leadership observer (native meta frontend serving flag)
         self.on_role_change = None
+        # optional commit observer (commit-gated native mirror flush)
+        self.on_commit: Optional[Callable[[int], None]] = None
+        # ReadIndex rule state: a new leader serves reads only after its
+        # no-op entry commits (see _become_leader / read_ready)
+        self.term_start_index = 0
+        self._announced_leader = False
         self.election_timeout = election_timeout_ms / 1000.0
         self.heartbeat = heartbeat_ms / 1000.0
         self.next_index: dict[int, int] = {}
@@ -217,6 +223,13 @@ class RaftNode:
     @property
     def is_leader(self) -> bool:
         return self.state == LEADER
+
+    @property
+    def read_ready(self) -> bool:
+        """Linearizable-read gate: leader whose current-term no-op has
+        committed (raft ReadIndex rule)."""
+        return self.state == LEADER and \
+            self.commit_index >= self.term_start_index
 
     @property
     def leader_addr(self) -> str:
@@ -331,12 +344,17 @@ class RaftNode:
             except Exception as e:  # noqa: BLE001
                 log.exception("backlog apply %d: %s", self.last_applied, e)
         # no-op entry in the new term: commits the previous-term backlog
-        # (raft's no-commit-of-old-terms rule needs a current-term entry)
-        self.append_local({"op": "noop", "op_id": self.log.last_index + 1})
-        if self.on_role_change is not None:
-            self.on_role_change(True)
+        # (raft's no-commit-of-old-terms rule needs a current-term entry).
+        # Serving READS must wait until it commits (ReadIndex rule): the
+        # on_role_change(True) announcement is deferred to that commit —
+        # see _advance_commit.
+        self.term_start_index = self.append_local(
+            {"op": "noop", "op_id": self.log.last_index + 1})
+        self._announced_leader = False
         log.info("node %d is LEADER (term %d, last_index %d)",
                  self.id, self.term, self.log.last_index)
+        if not self.peers:
+            self._advance_commit(self.log.last_index)
 
     def _become_follower(self, term: int):
         was_leader = self.state == LEADER
@@ -406,6 +424,13 @@ class RaftNode:
         self.commit_index = index
         self.log.flush()
         self._save_meta(fsync=False)   # advisory boot-replay watermark
+        if self.on_commit is not None:
+            self.on_commit(index)
+        if not self._announced_leader and self.state == LEADER \
+                and index >= self.term_start_index:
+            self._announced_leader = True
+            if self.on_role_change is not None:
+                self.on_role_change(True)
         remaining = []
         for idx, fut in self._commit_waiters:
             if idx <= index:

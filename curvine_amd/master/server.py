@@ -153,10 +153,12 @@ class Master:
             heartbeat_ms=self.conf.journal.heartbeat_interval_ms)
 
     def _apply_entry(self, e: dict) -> None:
-        """Raft follower apply path (leader applied at append time)."""
+        """Raft follower apply path (leader applied at append time).
+        op_id advances BEFORE the apply so the commit-gated native mirror
+        tags every mirror op with the entry actually being applied."""
+        self.journal.op_id = max(self.journal.op_id, e["op_id"])
         if not self.mounts.apply_entry(e):
             self.fs.fs_dir.apply_entry(e)
-        self.journal.op_id = max(self.journal.op_id, e["op_id"])
 
     def _snapshot_state(self) -> dict:
         state = self.fs.fs_dir.to_snapshot()

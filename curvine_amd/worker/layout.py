@@ -212,7 +212,9 @@ class FileLayout(BlockLayout):
         self.root = data_dir.path
         os.makedirs(os.path.join(self.root, self.WRITING), exist_ok=True)
         os.makedirs(os.path.join(self.root, self.FINAL), exist_ok=True)
+        import threading
         self._used = 0
+        self._used_lock = threading.Lock()
         self._fds: dict[int, object] = {}
 
     def _path(self, block_id: int, state: str) -> str:
@@ -225,9 +227,23 @@ class FileLayout(BlockLayout):
         return self._used
 
     def allocate(self, block_id: int, reserve: int) -> dict:
-        path = self._path(block_id, self.WRITING)
-        f = open(path, "wb")
-        self._used += reserve
+        # reserve against capacity FIRST (dir_state.rs:20 analog): a full
+        # SSD/HDD dir must raise so BlockStore.create_writer's tier
+        # fall-through can move to the next dir instead of overcommitting
+        # the disk
+        with self._used_lock:
+            if self._used + reserve > self.capacity:
+                raise err.CapacityExceeded(
+                    f"dir {self.root}: used {self._used} + reserve "
+                    f"{reserve} > capacity {self.capacity}")
+            self._used += reserve
+        try:
+            path = self._path(block_id, self.WRITING)
+            f = open(path, "wb")
+        except OSError:
+            with self._used_lock:
+                self._used -= reserve
+            raise
         return {"kind": "file", "tier": self.tier, "dir_id": self.dir_id,
                 "path": path, "block_id": block_id, "reserved": reserve,
                 "length": 0, "_f": f}
@@ -242,7 +258,8 @@ class FileLayout(BlockLayout):
         os.replace(meta["path"], final)
         meta["path"] = final
         meta["length"] = length
-        self._used -= meta["reserved"] - length
+        with self._used_lock:
+            self._used -= meta["reserved"] - length
         meta["reserved"] = length
 
     def deallocate(self, meta: dict) -> None:
@@ -256,7 +273,8 @@ class FileLayout(BlockLayout):
             os.remove(meta["path"])
         except FileNotFoundError:
             pass
-        self._used -= meta["reserved"]
+        with self._used_lock:
+            self._used -= meta["reserved"]
 
     def scan(self) -> list[dict]:
         out = []

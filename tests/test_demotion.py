@@ -62,3 +62,34 @@ def test_demote_skips_active_readers(tmp_path):
         assert moved == 1
     finally:
         store.close()
+
+
+def test_file_tier_capacity_fallthrough(tmp_path):
+    """a full SSD dir raises CapacityExceeded from
+    FileLayout.allocate so create_writer falls through to the next tier
+    instead of overcommitting the disk."""
+    from curvine_amd import errors as err
+
+    conf = WorkerConf(data_dirs=[f"[SSD:8MB]{tmp_path}/ssd",
+                                 f"[HDD:64MB]{tmp_path}/hdd"])
+    store = BlockStore(conf)
+    try:
+        w1 = store.create_writer(1, 6 << 20, "SSD")
+        assert w1.layout.tier == "SSD"
+        # SSD has 2 MB left: a 4 MB reservation must land on HDD
+        w2 = store.create_writer(2, 4 << 20, "SSD")
+        assert w2.layout.tier == "HDD"
+        # every dir full -> CapacityExceeded surfaces to the caller
+        with pytest.raises(err.CapacityExceeded):
+            store.create_writer(3, 256 << 20, "SSD")
+        # abort w1: reservation released, SSD takes new writes again
+        store.abort(1)
+        w4 = store.create_writer(4, 6 << 20, "SSD")
+        assert w4.layout.tier == "SSD"
+        ssd = next(l for l in store.layouts if l.tier == "SSD")
+        # finalize shrinks the reservation to the true length
+        w4.write(b"x" * (1 << 20))
+        store.finalize(4, 1 << 20)
+        assert ssd.used == 1 << 20
+    finally:
+        store.close()

@@ -293,3 +293,64 @@ def test_crashed_leader_uncommitted_tail_converges(loop, tmp_path):
         for m in live + [restarted]:
             await m.stop()
     run(loop, main())
+
+
+def test_native_reads_commit_gated(loop, tmp_path):
+    """the C++ meta mirror must only serve COMMITTED
+    state.  With followers down, a leader-side optimistic apply must not
+    be visible through the native frontend; once a majority returns and
+    the entry commits, it must be."""
+    async def main():
+        masters, ports, peers = await start_group(tmp_path)
+        leader = await wait_leader(masters)
+        nm = leader.native_meta
+        if nm is None:
+            pytest.skip("native meta frontend unavailable")
+        conn = ClusterConnector([f"127.0.0.1:{p}" for p in ports],
+                                timeout_ms=8000, retries=6)
+        await conn.rpc(RpcCode.Mkdir, {"path": "/gated",
+                                       "create_parents": True})
+        # committed state is natively visible
+        r = await conn.rpc(RpcCode.FileStatus, {"path": "/gated"})
+        assert r.header.get("status", r.header).get("file_type") is not None \
+            or r.header  # served
+        confs = {m.conf.journal.node_id: copy.deepcopy(m.conf)
+                 for m in masters}
+        followers = [m for m in masters if m is not leader]
+        for f in followers:
+            await f.stop()
+        # leader applies optimistically; commit can't advance
+        leader.fs.fs_dir.mkdir("/gated/uncommitted")
+        assert leader.fs.fs_dir.resolve("/gated/uncommitted") is not None
+        assert nm.mirror_gate is not None and nm.mirror_gate.pending, \
+            "mirror op was not gated"
+        # the native tree must NOT have it: ask through the wire on a
+        # fresh connection straight at the leader
+        from curvine_amd.rpc.client import RpcClient
+        c = await RpcClient("127.0.0.1", leader.rpc.port,
+                            timeout_ms=3000).connect()
+        r = await c.rpc(RpcCode.Exists, {"path": "/gated/uncommitted"})
+        assert r.header.get("exists") is False, \
+            "native frontend served uncommitted state"
+        await c.close()
+        # bring a follower back: majority -> commit -> flushed to mirror
+        fc = confs[followers[0].conf.journal.node_id]
+        fc.master.rpc_port = ports[fc.journal.node_id - 1]
+        back = await Master(fc).start()
+        for _ in range(100):
+            await asyncio.sleep(0.1)
+            if leader.raft.commit_index >= leader.raft.log.last_index \
+                    and not nm.mirror_gate.pending:
+                break
+        c = await RpcClient("127.0.0.1", leader.rpc.port,
+                            timeout_ms=3000).connect()
+        r = await c.rpc(RpcCode.Exists, {"path": "/gated/uncommitted"})
+        assert r.header.get("exists") is True
+        await c.close()
+        await conn.close()
+        for m in [leader, back, followers[1]]:
+            try:
+                await m.stop()
+            except Exception:  # noqa: BLE001 — follower[1] already stopped
+                pass
+    run(loop, main())
