@@ -1114,6 +1114,7 @@ static void device_sync(int device) {
 
 // native metadata RPC frontend (pure epoll/C++, no GPU involvement)
 #include "meta_server.cpp"
+#include "data_server.cpp"
 
 static py::dict device_mem_info(int device) {
   size_t free_b = 0, total_b = 0;
@@ -1183,6 +1184,19 @@ PYBIND11_MODULE(_native, m) {
   m.def("meta_block_remove_loc", &meta_block_remove_loc);
   m.def("meta_block_drop", &meta_block_drop);
   m.def("meta_take_access", &meta_take_access);
+  m.def("data_create", &data_create);
+  m.def("data_stop", &data_stop_srv);
+  m.def("data_block_publish", &data_block_publish);
+  m.def("data_block_drop", &data_block_drop);
+  m.def("data_block_refs", &data_block_refs);
+  m.def("data_write_register", &data_write_register);
+  m.def("data_write_unregister", &data_write_unregister);
+  m.def("data_forward_pop", &data_forward_pop);
+  m.def("data_eventfd", &data_eventfd);
+  m.def("data_send", &data_send);
+  m.def("data_stats", &data_stats);
+  m.def("data_read_into", &data_read_into);
+  m.def("data_write_from", &data_write_from);
   m.def("lz4_compress", &lz4_compress_py);
   m.def("lz4_decompress", &lz4_decompress_py);
   m.def("arena_lz4_decompress", &arena_lz4_decompress);

@@ -176,8 +176,43 @@ class BlockReaderLocal:
         self.reader.close()
 
 
+def _native_data_lib():
+    """The C++ data-plane client (csrc/data_server.cpp data_read_into /
+    data_write_from), or None when the extension lacks it."""
+    global _NATIVE_DATA
+    if _NATIVE_DATA is not False:
+        return _NATIVE_DATA
+    try:
+        from curvine_amd import native
+        lib = native.load()
+        _NATIVE_DATA = lib if hasattr(lib, "data_read_into") else None
+    except Exception:  # noqa: BLE001 — extension unavailable
+        _NATIVE_DATA = None
+    return _NATIVE_DATA
+
+
+_NATIVE_DATA: object = False
+
+
+def _raise_wire_error(hdr_bytes: bytes) -> None:
+    import msgpack
+
+    from curvine_amd.errors import FsError
+    try:
+        h = msgpack.unpackb(hdr_bytes, raw=False) if hdr_bytes else {}
+    except Exception:  # noqa: BLE001
+        h = {}
+    raise FsError.decode(h.get("error_code", 1),
+                         h.get("error_msg", "data stream error"))
+
+
 class BlockReaderRemote:
-    """Streaming ReadBlock: the server pushes chunks; we buffer in-order."""
+    """Streaming ReadBlock: the server pushes chunks; we buffer in-order.
+
+    Bulk reads take the native C++ client path (blocking socket loop with
+    the GIL released, chunks recv'd straight into the destination buffer);
+    the asyncio iterator remains for chunk-streaming consumers and as the
+    fallback."""
 
     def __init__(self, addr: WorkerAddress, block_id: int):
         self.addr = addr
@@ -202,6 +237,21 @@ class BlockReaderRemote:
             stream.close()
 
     async def read_into(self, off: int, out, out_off: int, n: int) -> int:
+        lib = _native_data_lib()
+        if lib is not None:
+            try:
+                loop = asyncio.get_event_loop()
+                status, hdr, got = await loop.run_in_executor(
+                    None, lib.data_read_into, self.addr.hostname,
+                    self.addr.rpc_port, self.block_id, off, n, out, out_off,
+                    4 << 20)
+                if status == int(Status.Error):
+                    _raise_wire_error(hdr)
+                if status == int(Status.Complete):
+                    return got
+                # torn stream: fall through to the asyncio path
+            except (RuntimeError, TypeError):
+                pass   # connect failed / non-buffer dst: asyncio fallback
         got = 0
         async for chunk in self.read_range(off, n):
             out[out_off + got:out_off + got + len(chunk)] = chunk
@@ -209,6 +259,11 @@ class BlockReaderRemote:
         return got
 
     async def read(self, off: int, n: int) -> bytes:
+        lib = _native_data_lib()
+        if lib is not None:
+            buf = bytearray(n)
+            got = await self.read_into(off, buf, 0, n)
+            return bytes(memoryview(buf)[:got])
         parts = []
         async for chunk in self.read_range(off, n):
             parts.append(chunk)

@@ -147,6 +147,10 @@ class WorkerConf:
     staging_buf_bytes: int = 4 << 20
     staging_buf_count: int = 8
     replication_concurrency: int = 4
+    # native C++ epoll data plane (csrc/data_server.cpp); falls back to
+    # the asyncio RpcServer when the extension is unavailable
+    native_data: bool = True
+    data_threads: int = 12
 
     def parsed_dirs(self) -> list[DataDir]:
         return [DataDir.parse(s) for s in self.data_dirs]

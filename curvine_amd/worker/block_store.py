@@ -54,6 +54,11 @@ class BlockStore:
         # incremental heartbeat deltas
         self._added: list[dict] = []
         self._removed: list[int] = []
+        # native data plane observer (worker/native_data.py): finalized
+        # blocks are published to the C++ read registry; deletes with
+        # in-flight native readers defer the layout free (reap_deferred)
+        self.data_plane = None
+        self._deferred_frees: list[tuple[int, object, dict]] = []
         self.scan()
 
     # ---------------- startup ----------------
@@ -139,6 +144,8 @@ class BlockStore:
             b.state = BlockState.FINALIZED
             b.generation += 1
             self._added.append({"block_id": block_id, "tier": b.layout.tier})
+        if self.data_plane is not None:
+            self.data_plane.publish(block_id, b.layout, b.meta)
         return b.layout.tier
 
     def block_crc(self, block_id: int):
@@ -202,10 +209,36 @@ class BlockStore:
         # caller holds self.lock
         self.blocks.pop(b.block_id, None)
         self._removed.append(b.block_id)
+        if self.data_plane is not None and \
+                self.data_plane.drop(b.block_id) > 0:
+            # native readers still streaming this extent: defer the free
+            # until they finish (reap_deferred polls refs)
+            self._deferred_frees.append((b.block_id, b.layout, b.meta))
+            return
         try:
             b.layout.deallocate(b.meta)
         except Exception as e:  # noqa: BLE001
             log.warning("deallocate block %d: %s", b.block_id, e)
+
+    def reap_deferred(self) -> int:
+        """Free deferred deletes whose native readers have drained
+        (called from the worker heartbeat loop)."""
+        if self.data_plane is None or not self._deferred_frees:
+            return 0
+        freed = 0
+        with self.lock:
+            pending, self._deferred_frees = self._deferred_frees, []
+        for bid, layout, meta in pending:
+            if self.data_plane.refs(bid) > 0:
+                with self.lock:
+                    self._deferred_frees.append((bid, layout, meta))
+                continue
+            try:
+                layout.deallocate(meta)
+                freed += 1
+            except Exception as e:  # noqa: BLE001
+                log.warning("deferred deallocate block %d: %s", bid, e)
+        return freed
 
     # ---------------- reporting ----------------
     def take_deltas(self) -> tuple[list[dict], list[int]]:

@@ -55,6 +55,14 @@ class Worker:
 
     # ---------------- lifecycle ----------------
     async def start(self) -> "Worker":
+        if self.conf.worker.native_data:
+            try:
+                from curvine_amd.worker.native_data import NativeDataFrontend
+                self.rpc = NativeDataFrontend(
+                    self, nthreads=self.conf.worker.data_threads)
+            except Exception as e:  # noqa: BLE001 — asyncio fallback
+                log.warning("native data frontend unavailable (%s); "
+                            "using asyncio rpc server", e)
         await self.rpc.start()
         self.conf.worker.rpc_port = self.rpc.port
         registry.register(self.worker_id, self.store)
@@ -109,6 +117,7 @@ class Worker:
                 await asyncio.sleep(interval)
                 await self._heartbeat_once()
                 tick += 1
+                self.store.reap_deferred()
                 if tick % 5 == 0:   # tier-pressure demotion sweep
                     loop = asyncio.get_event_loop()
                     await loop.run_in_executor(None,
