@@ -265,6 +265,58 @@ static void data_forward(DataServer* S, const std::shared_ptr<DataConn>& c,
   }
 }
 
+// ------------------------------------------------------- pinned bounce pool
+
+// hipHostMalloc costs ~1 ms: a per-device free-list of double-buffer
+// bounce pairs makes HBM stream opens O(microseconds) after warmup
+struct BouncePair {
+  void* pin[2] = {nullptr, nullptr};
+  hipEvent_t ev[2] = {};
+  int device = -1;
+};
+
+static constexpr size_t kBounceSz = 8 << 20;
+
+struct BouncePool {
+  std::mutex mu;
+  std::unordered_map<int, std::vector<BouncePair*>> free_by_dev;
+};
+static BouncePool g_bounce;
+
+static BouncePair* bounce_acquire(int device) {
+  {
+    std::lock_guard<std::mutex> g(g_bounce.mu);
+    auto& v = g_bounce.free_by_dev[device];
+    if (!v.empty()) {
+      BouncePair* b = v.back();
+      v.pop_back();
+      return b;
+    }
+  }
+  auto* b = new BouncePair();
+  b->device = device;
+  HIP_CHECK(hipSetDevice(device));
+  HIP_CHECK(hipHostMalloc(&b->pin[0], kBounceSz, hipHostMallocDefault));
+  HIP_CHECK(hipHostMalloc(&b->pin[1], kBounceSz, hipHostMallocDefault));
+  HIP_CHECK(hipEventCreateWithFlags(&b->ev[0], hipEventDisableTiming));
+  HIP_CHECK(hipEventCreateWithFlags(&b->ev[1], hipEventDisableTiming));
+  return b;
+}
+
+static void bounce_release(BouncePair* b) {
+  std::lock_guard<std::mutex> g(g_bounce.mu);
+  auto& v = g_bounce.free_by_dev[b->device];
+  if (v.size() >= 32) {
+    hipEventDestroy(b->ev[0]);
+    hipEventDestroy(b->ev[1]);
+    hipHostFree(b->pin[0]);
+    hipHostFree(b->pin[1]);
+    delete b;
+    return;
+  }
+  v.push_back(b);
+}
+
 // ---------------------------------------------------------------- read serve
 
 // false -> forward the frame to Python (unknown block / torn state)
@@ -331,16 +383,14 @@ static bool data_serve_read(DataServer* S, DataConn* c, const uint8_t* frame,
         pos += cn;
       }
     } else {
-      // HBM: double-buffered D2H into pinned bounce, overlapping sends
+      // HBM: double-buffered D2H into pooled pinned bounce buffers,
+      // overlapping the socket sends
       HIP_CHECK(hipSetDevice(a->device));
       hipStream_t s = thread_stream(a->device);
-      size_t bsz = std::min<int64_t>(chunk, 8 << 20);
-      void* pin[2] = {nullptr, nullptr};
-      hipEvent_t ev[2];
-      HIP_CHECK(hipHostMalloc(&pin[0], bsz, hipHostMallocDefault));
-      HIP_CHECK(hipHostMalloc(&pin[1], bsz, hipHostMallocDefault));
-      HIP_CHECK(hipEventCreateWithFlags(&ev[0], hipEventDisableTiming));
-      HIP_CHECK(hipEventCreateWithFlags(&ev[1], hipEventDisableTiming));
+      size_t bsz = std::min<int64_t>(chunk, kBounceSz);
+      BouncePair* bp = bounce_acquire(a->device);
+      void** pin = bp->pin;
+      hipEvent_t* ev = bp->ev;
       const uint8_t* src = (const uint8_t*)a->base + b->aoff + offset;
       int64_t nchunks = (n + bsz - 1) / int64_t(bsz);
       for (int64_t k = 0; k < nchunks && ok; k++) {
@@ -375,10 +425,7 @@ static bool data_serve_read(DataServer* S, DataConn* c, const uint8_t* frame,
         ok = frame_send_locked(c->fd, ph, (const uint8_t*)pin[last],
                                size_t(pn));
       }
-      hipEventDestroy(ev[0]);
-      hipEventDestroy(ev[1]);
-      hipHostFree(pin[0]);
-      hipHostFree(pin[1]);
+      bounce_release(bp);
     }
   } else {
     int fd = open(b->path.c_str(), O_RDONLY);
