@@ -159,7 +159,9 @@ class Worker:
 
     # ---------------- replication (worker side) ----------------
     async def _replicate(self, cmd: dict) -> None:
-        from curvine_amd.client.block_client import BlockWriterRemote
+        from curvine_amd.client.block_client import (BlockWriterRemote,
+                                                     _native_data_lib,
+                                                     _raise_wire_error)
         async with self._repl_sem:
             bid = cmd["block_id"]
             ok, error = True, ""
@@ -170,7 +172,23 @@ class Worker:
                 try:
                     targets = [WorkerAddress.from_dict(t)
                                for t in cmd.get("targets", [])]
+                    lib = _native_data_lib()
                     for t in targets:
+                        if lib is not None:
+                            # native push: one GIL-released streaming
+                            # write from a host staging buffer
+                            buf = bytearray(reader.length)
+                            await loop.run_in_executor(
+                                None, reader.read_into, 0, buf, 0,
+                                reader.length)
+                            status, hdr = await loop.run_in_executor(
+                                None, lib.data_write_from, t.hostname,
+                                t.rpc_port, bid, reader.length,
+                                cmd.get("tier", ""), buf, 0, reader.length,
+                                8 << 20, 4, False, reader.length)
+                            if status == 5:
+                                _raise_wire_error(hdr)
+                            continue
                         w = BlockWriterRemote(t, bid, reader.length,
                                               cmd.get("tier", ""))
                         pos = 0
