@@ -265,7 +265,8 @@ static bool mp_find_path(const uint8_t* p, const uint8_t* end,
 struct MetaNode {
   bool is_dir = false;
   uint32_t npairs = 0;
-  std::string blob;                      // msgpack pairs sans "path"
+  int64_t mtime = 0;   // kept OUT of blob: parent-mtime touches are O(1)
+  std::string blob;                      // msgpack pairs sans path/mtime
   std::map<std::string, int64_t> children;  // sorted == Python sorted()
   std::vector<std::pair<int64_t, int64_t>> blocks;  // (block_id, length)
 };
@@ -449,9 +450,11 @@ static int meta_resolve(MetaServer* S, const std::string& path,
 static void meta_append_status(std::string& o, const MetaNode& nd,
                                const std::string& path) {
   o.push_back(char(0xde));
-  wr_u16be(o, uint16_t(nd.npairs + 1));
+  wr_u16be(o, uint16_t(nd.npairs + 2));
   mp_str(o, "path", 4);
   mp_str(o, path);
+  mp_str(o, "mtime_ms", 8);
+  mp_uint(o, uint64_t(nd.mtime));
   o += nd.blob;
 }
 
@@ -794,7 +797,8 @@ static void meta_set_serving(int64_t sid, bool on) {
 }
 
 static void meta_upsert(int64_t sid, int64_t id, bool is_dir,
-                        py::bytes blob, uint32_t npairs, py::bytes blocks) {
+                        py::bytes blob, uint32_t npairs, py::bytes blocks,
+                        int64_t mtime) {
   MetaServer* S = meta_get(sid);
   std::string b = blob;
   std::string bb = blocks;  // packed little-endian (block_id, length) i64 pairs
@@ -802,12 +806,84 @@ static void meta_upsert(int64_t sid, int64_t id, bool is_dir,
   MetaNode& nd = S->nodes[id];
   nd.is_dir = is_dir;
   nd.npairs = npairs;
+  nd.mtime = mtime;
   nd.blob = std::move(b);
   nd.blocks.clear();
   const int64_t* p = (const int64_t*)bb.data();
   size_t n = bb.size() / 16;
   nd.blocks.reserve(n);
   for (size_t i = 0; i < n; i++) nd.blocks.emplace_back(p[2 * i], p[2 * i + 1]);
+}
+
+static void mp_kv_uint(std::string& o, const char* k, size_t kn, uint64_t v) {
+  mp_str(o, k, kn);
+  mp_uint(o, v);
+}
+
+// fast-path upsert: the FileStatus blob is packed HERE from positional
+// args (no Python dict + packb per mutation — the mutation-QPS hot path)
+static void meta_upsert_node(int64_t sid, int64_t id, bool is_dir,
+                             const std::string& name, int64_t file_type,
+                             int64_t length, bool is_complete,
+                             int64_t block_size, int64_t replicas,
+                             const std::string& storage_tier,
+                             int64_t mtime_ms, int64_t atime_ms, int64_t mode,
+                             int64_t uid, int64_t gid, int64_t ttl_ms,
+                             const std::string& ttl_action,
+                             const std::string& symlink_target,
+                             int64_t nlink, py::bytes blocks,
+                             py::object xattrs_blob) {
+  MetaServer* S = meta_get(sid);
+  std::string bb = blocks;
+  std::string blob;
+  blob.reserve(192 + name.size());
+  uint32_t npairs = 16;
+  mp_kv_uint(blob, "inode_id", 8, uint64_t(id));
+  mp_str(blob, "name", 4);
+  mp_str(blob, name);
+  mp_kv_uint(blob, "file_type", 9, uint64_t(file_type));
+  mp_kv_uint(blob, "length", 6, uint64_t(length));
+  mp_str(blob, "is_complete", 11);
+  blob.push_back(char(is_complete ? 0xc3 : 0xc2));
+  mp_kv_uint(blob, "block_size", 10, uint64_t(block_size));
+  mp_kv_uint(blob, "replicas", 8, uint64_t(replicas));
+  mp_str(blob, "storage_tier", 12);
+  mp_str(blob, storage_tier);
+  mp_kv_uint(blob, "atime_ms", 8, uint64_t(atime_ms));
+  mp_kv_uint(blob, "mode", 4, uint64_t(mode));
+  mp_kv_uint(blob, "uid", 3, uint64_t(uid));
+  mp_kv_uint(blob, "gid", 3, uint64_t(gid));
+  mp_kv_uint(blob, "ttl_ms", 6, uint64_t(ttl_ms));
+  mp_str(blob, "ttl_action", 10);
+  mp_str(blob, ttl_action);
+  mp_str(blob, "symlink_target", 14);
+  mp_str(blob, symlink_target);
+  mp_kv_uint(blob, "nlink", 5, uint64_t(nlink));
+  if (!xattrs_blob.is_none()) {
+    // pre-packed msgpack map under key "xattrs" (rare path)
+    std::string xb = py::bytes(xattrs_blob);
+    mp_str(blob, "xattrs", 6);
+    blob += xb;
+    npairs += 1;
+  }
+  std::unique_lock<std::shared_mutex> lk(S->tree_mu);
+  MetaNode& nd = S->nodes[id];
+  nd.is_dir = is_dir;
+  nd.npairs = npairs;
+  nd.mtime = mtime_ms;
+  nd.blob = std::move(blob);
+  nd.blocks.clear();
+  const int64_t* p = (const int64_t*)bb.data();
+  size_t n = bb.size() / 16;
+  nd.blocks.reserve(n);
+  for (size_t i = 0; i < n; i++) nd.blocks.emplace_back(p[2 * i], p[2 * i + 1]);
+}
+
+static void meta_touch(int64_t sid, int64_t id, int64_t mtime) {
+  MetaServer* S = meta_get(sid);
+  std::unique_lock<std::shared_mutex> lk(S->tree_mu);
+  auto it = S->nodes.find(id);
+  if (it != S->nodes.end()) it->second.mtime = mtime;
 }
 
 static void meta_add_child(int64_t sid, int64_t parent,

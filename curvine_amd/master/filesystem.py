@@ -55,9 +55,8 @@ class MasterFilesystem:
         node = self.fs_dir.mkdir(path, mode, create_parents)
         return self.fs_dir.status_of(node)
 
-    def create(self, path: str, block_size: int = 0, replicas: int = 0,
-               storage_tier: str = "", overwrite: bool = False,
-               mode: int = 0o644) -> FileStatus:
+    def _create_node(self, path: str, block_size: int, replicas: int,
+                     storage_tier: str, overwrite: bool, mode: int) -> Inode:
         node, removed = self.fs_dir.create(
             path,
             block_size or self.conf.master.block_size,
@@ -68,7 +67,22 @@ class MasterFilesystem:
         if removed:
             self.workers.schedule_block_delete(removed)
         self.writing[node.id] = {"since_ms": now_ms()}
+        return node
+
+    def create(self, path: str, block_size: int = 0, replicas: int = 0,
+               storage_tier: str = "", overwrite: bool = False,
+               mode: int = 0o644) -> FileStatus:
+        node = self._create_node(path, block_size, replicas, storage_tier,
+                                 overwrite, mode)
         return self.fs_dir.status_of(node, norm_path(path))
+
+    def create_dict(self, path: str, block_size: int = 0, replicas: int = 0,
+                    storage_tier: str = "", overwrite: bool = False,
+                    mode: int = 0o644) -> dict:
+        """Reply-dict variant (mutation-QPS hot path)."""
+        node = self._create_node(path, block_size, replicas, storage_tier,
+                                 overwrite, mode)
+        return self.fs_dir.status_dict(node, norm_path(path))
 
     def append(self, path: str) -> FileBlocks:
         node = self.fs_dir.must_resolve(path)
@@ -122,9 +136,9 @@ class MasterFilesystem:
             locations=[w.address for w in workers],
             tiers=[node.storage_tier] * len(workers))
 
-    def complete_file(self, path: str, length: int,
-                      block_lens: list[int] | None = None,
-                      commits: list[dict] | None = None) -> FileStatus:
+    def _complete_node(self, path: str, length: int,
+                       block_lens: list[int] | None,
+                       commits: list[dict] | None) -> Inode:
         node = self.fs_dir.must_resolve(path)
         self.fs_dir.complete_file(node, length, block_lens)
         self.writing.pop(node.id, None)
@@ -134,7 +148,19 @@ class MasterFilesystem:
             for wid, tier in zip(c.get("locations", []),
                                  c.get("tiers", []) or ["MEM"] * len(c.get("locations", []))):
                 self.workers.add_location(c["block_id"], wid, tier)
+        return node
+
+    def complete_file(self, path: str, length: int,
+                      block_lens: list[int] | None = None,
+                      commits: list[dict] | None = None) -> FileStatus:
+        node = self._complete_node(path, length, block_lens, commits)
         return self.fs_dir.status_of(node, norm_path(path))
+
+    def complete_file_dict(self, path: str, length: int,
+                           block_lens: list[int] | None = None,
+                           commits: list[dict] | None = None) -> dict:
+        node = self._complete_node(path, length, block_lens, commits)
+        return self.fs_dir.status_dict(node, norm_path(path))
 
     def delete(self, path: str, recursive: bool = False) -> int:
         removed = self.fs_dir.delete(path, recursive)

@@ -107,6 +107,10 @@ class MirrorFanout:
         for m in self.mirrors:
             m.upsert(node)
 
+    def touch(self, inode_id: int, mtime_ms: int) -> None:
+        for m in self.mirrors:
+            m.touch(inode_id, mtime_ms)
+
     def add_child(self, parent_id: int, name: str, child_id: int) -> None:
         for m in self.mirrors:
             m.add_child(parent_id, name, child_id)
@@ -176,6 +180,23 @@ class FsDir:
             ttl_ms=node.ttl_ms, ttl_action=node.ttl_action,
             symlink_target=node.symlink_target, nlink=node.nlink,
             xattrs={k: bytes(v) for k, v in node.xattrs.items()})
+
+    def status_dict(self, node: Inode, path: str | None = None) -> dict:
+        """Reply-shaped status dict without the FileStatus dataclass hop
+        (mutation-QPS hot path; keys == FileStatus.to_dict)."""
+        return {
+            "inode_id": node.id,
+            "path": path if path is not None else self.path_of(node.id),
+            "name": node.name, "file_type": int(node.file_type),
+            "length": node.length, "is_complete": node.complete,
+            "block_size": node.block_size, "replicas": node.replicas,
+            "storage_tier": node.storage_tier,
+            "mtime_ms": node.mtime_ms, "atime_ms": node.atime_ms,
+            "mode": node.mode, "uid": node.uid, "gid": node.gid,
+            "ttl_ms": node.ttl_ms, "ttl_action": node.ttl_action,
+            "symlink_target": node.symlink_target, "nlink": node.nlink,
+            "xattrs": node.xattrs,
+        }
 
     def iter_files(self) -> Iterator[Inode]:
         for node in self.inodes.values():
@@ -277,7 +298,7 @@ class FsDir:
         self.next_inode_id = max(self.next_inode_id, node.id)
         if self.mirror:
             self.mirror.upsert(node)
-            self.mirror.upsert(parent)          # mtime changed
+            self.mirror.touch(parent.id, parent.mtime_ms)
             self.mirror.add_child(parent.id, e["name"], node.id)
         return node, removed
 
@@ -359,7 +380,7 @@ class FsDir:
             parent.mtime_ms = e.get("ts") or now_ms()
             if self.mirror:
                 self.mirror.remove_child(parent.id, name)
-                self.mirror.upsert(parent)
+                self.mirror.touch(parent.id, parent.mtime_ms)
         if node.file_type == FileType.FILE and node.nlink > 1:
             node.nlink -= 1   # other hardlinked names keep the data
             if self.mirror:
@@ -430,10 +451,10 @@ class FsDir:
         if self.mirror:
             if src_parent is not None:
                 self.mirror.remove_child(src_parent.id, old_name)
-                self.mirror.upsert(src_parent)
+                self.mirror.touch(src_parent.id, src_parent.mtime_ms)
             self.mirror.upsert(node)            # name changed
             self.mirror.add_child(dst_parent.id, node.name, node.id)
-            self.mirror.upsert(dst_parent)
+            self.mirror.touch(dst_parent.id, dst_parent.mtime_ms)
         return removed
 
     def set_attr(self, node: Inode, **attrs) -> None:
@@ -525,7 +546,7 @@ class FsDir:
         parent.mtime_ms = e.get("ts") or now_ms()
         if self.mirror:
             self.mirror.upsert(node)
-            self.mirror.upsert(parent)
+            self.mirror.touch(parent.id, parent.mtime_ms)
             self.mirror.add_child(parent.id, e["dst_name"], node.id)
 
     def resize(self, node: Inode, new_length: int) -> list[int]:

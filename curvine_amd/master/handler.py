@@ -33,15 +33,26 @@ _MUTATIONS = {
 
 
 class MasterHandler:
+    # code int -> (RpcCode, op method name); built once at class init
+    _DISPATCH: dict = {}
+
     def __init__(self, master):
         self.master = master
         self.fs: MasterFilesystem = master.fs
+        if not MasterHandler._DISPATCH:
+            d = {}
+            for code in RpcCode:
+                name = f"op_{code.name.lower()}"
+                if hasattr(MasterHandler, name):
+                    d[int(code)] = (code, name)
+            MasterHandler._DISPATCH = d
 
     async def handle(self, msg: Message, conn) -> Optional[Message]:
-        code = RpcCode(msg.code) if msg.code in RpcCode._value2member_map_ else RpcCode.Undefined
-        fn = getattr(self, f"op_{code.name.lower()}", None)
-        if fn is None:
+        ent = self._DISPATCH.get(msg.code)
+        if ent is None:
             raise err.Unsupported(f"rpc code {msg.code}")
+        code, fn_name = ent
+        fn = getattr(self, fn_name)
         raft = self.master.raft
         if raft is not None and not raft.is_leader and code not in _READONLY_OK:
             raise err.NotLeader(f"leader={raft.leader_addr or ''}")
@@ -87,10 +98,12 @@ class MasterHandler:
         return {"status": st.to_dict()}
 
     def op_createfile(self, h, d):
-        st = self.fs.create(h["path"], h.get("block_size", 0),
-                            h.get("replicas", 0), h.get("storage_tier", ""),
-                            h.get("overwrite", False), h.get("mode", 0o644))
-        return {"status": st.to_dict()}
+        st = self.fs.create_dict(h["path"], h.get("block_size", 0),
+                                 h.get("replicas", 0),
+                                 h.get("storage_tier", ""),
+                                 h.get("overwrite", False),
+                                 h.get("mode", 0o644))
+        return {"status": st}
 
     def op_appendfile(self, h, d):
         return {"file_blocks": self.fs.append(h["path"]).to_dict()}
@@ -123,9 +136,10 @@ class MasterHandler:
         return {"block": lb.to_dict()}
 
     def op_completefile(self, h, d):
-        st = self.fs.complete_file(h["path"], h["length"], h.get("block_lens"),
-                                   h.get("commits"))
-        return {"status": st.to_dict()}
+        st = self.fs.complete_file_dict(h["path"], h["length"],
+                                        h.get("block_lens"),
+                                        h.get("commits"))
+        return {"status": st}
 
     def op_getblocklocations(self, h, d):
         return {"file_blocks": self.fs.get_block_locations(h["path"]).to_dict()}

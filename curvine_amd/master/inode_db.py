@@ -48,6 +48,10 @@ class SqliteInodeStore:
         self._dirty.add(node.id)
         self._deleted.discard(node.id)
 
+    def touch(self, inode_id: int, mtime_ms: int) -> None:
+        # mtime lives inside the serialized state: re-flush the row
+        self._dirty.add(inode_id)
+
     def add_child(self, parent_id: int, name: str, child_id: int) -> None:
         # children live inside the parent's serialized state
         self._dirty.add(parent_id)

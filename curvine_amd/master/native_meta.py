@@ -51,13 +51,14 @@ def _pack_pairs(d: dict) -> tuple[bytes, int]:
 
 def _node_blob(node) -> tuple[bytes, int]:
     """FileStatus fields (fs_dir.status_of / model.FileStatus.to_dict)
-    minus the lookup-dependent ``path``."""
+    minus the lookup-dependent ``path`` and ``mtime_ms`` (kept out of the
+    blob so parent-mtime touches never repack — passed separately)."""
     return _pack_pairs({
         "inode_id": node.id, "name": node.name,
         "file_type": int(node.file_type), "length": node.length,
         "is_complete": node.complete, "block_size": node.block_size,
         "replicas": node.replicas, "storage_tier": node.storage_tier,
-        "mtime_ms": node.mtime_ms, "atime_ms": node.atime_ms,
+        "atime_ms": node.atime_ms,
         "mode": node.mode, "uid": node.uid, "gid": node.gid,
         "ttl_ms": node.ttl_ms, "ttl_action": node.ttl_action,
         "symlink_target": node.symlink_target, "nlink": node.nlink,
@@ -83,13 +84,28 @@ class MetaMirror:
         self.sid = sid
 
     def upsert(self, node) -> None:
-        blob, n = _node_blob(node)
-        self._raw_upsert(node.id, node.is_dir, blob, n, _pack_blocks(node))
+        # positional fast path: the blob is packed in C++ (no Python dict
+        # + msgpack per mutation — the mutation-QPS hot path)
+        xb = None
+        if node.xattrs:
+            import msgpack as _mp
+            xb = _mp.packb({k: bytes(v) for k, v in node.xattrs.items()},
+                           use_bin_type=True)
+        self.lib.meta_upsert_node(
+            self.sid, node.id, node.is_dir, node.name, int(node.file_type),
+            node.length, node.complete, node.block_size, node.replicas,
+            node.storage_tier, node.mtime_ms, node.atime_ms, node.mode,
+            node.uid, node.gid, node.ttl_ms, node.ttl_action,
+            node.symlink_target, node.nlink, _pack_blocks(node), xb)
+
+    def touch(self, inode_id: int, mtime_ms: int) -> None:
+        self.lib.meta_touch(self.sid, inode_id, mtime_ms)
 
     def _raw_upsert(self, inode_id: int, is_dir: bool, blob: bytes,
-                    n: int, blocks: bytes) -> None:
+                    n: int, blocks: bytes, mtime: int) -> None:
         """Pre-serialized upsert (CommitGatedMirror replays these)."""
-        self.lib.meta_upsert(self.sid, inode_id, is_dir, blob, n, blocks)
+        self.lib.meta_upsert(self.sid, inode_id, is_dir, blob, n, blocks,
+                             mtime)
 
     def add_child(self, parent_id: int, name: str, child_id: int) -> None:
         self.lib.meta_add_child(self.sid, parent_id, name, child_id)
@@ -146,7 +162,10 @@ class CommitGatedMirror:
     def upsert(self, node) -> None:
         blob, n = _node_blob(node)
         self._gate(("_raw_upsert", node.id, node.is_dir, blob, n,
-                    _pack_blocks(node)))
+                    _pack_blocks(node), node.mtime_ms))
+
+    def touch(self, inode_id: int, mtime_ms: int) -> None:
+        self._gate(("touch", inode_id, mtime_ms))
 
     def add_child(self, parent_id: int, name: str, child_id: int) -> None:
         self._gate(("add_child", parent_id, name, child_id))
@@ -280,7 +299,8 @@ class NativeMetaFrontend:
         up, ac = self.lib.meta_upsert, self.lib.meta_add_child
         for node in fs_dir.inodes.values():
             blob, n = _node_blob(node)
-            up(self.sid, node.id, node.is_dir, blob, n, _pack_blocks(node))
+            up(self.sid, node.id, node.is_dir, blob, n, _pack_blocks(node),
+               node.mtime_ms)
             if node.children:
                 for name, cid in node.children.items():
                     ac(self.sid, node.id, name, cid)
