@@ -79,6 +79,13 @@ class Inode:
 
 
 def norm_path(path: str) -> str:
+    # fast path: already normalized (no "//", no "." or ".." segments —
+    # any such segment necessarily contains "/." — no trailing slash).
+    # Called several times per mutation; the rebuild below is the
+    # exception, not the rule.
+    if path.startswith("/") and "//" not in path and "/." not in path \
+            and (len(path) == 1 or path[-1] != "/"):
+        return path
     if not path.startswith("/"):
         raise err.InvalidPath(f"path must be absolute: {path!r}")
     parts = [p for p in path.split("/") if p and p != "."]

@@ -70,7 +70,8 @@ class MasterHandler:
             cached = self.master.retry_cache.get(rkey)
             if cached is not None:
                 return msg.reply(cached)
-        t0 = time.perf_counter()
+        auditing = self.fs.conf.master.audit_log
+        t0 = time.perf_counter() if auditing else 0.0
         try:
             op_before = self.master.journal.op_id
             reply = fn(msg.header, msg.data)
@@ -84,7 +85,7 @@ class MasterHandler:
                 self.master.retry_cache.put(rkey, reply or {})
             return msg.reply(reply or {})
         finally:
-            if self.fs.conf.master.audit_log:
+            if auditing:
                 audit.info("cmd=%s used_us=%d", code.name,
                            int((time.perf_counter() - t0) * 1e6))
 

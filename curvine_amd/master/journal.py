@@ -28,6 +28,8 @@ from typing import Callable, Iterator, Optional
 
 import msgpack
 
+from curvine_amd.model import now_ms
+
 log = logging.getLogger("curvine.journal")
 
 _FRAME = struct.Struct(">II")
@@ -118,7 +120,6 @@ class JournalWriter:
         if not self.enabled:
             return {}
         self.op_id += 1
-        from curvine_amd.model import now_ms
         entry = {"op": op, "op_id": self.op_id, "ts": now_ms(), **fields}
         buf = encode_entry(entry)
         if self._f is None or self._seg_bytes + len(buf) > self.segment_max:

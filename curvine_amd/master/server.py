@@ -40,23 +40,23 @@ class RetryCache:
 
     def __init__(self, size: int = 100_000, ttl_ms: int = 600_000):
         from collections import OrderedDict
+        import time as _t
         self._d: "OrderedDict[tuple, tuple[float, dict]]" = OrderedDict()
         self.size = size
         self.ttl = ttl_ms / 1000.0
+        self._now = _t.monotonic
 
     def get(self, key):
-        import time as _t
         ent = self._d.get(key)
         if ent is None:
             return None
-        if _t.monotonic() - ent[0] > self.ttl:
+        if self._now() - ent[0] > self.ttl:
             self._d.pop(key, None)
             return None
         return ent[1]
 
     def put(self, key, reply: dict) -> None:
-        import time as _t
-        self._d[key] = (_t.monotonic(), reply)
+        self._d[key] = (self._now(), reply)
         while len(self._d) > self.size:
             self._d.popitem(last=False)
 

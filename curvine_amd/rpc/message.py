@@ -42,6 +42,9 @@ class Status(IntEnum):
         return self in (Status.Open, Status.Running)
 
 
+_STATUS_BY_VAL = [Status(i) if i in Status._value2member_map_ else Status.Unary
+                  for i in range(16)]
+
 _req_ids = itertools.count(1)
 
 
@@ -49,7 +52,7 @@ def next_req_id() -> int:
     return next(_req_ids)
 
 
-@dataclass
+@dataclass(slots=True)
 class Message:
     code: int = 0
     req_status: Status = Status.Unary
@@ -79,8 +82,8 @@ class Message:
     def decode_proto(proto: bytes) -> tuple[int, int, "Message"]:
         hlen, dlen, code, status, req_id, seq_id = _HDR.unpack(proto)
         msg = Message(code=code,
-                      req_status=Status(status & 0xF),
-                      resp_status=Status((status >> 4) & 0xF),
+                      req_status=_STATUS_BY_VAL[status & 0xF],
+                      resp_status=_STATUS_BY_VAL[(status >> 4) & 0xF],
                       req_id=req_id, seq_id=seq_id)
         return hlen, dlen, msg
 
