@@ -53,6 +53,18 @@ class MasterHandler:
             raise err.Unsupported(f"rpc code {msg.code}")
         code, fn_name = ent
         fn = getattr(self, fn_name)
+        pbuf = conn.state.get("pbuf", False)
+        if msg.raw_header:
+            # candidate protobuf header (reference-client wire compat):
+            # decode via the transcribed schema and pin the connection
+            from curvine_amd.rpc import proto as _proto
+            decoded = _proto.decode_request(msg.code, msg.raw_header)
+            if decoded is None:
+                raise err.InvalidArgument(
+                    f"undecodable header for code {msg.code}")
+            msg.header = decoded
+            msg.raw_header = b""
+            pbuf = conn.state["pbuf"] = True
         raft = self.master.raft
         if raft is not None and not raft.is_leader and code not in _READONLY_OK:
             raise err.NotLeader(f"leader={raft.leader_addr or ''}")
@@ -83,7 +95,14 @@ class MasterHandler:
                 await raft.wait_commit(self.master.journal.op_id)
             if rkey is not None:
                 self.master.retry_cache.put(rkey, reply or {})
-            return msg.reply(reply or {})
+            out = msg.reply(reply or {})
+            if pbuf:
+                from curvine_amd.rpc import proto as _proto
+                enc = _proto.encode_response(msg.code, out.header)
+                if enc is not None:
+                    out.header = {}
+                    out.raw_header = enc
+            return out
         finally:
             if auditing:
                 audit.info("cmd=%s used_us=%d", code.name,

@@ -61,10 +61,14 @@ class Message:
     seq_id: int = 0
     header: dict = field(default_factory=dict)
     data: bytes = b""
+    # undecodable-as-msgpack header bytes (candidate protobuf header from
+    # a reference client; rpc/proto.py decides)
+    raw_header: bytes = b""
 
     # ---------------- encode / decode ----------------
     def encode(self) -> bytes:
-        hdr_bytes = msgpack.packb(self.header, use_bin_type=True) if self.header else b""
+        hdr_bytes = self.raw_header or (
+            msgpack.packb(self.header, use_bin_type=True) if self.header else b"")
         status = (int(self.resp_status) << 4) | int(self.req_status)
         proto = _HDR.pack(len(hdr_bytes), len(self.data), self.code & 0xFF,
                           status, self.req_id, self.seq_id)
@@ -72,7 +76,8 @@ class Message:
 
     def encode_parts(self) -> list[bytes]:
         """Zero-copy-ish: [proto+header, data] so large payloads aren't copied."""
-        hdr_bytes = msgpack.packb(self.header, use_bin_type=True) if self.header else b""
+        hdr_bytes = self.raw_header or (
+            msgpack.packb(self.header, use_bin_type=True) if self.header else b"")
         status = (int(self.resp_status) << 4) | int(self.req_status)
         proto = _HDR.pack(len(hdr_bytes), len(self.data), self.code & 0xFF,
                           status, self.req_id, self.seq_id)
@@ -88,7 +93,21 @@ class Message:
         return hlen, dlen, msg
 
     def set_header_bytes(self, b: bytes) -> None:
-        self.header = msgpack.unpackb(b, raw=False) if b else {}
+        """Decode a header; non-msgpack bytes (a protobuf header from a
+        reference peer, or garbage) are kept raw for the handler's
+        per-connection codec to interpret."""
+        if not b:
+            self.header = {}
+            return
+        try:
+            h = msgpack.unpackb(b, raw=False)
+        except Exception:  # noqa: BLE001 — candidate protobuf header
+            h = None
+        if isinstance(h, dict):
+            self.header = h
+        else:
+            self.header = {}
+            self.raw_header = bytes(b)
 
     # ---------------- helpers ----------------
     @staticmethod

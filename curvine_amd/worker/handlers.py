@@ -38,6 +38,15 @@ class WorkerHandler:
 
     async def handle(self, msg: Message, conn) -> Optional[Message]:
         code = msg.code
+        if msg.raw_header:
+            from curvine_amd.rpc import proto as _proto
+            decoded = _proto.decode_request(code, msg.raw_header)
+            if decoded is None:
+                raise err.InvalidArgument(
+                    f"undecodable header for code {code}")
+            msg.header = decoded
+            msg.raw_header = b""
+            conn.state["pbuf"] = True
         if code == int(RpcCode.WriteBlock):
             return await self._write_block(msg)
         if code == int(RpcCode.ReadBlock):
