@@ -949,9 +949,6 @@ static py::tuple data_read_into(const std::string& host, int port,
   int64_t got = 0;
   {
     py::gil_scoped_release rel;
-    int fd = dc_acquire(host, port);
-    if (fd < 0) throw std::runtime_error("data connect failed");
-    bool keep = false;
     uint64_t req = g_dc_req.fetch_add(1);
     // Open frame
     std::string h;
@@ -967,32 +964,46 @@ static py::tuple data_read_into(const std::string& host, int port,
     std::string head = meta_proto(uint32_t(h.size()), 0, kCodeReadBlock,
                                   uint8_t(1) /*Open req*/, req, 0);
     head += h;
-    if (fd_write_all(fd, head.data(), head.size())) {
-      // consume frames until Complete/Error
-      std::vector<uint8_t> hdr_buf;
-      for (;;) {
-        uint8_t proto[kMetaProto];
-        if (!dc_read_exact(fd, proto, kMetaProto)) break;
-        uint32_t hlen = rd_u32be(proto);
-        uint32_t dlen = rd_u32be(proto + 4);
-        if (hlen > kMetaMaxLen || dlen > kMetaMaxLen) break;
-        uint8_t st = proto[9] >> 4;
-        hdr_buf.resize(hlen);
-        if (hlen && !dc_read_exact(fd, hdr_buf.data(), hlen)) break;
-        if (dlen) {
-          if (got + dlen > cap - dst_off) break;  // overflow: fail
-          if (!dc_read_exact(fd, out + got, dlen)) break;
-          got += dlen;
-        }
-        if (st == 3 || st == 5) {  // Complete / Error
-          final_status = st;
-          err_hdr.assign((const char*)hdr_buf.data(), hlen);
-          keep = true;
-          break;
+    // a pooled fd can be stale (server restarted, port reused): retry
+    // ONCE on a fresh socket if nothing was consumed yet
+    for (int attempt = 0; attempt < 2; attempt++) {
+      int fd = attempt == 0 ? dc_acquire(host, port) : dc_connect(host, port);
+      if (fd < 0) {
+        if (attempt == 0) continue;
+        throw std::runtime_error("data connect failed");
+      }
+      bool keep = false;
+      bool progressed = false;
+      got = 0;
+      if (fd_write_all(fd, head.data(), head.size())) {
+        // consume frames until Complete/Error
+        std::vector<uint8_t> hdr_buf;
+        for (;;) {
+          uint8_t proto[kMetaProto];
+          if (!dc_read_exact(fd, proto, kMetaProto)) break;
+          progressed = true;
+          uint32_t hlen = rd_u32be(proto);
+          uint32_t dlen = rd_u32be(proto + 4);
+          if (hlen > kMetaMaxLen || dlen > kMetaMaxLen) break;
+          uint8_t st = proto[9] >> 4;
+          hdr_buf.resize(hlen);
+          if (hlen && !dc_read_exact(fd, hdr_buf.data(), hlen)) break;
+          if (dlen) {
+            if (got + dlen > int64_t(cap - dst_off)) break;  // overflow
+            if (!dc_read_exact(fd, out + got, dlen)) break;
+            got += dlen;
+          }
+          if (st == 3 || st == 5) {  // Complete / Error
+            final_status = st;
+            err_hdr.assign((const char*)hdr_buf.data(), hlen);
+            keep = true;
+            break;
+          }
         }
       }
+      dc_release(host, port, fd, keep);
+      if (keep || progressed) break;  // real outcome (or torn mid-stream)
     }
-    dc_release(host, port, fd, keep);
   }
   return py::make_tuple(int(final_status), py::bytes(err_hdr), got);
 }
@@ -1013,12 +1024,8 @@ static py::tuple data_write_from(const std::string& host, int port,
   uint8_t final_status = 5;
   {
     py::gil_scoped_release rel;
-    int fd = dc_acquire(host, port);
-    if (fd < 0) throw std::runtime_error("data connect failed");
-    bool keep = false;
     uint64_t req = g_dc_req.fetch_add(1);
     uint32_t seq = 0;
-    // Open
     std::string h;
     h.push_back(char(0x84));
     mp_str(h, "block_id", 8);
@@ -1032,7 +1039,9 @@ static py::tuple data_write_from(const std::string& host, int port,
     std::string head = meta_proto(uint32_t(h.size()), 0, kCodeWriteBlock,
                                   uint8_t(1), req, seq++);
     head += h;
-    auto read_reply = [&](uint8_t* st_out, std::string* hdr_out) -> bool {
+
+    auto read_reply = [&](int fd, uint8_t* st_out,
+                          std::string* hdr_out) -> bool {
       uint8_t proto[kMetaProto];
       if (!dc_read_exact(fd, proto, kMetaProto)) return false;
       uint32_t hlen = rd_u32be(proto);
@@ -1045,16 +1054,28 @@ static py::tuple data_write_from(const std::string& host, int port,
       if (hdr_out) hdr_out->assign((const char*)tmp.data(), hlen);
       return true;
     };
-    bool ok = fd_write_all(fd, head.data(), head.size());
+
+    // Open round-trip; a pooled fd can be stale (server restart, port
+    // reuse) so retry ONCE on a fresh socket before giving up
+    int fd = -1;
     uint8_t st = 0;
-    if (ok) ok = read_reply(&st, &fin_hdr);
-    if (ok && st == 5) {  // Open failed (e.g. BlockInWriting)
+    for (int attempt = 0; attempt < 2; attempt++) {
+      fd = attempt == 0 ? dc_acquire(host, port) : dc_connect(host, port);
+      if (fd < 0) continue;
+      if (fd_write_all(fd, head.data(), head.size()) &&
+          read_reply(fd, &st, &fin_hdr))
+        break;
+      close(fd);
+      fd = -1;
+    }
+    if (fd < 0) throw std::runtime_error("data write open failed");
+    if (st == 5) {  // Open rejected (e.g. BlockInWriting): clean stream
       final_status = 5;
       dc_release(host, port, fd, true);
-      goto done;
-    }
-    if (ok) {
+    } else {
       // stream chunks with a pipelined ack window
+      bool ok = true;
+      bool keep = false;
       int inflight = 0;
       int64_t pos = 0;
       while (pos < length && ok) {
@@ -1090,13 +1111,13 @@ static py::tuple data_write_from(const std::string& host, int port,
         pos += cn;
         inflight++;
         while (ok && inflight >= window) {
-          ok = read_reply(&st, nullptr);
+          ok = read_reply(fd, &st, nullptr);
           inflight--;
           if (st == 5) ok = false;
         }
       }
       while (ok && inflight > 0) {
-        ok = read_reply(&st, nullptr);
+        ok = read_reply(fd, &st, nullptr);
         inflight--;
         if (st == 5) ok = false;
       }
@@ -1109,15 +1130,14 @@ static py::tuple data_write_from(const std::string& host, int port,
                                     uint8_t(3) /*Complete*/, req, seq++);
         cf += fh;
         ok = fd_write_all(fd, cf.data(), cf.size());
-        if (ok) ok = read_reply(&st, &fin_hdr);
+        if (ok) ok = read_reply(fd, &st, &fin_hdr);
         if (ok) {
           final_status = st;
           keep = true;
         }
       }
+      dc_release(host, port, fd, keep);
     }
-    dc_release(host, port, fd, keep);
-  done:;
   }
   return py::make_tuple(int(final_status), py::bytes(fin_hdr));
 }
