@@ -879,6 +879,46 @@ static void meta_upsert_node(int64_t sid, int64_t id, bool is_dir,
   for (size_t i = 0; i < n; i++) nd.blocks.emplace_back(p[2 * i], p[2 * i + 1]);
 }
 
+// atomic node + children-edge insert for paged fault-ins: a concurrent
+// native resolve must never observe a directory with a partially filled
+// children map (that reads as FileNotFound).  children: repeated
+// (u32 name_len LE, name bytes, i64 child_id LE).
+static void meta_upsert_with_children(int64_t sid, int64_t id, bool is_dir,
+                                      py::bytes blob, uint32_t npairs,
+                                      py::bytes blocks, int64_t mtime,
+                                      py::bytes children) {
+  MetaServer* S = meta_get(sid);
+  std::string b = blob;
+  std::string bb = blocks;
+  std::string cb = children;
+  std::unique_lock<std::shared_mutex> lk(S->tree_mu);
+  MetaNode& nd = S->nodes[id];
+  nd.is_dir = is_dir;
+  nd.npairs = npairs;
+  nd.mtime = mtime;
+  nd.blob = std::move(b);
+  nd.blocks.clear();
+  const int64_t* p = (const int64_t*)bb.data();
+  size_t n = bb.size() / 16;
+  nd.blocks.reserve(n);
+  for (size_t i = 0; i < n; i++) nd.blocks.emplace_back(p[2 * i], p[2 * i + 1]);
+  nd.children.clear();
+  const uint8_t* cp = (const uint8_t*)cb.data();
+  const uint8_t* cend = cp + cb.size();
+  while (cend - cp >= 4) {
+    uint32_t ln = uint32_t(cp[0]) | (uint32_t(cp[1]) << 8) |
+                  (uint32_t(cp[2]) << 16) | (uint32_t(cp[3]) << 24);
+    cp += 4;
+    if (size_t(cend - cp) < ln + 8) break;
+    std::string name((const char*)cp, ln);
+    cp += ln;
+    int64_t cid;
+    memcpy(&cid, cp, 8);
+    cp += 8;
+    nd.children[std::move(name)] = cid;
+  }
+}
+
 static void meta_touch(int64_t sid, int64_t id, int64_t mtime) {
   MetaServer* S = meta_get(sid);
   std::unique_lock<std::shared_mutex> lk(S->tree_mu);

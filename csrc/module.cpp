@@ -1173,6 +1173,7 @@ PYBIND11_MODULE(_native, m) {
   m.def("meta_upsert", &meta_upsert);
   m.def("meta_upsert_node", &meta_upsert_node);
   m.def("meta_touch", &meta_touch);
+  m.def("meta_upsert_with_children", &meta_upsert_with_children);
   m.def("meta_add_child", &meta_add_child);
   m.def("meta_remove_child", &meta_remove_child);
   m.def("meta_drop", &meta_drop);

@@ -116,6 +116,9 @@ class MasterConf:
     # inodes batched per actor tick; restart = table scan + WAL tail.
     # Non-raft masters only (raft nodes rebuild from the raft log).
     inode_db: bool = True
+    # beyond-RAM namespace: cap the resident Python/native inode maps and
+    # page cold, flushed inodes to sqlite (0 = unlimited, paging off)
+    max_resident_inodes: int = 0
 
 
 @dataclass

@@ -222,6 +222,9 @@ class Master:
                             "using asyncio rpc server", e)
                 self.native_meta = None
         await self.rpc.start()
+        if self.inode_db is not None and self.native_meta is not None \
+                and self.conf.master.max_resident_inodes > 0:
+            self.inode_db.on_fault = self._native_fault_in
         self.conf.master.rpc_port = self.rpc.port
         if self.raft is not None:
             if self.native_meta is not None:
@@ -263,28 +266,89 @@ class Master:
             return op
 
         start = 0
+        paged = self.conf.master.inode_db and \
+            self.conf.master.max_resident_inodes > 0
         if self.conf.master.inode_db:
             import os as _os
             from curvine_amd.master.inode_db import SqliteInodeStore
             self.inode_db = SqliteInodeStore(
                 _os.path.join(self.conf.journal.journal_dir, "inodes.db"))
-            start = self.inode_db.load(self.fs.fs_dir, self.mounts) or 0
+            if paged:
+                # beyond-RAM mode: restore watermarks + root only; attach
+                # the mirror BEFORE the WAL tail so replayed mutations
+                # mark their rows dirty instead of needing a full resync
+                start = self.inode_db.load_paged(self.fs.fs_dir,
+                                                 self.mounts) or 0
+                self.inode_db.enable_paging(self.fs.fs_dir)
+                self._attach_inode_db_mirror()
+            else:
+                start = self.inode_db.load(self.fs.fs_dir, self.mounts) or 0
         self.fs.loader.load(apply, load_snap, start_op=start)
         if self.inode_db is not None:
-            # hook future mutations; WAL-tail entries replayed above were
-            # applied without the mirror, so mark everything once
             fs_dir = self.fs.fs_dir
+            if paged:
+                if fs_dir.journal.op_id > start:
+                    # a journal snapshot newer than the DB was loaded
+                    # around the mirror: flush every resident inode
+                    self.inode_db._dirty.update(dict.keys(fs_dir.inodes))
+                return
+            # non-paged: WAL-tail entries replayed above were applied
+            # without the mirror, so mark everything once
             if fs_dir.journal.op_id > start:
                 self.inode_db.resync(fs_dir)
-            cur = fs_dir.mirror
-            if cur is None:
-                fs_dir.mirror = self.inode_db
-            else:
-                from curvine_amd.master.fs_dir import MirrorFanout
-                ms = cur.mirrors if isinstance(cur, MirrorFanout) else [cur]
+            self._attach_inode_db_mirror()
+
+    def _native_evict(self, iid: int) -> None:
+        if self.native_meta is not None:
+            self.native_meta.lib.meta_drop(self.native_meta.sid, iid)
+
+    def _native_fault_in(self, node) -> None:
+        """Paged-in inodes return to the C++ tree so repeat lookups are
+        served natively again.  The node and its children edges go in
+        under ONE tree lock — a concurrent native resolve must never see
+        a directory with a partially filled children map (that would
+        serve a spurious FileNotFound)."""
+        nm = self.native_meta
+        if nm is None:
+            return
+        import struct as _st
+
+        from curvine_amd.master.native_meta import (_node_blob,
+                                                    _pack_blocks)
+        blob, n = _node_blob(node)
+        parts = []
+        if node.children:
+            for name, cid in node.children.items():
+                nb = name.encode()
+                parts.append(_st.pack("<I", len(nb)) + nb +
+                             _st.pack("<q", cid))
+        nm.lib.meta_upsert_with_children(
+            nm.sid, node.id, node.is_dir, blob, n, _pack_blocks(node),
+            node.mtime_ms, b"".join(parts))
+
+    def _attach_inode_db_mirror(self) -> None:
+        fs_dir = self.fs.fs_dir
+        cur = fs_dir.mirror
+        if cur is None:
+            fs_dir.mirror = self.inode_db
+        elif cur is not self.inode_db:
+            from curvine_amd.master.fs_dir import MirrorFanout
+            ms = cur.mirrors if isinstance(cur, MirrorFanout) else [cur]
+            if self.inode_db not in ms:
                 fs_dir.mirror = MirrorFanout(ms + [self.inode_db])
 
     def checkpoint(self) -> None:
+        if self.inode_db is not None and \
+                self.conf.master.max_resident_inodes > 0:
+            # paged mode: sqlite IS the checkpoint; a full-namespace
+            # snapshot would defeat the memory bound.  Purge the journal
+            # through the store's fully-flushed watermark only.
+            row = self.inode_db.conn.execute(
+                "SELECT v FROM meta WHERE k='op_id'").fetchone()
+            if row is not None:
+                self.journal.purge_through(
+                    int.from_bytes(row[0], "little"))
+            return
         state = self._snapshot_state()
         self.fs.loader.save_snapshot(state)
         if self.raft is not None:
@@ -324,20 +388,39 @@ class Master:
                     self.inode_db.flush(self.fs.fs_dir,
                                         self.mounts.to_snapshot(),
                                         self.journal.op_id)
+                    maxres = self.conf.master.max_resident_inodes
+                    if maxres > 0:
+                        self.inode_db.page_out(
+                            self.fs.fs_dir, set(self.fs.writing),
+                            maxres, self._native_evict)
                 if self.journal.op_id and tick % 60 == 0:
                     self.checkpoint()
             except Exception as e:  # noqa: BLE001
                 log.exception("master actor tick failed: %s", e)
 
     def _ttl_sweep(self) -> None:
-        """TTL subsystem analog (meta/inode/ttl/ttl_manager.rs:35-49)."""
+        """TTL subsystem analog (meta/inode/ttl/ttl_manager.rs:35-49).
+        Paged mode also consults the store's ttl_deadline column so
+        paged-out inodes still expire."""
         from curvine_amd.model import now_ms
         now = now_ms()
         expired = []
+        seen = set()
         for node in list(self.fs.fs_dir.inodes.values()):
             if node.ttl_ms > 0 and node.create_ms + node.ttl_ms < now \
                     and node.ttl_action in ("delete", "free"):
                 expired.append(node)
+                seen.add(node.id)
+        if self.inode_db is not None and \
+                self.conf.master.max_resident_inodes > 0:
+            for iid in self.inode_db.ttl_expired_ids(now):
+                if iid in seen:
+                    continue
+                node = self.fs.fs_dir.inodes.get(iid)   # faults in
+                if node is not None and node.ttl_ms > 0 and \
+                        node.create_ms + node.ttl_ms < now and \
+                        node.ttl_action in ("delete", "free"):
+                    expired.append(node)
         for node in expired:
             path = self.fs.fs_dir.path_of(node.id)
             try:
@@ -364,6 +447,17 @@ class Master:
         target = cap * self.conf.master.eviction_low_watermark
         files = [n for n in self.fs.fs_dir.iter_files()
                  if n.complete and n.blocks and n.id not in self.fs.writing]
+        if self.inode_db is not None and \
+                self.conf.master.max_resident_inodes > 0:
+            # paged mode: the coldest files may not be resident
+            res = {n.id for n in files}
+            for iid in self.inode_db.cold_file_ids():
+                if iid in res:
+                    continue
+                n = self.fs.fs_dir.inodes.get(iid)   # faults in
+                if n is not None and n.complete and n.blocks and \
+                        n.id not in self.fs.writing:
+                    files.append(n)
         if policy == "lfu":
             files.sort(key=lambda n: (n.access_count, n.atime_ms))
         else:
