@@ -186,6 +186,12 @@ class SyncWriteState:
         self._cur_async = None     # fallback async writers
         self._cur_pos = 0
         self._done = False
+        # shared-writer state (backend_handle.rs:288-310 analog): several
+        # FUSE handles may hold this writer concurrently
+        import threading as _th
+        self.wlock = _th.Lock()
+        self.refs = 1
+        self.native_handle = None   # FileHandle owning a native window
 
     def write(self, data, ptr: int | None = None) -> int:
         """`ptr`, when given, is the raw address of `data` inside a pinned
@@ -376,6 +382,10 @@ class CurvineFuseFs:
         self.nodes_lock = threading.Lock()
         self.next_node = 2
         self.handles: dict[int, FileHandle] = {}
+        # node_id -> live SyncWriteState shared across handles
+        self.shared_writers: dict[int, "SyncWriteState"] = {}
+        # serializes write-open registry check+create across channels
+        self.wopen_lock = threading.Lock()
         self.next_fh = 1
         self.handles_lock = threading.Lock()
         self.attr_cache: dict[int, tuple[FileStatus, float]] = {}
@@ -473,6 +483,8 @@ class CurvineFuseFs:
         from curvine_amd.native import load
         n = load().fuse_loop_unregister_write(self.session.native_id, h.fh)
         h.nw_registered = False
+        if h.writer is not None and h.writer.native_handle is h:
+            h.writer.native_handle = None
         if n and h.writer is not None:
             h.writer.advance(n)
             h.write_pos += n
@@ -488,7 +500,7 @@ class CurvineFuseFs:
         w = h.writer
         if self.session is None or self.session.native_id is None or \
                 w is None or w._done or w._cur is None or \
-                w._cur.meta.get("kind") != "arena":
+                w.refs > 1 or w._cur.meta.get("kind") != "arena":
             return
         room = w.block_size - w._cur_pos
         if room <= 0:
@@ -498,6 +510,7 @@ class CurvineFuseFs:
             self.session.native_id, h.fh, w._cur.layout.arena.handle,
             w._cur.meta["offset"] + w._cur_pos, room, h.write_pos)
         h.nw_registered = True
+        w.native_handle = h
 
     def new_handle(self, node_id: int, path: str) -> FileHandle:
         with self.handles_lock:
@@ -806,19 +819,26 @@ class CurvineFuseFs:
                     except Exception as e:  # noqa: BLE001
                         log.debug("native read registration failed: %s", e)
             else:
-                if flags & os.O_TRUNC:
-                    st = self.call(self.fs.client.create(path, overwrite=True))
-                    h.writer = SyncWriteState(self, st)
-                    h.write_pos = 0
-                elif flags & os.O_APPEND or accmode in (os.O_WRONLY, os.O_RDWR):
-                    st = self.call(self.fs.file_status(path))
-                    if st.length == 0:
-                        st = self.call(self.fs.client.create(path, overwrite=True))
-                        h.writer = SyncWriteState(self, st)
-                    else:
-                        fb = self.call(self.fs.client.append(path))
-                        h.writer = SyncWriteState(self, fb.status, fb.blocks)
-                    h.write_pos = h.writer.pos
+                # the registry check + append-lease acquisition must be
+                # atomic across FUSE channels, or two simultaneous
+                # write-opens both miss and the loser gets FileInWriting
+                self.wopen_lock.acquire()
+                try:
+                    shared = None
+                    if not flags & os.O_TRUNC:
+                        # POSIX allows several concurrent write-opens of
+                        # one file: share the live writer instead of
+                        # failing the append lease (shared-writer
+                        # semantics, backend_handle.rs:288-310)
+                        with self.handles_lock:
+                            sw = self.shared_writers.get(nodeid)
+                            if sw is not None and not sw._done:
+                                sw.refs += 1
+                                shared = sw
+                    self._wopen_branch(h, nodeid, path, flags, accmode,
+                                       shared)
+                finally:
+                    self.wopen_lock.release()
                 h.status = h.writer.status
                 if accmode == os.O_RDWR:
                     try:
@@ -832,6 +852,33 @@ class CurvineFuseFs:
             with self.handles_lock:
                 self.handles.pop(h.fh, None)
             raise
+
+    def _wopen_branch(self, h, nodeid, path, flags, accmode, shared):
+        import os
+        if shared is not None:
+            nh = shared.native_handle
+            if nh is not None:
+                with nh.lock:
+                    self._sync_native_write(nh)
+            h.writer = shared
+            h.write_pos = shared.pos
+        elif flags & os.O_TRUNC:
+            st = self.call(self.fs.client.create(path, overwrite=True))
+            h.writer = SyncWriteState(self, st)
+            h.write_pos = 0
+            with self.handles_lock:
+                self.shared_writers[nodeid] = h.writer
+        elif flags & os.O_APPEND or accmode in (os.O_WRONLY, os.O_RDWR):
+            st = self.call(self.fs.file_status(path))
+            if st.length == 0:
+                st = self.call(self.fs.client.create(path, overwrite=True))
+                h.writer = SyncWriteState(self, st)
+            else:
+                fb = self.call(self.fs.client.append(path))
+                h.writer = SyncWriteState(self, fb.status, fb.blocks)
+            h.write_pos = h.writer.pos
+            with self.handles_lock:
+                self.shared_writers[nodeid] = h.writer
 
     def op_read(self, nodeid, body, ctx):
         fh, offset, size, _rf, _lo, _fl, _ = abi.READ_IN.unpack_from(body, 0)
@@ -872,24 +919,34 @@ class CurvineFuseFs:
                 fb = self.call(self.fs.client.append(h.path))
                 h.writer = SyncWriteState(self, fb.status, fb.blocks)
                 h.write_pos = h.writer.pos
-            if offset > h.write_pos:
-                # forward seek: zero-fill the sparse hole
-                h.writer.write_zeros(offset - h.write_pos)
-                h.write_pos = offset
-            elif offset < h.write_pos:
-                # random write: rewrite the overlap in place, append any
-                # tail past the current end (linkers, rsync --inplace)
-                overlap = min(size, h.write_pos - offset)
-                h.writer.pwrite_back(offset, data[:overlap])
-                if overlap < size:
-                    h.writer.write(data[overlap:])
+                with self.handles_lock:
+                    self.shared_writers[h.node_id] = h.writer
+            w = h.writer
+            with w.wlock:
+                nh = w.native_handle
+                if nh is not None and nh is not h:
+                    # another handle has an open native append window:
+                    # fold it before touching the shared position
+                    self._sync_native_write(nh)
+                end = w.pos   # the SHARED append end (several handles
+                #               may interleave; h.write_pos is per-handle)
+                if offset > end:
+                    # forward seek: zero-fill the sparse hole
+                    w.write_zeros(offset - end)
+                elif offset < end:
+                    # random write: rewrite the overlap in place, append
+                    # any tail past the current end
+                    overlap = min(size, end - offset)
+                    w.pwrite_back(offset, data[:overlap])
+                    if overlap < size:
+                        w.write(data[overlap:])
                     h.write_pos = offset + size
+                    self._register_native_write(h)
+                    return abi.WRITE_OUT.pack(size, 0)
+                w.write(data, ptr=ptr)
+                h.write_pos = w.pos
+                # subsequent sequential WRITEs append GIL-free in C++
                 self._register_native_write(h)
-                return abi.WRITE_OUT.pack(size, 0)
-            h.writer.write(data, ptr=ptr)
-            h.write_pos += size
-            # subsequent sequential WRITEs append GIL-free in C++
-            self._register_native_write(h)
         return abi.WRITE_OUT.pack(size, 0)
 
     def op_flush(self, nodeid, body, ctx):
@@ -899,9 +956,12 @@ class CurvineFuseFs:
         fh, _u, _p, _lo = abi.FLUSH_IN.unpack_from(body, 0)
         h = self.handles.get(fh)
         if h is not None and h.writer is not None:
-            with h.lock:
+            with h.lock, h.writer.wlock:
                 self._sync_native_write(h)
-                if not h.writer._done:
+                if not h.writer._done and h.writer.refs <= 1:
+                    # last write handle: complete on close(2).  With
+                    # other handles still writing, completion waits for
+                    # the final release.
                     st = h.writer.complete()
                     self.cache_status(h.node_id, st)
         return b""
@@ -916,10 +976,16 @@ class CurvineFuseFs:
         if h is None:
             return b""
         if h.writer is not None:
-            with h.lock:
+            with h.lock, h.writer.wlock:
                 self._sync_native_write(h)
-                st = h.writer.complete()
-                self.cache_status(h.node_id, st)
+                h.writer.refs -= 1
+                if h.writer.refs <= 0:
+                    if not h.writer._done:
+                        st = h.writer.complete()
+                        self.cache_status(h.node_id, st)
+                    with self.handles_lock:
+                        if self.shared_writers.get(h.node_id) is h.writer:
+                            del self.shared_writers[h.node_id]
         if h.reader is not None:
             # unregister from the native loop BEFORE closing store readers
             # (registration pins the blocks via reader refcounts)

@@ -273,3 +273,46 @@ def test_truncate_storm_under_readers(mount):
             t.join(timeout=30)
     assert not errors
     assert os.path.getsize(p) == 1_000_000
+
+
+def test_concurrent_write_open_same_file(mount):
+    """POSIX: several threads open the SAME file O_WRONLY concurrently
+    (fio rand-write shape).  The FUSE layer shares one writer across the
+    handles (backend_handle.rs shared-writer analog) instead of failing
+    the append lease."""
+    import threading as _th
+    mnt, *_ = mount
+    p = f"{mnt}/shared.bin"
+    with open(p, "wb") as f:
+        f.write(b"\x00" * (1 << 20))
+    errs = []
+    slots = {}
+    lk = _th.Lock()
+
+    def w(t):
+        try:
+            fd = os.open(p, os.O_WRONLY)
+            try:
+                for k in range(8):
+                    slot = (t * 8 + k) % 16
+                    data = bytes([t * 16 + k]) * (64 << 10)
+                    with lk:
+                        slots[slot] = data
+                        os.pwrite(fd, data, slot * (64 << 10))
+            finally:
+                os.close(fd)
+        except Exception as e:  # noqa: BLE001
+            errs.append(e)
+
+    ts = [_th.Thread(target=w, args=(t,)) for t in range(4)]
+    for t in ts:
+        t.start()
+    for t in ts:
+        t.join()
+    assert not errs, errs[0]
+    with open(p, "rb") as f:
+        content = f.read()
+    assert len(content) == 1 << 20
+    for slot, data in slots.items():
+        assert content[slot * (64 << 10):(slot + 1) * (64 << 10)] == data, \
+            f"slot {slot} mismatch"
