@@ -27,6 +27,9 @@ def get_ufs(uri: str, properties: dict | None = None) -> UnderFs:
         return _CACHE[key]
     if uri.startswith("s3://") or uri.startswith("oss://"):
         fs: UnderFs = S3Ufs(uri, properties)
+    elif uri.startswith("hdfs://") or uri.startswith("webhdfs://"):
+        from curvine_amd.ufs.webhdfs import WebHdfsUfs
+        fs = WebHdfsUfs(uri, properties)
     elif uri.startswith("file://"):
         fs = LocalUfs(uri[len("file://"):])
     elif uri.startswith("/"):
