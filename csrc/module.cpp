@@ -1115,6 +1115,7 @@ static void device_sync(int device) {
 // native metadata RPC frontend (pure epoll/C++, no GPU involvement)
 #include "meta_server.cpp"
 #include "data_server.cpp"
+#include "sdk_abi.cpp"
 
 static py::dict device_mem_info(int device) {
   size_t free_b = 0, total_b = 0;

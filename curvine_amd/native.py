@@ -24,6 +24,7 @@ _SRC3 = os.path.join(_REPO, "csrc", "lz4.hip")
 _SRC4 = os.path.join(_REPO, "csrc", "fuse_loop.hip")
 _SRC5 = os.path.join(_REPO, "csrc", "meta_server.cpp")
 _SRC6 = os.path.join(_REPO, "csrc", "data_server.cpp")
+_SRC7 = os.path.join(_REPO, "csrc", "sdk_abi.cpp")
 _lock = threading.Lock()
 _mod = None
 
@@ -36,7 +37,8 @@ def build_native(force: bool = False) -> str:
                                         os.path.getmtime(_SRC3),
                                         os.path.getmtime(_SRC4),
                                         os.path.getmtime(_SRC5),
-                                        os.path.getmtime(_SRC6)):
+                                        os.path.getmtime(_SRC6),
+                                        os.path.getmtime(_SRC7)):
             return _SO
     import pybind11
     hipcc = os.environ.get("HIPCC", "/opt/rocm/bin/hipcc")
