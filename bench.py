@@ -430,6 +430,93 @@ def step_client_seq(args, rank, rt) -> int:
     return got
 
 
+def run_scale_sweep(args, rank, world, dist, torch, rt, has_gpu,
+                    one_step, barrier_sync):
+    """one command produces the 1/2/4/8 weak-scaling
+    curve (rank subgroups re-run the timed step; idle ranks wait at the
+    global barrier) AND an RCCL BlockDistributor.broadcast_file number,
+    written to profiles/scale_sweep_n<world>.json by rank 0.  At world=1
+    this is the pre-flight: the same code path, degenerate collectives."""
+    out = {"world": world, "workload": args.workload, "path": args.path,
+           "tier": "HBM" if has_gpu else "MEM(cpu)", "per_n": {}}
+    ns = [n for n in (1, 2, 4, 8) if n <= world]
+    groups = {}
+    for n in ns:
+        # new_group is collective: every rank participates in creation
+        groups[n] = (dist.new_group(ranks=list(range(n)))
+                     if dist is not None and n < world else None)
+    for n in ns:
+        g = groups[n]
+        if rank < n:
+            if dist is not None:
+                dist.barrier(group=g) if g is not None else dist.barrier()
+            if has_gpu:
+                torch.cuda.synchronize()
+            t0 = time.perf_counter()
+            bps = 0
+            for _ in range(args.steps):
+                bps = one_step()
+            if has_gpu:
+                torch.cuda.synchronize()
+            elapsed = time.perf_counter() - t0
+            if dist is not None and n > 1:
+                t = torch.tensor([elapsed], dtype=torch.float64)
+                dist.all_reduce(t, op=dist.ReduceOp.MAX, group=g)
+                elapsed = t.item()
+            if rank == 0:
+                total = bps * args.steps * n
+                out["per_n"][str(n)] = {
+                    "GiBps": round(total / elapsed / (1 << 30), 3),
+                    "ms_per_step": round(elapsed / args.steps * 1000, 2),
+                    "bytes_per_rank_step": bps,
+                }
+        if dist is not None:
+            dist.barrier()
+    # ---- RCCL broadcast_file (xGMI on GPUs, gloo on CPU) ----
+    if rt.worker is not None:
+        try:
+            from curvine_amd.parallel.distributor import BlockDistributor
+            local_rank = int(os.environ.get("LOCAL_RANK", rank))
+            if dist is None:
+                import torch.distributed as dist_mod
+                if not dist_mod.is_initialized():
+                    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+                    os.environ.setdefault("MASTER_PORT", "29571")
+                    dist_mod.init_process_group(
+                        "nccl" if has_gpu else "gloo", rank=0, world_size=1)
+            bd = BlockDistributor(device=local_rank if has_gpu else None)
+            path = "/bench/r0/f0"
+
+            def open_file(p):
+                return rt.call(rt.fs.client.open(p))
+
+            barrier_sync()
+            t0 = time.perf_counter()
+            sizes = bd.broadcast_file(open_file, rt.worker.store, path, 0)
+            barrier_sync()
+            dt = time.perf_counter() - t0
+            moved = sum(sizes.values())
+            if rank == 0:
+                out["broadcast_file"] = {
+                    "bytes": moved, "seconds": round(dt, 4),
+                    "GiBps_algo": round(moved / dt / (1 << 30), 3),
+                    "GiBps_bus": round(moved * max(1, world - 1) / dt
+                                       / (1 << 30), 3),
+                    "backend": "nccl(RCCL)" if has_gpu else "gloo",
+                    "blocks": len(sizes),
+                }
+        except Exception as e:  # noqa: BLE001 — sweep is best-effort
+            if rank == 0:
+                out["broadcast_file"] = {"error": str(e)}
+    if rank == 0:
+        os.makedirs("profiles", exist_ok=True)
+        dest = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                            "profiles", f"scale_sweep_n{world}.json")
+        with open(dest, "w") as f:
+            json.dump(out, f)
+        print("SCALE_SWEEP " + json.dumps(out), flush=True)
+
+
 def main():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
@@ -454,6 +541,11 @@ def main():
     p.add_argument("--staging-bytes", type=int, default=8 << 20)
     p.add_argument("--staging-count", type=int, default=8)
     p.add_argument("--fuse-channels", type=int, default=8)
+    p.add_argument("--scale-sweep", action="store_true",
+                   help="after the main measurement, run the 1/2/4/8 "
+                        "weak-scaling curve over rank subgroups plus a "
+                        "BlockDistributor.broadcast_file xGMI benchmark, "
+                        "writing profiles/scale_sweep_n<world>.json")
     args = p.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -588,6 +680,10 @@ def main():
             },
         }
         print(json.dumps(result), flush=True)
+
+    if args.scale_sweep:
+        run_scale_sweep(args, rank, world, dist, torch, rt, has_gpu,
+                        one_step, barrier_sync)
 
     # teardown
     if daemon_proc is not None:
