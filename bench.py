@@ -265,7 +265,8 @@ def step_fuse_rand4k(args, rank, mnt, lat_out: list) -> int:
     alll = sorted(x for l in lats for x in l)
     if alll:
         lat_out.append({"p50_us": alll[len(alll) // 2],
-                        "p99_us": alll[int(len(alll) * 0.99)]})
+                        "p99_us": alll[int(len(alll) * 0.99)],
+                        "convention": "per-op pread wall time"})
     return sum(total)
 
 
@@ -363,7 +364,8 @@ def step_client_rand4k(args, rank, rt, lat_out: list) -> int:
     alll = sorted(x for l in lats for x in l)
     if alll:
         lat_out.append({"p50_us": alll[len(alll) // 2],
-                        "p99_us": alll[int(len(alll) * 0.99)]})
+                        "p99_us": alll[int(len(alll) * 0.99)],
+                        "convention": "iodepth-amortized (batch wall / depth)"})
     return sum(total)
 
 

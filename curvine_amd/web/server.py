@@ -18,27 +18,93 @@ from typing import Optional
 log = logging.getLogger("curvine.web")
 
 _PAGE = """<!doctype html><html><head><title>curvine-amd</title>
-<style>body{font-family:monospace;margin:2em}table{border-collapse:collapse}
-td,th{border:1px solid #999;padding:4px 8px;text-align:left}</style></head>
-<body><h2>curvine-amd cluster</h2><div id=info></div>
-<h3>workers</h3><table id=w><tr><th>id</th><th>addr</th><th>device</th>
-<th>used</th><th>capacity</th><th>tiers</th></tr></table>
+<meta charset=utf-8>
+<style>
+body{font-family:-apple-system,'Segoe UI',Roboto,monospace;margin:0;background:#0d1117;color:#c9d1d9}
+header{background:#161b22;padding:12px 24px;display:flex;align-items:center;gap:16px;border-bottom:1px solid #30363d}
+header h1{font-size:16px;margin:0;color:#e6edf3}
+header .pill{background:#1f6feb22;color:#58a6ff;border:1px solid #1f6feb66;border-radius:12px;padding:2px 10px;font-size:12px}
+main{padding:16px 24px;display:grid;grid-template-columns:1fr 1fr;gap:16px;max-width:1200px}
+section{background:#161b22;border:1px solid #30363d;border-radius:8px;padding:12px 16px}
+section h2{font-size:13px;margin:0 0 8px;color:#8b949e;text-transform:uppercase;letter-spacing:.08em}
+table{border-collapse:collapse;width:100%;font-size:13px}
+td,th{border-bottom:1px solid #21262d;padding:4px 8px;text-align:left}
+th{color:#8b949e;font-weight:600}
+.bar{background:#21262d;border-radius:4px;height:8px;overflow:hidden}
+.bar i{display:block;height:100%;background:#238636}
+.crumb a{color:#58a6ff;text-decoration:none;cursor:pointer}
+#browse td a{color:#58a6ff;text-decoration:none;cursor:pointer}
+.num{text-align:right;font-variant-numeric:tabular-nums}
+.wide{grid-column:1/3}
+</style></head><body>
+<header><h1>curvine-amd</h1><span class=pill id=role>master</span>
+<span id=summary style="font-size:13px;color:#8b949e"></span></header>
+<main>
+<section class=wide><h2>cluster</h2><div id=capbar class=bar><i style="width:0"></i></div>
+<div id=info style="margin-top:6px;font-size:13px"></div></section>
+<section class=wide><h2>workers</h2><table id=w><thead><tr><th>id</th><th>addr</th><th>gpu</th><th class=num>used</th><th class=num>capacity</th><th>tiers</th></tr></thead><tbody></tbody></table></section>
+<section class=wide><h2>namespace <span class=crumb id=crumb></span></h2>
+<table id=browse><thead><tr><th>name</th><th>type</th><th class=num>size</th><th>tier</th><th class=num>mtime</th></tr></thead><tbody></tbody></table></section>
+<section><h2>mounts</h2><table id=mounts><tbody></tbody></table></section>
+<section><h2>jobs</h2><table id=jobs><tbody></tbody></table></section>
+<section><h2>raft</h2><div id=raft style="font-size:13px">standalone</div></section>
+<section><h2>fuse</h2><div id=fuse style="font-size:13px">-</div></section>
+</main>
 <script>
-fetch('/api/info').then(r=>r.json()).then(d=>{
- document.getElementById('info').innerText =
-  `inodes ${d.inode_num}  blocks ${d.block_num}  used ${(d.used/2**30).toFixed(2)} / ${(d.capacity/2**30).toFixed(2)} GiB`;
- const t=document.getElementById('w');
- for(const w of d.live_workers){const r=t.insertRow();
-  const a=w.address;
-  r.insertCell().innerText=a.worker_id;
-  r.insertCell().innerText=`${a.hostname}:${a.rpc_port}`;
-  r.insertCell().innerText=a.device_id;
-  const used=w.storages.reduce((s,x)=>s+x.used,0);
-  const cap=w.storages.reduce((s,x)=>s+x.capacity,0);
-  r.insertCell().innerText=(used/2**30).toFixed(2)+' GiB';
-  r.insertCell().innerText=(cap/2**30).toFixed(2)+' GiB';
-  r.insertCell().innerText=w.storages.map(s=>`${s.tier}:${(s.capacity/2**30).toFixed(0)}G`).join(' ');
- }});
+const gib=b=>(b/2**30).toFixed(2)+' GiB';
+const J=u=>fetch(u).then(r=>r.ok?r.json():null).catch(()=>null);
+let cwd='/';
+async function refresh(){
+ const d=await J('/api/info');
+ if(d){
+  const pct=d.capacity? (100*d.used/d.capacity):0;
+  document.querySelector('#capbar i').style.width=pct.toFixed(1)+'%';
+  document.getElementById('info').innerText=
+   `${d.inode_num} inodes, ${d.block_num} blocks, ${gib(d.used)} of ${gib(d.capacity)} used (${pct.toFixed(1)}%)`;
+  document.getElementById('summary').innerText=
+   `${(d.live_workers||[]).length} workers live`;
+  const tb=document.querySelector('#w tbody');tb.innerHTML='';
+  for(const w of d.live_workers||[]){const r=tb.insertRow();const a=w.address;
+   const used=w.storages.reduce((s,x)=>s+x.used,0),cap=w.storages.reduce((s,x)=>s+x.capacity,0);
+   r.innerHTML=`<td>${a.worker_id}</td><td>${a.hostname}:${a.rpc_port}</td><td>${a.device_id}</td>`+
+    `<td class=num>${gib(used)}</td><td class=num>${gib(cap)}</td>`+
+    `<td>${w.storages.map(s=>s.tier+':'+(s.capacity/2**30).toFixed(0)+'G').join(' ')}</td>`;}
+ }
+ const m=await J('/api/mounts');
+ if(m){const tb=document.querySelector('#mounts tbody');tb.innerHTML='';
+  for(const x of m){const r=tb.insertRow();
+   r.innerHTML=`<td>${x.cv_path||x.path||''}</td><td>${x.ufs_path||''}</td><td>${x.cache_mode||''}</td>`;}
+  if(!m.length)tb.innerHTML='<tr><td style=color:#8b949e>none</td></tr>';}
+ const j=await J('/api/jobs');
+ if(j){const tb=document.querySelector('#jobs tbody');tb.innerHTML='';
+  const ids=Object.keys(j);
+  for(const id of ids.slice(-10)){const x=j[id];const r=tb.insertRow();
+   r.innerHTML=`<td>${id}</td><td>${x.state}</td><td class=num>${x.done}/${x.total}</td>`;}
+  if(!ids.length)tb.innerHTML='<tr><td style=color:#8b949e>none</td></tr>';}
+ const rf=await J('/api/raft');
+ if(rf)document.getElementById('raft').innerText=
+  `node ${rf.id} ${rf.state}, term ${rf.term}, leader ${rf.leader}, commit ${rf.commit}/${rf.last_index}`;
+ const fu=await J('/api/fuse');
+ if(fu)document.getElementById('fuse').innerText=JSON.stringify(fu).slice(0,400);
+ browse(cwd);
+}
+async function browse(p){
+ cwd=p;
+ const parts=p.split('/').filter(x=>x);let acc='';
+ let html='<a onclick="browse(String.fromCharCode(47))">/</a> ';
+ for(const x of parts){acc+='/'+x;html+=`<a onclick="browse('${acc}')">${x}</a> / `;}
+ document.getElementById('crumb').innerHTML=html;
+ const sts=await J('/api/browse?path='+encodeURIComponent(p));
+ const tb=document.querySelector('#browse tbody');tb.innerHTML='';
+ if(!sts)return;
+ for(const s of sts){const r=tb.insertRow();
+  const dir=s.file_type===1;
+  const name=dir?`<a onclick="browse('${s.path}')">${s.name}/</a>`:s.name;
+  r.innerHTML=`<td>${name}</td><td>${dir?'dir':(s.file_type===2?'link':'file')}</td>`+
+   `<td class=num>${dir?'':gib(s.length)}</td><td>${s.storage_tier||''}</td>`+
+   `<td class=num>${new Date(s.mtime_ms).toISOString().slice(0,19)}</td>`;}
+}
+refresh();setInterval(refresh,3000);
 </script></body></html>"""
 
 
