@@ -527,19 +527,19 @@ def main():
     p.add_argument("--path", choices=["fuse", "client"], default="fuse")
     p.add_argument("--workload", choices=["seqread", "randread4k"],
                    default="seqread")
-    p.add_argument("--files", type=int, default=8)
+    p.add_argument("--files", type=int, default=16)
     p.add_argument("--file-size", type=int, default=1 << 30)
     p.add_argument("--block-size", type=int, default=256 << 20)
-    p.add_argument("--read-chunk", type=int, default=1 << 20)
-    p.add_argument("--threads", type=int, default=16)
+    p.add_argument("--read-chunk", type=int, default=4 << 20)
+    p.add_argument("--threads", type=int, default=32)
     p.add_argument("--rand-reads", type=int, default=200_000)
     p.add_argument("--iodepth", type=int, default=64,
                    help="queue depth per thread for randread4k (fio iodepth)")
-    p.add_argument("--seq-batch", type=int, default=4,
+    p.add_argument("--seq-batch", type=int, default=1,
                    help="chunks per sync in seqread (DMA pipelining depth)")
     p.add_argument("--no-short-circuit", action="store_true",
                    help="force the streaming worker-RPC read path")
-    p.add_argument("--hbm-gb", type=int, default=16)
+    p.add_argument("--hbm-gb", type=int, default=32)
     p.add_argument("--staging-bytes", type=int, default=8 << 20)
     p.add_argument("--staging-count", type=int, default=8)
     p.add_argument("--fuse-channels", type=int, default=8)
