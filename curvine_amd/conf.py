@@ -215,6 +215,9 @@ class JobConf:
     task_chunk_size: int = 8 << 20
     store: str = "memory"  # memory | sqlite
     store_path: str = "/tmp/curvine/jobs.db"
+    # standalone transfer service "host:port" ("" = jobs run embedded in
+    # the master, the reference's other deployment shape)
+    service_addr: str = ""
 
 
 @dataclass

@@ -239,17 +239,39 @@ class MasterHandler:
         return {"mount": mi.to_dict() if mi else None}
 
     # ---------------- jobs ----------------
+    async def _forward_transfer(self, code, h):
+        """Standalone transfer-service deployment: the master proxies the
+        job surface to the external service (conf.job.service_addr)."""
+        conn = self.master.transfer_conn()
+        r = await conn.rpc(code, h)
+        return r.header
+
     def op_submitjob(self, h, d):
+        if self.master.conf.job.service_addr:
+            return self._forward_transfer(RpcCode.SubmitJob, h)
         return self.master.jobs.submit(h)
 
     def op_getjobstatus(self, h, d):
+        if self.master.conf.job.service_addr:
+            return self._forward_transfer(RpcCode.GetJobStatus, h)
         return self.master.jobs.status(h["job_id"])
 
     def op_canceljob(self, h, d):
+        if self.master.conf.job.service_addr:
+            return self._forward_transfer(RpcCode.CancelJob, h)
         return self.master.jobs.cancel(h["job_id"])
 
     def op_reporttask(self, h, d):
+        if self.master.conf.job.service_addr:
+            return self._forward_transfer(RpcCode.ReportTask, h)
         self.master.jobs.report_task(h)
+        return {}
+
+    def op_submittask(self, h, d):
+        """Transfer-service -> master: enqueue one worker command for
+        delivery on that worker's next heartbeat (scheduler.rs:43
+        dispatch analog)."""
+        self.fs.workers.add_command(h["worker_id"], h["command"])
         return {}
 
     # ---------------- worker plane ----------------

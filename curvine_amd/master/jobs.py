@@ -50,21 +50,38 @@ class JobStore:
 
 
 class JobManager:
-    def __init__(self, master):
+    """Runs embedded in the master (`master` set) OR inside the
+    standalone transfer service (`master=None`; the service supplies
+    mounts/workers over RPC and calls new_job/plan directly)."""
+
+    def __init__(self, master, conf=None):
         self.master = master
-        self.store = JobStore(master.conf.job.store, master.conf.job.store_path)
+        conf = conf if conf is not None else master.conf
+        self.store = JobStore(conf.job.store, conf.job.store_path)
         self.jobs: dict[str, dict] = self.store.load_all()
         self.next_id = max((int(j.rsplit("-", 1)[-1]) for j in self.jobs), default=0)
 
     def submit(self, h: dict) -> dict:
-        """h: {path, recursive, replicas}.  The path must be under a mount
-        (or an existing cache dir for re-replication loads)."""
+        """Embedded path. h: {path, recursive, replicas}; the path must
+        be under a mount (or an existing cache dir for re-replication
+        loads)."""
+        mount = self.master.mounts.lookup(h["path"])
+        job = self.new_job(h, mount)
+        worker_ids = [w.address.worker_id
+                      for w in self.master.fs.workers.live_workers()]
+        self.plan(job, mount, worker_ids)
+        for task in job["tasks"].values():
+            self.master.fs.workers.add_command(task["worker"], {
+                "cmd": "load_task", **task})
+        self.store.save(job)
+        return {"job_id": job["job_id"], "state": job["state"],
+                "total": job["total"]}
+
+    def new_job(self, h: dict, mount) -> dict:
         self.next_id += 1
         job_id = f"job-{self.next_id}"
-        path = h["path"]
-        mount = self.master.mounts.lookup(path)
         job = {
-            "job_id": job_id, "path": path, "state": "planning",
+            "job_id": job_id, "path": h["path"], "state": "planning",
             "submitted_ms": int(time.time() * 1000),
             "tasks": {}, "done": 0, "failed": 0, "total": 0,
             "mount": mount.to_dict() if mount else None,
@@ -72,46 +89,43 @@ class JobManager:
             "recursive": h.get("recursive", True),
         }
         self.jobs[job_id] = job
-        self._plan(job)
-        self.store.save(job)
-        return {"job_id": job_id, "state": job["state"], "total": job["total"]}
+        return job
 
-    def _plan(self, job: dict) -> None:
-        """List the UFS and emit one load task per file to a worker."""
-        mount = job["mount"]
+    def plan(self, job: dict, mount, worker_ids: list[int]) -> None:
+        """List the UFS and emit one load task per file round-robin over
+        the workers (transfer/planner.rs + scheduler.rs analog).  The
+        caller delivers each task to its worker."""
         if mount is None:
             job["state"] = "failed"
             job["error"] = f"path {job['path']} is not under a mount"
             return
+        m = mount.to_dict() if hasattr(mount, "to_dict") else mount
         from curvine_amd.ufs import get_ufs
         try:
-            ufs = get_ufs(mount["ufs_path"], mount.get("properties", {}))
-            rel = job["path"][len(mount["curvine_path"]):] or "/"
+            ufs = get_ufs(m["ufs_path"], m.get("properties", {}))
+            rel = job["path"][len(m["curvine_path"]):] or "/"
             files = ufs.list_files(rel, recursive=job["recursive"])
         except Exception as e:  # noqa: BLE001
             job["state"] = "failed"
             job["error"] = str(e)
             return
-        workers = self.master.fs.workers.live_workers()
-        if not workers:
+        if not worker_ids:
             job["state"] = "failed"
             job["error"] = "no live workers"
             return
         job["total"] = len(files)
         job["state"] = "running" if files else "completed"
         for i, f in enumerate(files):
-            w = workers[i % len(workers)]
+            wid = worker_ids[i % len(worker_ids)]
             task_id = f"{job['job_id']}-t{i}"
-            cv_path = mount["curvine_path"].rstrip("/") + f["path"]
-            task = {"task_id": task_id, "ufs_path": mount["ufs_path"],
+            cv_path = m["curvine_path"].rstrip("/") + f["path"]
+            task = {"task_id": task_id, "ufs_path": m["ufs_path"],
                     "ufs_rel": f["path"], "cv_path": cv_path,
                     "length": f["length"], "state": "assigned",
-                    "worker": w.address.worker_id,
-                    "properties": mount.get("properties", {}),
+                    "worker": wid,
+                    "properties": m.get("properties", {}),
                     "replicas": job["replicas"]}
             job["tasks"][task_id] = task
-            self.master.fs.workers.add_command(w.address.worker_id, {
-                "cmd": "load_task", **task})
 
     def status(self, job_id: str) -> dict:
         job = self.jobs.get(job_id)

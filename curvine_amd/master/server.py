@@ -134,6 +134,16 @@ class Master:
         self._actor_task: Optional[asyncio.Task] = None
         self._stopped = asyncio.Event()
         self._mutation_count = 0
+        self._transfer_conn = None
+
+    def transfer_conn(self):
+        """Pooled connection to the standalone transfer service."""
+        if self._transfer_conn is None:
+            from curvine_amd.rpc.client import ClusterConnector
+            self._transfer_conn = ClusterConnector(
+                [self.conf.job.service_addr],
+                self.conf.client.rpc_timeout_ms)
+        return self._transfer_conn
 
     def _make_raft(self):
         from curvine_amd.master.raft import RaftNode

@@ -32,6 +32,11 @@ async def run(args) -> None:
         if args.data_dir:
             conf.worker.data_dirs = args.data_dir
         servers.append(await Worker(conf, device_id=args.device).start())
+    if args.service == "transfer":
+        from curvine_amd.transfer import TransferService
+        svc = await TransferService(conf).start()
+        print(f"TRANSFER_PORT={svc.port}", flush=True)
+        servers.append(svc)
     if args.web:
         from curvine_amd.master.server import Master
         from curvine_amd.web.server import WebServer
@@ -52,7 +57,7 @@ async def run(args) -> None:
 
 def main(argv=None) -> int:
     p = argparse.ArgumentParser(prog="curvine-server")
-    p.add_argument("--service", choices=["master", "worker", "all"],
+    p.add_argument("--service", choices=["master", "worker", "all", "transfer"],
                    default="all")
     p.add_argument("--conf", default=None)
     p.add_argument("--device", type=int, default=-1)
