@@ -91,6 +91,7 @@ struct DataBlock {
   uint64_t aoff = 0;
   int64_t len = 0;
   std::string path;
+  bool direct = false;   // file kind: serve with O_DIRECT aligned preads
   std::atomic<int> refs{0};
   std::atomic<bool> dead{false};
 };
@@ -427,6 +428,47 @@ static bool data_serve_read(DataServer* S, DataConn* c, const uint8_t* frame,
       }
       bounce_release(bp);
     }
+  } else if (b->direct) {
+    // NVMe page-cache bypass: aligned O_DIRECT preads into an aligned
+    // bounce, sent with sendmsg (sendfile needs the page cache)
+    int fd = open(b->path.c_str(), O_RDONLY | O_DIRECT);
+    if (fd < 0) fd = open(b->path.c_str(), O_RDONLY);  // fs w/o DIO
+    if (fd < 0) {
+      ok = false;
+    } else {
+      constexpr int64_t kAlign = 4096;
+      size_t bsz = size_t(std::min<int64_t>(chunk + 2 * kAlign, 16 << 20));
+      void* abuf = nullptr;
+      if (posix_memalign(&abuf, kAlign, bsz) != 0) abuf = nullptr;
+      int64_t pos = 0;
+      while (abuf && pos < n && ok) {
+        int64_t want = std::min<int64_t>(chunk, n - pos);
+        int64_t fo = offset + pos;
+        int64_t lo = fo & ~(kAlign - 1);
+        int64_t span = fo + want - lo;
+        span = (span + kAlign - 1) & ~(kAlign - 1);
+        int64_t got = 0;
+        while (got < span) {
+          ssize_t r = pread(fd, (uint8_t*)abuf + got, size_t(span - got),
+                            off_t(lo + got));
+          if (r < 0 && errno == EINTR) continue;
+          if (r <= 0) break;   // EOF tail (unaligned file end)
+          got += r;
+        }
+        int64_t have = std::min<int64_t>(want, got - (fo - lo));
+        if (have < want) { ok = false; break; }
+        std::string ph = meta_proto(0, uint32_t(want), kCodeReadBlock,
+                                    uint8_t((2 << 4) | req_status), req_id,
+                                    seq);
+        ok = frame_send_locked(c->fd, ph,
+                               (const uint8_t*)abuf + (fo - lo),
+                               size_t(want));
+        pos += want;
+      }
+      if (!abuf) ok = false;
+      free(abuf);
+      close(fd);
+    }
   } else {
     int fd = open(b->path.c_str(), O_RDONLY);
     if (fd < 0) {
@@ -720,7 +762,8 @@ static void data_stop_srv(int64_t sid) {
 
 static void data_block_publish(int64_t sid, int64_t block_id, int kind,
                                int arena, uint64_t aoff, int64_t len,
-                               const std::string& path) {
+                               const std::string& path,
+                               bool direct = false) {
   DataServer* S = data_get(sid);
   auto b = std::make_shared<DataBlock>();
   b->kind = kind;
@@ -728,6 +771,7 @@ static void data_block_publish(int64_t sid, int64_t block_id, int kind,
   b->aoff = aoff;
   b->len = len;
   b->path = path;
+  b->direct = direct;
   std::unique_lock<std::shared_mutex> lk(S->blk_mu);
   S->blocks[block_id] = std::move(b);
 }

@@ -1190,7 +1190,10 @@ PYBIND11_MODULE(_native, m) {
   m.def("meta_take_access", &meta_take_access);
   m.def("data_create", &data_create);
   m.def("data_stop", &data_stop_srv);
-  m.def("data_block_publish", &data_block_publish);
+  m.def("data_block_publish", &data_block_publish, py::arg("sid"),
+        py::arg("block_id"), py::arg("kind"), py::arg("arena"),
+        py::arg("aoff"), py::arg("len"), py::arg("path"),
+        py::arg("direct") = false);
   m.def("data_block_drop", &data_block_drop);
   m.def("data_block_refs", &data_block_refs);
   m.def("data_write_register", &data_write_register);

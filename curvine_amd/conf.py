@@ -66,23 +66,33 @@ class DataDir:
     capacity: int = 1 << 30
     path: str = "/tmp/curvine/data"
     device_id: int = 0  # GPU ordinal for HBM dirs
+    # O_DIRECT reads on SSD/HDD dirs (NVMe page-cache bypass; the
+    # reference's SPDK-tier analog at the kernel-API level):
+    # "[SSD:1TB:direct]/nvme0"
+    o_direct: bool = False
 
     @staticmethod
     def parse(s: str) -> "DataDir":
-        """Parse "[MEM:30GB]/path" / "[HBM:200GB:0]label" / bare path."""
+        """Parse "[MEM:30GB]/path" / "[HBM:200GB:0]label" /
+        "[SSD:1TB:direct]/nvme" / bare path."""
         m = _DATA_DIR_RE.match(s)
         if not m:
             return DataDir(tier=TIER_SSD, capacity=0, path=s)
         tier = m.group(1).upper()
         if tier not in TIERS:
             raise ValueError(f"unknown storage tier in {s!r}")
-        cap, dev = 1 << 30, 0
+        cap, dev, direct = 1 << 30, 0, False
         if m.group(2):
             parts = m.group(2).split(":")
             cap = parse_bytes(parts[0])
-            if len(parts) > 1:
-                dev = int(parts[1])
-        return DataDir(tier=tier, capacity=cap, path=m.group(3) or f"/tmp/curvine/{tier.lower()}", device_id=dev)
+            for tok in parts[1:]:
+                if tok.lower() == "direct":
+                    direct = True
+                elif tok:
+                    dev = int(tok)
+        return DataDir(tier=tier, capacity=cap,
+                       path=m.group(3) or f"/tmp/curvine/{tier.lower()}",
+                       device_id=dev, o_direct=direct)
 
 
 @dataclass

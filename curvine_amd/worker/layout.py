@@ -206,16 +206,63 @@ class FileLayout(BlockLayout):
     WRITING = "writing"
     FINAL = "final"
 
+    ALIGN = 4096   # O_DIRECT sector/buffer alignment
+
     def __init__(self, data_dir: DataDir, dir_id: int):
         super().__init__(data_dir, dir_id)
         self.tier = data_dir.tier
         self.root = data_dir.path
+        # O_DIRECT reads (page-cache bypass on NVMe; writes stay buffered
+        # with fsync at finalize + DONTNEED so the cache stays clean —
+        # the read path is what a cache serves)
+        self.direct = data_dir.o_direct
         os.makedirs(os.path.join(self.root, self.WRITING), exist_ok=True)
         os.makedirs(os.path.join(self.root, self.FINAL), exist_ok=True)
+        if self.direct:
+            self.direct = self._probe_direct()
         import threading
         self._used = 0
         self._used_lock = threading.Lock()
         self._fds: dict[int, object] = {}
+
+    def _probe_direct(self) -> bool:
+        """tmpfs and some overlays reject O_DIRECT: probe once and fall
+        back to buffered reads with a warning instead of failing every
+        read."""
+        probe = os.path.join(self.root, ".direct_probe")
+        try:
+            with open(probe, "wb") as f:
+                f.write(b"\0" * self.ALIGN)
+            fd = os.open(probe, os.O_RDONLY | os.O_DIRECT)
+            os.close(fd)
+            return True
+        except OSError as e:
+            import logging
+            logging.getLogger("curvine.layout").warning(
+                "dir %s: O_DIRECT unsupported (%s); buffered reads",
+                self.root, e)
+            return False
+        finally:
+            try:
+                os.remove(probe)
+            except OSError:
+                pass
+
+    def _pread_direct(self, path: str, off: int, n: int) -> bytes:
+        """Aligned O_DIRECT pread covering [off, off+n) (bounce through
+        an mmap page-aligned buffer; short tail reads clamp at EOF)."""
+        import mmap
+        a = self.ALIGN
+        lo = off - (off % a)
+        span = off + n - lo
+        span = span + (-span % a)
+        fd = os.open(path, os.O_RDONLY | os.O_DIRECT)
+        try:
+            buf = mmap.mmap(-1, span)
+            got = os.preadv(fd, [buf], lo)
+            return buf[off - lo:min(off - lo + n, got)]
+        finally:
+            os.close(fd)
 
     def _path(self, block_id: int, state: str) -> str:
         sub = os.path.join(self.root, state, f"{block_id % 256:02x}")
@@ -253,6 +300,13 @@ class FileLayout(BlockLayout):
         if f:
             f.flush()
             os.fsync(f.fileno())
+            if self.direct:
+                # reads bypass the cache: drop the write-side pages now
+                try:
+                    os.posix_fadvise(f.fileno(), 0, 0,
+                                     os.POSIX_FADV_DONTNEED)
+                except OSError:
+                    pass
             f.close()
         final = self._path(meta["block_id"], self.FINAL)
         os.replace(meta["path"], final)
@@ -312,11 +366,17 @@ class FileLayout(BlockLayout):
         raise err.Unsupported("ptr write to file layout")
 
     def _read_at(self, meta, off, n):
+        if self.direct:
+            return self._pread_direct(meta["path"], off, n)
         with open(meta["path"], "rb") as f:
             f.seek(off)
             return f.read(n)
 
     def _read_into(self, meta, off, out, out_off, n):
+        if self.direct:
+            data = self._pread_direct(meta["path"], off, n)
+            out[out_off:out_off + len(data)] = data
+            return
         with open(meta["path"], "rb") as f:
             f.seek(off)
             mv = memoryview(out)[out_off:out_off + n]

@@ -42,7 +42,7 @@ class DataPlane:
             else:
                 self.lib.data_block_publish(
                     self.sid, block_id, 1, -1, 0, meta["length"],
-                    meta["path"])
+                    meta["path"], getattr(layout, "direct", False))
         except Exception as e:  # noqa: BLE001 — registry is a cache
             log.warning("publish block %d: %s", block_id, e)
 

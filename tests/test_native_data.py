@@ -177,3 +177,56 @@ def test_delete_deferred_until_native_readers_drain(tmp_path):
             assert st["published_blocks"] == 0
             await fs.close()
     _run(main())
+
+
+def test_direct_io_file_tier(tmp_path):
+    """[SSD:...:direct] dir: O_DIRECT reads on both the short-circuit
+    (FileLayout aligned bounce) and the native remote plane (aligned
+    pread + sendmsg instead of sendfile), including unaligned offsets
+    and tails."""
+    async def main():
+        conf = _remote_conf(tmp_path)
+        conf.worker.data_dirs = [f"[SSD:256MB:direct]{tmp_path}/nvme"]
+        conf.client.storage_tier = "SSD"
+        async with MiniCluster(conf=conf, tmp_dir=str(tmp_path)) as mc:
+            layout = mc.workers[0].store.layouts[0]
+            assert layout.direct, "O_DIRECT probe failed on this fs"
+            fe = _native_frontend(mc.workers[0])
+            fs = mc.fs()
+            data = os.urandom((5 << 20) + 777)   # unaligned tail
+            await fs.write_all("/dio/a.bin", data, storage_tier="SSD")
+            # remote plane (native O_DIRECT serve)
+            back = await fs.read_all("/dio/a.bin")
+            assert back == data
+            st = fe.stats()
+            assert st["served_reads"] >= 1, st
+            # unaligned ranged reads through the short-circuit reader
+            # (file spans two 4 MiB blocks; ranges stay within block 0)
+            info = await fs.client.open("/dio/a.bin")
+            store = mc.workers[0].store
+            r = store.open_reader(info.blocks[0].block.block_id)
+            try:
+                assert r.read(0, 10) == data[:10]
+                assert r.read(4095, 4098) == data[4095:4095 + 4098]
+                assert r.read(1234567, 54321) == \
+                    data[1234567:1234567 + 54321]
+            finally:
+                r.close()
+            # block 1 carries the unaligned 777-byte file tail
+            b1 = info.blocks[1]
+            r = store.open_reader(b1.block.block_id)
+            try:
+                assert r.length == b1.block.length
+                assert r.read(r.length - 100, 200) == \
+                    data[b1.offset + r.length - 100:
+                         b1.offset + r.length]
+            finally:
+                r.close()
+            # remote ranged read at an unaligned offset
+            from curvine_amd.client.block_client import BlockReaderRemote
+            br = BlockReaderRemote(info.blocks[0].locations[0],
+                                   info.blocks[0].block.block_id)
+            got = await br.read(4097, 100_003)
+            assert got == data[4097:4097 + 100_003]
+            await fs.close()
+    _run(main())
