@@ -27,6 +27,14 @@ def get_ufs(uri: str, properties: dict | None = None) -> UnderFs:
         return _CACHE[key]
     if uri.startswith("s3://") or uri.startswith("oss://"):
         fs: UnderFs = S3Ufs(uri, properties)
+    elif uri.startswith("gs://"):
+        # GCS speaks the S3 XML protocol in interoperability mode (HMAC
+        # keys sign with SigV4 exactly like S3); default the endpoint
+        fs = S3Ufs(uri, {"endpoint": "https://storage.googleapis.com",
+                         **properties})
+    elif uri.startswith("az://") or uri.startswith("abfs://"):
+        from curvine_amd.ufs.azure import AzureUfs
+        fs = AzureUfs(uri, properties)
     elif uri.startswith("hdfs://") or uri.startswith("webhdfs://"):
         from curvine_amd.ufs.webhdfs import WebHdfsUfs
         fs = WebHdfsUfs(uri, properties)
