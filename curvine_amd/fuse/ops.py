@@ -394,6 +394,9 @@ class CurvineFuseFs:
         # POSIX advisory locks: (node_id) -> list of (start, end, type, owner, pid)
         self.plocks: dict[int, list] = {}
         self.plock_mu = threading.Lock()
+        # kernel INTERRUPT targets: uniques of in-flight requests the
+        # kernel asked to abort (consumed by blocking SETLKW polls)
+        self.interrupted: set[int] = set()
 
     # ---------------- helpers ----------------
     def call(self, coro, timeout: float = 120.0):
@@ -1149,6 +1152,14 @@ class CurvineFuseFs:
         raise OSError(errno.EINVAL, "whence")
 
     def op_interrupt(self, nodeid, body, ctx):
+        """INTERRUPT carries the unique of an in-flight request; blocked
+        SETLKW waiters poll self.interrupted and abort with EINTR."""
+        import struct as _st
+        (unique,) = _st.unpack_from("<Q", body, 0)
+        with self.plock_mu:
+            self.interrupted.add(unique)
+            if len(self.interrupted) > 4096:
+                self.interrupted.clear()   # stale uniques: reset
         return None
 
     # ---------------- POSIX advisory locks ----------------
@@ -1174,7 +1185,8 @@ class CurvineFuseFs:
     def op_setlk(self, nodeid, body, ctx, wait=False):
         fh, owner, start, end, ltype, pid, _fl, _ = abi.LK_IN.unpack_from(body, 0)
         end = end or (1 << 63)
-        deadline = time.monotonic() + 30 if wait else 0
+        unique = ctx[3] if len(ctx) > 3 else 0
+        deadline = time.monotonic() + 300 if wait else 0
         while True:
             with self.plock_mu:
                 if ltype == self.F_UNLCK:
@@ -1188,12 +1200,37 @@ class CurvineFuseFs:
                     self.plocks.setdefault(nodeid, []).append(
                         (start, end, ltype, owner, pid))
                     return b""
+                if unique and unique in self.interrupted:
+                    # the kernel INTERRUPTed this request (signal on the
+                    # blocked thread): abort the wait
+                    self.interrupted.discard(unique)
+                    raise OSError(errno.EINTR, "lock wait interrupted")
             if not wait or time.monotonic() > deadline:
                 raise OSError(errno.EAGAIN, "lock conflict")
             time.sleep(0.02)
 
     def op_setlkw(self, nodeid, body, ctx):
-        return self.op_setlk(nodeid, body, ctx, wait=True)
+        """Blocking lock waits run on their own thread (reply deferred):
+        a wait parked on the channel thread would wedge that channel —
+        including the INTERRUPT that is supposed to abort the wait."""
+        ch = ctx[4] if len(ctx) > 4 else None
+        unique = ctx[3] if len(ctx) > 3 else 0
+        if ch is None:
+            return self.op_setlk(nodeid, body, ctx, wait=True)
+        body_copy = bytes(body)
+
+        def waiter():
+            try:
+                res = self.op_setlk(nodeid, body_copy, ctx, wait=True)
+                ch.reply(unique, res)
+            except OSError as e:
+                ch.reply_error(unique, e.errno or errno.EIO)
+            except Exception:  # noqa: BLE001
+                ch.reply_error(unique, errno.EIO)
+
+        threading.Thread(target=waiter, daemon=True,
+                         name="fuse-setlkw").start()
+        return None
 
     # ---------------- hot-upgrade state persist/restore ----------------
     # (fs/file_system.rs:219-223 + state/backend_handle.rs:288-310 +

@@ -316,3 +316,67 @@ def test_concurrent_write_open_same_file(mount):
     for slot, data in slots.items():
         assert content[slot * (64 << 10):(slot + 1) * (64 << 10)] == data, \
             f"slot {slot} mismatch"
+
+
+def test_setlkw_interrupted_by_signal(mount):
+    """A blocked F_SETLKW whose process is signalled must be aborted in
+    the DAEMON too (kernel INTERRUPT -> op_interrupt -> waiter thread
+    exits with EINTR).  Without that, a ghost waiter thread grabs the
+    lock for the dead process when the holder releases, and the file
+    stays locked forever."""
+    import fcntl
+    import time as _time
+    mnt, *_ = mount
+    p = f"{mnt}/lk.bin"
+    with open(p, "wb") as f:
+        f.write(b"x" * 100)
+    hold = (
+        "import fcntl, sys, time\n"
+        f"f = open({p!r}, 'r+b')\n"
+        "fcntl.lockf(f, fcntl.LOCK_EX)\n"
+        "print('LOCKED', flush=True)\n"
+        "sys.stdin.readline()\n"          # release on demand
+        "fcntl.lockf(f, fcntl.LOCK_UN)\n"
+        "print('RELEASED', flush=True)\n"
+        "time.sleep(5)\n")
+    holder = subprocess.Popen([sys.executable, "-c", hold],
+                              stdin=subprocess.PIPE,
+                              stdout=subprocess.PIPE, text=True)
+    try:
+        assert holder.stdout.readline().strip() == "LOCKED"
+        # waiter blocks in F_SETLKW; SIGALRM handler raises -> process
+        # exits mid-wait; the kernel INTERRUPTs the in-flight request
+        wprog = (
+            "import fcntl, signal, sys\n"
+            "def h(*a): raise KeyboardInterrupt\n"
+            "signal.signal(signal.SIGALRM, h)\n"
+            f"f = open({p!r}, 'r+b')\n"
+            "signal.alarm(1)\n"
+            "try:\n"
+            "    fcntl.lockf(f, fcntl.LOCK_EX)\n"
+            "    print('ACQUIRED', flush=True)\n"
+            "except KeyboardInterrupt:\n"
+            "    print('INTERRUPTED', flush=True)\n")
+        waiter = subprocess.run([sys.executable, "-c", wprog],
+                                capture_output=True, text=True, timeout=20)
+        assert waiter.stdout.strip() == "INTERRUPTED", \
+            (waiter.stdout, waiter.stderr)
+        _time.sleep(0.3)   # let the INTERRUPT reach the daemon
+        holder.stdin.write("go\n")
+        holder.stdin.flush()
+        assert holder.stdout.readline().strip() == "RELEASED"
+        # no ghost: a fresh exclusive lock must succeed promptly
+        probe = (
+            "import fcntl, time\n"
+            f"f = open({p!r}, 'r+b')\n"
+            "t0 = time.monotonic()\n"
+            "fcntl.lockf(f, fcntl.LOCK_EX)\n"
+            "print('OK', round(time.monotonic()-t0, 2), flush=True)\n")
+        r = subprocess.run([sys.executable, "-c", probe],
+                           capture_output=True, text=True, timeout=15)
+        out = r.stdout.split()
+        assert out and out[0] == "OK", (r.stdout, r.stderr)
+        assert float(out[1]) < 5, f"ghost waiter held the lock: {out}"
+    finally:
+        holder.terminate()
+        holder.wait(timeout=10)
