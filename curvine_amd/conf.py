@@ -160,6 +160,11 @@ class WorkerConf:
     staging_buf_bytes: int = 4 << 20
     staging_buf_count: int = 8
     replication_concurrency: int = 4
+    # fsync block files at finalize.  Off by default: this is a CACHE —
+    # a crash loses re-fetchable blocks, and the publish-time CRC catches
+    # torn files at read/verify time.  Turn on for cache-as-of-record
+    # deployments.
+    fsync_on_finalize: bool = False
     # native C++ epoll data plane (csrc/data_server.cpp); falls back to
     # the asyncio RpcServer when the extension is unavailable
     native_data: bool = True

@@ -45,8 +45,10 @@ class BlockStore:
         self.conf = conf
         self.layouts: list[BlockLayout] = []
         for i, dd in enumerate(conf.parsed_dirs()):
-            self.layouts.append(make_layout(dd, i, conf.staging_buf_bytes,
-                                            conf.staging_buf_count))
+            layout = make_layout(dd, i, conf.staging_buf_bytes,
+                                 conf.staging_buf_count)
+            layout.fsync_on_finalize = conf.fsync_on_finalize
+            self.layouts.append(layout)
         # hottest-first iteration order
         self.layouts.sort(key=lambda l: TIER_ORDER.get(l.tier, 9))
         self.blocks: dict[int, _Block] = {}

@@ -295,11 +295,14 @@ class FileLayout(BlockLayout):
                 "path": path, "block_id": block_id, "reserved": reserve,
                 "length": 0, "_f": f}
 
+    fsync_on_finalize = False   # set from WorkerConf by BlockStore
+
     def finalize(self, meta: dict, length: int) -> None:
         f = meta.pop("_f", None)
         if f:
             f.flush()
-            os.fsync(f.fileno())
+            if self.fsync_on_finalize:
+                os.fsync(f.fileno())
             if self.direct:
                 # reads bypass the cache: drop the write-side pages now
                 try:
