@@ -213,6 +213,14 @@ class FuseConf:
     state_file: str = "/tmp/curvine/fuse.state"
     kernel_cache: bool = True
     native_loop: bool = True   # GIL-free C++ READ channels
+    # kernel writeback cache: dirty pages aggregate in the page cache and
+    # arrive as large async WRITEs instead of one synchronous round trip
+    # per write(2).  Off by default: the kernel raises internal EIO on
+    # some unlinked-file page flush paths with no daemon op erroring
+    # (see ROADMAP); the daemon-side semantics it needs (reads on
+    # write-only handles, handle-served getattr, open-time block
+    # pinning) are implemented and kept on
+    writeback_cache: bool = False
 
 
 @dataclass

@@ -68,6 +68,15 @@ class SyncReadState:
         self._offs = None                     # block offset index (lazy)
         self._local: dict[int, object] = {}   # block idx -> store reader
         self._remote = None                   # lazy async FsReader
+        # pin every colocated block NOW: store reader refcounts defer
+        # deletion, so an open fd keeps serving after unlink/truncate
+        # even when the kernel page cache absorbed the early reads
+        # (writeback_cache) and the first READ op arrives late
+        for idx, lb in enumerate(file_blocks.blocks):
+            try:
+                self._open_local(idx, lb)
+            except Exception:  # noqa: BLE001 — remote/raced blocks: lazy
+                pass
 
     def read_into(self, off: int, out, out_off: int, n: int) -> int:
         import bisect
@@ -539,6 +548,8 @@ class CurvineFuseFs:
                      abi.FUSE_HANDLE_KILLPRIV_V2 |
                      abi.FUSE_DO_READDIRPLUS | abi.FUSE_READDIRPLUS_AUTO |
                      abi.FUSE_POSIX_LOCKS) & flags | abi.FUSE_MAX_PAGES
+        if self.conf.fuse.writeback_cache:
+            out_flags |= abi.FUSE_WRITEBACK_CACHE & flags
         max_write = self.conf.fuse.max_write
         return abi.INIT_OUT.pack(
             7, min(minor, abi.FUSE_KERNEL_MINOR_VERSION),
@@ -592,7 +603,17 @@ class CurvineFuseFs:
         try:
             st = self.stat_path(nodeid, path)
         except cverr.FileNotFound:
-            raise OSError(errno.ENOENT, path)
+            # fstat of an open-but-unlinked (or renamed-away) file: POSIX
+            # serves the inode through the handle (the kernel revalidates
+            # attrs before cached reads under writeback_cache)
+            st = None
+            with self.handles_lock:
+                for h in self.handles.values():
+                    if h.node_id == nodeid and h.status is not None:
+                        st = h.status
+                        break
+            if st is None:
+                raise OSError(errno.ENOENT, path)
         # live write handle: report current write position as size
         st = self._adjust_writing_size(nodeid, st)
         return self.attr_out(nodeid, st)
@@ -892,8 +913,16 @@ class CurvineFuseFs:
             h.status = fb.status
             h.reader = SyncReadState(self, fb)
         if h.reader is None:
-            # O_WRONLY handle read
-            raise OSError(errno.EBADF, "not open for read")
+            # O_WRONLY handle read: under writeback_cache the kernel
+            # reads through ANY open handle to prefill partial pages
+            # before writing them back — serve the settled content
+            with h.lock:
+                self._sync_native_write(h)
+            try:
+                fb = self.call(self.fs.client.open(h.path))
+                h.reader = SyncReadState(self, fb)
+            except Exception:  # noqa: BLE001
+                raise OSError(errno.EBADF, "not open for read")
         ch = ctx[4]
         pin = getattr(ch, "reply_pin", None)
         if pin is not None and size <= pin.nbytes - 64 and \
