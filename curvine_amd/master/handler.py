@@ -85,6 +85,15 @@ class MasterHandler:
         auditing = self.fs.conf.master.audit_log
         t0 = time.perf_counter() if auditing else 0.0
         try:
+            if pbuf:
+                # reference peers expect errors in the DATA section in
+                # the ErrorEncoder layout, not a msgpack header
+                try:
+                    return await self._handle_pbuf(msg, fn, raft, rkey)
+                finally:
+                    if auditing:
+                        audit.info("cmd=%s used_us=%d", code.name,
+                                   int((time.perf_counter() - t0) * 1e6))
             op_before = self.master.journal.op_id
             reply = fn(msg.header, msg.data)
             if asyncio.iscoroutine(reply):
@@ -107,6 +116,30 @@ class MasterHandler:
             if auditing:
                 audit.info("cmd=%s used_us=%d", code.name,
                            int((time.perf_counter() - t0) * 1e6))
+
+    async def _handle_pbuf(self, msg, fn, raft, rkey):
+        from curvine_amd.rpc import proto as _proto
+        from curvine_amd.rpc.message import Status
+        try:
+            op_before = self.master.journal.op_id
+            reply = fn(msg.header, msg.data)
+            if asyncio.iscoroutine(reply):
+                reply = await reply
+            if raft is not None and raft.is_leader \
+                    and self.master.journal.op_id > op_before:
+                await raft.wait_commit(self.master.journal.op_id)
+        except Exception as e:  # noqa: BLE001 — reference error wire
+            out = msg.reply(resp_status=Status.Error)
+            out.data = _proto.encode_error(e)
+            return out
+        if rkey is not None:
+            self.master.retry_cache.put(rkey, reply or {})
+        out = msg.reply(reply or {})
+        enc = _proto.encode_response(msg.code, out.header)
+        if enc is not None:
+            out.header = {}
+            out.raw_header = enc
+        return out
 
     # ---------------- filesystem ----------------
     def op_heartbeat(self, h, d):

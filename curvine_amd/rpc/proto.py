@@ -656,6 +656,50 @@ CODECS: dict[int, tuple] = {
 }
 
 
+# ---------------------------------------------------------- error frames
+# The reference carries errors in the DATA section (rpc_message.rs:193-197
+# error_ext), encoded by its ErrorEncoder (error_encoder.rs:24-50):
+#   i32 kind BE | u32 msg_len BE | msg utf8 | u32 data_len BE | data
+# with kind from curvine-error ErrorKind (fs_error.rs:38-79).  Our typed
+# errors map onto those kinds; unmapped ones ride as Common (10000).
+
+_ERROR_KINDS = {
+    "NotLeader": 2, "RpcTimeout": 4,
+    "FileAlreadyExists": 7, "FileNotFound": 8,
+    "ParentNotDir": 10, "DirNotEmpty": 11,
+    "ChecksumMismatch": 12, "BlockInWriting": 13,
+    "InvalidPath": 16, "CapacityExceeded": 17,
+    "Unsupported": 19, "UfsError": 20, "Expired": 21,
+    "JobNotFound": 23, "IsDirectory": 26, "NotDirectory": 27,
+    "InvalidArgument": 28, "BlockNotFound": 29,
+    "NoAvailableWorker": 31,
+}
+_KIND_NAMES = {v: k for k, v in _ERROR_KINDS.items()}
+
+
+def encode_error(e: Exception) -> bytes:
+    import struct as _st
+    kind = _ERROR_KINDS.get(type(e).__name__, 10000)
+    msg = str(e).encode()
+    return _st.pack(">i", kind) + _st.pack(">I", len(msg)) + msg + \
+        _st.pack(">I", 0)
+
+
+def decode_error(raw: bytes):
+    """Parse a reference error blob -> (our typed exception)."""
+    import struct as _st
+
+    from curvine_amd import errors as _err
+    if len(raw) < 8:
+        return _err.FsError(raw.decode(errors="replace") or "rpc error")
+    (kind,) = _st.unpack_from(">i", raw, 0)
+    (n,) = _st.unpack_from(">I", raw, 4)
+    msg = raw[8:8 + n].decode(errors="replace")
+    name = _KIND_NAMES.get(kind)
+    cls = getattr(_err, name, None) if name else None
+    return (cls or _err.FsError)(msg)
+
+
 def decode_request(code: int, raw: bytes) -> Optional[dict]:
     """Parse a protobuf request header into the internal dict, or None
     when the code has no protobuf codec / the bytes don't parse."""

@@ -47,6 +47,27 @@ class WorkerHandler:
             msg.header = decoded
             msg.raw_header = b""
             conn.state["pbuf"] = True
+        if conn.state.get("pbuf"):
+            # reference peers: errors ride the DATA section in the
+            # ErrorEncoder layout; replies get protobuf headers
+            from curvine_amd.rpc import proto as _proto
+            from curvine_amd.rpc.message import Status
+            try:
+                out = await self._handle_inner(msg, conn)
+            except Exception as e:  # noqa: BLE001
+                out = msg.reply(resp_status=Status.Error)
+                out.data = _proto.encode_error(e)
+                return out
+            if out is not None and out.header:
+                enc = _proto.encode_response(code, out.header)
+                if enc is not None:
+                    out.header = {}
+                    out.raw_header = enc
+            return out
+        return await self._handle_inner(msg, conn)
+
+    async def _handle_inner(self, msg: Message, conn) -> Optional[Message]:
+        code = msg.code
         if code == int(RpcCode.WriteBlock):
             return await self._write_block(msg)
         if code == int(RpcCode.ReadBlock):

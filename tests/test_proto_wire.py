@@ -91,6 +91,20 @@ def test_status_response_roundtrip():
     assert back["xattrs"] == {"user.k": b"v"}
 
 
+def test_error_wire_golden():
+    """ErrorEncoder layout (error_encoder.rs:24-50): i32 kind BE,
+    u32 len BE, msg, u32 data_len BE — FileNotFound kind is 8
+    (fs_error.rs:47)."""
+    from curvine_amd import errors as cverr
+    raw = proto.encode_error(cverr.FileNotFound("/x"))
+    assert raw == bytes.fromhex("00000008" "00000002" "2f78" "00000000"), \
+        raw.hex()
+    back = proto.decode_error(raw)
+    assert isinstance(back, cverr.FileNotFound) and str(back) == "/x"
+    raw2 = proto.encode_error(RuntimeError("boom"))
+    assert raw2[:4] == (10000).to_bytes(4, "big")
+
+
 def test_hbm_tier_maps_to_mem_on_wire():
     st = {"inode_id": 1, "path": "/f", "name": "f", "file_type": 0,
           "length": 0, "is_complete": True, "block_size": 1, "replicas": 1,
@@ -123,7 +137,9 @@ class _PbClient:
         body = await self.r.readexactly(hlen + dlen)
         assert req_id == self.req_id
         if (status >> 4) == 5:
-            raise RuntimeError(f"error reply: {body!r}")
+            # reference error wire: DATA section, ErrorEncoder layout
+            assert hlen == 0, "error reply must not carry a header"
+            raise proto.decode_error(body)
         out = proto.decode_response(int(code), body[:hlen])
         assert out is not None, "reply did not parse as protobuf"
         return out
@@ -167,6 +183,13 @@ def test_protobuf_client_against_live_master(tmp_path):
                                           "recursive": False})
             assert (await c.call(RpcCode.Exists,
                                  {"path": "/pb/g"}))["exists"] is False
+            # typed errors cross the wire in the reference binary layout
+            from curvine_amd import errors as cverr
+            try:
+                await c.call(RpcCode.FileStatus, {"path": "/pb/missing"})
+                raise AssertionError("expected FileNotFound")
+            except cverr.FileNotFound as e:
+                assert "missing" in str(e)
             w.close()
             await w.wait_closed()
 
