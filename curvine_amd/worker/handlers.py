@@ -162,8 +162,16 @@ class WorkerHandler:
             length = min(h.get("length", reader.length),
                          reader.length - offset)
             chunk = h.get("chunk_size", 1 << 20)
-            await conn.send(msg.reply({"length": reader.length},
-                                      resp_status=Status.Running))
+            ack = msg.reply({"length": reader.length},
+                            resp_status=Status.Running)
+            if conn.state.get("pbuf"):
+                # reference peers: BlockReadResponse protobuf open-ack
+                from curvine_amd.rpc import proto as _proto
+                enc = _proto.encode_response(msg.code, ack.header)
+                if enc is not None:
+                    ack.header = {}
+                    ack.raw_header = enc
+            await conn.send(ack)
             pos = 0
             t0 = time.perf_counter()
             # read wide spans per executor hop (the thread handoff costs

@@ -202,3 +202,52 @@ def test_protobuf_client_against_live_master(tmp_path):
         sf.shutdown()
     finally:
         smc.stop()
+
+
+def test_protobuf_block_read_stream(tmp_path):
+    """A protobuf peer streams a block read from the worker: protobuf
+    BlockReadRequest open, protobuf BlockReadResponse ack, raw data
+    frames, Complete."""
+    import asyncio
+    import os
+
+    from curvine_amd.testing import MiniCluster
+
+    async def main():
+        async with MiniCluster(tmp_dir=str(tmp_path / "cv")) as mc:
+            fs = mc.fs()
+            data = os.urandom(2 << 20)
+            await fs.write_all("/pbw/f.bin", data)
+            info = await fs.client.open("/pbw/f.bin")
+            lb = info.blocks[0]
+            r, w = await asyncio.open_connection(
+                "127.0.0.1", lb.locations[0].rpc_port)
+            raw = proto.encode_request(int(RpcCode.ReadBlock), {
+                "block_id": lb.block.block_id, "offset": 0,
+                "length": lb.block.length, "chunk_size": 1 << 20})
+            w.write(_HDR.pack(len(raw), 0, int(RpcCode.ReadBlock),
+                              1, 4242, 0) + raw)
+            await w.drain()
+            got = b""
+            acked = False
+            while True:
+                hdr = await r.readexactly(22)
+                hlen, dlen, code, status, req_id, seq = _HDR.unpack(hdr)
+                body = await r.readexactly(hlen + dlen)
+                rs = status >> 4
+                assert rs != 5, body
+                if hlen and not acked:
+                    out = proto.decode_response(int(RpcCode.ReadBlock),
+                                                body[:hlen])
+                    assert out and out["length"] == lb.block.length
+                    acked = True
+                got += body[hlen:]
+                if rs == 3:
+                    break
+            assert acked, "no protobuf open-ack"
+            assert got == data[:lb.block.length]
+            w.close()
+            await w.wait_closed()
+            await fs.close()
+
+    asyncio.new_event_loop().run_until_complete(main())
