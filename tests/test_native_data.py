@@ -230,3 +230,44 @@ def test_direct_io_file_tier(tmp_path):
             assert got == data[4097:4097 + 100_003]
             await fs.close()
     _run(main())
+
+
+@pytest.mark.gpu
+def test_remote_read_write_native_hbm_tier(tmp_path):
+    """GPU: the native data plane serves HBM-tier blocks over the wire —
+    double-buffered pinned D2H on reads, H2D consumption on writes —
+    byte-exact against the source."""
+    from curvine_amd import native
+    if not native.gpu_available():
+        pytest.skip("no GPU")
+
+    async def main():
+        conf = _remote_conf(tmp_path)
+        conf.worker.data_dirs = [f"[HBM:1GB:0]gpu0"]
+        conf.client.storage_tier = "HBM"
+        async with MiniCluster(conf=conf, tmp_dir=str(tmp_path)) as mc:
+            fe = _native_frontend(mc.workers[0])
+            fs = mc.fs()
+            data = os.urandom((48 << 20) + 12345)
+            await fs.write_all("/hbm/a.bin", data, storage_tier="HBM")
+            back = await fs.read_all("/hbm/a.bin")
+            assert back == data
+            st = fe.stats()
+            assert st["served_reads"] >= 1, st
+            assert st["served_read_bytes"] >= len(data), st
+            # explicit remote write stream lands in HBM and reads back
+            from curvine_amd.client.block_client import (BlockReaderRemote,
+                                                         BlockWriterRemote)
+            addr = mc.workers[0].address()
+            wdata = os.urandom(16 << 20)
+            w = BlockWriterRemote(addr, 777003, len(wdata), "HBM")
+            await w.write(wdata[:8 << 20])
+            await w.write(wdata[8 << 20:])
+            assert await w.commit(len(wdata)) == "HBM"
+            r = BlockReaderRemote(addr, 777003)
+            assert await r.read(0, len(wdata)) == wdata
+            # ranged unaligned read through the bounce pipeline
+            assert await r.read(1234567, 4 << 20) == \
+                wdata[1234567:1234567 + (4 << 20)]
+            await fs.close()
+    _run(main())
