@@ -643,6 +643,11 @@ static void data_readable(DataServer* S, const std::shared_ptr<DataConn>& c) {
       std::memmove(c->rbuf.data(), c->rbuf.data() + off, c->rlen - off);
       c->rlen -= off;
     }
+    if (c->rlen == 0 && c->rbuf.capacity() > (1u << 20)) {
+      // a 16 MiB write frame would otherwise pin its buffer per conn
+      c->rbuf.clear();
+      c->rbuf.shrink_to_fit();
+    }
     if (c->dead.load()) {
       data_close_conn(S, c);
       return;

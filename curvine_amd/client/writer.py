@@ -31,6 +31,8 @@ class FsWriter:
         self._block_addrs: list[tuple] = []   # (block_id, locations, tiers)
         self._rw: dict[int, list] = {}        # rewrite writers per block idx
         self._commits: list[dict] = []   # block locations to report on complete
+        self._exclude: set[int] = set()  # workers that failed block opens
+        self._abandoned = 0
         self.pos = 0
         self._closed = False
         # end-to-end write integrity: keep a running CRC32C per block and
@@ -65,7 +67,27 @@ class FsWriter:
             take = min(room, len(buf) - off)
             chunk = bytes(buf[off:off + take])
             # star fan-out: all replicas in parallel
-            await asyncio.gather(*[w.write(chunk) for w in self._writers])
+            results = await asyncio.gather(
+                *[w.write(chunk) for w in self._writers],
+                return_exceptions=True)
+            failed = [(self._block.locations[i].worker_id, r)
+                      for i, r in enumerate(results)
+                      if isinstance(r, BaseException)]
+            if failed:
+                retriable = all(isinstance(r, (ConnectionError, OSError,
+                                               err.ConnectError,
+                                               err.RpcTimeout))
+                                for _, r in failed)
+                if self._block_pos != 0 or not retriable:
+                    raise failed[0][1]  # bytes already placed / hard error
+                # replica target(s) died before any byte landed: abandon
+                # this block (stays 0-length in metadata — readers skip
+                # zero-length blocks) and re-place excluding only the
+                # failed workers (AddBlockRequest.exclude_workers flow,
+                # block_writer.rs replacement-worker analog)
+                await self._abandon_block(
+                    failed[0][1], [wid for wid, _ in failed])
+                continue
             if self._crc_on:
                 from curvine_amd import native
                 self._blk_crc = native.crc32c(chunk, self._blk_crc)
@@ -74,8 +96,28 @@ class FsWriter:
             if self._block_pos >= self.block_size:
                 await self._commit_block()
 
+    async def _abandon_block(self, exc: BaseException,
+                             failed_ids: list[int]) -> None:
+        for w in self._writers:
+            try:
+                await w.abort()
+            except Exception:  # noqa: BLE001
+                pass
+        self._exclude.update(failed_ids)
+        self._abandoned += 1
+        if self._abandoned > 3:
+            raise exc
+        self._block_lens.append(0)
+        self._commits.append({"block_id": self._block.block.block_id,
+                              "locations": [], "tiers": []})
+        self._block_addrs.append((self._block.block.block_id, [], []))
+        self._block = None
+        self._writers = []
+        self._block_pos = 0
+
     async def _next_block(self) -> None:
-        lb = await self.client.add_block(self.path)
+        lb = await self.client.add_block(
+            self.path, exclude_workers=sorted(self._exclude) or None)
         self._block = lb
         self._block_pos = 0
         if not lb.locations:
