@@ -390,6 +390,16 @@ static bool data_serve_read(DataServer* S, DataConn* c, const uint8_t* frame,
       hipStream_t s = thread_stream(a->device);
       size_t bsz = std::min<int64_t>(chunk, kBounceSz);
       BouncePair* bp = bounce_acquire(a->device);
+      struct BounceGuard {
+        BouncePair* b;
+        hipStream_t s;
+        ~BounceGuard() {
+          // an exception can leave an async D2H in flight targeting the
+          // pinned buffers: drain before the pair goes back to the pool
+          hipStreamSynchronize(s);
+          bounce_release(b);
+        }
+      } bg{bp, s};
       void** pin = bp->pin;
       hipEvent_t* ev = bp->ev;
       const uint8_t* src = (const uint8_t*)a->base + b->aoff + offset;
@@ -426,7 +436,6 @@ static bool data_serve_read(DataServer* S, DataConn* c, const uint8_t* frame,
         ok = frame_send_locked(c->fd, ph, (const uint8_t*)pin[last],
                                size_t(pn));
       }
-      bounce_release(bp);
     }
   } else if (b->direct) {
     // NVMe page-cache bypass: aligned O_DIRECT preads into an aligned
@@ -533,7 +542,11 @@ static bool data_serve_write(DataServer* S, DataConn* c, const uint8_t* frame,
         ssize_t w = pwrite(ws->fd, p, left, fo);
         if (w < 0) {
           if (errno == EINTR) continue;
-          return false;  // forward: Python will surface the error
+          // bytes may already be placed: forwarding the frame to Python
+          // would re-write it at a stale position.  Fail the stream —
+          // the client aborts and re-places the block elsewhere.
+          c->dead.store(true);
+          return true;
         }
         p += w;
         fo += w;

@@ -14,6 +14,10 @@
 // lib_fs_writer.rs): int64 handles for filesystem, reader, writer; every
 // call returns 0 / positive on success, -errcode on failure;
 // cv_last_error copies the thread-local message.
+//
+// Concurrency contract (same as the JNI SDK's usage): a handle is driven
+// by one thread at a time; different handles may be used concurrently;
+// close readers/writers before their filesystem handle.
 
 #include <cstdarg>
 
