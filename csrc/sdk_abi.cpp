@@ -231,6 +231,7 @@ struct CvReader {
 struct CvWriter {
   CvFilesystem* fs;
   std::string path;
+  int64_t block_size = 64 << 20;   // from the create reply
   int64_t pos = 0;
   // current block stream
   int dfd = -1;
@@ -570,6 +571,11 @@ extern "C" int64_t cv_create(int64_t h, const char* path, int overwrite) {
   auto* w = new CvWriter();
   w->fs = f;
   w->path = path;
+  const CvVal* st = v.get("status");
+  if (st) {
+    int64_t bs = st->geti("block_size");
+    if (bs > 0) w->block_size = bs;
+  }
   return cvh_put(w);
 }
 
@@ -633,7 +639,7 @@ static int cvw_open_block(CvWriter* w) {
   w->seq = 0;
   w->inflight = 0;
   std::string hmap = CvMap().kv_int("block_id", w->cur_block)
-                         .kv_int("reserve", w->fs->block_size)
+                         .kv_int("reserve", w->block_size)
                          .kv_str("tier", w->fs->tier).done();
   std::string frame =
       meta_proto(uint32_t(hmap.size()), 0, kCodeWriteBlock, 1, w->req,
@@ -658,7 +664,7 @@ extern "C" int64_t cv_write(int64_t wh, const void* buf, int64_t n) {
       int rc = cvw_open_block(w);
       if (rc) return rc;
     }
-    int64_t room = w->fs->block_size - w->cur_len;
+    int64_t room = w->block_size - w->cur_len;
     int64_t take = std::min<int64_t>({left, room, 4 << 20});
     std::string ph = meta_proto(0, uint32_t(take), kCodeWriteBlock, 2,
                                 w->req, w->seq++);
@@ -700,7 +706,7 @@ extern "C" int64_t cv_write(int64_t wh, const void* buf, int64_t n) {
     left -= take;
     w->cur_len += take;
     w->pos += take;
-    if (w->cur_len >= w->fs->block_size) {
+    if (w->cur_len >= w->block_size) {
       int rc = cvw_finish_block(w);
       if (rc) return rc;
     }

@@ -61,8 +61,9 @@ def test_c_abi_end_to_end(lib, cluster):
         assert lib.cv_get_status(fs, b"/cabi/sub", ctypes.byref(st)) == 0
         assert st.file_type == 1   # dir
 
-        # streaming write crossing a block boundary (64 MiB blocks ->
-        # use a smallish payload but multiple cv_write calls)
+        # streaming write crossing block boundaries (the writer adopts
+        # the master's block size from the create reply — 4 MiB in the
+        # test conf, so 9 MiB = 3 blocks)
         data = os.urandom(9 << 20)
         w = lib.cv_create(fs, b"/cabi/file.bin", 1)
         assert w > 0, _err(lib)
