@@ -90,6 +90,10 @@ def main(argv=None) -> int:
         conf.worker.data_dirs = args.data_dir
     conf.worker.rpc_port = 0
     conf.fuse.mnt_path = args.mnt
+    import os as _os
+    if _os.environ.get("CURVINE_FUSE_WRITEBACK"):
+        conf.fuse.writeback_cache = \
+            _os.environ["CURVINE_FUSE_WRITEBACK"] != "0"
 
     daemon = FuseDaemon(conf, args.mnt, embed_worker=args.embed_worker,
                         device_id=args.device).start(takeover=args.takeover)

@@ -35,10 +35,13 @@ def mount(tmp_path):
     # never stat a dead fuse mount (uninterruptible hang)
     mnt = f"/tmp/curvine-fuse-test-{os.getpid()}"
     master = f"127.0.0.1:{smc.master.rpc.port}"
+    dbg = os.environ.get("CURVINE_FUSE_DEBUG_LOG")
     proc = subprocess.Popen(
         [sys.executable, "-m", "curvine_amd.fuse", "--mnt", mnt,
-         "--master", master, "--log-level", "WARNING"],
-        stdout=subprocess.PIPE, stderr=subprocess.DEVNULL, text=True,
+         "--master", master, "--log-level",
+         "DEBUG" if dbg else "WARNING"],
+        stdout=subprocess.PIPE,
+        stderr=open(dbg, "a") if dbg else subprocess.DEVNULL, text=True,
         cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
     line = proc.stdout.readline()
     assert line.startswith("READY"), f"fuse daemon failed: {line!r}"
