@@ -256,6 +256,28 @@ def cmd_load_cancel(fs, args):
     print(json.dumps(fs.call(fs.fs.client.cancel_job(args.job_id))))
 
 
+def cmd_load_retry(fs, args):
+    """Re-dispatch failed tasks of a load/transfer job (RetryTransfer)."""
+    from curvine_amd.rpc.codes import RpcCode
+
+    async def go():
+        r = await fs.fs.client.connector.rpc(RpcCode.RetryTransfer,
+                                             {"job_id": args.job_id})
+        return r.header
+    print(json.dumps(fs.call(go())))
+
+
+def cmd_transfers(fs, args):
+    """List transfer/load jobs (ListTransfers)."""
+    from curvine_amd.rpc.codes import RpcCode
+
+    async def go():
+        r = await fs.fs.client.connector.rpc(RpcCode.ListTransfers, {})
+        return r.header
+    for j in fs.call(go()).get("transfers", fs.call(go()).get("jobs", [])):
+        print(json.dumps(j))
+
+
 def cmd_mount(fs, args):
     props = dict(kv.split("=", 1) for kv in (args.option or []))
     mi = fs.mount(args.curvine_path, args.ufs_path, props,
@@ -393,6 +415,8 @@ def build_parser() -> argparse.ArgumentParser:
         A("--wait", action="store_true"))
     add("load-status", cmd_load_status, A("job_id"))
     add("load-cancel", cmd_load_cancel, A("job_id"))
+    add("load-retry", cmd_load_retry, A("job_id"))
+    add("transfers", cmd_transfers)
     add("mount", cmd_mount, A("ufs_path"), A("curvine_path"),
         A("-o", "--option", action="append"),
         A("--cache-mode", default="cache", choices=["cache", "fs"]),

@@ -210,12 +210,24 @@ class WebServer:
                 lines.append(f"curvine_master_{k} {info[k]}")
             lines.append(
                 f"curvine_master_live_workers {len(info['live_workers'])}")
+        if self.master and getattr(self.master, "native_meta", None):
+            for k, v in self.master.native_meta.stats().items():
+                if isinstance(v, (int, float)):
+                    lines.append(f"curvine_meta_frontend_{k} {v}")
         if self.worker:
             for s in self.worker.store.storages():
                 lbl = f'{{tier="{s.tier}",dir="{s.dir_id}"}}'
                 lines.append(f"curvine_worker_capacity_bytes{lbl} {s.capacity}")
                 lines.append(f"curvine_worker_used_bytes{lbl} {s.used}")
                 lines.append(f"curvine_worker_blocks{lbl} {s.block_num}")
+            stats_fn = getattr(self.worker.rpc, "stats", None)
+            if stats_fn:
+                try:
+                    for k, v in stats_fn().items():
+                        if isinstance(v, (int, float)):
+                            lines.append(f"curvine_data_plane_{k} {v}")
+                except Exception:  # noqa: BLE001
+                    pass
         if self.fuse_session:
             for op, d in self.fuse_session.stats().items():
                 if not isinstance(d, dict) or "count" not in d:

@@ -79,3 +79,36 @@ def test_validate_subcommand(tmp_path):
     bad.write_text('[worker]\ndata_dirs = ["[HBM]gpu0"]\n'
                    '[journal]\npeers = ["1@h:1", "2@h:2"]\n')
     assert main(["validate", str(bad)]) == 1
+
+
+def test_cv_transfers_and_retry(tmp_path, capsys):
+    """cv transfers / load-retry surface (ListTransfers/RetryTransfer)."""
+    import json as _json
+    import os
+    import sys
+    sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+    from curvine_amd.cli.cv import main as cv_main
+    from curvine_amd.testing import SyncMiniCluster
+
+    smc = SyncMiniCluster(tmp_dir=str(tmp_path / "cv")).start()
+    try:
+        master = f"127.0.0.1:{smc.master.rpc.port}"
+        # mount a local UFS + load it so a job exists
+        src = tmp_path / "src"
+        src.mkdir()
+        (src / "a.bin").write_bytes(b"x" * 1000)
+        assert cv_main(["--master", master, "mount", f"file://{src}",
+                        "/warm"]) == 0
+        assert cv_main(["--master", master, "load", "/warm"]) == 0
+        out = capsys.readouterr().out
+        job_id = _json.loads(out.strip().split("\n")[-1])["job_id"]
+        assert cv_main(["--master", master, "transfers"]) == 0
+        out = capsys.readouterr().out
+        assert any(_json.loads(l)["job_id"] == job_id
+                   for l in out.strip().split("\n") if l.startswith("{"))
+        assert cv_main(["--master", master, "load-retry", job_id]) == 0
+        out = capsys.readouterr().out
+        assert _json.loads(out.strip().split("\n")[-1])["job_id"] == job_id
+    finally:
+        smc.stop()
