@@ -398,12 +398,9 @@ class MasterHandler:
             for j in self.master.jobs.jobs.values()]}
 
     def op_retrytransfer(self, h, d):
-        job = self.master.jobs.jobs.get(h.get("job_id"))
-        if job is None:
-            raise err.JobNotFound(str(h.get("job_id")))
-        return self.master.jobs.submit({"path": job["path"],
-                                        "recursive": job["recursive"],
-                                        "replicas": job["replicas"]})
+        if self.master.conf.job.service_addr:
+            return self._forward_transfer(RpcCode.RetryTransfer, h)
+        return self.master.jobs.retry(h.get("job_id", ""))
 
     # ---------------- raft protocol ----------------
     def op_raftvote(self, h, d):

@@ -127,6 +127,27 @@ class JobManager:
                     "replicas": job["replicas"]}
             job["tasks"][task_id] = task
 
+    def retry(self, job_id: str) -> dict:
+        """Re-dispatch FAILED tasks of an existing job in place
+        (RetryTransfer semantics — not a fresh submission)."""
+        job = self.jobs.get(job_id)
+        if job is None:
+            raise err.JobNotFound(job_id)
+        retried = 0
+        for task in job["tasks"].values():
+            if task["state"] == "failed":
+                task["state"] = "assigned"
+                job["failed"] -= 1
+                if self.master is not None:
+                    self.master.fs.workers.add_command(
+                        task["worker"], {"cmd": "load_task", **task})
+                retried += 1
+        if retried and job["state"] in ("completed_with_failures", "failed"):
+            job["state"] = "running"
+        self.store.save(job)
+        return {"job_id": job_id, "retried": retried,
+                "state": job["state"]}
+
     def status(self, job_id: str) -> dict:
         job = self.jobs.get(job_id)
         if job is None:

@@ -157,20 +157,10 @@ class TransferService:
                 "total": job["total"]}
 
     async def retry(self, job_id: str) -> dict:
-        job = self.jobs.jobs.get(job_id)
-        if job is None:
-            raise err.JobNotFound(job_id)
-        retried = 0
+        out = self.jobs.retry(job_id)   # master=None: state walk only
+        job = self.jobs.jobs[job_id]
         for task in job["tasks"].values():
-            if task["state"] == "failed":
-                task["state"] = "assigned"
-                job["failed"] -= 1
+            if task["state"] == "assigned" and out["retried"]:
                 await self.env.add_command(task["worker"],
                                            {"cmd": "load_task", **task})
-                retried += 1
-        if retried and job["state"] in ("completed_with_failures",
-                                        "failed"):
-            job["state"] = "running"
-        self.jobs.store.save(job)
-        return {"job_id": job_id, "retried": retried,
-                "state": job["state"]}
+        return out
