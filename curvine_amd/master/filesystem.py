@@ -80,9 +80,10 @@ class MasterFilesystem:
                     storage_tier: str = "", overwrite: bool = False,
                     mode: int = 0o644) -> dict:
         """Reply-dict variant (mutation-QPS hot path)."""
+        path = norm_path(path)
         node = self._create_node(path, block_size, replicas, storage_tier,
                                  overwrite, mode)
-        return self.fs_dir.status_dict(node, norm_path(path))
+        return self.fs_dir.status_dict(node, path)
 
     def append(self, path: str) -> FileBlocks:
         node = self.fs_dir.must_resolve(path)

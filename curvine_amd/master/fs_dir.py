@@ -26,7 +26,7 @@ ROOT_ID = 1
 
 
 class Inode:
-    __slots__ = ("id", "parent_id", "name", "file_type", "children",
+    __slots__ = ("id", "parent_id", "name", "file_type", "is_dir", "children",
                  "length", "blocks", "block_size", "replicas", "storage_tier",
                  "complete", "mtime_ms", "atime_ms", "mode", "uid", "gid",
                  "ttl_ms", "ttl_action", "symlink_target", "nlink", "xattrs",
@@ -37,7 +37,10 @@ class Inode:
         self.parent_id = 0
         self.name = name
         self.file_type = file_type
-        self.children: Optional[dict[str, int]] = {} if file_type == FileType.DIR else None
+        # plain attribute, not a property: checked several times per
+        # mutation and file_type never changes after construction
+        self.is_dir = file_type == FileType.DIR
+        self.children: Optional[dict[str, int]] = {} if self.is_dir else None
         self.length = 0
         self.blocks: list[list[int]] = []   # [block_id, length]
         self.block_size = 64 << 20
@@ -58,10 +61,6 @@ class Inode:
         self.xattrs: dict[str, bytes] = {}
         self.access_count = 0
 
-    @property
-    def is_dir(self) -> bool:
-        return self.file_type == FileType.DIR
-
     def to_state(self) -> dict:
         return {s: (getattr(self, s) if s != "children" else
                     (dict(self.children) if self.children is not None else None))
@@ -73,7 +72,8 @@ class Inode:
         for s in Inode.__slots__:
             if s in d:
                 setattr(ino, s, d[s])
-        if ino.file_type == FileType.DIR and ino.children is None:
+        ino.is_dir = ino.file_type == FileType.DIR
+        if ino.is_dir and ino.children is None:
             ino.children = {}
         return ino
 
@@ -257,13 +257,6 @@ class FsDir:
                file_type: int = int(FileType.FILE)) -> tuple[Inode, list[int]]:
         """Returns (inode, blocks_to_delete_of_overwritten_file)."""
         path = norm_path(path)
-        existing = self.resolve(path)
-        removed_blocks: list[int] = []
-        if existing is not None:
-            if existing.is_dir:
-                raise err.IsDirectory(path)
-            if not overwrite:
-                raise err.FileAlreadyExists(path)
         parent_path, name = split_path(path)
         parent = self.resolve(parent_path)
         if parent is None:
@@ -272,6 +265,14 @@ class FsDir:
             parent = self.mkdir(parent_path, 0o755, True)
         if not parent.is_dir:
             raise err.NotDirectory(parent_path)
+        removed_blocks: list[int] = []
+        child_id = parent.children.get(name)
+        existing = self.inodes.get(child_id) if child_id is not None else None
+        if existing is not None:
+            if existing.is_dir:
+                raise err.IsDirectory(path)
+            if not overwrite:
+                raise err.FileAlreadyExists(path)
         entry = self.journal.log(
             Op.CREATE, parent_id=parent.id, name=name,
             inode_id=self.next_inode_id + 1, block_size=block_size,
