@@ -108,6 +108,14 @@ class SqliteInodeStore:
         self.conn.execute("PRAGMA synchronous=NORMAL")
         self._dirty: set[int] = set()
         self._deleted: set[int] = set()
+        # async-flush state: rows snapshot on the master thread, the
+        # sqlite transaction runs on a dedicated writer thread with its
+        # own connection (WAL: one writer + concurrent readers), so the
+        # actor tick never stalls mutations behind a 20k-row commit
+        self._inflight: set[int] = set()
+        self._pool = None
+        self._commit_f = None
+        self._wconn = None
 
     # ---------------- mirror interface (same as MetaMirror) ----------------
     def upsert(self, node) -> None:
@@ -148,15 +156,75 @@ class SqliteInodeStore:
     MAX_BATCH = 20_000   # bound the per-tick pause on mutation storms
 
     def flush(self, fs_dir, mounts_state, op_id: int) -> int:
-        """One transaction: dirty upserts + deletes + watermarks, capped
-        at MAX_BATCH rows (the remainder stays dirty for the next tick —
-        correctness is carried by the WAL tail + restart reconcile).
-        Returns the number of rows written."""
+        """Synchronous flush: snapshot + one transaction on the calling
+        thread (shutdown / checkpoint path).  Waits out any in-flight
+        async commit first so row versions never go backwards."""
+        self.wait_flush()
+        snap = self._snapshot(fs_dir, mounts_state, op_id)
+        if snap is None:
+            return 0
+        try:
+            return self._commit(self.conn, snap)
+        except Exception:
+            self._recover(snap)
+            raise
+        finally:
+            self._inflight.clear()
+
+    def flush_async(self, fs_dir, mounts_state, op_id: int) -> bool:
+        """Actor-tick flush: snapshot rows on the caller (master) thread —
+        the consistent-read part — then commit them on the writer thread.
+        Single-flight: a tick that lands while a commit is still running
+        skips (the dirt just waits for the next tick)."""
+        if self._commit_f is not None and not self._commit_f.done():
+            return False
+        snap = self._snapshot(fs_dir, mounts_state, op_id)
+        if snap is None:
+            return False
+        if self._pool is None:
+            from concurrent.futures import ThreadPoolExecutor
+            self._pool = ThreadPoolExecutor(
+                max_workers=1, thread_name_prefix="inode-db-flush")
+        self._commit_f = self._pool.submit(self._commit_async, snap)
+        return True
+
+    def wait_flush(self) -> None:
+        """Block until any in-flight async commit lands."""
+        f, self._commit_f = self._commit_f, None
+        if f is not None:
+            try:
+                f.result(timeout=60)
+            except Exception:  # noqa: BLE001 — already recovered to dirty
+                pass
+
+    def _commit_async(self, snap) -> int:
+        if self._wconn is None:
+            self._wconn = sqlite3.connect(self.path, check_same_thread=False)
+            self._wconn.execute("PRAGMA synchronous=NORMAL")
+        try:
+            return self._commit(self._wconn, snap)
+        except Exception:  # noqa: BLE001
+            log.exception("async inode-db flush failed; rows stay dirty")
+            self._recover(snap)
+            return 0
+        finally:
+            self._inflight.clear()
+
+    def _recover(self, snap) -> None:
+        rows, block_rows, dels, batch, partial, op_id, mounts_state = snap
+        self._dirty.update(batch)
+        self._deleted.update(d[0] for d in dels)
+
+    def _snapshot(self, fs_dir, mounts_state, op_id: int):
+        """Consistent snapshot of up to MAX_BATCH dirty rows; must run on
+        the mutating (master) thread.  Clears the taken dirt — a failure
+        path re-adds it — and marks it in flight so page_out will not
+        evict a node whose row is not committed yet."""
         if not self._dirty and not self._deleted:
             cur = self.conn.execute("SELECT v FROM meta WHERE k='op_id'")
             row = cur.fetchone()
             if row is not None and int.from_bytes(row[0], "little") == op_id:
-                return 0
+                return None
         rows = []
         batch = []
         block_rows = []
@@ -178,40 +246,46 @@ class SqliteInodeStore:
                 break
         partial = len(batch) < len(self._dirty)
         dels = [(iid,) for iid in self._deleted]
-        with self.conn:
+        self._dirty.difference_update(batch)
+        self._deleted.clear()
+        self._inflight = set(batch) | {d[0] for d in dels}
+        # watermarks are part of the snapshot (the writer thread must not
+        # read fs_dir)
+        mstate = (fs_dir.next_inode_id, fs_dir.next_block_id, mounts_state)
+        return rows, block_rows, dels, batch, partial, op_id, mstate
+
+    def _commit(self, conn, snap) -> int:
+        rows, block_rows, dels, batch, partial, op_id, mstate = snap
+        next_inode_id, next_block_id, mounts_state = mstate
+        with conn:
             if rows:
-                self.conn.executemany(
+                conn.executemany(
                     "REPLACE INTO inodes (id, state, atime, ttl_deadline,"
                     " is_file) VALUES (?, ?, ?, ?, ?)", rows)
                 # block index rows: drop-then-insert per dirty inode so
                 # truncated files lose their stale entries
-                self.conn.executemany(
+                conn.executemany(
                     "DELETE FROM blocks WHERE inode_id=?",
                     [(r[0],) for r in rows])
                 if block_rows:
-                    self.conn.executemany(
+                    conn.executemany(
                         "REPLACE INTO blocks (block_id, inode_id)"
                         " VALUES (?, ?)", block_rows)
             if dels:
-                self.conn.executemany("DELETE FROM inodes WHERE id=?", dels)
-                self.conn.executemany(
+                conn.executemany("DELETE FROM inodes WHERE id=?", dels)
+                conn.executemany(
                     "DELETE FROM blocks WHERE inode_id=?", dels)
-            meta = [("next_inode_id",
-                     fs_dir.next_inode_id.to_bytes(8, "little")),
-                    ("next_block_id",
-                     fs_dir.next_block_id.to_bytes(8, "little")),
+            meta = [("next_inode_id", next_inode_id.to_bytes(8, "little")),
+                    ("next_block_id", next_block_id.to_bytes(8, "little")),
                     ("mounts", msgpack.packb(mounts_state,
                                              use_bin_type=True))]
             if not partial:
                 # advance the restart watermark only once every dirty
                 # inode as of this op is actually on disk
                 meta.append(("op_id", op_id.to_bytes(8, "little")))
-            self.conn.executemany(
+            conn.executemany(
                 "REPLACE INTO meta (k, v) VALUES (?, ?)", meta)
-        n = len(rows) + len(dels)
-        self._dirty.difference_update(batch)
-        self._deleted.clear()
-        return n
+        return len(rows) + len(dels)
 
     def load(self, fs_dir, mounts) -> Optional[int]:
         """Populate fs_dir (and the mount table) from the DB; returns the
@@ -310,7 +384,8 @@ class SqliteInodeStore:
         target = max(1, int(max_resident * 0.9))
         cands = [n for iid, n in inodes.items()
                  if iid != ROOT_ID and iid not in self._dirty
-                 and iid not in self._deleted and iid not in protected]
+                 and iid not in self._deleted and iid not in self._inflight
+                 and iid not in protected]
         cands.sort(key=lambda n: n.atime_ms)
         evicted = 0
         bindex = fs_dir.block_index
@@ -346,7 +421,14 @@ class SqliteInodeStore:
         self._dirty.update(live)
 
     def close(self) -> None:
-        try:
-            self.conn.close()
-        except Exception:  # noqa: BLE001
-            pass
+        self.wait_flush()
+        if self._pool is not None:
+            self._pool.shutdown(wait=True)
+            self._pool = None
+        for c in (self._wconn, self.conn):
+            try:
+                if c is not None:
+                    c.close()
+            except Exception:  # noqa: BLE001
+                pass
+        self._wconn = None

@@ -395,9 +395,12 @@ class Master:
                 self._ttl_sweep()
                 self._eviction_sweep()
                 if self.inode_db is not None:
-                    self.inode_db.flush(self.fs.fs_dir,
-                                        self.mounts.to_snapshot(),
-                                        self.journal.op_id)
+                    # snapshot on this thread, commit on the store's
+                    # writer thread: the tick never stalls mutations
+                    # behind the sqlite transaction
+                    self.inode_db.flush_async(self.fs.fs_dir,
+                                              self.mounts.to_snapshot(),
+                                              self.journal.op_id)
                     maxres = self.conf.master.max_resident_inodes
                     if maxres > 0:
                         self.inode_db.page_out(

@@ -293,3 +293,68 @@ def test_paged_namespace_beyond_resident_cap(tmp_path):
         await m.stop()
 
     _a.new_event_loop().run_until_complete(phase3())
+
+
+def test_async_flush_semantics(tmp_path):
+    """Off-thread commit: rows land, re-dirtied nodes survive an
+    in-flight commit, a failed commit recovers dirt, and the sync
+    shutdown flush waits out the writer thread."""
+    import sqlite3
+    import threading
+
+    from curvine_amd.master.fs_dir import FsDir
+    from curvine_amd.master.inode_db import SqliteInodeStore
+    from curvine_amd.master.journal import JournalWriter
+
+    jd = str(tmp_path / "j")
+    fs_dir = FsDir(JournalWriter(jd))
+    store = SqliteInodeStore(str(tmp_path / "inodes.db"))
+    fs_dir.mirror = store
+    store._dirty.update(dict.keys(fs_dir.inodes))
+
+    for i in range(50):
+        fs_dir.create(f"/a/f{i}", 1 << 20, 1, "MEM")
+
+    # async flush commits on the writer thread
+    assert store.flush_async(fs_dir, {}, fs_dir.journal.op_id)
+    store.wait_flush()
+    n = store.conn.execute("SELECT COUNT(*) FROM inodes").fetchone()[0]
+    assert n == 52  # root + /a + 50 files
+    assert not store._dirty and not store._inflight
+
+    # a node dirtied while the commit is in flight stays dirty after it
+    gate = threading.Event()
+    orig = store._commit
+
+    def slow_commit(conn, snap):
+        gate.wait(5)
+        return orig(conn, snap)
+
+    store._commit = slow_commit
+    fs_dir.create("/a/late1", 1 << 20, 1, "MEM")
+    assert store.flush_async(fs_dir, {}, fs_dir.journal.op_id)
+    fs_dir.create("/a/late2", 1 << 20, 1, "MEM")   # dirtied mid-flight
+    late2 = fs_dir.resolve("/a/late2").id
+    assert store.flush_async(fs_dir, {}, fs_dir.journal.op_id) is False
+    gate.set()
+    store.wait_flush()
+    assert late2 in store._dirty          # waits for the next tick
+    store._commit = orig
+
+    # a failing commit re-adds its batch to the dirty set
+    def boom(conn, snap):
+        raise sqlite3.OperationalError("disk I/O error (injected)")
+
+    store._commit = boom
+    fs_dir.create("/a/fail", 1 << 20, 1, "MEM")
+    fid = fs_dir.resolve("/a/fail").id
+    assert store.flush_async(fs_dir, {}, fs_dir.journal.op_id)
+    store.wait_flush()
+    assert fid in store._dirty
+    store._commit = orig
+
+    # sync flush (shutdown path) drains everything
+    store.flush(fs_dir, {}, fs_dir.journal.op_id)
+    n = store.conn.execute("SELECT COUNT(*) FROM inodes").fetchone()[0]
+    assert n == 55
+    store.close()
