@@ -135,7 +135,25 @@ class FsClient:
         return h.get("freed_blocks", 0)
 
     async def get_master_info(self) -> dict:
-        return await self._rpc(RpcCode.GetFilesystemInfo, {})
+        # the client handshake: report our version (component_info=1000
+        # analog), warn (deduped) about incompatible workers the master
+        # advertises (worker_precheck.rs behavior — always diagnose on
+        # the client side, never rejects)
+        from curvine_amd.compat import (CompatibilityPolicy, PeerWarnDedup,
+                                        component_info)
+        info = await self._rpc(RpcCode.GetFilesystemInfo,
+                               {"component_info": component_info("client")})
+        if not hasattr(self, "_worker_precheck"):
+            self._worker_precheck = PeerWarnDedup("worker")
+            self._precheck_policy = CompatibilityPolicy()
+        pol = self._precheck_policy
+        for w in info.get("live_workers", []):
+            ci = w.get("component_info")
+            if pol.should_evaluate(ci is not None) and ci is not None:
+                self._worker_precheck.warn(
+                    w.get("address", {}).get("worker_id"),
+                    pol.check_worker(ci))
+        return info
 
     # ---------------- mounts / jobs ----------------
     async def mount(self, curvine_path: str, ufs_path: str,

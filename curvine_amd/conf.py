@@ -244,6 +244,17 @@ class JobConf:
 
 
 @dataclass
+class CompatibilityConf:
+    """Peer version policy (compatibility_conf.rs analog).  Defaults are
+    intentionally lenient: diagnose mode with no bounds, so old
+    components are never rejected without explicit configuration."""
+    mode: str = "diagnose"          # diagnose | enforce
+    min_worker_version: str = ""    # lowest worker release accepted
+    min_client_version: str = ""    # lowest client release accepted
+    blocked_versions: list = field(default_factory=list)
+
+
+@dataclass
 class ClusterConf:
     cluster_id: str = "curvine-amd"
     master: MasterConf = field(default_factory=MasterConf)
@@ -253,6 +264,8 @@ class ClusterConf:
     fuse: FuseConf = field(default_factory=FuseConf)
     ufs: UfsConf = field(default_factory=UfsConf)
     job: JobConf = field(default_factory=JobConf)
+    compatibility: CompatibilityConf = field(
+        default_factory=CompatibilityConf)
     testing: bool = False
 
     # ---------------- loading ----------------

@@ -46,6 +46,7 @@ class ErrorCode(enum.IntEnum):
     RETRY = 31
     NOT_EMPTY = 32
     CONNECT = 33
+    INCOMPATIBLE_VERSION = 34
 
 
 class FsError(Exception):
@@ -113,6 +114,8 @@ OutOfRange = _err("OutOfRange", ErrorCode.OUT_OF_RANGE)
 JobNotFound = _err("JobNotFound", ErrorCode.JOB_NOT_FOUND)
 RetryError = _err("RetryError", ErrorCode.RETRY)
 ConnectError = _err("ConnectError", ErrorCode.CONNECT)
+IncompatibleVersion = _err("IncompatibleVersion",
+                           ErrorCode.INCOMPATIBLE_VERSION)
 
 _CODE_TO_CLASS: dict[ErrorCode, type] = {
     c.code: c  # type: ignore[attr-defined]
@@ -123,7 +126,7 @@ _CODE_TO_CLASS: dict[ErrorCode, type] = {
         CapacityExceeded, ChecksumMismatch, Unsupported, RpcTimeout,
         Cancelled, NotLeader, MountNotFound, UfsError, QuotaExceeded,
         PermissionDenied, StaleGeneration, OutOfRange, JobNotFound,
-        RetryError, ConnectError,
+        RetryError, ConnectError, IncompatibleVersion,
     ]
 }
 

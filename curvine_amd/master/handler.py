@@ -202,7 +202,20 @@ class MasterHandler:
         return {"file_blocks": self.fs.get_block_locations(h["path"]).to_dict()}
 
     def op_getfilesysteminfo(self, h, d):
-        return self.fs.master_info()
+        # client handshake carries its version report here (the
+        # reference's GetFilesystemInfoRequest.component_info = 1000)
+        ci = h.get("component_info")
+        pol = self.master.compat_policy
+        if pol.should_evaluate(ci is not None):
+            v = pol.check_client(ci)
+            self.master.compat_warn_clients.warn(
+                (ci or {}).get("component", "client"), v)
+            if v.rejects(pol.mode):
+                raise err.IncompatibleVersion(f"client: {v.describe()}")
+        out = self.fs.master_info()
+        from curvine_amd.compat import component_info
+        out["component_info"] = component_info("master")
+        return out
 
     def op_setattr(self, h, d):
         st = self.fs.set_attr(h["path"], **{
@@ -314,6 +327,14 @@ class MasterHandler:
     # ---------------- worker plane ----------------
     def op_workerheartbeat(self, h, d):
         info = WorkerInfo.from_dict(h["worker"])
+        pol = self.master.compat_policy
+        if pol.should_evaluate(info.component_info is not None):
+            v = pol.check_worker(info.component_info)
+            self.master.compat_warn_workers.warn(
+                info.address.worker_id, v)
+            if v.rejects(pol.mode):
+                raise err.IncompatibleVersion(
+                    f"worker {info.address.worker_id}: {v.describe()}")
         cmds = self.fs.worker_heartbeat(info, h.get("added_blocks", []),
                                         h.get("removed_blocks", []))
         return {"commands": cmds}

@@ -131,6 +131,11 @@ class Master:
                              conf.master.rpc_port, self.rpc_service)
         self.native_meta = None   # NativeMetaFrontend when enabled
         self.inode_db = None      # SqliteInodeStore when enabled
+        from curvine_amd.compat import CompatibilityPolicy, PeerWarnDedup
+        self.compat_policy = CompatibilityPolicy.from_conf(
+            conf.compatibility)
+        self.compat_warn_workers = PeerWarnDedup("worker")
+        self.compat_warn_clients = PeerWarnDedup("client")
         self._actor_task: Optional[asyncio.Task] = None
         self._stopped = asyncio.Event()
         self._mutation_count = 0

@@ -74,6 +74,9 @@ class WorkerInfo:
     state: int = int(WorkerState.LIVE)
     storages: list[StorageInfo] = field(default_factory=list)
     last_heartbeat_ms: int = 0
+    # structured version report (ComponentInfoProto analog, wire field
+    # 1000); None = legacy peer, treated as unknown by the policy
+    component_info: dict | None = None
 
     @property
     def capacity(self) -> int:
@@ -98,6 +101,7 @@ class WorkerInfo:
         w.state = d.get("state", 0)
         w.storages = [StorageInfo(**s) for s in d.get("storages", [])]
         w.last_heartbeat_ms = d.get("last_heartbeat_ms", 0)
+        w.component_info = d.get("component_info")
         return w
 
 

@@ -95,7 +95,10 @@ class Worker:
                              device_id=self.device_id)
 
     def info(self) -> WorkerInfo:
-        return WorkerInfo(address=self.address(), storages=self.store.storages())
+        from curvine_amd.compat import component_info
+        return WorkerInfo(address=self.address(),
+                          storages=self.store.storages(),
+                          component_info=component_info("worker"))
 
     # ---------------- heartbeats ----------------
     async def _connect_master(self) -> None:
