@@ -69,7 +69,10 @@ def build_conf(args, tmp, rank):
     conf.master.block_size = args.block_size
     conf.client.block_size = args.block_size
     conf.master.heartbeat_check_ms = 1000
-    conf.worker.heartbeat_interval_ms = 1000
+    # seqwrite overwrites each step; block frees ride the heartbeat
+    # command channel, so keep it snappy to bound transient capacity use
+    conf.worker.heartbeat_interval_ms = \
+        250 if args.workload == "seqwrite" else 1000
     conf.client.write_chunk_size = 4 << 20
     conf.client.read_chunk_size = args.read_chunk
     conf.worker.staging_buf_bytes = args.staging_bytes
@@ -432,6 +435,67 @@ def step_client_seq(args, rank, rt) -> int:
     return got
 
 
+def step_client_seqwrite(args, rank, rt) -> int:
+    """Sequential writes via the client path: one concurrent writer per
+    file (short-circuit into the in-process worker's arena), overwrite
+    per step so capacity stays bounded.  Metric = ingest GiB/s (the
+    reference's fio write suite analog, build/tests/fio-test.sh)."""
+    async def run():
+        import numpy as np
+        base = np.random.default_rng(4321 + rank).integers(
+            0, 256, size=args.read_chunk, dtype=np.uint8).tobytes()
+
+        async def write_file(i):
+            w = await rt.fs.create(f"/bench/r{rank}/w{i}", overwrite=True,
+                                   storage_tier="HBM")
+            pos = 0
+            while pos < args.file_size:
+                n = min(args.read_chunk, args.file_size - pos)
+                await w.write(base if n == len(base) else base[:n])
+                pos += n
+            await w.complete()
+            return pos
+        totals = await asyncio.gather(
+            *(write_file(i) for i in range(args.files)))
+        return sum(totals)
+    return rt.call(run())
+
+
+def step_fuse_seqwrite(args, rank, mnt) -> int:
+    """T threads, each rewriting its share of files through the kernel
+    mount in read_chunk chunks."""
+    import numpy as np
+    base = np.random.default_rng(4321 + rank).integers(
+        0, 256, size=args.read_chunk, dtype=np.uint8).tobytes()
+    os.makedirs(f"{mnt}/bench/r{rank}", exist_ok=True)
+    paths = [f"{mnt}/bench/r{rank}/w{i}" for i in range(args.files)]
+    total = [0] * args.threads
+    errs = []
+
+    def worker(t):
+        try:
+            for i in range(t, len(paths), args.threads):
+                with open(paths[i], "wb") as f:
+                    pos = 0
+                    while pos < args.file_size:
+                        n = min(args.read_chunk, args.file_size - pos)
+                        f.write(base if n == len(base) else base[:n])
+                        pos += n
+                total[t] += args.file_size
+        except Exception as e:  # noqa: BLE001
+            errs.append(e)
+
+    ths = [threading.Thread(target=worker, args=(t,))
+           for t in range(args.threads)]
+    for t in ths:
+        t.start()
+    for t in ths:
+        t.join()
+    if errs:
+        raise errs[0]
+    return sum(total)
+
+
 def run_scale_sweep(args, rank, world, dist, torch, rt, has_gpu,
                     one_step, barrier_sync):
     """one command produces the 1/2/4/8 weak-scaling
@@ -525,7 +589,8 @@ def main():
     p.add_argument("--steps", type=int, default=5)
     p.add_argument("--warmup", type=int, default=2)
     p.add_argument("--path", choices=["fuse", "client"], default="fuse")
-    p.add_argument("--workload", choices=["seqread", "randread4k"],
+    p.add_argument("--workload",
+                   choices=["seqread", "randread4k", "seqwrite"],
                    default="seqread")
     p.add_argument("--files", type=int, default=16)
     p.add_argument("--file-size", type=int, default=1 << 30)
@@ -591,6 +656,9 @@ def main():
         args.files, args.file_size = 4, 32 << 20   # CPU plumbing scale
         args.hbm_gb = 2
         args.rand_reads = 20_000
+        # a writer reserves a whole block per open block: size blocks to
+        # the plumbing-scale files so reserves match the data
+        args.block_size = min(args.block_size, args.file_size)
 
     def barrier_sync():
         if dist is not None:
@@ -601,13 +669,18 @@ def main():
     rt, conf, daemon_proc, mnt = setup(args, rank, world, dist, has_gpu)
     log(rank, f"cluster up ({args.path}); writing "
         f"{args.files}x{args.file_size >> 20}MiB")
-    t0 = time.perf_counter()
-    write_dataset(args, rank, rt, mnt)
-    log(rank, f"dataset written in {time.perf_counter() - t0:.1f}s")
+    if args.workload != "seqwrite":   # writing IS the seqwrite workload
+        t0 = time.perf_counter()
+        write_dataset(args, rank, rt, mnt)
+        log(rank, f"dataset written in {time.perf_counter() - t0:.1f}s")
 
     lat_out: list = []
 
     def one_step():
+        if args.workload == "seqwrite":
+            if args.path == "fuse":
+                return step_fuse_seqwrite(args, rank, mnt)
+            return step_client_seqwrite(args, rank, rt)
         if args.workload == "randread4k":
             if args.path != "fuse":
                 return step_client_rand4k(args, rank, rt, lat_out)
@@ -644,6 +717,11 @@ def main():
             # the reference publishes no 4 KiB IOPS figure (only 256 KiB
             # random GiB/s and a ~100 µs latency class) — no ratio to quote
             vs = None
+        elif args.workload == "seqwrite":
+            metric = ("fuse_seq_write_GiBps" if args.path == "fuse"
+                      else "cached_seq_write_GiBps")
+            value, unit = round(gibps, 3), "GiB/s"
+            vs = None   # the reference publishes no write GiB/s figure
         else:
             metric = ("fuse_seq_read_GiBps" if args.path == "fuse"
                       else "cached_seq_read_GiBps")

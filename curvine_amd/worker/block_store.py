@@ -96,8 +96,17 @@ class BlockStore:
             try:
                 meta = layout.allocate(block_id, reserve)
             except err.CapacityExceeded as e:
-                last_exc = e
-                continue
+                # allocation pressure: reclaim drained deferred deletes
+                # now instead of waiting for the heartbeat tick, then
+                # retry this tier once
+                if self.reap_deferred() == 0:
+                    last_exc = e
+                    continue
+                try:
+                    meta = layout.allocate(block_id, reserve)
+                except err.CapacityExceeded as e2:
+                    last_exc = e2
+                    continue
             b = _Block(block_id, layout, meta)
             with self.lock:
                 if block_id in self.blocks:   # raced: rollback
