@@ -266,57 +266,9 @@ static void data_forward(DataServer* S, const std::shared_ptr<DataConn>& c,
   }
 }
 
-// ------------------------------------------------------- pinned bounce pool
-
-// hipHostMalloc costs ~1 ms: a per-device free-list of double-buffer
-// bounce pairs makes HBM stream opens O(microseconds) after warmup
-struct BouncePair {
-  void* pin[2] = {nullptr, nullptr};
-  hipEvent_t ev[2] = {};
-  int device = -1;
-};
-
-static constexpr size_t kBounceSz = 8 << 20;
-
-struct BouncePool {
-  std::mutex mu;
-  std::unordered_map<int, std::vector<BouncePair*>> free_by_dev;
-};
-static BouncePool g_bounce;
-
-static BouncePair* bounce_acquire(int device) {
-  {
-    std::lock_guard<std::mutex> g(g_bounce.mu);
-    auto& v = g_bounce.free_by_dev[device];
-    if (!v.empty()) {
-      BouncePair* b = v.back();
-      v.pop_back();
-      return b;
-    }
-  }
-  auto* b = new BouncePair();
-  b->device = device;
-  HIP_CHECK(hipSetDevice(device));
-  HIP_CHECK(hipHostMalloc(&b->pin[0], kBounceSz, hipHostMallocDefault));
-  HIP_CHECK(hipHostMalloc(&b->pin[1], kBounceSz, hipHostMallocDefault));
-  HIP_CHECK(hipEventCreateWithFlags(&b->ev[0], hipEventDisableTiming));
-  HIP_CHECK(hipEventCreateWithFlags(&b->ev[1], hipEventDisableTiming));
-  return b;
-}
-
-static void bounce_release(BouncePair* b) {
-  std::lock_guard<std::mutex> g(g_bounce.mu);
-  auto& v = g_bounce.free_by_dev[b->device];
-  if (v.size() >= 32) {
-    hipEventDestroy(b->ev[0]);
-    hipEventDestroy(b->ev[1]);
-    hipHostFree(b->pin[0]);
-    hipHostFree(b->pin[1]);
-    delete b;
-    return;
-  }
-  v.push_back(b);
-}
+// The pinned bounce pool (BouncePair/bounce_acquire/bounce_release) is
+// shared module infrastructure — defined in module.cpp ahead of the
+// arena copy engines, used here for HBM stream serving.
 
 // ---------------------------------------------------------------- read serve
 

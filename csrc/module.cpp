@@ -309,6 +309,58 @@ static bool is_pinned_host(const void* p) {
 // a pooled stream (parallel across caller threads, no global lock)
 static const uint64_t RING_THRESHOLD = 2u << 20;
 
+// ------------------------------------------------------- pinned bounce pool
+
+// hipHostMalloc costs ~1 ms: a per-device free-list of double-buffer
+// bounce pairs makes HBM stream opens O(microseconds) after warmup
+struct BouncePair {
+  void* pin[2] = {nullptr, nullptr};
+  hipEvent_t ev[2] = {};
+  int device = -1;
+};
+
+static constexpr size_t kBounceSz = 8 << 20;
+
+struct BouncePool {
+  std::mutex mu;
+  std::unordered_map<int, std::vector<BouncePair*>> free_by_dev;
+};
+static BouncePool g_bounce;
+
+static BouncePair* bounce_acquire(int device) {
+  {
+    std::lock_guard<std::mutex> g(g_bounce.mu);
+    auto& v = g_bounce.free_by_dev[device];
+    if (!v.empty()) {
+      BouncePair* b = v.back();
+      v.pop_back();
+      return b;
+    }
+  }
+  auto* b = new BouncePair();
+  b->device = device;
+  HIP_CHECK(hipSetDevice(device));
+  HIP_CHECK(hipHostMalloc(&b->pin[0], kBounceSz, hipHostMallocDefault));
+  HIP_CHECK(hipHostMalloc(&b->pin[1], kBounceSz, hipHostMallocDefault));
+  HIP_CHECK(hipEventCreateWithFlags(&b->ev[0], hipEventDisableTiming));
+  HIP_CHECK(hipEventCreateWithFlags(&b->ev[1], hipEventDisableTiming));
+  return b;
+}
+
+static void bounce_release(BouncePair* b) {
+  std::lock_guard<std::mutex> g(g_bounce.mu);
+  auto& v = g_bounce.free_by_dev[b->device];
+  if (v.size() >= 32) {
+    hipEventDestroy(b->ev[0]);
+    hipEventDestroy(b->ev[1]);
+    hipHostFree(b->pin[0]);
+    hipHostFree(b->pin[1]);
+    delete b;
+    return;
+  }
+  v.push_back(b);
+}
+
 // per-caller-thread stream: zero contention for concurrent small copies
 // (the 4K-IOPS path).  hipMemcpy on the null stream and ROCm's internal
 // pageable-staging lock both serialize; a thread_local stream does not.
@@ -342,60 +394,81 @@ static void dev_write_direct(Arena* a, uint64_t off, const uint8_t* src,
 // device -> host buffer, chunked through the pinned ring: D2H DMA of chunk
 // k overlaps the CPU memcpy of chunk k-1 (the staging pipeline of
 // BASELINE.json's "pinned hipMemcpyAsync on a side stream").
+// device -> unpinned host, chunked through a POOLED pinned double-buffer
+// on the caller's stream (D2H DMA of chunk k overlaps the memcpy-out of
+// chunk k-1); concurrent readers no longer serialize on one arena ring.
 static void dev_read(Arena* a, uint64_t off, uint8_t* dst, uint64_t n) {
   if (n <= RING_THRESHOLD || is_pinned_host(dst)) {
     dev_read_direct(a, off, dst, n);
     return;
   }
-  std::lock_guard<std::mutex> g(a->mu);
   HIP_CHECK(hipSetDevice(a->device));
+  BouncePair* bp = bounce_acquire(a->device);
+  hipStream_t s = thread_stream(a->device);
   const uint8_t* src = (const uint8_t*)a->base + off;
-  size_t nb = a->pin.size();
-  uint64_t nchunks = (n + a->pin_sz - 1) / a->pin_sz;
-  // issue up to nb asynchronous D2H copies ahead, drain in order
-  for (uint64_t c = 0; c < nchunks; ++c) {
-    uint64_t coff = c * a->pin_sz;
-    uint64_t clen = std::min<uint64_t>(a->pin_sz, n - coff);
-    size_t slot = c % nb;
-    if (c >= nb) HIP_CHECK(hipEventSynchronize(a->ev[slot]));
-    HIP_CHECK(hipMemcpyAsync(a->pin[slot], src + coff, clen,
-                             hipMemcpyDeviceToHost, a->stream));
-    HIP_CHECK(hipEventRecord(a->ev[slot], a->stream));
-    if (c + 1 == nchunks || ((c + 1) % nb) == 0) {
-      // drain the window and memcpy out
-      uint64_t first = (c / nb) * nb;
-      for (uint64_t d = first; d <= c; ++d) {
-        size_t ds = d % nb;
-        HIP_CHECK(hipEventSynchronize(a->ev[ds]));
-        uint64_t doff = d * a->pin_sz;
-        std::memcpy(dst + doff, a->pin[ds],
-                    std::min<uint64_t>(a->pin_sz, n - doff));
+  uint64_t nchunks = (n + kBounceSz - 1) / kBounceSz;
+  try {
+    for (uint64_t c = 0; c < nchunks; ++c) {
+      uint64_t coff = c * kBounceSz;
+      uint64_t clen = std::min<uint64_t>(kBounceSz, n - coff);
+      int slot = (int)(c & 1);
+      HIP_CHECK(hipMemcpyAsync(bp->pin[slot], src + coff, clen,
+                               hipMemcpyDeviceToHost, s));
+      HIP_CHECK(hipEventRecord(bp->ev[slot], s));
+      if (c > 0) {
+        // drain the previous chunk while this one is in flight
+        int prev = (int)((c - 1) & 1);
+        HIP_CHECK(hipEventSynchronize(bp->ev[prev]));
+        uint64_t poff = (c - 1) * kBounceSz;
+        std::memcpy(dst + poff, bp->pin[prev],
+                    std::min<uint64_t>(kBounceSz, n - poff));
       }
     }
+    int last = (int)((nchunks - 1) & 1);
+    HIP_CHECK(hipEventSynchronize(bp->ev[last]));
+    uint64_t loff = (nchunks - 1) * kBounceSz;
+    std::memcpy(dst + loff, bp->pin[last],
+                std::min<uint64_t>(kBounceSz, n - loff));
+  } catch (...) {
+    (void)hipStreamSynchronize(s);   // no in-flight DMA may outlive bp
+    bounce_release(bp);
+    throw;
   }
+  bounce_release(bp);
 }
 
+// host -> device, chunked through a POOLED pinned double-buffer on the
+// caller's stream: concurrent writers (multi-file ingest, replication
+// pushes) each pipeline their own H2D DMAs instead of serializing on
+// one arena-wide staging ring.
 static void dev_write(Arena* a, uint64_t off, const uint8_t* src, uint64_t n) {
   if (n <= RING_THRESHOLD || is_pinned_host(src)) {
     dev_write_direct(a, off, src, n);
     return;
   }
-  std::lock_guard<std::mutex> g(a->mu);
   HIP_CHECK(hipSetDevice(a->device));
+  BouncePair* bp = bounce_acquire(a->device);
+  hipStream_t s = thread_stream(a->device);
   uint8_t* dst = (uint8_t*)a->base + off;
-  size_t nb = a->pin.size();
-  uint64_t nchunks = (n + a->pin_sz - 1) / a->pin_sz;
-  for (uint64_t c = 0; c < nchunks; ++c) {
-    uint64_t coff = c * a->pin_sz;
-    uint64_t clen = std::min<uint64_t>(a->pin_sz, n - coff);
-    size_t slot = c % nb;
-    if (c >= nb) HIP_CHECK(hipEventSynchronize(a->ev[slot]));
-    std::memcpy(a->pin[slot], src + coff, clen);
-    HIP_CHECK(hipMemcpyAsync(dst + coff, a->pin[slot], clen,
-                             hipMemcpyHostToDevice, a->stream));
-    HIP_CHECK(hipEventRecord(a->ev[slot], a->stream));
+  uint64_t nchunks = (n + kBounceSz - 1) / kBounceSz;
+  try {
+    for (uint64_t c = 0; c < nchunks; ++c) {
+      uint64_t coff = c * kBounceSz;
+      uint64_t clen = std::min<uint64_t>(kBounceSz, n - coff);
+      int slot = (int)(c & 1);
+      if (c >= 2) HIP_CHECK(hipEventSynchronize(bp->ev[slot]));
+      std::memcpy(bp->pin[slot], src + coff, clen);
+      HIP_CHECK(hipMemcpyAsync(dst + coff, bp->pin[slot], clen,
+                               hipMemcpyHostToDevice, s));
+      HIP_CHECK(hipEventRecord(bp->ev[slot], s));
+    }
+    HIP_CHECK(hipStreamSynchronize(s));
+  } catch (...) {
+    (void)hipStreamSynchronize(s);   // no in-flight DMA may outlive bp
+    bounce_release(bp);
+    throw;
   }
-  HIP_CHECK(hipStreamSynchronize(a->stream));
+  bounce_release(bp);
 }
 
 static void arena_read(int h, uint64_t off, py::buffer buf, uint64_t buf_off,
