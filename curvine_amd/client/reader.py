@@ -37,6 +37,24 @@ class ReadDetector:
         return self.seq_count >= self.rand_count
 
 
+
+def _hole_span(offs, blocks, length, pos) -> int:
+    """Length of the zero hole at file offset pos, or 0 when pos falls
+    inside a cached block extent.  Metadata length may exceed block
+    coverage (extending truncate, sparse tails): the uncovered range
+    reads back as zeros, like the reference's hole synthesis
+    (block_reader_hole.rs)."""
+    import bisect
+    idx = bisect.bisect_right(offs, pos) - 1
+    nxt = idx + 1
+    if idx >= 0:
+        lb = blocks[idx]
+        if pos < lb.offset + lb.block.length:
+            return 0
+    end = offs[nxt] if nxt < len(offs) else length
+    return max(0, end - pos)
+
+
 class SyncLocalReader:
     """Synchronous short-circuit reader: all blocks resolved to in-process
     store readers at construction; pread_* are plain function calls safe
@@ -97,31 +115,42 @@ class SyncLocalReader:
         n = max(0, min(n, self.length - off))
         got = 0
         while got < n:
-            idx = bisect.bisect_right(self._offs, off + got) - 1
+            pos = off + got
+            hole = _hole_span(self._offs, self.fb.blocks, self.length, pos)
+            if hole > 0:
+                fill = min(hole, n - got)
+                out[out_off + got:out_off + got + fill] = b"\x00" * fill
+                got += fill
+                continue
+            idx = bisect.bisect_right(self._offs, pos) - 1
             if idx < 0 or idx >= len(self._readers):
                 break
             lb = self.fb.blocks[idx]
-            boff = off + got - lb.offset
+            boff = pos - lb.offset
             want = min(n - got, lb.block.length - boff)
-            if want <= 0:
-                break
             got += self._readers[idx].read_into(boff, out, out_off + got, want)
         return got
 
     def pread_into_ptr(self, off: int, ptr: int, n: int) -> int:
         """Pinned/device-pointer destination (direct DMA)."""
         import bisect
+        import ctypes
         n = max(0, min(n, self.length - off))
         got = 0
         while got < n:
-            idx = bisect.bisect_right(self._offs, off + got) - 1
+            pos = off + got
+            hole = _hole_span(self._offs, self.fb.blocks, self.length, pos)
+            if hole > 0:
+                fill = min(hole, n - got)
+                ctypes.memset(ptr + got, 0, fill)
+                got += fill
+                continue
+            idx = bisect.bisect_right(self._offs, pos) - 1
             if idx < 0 or idx >= len(self._readers):
                 break
             lb = self.fb.blocks[idx]
-            boff = off + got - lb.offset
+            boff = pos - lb.offset
             want = min(n - got, lb.block.length - boff)
-            if want <= 0:
-                break
             got += self._readers[idx].read_to_ptr(boff, ptr + got, want, False)
         return got
 
@@ -341,7 +370,15 @@ class FsReader:
             return await self._pread_parallel(off, out, out_off, n)
         got = 0
         while got < n:
-            idx, lb, boff = self._block_at(off + got)
+            pos = off + got
+            hole = _hole_span(self._block_offsets(), self.fb.blocks,
+                              self.length, pos)
+            if hole > 0:
+                fill = min(hole, n - got)
+                out[out_off + got:out_off + got + fill] = b"\x00" * fill
+                got += fill
+                continue
+            idx, lb, boff = self._block_at(pos)
             want = min(n - got, lb.block.length - boff)
             if want <= 0:
                 break
@@ -352,6 +389,13 @@ class FsReader:
                 break
             got += rn
         return got
+
+    def _block_offsets(self) -> list:
+        offs = getattr(self, "_block_offs", None)
+        if offs is None or len(offs) != len(self.fb.blocks):
+            offs = [b.offset for b in self.fb.blocks]
+            self._block_offs = offs
+        return offs
 
     async def _read_with_fallback(self, idx, reader, boff, out, out_off, want):
         try:
@@ -380,7 +424,15 @@ class FsReader:
     async def _pread_slice(self, off, out, out_off, n) -> int:
         got = 0
         while got < n:
-            idx, lb, boff = self._block_at(off + got)
+            pos = off + got
+            hole = _hole_span(self._block_offsets(), self.fb.blocks,
+                              self.length, pos)
+            if hole > 0:
+                fill = min(hole, n - got)
+                out[out_off + got:out_off + got + fill] = b"\x00" * fill
+                got += fill
+                continue
+            idx, lb, boff = self._block_at(pos)
             want = min(n - got, lb.block.length - boff)
             if want <= 0:
                 break

@@ -80,26 +80,33 @@ class SyncReadState:
 
     def read_into(self, off: int, out, out_off: int, n: int) -> int:
         import bisect
+
+        from curvine_amd.client.reader import _hole_span
         n = max(0, min(n, self.length - off))
         if self._offs is None:
             self._offs = [b.offset for b in self.fb.blocks]
         got = 0
         while got < n:
-            idx = bisect.bisect_right(self._offs, off + got) - 1
+            pos = off + got
+            hole = _hole_span(self._offs, self.fb.blocks, self.length, pos)
+            if hole > 0:
+                fill = min(hole, n - got)
+                out[out_off + got:out_off + got + fill] = b"\x00" * fill
+                got += fill
+                continue
+            idx = bisect.bisect_right(self._offs, pos) - 1
             if idx < 0 or idx >= len(self.fb.blocks):
                 break
             lb = self.fb.blocks[idx]
-            boff = off + got - lb.offset
+            boff = pos - lb.offset
             want = min(n - got, lb.block.length - boff)
-            if want <= 0:
-                break
             r = self._local.get(idx)
             if r is None:
                 r = self._open_local(idx, lb)
             if r is not None:
                 got += r.read_into(boff, out, out_off + got, want)
             else:
-                got += self._read_remote(off + got, out, out_off + got, want)
+                got += self._read_remote(pos, out, out_off + got, want)
         return got
 
     def _open_local(self, idx: int, lb):
@@ -120,19 +127,27 @@ class SyncReadState:
         blocks DMA directly to `ptr` (hipMemcpyAsync D2H onto a pooled
         stream); file/remote blocks fall back to the view."""
         import bisect
+        import ctypes
+
+        from curvine_amd.client.reader import _hole_span
         n = max(0, min(n, self.length - off))
         if self._offs is None:
             self._offs = [b.offset for b in self.fb.blocks]
         got = 0
         while got < n:
-            idx = bisect.bisect_right(self._offs, off + got) - 1
+            pos = off + got
+            hole = _hole_span(self._offs, self.fb.blocks, self.length, pos)
+            if hole > 0:
+                fill = min(hole, n - got)
+                ctypes.memset(ptr + got, 0, fill)
+                got += fill
+                continue
+            idx = bisect.bisect_right(self._offs, pos) - 1
             if idx < 0 or idx >= len(self.fb.blocks):
                 break
             lb = self.fb.blocks[idx]
-            boff = off + got - lb.offset
+            boff = pos - lb.offset
             want = min(n - got, lb.block.length - boff)
-            if want <= 0:
-                break
             r = self._local.get(idx)
             if r is None:
                 r = self._open_local(idx, lb)
@@ -141,7 +156,7 @@ class SyncReadState:
             elif r is not None:
                 got += r.read_into(boff, view, got, want)
             else:
-                got += self._read_remote(off + got, view, got, want)
+                got += self._read_remote(pos, view, got, want)
         return got
 
     def _read_remote(self, off: int, out, out_off: int, n: int) -> int:
@@ -624,8 +639,12 @@ class CurvineFuseFs:
                 if h.node_id == nodeid and h.writer is not None:
                     import copy
                     st2 = copy.copy(st)
+                    # the SHARED append end can be past this handle's own
+                    # position (another handle appended, or an extending
+                    # truncate zero-filled through the shared writer)
                     st2.length = max(st.length,
-                                     h.write_pos + self._native_write_extra(h))
+                                     h.write_pos + self._native_write_extra(h),
+                                     getattr(h.writer, "pos", 0))
                     return st2
         return st
 
@@ -671,7 +690,21 @@ class CurvineFuseFs:
             return cur
         if size < cur.length:
             return self.call(self.fs.resize(path, size))
-        raise OSError(errno.ENOTSUP, "extending truncate not supported yet")
+        # extending truncate (ftruncate growth / posix_fallocate): with
+        # an open writer extend its sparse watermark so close() keeps
+        # the new size; otherwise grow the metadata length — the tail
+        # reads back as a hole (BlockReaderHole zeros)
+        with self.handles_lock:
+            shared = self.shared_writers.get(nodeid)
+        if shared is not None:
+            with shared.wlock:
+                nh = shared.native_handle
+                if nh is not None:
+                    self._sync_native_write(nh)
+                if size > shared.pos:
+                    shared.write_zeros(size - shared.pos)
+            return self.stat_path(nodeid, path)
+        return self.call(self.fs.resize(path, size))
 
     def op_mkdir(self, nodeid, body, ctx):
         mode, _umask = abi.MKDIR_IN.unpack_from(body, 0)
@@ -819,6 +852,10 @@ class CurvineFuseFs:
         h.status = st
         node = self.child_node(nodeid, name)
         h.node_id = node.id
+        # register like op_open's write branch: later write-opens (and
+        # extending truncate) must find this live writer
+        with self.handles_lock:
+            self.shared_writers[node.id] = h.writer
         self.invalidate(node.id)
         entry = self.entry_out(node, st)
         open_out = abi.OPEN_OUT.pack(h.fh, 0, 0)
@@ -1163,8 +1200,19 @@ class CurvineFuseFs:
 
     def op_fallocate(self, nodeid, body, ctx):
         fh, offset, length, mode, _ = abi.FALLOCATE_IN.unpack_from(body, 0)
+        FALLOC_FL_KEEP_SIZE = 0x01
+        if mode == FALLOC_FL_KEEP_SIZE:
+            return b""   # preallocation hint; arenas reserve per block
         if mode != 0:
             raise OSError(errno.ENOTSUP, "fallocate mode")
+        # plain fallocate extends the file when offset+length > size
+        # (posix_fallocate contract; write_handler.rs resize analog)
+        path = self.node_path(nodeid)
+        st = self.stat_path(nodeid, path)
+        st = self._adjust_writing_size(nodeid, st)
+        if offset + length > st.length:
+            self.invalidate(nodeid)
+            self._truncate(nodeid, path, offset + length, fh)
         return b""
 
     def op_lseek(self, nodeid, body, ctx):

@@ -380,3 +380,71 @@ def test_setlkw_interrupted_by_signal(mount):
     finally:
         holder.terminate()
         holder.wait(timeout=10)
+
+
+def test_truncate_extend_closed_file(mount):
+    """ftruncate/truncate growth on a closed file: size grows, the tail
+    reads back as zeros (hole)."""
+    mnt = mount[0]
+    p = f"{mnt}/extend.bin"
+    with open(p, "wb") as f:
+        f.write(b"abc")
+    os.truncate(p, 4096)
+    assert os.stat(p).st_size == 4096
+    data = open(p, "rb").read()
+    assert data[:3] == b"abc" and len(data) == 4096
+    assert data[3:] == b"\0" * 4093
+
+
+def test_ftruncate_extend_open_writer(mount):
+    """Growth via an open write handle persists across close."""
+    mnt = mount[0]
+    p = f"{mnt}/extend2.bin"
+    with open(p, "wb") as f:
+        f.write(b"xy")
+        f.flush()
+        os.ftruncate(f.fileno(), 1 << 20)
+        assert os.fstat(f.fileno()).st_size == 1 << 20
+    assert os.stat(p).st_size == 1 << 20
+    data = open(p, "rb").read()
+    assert data[:2] == b"xy" and data[2:] == b"\0" * ((1 << 20) - 2)
+
+
+def test_posix_fallocate_extends(mount):
+    """posix_fallocate allocates AND grows the size; KEEP_SIZE does not
+    change the size."""
+    mnt = mount[0]
+    p = f"{mnt}/falloc.bin"
+    with open(p, "wb") as f:
+        f.write(b"data")
+        f.flush()
+        os.posix_fallocate(f.fileno(), 0, 8192)
+        assert os.fstat(f.fileno()).st_size == 8192
+    assert os.stat(p).st_size == 8192
+    assert open(p, "rb").read()[:4] == b"data"
+
+    q = f"{mnt}/falloc_keep.bin"
+    libc = ctypes.CDLL(None, use_errno=True)
+    with open(q, "wb") as f:
+        f.write(b"1234")
+        f.flush()
+        FALLOC_FL_KEEP_SIZE = 0x01
+        rc = libc.fallocate(f.fileno(), FALLOC_FL_KEEP_SIZE,
+                            ctypes.c_long(0), ctypes.c_long(1 << 16))
+        assert rc == 0, os.strerror(ctypes.get_errno())
+        assert os.fstat(f.fileno()).st_size == 4
+
+
+def test_o_append_positioning(mount):
+    """O_APPEND writes land at EOF even after another handle extends the
+    file."""
+    mnt = mount[0]
+    p = f"{mnt}/append.bin"
+    with open(p, "wb") as f:
+        f.write(b"base")
+    fd = os.open(p, os.O_WRONLY | os.O_APPEND)
+    try:
+        os.write(fd, b"-tail")
+    finally:
+        os.close(fd)
+    assert open(p, "rb").read() == b"base-tail"
