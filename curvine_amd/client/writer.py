@@ -46,6 +46,15 @@ class FsWriter:
             raise err.FsError("writer closed")
         data = memoryview(data)
         total = len(data)
+        # zero-copy fast path: with nothing buffered, large inputs go to
+        # the block writers directly — no bytearray staging copy, no
+        # bytes() slice copy; the event loop thread does no memcpy at all
+        # (the copies run on executor threads / in C++ with the GIL
+        # released), which is what bounds multi-file ingest throughput
+        if not self._buf and total >= self.chunk_size:
+            cut = total - (total % self.chunk_size)
+            await self._write_view(data[:cut])
+            data = data[cut:]
         while len(data) > 0:
             room = self.chunk_size - len(self._buf)
             take = min(room, len(data))
@@ -59,13 +68,16 @@ class FsWriter:
     async def _flush_chunk(self) -> None:
         buf = self._buf
         self._buf = bytearray()
+        await self._write_view(memoryview(buf))
+
+    async def _write_view(self, buf) -> None:
         off = 0
         while off < len(buf):
             if self._block is None:
                 await self._next_block()
             room = self.block_size - self._block_pos
             take = min(room, len(buf) - off)
-            chunk = bytes(buf[off:off + take])
+            chunk = buf[off:off + take]
             # star fan-out: all replicas in parallel
             results = await asyncio.gather(
                 *[w.write(chunk) for w in self._writers],
