@@ -406,11 +406,13 @@ static void dev_read(Arena* a, uint64_t off, uint8_t* dst, uint64_t n) {
   BouncePair* bp = bounce_acquire(a->device);
   hipStream_t s = thread_stream(a->device);
   const uint8_t* src = (const uint8_t*)a->base + off;
-  uint64_t nchunks = (n + kBounceSz - 1) / kBounceSz;
+  uint64_t cs = std::min<uint64_t>(kBounceSz,
+                                   std::max<uint64_t>(1u << 20, (n + 1) / 2));
+  uint64_t nchunks = (n + cs - 1) / cs;
   try {
     for (uint64_t c = 0; c < nchunks; ++c) {
-      uint64_t coff = c * kBounceSz;
-      uint64_t clen = std::min<uint64_t>(kBounceSz, n - coff);
+      uint64_t coff = c * cs;
+      uint64_t clen = std::min<uint64_t>(cs, n - coff);
       int slot = (int)(c & 1);
       HIP_CHECK(hipMemcpyAsync(bp->pin[slot], src + coff, clen,
                                hipMemcpyDeviceToHost, s));
@@ -419,16 +421,16 @@ static void dev_read(Arena* a, uint64_t off, uint8_t* dst, uint64_t n) {
         // drain the previous chunk while this one is in flight
         int prev = (int)((c - 1) & 1);
         HIP_CHECK(hipEventSynchronize(bp->ev[prev]));
-        uint64_t poff = (c - 1) * kBounceSz;
+        uint64_t poff = (c - 1) * cs;
         std::memcpy(dst + poff, bp->pin[prev],
-                    std::min<uint64_t>(kBounceSz, n - poff));
+                    std::min<uint64_t>(cs, n - poff));
       }
     }
     int last = (int)((nchunks - 1) & 1);
     HIP_CHECK(hipEventSynchronize(bp->ev[last]));
-    uint64_t loff = (nchunks - 1) * kBounceSz;
+    uint64_t loff = (nchunks - 1) * cs;
     std::memcpy(dst + loff, bp->pin[last],
-                std::min<uint64_t>(kBounceSz, n - loff));
+                std::min<uint64_t>(cs, n - loff));
   } catch (...) {
     (void)hipStreamSynchronize(s);   // no in-flight DMA may outlive bp
     bounce_release(bp);
@@ -450,11 +452,15 @@ static void dev_write(Arena* a, uint64_t off, const uint8_t* src, uint64_t n) {
   BouncePair* bp = bounce_acquire(a->device);
   hipStream_t s = thread_stream(a->device);
   uint8_t* dst = (uint8_t*)a->base + off;
-  uint64_t nchunks = (n + kBounceSz - 1) / kBounceSz;
+  // chunk so every call gets >= 2 chunks: the staging memcpy of chunk
+  // k overlaps the H2D DMA of chunk k-1 even for a single 4 MiB write
+  uint64_t cs = std::min<uint64_t>(kBounceSz,
+                                   std::max<uint64_t>(1u << 20, (n + 1) / 2));
+  uint64_t nchunks = (n + cs - 1) / cs;
   try {
     for (uint64_t c = 0; c < nchunks; ++c) {
-      uint64_t coff = c * kBounceSz;
-      uint64_t clen = std::min<uint64_t>(kBounceSz, n - coff);
+      uint64_t coff = c * cs;
+      uint64_t clen = std::min<uint64_t>(cs, n - coff);
       int slot = (int)(c & 1);
       if (c >= 2) HIP_CHECK(hipEventSynchronize(bp->ev[slot]));
       std::memcpy(bp->pin[slot], src + coff, clen);
