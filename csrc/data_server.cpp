@@ -1022,6 +1022,218 @@ static py::tuple data_read_into(const std::string& host, int port,
   return py::make_tuple(int(final_status), py::bytes(err_hdr), got);
 }
 
+// ------------------------------------------------ streaming write session
+//
+// The client-side analog of the reference's BlockWriter remote stream
+// (block_writer.rs open/write/commit over one connection), kept in C++
+// so FsWriter's executor threads stream chunks GIL-free: dw_open does
+// the Open round-trip on a pooled conn, dw_write sends one windowed
+// chunk frame, dw_commit completes (or dw_abort cancels) and returns
+// the final header.
+
+struct DwSession {
+  std::string host;
+  int port = 0;
+  int fd = -1;
+  uint64_t req = 0;
+  uint32_t seq = 0;
+  int inflight = 0;
+  int window = 8;
+  bool failed = false;
+};
+
+static std::mutex g_dw_mu;
+static std::unordered_map<int64_t, DwSession> g_dw;
+static int64_t g_dw_next = 1;
+
+static bool dw_read_reply(int fd, uint8_t* st_out, std::string* hdr_out) {
+  uint8_t proto[kMetaProto];
+  if (!dc_read_exact(fd, proto, kMetaProto)) return false;
+  uint32_t hlen = rd_u32be(proto);
+  uint32_t dlen = rd_u32be(proto + 4);
+  if (hlen > kMetaMaxLen || dlen > kMetaMaxLen) return false;
+  std::vector<uint8_t> tmp(hlen + dlen);
+  if ((hlen + dlen) && !dc_read_exact(fd, tmp.data(), hlen + dlen))
+    return false;
+  *st_out = proto[9] >> 4;
+  if (hdr_out) hdr_out->assign((const char*)tmp.data(), hlen);
+  return true;
+}
+
+// (handle>0, status, header) — handle 0 on connect failure or rejection
+static py::tuple dw_open(const std::string& host, int port, int64_t block_id,
+                         int64_t reserve, const std::string& tier,
+                         bool reopen, int window) {
+  std::string hdr;
+  uint8_t st = 5;
+  int fd = -1;
+  uint64_t req = 0;
+  {
+    py::gil_scoped_release rel;
+    req = g_dc_req.fetch_add(1);
+    std::string h;
+    h.push_back(char(0x84));
+    mp_str(h, "block_id", 8);
+    mp_uint(h, uint64_t(block_id));
+    mp_str(h, "reserve", 7);
+    mp_uint(h, uint64_t(reserve));
+    mp_str(h, "tier", 4);
+    mp_str(h, tier);
+    mp_str(h, "reopen", 6);
+    h.push_back(char(reopen ? 0xc3 : 0xc2));
+    std::string head = meta_proto(uint32_t(h.size()), 0, kCodeWriteBlock,
+                                  uint8_t(1), req, 0);
+    head += h;
+    for (int attempt = 0; attempt < 2; attempt++) {
+      fd = attempt == 0 ? dc_acquire(host, port) : dc_connect(host, port);
+      if (fd < 0) continue;
+      if (fd_write_all(fd, head.data(), head.size()) &&
+          dw_read_reply(fd, &st, &hdr))
+        break;
+      close(fd);
+      fd = -1;
+    }
+  }
+  if (fd < 0) return py::make_tuple(int64_t(0), 5, py::bytes());
+  if (st == 5) {  // rejected: the stream is clean, keep the conn pooled
+    dc_release(host, port, fd, true);
+    return py::make_tuple(int64_t(0), 5, py::bytes(hdr));
+  }
+  std::lock_guard<std::mutex> g(g_dw_mu);
+  int64_t id = g_dw_next++;
+  DwSession& ses = g_dw[id];
+  ses.host = host;
+  ses.port = port;
+  ses.fd = fd;
+  ses.req = req;
+  ses.seq = 1;
+  ses.window = window > 0 ? window : 8;
+  return py::make_tuple(id, int(st), py::bytes(hdr));
+}
+
+static DwSession* dw_get(int64_t id) {
+  std::lock_guard<std::mutex> g(g_dw_mu);
+  auto it = g_dw.find(id);
+  return it == g_dw.end() ? nullptr : &it->second;
+}
+
+// send one data frame (chunked internally), draining acks past the
+// window.  Returns false when the stream failed (caller aborts).
+static bool dw_write(int64_t id, py::buffer src, uint64_t src_off,
+                     int64_t n, int64_t chunk) {
+  DwSession* ses = dw_get(id);
+  if (!ses || ses->failed) return false;
+  py::buffer_info info = src.request(false);
+  uint64_t cap = (uint64_t)info.size * (uint64_t)info.itemsize;
+  if (src_off + uint64_t(n) > cap)
+    throw std::runtime_error("src buffer too small");
+  const uint8_t* in = (const uint8_t*)info.ptr + src_off;
+  bool ok = true;
+  {
+    py::gil_scoped_release rel;
+    int64_t pos = 0;
+    if (chunk <= 0) chunk = 4 << 20;
+    while (pos < n && ok) {
+      int64_t cn = std::min<int64_t>(chunk, n - pos);
+      std::string ph = meta_proto(0, uint32_t(cn), kCodeWriteBlock,
+                                  uint8_t(2), ses->req, ses->seq++);
+      struct iovec iov[2] = {{(void*)ph.data(), ph.size()},
+                             {(void*)(in + pos), size_t(cn)}};
+      struct msghdr mh = {};
+      mh.msg_iov = iov;
+      mh.msg_iovlen = 2;
+      size_t sent = 0, total = ph.size() + size_t(cn);
+      while (sent < total && ok) {
+        ssize_t w = sendmsg(ses->fd, &mh, MSG_NOSIGNAL);
+        if (w < 0) {
+          if (errno == EINTR) continue;
+          ok = false;
+          break;
+        }
+        sent += size_t(w);
+        size_t adv = size_t(w);
+        for (int i = 0; i < 2 && adv; i++) {
+          size_t take = std::min(adv, iov[i].iov_len);
+          iov[i].iov_base = (char*)iov[i].iov_base + take;
+          iov[i].iov_len -= take;
+          adv -= take;
+        }
+        while (mh.msg_iovlen && mh.msg_iov->iov_len == 0) {
+          mh.msg_iov++;
+          mh.msg_iovlen--;
+        }
+      }
+      pos += cn;
+      ses->inflight++;
+      while (ok && ses->inflight >= ses->window) {
+        uint8_t st = 0;
+        ok = dw_read_reply(ses->fd, &st, nullptr);
+        ses->inflight--;
+        if (st == 5) ok = false;
+      }
+    }
+  }
+  if (!ok) ses->failed = true;
+  return ok;
+}
+
+static void dw_close(int64_t id, bool keep) {
+  std::lock_guard<std::mutex> g(g_dw_mu);
+  auto it = g_dw.find(id);
+  if (it == g_dw.end()) return;
+  dc_release(it->second.host, it->second.port, it->second.fd, keep);
+  g_dw.erase(it);
+}
+
+// (status, final_header): drain acks, send Complete (or no_finalize
+// close), read the final reply
+static py::tuple dw_commit(int64_t id, int64_t length, bool no_finalize) {
+  DwSession* ses = dw_get(id);
+  if (!ses) return py::make_tuple(5, py::bytes());
+  std::string hdr;
+  uint8_t st = 5;
+  bool ok = !ses->failed;
+  {
+    py::gil_scoped_release rel;
+    while (ok && ses->inflight > 0) {
+      uint8_t ast = 0;
+      ok = dw_read_reply(ses->fd, &ast, nullptr);
+      ses->inflight--;
+      if (ast == 5) ok = false;
+    }
+    if (ok) {
+      std::string fh;
+      fh.push_back(char(0x81));
+      if (no_finalize) {
+        mp_str(fh, "no_finalize", 11);
+        fh.push_back(char(0xc3));
+      } else {
+        mp_str(fh, "length", 6);
+        mp_uint(fh, uint64_t(length));
+      }
+      std::string cf = meta_proto(uint32_t(fh.size()), 0, kCodeWriteBlock,
+                                  uint8_t(3), ses->req, ses->seq++);
+      cf += fh;
+      ok = fd_write_all(ses->fd, cf.data(), cf.size());
+      if (ok) ok = dw_read_reply(ses->fd, &st, &hdr);
+    }
+  }
+  dw_close(id, ok && st != 5);
+  return py::make_tuple(ok ? int(st) : 5, py::bytes(hdr));
+}
+
+static void dw_abort(int64_t id) {
+  DwSession* ses = dw_get(id);
+  if (!ses) return;
+  {
+    py::gil_scoped_release rel;
+    std::string cf = meta_proto(0, 0, kCodeWriteBlock, uint8_t(4) /*Cancel*/,
+                                ses->req, ses->seq++);
+    (void)fd_write_all(ses->fd, cf.data(), cf.size());
+  }
+  dw_close(id, false);
+}
+
 // returns (resp_status, final_header_bytes)
 static py::tuple data_write_from(const std::string& host, int port,
                                  int64_t block_id, int64_t reserve,

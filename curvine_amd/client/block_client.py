@@ -99,7 +99,9 @@ class BlockWriterRemote:
 
     async def write(self, data) -> None:
         await self._ensure_open()
-        await self.stream.send({}, bytes(data), Status.Running)
+        # no bytes() copy: the transport copies what it cannot send
+        # immediately, so a memoryview is safe to pass through
+        await self.stream.send({}, data, Status.Running)
         self.inflight += 1
         self.pos += len(data)
         while self.inflight >= self.WINDOW:
@@ -108,7 +110,7 @@ class BlockWriterRemote:
 
     async def pwrite(self, off: int, data) -> None:
         await self._ensure_open()
-        await self.stream.send({"off": off}, bytes(data), Status.Running)
+        await self.stream.send({"off": off}, data, Status.Running)
         self.inflight += 1
         while self.inflight >= self.WINDOW:
             await self.stream.recv()
