@@ -1297,6 +1297,10 @@ PYBIND11_MODULE(_native, m) {
   m.def("data_stats", &data_stats);
   m.def("data_read_into", &data_read_into);
   m.def("data_write_from", &data_write_from);
+  m.def("dw_open", &dw_open);
+  m.def("dw_write", &dw_write);
+  m.def("dw_commit", &dw_commit);
+  m.def("dw_abort", &dw_abort);
   m.def("lz4_compress", &lz4_compress_py);
   m.def("lz4_decompress", &lz4_decompress_py);
   m.def("arena_lz4_decompress", &arena_lz4_decompress);

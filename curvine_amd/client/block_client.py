@@ -73,7 +73,10 @@ class BlockWriterLocal:
 
 
 class BlockWriterRemote:
-    """Streaming WriteBlock over RPC with a pipelined ack window."""
+    """Streaming WriteBlock: bulk appends take the native C++ session
+    (dw_open/dw_write in csrc/data_server.cpp — pooled conn, GIL-free
+    sendmsg loop with a pipelined ack window, one executor hop per
+    chunk); the asyncio stream remains for pwrite and as the fallback."""
 
     WINDOW = 4
 
@@ -87,6 +90,8 @@ class BlockWriterRemote:
         self.stream: Optional[RpcStream] = None
         self.inflight = 0
         self.pos = 0
+        self._dw = None         # native session id
+        self._dw_tried = False
 
     async def _ensure_open(self) -> None:
         if self.stream is not None:
@@ -97,7 +102,41 @@ class BlockWriterRemote:
             {"block_id": self.block_id, "reserve": self.reserve,
              "tier": self.tier, "reopen": self.reopen}, status=Status.Open)
 
+    async def _ensure_native(self) -> bool:
+        """Open the native write session once; False -> asyncio path."""
+        if self._dw is not None:
+            return True
+        if self._dw_tried or self.stream is not None:
+            return False
+        self._dw_tried = True
+        lib = _native_data_lib()
+        if lib is None or not hasattr(lib, "dw_open"):
+            return False
+        loop = asyncio.get_running_loop()
+        hid, st, hdr = await loop.run_in_executor(
+            None, lib.dw_open, self.addr.hostname, self.addr.rpc_port,
+            self.block_id, self.reserve, self.tier, self.reopen,
+            self.WINDOW)
+        if st == 5 and hdr:
+            _raise_wire_error(hdr)
+        if hid <= 0:
+            return False
+        self._dw = hid
+        return True
+
     async def write(self, data) -> None:
+        if await self._ensure_native():
+            lib = _native_data_lib()
+            loop = asyncio.get_running_loop()
+            ok = await loop.run_in_executor(
+                None, lib.dw_write, self._dw, data, 0, len(data), 4 << 20)
+            if not ok:
+                lib.dw_abort(self._dw)
+                self._dw = None
+                raise ConnectError(
+                    f"native write stream to {self.addr.hostname} failed")
+            self.pos += len(data)
+            return
         await self._ensure_open()
         # no bytes() copy: the transport copies what it cannot send
         # immediately, so a memoryview is safe to pass through
@@ -109,6 +148,11 @@ class BlockWriterRemote:
             self.inflight -= 1
 
     async def pwrite(self, off: int, data) -> None:
+        if self._dw is not None:
+            # positional rewrites need the richer asyncio stream; a
+            # writer does either bulk appends or rewrites, not both on
+            # one handle (FsWriter reopens for rewrites)
+            await self._dw_finish(no_finalize=True)
         await self._ensure_open()
         await self.stream.send({"off": off}, data, Status.Running)
         self.inflight += 1
@@ -116,7 +160,27 @@ class BlockWriterRemote:
             await self.stream.recv()
             self.inflight -= 1
 
+    async def _dw_finish(self, length: int | None = None,
+                         no_finalize: bool = False) -> dict:
+        import msgpack
+        lib = _native_data_lib()
+        loop = asyncio.get_running_loop()
+        hid, self._dw = self._dw, None
+        st, hdr = await loop.run_in_executor(
+            None, lib.dw_commit, hid, length if length is not None else -1,
+            no_finalize)
+        if st == 5:
+            _raise_wire_error(hdr)
+        try:
+            return msgpack.unpackb(hdr, raw=False) if hdr else {}
+        except Exception:  # noqa: BLE001
+            return {}
+
     async def commit(self, length: int | None = None) -> str:
+        if self._dw is not None:
+            h = await self._dw_finish(length, no_finalize=length is None)
+            self.last_crc = h.get("crc32c")
+            return h.get("tier", "")
         await self._ensure_open()
         while self.inflight > 0:
             await self.stream.recv()
@@ -129,6 +193,14 @@ class BlockWriterRemote:
         return reply.header.get("tier", "")
 
     async def abort(self) -> None:
+        if self._dw is not None:
+            lib = _native_data_lib()
+            hid, self._dw = self._dw, None
+            try:
+                await asyncio.get_running_loop().run_in_executor(
+                    None, lib.dw_abort, hid)
+            except Exception:  # noqa: BLE001
+                pass
         if self.stream is not None:
             try:
                 await self.stream.send({}, b"", Status.Cancel)

@@ -60,6 +60,9 @@ def test_remote_read_write_native_mem_tier(tmp_path):
             while pos < len(wdata):
                 await w.write(wdata[pos:pos + (1 << 20)])
                 pos += 1 << 20
+            from curvine_amd.client.block_client import _native_data_lib
+            if hasattr(_native_data_lib(), "dw_open"):
+                assert w._dw is not None   # native streaming session in use
             tier = await w.commit(len(wdata))
             assert tier == "MEM"
             st = fe.stats()
