@@ -1177,6 +1177,26 @@ static bool dw_write(int64_t id, py::buffer src, uint64_t src_off,
   return ok;
 }
 
+// read acks until nothing is in flight: callers that are about to
+// rewrite bytes on a SEPARATE reopen stream must know the appends have
+// been consumed by the server first (two connections are unordered)
+static bool dw_drain(int64_t id) {
+  DwSession* ses = dw_get(id);
+  if (!ses || ses->failed) return false;
+  bool ok = true;
+  {
+    py::gil_scoped_release rel;
+    while (ok && ses->inflight > 0) {
+      uint8_t st = 0;
+      ok = dw_read_reply(ses->fd, &st, nullptr);
+      ses->inflight--;
+      if (st == 5) ok = false;
+    }
+  }
+  if (!ok) ses->failed = true;
+  return ok;
+}
+
 static void dw_close(int64_t id, bool keep) {
   std::lock_guard<std::mutex> g(g_dw_mu);
   auto it = g_dw.find(id);

@@ -1299,6 +1299,7 @@ PYBIND11_MODULE(_native, m) {
   m.def("data_write_from", &data_write_from);
   m.def("dw_open", &dw_open);
   m.def("dw_write", &dw_write);
+  m.def("dw_drain", &dw_drain);
   m.def("dw_commit", &dw_commit);
   m.def("dw_abort", &dw_abort);
   m.def("lz4_compress", &lz4_compress_py);

@@ -149,10 +149,26 @@ class BlockWriterRemote:
 
     async def pwrite(self, off: int, data) -> None:
         if self._dw is not None:
-            # positional rewrites need the richer asyncio stream; a
-            # writer does either bulk appends or rewrites, not both on
-            # one handle (FsWriter reopens for rewrites)
-            await self._dw_finish(no_finalize=True)
+            # keep the native append session; rewrites ride a separate
+            # reopen stream.  Drain the append window first — the two
+            # connections are unordered, so the bytes being rewritten
+            # must already be consumed by the server
+            lib = _native_data_lib()
+            ok = await asyncio.get_running_loop().run_in_executor(
+                None, lib.dw_drain, self._dw)
+            if not ok:
+                lib.dw_abort(self._dw)
+                self._dw = None
+                raise ConnectError(
+                    f"native write stream to {self.addr.hostname} failed")
+            if self.stream is None:
+                client = await factory().get(self.addr.hostname,
+                                             self.addr.rpc_port)
+                self.stream = client.stream(RpcCode.WriteBlock)
+                await self.stream.call(
+                    {"block_id": self.block_id, "reserve": 0,
+                     "tier": self.tier, "reopen": True},
+                    status=Status.Open)
         await self._ensure_open()
         await self.stream.send({"off": off}, data, Status.Running)
         self.inflight += 1
@@ -178,6 +194,16 @@ class BlockWriterRemote:
 
     async def commit(self, length: int | None = None) -> str:
         if self._dw is not None:
+            if self.stream is not None:
+                # close the rewrite side stream first, without finalize
+                while self.inflight > 0:
+                    await self.stream.recv()
+                    self.inflight -= 1
+                await self.stream.send({"no_finalize": True}, b"",
+                                       Status.Complete)
+                await self.stream.recv()
+                self.stream.close()
+                self.stream = None
             h = await self._dw_finish(length, no_finalize=length is None)
             self.last_crc = h.get("crc32c")
             return h.get("tier", "")
