@@ -263,3 +263,40 @@ def test_fswriter_pwrite_at(tmp_path):
         await mc.stop()
 
     asyncio.new_event_loop().run_until_complete(main())
+
+
+def test_resize_extend_reads_tail_hole(loop, tmp_path):
+    """Growing a file past its cached blocks (extending truncate): the
+    uncovered tail reads back as zeros through both the async reader
+    and the sync short-circuit reader."""
+    async def main():
+        async with MiniCluster(tmp_dir=str(tmp_path)) as mc:
+            fs = mc.fs()
+            data = os.urandom(3 << 20)
+            await fs.write_all("/grow.bin", data)
+            await fs.resize("/grow.bin", 8 << 20)
+            st = await fs.file_status("/grow.bin")
+            assert st.length == 8 << 20
+
+            back = await fs.read_all("/grow.bin")
+            assert len(back) == 8 << 20
+            assert back[:len(data)] == data
+            assert back[len(data):] == b"\0" * ((8 << 20) - len(data))
+
+            # ranged read fully inside the hole
+            r = await fs.open("/grow.bin")
+            mid = await r.pread(5 << 20, 1 << 20)
+            assert mid == b"\0" * (1 << 20)
+            # straddling read
+            straddle = await r.pread((3 << 20) - 7, 100)
+            assert straddle == data[-7:] + b"\0" * 93
+
+            # sync short-circuit reader sees the same
+            sr = r.to_sync()
+            buf = bytearray(1 << 20)
+            got = sr.pread_into(6 << 20, buf, 0, 1 << 20)
+            assert got == 1 << 20 and bytes(buf) == b"\0" * (1 << 20)
+            sr.close()
+            r.close()
+            await fs.close()
+    run(loop, main())
