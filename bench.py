@@ -125,9 +125,11 @@ def setup(args, rank, world, dist, has_gpu):
     elif args.no_short_circuit:
         # faithful remote plane: the worker is its OWN process, every
         # byte crosses its streaming RPC over loopback
+        hb = 250 if args.workload == "seqwrite" else 1000
         wp = subprocess.Popen(
             [sys.executable, "-m", "curvine_amd.server_main",
              "--service", "worker", "--master-port", str(master_port),
+             "--worker-port", "0", "--heartbeat-ms", str(hb),
              "--device", str(local_rank if has_gpu else -1),
              "--log-level", "WARNING"] +
             [f"--data-dir={d}" for d in data_dirs],
@@ -667,6 +669,36 @@ def main():
             torch.cuda.synchronize()
 
     rt, conf, daemon_proc, mnt = setup(args, rank, world, dist, has_gpu)
+    try:
+        _run_bench(args, rank, world, dist, torch, rt, mnt, has_gpu,
+                   barrier_sync)
+    finally:
+        # a failed step must not strand the worker/daemon subprocess
+        # (a stranded fixed-port worker breaks the NEXT bench run)
+        if daemon_proc is not None:
+            daemon_proc.terminate()
+            try:
+                daemon_proc.wait(timeout=10)
+            except subprocess.TimeoutExpired:
+                daemon_proc.kill()
+        async def teardown():
+            if rt.fs:
+                await rt.fs.close()
+            if rt.worker:
+                await rt.worker.stop()
+            if rt.master:
+                await rt.master.stop()
+        try:
+            rt.call(teardown())
+        except Exception:  # noqa: BLE001 — best-effort teardown
+            pass
+        rt.stop()
+        if dist is not None:
+            dist.destroy_process_group()
+
+
+def _run_bench(args, rank, world, dist, torch, rt, mnt, has_gpu,
+               barrier_sync):
     log(rank, f"cluster up ({args.path}); writing "
         f"{args.files}x{args.file_size >> 20}MiB")
     if args.workload != "seqwrite":   # writing IS the seqwrite workload
@@ -765,24 +797,6 @@ def main():
         run_scale_sweep(args, rank, world, dist, torch, rt, has_gpu,
                         one_step, barrier_sync)
 
-    # teardown
-    if daemon_proc is not None:
-        daemon_proc.terminate()
-        try:
-            daemon_proc.wait(timeout=10)
-        except subprocess.TimeoutExpired:
-            daemon_proc.kill()
-    async def teardown():
-        if rt.fs:
-            await rt.fs.close()
-        if rt.worker:
-            await rt.worker.stop()
-        if rt.master:
-            await rt.master.stop()
-    rt.call(teardown())
-    rt.stop()
-    if dist is not None:
-        dist.destroy_process_group()
 
 
 if __name__ == "__main__":

@@ -73,6 +73,7 @@ struct FuseLoop {
   std::mutex fwd_mu;
   std::condition_variable fwd_cv;
   std::atomic<bool> stopping{false};
+  std::atomic<uint64_t> reply_errors{0};
   std::atomic<uint64_t> native_reads{0}, native_writes{0}, forwarded{0};
 };
 
@@ -87,6 +88,22 @@ static FuseLoop* get_loop(int id) {
 }
 
 // serve one READ natively; returns false if the fh is not registered
+
+// /dev/fuse replies are all-or-nothing; EINTR retries, ENOENT means the
+// kernel aborted the request (not an error for us).  Anything else is
+// counted — a silently dropped reply would wedge the application's read.
+static bool fuse_reply_write(FuseLoop* L, int fd, const uint8_t* buf,
+                             size_t len) {
+  for (;;) {
+    ssize_t w = write(fd, buf, len);
+    if (w >= 0) return true;
+    if (errno == EINTR) continue;
+    if (errno == ENOENT) return true;   // request aborted by the kernel
+    L->reply_errors.fetch_add(1, std::memory_order_relaxed);
+    return false;
+  }
+}
+
 static bool serve_read(FuseLoop* L, int fd, const FuseInHeader* h,
                        const FuseReadIn* r, uint8_t* reply) {
   std::shared_lock<std::shared_mutex> lk(L->hmu);
@@ -131,8 +148,7 @@ static bool serve_read(FuseLoop* L, int fd, const FuseInHeader* h,
   oh->len = (uint32_t)(sizeof(FuseOutHeader) + got);
   oh->error = 0;
   oh->unique = h->unique;
-  ssize_t w = write(fd, reply, oh->len);
-  (void)w;
+  if (!fuse_reply_write(L, fd, reply, oh->len)) return true;  // counted
   L->native_reads.fetch_add(1, std::memory_order_relaxed);
   return true;
 }
@@ -173,8 +189,8 @@ static bool serve_write(FuseLoop* L, int fd, const FuseInHeader* h,
   rep.oh.unique = h->unique;
   rep.wo.size = w->size;
   rep.wo.pad = 0;
-  ssize_t wr = write(fd, &rep, sizeof(rep));
-  (void)wr;
+  if (!fuse_reply_write(L, fd, (const uint8_t*)&rep, sizeof(rep)))
+    return true;   // bytes are in the arena; the reply loss is counted
   L->native_writes.fetch_add(1, std::memory_order_relaxed);
   return true;
 }
@@ -345,6 +361,7 @@ static py::dict fuse_loop_stats(int id) {
   d["native_reads"] = L->native_reads.load();
   d["native_writes"] = L->native_writes.load();
   d["forwarded"] = L->forwarded.load();
+  d["reply_errors"] = L->reply_errors.load();
   {
     std::shared_lock<std::shared_mutex> lk(L->hmu);
     d["registered_handles"] = L->handles.size();

@@ -364,14 +364,31 @@ static void bounce_release(BouncePair* b) {
 // per-caller-thread stream: zero contention for concurrent small copies
 // (the 4K-IOPS path).  hipMemcpy on the null stream and ROCm's internal
 // pageable-staging lock both serialize; a thread_local stream does not.
-static hipStream_t thread_stream(int device) {
-  thread_local hipStream_t tls[64] = {};
-  if (device < 0 || device >= 64) throw std::runtime_error("bad device");
-  if (!tls[device]) {
-    HIP_CHECK(hipSetDevice(device));
-    HIP_CHECK(hipStreamCreateWithFlags(&tls[device], hipStreamNonBlocking));
+// Streams are destroyed when their thread exits (no leak under thread
+// churn) — except during process teardown, when the HIP runtime may
+// already be gone.
+static std::atomic<bool> g_process_exiting{false};
+static struct ExitFlagSetter {
+  ~ExitFlagSetter() { g_process_exiting.store(true); }
+} g_exit_flag_setter;
+
+struct TlsStreams {
+  hipStream_t s[64] = {};
+  ~TlsStreams() {
+    if (g_process_exiting.load()) return;
+    for (auto x : s)
+      if (x) (void)hipStreamDestroy(x);
   }
-  return tls[device];
+};
+
+static hipStream_t thread_stream(int device) {
+  thread_local TlsStreams tls;
+  if (device < 0 || device >= 64) throw std::runtime_error("bad device");
+  if (!tls.s[device]) {
+    HIP_CHECK(hipSetDevice(device));
+    HIP_CHECK(hipStreamCreateWithFlags(&tls.s[device], hipStreamNonBlocking));
+  }
+  return tls.s[device];
 }
 
 static void dev_read_direct(Arena* a, uint64_t off, uint8_t* dst, uint64_t n) {

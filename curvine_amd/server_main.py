@@ -31,6 +31,10 @@ async def run(args) -> None:
         from curvine_amd.worker.server import Worker
         if args.data_dir:
             conf.worker.data_dirs = args.data_dir
+        if args.worker_port >= 0:
+            conf.worker.rpc_port = args.worker_port
+        if args.heartbeat_ms > 0:
+            conf.worker.heartbeat_interval_ms = args.heartbeat_ms
         servers.append(await Worker(conf, device_id=args.device).start())
     if args.service == "transfer":
         from curvine_amd.transfer import TransferService
@@ -62,6 +66,9 @@ def main(argv=None) -> int:
     p.add_argument("--conf", default=None)
     p.add_argument("--device", type=int, default=-1)
     p.add_argument("--master-port", type=int, default=0)
+    p.add_argument("--worker-port", type=int, default=-1,
+                   help="worker RPC port (0 = ephemeral)")
+    p.add_argument("--heartbeat-ms", type=int, default=0)
     p.add_argument("--data-dir", action="append", default=[])
     p.add_argument("--web", action="store_true")
     p.add_argument("--log-level", default="INFO")
