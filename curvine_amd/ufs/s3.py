@@ -10,6 +10,8 @@ from __future__ import annotations
 
 import datetime
 import hashlib
+import time
+import random
 import hmac
 import urllib.parse
 import xml.etree.ElementTree as ET
@@ -151,18 +153,40 @@ class S3Ufs(UnderFs):
     def _key(self, path: str) -> str:
         return f"{self.prefix}/{path.strip('/')}"
 
+    # transient statuses S3 documents for client retry (throttling and
+    # internal errors), plus connection-level failures
+    _RETRY_STATUS = frozenset({429, 500, 502, 503, 504})
+    _RETRIES = 4
+
     def _req(self, method: str, url: str, data: bytes = b"",
              headers: Optional[dict] = None):
         payload_sha = hashlib.sha256(data).hexdigest()
-        hdrs = sigv4_headers(method, url, self.region, self.access_key,
-                             self.secret_key, payload_sha, extra_headers=headers)
-        r = self.session.request(method, url, data=data or None, headers=hdrs,
-                                 timeout=60)
-        if r.status_code == 404:
-            raise err.FileNotFound(url)
-        if r.status_code >= 300:
-            raise err.UfsError(f"s3 {method} {url}: {r.status_code} {r.text[:200]}")
-        return r
+        last: Exception = err.UfsError(f"s3 {method} {url}: no attempt")
+        for attempt in range(self._RETRIES + 1):
+            if attempt:
+                # exponential backoff with jitter, capped at 8 s
+                time.sleep(min(8.0, (2.0 ** (attempt - 1)) * 0.25)
+                           * (0.5 + random.random()))
+            hdrs = sigv4_headers(method, url, self.region, self.access_key,
+                                 self.secret_key, payload_sha,
+                                 extra_headers=headers)
+            try:
+                r = self.session.request(method, url, data=data or None,
+                                         headers=hdrs, timeout=60)
+            except Exception as e:  # noqa: BLE001 — conn reset/timeout
+                last = err.UfsError(f"s3 {method} {url}: {e}")
+                continue
+            if r.status_code == 404:
+                raise err.FileNotFound(url)
+            if r.status_code in self._RETRY_STATUS:
+                last = err.UfsError(f"s3 {method} {url}: {r.status_code} "
+                                    f"{r.text[:200]}")
+                continue
+            if r.status_code >= 300:
+                raise err.UfsError(f"s3 {method} {url}: {r.status_code} "
+                                   f"{r.text[:200]}")
+            return r
+        raise last
 
     def _get_range(self, key: str, start: int, end: int) -> bytes:
         r = self._req("GET", self._url(key), headers={"range": f"bytes={start}-{end}"})

@@ -15,6 +15,11 @@ class FakeS3:
         self.objects: dict[str, bytes] = {}     # "bucket/key" -> bytes
         self.uploads: dict[str, dict[int, bytes]] = {}
         self.max_keys = max_keys
+        # fault injection: fail the next N requests with this status
+        # (throttling/5xx retry tests)
+        self.fail_next = 0
+        self.fail_status = 503
+        self.requests_seen = 0
         store = self
 
         class Handler(BaseHTTPRequestHandler):
@@ -28,6 +33,14 @@ class FakeS3:
                 return urllib.parse.unquote(parts.path.lstrip("/")), \
                     urllib.parse.parse_qs(parts.query,
                                           keep_blank_values=True)
+
+            def _maybe_fail(self) -> bool:
+                store.requests_seen += 1
+                if store.fail_next > 0:
+                    store.fail_next -= 1
+                    self._reply(store.fail_status, b"injected failure")
+                    return True
+                return False
 
             def _reply(self, code: int, body: bytes = b"", headers=None):
                 self.send_response(code)
@@ -43,6 +56,8 @@ class FakeS3:
                 return self.rfile.read(n) if n else b""
 
             def do_PUT(self):
+                if self._maybe_fail():
+                    return
                 assert self.headers.get("authorization", "").startswith(
                     "AWS4-HMAC-SHA256"), "unsigned request"
                 key, q = self._key()
@@ -79,6 +94,8 @@ class FakeS3:
                 self._reply(400)
 
             def do_GET(self):
+                if self._maybe_fail():
+                    return
                 key, q = self._key()
                 if "list-type" in q:
                     return self._list(key.rstrip("/"), q)

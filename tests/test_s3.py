@@ -153,3 +153,29 @@ def test_s3_load_job_into_cache(s3, tmp_path):
             await mc.stop()
 
     asyncio.new_event_loop().run_until_complete(main())
+
+
+def test_s3_transient_errors_are_retried(s3):
+    """429/5xx and connection failures back off and retry (the S3
+    contract for throttling); hard 4xx does not retry."""
+    from curvine_amd import errors as err
+    from curvine_amd.ufs.s3 import S3Ufs
+
+    ufs = S3Ufs("s3://b/pre", _props(s3))
+    ufs._RETRIES = 3
+    with ufs.create("obj.bin") as w:
+        w.write(b"payload")
+
+    s3.fail_next, s3.fail_status = 2, 503
+    assert ufs.open("obj.bin").read(7) == b"payload"   # survives two 503s
+
+    s3.fail_next, s3.fail_status = 2, 429
+    assert ufs.open("obj.bin").read(7) == b"payload"
+
+    # more consecutive failures than retries -> typed UfsError
+    s3.fail_next, s3.fail_status = 10, 500
+    before = s3.requests_seen
+    with pytest.raises(err.UfsError):
+        ufs.open("obj.bin").read(7)
+    assert s3.requests_seen - before == ufs._RETRIES + 1
+    s3.fail_next = 0
