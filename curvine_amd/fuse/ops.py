@@ -1220,13 +1220,30 @@ class CurvineFuseFs:
         h = self.get_handle(fh)
         length = h.status.length if h.status else 0
         SEEK_DATA, SEEK_HOLE = 3, 4
+        if whence not in (SEEK_DATA, SEEK_HOLE):
+            raise OSError(errno.EINVAL, "whence")
+        if offset >= length:
+            raise OSError(errno.ENXIO, "past eof")
+        # consult the open reader's extent map: ranges the cached blocks
+        # do not cover are holes (sparse files, extending truncate)
+        blocks = getattr(getattr(h, "reader", None), "fb", None)
+        blocks = blocks.blocks if blocks is not None else []
         if whence == SEEK_DATA:
-            if offset >= length:
-                raise OSError(errno.ENXIO, "past eof")
-            return abi.LSEEK_OUT.pack(offset)
-        if whence == SEEK_HOLE:
-            return abi.LSEEK_OUT.pack(length)
-        raise OSError(errno.EINVAL, "whence")
+            for lb in blocks:
+                end = lb.offset + lb.block.length
+                if offset < end:
+                    return abi.LSEEK_OUT.pack(max(offset, lb.offset))
+            raise OSError(errno.ENXIO, "in trailing hole")
+        # SEEK_HOLE: first uncovered byte at or after offset
+        pos = offset
+        for lb in blocks:
+            end = lb.offset + lb.block.length
+            if pos < lb.offset:
+                return abi.LSEEK_OUT.pack(pos)     # gap hole
+            if pos < end:
+                pos = end
+        return abi.LSEEK_OUT.pack(min(pos, length)
+                                  if pos < length else length)
 
     def op_interrupt(self, nodeid, body, ctx):
         """INTERRUPT carries the unique of an in-flight request; blocked

@@ -448,3 +448,101 @@ def test_o_append_positioning(mount):
     finally:
         os.close(fd)
     assert open(p, "rb").read() == b"base-tail"
+
+
+def test_seek_hole_data(mount):
+    """SEEK_HOLE/SEEK_DATA reflect the cached extent map: the tail of an
+    extended file is a hole; dense files report one data segment."""
+    mnt = mount[0]
+    p = f"{mnt}/holes.bin"
+    with open(p, "wb") as f:
+        f.write(b"x" * 8192)
+    os.truncate(p, 1 << 20)          # tail hole [8192, 1 MiB)
+
+    fd = os.open(p, os.O_RDONLY)
+    try:
+        assert os.lseek(fd, 0, os.SEEK_DATA) == 0
+        hole = os.lseek(fd, 0, os.SEEK_HOLE)
+        assert hole == 8192
+        # SEEK_DATA inside the trailing hole: ENXIO
+        with pytest.raises(OSError) as ei:
+            os.lseek(fd, 8192, os.SEEK_DATA)
+        assert ei.value.errno == errno.ENXIO
+        # past EOF: ENXIO
+        with pytest.raises(OSError):
+            os.lseek(fd, 2 << 20, os.SEEK_HOLE)
+    finally:
+        os.close(fd)
+
+    q = f"{mnt}/dense.bin"
+    with open(q, "wb") as f:
+        f.write(b"y" * 4096)
+    fd = os.open(q, os.O_RDONLY)
+    try:
+        assert os.lseek(fd, 0, os.SEEK_HOLE) == 4096   # hole == EOF
+        assert os.lseek(fd, 100, os.SEEK_DATA) == 100
+    finally:
+        os.close(fd)
+
+
+def test_xattr_corpus(mount):
+    """set/get/list/remove xattrs, ENODATA on missing, replace
+    semantics, many attributes listed back."""
+    mnt = mount[0]
+    p = f"{mnt}/xa.bin"
+    open(p, "wb").write(b"z")
+    os.setxattr(p, "user.one", b"1")
+    os.setxattr(p, "user.two", b"22")
+    assert os.getxattr(p, "user.one") == b"1"
+    os.setxattr(p, "user.one", b"replaced")
+    assert os.getxattr(p, "user.one") == b"replaced"
+    names = set(os.listxattr(p))
+    assert {"user.one", "user.two"} <= names
+    os.removexattr(p, "user.two")
+    assert "user.two" not in set(os.listxattr(p))
+    with pytest.raises(OSError) as ei:
+        os.getxattr(p, "user.two")
+    assert ei.value.errno in (errno.ENODATA, 61)
+    with pytest.raises(OSError):
+        os.removexattr(p, "user.gone")
+    # a pile of attributes round-trips through listxattr paging
+    for i in range(40):
+        os.setxattr(p, f"user.k{i:02d}", str(i).encode())
+    got = set(os.listxattr(p))
+    assert all(f"user.k{i:02d}" in got for i in range(40))
+
+
+def test_readdir_stable_under_mutation(mount):
+    """Listing a directory while entries are created/deleted never
+    crashes and yields each stable entry exactly once."""
+    mnt = mount[0]
+    d = f"{mnt}/churn"
+    os.mkdir(d)
+    stable = {f"s{i}" for i in range(50)}
+    for name in stable:
+        open(f"{d}/{name}", "wb").close()
+    import threading
+
+    stop = threading.Event()
+
+    def churn():
+        i = 0
+        while not stop.is_set():
+            name = f"{d}/tmp{i % 7}"
+            try:
+                open(name, "wb").close()
+                os.unlink(name)
+            except OSError:
+                pass
+            i += 1
+
+    th = threading.Thread(target=churn)
+    th.start()
+    try:
+        for _ in range(20):
+            seen = os.listdir(d)
+            assert len(seen) == len(set(seen))   # no duplicates
+            assert stable <= set(seen)
+    finally:
+        stop.set()
+        th.join()
