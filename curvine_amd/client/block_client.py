@@ -218,6 +218,18 @@ class BlockWriterRemote:
         self.last_crc = reply.header.get("crc32c")
         return reply.header.get("tier", "")
 
+    def __del__(self):
+        # a writer dropped without commit/abort must not leak its native
+        # session (pooled fd + C++ entry); socket-only, safe from GC
+        dw = getattr(self, "_dw", None)
+        if dw is not None:
+            try:
+                lib = _native_data_lib()
+                if lib is not None:
+                    lib.dw_abort(dw)
+            except Exception:  # noqa: BLE001 — interpreter teardown
+                pass
+
     async def abort(self) -> None:
         if self._dw is not None:
             lib = _native_data_lib()
