@@ -105,10 +105,59 @@ struct WriteSess {
   std::atomic<uint64_t> pos{0};  // append watermark (finalize default len)
 };
 
+// Receive buffer that can be upgraded to PINNED host memory: once an
+// HBM write session opens on a conn, frames land in hipHostMalloc'd
+// pages so dev_write DMAs straight from the receive buffer (no staging
+// memcpy — the socket recv is the only host copy).
+struct RBuf {
+  uint8_t* p = nullptr;
+  size_t cap = 0;
+  bool pinned = false;
+  std::atomic<bool> want_pinned{false};   // set from the register path
+
+  uint8_t* data() { return p; }
+  size_t capacity() const { return cap; }
+
+  void grow(size_t need, size_t keep) {
+    bool wp = want_pinned.load(std::memory_order_relaxed);
+    bool to_pin = wp && !pinned;
+    if (need <= cap && !to_pin) return;
+    size_t ncap = cap ? cap : 256u << 10;
+    while (ncap < need) ncap *= 2;
+    uint8_t* np = nullptr;
+    bool np_pinned = false;
+    if (wp &&
+        hipHostMalloc((void**)&np, ncap, hipHostMallocDefault) ==
+            hipSuccess) {
+      np_pinned = true;
+    } else {
+      (void)hipGetLastError();
+      np = (uint8_t*)std::malloc(ncap);
+      if (!np) throw std::bad_alloc();
+    }
+    if (keep && p) std::memcpy(np, p, keep);
+    release();
+    p = np;
+    cap = ncap;
+    pinned = np_pinned;
+  }
+
+  void release() {
+    if (!p) return;
+    if (pinned) (void)hipHostFree(p);
+    else std::free(p);
+    p = nullptr;
+    cap = 0;
+    pinned = false;
+  }
+
+  ~RBuf() { release(); }
+};
+
 struct DataConn {
   int fd = -1;
   uint64_t id = 0;
-  std::vector<uint8_t> rbuf;
+  RBuf rbuf;
   size_t rlen = 0;
   std::mutex wmu;
   std::atomic<bool> dead{false};
@@ -576,7 +625,7 @@ static void data_readable(DataServer* S, const std::shared_ptr<DataConn>& c) {
       size_t total = kMetaProto + hlen + dlen;
       if (total > c->rlen) want = total - c->rlen;
     }
-    if (c->rbuf.size() < c->rlen + want) c->rbuf.resize(c->rlen + want);
+    c->rbuf.grow(c->rlen + want, c->rlen);
     ssize_t n = recv(c->fd, c->rbuf.data() + c->rlen, want, 0);
     if (n == 0) {
       data_close_conn(S, c);
@@ -608,10 +657,11 @@ static void data_readable(DataServer* S, const std::shared_ptr<DataConn>& c) {
       std::memmove(c->rbuf.data(), c->rbuf.data() + off, c->rlen - off);
       c->rlen -= off;
     }
-    if (c->rlen == 0 && c->rbuf.capacity() > (1u << 20)) {
-      // a 16 MiB write frame would otherwise pin its buffer per conn
-      c->rbuf.clear();
-      c->rbuf.shrink_to_fit();
+    if (c->rlen == 0 && c->rbuf.capacity() > (8u << 20) && !c->rbuf.pinned) {
+      // a 16 MiB write frame would otherwise hold its buffer per conn
+      // (pinned buffers stay: re-pinning costs ~1 ms and an HBM write
+      // session is usually followed by more)
+      c->rbuf.release();
     }
     if (c->dead.load()) {
       data_close_conn(S, c);
@@ -790,6 +840,11 @@ static void data_write_register(int64_t sid, uint64_t conn_id, uint64_t req_id,
   ws->reserved = reserved;
   ws->fd = fd >= 0 ? dup(fd) : -1;
   ws->pos.store(pos);
+  if (kind == 0 && arena >= 0 && get_arena(arena)->is_dev()) {
+    // HBM destination: upgrade this conn's receive buffer to pinned
+    // pages at its next growth so dev_write DMAs straight from it
+    c->rbuf.want_pinned.store(true, std::memory_order_relaxed);
+  }
   std::lock_guard<std::mutex> g(c->smu);
   c->sess[req_id] = std::move(ws);
 }
