@@ -52,7 +52,9 @@ _ENUMS_BASED1 = {
     "StorageStateProto": ["CV", "UFS", "BOTH"],
     "BlockReportStatusProto": ["FINALIZED", "WRITING", "DELETED"],
 }
-_ENUMS0 = {"HeartbeatStatusProto": ["START", "RUNNING", "END"]}
+_ENUMS0 = {"HeartbeatStatusProto": ["START", "RUNNING", "END"],
+           # UNKNOWN is the zero value; consumers treat it as DIAGNOSE
+           "CompatibilityModeProto": ["UNKNOWN", "DIAGNOSE", "ENFORCE"]}
 
 _MESSAGES = {
     # ---- common.proto
@@ -91,6 +93,72 @@ _MESSAGES = {
         ("ip_addr", 3, "string", "required"),
         ("rpc_port", 4, "uint32", "required"),
         ("web_port", 5, "uint32", "required"),
+    ],
+    "StorageInfoProto": [
+        ("dir_id", 1, "uint32", "required"),
+        ("storage_id", 2, "string", "required"),
+        ("failed", 3, "bool", "required"),
+        ("capacity", 4, "int64", "required"),
+        ("available", 5, "int64", "required"),
+        ("fs_used", 6, "int64", "required"),
+        ("non_fs_used", 7, "int64", "required"),
+        ("reserved_bytes", 8, "int64", "required"),
+        ("storage_type", 9, "enum", "required", "StorageTypeProto"),
+        ("block_num", 10, "int64", "required"),
+        ("dir_path", 11, "string", "required"),
+    ],
+    "ComponentInfoProto": [
+        ("component", 1, "string", "optional"),
+        ("release_version", 2, "string", "optional"),
+        ("git_commit", 3, "string", "optional"),
+        ("git_tag", 4, "string", "optional"),
+        ("git_branch", 5, "string", "optional"),
+        ("protocol_version", 6, "uint32", "optional"),
+        ("min_protocol_version", 7, "uint32", "optional"),
+        ("capabilities", 8, "string", "repeated"),
+    ],
+    "ServerCompatibilityInfoProto": [
+        ("server", 1, "msg", "required", "ComponentInfoProto"),
+        ("min_worker_version", 2, "string", "optional"),
+        ("min_client_version", 3, "string", "optional"),
+        ("compatibility_mode", 4, "enum", "required",
+         "CompatibilityModeProto"),
+        ("blocked_versions", 5, "string", "repeated"),
+    ],
+    "WorkerInfoProto": [
+        ("address", 1, "msg", "required", "WorkerAddressProto"),
+        ("capacity", 2, "int64", "required"),
+        ("available", 3, "int64", "required"),
+        ("fs_used", 4, "int64", "required"),
+        ("non_fs_used", 5, "int64", "required"),
+        ("last_update", 6, "uint64", "required"),
+        ("storage_map", 7, "map_string_msg", "repeated",
+         "StorageInfoProto"),
+        ("reserved_bytes", 8, "int64", "required"),
+        ("component_info", 1000, "msg", "optional", "ComponentInfoProto"),
+    ],
+    "GetFilesystemInfoRequest": [
+        ("component_info", 1000, "msg", "optional", "ComponentInfoProto"),
+    ],
+    "GetFilesystemInfoResponse": [
+        ("active_master", 1, "string", "required"),
+        ("journal_nodes", 2, "string", "repeated"),
+        ("inode_dir_num", 3, "int64", "required"),
+        ("inode_file_num", 4, "int64", "required"),
+        ("block_num", 5, "int64", "required"),
+        ("capacity", 6, "int64", "required"),
+        ("available", 7, "int64", "required"),
+        ("fs_used", 8, "int64", "required"),
+        ("non_fs_used", 9, "int64", "required"),
+        ("reserved_bytes", 10, "int64", "required"),
+        ("live_workers", 11, "msg", "repeated", "WorkerInfoProto"),
+        ("blacklist_workers", 12, "msg", "repeated", "WorkerInfoProto"),
+        ("decommission_workers", 13, "msg", "repeated", "WorkerInfoProto"),
+        ("lost_workers", 14, "msg", "repeated", "WorkerInfoProto"),
+        ("allocatable_capacity", 15, "int64", "optional"),
+        ("allocatable_available", 16, "int64", "optional"),
+        ("compatibility", 1000, "msg", "optional",
+         "ServerCompatibilityInfoProto"),
     ],
     "FileAllocOptsProto": [
         ("truncate", 1, "bool", "required"),
@@ -311,7 +379,23 @@ def _build():
             fld.name = fname
             fld.number = num
             fld.label = _L[label]
-            if ftype == "map_string_bytes":
+            if ftype == "map_string_msg":
+                # map<string, Msg>: wire-identical to a repeated nested
+                # entry message with the map_entry option
+                entry = m.nested_type.add()
+                entry.name = _camel(fname) + "Entry"
+                entry.options.map_entry = True
+                k = entry.field.add()
+                k.name, k.number = "key", 1
+                k.label, k.type = F.LABEL_OPTIONAL, F.TYPE_STRING
+                v = entry.field.add()
+                v.name, v.number = "value", 2
+                v.label, v.type = F.LABEL_OPTIONAL, F.TYPE_MESSAGE
+                v.type_name = f".proto.{spec[4]}"
+                fld.label = F.LABEL_REPEATED
+                fld.type = F.TYPE_MESSAGE
+                fld.type_name = f".proto.{mname}.{entry.name}"
+            elif ftype == "map_string_bytes":
                 # map<string, bytes>: nested auto-generated entry message
                 entry = m.nested_type.add()
                 entry.name = _camel(fname) + "Entry"
@@ -623,6 +707,80 @@ def _bread_resp(h: dict, msg) -> None:
     msg.storage_type = 0
 
 
+def _ci_to_dict(ci) -> dict:
+    return {"component": ci.component,
+            "release_version": ci.release_version,
+            "protocol_version": ci.protocol_version or 1}
+
+
+def _ci_fill(msg, d: dict) -> None:
+    msg.component = d.get("component", "")
+    msg.release_version = d.get("release_version", "")
+    msg.protocol_version = d.get("protocol_version", 1)
+
+
+def _fsinfo_req(msg) -> dict:
+    out: dict = {}
+    if msg.HasField("component_info"):
+        out["component_info"] = _ci_to_dict(msg.component_info)
+    return out
+
+
+def _fsinfo_resp(h: dict, msg) -> None:
+    msg.active_master = h.get("cluster_id", "")
+    msg.inode_dir_num = 0
+    msg.inode_file_num = h.get("inode_num", 0)
+    msg.block_num = h.get("block_num", 0)
+    cap = h.get("capacity", 0)
+    used = h.get("used", 0)
+    msg.capacity = cap
+    msg.available = max(0, cap - used)
+    msg.fs_used = used
+    msg.non_fs_used = 0
+    msg.reserved_bytes = 0
+    msg.allocatable_capacity = cap
+    msg.allocatable_available = max(0, cap - used)
+    for w in h.get("live_workers", []):
+        wp = msg.live_workers.add()
+        a = w.get("address", {})
+        wp.address.worker_id = a.get("worker_id", 0)
+        wp.address.hostname = a.get("hostname", "")
+        wp.address.ip_addr = a.get("hostname", "")
+        wp.address.rpc_port = a.get("rpc_port", 0)
+        wp.address.web_port = 0
+        wcap = wused = 0
+        for st in w.get("storages", []):
+            sp = wp.storage_map[f"{st.get('tier', 'MEM')}-"
+                                f"{st.get('dir_id', 0)}"]
+            sp.dir_id = st.get("dir_id", 0)
+            sp.storage_id = f"{st.get('tier', 'MEM')}-{st.get('dir_id', 0)}"
+            sp.failed = False
+            sp.capacity = st.get("capacity", 0)
+            sp.fs_used = st.get("used", 0)
+            sp.available = max(0, sp.capacity - sp.fs_used)
+            sp.non_fs_used = 0
+            sp.reserved_bytes = 0
+            sp.storage_type = _TIER_TO_WIRE.get(st.get("tier", "MEM"), 0)
+            sp.block_num = st.get("block_num", 0)
+            sp.dir_path = ""
+            wcap += sp.capacity
+            wused += sp.fs_used
+        wp.capacity = wcap
+        wp.available = max(0, wcap - wused)
+        wp.fs_used = wused
+        wp.non_fs_used = 0
+        wp.reserved_bytes = 0
+        wp.last_update = w.get("last_heartbeat_ms", 0)
+        ci = w.get("component_info")
+        if ci:
+            _ci_fill(wp.component_info, ci)
+    ci = h.get("component_info")
+    if ci:
+        # advertise the master's compatibility contract (handshake)
+        _ci_fill(msg.compatibility.server, ci)
+        msg.compatibility.compatibility_mode = 1   # DIAGNOSE
+
+
 CODECS: dict[int, tuple] = {
     int(RpcCode.Mkdir): (M["MkdirRequest"], _mk_req,
                          M["MkdirResponse"], _mk_resp),
@@ -649,6 +807,9 @@ CODECS: dict[int, tuple] = {
         M["GetBlockLocationsResponse"], _block_locs_resp),
     int(RpcCode.Free): (M["FreeRequest"], _free_req,
                         M["FreeResponse"], _free_resp),
+    int(RpcCode.GetFilesystemInfo): (
+        M["GetFilesystemInfoRequest"], _fsinfo_req,
+        M["GetFilesystemInfoResponse"], _fsinfo_resp),
     int(RpcCode.WriteBlock): (M["BlockWriteRequest"], _bwrite_req,
                               M["BlockWriteResponse"], _bwrite_resp),
     int(RpcCode.ReadBlock): (M["BlockReadRequest"], _bread_req,

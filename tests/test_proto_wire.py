@@ -251,3 +251,45 @@ def test_protobuf_block_read_stream(tmp_path):
             await fs.close()
 
     asyncio.new_event_loop().run_until_complete(main())
+
+
+def test_fsinfo_handshake_pbuf_roundtrip():
+    """GetFilesystemInfo over the protobuf wire: a reference-shaped
+    client request with component_info=1000 decodes, and the response
+    carries workers + the master's compatibility contract."""
+    from curvine_amd.compat import component_info
+    from curvine_amd.rpc import proto
+    from curvine_amd.rpc.codes import RpcCode
+
+    req = proto.M["GetFilesystemInfoRequest"]()
+    proto._ci_fill(req.component_info, component_info("client"))
+    d = proto.decode_request(int(RpcCode.GetFilesystemInfo),
+                             req.SerializeToString())
+    assert d["component_info"]["component"] == "client"
+    assert d["component_info"]["protocol_version"] == 1
+    # legacy client (no component_info): empty dict, never None
+    assert proto.decode_request(
+        int(RpcCode.GetFilesystemInfo),
+        proto.M["GetFilesystemInfoRequest"]().SerializeToString()) == {}
+
+    hdr = {"cluster_id": "cv", "inode_num": 5, "block_num": 2,
+           "capacity": 100, "used": 10,
+           "component_info": component_info("master"),
+           "live_workers": [{
+               "address": {"worker_id": 7, "hostname": "h", "rpc_port": 1},
+               "storages": [{"tier": "HBM", "dir_id": 0, "capacity": 50,
+                             "used": 5, "block_num": 1}],
+               "last_heartbeat_ms": 123,
+               "component_info": component_info("worker")}]}
+    enc = proto.encode_response(int(RpcCode.GetFilesystemInfo), hdr)
+    m = proto.M["GetFilesystemInfoResponse"]()
+    m.ParseFromString(enc)
+    assert m.active_master == "cv"
+    assert m.capacity == 100 and m.available == 90
+    w = m.live_workers[0]
+    assert w.address.worker_id == 7
+    assert w.component_info.release_version
+    assert w.storage_map["HBM-0"].capacity == 50
+    # HBM rides the MEM storage type on the wire (documented deviation)
+    assert m.compatibility.server.component == "master"
+    assert m.compatibility.compatibility_mode == 1   # DIAGNOSE
