@@ -78,7 +78,10 @@ class BlockWriterRemote:
     sendmsg loop with a pipelined ack window, one executor hop per
     chunk); the asyncio stream remains for pwrite and as the fallback."""
 
-    WINDOW = 4
+    import os as _os
+    WINDOW = int(_os.environ.get("CURVINE_DW_WINDOW", "4"))
+    DW_CHUNK = int(_os.environ.get("CURVINE_DW_CHUNK", str(4 << 20)))
+    del _os
 
     def __init__(self, addr: WorkerAddress, block_id: int, reserve: int,
                  tier: str, reopen: bool = False):
@@ -129,7 +132,8 @@ class BlockWriterRemote:
             lib = _native_data_lib()
             loop = asyncio.get_running_loop()
             ok = await loop.run_in_executor(
-                None, lib.dw_write, self._dw, data, 0, len(data), 4 << 20)
+                None, lib.dw_write, self._dw, data, 0, len(data),
+                self.DW_CHUNK)
             if not ok:
                 lib.dw_abort(self._dw)
                 self._dw = None
