@@ -358,3 +358,48 @@ def test_async_flush_semantics(tmp_path):
     n = store.conn.execute("SELECT COUNT(*) FROM inodes").fetchone()[0]
     assert n == 55
     store.close()
+
+
+def test_page_out_skips_inflight_rows(tmp_path):
+    """An inode whose row is snapshot but not yet committed must not be
+    evicted: a fault-in would read the stale row."""
+    import threading
+
+    from curvine_amd.master.fs_dir import FsDir
+    from curvine_amd.master.inode_db import SqliteInodeStore
+    from curvine_amd.master.journal import JournalWriter
+
+    fs_dir = FsDir(JournalWriter(str(tmp_path / "j")))
+    store = SqliteInodeStore(str(tmp_path / "inodes.db"))
+    fs_dir.mirror = store
+    store._dirty.update(dict.keys(fs_dir.inodes))
+    for i in range(200):
+        fs_dir.create(f"/p/f{i}", 1 << 20, 1, "MEM")
+    store.flush(fs_dir, {}, fs_dir.journal.op_id)   # everything durable
+    store.enable_paging(fs_dir)
+
+    # dirty a batch again, gate the commit, snapshot
+    for i in range(50):
+        fs_dir.resolve(f"/p/f{i}").atime_ms += 1
+        store._dirty.add(fs_dir.resolve(f"/p/f{i}").id)
+    gate = threading.Event()
+    orig = store._commit
+
+    def slow(conn, snap):
+        gate.wait(10)
+        return orig(conn, snap)
+
+    store._commit = slow
+    assert store.flush_async(fs_dir, {}, fs_dir.journal.op_id)
+    inflight = set(store._inflight)
+    assert inflight, "snapshot should have taken the dirty batch"
+
+    evicted = store.page_out(fs_dir, set(), max_resident=20)
+    assert evicted > 0
+    for iid in inflight:
+        assert dict.__contains__(fs_dir.inodes, iid), \
+            "in-flight row evicted before its commit landed"
+    gate.set()
+    store.wait_flush()
+    store._commit = orig
+    store.close()
