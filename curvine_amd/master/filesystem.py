@@ -276,12 +276,19 @@ class MasterFilesystem:
 
     # ---------------- info ----------------
     def master_info(self) -> dict:
+        from curvine_amd.model import WorkerState
         live = self.workers.live_workers()
+        by_state: dict[int, list] = {}
+        for w in self.workers.workers.values():
+            by_state.setdefault(int(w.state), []).append(w.to_dict())
         return {
             "cluster_id": self.conf.cluster_id,
             "inode_num": len(self.fs_dir.inodes),
             "block_num": len(self.fs_dir.block_index),
             "live_workers": [w.to_dict() for w in live],
+            "decommission_workers":
+                by_state.get(int(WorkerState.DECOMMISSIONING), []),
+            "lost_workers": by_state.get(int(WorkerState.LOST), []),
             "capacity": self.workers.total_capacity(),
             "used": self.workers.total_used(),
             "block_size": self.conf.master.block_size,

@@ -740,45 +740,50 @@ def _fsinfo_resp(h: dict, msg) -> None:
     msg.reserved_bytes = 0
     msg.allocatable_capacity = cap
     msg.allocatable_available = max(0, cap - used)
-    for w in h.get("live_workers", []):
-        wp = msg.live_workers.add()
-        a = w.get("address", {})
-        wp.address.worker_id = a.get("worker_id", 0)
-        wp.address.hostname = a.get("hostname", "")
-        wp.address.ip_addr = a.get("hostname", "")
-        wp.address.rpc_port = a.get("rpc_port", 0)
-        wp.address.web_port = 0
-        wcap = wused = 0
-        for st in w.get("storages", []):
-            sp = wp.storage_map[f"{st.get('tier', 'MEM')}-"
-                                f"{st.get('dir_id', 0)}"]
-            sp.dir_id = st.get("dir_id", 0)
-            sp.storage_id = f"{st.get('tier', 'MEM')}-{st.get('dir_id', 0)}"
-            sp.failed = False
-            sp.capacity = st.get("capacity", 0)
-            sp.fs_used = st.get("used", 0)
-            sp.available = max(0, sp.capacity - sp.fs_used)
-            sp.non_fs_used = 0
-            sp.reserved_bytes = 0
-            sp.storage_type = _TIER_TO_WIRE.get(st.get("tier", "MEM"), 0)
-            sp.block_num = st.get("block_num", 0)
-            sp.dir_path = ""
-            wcap += sp.capacity
-            wused += sp.fs_used
-        wp.capacity = wcap
-        wp.available = max(0, wcap - wused)
-        wp.fs_used = wused
-        wp.non_fs_used = 0
-        wp.reserved_bytes = 0
-        wp.last_update = w.get("last_heartbeat_ms", 0)
-        ci = w.get("component_info")
-        if ci:
-            _ci_fill(wp.component_info, ci)
+    for field_name in ("live_workers", "decommission_workers",
+                       "lost_workers"):
+        for w in h.get(field_name, []):
+            _fill_worker(getattr(msg, field_name).add(), w)
     ci = h.get("component_info")
     if ci:
         # advertise the master's compatibility contract (handshake)
         _ci_fill(msg.compatibility.server, ci)
         msg.compatibility.compatibility_mode = 1   # DIAGNOSE
+
+
+def _fill_worker(wp, w: dict) -> None:
+    a = w.get("address", {})
+    wp.address.worker_id = a.get("worker_id", 0)
+    wp.address.hostname = a.get("hostname", "")
+    wp.address.ip_addr = a.get("hostname", "")
+    wp.address.rpc_port = a.get("rpc_port", 0)
+    wp.address.web_port = 0
+    wcap = wused = 0
+    for st in w.get("storages", []):
+        sp = wp.storage_map[f"{st.get('tier', 'MEM')}-"
+                            f"{st.get('dir_id', 0)}"]
+        sp.dir_id = st.get("dir_id", 0)
+        sp.storage_id = f"{st.get('tier', 'MEM')}-{st.get('dir_id', 0)}"
+        sp.failed = False
+        sp.capacity = st.get("capacity", 0)
+        sp.fs_used = st.get("used", 0)
+        sp.available = max(0, sp.capacity - sp.fs_used)
+        sp.non_fs_used = 0
+        sp.reserved_bytes = 0
+        sp.storage_type = _TIER_TO_WIRE.get(st.get("tier", "MEM"), 0)
+        sp.block_num = st.get("block_num", 0)
+        sp.dir_path = ""
+        wcap += sp.capacity
+        wused += sp.fs_used
+    wp.capacity = wcap
+    wp.available = max(0, wcap - wused)
+    wp.fs_used = wused
+    wp.non_fs_used = 0
+    wp.reserved_bytes = 0
+    wp.last_update = w.get("last_heartbeat_ms", 0)
+    ci = w.get("component_info")
+    if ci:
+        _ci_fill(wp.component_info, ci)
 
 
 CODECS: dict[int, tuple] = {
