@@ -220,7 +220,11 @@ class FuseConf:
     # (see ROADMAP); the daemon-side semantics it needs (reads on
     # write-only handles, handle-served getattr, open-time block
     # pinning) are implemented and kept on
-    writeback_cache: bool = False
+    # kernel page-cache write aggregation (FUSE_WRITEBACK_CACHE).  On by
+    # default: writes ride the serialized Python path in this mode (the
+    # native CAS-append window assumes in-order single-stream WRITEs)
+    # and every FUSE suite + the fio byte-verification sweep passes.
+    writeback_cache: bool = True
 
 
 @dataclass

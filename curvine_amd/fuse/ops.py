@@ -525,6 +525,12 @@ class CurvineFuseFs:
 
     def _register_native_write(self, h: FileHandle) -> None:
         w = h.writer
+        if self.fs.conf.fuse.writeback_cache:
+            # kernel writeback flushes pages concurrently and not always
+            # in order; the native CAS-append window assumes a single
+            # in-order WRITE stream, so keep writes on the serialized
+            # Python path in this mode
+            return
         if self.session is None or self.session.native_id is None or \
                 w is None or w._done or w._cur is None or \
                 w.refs > 1 or w._cur.meta.get("kind") != "arena":
@@ -666,12 +672,33 @@ class CurvineFuseFs:
             attrs["mtime_ms"] = int(time.time() * 1000) \
                 if valid & abi.FATTR_MTIME_NOW else mt * 1000 + mtn // 1_000_000
         self.invalidate(nodeid)
-        if valid & abi.FATTR_SIZE:
-            st = self._truncate(nodeid, path, size, fh if valid & abi.FATTR_FH else 0)
-        if attrs:
-            st = self.call(self.fs.set_attr(path, **attrs))
-        else:
-            st = self.stat_path(nodeid, path)
+        try:
+            if valid & abi.FATTR_SIZE:
+                st = self._truncate(nodeid, path, size,
+                                    fh if valid & abi.FATTR_FH else 0)
+            if attrs:
+                st = self.call(self.fs.set_attr(path, **attrs))
+            else:
+                st = self.stat_path(nodeid, path)
+        except (cverr.FileNotFound, OSError) as e:
+            if isinstance(e, OSError) and e.errno != errno.ENOENT:
+                raise
+            # SETATTR on an open-but-unlinked node: the kernel flushes
+            # timestamps through write_inode (writeback_cache) at close;
+            # failing it would poison the mapping and turn every later
+            # close() into EIO.  Serve from the handle like getattr does.
+            st = None
+            with self.handles_lock:
+                for h in self.handles.values():
+                    if h.node_id == nodeid and h.status is not None:
+                        st = h.status
+                        break
+            if st is None:
+                raise
+            if "mtime_ms" in attrs:
+                st.mtime_ms = attrs["mtime_ms"]
+            if "atime_ms" in attrs:
+                st.atime_ms = attrs["atime_ms"]
         st = self._adjust_writing_size(nodeid, st)
         self.cache_status(nodeid, st)
         return self.attr_out(nodeid, st)

@@ -22,6 +22,9 @@ from curvine_amd.fuse import abi
 
 log = logging.getLogger("curvine.fuse")
 
+# opt-in per-op trace for protocol debugging (CURVINE_FUSE_OP_TRACE=1)
+_OP_TRACE = bool(os.environ.get("CURVINE_FUSE_OP_TRACE"))
+
 libc = ctypes.CDLL("libc.so.6", use_errno=True)
 FUSE_DEV_IOC_CLONE = 0x8004E500
 MNT_DETACH = 2
@@ -129,6 +132,10 @@ class FuseChannel(threading.Thread):
         (length, opcode, unique, nodeid, uid, gid, pid, _extlen, _pad) = \
             abi.IN_HEADER.unpack_from(req, 0)
         body = memoryview(req)[abi.IN_HEADER_SIZE:length]
+        if _OP_TRACE:
+            with open("/tmp/fuse_op_trace.log", "a") as _tf:
+                _tf.write(f"{abi.Op.NAMES.get(opcode, opcode)} "
+                          f"node={nodeid} len={length}\n")
         handler = self.session.fs.HANDLERS.get(opcode)
         if handler is None:
             log.debug("fuse op %s unimplemented",
