@@ -24,6 +24,15 @@ class FsClient:
         # in-process worker colocated with this client (embedded mode)
         self.local_worker_id = -1
 
+    async def report_metrics(self, metrics: dict,
+                             kind: str = "client") -> None:
+        """Push a metrics snapshot to the master (MetricsReport code 60,
+        the reference's client metrics channel)."""
+        import os as _os
+        await self._rpc(RpcCode.MetricsReport, {
+            "client_id": f"{self.client_host}:{_os.getpid()}",
+            "kind": kind, "metrics": metrics})
+
     async def close(self) -> None:
         await self.connector.close()
 

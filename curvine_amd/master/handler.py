@@ -359,6 +359,17 @@ class MasterHandler:
         return {}
 
     def op_metricsreport(self, h, d):
+        """Clients push their metrics snapshots (MetricsReport, code 60)
+        for cluster-wide visibility; bounded per-client store exposed on
+        the master's /metrics and /api/client-metrics."""
+        cid = str(h.get("client_id", ""))[:128]
+        if cid:
+            store = self.master.client_metrics
+            store[cid] = {"ts_ms": int(time.time() * 1000),
+                          "kind": str(h.get("kind", "client"))[:32],
+                          "metrics": h.get("metrics") or {}}
+            while len(store) > 256:   # bounded: drop the oldest
+                store.pop(min(store, key=lambda k: store[k]["ts_ms"]))
         return {}
 
     def op_getmetadatasnapshotpage(self, h, d):

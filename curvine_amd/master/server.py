@@ -131,6 +131,8 @@ class Master:
                              conf.master.rpc_port, self.rpc_service)
         self.native_meta = None   # NativeMetaFrontend when enabled
         self.inode_db = None      # SqliteInodeStore when enabled
+        # client-pushed metrics snapshots (MetricsReport, code 60)
+        self.client_metrics: dict[str, dict] = {}
         from curvine_amd.compat import CompatibilityPolicy, PeerWarnDedup
         self.compat_policy = CompatibilityPolicy.from_conf(
             conf.compatibility)

@@ -180,6 +180,8 @@ class WebServer:
                 return self._json({jid: {k: j[k] for k in
                                          ("state", "total", "done", "failed")}
                                    for jid, j in self.master.jobs.jobs.items()})
+            if path == "/api/client-metrics" and self.master:
+                return self._json(getattr(self.master, "client_metrics", {}))
             if path == "/api/raft" and self.master and self.master.raft:
                 r = self.master.raft
                 return self._json({
