@@ -225,6 +225,9 @@ class FuseConf:
     # native CAS-append window assumes in-order single-stream WRITEs)
     # and every FUSE suite + the fio byte-verification sweep passes.
     writeback_cache: bool = True
+    # push per-op FUSE stats to the master every N seconds via
+    # MetricsReport (0 = off)
+    metrics_report_s: int = 30
 
 
 @dataclass

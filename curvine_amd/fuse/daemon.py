@@ -74,7 +74,27 @@ class FuseDaemon:
             native_loop=self.conf.fuse.native_loop).start(
                 session_fd=session_fd)
         self._start_control_server()
+        if self.conf.fuse.metrics_report_s > 0:
+            self._metrics_task = asyncio.run_coroutine_threadsafe(
+                self._metrics_report_loop(), self.loop)
         return self
+
+    async def _metrics_report_loop(self) -> None:
+        """Push the per-op FUSE stats to the master every interval
+        (MetricsReport code 60 — cluster-wide FUSE visibility)."""
+        interval = self.conf.fuse.metrics_report_s
+        while True:
+            await asyncio.sleep(interval)
+            try:
+                if self.session is not None:
+                    await self.fs.client.report_metrics(
+                        {"mnt": self.mnt_path,
+                         "fuse_op_stats": self.session.stats()},
+                        kind="fuse")
+            except asyncio.CancelledError:
+                return
+            except Exception:  # noqa: BLE001 — master briefly away
+                pass
 
     # ---------------- hot upgrade (fd + state handover) ----------------
     def _request_takeover(self) -> tuple[int, dict]:
