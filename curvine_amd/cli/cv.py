@@ -274,7 +274,8 @@ def cmd_transfers(fs, args):
     async def go():
         r = await fs.fs.client.connector.rpc(RpcCode.ListTransfers, {})
         return r.header
-    for j in fs.call(go()).get("transfers", fs.call(go()).get("jobs", [])):
+    h = fs.call(go())
+    for j in h.get("transfers") or h.get("jobs") or []:
         print(json.dumps(j))
 
 
