@@ -136,9 +136,15 @@ class RaftNode:
                  load_snapshot: Callable[[dict], None],
                  rebuild: Callable[[], None],
                  election_timeout_ms: int = 1500,
-                 heartbeat_ms: int = 300):
+                 heartbeat_ms: int = 300,
+                 learners: Optional[set] = None):
         self.id = node_id
         self.peers = {pid: addr for pid, addr in peers.items() if pid != node_id}
+        # LEARNERS (non-voting members): replicated to like any peer but
+        # excluded from elections and the commit quorum — a warm standby
+        # or read replica that never blocks or steals leadership
+        self.learners: set = set(learners or ())
+        self.is_learner = self.id in self.learners
         self.state = FOLLOWER
         self.term = 0
         self.voted_for: Optional[int] = None
@@ -246,7 +252,7 @@ class RaftNode:
                 now = time.monotonic()
                 if self.state == LEADER:
                     await self._broadcast_append()
-                elif now - self._last_heard > \
+                elif not self.is_learner and now - self._last_heard > \
                         self.election_timeout * (1 + random.random()):
                     await self._run_election()
             except asyncio.CancelledError:
@@ -315,7 +321,8 @@ class RaftNode:
             except Exception:  # noqa: BLE001
                 return None
 
-        results = await asyncio.gather(*[ask(p) for p in self.peers])
+        voting = [p for p in self.peers if p not in self.learners]
+        results = await asyncio.gather(*[ask(p) for p in voting])
         if self.term != term or self.state != CANDIDATE:
             return
         for r in results:
@@ -326,7 +333,7 @@ class RaftNode:
                 return
             if r.get("granted"):
                 votes += 1
-        if votes * 2 > len(self.peers) + 1:
+        if votes * 2 > len(voting) + 1:
             self._become_leader()
 
     def _become_leader(self):
@@ -411,9 +418,11 @@ class RaftNode:
             return
         await asyncio.gather(*[self._append_to(p) for p in self.peers],
                              return_exceptions=True)
-        # commit = median of match indexes (including self)
+        # commit = median of VOTING match indexes (including self);
+        # learners replicate but never count toward the quorum
+        voting = [p for p in self.peers if p not in self.learners]
         matches = sorted([self.log.last_index] +
-                         [self.match_index.get(p, 0) for p in self.peers],
+                         [self.match_index.get(p, 0) for p in voting],
                         reverse=True)
         majority = matches[len(matches) // 2]
         if majority > self.commit_index and \
