@@ -146,6 +146,10 @@ class JournalConf:
     rpc_port: int = 8996
     election_timeout_ms: int = 1500
     heartbeat_interval_ms: int = 300
+    # non-voting member ids: replicated to like any peer but excluded
+    # from elections and the commit quorum (raft learners — warm
+    # standbys / read replicas)
+    learners: list[int] = field(default_factory=list)
 
 
 @dataclass

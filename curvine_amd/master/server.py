@@ -167,7 +167,8 @@ class Master:
             load_snapshot=self._load_snapshot_state,
             rebuild=self._rebuild_state,
             election_timeout_ms=self.conf.journal.election_timeout_ms,
-            heartbeat_ms=self.conf.journal.heartbeat_interval_ms)
+            heartbeat_ms=self.conf.journal.heartbeat_interval_ms,
+            learners={int(x) for x in self.conf.journal.learners})
 
     def _apply_entry(self, e: dict) -> None:
         """Raft follower apply path (leader applied at append time).

@@ -354,3 +354,66 @@ def test_native_reads_commit_gated(loop, tmp_path):
             except Exception:  # noqa: BLE001 — follower[1] already stopped
                 pass
     run(loop, main())
+
+
+def test_learner_replicates_without_voting(loop, tmp_path):
+    """A learner (non-voting member) receives the replicated log but
+    never becomes leader, never counts toward the commit quorum, and a
+    2-voter+1-learner group keeps committing with the learner down."""
+    async def main():
+        import socket
+        ports = []
+        for _ in range(3):
+            s = socket.socket()
+            s.bind(("127.0.0.1", 0))
+            ports.append(s.getsockname()[1])
+            s.close()
+        peers = [f"{i + 1}@127.0.0.1:{p}" for i, p in enumerate(ports)]
+        masters = []
+        for i, p in enumerate(ports):
+            conf = ClusterConf()
+            conf.master.rpc_port = p
+            conf.journal.journal_dir = str(tmp_path / f"m{i + 1}")
+            conf.journal.peers = list(peers)
+            conf.journal.node_id = i + 1
+            conf.journal.learners = [3]      # node 3 is the learner
+            conf.journal.election_timeout_ms = 400
+            conf.journal.heartbeat_interval_ms = 100
+            masters.append(await Master(conf).start())
+        try:
+            for _ in range(60):
+                leaders = [m for m in masters if m.raft.is_leader]
+                if leaders:
+                    break
+                await asyncio.sleep(0.1)
+            assert leaders, "no leader elected"
+            leader = leaders[0]
+            assert leader.raft.id != 3, "a learner must never lead"
+
+            conn = ClusterConnector(
+                [f"127.0.0.1:{p}" for p in ports], 5000)
+            await conn.rpc(RpcCode.Mkdir, {"path": "/lrn"})
+            await conn.rpc(RpcCode.CreateFile, {"path": "/lrn/a"})
+            await conn.rpc(RpcCode.CompleteFile,
+                           {"path": "/lrn/a", "length": 0})
+            # the learner replicates the committed namespace
+            learner = masters[2]
+            for _ in range(50):
+                if learner.fs.fs_dir.resolve("/lrn/a") is not None:
+                    break
+                await asyncio.sleep(0.1)
+            assert learner.fs.fs_dir.resolve("/lrn/a") is not None
+
+            # learner down: the 2 voters still commit (quorum excludes it)
+            await learner.stop()
+            await conn.rpc(RpcCode.Mkdir, {"path": "/lrn2"})
+            r = await conn.rpc(RpcCode.Exists, {"path": "/lrn2"})
+            assert r.header["exists"]
+            await conn.close()
+        finally:
+            for m in masters:
+                try:
+                    await m.stop()
+                except Exception:  # noqa: BLE001
+                    pass
+    run(loop, main())
