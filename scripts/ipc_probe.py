@@ -15,6 +15,7 @@ def check(rc, what):
 
 if len(sys.argv) > 1 and sys.argv[1] == "child":
     h = bytes.fromhex(sys.argv[2])
+    check(libhip.hipSetDevice(0), "child setdev")
     ptr = ctypes.c_void_p()
     buf = ctypes.create_string_buffer(h, HANDLE_SZ)
     rc = libhip.hipIpcOpenMemHandle(ctypes.byref(ptr), buf,
