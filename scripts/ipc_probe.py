@@ -9,6 +9,12 @@ import sys
 libhip = ctypes.CDLL("libamdhip64.so")
 HANDLE_SZ = 64
 
+
+class IpcHandle(ctypes.Structure):
+    # hipIpcMemHandle_t: 64 reserved bytes, passed BY VALUE to open
+    _fields_ = [("reserved", ctypes.c_char * HANDLE_SZ)]
+
+
 def check(rc, what):
     if rc != 0:
         raise RuntimeError(f"{what} -> {rc}")
@@ -17,8 +23,11 @@ if len(sys.argv) > 1 and sys.argv[1] == "child":
     h = bytes.fromhex(sys.argv[2])
     check(libhip.hipSetDevice(0), "child setdev")
     ptr = ctypes.c_void_p()
-    buf = ctypes.create_string_buffer(h, HANDLE_SZ)
-    rc = libhip.hipIpcOpenMemHandle(ctypes.byref(ptr), buf,
+    hd = IpcHandle()
+    ctypes.memmove(hd.reserved, h, HANDLE_SZ)
+    libhip.hipIpcOpenMemHandle.argtypes = [
+        ctypes.POINTER(ctypes.c_void_p), IpcHandle, ctypes.c_uint]
+    rc = libhip.hipIpcOpenMemHandle(ctypes.byref(ptr), hd,
                                     ctypes.c_uint(1))  # lazy peer access
     check(rc, "hipIpcOpenMemHandle")
     out = ctypes.create_string_buffer(16)
