@@ -182,6 +182,9 @@ struct Arena {
   void* scratch = nullptr;
   size_t scratch_sz = 0;
   bool crc_tables_uploaded = false;
+  // true when base came from hipIpcOpenMemHandle (another process's
+  // arena): destroy closes the mapping instead of freeing
+  bool ipc_imported = false;
   std::mutex mu;
 
   bool is_dev() const { return device >= 0; }
@@ -262,7 +265,8 @@ static void arena_destroy(int h) {
     for (auto* p : a->pin) hipHostFree(p);
     for (auto e : a->ev) hipEventDestroy(e);
     if (a->scratch) hipFree(a->scratch);
-    hipFree(a->base);
+    if (a->ipc_imported) hipIpcCloseMemHandle(a->base);
+    else hipFree(a->base);
     hipStreamDestroy(a->stream);
     hipStreamDestroy(a->kstream);
     for (int i = 0; i < SPOOL; ++i)
@@ -273,6 +277,55 @@ static void arena_destroy(int h) {
     std::free(a->base);
   }
   delete a;
+}
+
+// ---- cross-process arena sharing (hipIpc, dmabuf mode) ----
+//
+// A worker exports its device arena once; a colocated client process
+// (FUSE daemon, another engine) opens the handle and reads published
+// extents with direct D2H DMA in ITS OWN process — the production
+// short-circuit for the one-process-per-GPU deployment shape.
+
+static py::bytes arena_ipc_handle(int h) {
+  Arena* a = get_arena(h);
+  if (!a->is_dev())
+    throw std::runtime_error("ipc export: device arenas only");
+  hipIpcMemHandle_t ih;
+  HIP_CHECK(hipSetDevice(a->device));
+  HIP_CHECK(hipIpcGetMemHandle(&ih, a->base));
+  return py::bytes((const char*)&ih, sizeof(ih));
+}
+
+static int arena_ipc_open(py::bytes handle, size_t cap, int device) {
+  std::string hb = handle;
+  if (hb.size() != sizeof(hipIpcMemHandle_t))
+    throw std::runtime_error("ipc open: bad handle size");
+  if (device < 0 || device >= device_count())
+    throw std::runtime_error("ipc open: bad device");
+  hipIpcMemHandle_t ih;
+  std::memcpy(&ih, hb.data(), sizeof(ih));
+  auto* a = new Arena();
+  a->device = device;
+  a->cap = cap;
+  a->ipc_imported = true;
+  HIP_CHECK(hipSetDevice(device));
+  hipError_t e = hipIpcOpenMemHandle(&a->base, ih,
+                                     hipIpcMemLazyEnablePeerAccess);
+  if (e != hipSuccess) {
+    delete a;
+    throw std::runtime_error(std::string("hipIpcOpenMemHandle: ") +
+                             hipGetErrorString(e));
+  }
+  HIP_CHECK(hipStreamCreateWithFlags(&a->stream, hipStreamNonBlocking));
+  HIP_CHECK(hipStreamCreateWithFlags(&a->kstream, hipStreamNonBlocking));
+  for (int i = 0; i < SPOOL; ++i)
+    HIP_CHECK(hipStreamCreateWithFlags(&a->spool[i], hipStreamNonBlocking));
+  // no staging ring: reads ride the pooled bounce pairs / pinned dests
+  std::lock_guard<std::mutex> g(g_arenas_mu);
+  for (size_t i = 0; i < g_arenas.size(); ++i)
+    if (!g_arenas[i]) { g_arenas[i] = a; return (int)i; }
+  g_arenas.push_back(a);
+  return (int)g_arenas.size() - 1;
 }
 
 static void ensure_scratch(Arena* a, size_t n) {

@@ -405,3 +405,149 @@ class BlockReaderHole:
 
     def close(self) -> None:
         pass
+
+
+# --------------------------------------------------------------------------
+# Cross-process HBM short-circuit (hipIpc)
+# --------------------------------------------------------------------------
+#
+# A colocated worker in ANOTHER process discloses its device arena via
+# ShortCircuitInfo (hipIpc handle + extent); the client maps the arena
+# once per (worker, arena) and serves reads with direct D2H DMA in its
+# own process — no socket in the byte path.  The worker pins the block
+# (store reader lease) while this reader lives, so delete/demote defer
+# exactly like the in-process short circuit.
+
+class _IpcArenaPool:
+    """(host, port, arena_handle) -> mapped native Arena, per process."""
+
+    def __init__(self):
+        import threading
+        self._mu = threading.Lock()
+        self._arenas: dict = {}
+
+    def get(self, key, ipc_handle: bytes, cap: int, device: int):
+        from curvine_amd import native
+        with self._mu:
+            a = self._arenas.get(key)
+            if a is None:
+                a = native.Arena.from_ipc(ipc_handle, cap, device)
+                self._arenas[key] = a
+            return a
+
+
+_IPC_POOL = _IpcArenaPool()
+
+_LOCAL_HOSTS = None
+
+
+def _is_local_host(hostname: str) -> bool:
+    global _LOCAL_HOSTS
+    if _LOCAL_HOSTS is None:
+        import socket
+        _LOCAL_HOSTS = {"127.0.0.1", "localhost", socket.gethostname()}
+    return hostname in _LOCAL_HOSTS
+
+
+class BlockReaderIpc:
+    """Store-reader-compatible view of a block living in ANOTHER
+    process's HBM arena (same interface as BlockStore.open_reader's
+    result: read_into/read_to_ptr/crc32c/meta/layout.arena/close)."""
+
+    def __init__(self, arena, info: dict, unpin):
+        from types import SimpleNamespace
+        self._arena = arena
+        self._off = info["offset"]
+        self.length = info["length"]
+        self.meta = {"kind": "arena", "offset": info["offset"],
+                     "length": info["length"]}
+        self.layout = SimpleNamespace(arena=arena)
+        self._unpin = unpin
+
+    def read_into(self, off: int, out, out_off: int, n: int) -> int:
+        n = min(n, self.length - off)
+        if n <= 0:
+            return 0
+        self._arena.read(self._off + off, out, out_off, n)
+        return n
+
+    def read_to_ptr(self, off: int, ptr: int, n: int, device: bool) -> int:
+        n = min(n, self.length - off)
+        if n <= 0:
+            return 0
+        self._arena.read_to_ptr(self._off + off, ptr, n, device)
+        return n
+
+    def crc32c(self, off: int, n: int) -> int:
+        return self._arena.crc32c(self._off + off, n)
+
+    async def read(self, off: int, n: int) -> bytes:
+        out = bytearray(min(n, self.length - off))
+        self.read_into(off, out, 0, len(out))
+        return bytes(out)
+
+    async def read_into_async(self, off: int, out, out_off: int, n: int):
+        return self.read_into(off, out, out_off, n)
+
+    def close(self) -> None:
+        u, self._unpin = self._unpin, None
+        if u is not None:
+            try:
+                u()
+            except Exception:  # noqa: BLE001
+                pass
+
+
+class AsyncIpcReader:
+    """Async adapter over BlockReaderIpc for the FsReader block-reader
+    interface (DMA runs on an executor thread)."""
+
+    def __init__(self, r: "BlockReaderIpc"):
+        self._r = r
+        self.length = r.length
+
+    async def read_into(self, off: int, out, out_off: int, n: int) -> int:
+        loop = asyncio.get_running_loop()
+        return await loop.run_in_executor(None, self._r.read_into,
+                                          off, out, out_off, n)
+
+    def close(self) -> None:
+        self._r.close()
+
+
+async def open_ipc_reader(client, addr: WorkerAddress,
+                          block_id: int):
+    """Map a colocated (same-host, other-process) worker's HBM block via
+    hipIpc, pinning it on the worker.  None when not applicable."""
+    from curvine_amd import native
+    if not _is_local_host(addr.hostname) or not native.gpu_available():
+        return None
+    try:
+        c = await factory().get(addr.hostname, addr.rpc_port)
+        r = await c.rpc(RpcCode.ShortCircuitInfo,
+                        {"block_id": block_id, "pin": True})
+        info = r.header.get("info", {})
+        if info.get("kind") != "arena" or not info.get("ipc"):
+            tok = info.get("pin_token")
+            if tok:
+                await c.rpc(RpcCode.UnpinBlock, {"token": tok})
+            return None
+        arena = _IPC_POOL.get((addr.hostname, addr.rpc_port,
+                               info["arena_handle"]),
+                              info["ipc"], info["cap"], info["device"])
+        token = info.get("pin_token")
+
+        def unpin():
+            async def go():
+                try:
+                    cc = await factory().get(addr.hostname, addr.rpc_port)
+                    await cc.rpc(RpcCode.UnpinBlock, {"token": token})
+                except Exception:  # noqa: BLE001
+                    pass
+            try:
+                asyncio.get_running_loop().create_task(go())
+            except RuntimeError:
+                pass   # no loop on this thread: the lease expires
+        return BlockReaderIpc(arena, info, unpin if token else None)
+    except Exception:  # noqa: BLE001 — fall back to the remote stream
+        return None

@@ -63,7 +63,7 @@ class SyncLocalReader:
     scheduling per op).  Remote blocks -> construction fails (caller keeps
     the async FsReader)."""
 
-    def __init__(self, file_blocks: FileBlocks):
+    def __init__(self, file_blocks: FileBlocks, ipc_resolver=None):
         from curvine_amd.worker import registry
         self.fb = file_blocks
         self.length = file_blocks.status.length
@@ -78,6 +78,9 @@ class SyncLocalReader:
                     if store is not None:
                         r = store.open_reader(lb.block.block_id)
                         break
+                if r is None and ipc_resolver is not None:
+                    # colocated other-process worker: hipIpc-mapped view
+                    r = ipc_resolver(lb)
                 if r is None:
                     raise err.BlockNotFound(
                         f"block {lb.block.block_id} has no in-process replica")
@@ -292,6 +295,10 @@ class FsReader:
     def __init__(self, client: FsClient, file_blocks: FileBlocks):
         self.client = client
         self.fb = file_blocks
+        try:
+            self._loop = __import__("asyncio").get_running_loop()
+        except RuntimeError:
+            self._loop = None
         self.status = file_blocks.status
         self.length = self.status.length
         self.pos = 0
@@ -342,6 +349,16 @@ class FsReader:
                         return BlockReaderLocal(store, lb.block.block_id)
                     except Exception:  # noqa: BLE001
                         self.failed_workers.add(addr.worker_id)
+        # cross-process HBM short-circuit: a colocated worker in another
+        # process discloses its arena via hipIpc — direct DMA reads here
+        if self.client.conf.client.short_circuit:
+            from curvine_amd.client.block_client import (AsyncIpcReader,
+                                                         open_ipc_reader)
+            for addr in candidates:
+                ipc = await open_ipc_reader(self.client, addr,
+                                            lb.block.block_id)
+                if ipc is not None:
+                    return AsyncIpcReader(ipc)
         last: Exception = err.BlockNotFound(str(lb.block.block_id))
         for addr in candidates:
             try:
@@ -485,8 +502,32 @@ class FsReader:
 
     def to_sync(self) -> SyncLocalReader:
         """Short-circuit sync view (raises if any block lacks a local
-        in-process replica)."""
-        return SyncLocalReader(self.fb)
+        in-process OR hipIpc-mapped replica)."""
+        return SyncLocalReader(self.fb, ipc_resolver=self._ipc_resolver)
+
+    def _ipc_resolver(self, lb):
+        """Sync bridge for SyncLocalReader: map a colocated
+        other-process HBM block via the client's event loop."""
+        import asyncio as _a
+        if not self.client.conf.client.short_circuit or self._loop is None:
+            return None
+        try:
+            if _a.get_running_loop() is self._loop:
+                return None   # would deadlock; caller stays async
+        except RuntimeError:
+            pass
+        from curvine_amd.client.block_client import open_ipc_reader
+        for addr in lb.locations:
+            try:
+                fut = _a.run_coroutine_threadsafe(
+                    open_ipc_reader(self.client, addr, lb.block.block_id),
+                    self._loop)
+                ipc = fut.result(30)
+            except Exception:  # noqa: BLE001
+                continue
+            if ipc is not None:
+                return ipc
+        return None
 
     def close(self) -> None:
         for r in self._readers.values():

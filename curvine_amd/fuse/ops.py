@@ -120,6 +120,19 @@ class SyncReadState:
                     return r
                 except Exception:  # noqa: BLE001
                     continue
+        # colocated worker in ANOTHER process (the production daemon
+        # shape): hipIpc-map its HBM arena for direct DMA reads — the
+        # same native read path as the in-process registry hit
+        try:
+            from curvine_amd.client.block_client import open_ipc_reader
+            for addr in lb.locations:
+                r = self.fs.call(open_ipc_reader(
+                    self.fs.fs.client, addr, lb.block.block_id), 30)
+                if r is not None:
+                    self._local[idx] = r
+                    return r
+        except Exception:  # noqa: BLE001 — fall back to the remote path
+            pass
         return None
 
     def read_into_px(self, off: int, ptr: int, view, n: int) -> int:

@@ -150,6 +150,22 @@ class Arena:
     def base_ptr(self) -> int:
         return self._n.arena_base_ptr(self.handle)
 
+    def ipc_handle(self) -> bytes:
+        """hipIpc export of a device arena (64-byte handle another
+        process opens with Arena.from_ipc)."""
+        return self._n.arena_ipc_handle(self.handle)
+
+    @classmethod
+    def from_ipc(cls, handle: bytes, capacity: int, device: int) -> "Arena":
+        """Map another process's device arena (hipIpcOpenMemHandle)."""
+        self = cls.__new__(cls)
+        self._n = load()
+        self.device = device
+        self.capacity = capacity
+        self.handle = self._n.arena_ipc_open(handle, capacity, device)
+        self._closed = False
+        return self
+
     def close(self) -> None:
         if not self._closed:
             self._n.arena_destroy(self.handle)

@@ -96,5 +96,6 @@ class RpcCode(IntEnum):
 
     # ---- curvine_amd additions ----
     ShortCircuitInfo = 100     # disclose local block location (path or HBM extent)
+    UnpinBlock = 101           # release a ShortCircuitInfo pin lease
     DeviceShortCircuit = 101   # hipIpc/dmabuf handle for cross-process GPU read
     RcclGroupSetup = 102       # establish an RCCL communicator for bulk distribution

@@ -75,7 +75,28 @@ class WorkerHandler:
         if code == int(RpcCode.WriteBlocksBatch):
             return await self._write_batch(msg)
         if code == int(RpcCode.ShortCircuitInfo):
-            return msg.reply({"info": self.store.local_info(msg.header["block_id"])})
+            h = msg.header
+            info = self.store.local_info(h["block_id"])
+            if h.get("pin"):
+                # hold a store reader so delete/demote defer while an
+                # out-of-process short-circuit reader streams the extent;
+                # leased — the heartbeat loop reaps expired pins
+                import time as _t
+                import uuid as _u
+                r = self.store.open_reader(h["block_id"])
+                token = _u.uuid4().hex
+                lease_s = min(3600, int(h.get("lease_ms", 300_000)) / 1000)
+                self.worker.pins[token] = (r, _t.monotonic() + lease_s)
+                info["pin_token"] = token
+            return msg.reply({"info": info})
+        if code == int(RpcCode.UnpinBlock):
+            ent = self.worker.pins.pop(msg.header.get("token", ""), None)
+            if ent is not None:
+                try:
+                    ent[0].close()
+                except Exception:  # noqa: BLE001
+                    pass
+            return msg.reply({})
         if code == int(RpcCode.Heartbeat):
             return msg.reply({"worker_id": self.worker.worker_id})
         raise err.Unsupported(f"worker rpc code {code}")

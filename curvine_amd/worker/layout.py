@@ -191,10 +191,22 @@ class ArenaLayout(BlockLayout):
         return self.arena.crc32c(meta["offset"] + off, n)
 
     def local_info(self, meta: dict) -> dict:
-        return {"kind": "arena", "tier": self.tier,
+        info = {"kind": "arena", "tier": self.tier,
                 "arena_handle": self.arena.handle,
                 "device": self.arena.device,
                 "offset": meta["offset"], "length": meta["length"]}
+        if self.arena.device >= 0:
+            # cross-process short-circuit: a colocated client opens the
+            # hipIpc handle and DMA-reads the extent in its own process
+            if not hasattr(self, "_ipc"):
+                try:
+                    self._ipc = self.arena.ipc_handle()
+                except Exception:  # noqa: BLE001 — pool without IPC
+                    self._ipc = None
+            if self._ipc:
+                info["ipc"] = self._ipc
+                info["cap"] = self.arena.capacity
+        return info
 
     def close(self) -> None:
         self.arena.close()

@@ -48,6 +48,8 @@ class Worker:
                              conf.worker.rpc_port, WorkerService(self))
         self._hb_task: Optional[asyncio.Task] = None
         self._master: Optional[RpcClient] = None
+        # ShortCircuitInfo pin leases: token -> (store reader, deadline)
+        self.pins: dict[str, tuple] = {}
         self._stopped = asyncio.Event()
         self._repl_sem = asyncio.Semaphore(conf.worker.replication_concurrency)
         self._load_sem = asyncio.Semaphore(conf.job.worker_task_concurrency)
@@ -121,6 +123,14 @@ class Worker:
                 await self._heartbeat_once()
                 tick += 1
                 self.store.reap_deferred()
+                now = __import__("time").monotonic()
+                for tok in [t for t, (_, dl) in self.pins.items()
+                            if dl < now]:
+                    r, _ = self.pins.pop(tok)
+                    try:
+                        r.close()
+                    except Exception:  # noqa: BLE001
+                        pass
                 if tick % 5 == 0:   # tier-pressure demotion sweep
                     loop = asyncio.get_event_loop()
                     await loop.run_in_executor(None,

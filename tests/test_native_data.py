@@ -274,3 +274,78 @@ def test_remote_read_write_native_hbm_tier(tmp_path):
                 wdata[1234567:1234567 + (4 << 20)]
             await fs.close()
     _run(main())
+
+
+@pytest.mark.gpu
+def test_ipc_cross_process_short_circuit(tmp_path):
+    """GPU: with the worker in its OWN process, a colocated client maps
+    the worker's HBM arena over hipIpc and reads with direct DMA — the
+    production short-circuit.  Pins defer deletion while reading."""
+    import subprocess
+    import sys
+
+    from curvine_amd import native
+    if not native.gpu_available():
+        pytest.skip("no GPU")
+
+    async def main():
+        from curvine_amd.client.block_client import (AsyncIpcReader,
+                                                     BlockReaderIpc)
+        from curvine_amd.client.filesystem import CurvineFileSystem
+        from curvine_amd.master.server import Master
+        from curvine_amd.testing import test_conf
+
+        conf = test_conf(str(tmp_path))
+        conf.client.storage_tier = "HBM"
+        m = await Master(conf).start()
+        wp = subprocess.Popen(
+            [sys.executable, "-m", "curvine_amd.server_main",
+             "--service", "worker", "--master-port", str(m.rpc.port),
+             "--worker-port", "0", "--heartbeat-ms", "300",
+             "--device", "0", "--data-dir", "[HBM:1GB:0]gpu0",
+             "--log-level", "WARNING"],
+            cwd=str(tmp_path.parent))
+        try:
+            conf.client.master_addrs = [f"127.0.0.1:{m.rpc.port}"]
+            fs = CurvineFileSystem(conf)
+            for _ in range(100):
+                info = await fs.client.get_master_info()
+                if info["live_workers"]:
+                    break
+                await asyncio.sleep(0.2)
+            assert info["live_workers"], "worker never registered"
+
+            data = os.urandom((12 << 20) + 777)
+            await fs.write_all("/ipc/a.bin", data, storage_tier="HBM")
+
+            r = await fs.open("/ipc/a.bin")
+            back = await r.read_all() if hasattr(r, "read_all") else \
+                await r.pread(0, r.length)
+            assert back == data
+            # the cached block readers must be the hipIpc path
+            kinds = {type(x).__name__ for x in r._readers.values()}
+            assert kinds == {"AsyncIpcReader"}, kinds
+
+            # sync short-circuit view + native registered batch reads
+            sr = r.to_sync()
+            buf = bytearray(1 << 20)
+            got = sr.pread_into(3 << 20, buf, 0, 1 << 20)
+            assert got == 1 << 20
+            assert bytes(buf) == data[3 << 20:4 << 20]
+
+            # delete while pinned: the extent defers; reads keep serving
+            await fs.delete("/ipc/a.bin")
+            await asyncio.sleep(0.8)    # heartbeat delivers the delete
+            got = sr.pread_into(5 << 20, buf, 0, 1 << 20)
+            assert got == 1 << 20 and bytes(buf) == data[5 << 20:6 << 20]
+            sr.close()
+            r.close()
+            await fs.close()
+        finally:
+            wp.terminate()
+            try:
+                wp.wait(timeout=10)
+            except subprocess.TimeoutExpired:
+                wp.kill()
+            await m.stop()
+    _run(main())
