@@ -304,7 +304,7 @@ def test_ipc_cross_process_short_circuit(tmp_path):
              "--worker-port", "0", "--heartbeat-ms", "300",
              "--device", "0", "--data-dir", "[HBM:1GB:0]gpu0",
              "--log-level", "WARNING"],
-            cwd=str(tmp_path.parent))
+            cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
         try:
             conf.client.master_addrs = [f"127.0.0.1:{m.rpc.port}"]
             fs = CurvineFileSystem(conf)
