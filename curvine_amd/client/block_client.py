@@ -549,5 +549,9 @@ async def open_ipc_reader(client, addr: WorkerAddress,
             except RuntimeError:
                 pass   # no loop on this thread: the lease expires
         return BlockReaderIpc(arena, info, unpin if token else None)
-    except Exception:  # noqa: BLE001 — fall back to the remote stream
+    except Exception as e:  # noqa: BLE001 — fall back to the remote stream
+        import logging
+        logging.getLogger("curvine.client").debug(
+            "ipc short-circuit unavailable for block %d at %s: %s",
+            block_id, addr.hostname, e)
         return None

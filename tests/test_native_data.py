@@ -318,6 +318,17 @@ def test_ipc_cross_process_short_circuit(tmp_path):
             data = os.urandom((12 << 20) + 777)
             await fs.write_all("/ipc/a.bin", data, storage_tier="HBM")
 
+            # surface the disclosure for debugging
+            from curvine_amd.client.block_client import factory
+            w0 = info["live_workers"][0]["address"]
+            c = await factory().get(w0["hostname"], w0["rpc_port"])
+            from curvine_amd.rpc.codes import RpcCode as _RC
+            fb = await fs.client.open("/ipc/a.bin")
+            sci = await c.rpc(_RC.ShortCircuitInfo,
+                              {"block_id": fb.blocks[0].block.block_id})
+            print("SCI:", {k: (len(v) if isinstance(v, bytes) else v)
+                           for k, v in sci.header.get("info", {}).items()})
+
             r = await fs.open("/ipc/a.bin")
             back = await r.read_all() if hasattr(r, "read_all") else \
                 await r.pread(0, r.length)
