@@ -201,11 +201,15 @@ class ArenaLayout(BlockLayout):
             if not hasattr(self, "_ipc"):
                 try:
                     self._ipc = self.arena.ipc_handle()
-                except Exception:  # noqa: BLE001 — pool without IPC
+                    self._ipc_err = None
+                except Exception as e:  # noqa: BLE001 — pool without IPC
                     self._ipc = None
+                    self._ipc_err = f"{type(e).__name__}: {e}"
             if self._ipc:
                 info["ipc"] = self._ipc
                 info["cap"] = self.arena.capacity
+            elif self._ipc_err:
+                info["ipc_error"] = self._ipc_err
         return info
 
     def close(self) -> None:
