@@ -1296,6 +1296,8 @@ PYBIND11_MODULE(_native, m) {
   m.def("device_sync", &device_sync);
   m.def("memcpy_h2d", &memcpy_h2d);
   m.def("device_mem_info", &device_mem_info);
+  m.def("arena_ipc_handle", &arena_ipc_handle);
+  m.def("arena_ipc_open", &arena_ipc_open);
   m.def("arena_create", &arena_create, py::arg("device"), py::arg("capacity"),
         py::arg("staging_bytes") = 4 << 20, py::arg("staging_count") = 8,
         py::arg("host_pinned") = false);
