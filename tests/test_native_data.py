@@ -338,7 +338,9 @@ def test_ipc_cross_process_short_circuit(tmp_path):
             assert kinds == {"AsyncIpcReader"}, kinds
 
             # sync short-circuit view + native registered batch reads
-            sr = r.to_sync()
+            # (constructed OFF the loop thread, as real sync callers are)
+            sr = await asyncio.get_running_loop().run_in_executor(
+                None, r.to_sync)
             buf = bytearray(1 << 20)
             got = sr.pread_into(3 << 20, buf, 0, 1 << 20)
             assert got == 1 << 20
