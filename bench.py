@@ -122,7 +122,7 @@ def setup(args, rank, world, dist, has_gpu):
         line = daemon_proc.stdout.readline()
         if not line.startswith("READY"):
             raise RuntimeError(f"cv-fuse failed to start: {line!r}")
-    elif args.no_short_circuit:
+    elif args.no_short_circuit or args.separate_worker:
         # faithful remote plane: the worker is its OWN process, every
         # byte crosses its streaming RPC over loopback
         hb = 250 if args.workload == "seqwrite" else 1000
@@ -285,12 +285,13 @@ def _sync_readers(args, rank, rt):
     return readers
 
 
-def step_client_seq_remote(args, rank, rt) -> int:
+def step_client_seq_remote(args, rank, rt, disable_sc: bool = True) -> int:
     """Sequential reads with short-circuit DISABLED: every byte crosses
     the worker's streaming RPC (the inter-node data plane, exercised
     over loopback — one concurrent stream per file)."""
     async def run():
-        rt.fs.client.conf.client.short_circuit = False
+        if disable_sc:
+            rt.fs.client.conf.client.short_circuit = False
         readers = [await rt.fs.open(f"/bench/r{rank}/f{i}")
                    for i in range(args.files)]
 
@@ -376,10 +377,16 @@ def step_client_rand4k(args, rank, rt, lat_out: list) -> int:
 
 def step_client_seq(args, rank, rt) -> int:
     """Sequential reads via the SYNC short-circuit path: OS threads with
-    pinned destination buffers (direct D2H DMA, no event loop per op)."""
+    pinned destination buffers (direct D2H DMA, no event loop per op).
+    Covers in-process registry hits AND hipIpc-mapped other-process
+    arenas; with neither available (CPU separate-worker), falls back to
+    the async streaming path without touching the short-circuit conf."""
     from curvine_amd import native
     use_pinned = native.gpu_available()
-    readers = _sync_readers(args, rank, rt)
+    try:
+        readers = _sync_readers(args, rank, rt)
+    except Exception:
+        return step_client_seq_remote(args, rank, rt, disable_sc=False)
     nthreads = min(args.threads, max(1, len(readers)))
     total = [0] * nthreads
     errs = []
@@ -604,6 +611,9 @@ def main():
                    help="queue depth per thread for randread4k (fio iodepth)")
     p.add_argument("--seq-batch", type=int, default=1,
                    help="chunks per sync in seqread (DMA pipelining depth)")
+    p.add_argument("--separate-worker", action="store_true",
+                   help="worker in its own process with short-circuit ON "
+                        "(hipIpc cross-process arena mapping)")
     p.add_argument("--no-short-circuit", action="store_true",
                    help="force the streaming worker-RPC read path")
     p.add_argument("--hbm-gb", type=int, default=32)
