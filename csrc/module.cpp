@@ -664,6 +664,7 @@ static void arena_fill(int h, uint64_t off, uint64_t n, int value) {
   }
   std::lock_guard<std::mutex> g(a->mu);
   HIP_CHECK(hipSetDevice(a->device));
+  (void)hipGetLastError();   // clear stale per-thread state
   hipLaunchKernelGGL(fill_kernel, dim3(grid_for(n / 16 + 1)), dim3(WG), 0,
                      a->kstream, (uint8_t*)a->base + off, (uint8_t)value, n);
   HIP_CHECK(hipGetLastError());
@@ -692,6 +693,7 @@ static uint32_t arena_crc32c(int h, uint64_t off, uint64_t n) {
     uint32_t n_wg = (uint32_t)((n_sub + WG - 1) / WG);
     ensure_scratch(a, n_wg * sizeof(uint32_t) + a->pin_sz);
     uint32_t* d_out = (uint32_t*)a->scratch;
+    (void)hipGetLastError();   // clear stale per-thread state
     hipLaunchKernelGGL(crc32c_kernel, dim3(n_wg), dim3(WG), 0, a->kstream, p,
                        n_sub, d_out);
     HIP_CHECK(hipGetLastError());
@@ -786,6 +788,7 @@ static void arena_gather(int h, std::vector<std::pair<uint64_t, uint64_t>> ext,
                            tile_off.size() * sizeof(uint64_t),
                            hipMemcpyHostToDevice, a->kstream));
   int grid = (int)std::min<size_t>(tile_ext.size(), 8192);
+  (void)hipGetLastError();   // clear stale per-thread state
   hipLaunchKernelGGL(copy_extents_kernel, dim3(grid), dim3(WG), 0, a->kstream,
                      (const uint8_t*)a->base, d_pack, d_ext, d_te, d_to,
                      (uint32_t)tile_ext.size());
@@ -860,6 +863,7 @@ static void arena_gather_ptr(
                            tile_off.size() * sizeof(uint64_t),
                            hipMemcpyHostToDevice, a->kstream));
   int grid = (int)std::min<size_t>(tile_ext.size(), 8192);
+  (void)hipGetLastError();   // clear stale per-thread state
   hipLaunchKernelGGL(copy_extents_kernel, dim3(grid), dim3(WG), 0, a->kstream,
                      (const uint8_t*)a->base, dst, d_ext, d_te, d_to,
                      (uint32_t)tile_ext.size());
@@ -1124,6 +1128,7 @@ static uint64_t arena_lz4_decompress(int ah, uint64_t dst_off, py::buffer buf) {
                            hipMemcpyHostToDevice, a->kstream));
   HIP_CHECK(hipMemsetAsync(d_err, 0, 4, a->kstream));
   int grid = (int)std::min<uint32_t>(h.n_chunks, 16384);
+  (void)hipGetLastError();   // clear stale per-thread state
   hipLaunchKernelGGL(lz4_decompress_kernel, dim3(grid), dim3(64), 0,
                      a->kstream, d_comp, d_off, d_len,
                      (uint8_t*)a->base + dst_off, h.chunk_size, h.raw_size,
